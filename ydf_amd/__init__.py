@@ -1,0 +1,56 @@
+"""ydf_amd — an MI355X-native decision-forests framework.
+
+A from-scratch reimplementation of the capability surface of Yggdrasil
+Decision Forests (google/yggdrasil-decision-forests) designed for AMD
+Instinct MI355X (gfx950/CDNA4): binned GPU-resident column store,
+hand-written HIP kernels for the training and serving hot paths, and
+RCCL-over-xGMI data-parallel training.
+
+Public API mirrors the `ydf` Python package
+(reference port/python/ydf/__init__.py:38-128).
+"""
+
+__version__ = "0.1.0"
+
+# Learners
+from ydf_amd.learner.generic_learner import GenericLearner
+from ydf_amd.learner.specialized_learners import (
+    CartLearner,
+    DecisionTreeLearner,
+    GradientBoostedTreesLearner,
+    IsolationForestLearner,
+    RandomForestLearner,
+)
+
+# Models
+from ydf_amd.model.generic_model import GenericModel, ModelIOOptions
+from ydf_amd.model.specialized import (
+    CARTModel,
+    DecisionForestModel,
+    GradientBoostedTreesModel,
+    IsolationForestModel,
+    RandomForestModel,
+)
+from ydf_amd.model.model_lib import (
+    deserialize_model,
+    load_model,
+    serialize_model,
+)
+from ydf_amd.model import tree
+
+# Dataset
+from ydf_amd.dataset.dataset import VerticalDataset, create_vertical_dataset
+from ydf_amd.dataset.dataspec import (
+    Column,
+    DataSpecification,
+    Semantic,
+    Task,
+)
+
+Feature = Column
+
+# Metric
+from ydf_amd.metric.metric import Evaluation, evaluate_predictions
+
+# Utilities
+from ydf_amd.utils.log import strict, verbose

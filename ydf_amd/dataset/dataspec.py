@@ -1,0 +1,170 @@
+"""Data specification: column semantics, vocabularies, binning boundaries.
+
+Capability analogue of the reference's DataSpecification proto +
+dataspec inference (yggdrasil_decision_forests/dataset/data_spec.proto:49,
+data_spec_inference.h:55), redesigned around a GPU-ready binned column
+store: every numerical feature carries quantile bin boundaries so the
+training path can materialize the uint8 binned matrix directly on device.
+"""
+from __future__ import annotations
+
+import dataclasses
+import enum
+from typing import Dict, List, Optional
+
+import numpy as np
+
+
+class Semantic(enum.Enum):
+    """Column semantic (reference data_spec.proto ColumnType)."""
+
+    NUMERICAL = 1
+    CATEGORICAL = 2
+    BOOLEAN = 3
+    HASH = 4
+    CATEGORICAL_SET = 5
+    DISCRETIZED_NUMERICAL = 6
+
+
+class Task(enum.Enum):
+    """Learning task (reference abstract_model.proto Task)."""
+
+    CLASSIFICATION = 1
+    REGRESSION = 2
+    RANKING = 3
+    CATEGORICAL_UPLIFT = 4
+    NUMERICAL_UPLIFT = 5
+    ANOMALY_DETECTION = 6
+
+
+@dataclasses.dataclass
+class Column:
+    """User-facing feature declaration (mirrors ydf.Column)."""
+
+    name: str
+    semantic: Optional[Semantic] = None
+
+
+# Out-of-vocabulary item: index 0 of every categorical vocabulary
+# (reference convention, data_spec.proto VocabValue).
+OOV_ITEM = "<OOD>"
+
+
+@dataclasses.dataclass
+class ColumnSpec:
+    """Inferred per-column specification."""
+
+    name: str
+    semantic: Semantic
+    # CATEGORICAL: vocabulary, index -> item (index 0 is OOV).
+    vocab: Optional[List[str]] = None
+    # NUMERICAL: statistics + quantile bin boundaries (ascending, <=255).
+    mean: float = 0.0
+    min_value: float = 0.0
+    max_value: float = 0.0
+    num_nas: int = 0
+    boundaries: Optional[np.ndarray] = None
+
+    @property
+    def vocab_size(self) -> int:
+        return len(self.vocab) if self.vocab is not None else 0
+
+    def to_json(self) -> dict:
+        d = {
+            "name": self.name,
+            "semantic": self.semantic.name,
+            "mean": float(self.mean),
+            "min_value": float(self.min_value),
+            "max_value": float(self.max_value),
+            "num_nas": int(self.num_nas),
+        }
+        if self.vocab is not None:
+            d["vocab"] = list(self.vocab)
+        if self.boundaries is not None:
+            d["boundaries"] = [float(v) for v in self.boundaries]
+        return d
+
+    @classmethod
+    def from_json(cls, d: dict) -> "ColumnSpec":
+        return cls(
+            name=d["name"],
+            semantic=Semantic[d["semantic"]],
+            vocab=d.get("vocab"),
+            mean=d.get("mean", 0.0),
+            min_value=d.get("min_value", 0.0),
+            max_value=d.get("max_value", 0.0),
+            num_nas=d.get("num_nas", 0),
+            boundaries=np.asarray(d["boundaries"], dtype=np.float32)
+            if "boundaries" in d
+            else None,
+        )
+
+
+@dataclasses.dataclass
+class DataSpecification:
+    """Full dataset schema: ordered columns + label column index."""
+
+    columns: List[ColumnSpec]
+    label: Optional[str] = None
+
+    def column(self, name: str) -> ColumnSpec:
+        for c in self.columns:
+            if c.name == name:
+                return c
+        raise KeyError(f"no column named {name!r}")
+
+    @property
+    def feature_columns(self) -> List[ColumnSpec]:
+        return [c for c in self.columns if c.name != self.label]
+
+    @property
+    def label_column(self) -> ColumnSpec:
+        assert self.label is not None
+        return self.column(self.label)
+
+    def to_json(self) -> dict:
+        return {
+            "columns": [c.to_json() for c in self.columns],
+            "label": self.label,
+        }
+
+    @classmethod
+    def from_json(cls, d: dict) -> "DataSpecification":
+        return cls(
+            columns=[ColumnSpec.from_json(c) for c in d["columns"]],
+            label=d.get("label"),
+        )
+
+
+def categorical_vocab(values: np.ndarray, max_vocab_count: int = 2000,
+                      min_vocab_frequency: int = 1) -> List[str]:
+    """Vocabulary sorted by descending frequency then item value.
+
+    Matches the reference's ordering (most frequent first, index 0 reserved
+    for out-of-vocabulary).
+    """
+    items, counts = np.unique(values.astype(str), return_counts=True)
+    order = np.lexsort((items, -counts))
+    vocab = [OOV_ITEM]
+    for idx in order[: max_vocab_count - 1]:
+        if counts[idx] >= min_vocab_frequency:
+            vocab.append(str(items[idx]))
+    return vocab
+
+
+def numerical_boundaries(values: np.ndarray, max_bins: int = 256,
+                         max_sample: int = 1_000_000) -> np.ndarray:
+    """Quantile bin boundaries (ascending, deduplicated, <= max_bins-1 cuts).
+
+    The GPU-ready analogue of the reference's DISCRETIZED_NUMERICAL /
+    dataset-cache binning (dataset_cache.h:15-58)."""
+    v = values[np.isfinite(values)]
+    if v.size == 0:
+        return np.zeros((0,), dtype=np.float32)
+    if v.size > max_sample:
+        rng = np.random.RandomState(1234)
+        v = v[rng.randint(0, v.size, max_sample)]
+    qs = np.linspace(0.0, 1.0, max_bins + 1)[1:-1]
+    cuts = np.quantile(v, qs).astype(np.float64)
+    cuts = np.unique(cuts)
+    return cuts.astype(np.float32)

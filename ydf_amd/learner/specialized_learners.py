@@ -1,0 +1,353 @@
+"""Concrete learners: GradientBoostedTrees, RandomForest, Cart,
+IsolationForest.
+
+API mirrors PYDF (port/python/ydf/learner/specialized_learners_pre_generated
+.py signatures; hyperparameter names + defaults from SURVEY.md Appendix A,
+i.e. the reference proto defaults). The training hot paths run through the
+HIP/gfx950 kernels in ydf_amd/ops on GPU or their C++ CPU twins.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional, Sequence, Union
+
+import numpy as np
+import torch
+
+from ydf_amd.dataset.dataset import VerticalDataset
+from ydf_amd.dataset.dataspec import Column, Task
+from ydf_amd.learner import trainer as trainer_lib
+from ydf_amd.learner.generic_learner import GenericLearner
+from ydf_amd.model.forest import FlatForest, build_flat_forest
+from ydf_amd.model.specialized import (GradientBoostedTreesModel,
+                                       IsolationForestModel,
+                                       RandomForestModel)
+from ydf_amd.utils.log import info
+
+
+class GradientBoostedTreesLearner(GenericLearner):
+    """GBT learner (reference learner/gradient_boosted_trees/; Python
+    surface of ydf.GradientBoostedTreesLearner).
+
+    Notes vs the reference: numerical features are always trained on the
+    256-bin quantile-discretized representation (the reference's
+    force_numerical_discretization path); split gain always uses the
+    second-order formulation (reference use_hessian_gain)."""
+
+    def __init__(self, label: str, task: Task = Task.CLASSIFICATION,
+                 features: Optional[Sequence[Union[str, Column]]] = None,
+                 num_trees: int = 300, max_depth: int = 6,
+                 shrinkage: float = 0.1, subsample: float = 1.0,
+                 min_examples: int = 5, l2_regularization: float = 0.0,
+                 min_sum_hessian_in_leaf: float = 1e-3,
+                 validation_ratio: float = 0.1,
+                 early_stopping: str = "LOSS_INCREASE",
+                 early_stopping_num_trees_look_ahead: int = 30,
+                 early_stopping_initial_iteration: int = 10,
+                 num_candidate_attributes_ratio: float = -1.0,
+                 use_hessian_gain: bool = True,
+                 apply_link_function: bool = True,
+                 random_seed: int = 123456, **kwargs):
+        super().__init__(label=label, task=task, features=features,
+                         random_seed=random_seed, **kwargs)
+        self.hyperparameters = dict(
+            num_trees=num_trees, max_depth=max_depth, shrinkage=shrinkage,
+            subsample=subsample, min_examples=min_examples,
+            l2_regularization=l2_regularization,
+            min_sum_hessian_in_leaf=min_sum_hessian_in_leaf,
+            validation_ratio=validation_ratio, early_stopping=early_stopping,
+            early_stopping_num_trees_look_ahead=(
+                early_stopping_num_trees_look_ahead),
+            early_stopping_initial_iteration=early_stopping_initial_iteration,
+            num_candidate_attributes_ratio=num_candidate_attributes_ratio,
+            use_hessian_gain=use_hessian_gain,
+            apply_link_function=apply_link_function,
+        )
+
+    def train(self, data, valid=None, verbose=None
+              ) -> GradientBoostedTreesModel:
+        hp = self.hyperparameters
+        device = self._resolve_device()
+        ds, bins, labels, bnd = self._prepare(data, device)
+        if labels is None:
+            raise ValueError(f"label column {self.label!r} missing")
+        classes = self._label_classes(ds)
+        n_classes = len(classes) if classes else 2
+        if self._task == Task.CLASSIFICATION:
+            loss = (trainer_lib.LOSS_MULTINOMIAL if n_classes > 2
+                    else trainer_lib.LOSS_BINOMIAL)
+        elif self._task == Task.REGRESSION:
+            loss = trainer_lib.LOSS_SQUARED_ERROR
+        else:
+            raise NotImplementedError(
+                f"GBT task {self._task} not yet supported")
+
+        # validation split (reference validation_set_ratio,
+        # gradient_boosted_trees.cc:1243)
+        valid_bins = valid_labels = None
+        vr = hp["validation_ratio"]
+        if valid is not None:
+            vds, valid_bins, valid_labels, _ = self._prepare_valid(
+                valid, ds, device)
+        elif vr > 0.0 and hp["early_stopping"] != "NONE":
+            N = bins.shape[1]
+            rng = np.random.RandomState(self.random_seed)
+            perm = rng.permutation(N)
+            n_valid = max(1, int(N * vr)) if N > 10 else 0
+            if n_valid:
+                vi = torch.from_numpy(perm[:n_valid].copy()).to(device)
+                ti = torch.from_numpy(perm[n_valid:].copy()).to(device)
+                valid_bins = bins[:, vi].contiguous()
+                valid_labels = labels[vi].contiguous()
+                bins = bins[:, ti].contiguous()
+                labels = labels[ti].contiguous()
+
+        F = bins.shape[0]
+        ncand = 0
+        if hp["num_candidate_attributes_ratio"] > 0:
+            ncand = max(1, int(round(hp["num_candidate_attributes_ratio"]
+                                     * F)))
+        cfg = trainer_lib.TrainerConfig(
+            loss=loss, num_trees=hp["num_trees"], max_depth=hp["max_depth"],
+            shrinkage=hp["shrinkage"], lambda_l2=hp["l2_regularization"],
+            min_examples=hp["min_examples"],
+            min_hessian=hp["min_sum_hessian_in_leaf"],
+            subsample=hp["subsample"], n_classes=n_classes,
+            seed=self.random_seed, num_candidate_features=ncand,
+            early_stopping=(hp["early_stopping"] != "NONE"
+                            and valid_bins is not None),
+            early_stopping_num_trees_look_ahead=(
+                hp["early_stopping_num_trees_look_ahead"]),
+            early_stopping_initial_iteration=(
+                hp["early_stopping_initial_iteration"]),
+        )
+        t = trainer_lib.ForestTrainer(bins, labels, cfg,
+                                      valid_bins=valid_bins,
+                                      valid_labels=valid_labels)
+        trees, init_preds, logs = trainer_lib.train_gbt(t, log=info)
+        flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"])
+        C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
+        activation = "identity"
+        if hp["apply_link_function"]:
+            if loss == trainer_lib.LOSS_BINOMIAL:
+                activation = "sigmoid"
+            elif loss == trainer_lib.LOSS_MULTINOMIAL:
+                activation = "softmax"
+        model = GradientBoostedTreesModel(
+            forest=flat, dataspec=ds.dataspec, task=self._task,
+            label_classes=classes, init_predictions=init_preds,
+            num_trees_per_iter=C, activation=activation)
+        model.training_logs = logs
+        return model
+
+    def _prepare_valid(self, valid, train_ds: VerticalDataset, device):
+        """Bins a user-provided validation dataset with the TRAIN dataspec."""
+        from ydf_amd.dataset.dataset import create_vertical_dataset
+
+        vds = create_vertical_dataset(valid, dataspec=train_ds.dataspec)
+        bnd = None
+        from ydf_amd.model.forest import padded_boundaries
+        from ydf_amd import ops
+
+        bnd = padded_boundaries(vds.dataspec.feature_columns)
+        X = torch.from_numpy(np.ascontiguousarray(vds.X)).to(device)
+        bnd_t = torch.from_numpy(bnd).to(device)
+        bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
+        ops.bin_data(X, bnd_t, bins)
+        labels = torch.from_numpy(
+            np.ascontiguousarray(vds.label_values)).to(device)
+        return vds, bins, labels, bnd
+
+
+class RandomForestLearner(GenericLearner):
+    """Random forest learner (reference learner/random_forest/
+    random_forest.cc:917 bagging loop).
+
+    Deviations (documented): bootstrap resampling is Poisson(1)-approximated
+    (same expectation as sampling-with-replacement); classification
+    aggregates per-tree leaf probabilities (winner_take_all=False
+    semantics), which is the reference's recommended setting for
+    calibrated probabilities."""
+
+    def __init__(self, label: Optional[str] = None,
+                 task: Task = Task.CLASSIFICATION,
+                 features: Optional[Sequence[Union[str, Column]]] = None,
+                 num_trees: int = 300, max_depth: int = 16,
+                 min_examples: int = 5,
+                 bootstrap_training_dataset: bool = True,
+                 bootstrap_size_ratio: float = 1.0,
+                 num_candidate_attributes: int = 0,
+                 num_candidate_attributes_ratio: float = -1.0,
+                 winner_take_all: bool = False,
+                 compute_oob_performances: bool = False,
+                 random_seed: int = 123456, **kwargs):
+        super().__init__(label=label, task=task, features=features,
+                         random_seed=random_seed, **kwargs)
+        self.hyperparameters = dict(
+            num_trees=num_trees, max_depth=max_depth,
+            min_examples=min_examples,
+            bootstrap_training_dataset=bootstrap_training_dataset,
+            bootstrap_size_ratio=bootstrap_size_ratio,
+            num_candidate_attributes=num_candidate_attributes,
+            num_candidate_attributes_ratio=num_candidate_attributes_ratio,
+            winner_take_all=winner_take_all,
+            compute_oob_performances=compute_oob_performances,
+        )
+
+    def _num_candidate(self, F: int) -> int:
+        hp = self.hyperparameters
+        if hp["num_candidate_attributes_ratio"] > 0:
+            return max(1, int(round(hp["num_candidate_attributes_ratio"]
+                                    * F)))
+        k = hp["num_candidate_attributes"]
+        if k > 0:
+            return min(k, F)
+        if k == -1:
+            return F
+        # k == 0: reference default — sqrt(F) classification, F/3 regression
+        if self._task == Task.CLASSIFICATION:
+            return max(1, int(math.sqrt(F) + 0.5))
+        return max(1, F // 3)
+
+    def train(self, data, valid=None, verbose=None) -> RandomForestModel:
+        hp = self.hyperparameters
+        device = self._resolve_device()
+        ds, bins, labels, bnd = self._prepare(data, device)
+        if labels is None:
+            raise ValueError(f"label column {self.label!r} missing")
+        classes = self._label_classes(ds) \
+            if self._task == Task.CLASSIFICATION else None
+        n_classes = len(classes) if classes else 2
+        F = bins.shape[0]
+        cfg = trainer_lib.TrainerConfig(
+            loss=trainer_lib.LOSS_RF, num_trees=hp["num_trees"],
+            max_depth=hp["max_depth"], shrinkage=1.0, lambda_l2=0.0,
+            min_examples=hp["min_examples"], min_hessian=0.0,
+            n_classes=n_classes, seed=self.random_seed,
+            bootstrap=hp["bootstrap_training_dataset"],
+            num_candidate_features=self._num_candidate(F),
+        )
+        t = trainer_lib.ForestTrainer(bins, labels, cfg)
+        trees = trainer_lib.train_rf(t, log=info)
+        flat = build_flat_forest(trees, bnd, leaf_scale=1.0)
+        C = n_classes if (classes and n_classes > 2) else 1
+        model = RandomForestModel(
+            forest=flat, dataspec=ds.dataspec, task=self._task,
+            label_classes=classes, init_predictions=[0.0] * max(C, 1),
+            num_trees_per_iter=C, activation="identity")
+        return model
+
+
+class CartLearner(RandomForestLearner):
+    """CART: a single tree, no bagging, all features as candidates
+    (reference learner/cart/cart.h:44; validation-set pruning TODO)."""
+
+    def __init__(self, label: Optional[str] = None,
+                 task: Task = Task.CLASSIFICATION,
+                 max_depth: int = 16, min_examples: int = 5,
+                 validation_ratio: float = 0.1, **kwargs):
+        kwargs.setdefault("num_trees", 1)
+        kwargs.setdefault("bootstrap_training_dataset", False)
+        kwargs.setdefault("num_candidate_attributes", -1)
+        super().__init__(label=label, task=task, max_depth=max_depth,
+                         min_examples=min_examples, **kwargs)
+        self.hyperparameters["validation_ratio"] = validation_ratio
+
+
+class IsolationForestLearner(GenericLearner):
+    """Isolation forest (reference learner/isolation_forest/
+    isolation_forest.h:40): subsampled trees with uniform random
+    axis-aligned splits; max depth = ceil(log2(subsample)) per
+    isolation_forest.cc:670. Trees are built host-side (tiny subsamples);
+    scoring runs through the batch inference kernels."""
+
+    def __init__(self, label: Optional[str] = None,
+                 task: Task = Task.ANOMALY_DETECTION,
+                 features: Optional[Sequence[Union[str, Column]]] = None,
+                 num_trees: int = 300, subsample_count: int = 256,
+                 subsample_ratio: Optional[float] = None,
+                 max_depth: int = -2, random_seed: int = 123456, **kwargs):
+        super().__init__(label=label, task=task, features=features,
+                         random_seed=random_seed, **kwargs)
+        self.hyperparameters = dict(
+            num_trees=num_trees, subsample_count=subsample_count,
+            subsample_ratio=subsample_ratio, max_depth=max_depth)
+
+    def train(self, data, valid=None, verbose=None) -> IsolationForestModel:
+        from ydf_amd.dataset.dataset import create_vertical_dataset
+
+        hp = self.hyperparameters
+        if isinstance(data, VerticalDataset):
+            ds = data
+        else:
+            ds = create_vertical_dataset(data, label=self.label,
+                                         task=self._task,
+                                         features=self.features)
+        X = ds.X  # [F, N]
+        F, N = X.shape
+        rng = np.random.RandomState(self.random_seed)
+        if hp["subsample_ratio"] is not None:
+            sub = max(2, int(N * hp["subsample_ratio"]))
+        else:
+            sub = min(hp["subsample_count"], N)
+        max_depth = hp["max_depth"]
+        if max_depth < 0:  # -2: reference default ceil(log2(subsample))
+            max_depth = max(1, int(math.ceil(math.log2(max(sub, 2)))))
+
+        c = IsolationForestModel.expected_path_length
+        feats, thrs, lefts, roots = [], [], [], []
+
+        def new_node() -> int:
+            feats.append(-1)
+            thrs.append(0.0)
+            lefts.append(0)
+            return len(feats) - 1
+
+        def build(root_rows: np.ndarray) -> None:
+            # children are allocated as adjacent pairs (flat layout needs
+            # right = left + 1)
+            root = new_node()
+            stack = [(root, root_rows, 0)]
+            while stack:
+                my, rows, depth = stack.pop()
+                n = len(rows)
+                split = None
+                if depth < max_depth and n > 1:
+                    # random feature with a non-constant range, uniform cut
+                    for _ in range(8):
+                        f = rng.randint(F)
+                        vals = X[f, rows]
+                        lo, hi = float(vals.min()), float(vals.max())
+                        if hi > lo:
+                            split = (f, float(rng.uniform(lo, hi)), vals)
+                            break
+                if split is None:
+                    thrs[my] = float(depth + c(n))
+                    continue
+                f, cut, vals = split
+                go_right = vals > cut
+                feats[my] = f
+                thrs[my] = cut
+                li = new_node()
+                ri = new_node()
+                lefts[my] = li
+                stack.append((li, rows[~go_right], depth + 1))
+                stack.append((ri, rows[go_right], depth + 1))
+
+        for _ in range(hp["num_trees"]):
+            rows = rng.choice(N, size=sub, replace=False)
+            roots.append(len(feats))
+            build(rows)
+
+        flat = FlatForest(feat=np.asarray(feats, np.int32),
+                          thr=np.asarray(thrs, np.float32),
+                          left=np.asarray(lefts, np.int32),
+                          roots=np.asarray(roots, np.int32))
+        return IsolationForestModel(
+            forest=flat, dataspec=ds.dataspec, task=self._task,
+            label_classes=None, init_predictions=[0.0],
+            num_trees_per_iter=1, activation="isolation",
+            num_examples_per_tree=sub)
+
+
+DecisionTreeLearner = CartLearner

@@ -1,0 +1,399 @@
+"""Level-wise histogram forest trainer (GBT / RF / CART cores).
+
+MI355X-native redesign of the reference training stack
+(learner/gradient_boosted_trees/gradient_boosted_trees.cc:1460 boosting loop,
+learner/decision_tree/training.cc:4739 DecisionTreeTrain,
+learner/random_forest/random_forest.cc:917 bagging): trees grow level-wise
+("open nodes" like the reference's distributed layer-wise growth,
+distributed_decision_tree/training.h:145) over GPU-resident binned columns.
+Per level the trainer builds {sum_g,sum_h,count} histograms with the
+hand-written HIP kernels, all-reduces them across data-parallel ranks (RCCL
+over xGMI — replacing the reference's gRPC MergeBestSplits protocol,
+distributed_gradient_boosted_trees.cc:1246), and selects splits redundantly
+on every rank, so no split/bitmap exchange is ever needed: rows never move.
+
+Works identically on CPU tensors (C++ ops; `gloo` all-reduce) for tests and
+on CUDA/ROCm device tensors (HIP kernels; RCCL).
+"""
+from __future__ import annotations
+
+import dataclasses
+import math
+from typing import List, Optional
+
+import numpy as np
+import torch
+
+from ydf_amd import ops
+
+LOSS_SQUARED_ERROR = 2
+LOSS_BINOMIAL = 1
+LOSS_MULTINOMIAL = 3
+LOSS_RF = 100  # weighted-target mode (RF/CART): not a GBT loss
+
+
+@dataclasses.dataclass
+class TrainerConfig:
+    loss: int = LOSS_BINOMIAL
+    num_trees: int = 300
+    max_depth: int = 6
+    shrinkage: float = 0.1
+    lambda_l2: float = 0.0
+    min_examples: int = 5
+    min_hessian: float = 1e-3
+    min_gain: float = 0.0
+    subsample: float = 1.0
+    n_classes: int = 2           # multinomial only
+    seed: int = 123456
+    # RF-specific
+    bootstrap: bool = False
+    num_candidate_features: int = 0  # 0 = all features
+    # early stopping (GBT; reference gradient_boosted_trees.proto:151-172)
+    early_stopping: bool = False
+    early_stopping_num_trees_look_ahead: int = 30
+    early_stopping_initial_iteration: int = 10
+    # device memory budget for the per-level histogram buffer
+    hist_budget_bytes: int = 1 << 31
+
+
+@dataclasses.dataclass
+class HostTree:
+    """One trained tree copied to host, complete-array form."""
+
+    feat: np.ndarray        # [total_nodes] i32, -1 = leaf/unused
+    bin: np.ndarray         # [total_nodes] i32 (split bin)
+    leaf_value: np.ndarray  # [total_nodes] f32 (unscaled -G/(H+l2))
+    counts: np.ndarray      # [total_nodes] f32
+    max_depth: int
+
+
+def _dist_ok() -> bool:
+    return torch.distributed.is_available() and \
+        torch.distributed.is_initialized() and \
+        torch.distributed.get_world_size() > 1
+
+
+class ForestTrainer:
+    """Grows histogram trees on one device (optionally data-parallel)."""
+
+    def __init__(self, bins: torch.Tensor, labels: torch.Tensor,
+                 cfg: TrainerConfig,
+                 valid_bins: Optional[torch.Tensor] = None,
+                 valid_labels: Optional[torch.Tensor] = None):
+        assert bins.dtype == torch.uint8 and bins.dim() == 2
+        self.bins = bins
+        self.labels = labels
+        self.cfg = cfg
+        self.device = bins.device
+        self.F, self.N = bins.shape
+        self.valid_bins = valid_bins
+        self.valid_labels = valid_labels
+        self.distributed = _dist_ok()
+
+        d = cfg.max_depth
+        self.total_nodes = (1 << (d + 1)) - 1
+        n_bins = ops.MAX_BINS
+        self.n_bins = n_bins
+        slot_bytes = self.F * n_bins * 3 * 4
+        self.max_slots = max(1, min(1 << d, cfg.hist_budget_bytes // slot_bytes))
+
+        dev = self.device
+        self.gh = torch.empty((self.N, 2), dtype=torch.float32, device=dev)
+        self.node_ids = torch.empty(self.N, dtype=torch.int32, device=dev)
+        self.hist = torch.empty((self.max_slots, self.F, n_bins, 3),
+                                dtype=torch.float32, device=dev)
+        self.node_stats = torch.zeros((self.total_nodes, 3),
+                                      dtype=torch.float32, device=dev)
+        self.leaf_vals = torch.empty(self.total_nodes, dtype=torch.float32,
+                                     device=dev)
+        self.tree_feat = torch.empty(self.total_nodes, dtype=torch.int32,
+                                     device=dev)
+        self.tree_bin = torch.empty(self.total_nodes, dtype=torch.int32,
+                                    device=dev)
+        self.bg_nf = torch.empty((self.max_slots, self.F),
+                                 dtype=torch.float32, device=dev)
+        self.bb_nf = torch.empty((self.max_slots, self.F), dtype=torch.int32,
+                                 device=dev)
+        self.best_feat = torch.empty(1 << d, dtype=torch.int32, device=dev)
+        self.best_bin = torch.empty(1 << d, dtype=torch.int32, device=dev)
+        self.best_gain = torch.empty(1 << d, dtype=torch.float32, device=dev)
+        # identity slot map reused when replaying a tree over other rows
+        self.arange_buf = torch.arange(1 << d, dtype=torch.int32, device=dev)
+        if valid_bins is not None:
+            self.valid_node_ids = torch.empty(valid_bins.shape[1],
+                                              dtype=torch.int32, device=dev)
+        self.rng = np.random.RandomState(cfg.seed)
+
+    # -- helpers ----------------------------------------------------------
+    def _allreduce(self, t: torch.Tensor):
+        if self.distributed:
+            torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.SUM)
+
+    def _feat_mask(self, n_active: int, tree_idx: int,
+                   level: int) -> Optional[torch.Tensor]:
+        k = self.cfg.num_candidate_features
+        if k <= 0 or k >= self.F:
+            return None
+        # Generated on host so every data-parallel rank draws the SAME mask
+        # (the reference samples candidate attributes per node too,
+        # decision_tree/training.cc num_candidate_attributes).
+        rs = np.random.RandomState(
+            (self.cfg.seed * 1000003 + tree_idx * 8191 + level) % (1 << 31))
+        mask = np.zeros((n_active, self.F), dtype=np.uint8)
+        for s in range(n_active):
+            idx = rs.choice(self.F, size=k, replace=False)
+            mask[s, idx] = 1
+        return torch.from_numpy(mask).to(self.device)
+
+    # -- one tree ---------------------------------------------------------
+    def grow_tree(self, tree_idx: int,
+                  sample_mask: Optional[torch.Tensor] = None) -> HostTree:
+        """Grows one tree from self.gh; returns it as host arrays.
+
+        On exit self.node_ids holds the final (sampled-rows) assignment and
+        self.leaf_vals the per-node values; callers apply update_preds.
+        """
+        cfg = self.cfg
+        self.tree_feat.fill_(-1)
+        self.tree_bin.zero_()
+        self.node_stats.zero_()
+        if sample_mask is None:
+            self.node_ids.zero_()
+        else:
+            # -1 parks out-of-sample rows outside every level
+            self.node_ids.copy_(
+                torch.where(sample_mask,
+                            torch.zeros((), dtype=torch.int32,
+                                        device=self.device),
+                            torch.full((), -1, dtype=torch.int32,
+                                       device=self.device)))
+
+        active_abs = np.array([0], dtype=np.int64)
+        for level in range(cfg.max_depth):
+            level_base = (1 << level) - 1
+            level_size = 1 << level
+            n_active = len(active_abs)
+            if n_active == 0:
+                break
+            active_abs_t = torch.from_numpy(
+                active_abs.astype(np.int32)).to(self.device)
+            slot_map = torch.full((level_size,), -1, dtype=torch.int32,
+                                  device=self.device)
+            slot_map[torch.from_numpy(active_abs - level_base).to(
+                self.device)] = self.arange_buf[:n_active]
+            feat_mask = self._feat_mask(n_active, tree_idx, level)
+
+            for s0 in range(0, n_active, self.max_slots):
+                ns = min(self.max_slots, n_active - s0)
+                hist_view = self.hist[:ns]
+                hist_view.zero_()
+                ops.hist_build(self.bins, self.gh, self.node_ids, slot_map,
+                               hist_view, level_base, level_size, s0, ns)
+                self._allreduce(hist_view)
+                ops.split_scan(hist_view, active_abs_t, self.node_stats,
+                               self.bg_nf, self.bb_nf, self.best_feat,
+                               self.best_bin, self.best_gain, s0, ns,
+                               cfg.lambda_l2, cfg.min_hessian,
+                               cfg.min_examples, cfg.min_gain,
+                               feat_mask=feat_mask)
+
+            # record the level's splits into the complete-tree arrays
+            idx64 = torch.from_numpy(active_abs).to(self.device)
+            self.tree_feat[idx64] = self.best_feat[:n_active]
+            self.tree_bin[idx64] = self.best_bin[:n_active]
+            ops.update_node_ids(self.bins, self.node_ids, slot_map,
+                                self.best_feat, self.best_bin, level_base,
+                                level_size)
+
+            if level + 1 < cfg.max_depth:
+                # choose next level's open nodes (host sync; deterministic
+                # across ranks because histograms were all-reduced)
+                bf = self.best_feat[:n_active].cpu().numpy()
+                split_abs = active_abs[bf >= 0]
+                if len(split_abs) == 0:
+                    active_abs = np.array([], dtype=np.int64)
+                    continue
+                children = np.concatenate([2 * split_abs + 1,
+                                           2 * split_abs + 2])
+                ns_view = self.node_stats.view(-1, 3)
+                ccounts = ns_view[torch.from_numpy(children).to(self.device),
+                                  2].cpu().numpy()
+                need = max(2 * cfg.min_examples, 2)
+                active_abs = np.sort(children[ccounts >= need])
+
+        ops.leaf_values(self.node_stats, self.leaf_vals, cfg.lambda_l2)
+        # .copy(): on CPU .cpu().numpy() aliases the (reused) buffers
+        return HostTree(
+            feat=self.tree_feat.cpu().numpy().copy(),
+            bin=self.tree_bin.cpu().numpy().copy(),
+            leaf_value=self.leaf_vals.cpu().numpy().copy(),
+            counts=self.node_stats[:, 2].cpu().numpy().copy(),
+            max_depth=cfg.max_depth,
+        )
+
+    def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor):
+        """Routes arbitrary rows through the latest tree (device arrays)."""
+        node_ids.zero_()
+        for level in range(self.cfg.max_depth):
+            level_base = (1 << level) - 1
+            level_size = 1 << level
+            ops.update_node_ids(
+                bins, node_ids, self.arange_buf[:level_size],
+                self.tree_feat[level_base:level_base + level_size],
+                self.tree_bin[level_base:level_base + level_size],
+                level_base, level_size)
+
+
+def train_gbt(trainer: ForestTrainer, log=None):
+    """The boosting loop (reference gradient_boosted_trees.cc:1460).
+
+    Returns (trees, init_preds, training_logs). For multinomial loss,
+    trees are interleaved per class: tree t belongs to class t % n_classes
+    (reference num_trees_per_iter semantics).
+    """
+    cfg = trainer.cfg
+    dev = trainer.device
+    N = trainer.N
+    y = trainer.labels
+    multi = cfg.loss == LOSS_MULTINOMIAL
+    C = cfg.n_classes if multi else 1
+
+    # initial predictions (reference loss->InitialPredictions,
+    # gradient_boosted_trees.cc:1329)
+    counts = torch.tensor([float(N)], device=dev)
+    if cfg.loss == LOSS_BINOMIAL:
+        s = torch.stack([y.sum(), counts[0]])
+        trainer._allreduce(s)
+        p = (s[0] / s[1]).clamp(1e-6, 1 - 1e-6)
+        init = float(torch.log(p / (1 - p)).item())
+        init_preds = [init]
+    elif cfg.loss == LOSS_SQUARED_ERROR:
+        s = torch.stack([y.sum(), counts[0]])
+        trainer._allreduce(s)
+        init = float((s[0] / s[1]).item())
+        init_preds = [init]
+    else:  # multinomial: zeros
+        init = 0.0
+        init_preds = [0.0] * C
+
+    preds = torch.full((C, N), 0.0, dtype=torch.float32, device=dev)
+    for c in range(C):
+        preds[c].fill_(init_preds[c])
+
+    has_valid = trainer.valid_bins is not None
+    if has_valid:
+        NV = trainer.valid_bins.shape[1]
+        valid_preds = torch.full((C, NV), 0.0, dtype=torch.float32,
+                                 device=dev)
+        for c in range(C):
+            valid_preds[c].fill_(init_preds[c])
+        loss_buf = torch.zeros(2, dtype=torch.float32, device=dev)
+
+    trees: List[HostTree] = []
+    logs = []
+    best_loss = math.inf
+    best_num_trees = 0
+    n_iters = cfg.num_trees
+    for it in range(n_iters):
+        sample_mask = None
+        if cfg.subsample < 1.0:
+            sample_mask = (
+                torch.from_numpy(
+                    trainer.rng.random_sample(N).astype(np.float32))
+                .to(dev) < cfg.subsample)
+        for c in range(C):
+            pc = preds[c]
+            if multi:
+                ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
+            else:
+                ops.grad_hess(pc, y, trainer.gh, cfg.loss)
+            tree = trainer.grow_tree(it * C + c, sample_mask)
+            trees.append(tree)
+            if sample_mask is not None:
+                trainer.route_rows(trainer.bins, trainer.node_ids)
+            ops.update_preds(pc, trainer.node_ids, trainer.leaf_vals,
+                             cfg.shrinkage)
+            if has_valid:
+                trainer.route_rows(trainer.valid_bins,
+                                   trainer.valid_node_ids)
+                ops.update_preds(valid_preds[c], trainer.valid_node_ids,
+                                 trainer.leaf_vals, cfg.shrinkage)
+        if has_valid:
+            vloss = _eval_loss(trainer, valid_preds, trainer.valid_labels,
+                               cfg, loss_buf)
+            logs.append({"iteration": it + 1, "valid_loss": vloss})
+            if vloss < best_loss:
+                best_loss = vloss
+                best_num_trees = (it + 1) * C
+            if (cfg.early_stopping
+                    and it + 1 >= cfg.early_stopping_initial_iteration
+                    and (it + 1) * C - best_num_trees >=
+                    cfg.early_stopping_num_trees_look_ahead * C):
+                if log:
+                    log(f"early stop at iteration {it + 1} "
+                        f"(best={best_num_trees // C})")
+                break
+    if has_valid and cfg.early_stopping and best_num_trees > 0:
+        trees = trees[:best_num_trees]
+    return trees, init_preds, logs
+
+
+def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
+    if cfg.loss == LOSS_BINOMIAL:
+        loss_buf.zero_()
+        ops.binary_logloss(preds[0], labels, loss_buf)
+        n = torch.tensor([float(labels.numel())], device=preds.device)
+        s = torch.cat([loss_buf[:1], n])
+        trainer._allreduce(s)
+        return float((s[0] / s[1]).item())
+    if cfg.loss == LOSS_SQUARED_ERROR:
+        s = torch.stack([((preds[0] - labels) ** 2).sum(),
+                         torch.tensor(float(labels.numel()),
+                                      device=preds.device)])
+        trainer._allreduce(s)
+        return float((s[0] / s[1]).item())
+    # multinomial cross-entropy
+    lse = torch.logsumexp(preds, dim=0)
+    idx = labels.long().clamp_(0, preds.shape[0] - 1)
+    picked = preds.gather(0, idx.view(1, -1)).view(-1)
+    s = torch.stack([(lse - picked).sum(),
+                     torch.tensor(float(labels.numel()),
+                                  device=preds.device)])
+    trainer._allreduce(s)
+    return float((s[0] / s[1]).item())
+
+
+def train_rf(trainer: ForestTrainer, log=None):
+    """Random-forest bagging loop (reference random_forest.cc:917).
+
+    Binary classification / regression: target mean leaves. Multi-class:
+    one tree per class per iteration (probability forest). Bootstrap is
+    Poisson(1)-approximated (documented deviation from the reference's exact
+    multinomial resampling; same expectation).
+    """
+    cfg = trainer.cfg
+    dev = trainer.device
+    N = trainer.N
+    multi = cfg.n_classes > 2 and cfg.loss == LOSS_RF
+    C = cfg.n_classes if multi else 1
+    trees: List[HostTree] = []
+    onehot = None
+    for it in range(cfg.num_trees):
+        weights = None
+        if cfg.bootstrap:
+            w = trainer.rng.poisson(
+                1.0, size=N).astype(np.float32)
+            weights = torch.from_numpy(w).to(dev)
+        for c in range(C):
+            if multi:
+                if onehot is None:
+                    onehot = torch.empty(N, dtype=torch.float32, device=dev)
+                onehot.copy_((trainer.labels == c).float())
+                ops.weighted_target(onehot, weights, trainer.gh)
+            else:
+                ops.weighted_target(trainer.labels, weights, trainer.gh)
+            tree = trainer.grow_tree(it * C + c)
+            trees.append(tree)
+        if log and (it + 1) % 100 == 0:
+            log(f"trained {it + 1}/{cfg.num_trees} trees")
+    return trees

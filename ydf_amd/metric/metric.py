@@ -1,0 +1,161 @@
+"""Evaluation metrics (capability analogue of the reference metric layer,
+yggdrasil_decision_forests/metric/metric.h:42-177: accuracy, AUC, PR-AUC,
+log-loss, RMSE/MAE, confusion matrix; bootstrap CIs to come)."""
+from __future__ import annotations
+
+import dataclasses
+from typing import Dict, Optional
+
+import numpy as np
+
+
+def accuracy(labels: np.ndarray, pred_classes: np.ndarray) -> float:
+    return float((labels == pred_classes).mean()) if len(labels) else 0.0
+
+
+def confusion_matrix(labels: np.ndarray, pred_classes: np.ndarray,
+                     n_classes: int) -> np.ndarray:
+    cm = np.zeros((n_classes, n_classes), dtype=np.int64)
+    np.add.at(cm, (labels.astype(np.int64), pred_classes.astype(np.int64)), 1)
+    return cm
+
+
+def roc_auc(labels: np.ndarray, scores: np.ndarray) -> float:
+    """Rank-based AUC (equivalent to the trapezoidal ROC integral used by
+    the reference, metric.h:150)."""
+    labels = np.asarray(labels, dtype=bool)
+    n_pos = int(labels.sum())
+    n_neg = len(labels) - n_pos
+    if n_pos == 0 or n_neg == 0:
+        return float("nan")
+    order = np.argsort(scores, kind="mergesort")
+    ranks = np.empty(len(scores), dtype=np.float64)
+    ranks[order] = np.arange(1, len(scores) + 1)
+    # average ranks for ties
+    sorted_scores = scores[order]
+    uniq, inv, cnt = np.unique(sorted_scores, return_inverse=True,
+                               return_counts=True)
+    cum = np.cumsum(cnt)
+    avg_rank = (cum - (cnt - 1) / 2.0)
+    ranks[order] = avg_rank[inv]
+    sum_pos = ranks[labels].sum()
+    return float((sum_pos - n_pos * (n_pos + 1) / 2.0) / (n_pos * n_neg))
+
+
+def pr_auc(labels: np.ndarray, scores: np.ndarray) -> float:
+    labels = np.asarray(labels, dtype=bool)
+    order = np.argsort(-scores, kind="mergesort")
+    tp = np.cumsum(labels[order])
+    fp = np.cumsum(~labels[order])
+    n_pos = int(labels.sum())
+    if n_pos == 0:
+        return float("nan")
+    precision = tp / np.maximum(tp + fp, 1)
+    recall = tp / n_pos
+    # step-wise integration
+    dr = np.diff(np.concatenate([[0.0], recall]))
+    return float((precision * dr).sum())
+
+
+def log_loss(labels: np.ndarray, probs: np.ndarray,
+             eps: float = 1e-12) -> float:
+    """labels: int class idx; probs: [N] (binary, P(class1)) or [N,C]."""
+    if probs.ndim == 1:
+        p = np.clip(probs, eps, 1 - eps)
+        y = labels.astype(np.float64)
+        return float(-(y * np.log(p) + (1 - y) * np.log(1 - p)).mean())
+    p = np.clip(probs[np.arange(len(labels)), labels.astype(int)], eps, 1.0)
+    return float(-np.log(p).mean())
+
+
+def rmse(labels: np.ndarray, preds: np.ndarray) -> float:
+    return float(np.sqrt(((labels - preds) ** 2).mean()))
+
+
+def mae(labels: np.ndarray, preds: np.ndarray) -> float:
+    return float(np.abs(labels - preds).mean())
+
+
+def ndcg(labels: np.ndarray, scores: np.ndarray, groups: np.ndarray,
+         truncation: int = 5) -> float:
+    """Mean NDCG@truncation over ranking groups (reference metric/ranking)."""
+    total, n_groups = 0.0, 0
+    for g in np.unique(groups):
+        m = groups == g
+        rel = labels[m]
+        sc = scores[m]
+        if len(rel) == 0:
+            continue
+        order = np.argsort(-sc, kind="mergesort")
+        gains = (2.0 ** rel[order][:truncation] - 1)
+        discounts = 1.0 / np.log2(np.arange(2, len(gains) + 2))
+        dcg = float((gains * discounts).sum())
+        iorder = np.argsort(-rel, kind="mergesort")
+        igains = (2.0 ** rel[iorder][:truncation] - 1)
+        idcg = float((igains * discounts[:len(igains)]).sum())
+        if idcg > 0:
+            total += dcg / idcg
+            n_groups += 1
+    return total / n_groups if n_groups else float("nan")
+
+
+@dataclasses.dataclass
+class Evaluation:
+    """Evaluation report (mirrors ydf.metric.Evaluation fields)."""
+
+    num_examples: int = 0
+    accuracy: Optional[float] = None
+    loss: Optional[float] = None
+    auc: Optional[float] = None
+    pr_auc: Optional[float] = None
+    rmse: Optional[float] = None
+    mae: Optional[float] = None
+    ndcg: Optional[float] = None
+    confusion: Optional[np.ndarray] = None
+
+    def to_dict(self) -> Dict:
+        d = {"num_examples": self.num_examples}
+        for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae", "ndcg"):
+            v = getattr(self, k)
+            if v is not None:
+                d[k] = v
+        return d
+
+    def __str__(self) -> str:
+        parts = [f"num examples: {self.num_examples}"]
+        for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae", "ndcg"):
+            v = getattr(self, k)
+            if v is not None:
+                parts.append(f"{k}: {v:.6g}")
+        return "\n".join(parts)
+
+    def _repr_html_(self) -> str:
+        rows = "".join(
+            f"<tr><td>{k}</td><td>{v:.6g}</td></tr>"
+            if isinstance(v, float) else f"<tr><td>{k}</td><td>{v}</td></tr>"
+            for k, v in self.to_dict().items())
+        return f"<table>{rows}</table>"
+
+
+def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
+                         task, n_classes: int = 2) -> Evaluation:
+    """Evaluates raw predictions (mirrors ydf.evaluate_predictions)."""
+    from ydf_amd.dataset.dataspec import Task
+
+    ev = Evaluation(num_examples=len(labels))
+    if task == Task.CLASSIFICATION:
+        if predictions.ndim == 1:
+            pred_cls = (predictions >= 0.5).astype(np.int64)
+            ev.auc = roc_auc(labels > 0.5, predictions)
+            ev.pr_auc = pr_auc(labels > 0.5, predictions)
+        else:
+            pred_cls = predictions.argmax(axis=1)
+        ev.accuracy = accuracy(labels.astype(np.int64), pred_cls)
+        ev.loss = log_loss(labels, predictions)
+        ev.confusion = confusion_matrix(labels.astype(np.int64), pred_cls,
+                                        n_classes)
+    elif task == Task.REGRESSION:
+        ev.rmse = rmse(labels, predictions)
+        ev.mae = mae(labels, predictions)
+        ev.loss = ev.rmse ** 2
+    return ev

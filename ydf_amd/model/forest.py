@@ -1,0 +1,117 @@
+"""Flat forest: the serving-side node-array representation.
+
+Capability analogue of the reference's flat-node serving models
+(serving/decision_forest/decision_forest_serving.h:94 GenericNode +
+FlatNodeModel :201), laid out SoA for the MI355X inference kernels:
+  feat[n]  i32  split feature (-1 = leaf)
+  thr[n]   f32  split threshold (x > thr -> right) or leaf value
+  left[n]  i32  left child index (right = left + 1)
+  roots[t] i32  root of tree t
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import List, Optional
+
+import numpy as np
+
+from ydf_amd.learner.trainer import HostTree
+
+
+@dataclasses.dataclass
+class FlatForest:
+    feat: np.ndarray   # i32 [total]
+    thr: np.ndarray    # f32 [total]
+    left: np.ndarray   # i32 [total]
+    roots: np.ndarray  # i32 [n_trees]
+
+    @property
+    def n_trees(self) -> int:
+        return len(self.roots)
+
+    @property
+    def n_nodes(self) -> int:
+        return len(self.feat)
+
+    def tree_slice(self, t: int):
+        lo = self.roots[t]
+        hi = self.roots[t + 1] if t + 1 < self.n_trees else self.n_nodes
+        return lo, hi
+
+
+def _reachable(feat: np.ndarray, max_depth: int):
+    total = feat.size
+    reach = np.zeros(total, dtype=bool)
+    internal = np.zeros(total, dtype=bool)
+    reach[0] = True
+    for level in range(max_depth):
+        base = (1 << level) - 1
+        size = 1 << level
+        lv = slice(base, base + size)
+        is_int = reach[lv] & (feat[lv] >= 0)
+        internal[lv] = is_int
+        idx = np.nonzero(is_int)[0] + base
+        if idx.size:
+            reach[2 * idx + 1] = True
+            reach[2 * idx + 2] = True
+    return reach, internal
+
+
+def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
+                      leaf_scale: float = 1.0):
+    """Compacts a complete-array HostTree into flat (feat, thr, left) arrays.
+
+    boundaries: padded [F, n_cuts] cut matrix; split threshold = cut[bin]
+    (binning guarantees "bin > b" <=> "x > cut[b]").
+    """
+    reach, internal = _reachable(tree.feat, tree.max_depth)
+    nodes = np.nonzero(reach)[0]  # ascending = level order
+    total = tree.feat.size
+    new_idx = np.full(total + 2, -1, dtype=np.int32)
+    new_idx[nodes] = np.arange(len(nodes), dtype=np.int32)
+    n_int = internal[nodes]
+    feat = np.where(n_int, tree.feat[nodes], -1).astype(np.int32)
+    lchild = np.minimum(2 * nodes + 1, total + 1)
+    left = np.where(n_int, new_idx[lchild], 0).astype(np.int32)
+    thr = np.empty(len(nodes), dtype=np.float32)
+    ii = np.nonzero(n_int)[0]
+    if ii.size:
+        thr[ii] = boundaries[tree.feat[nodes[ii]], tree.bin[nodes[ii]]]
+    li = np.nonzero(~n_int)[0]
+    thr[li] = tree.leaf_value[nodes[li]] * leaf_scale
+    return feat, thr, left
+
+
+def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
+                      leaf_scale: float = 1.0) -> FlatForest:
+    feats, thrs, lefts, roots = [], [], [], []
+    off = 0
+    for t in trees:
+        f, th, lf = host_tree_to_flat(t, boundaries, leaf_scale)
+        lf = np.where(f >= 0, lf + off, 0)
+        roots.append(off)
+        off += len(f)
+        feats.append(f)
+        thrs.append(th)
+        lefts.append(lf)
+    return FlatForest(
+        feat=np.concatenate(feats) if feats else np.zeros(0, np.int32),
+        thr=np.concatenate(thrs) if thrs else np.zeros(0, np.float32),
+        left=np.concatenate(lefts) if lefts else np.zeros(0, np.int32),
+        roots=np.asarray(roots, dtype=np.int32),
+    )
+
+
+def padded_boundaries(specs, max_bins: int = 256) -> np.ndarray:
+    """Stacks per-column ragged cut lists into a dense [F, n_cuts] matrix,
+    padding with +inf (padding adds no cut below any finite value, so bin
+    assignment is unchanged)."""
+    n_cuts = max(1, max((len(s.boundaries) if s.boundaries is not None else 0)
+                        for s in specs))
+    n_cuts = min(n_cuts, max_bins - 1)
+    out = np.full((len(specs), n_cuts), np.inf, dtype=np.float32)
+    for i, s in enumerate(specs):
+        b = s.boundaries
+        if b is not None and len(b):
+            out[i, :min(len(b), n_cuts)] = b[:n_cuts]
+    return out

@@ -1,0 +1,283 @@
+"""GenericModel: prediction, evaluation, persistence.
+
+Capability analogue of the reference AbstractModel
+(model/abstract_model.h:63) + the PYDF GenericModel surface
+(port/python/ydf/model/generic_model.py: predict:438, evaluate:552,
+save:358, describe:277, benchmark:320). Serving runs through the flat-forest
+HIP kernel on GPU (ydf_amd/serving) or its C++ CPU twin.
+"""
+from __future__ import annotations
+
+import dataclasses
+import json
+import os
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ydf_amd import ops
+from ydf_amd.dataset.dataset import (VerticalDataset, _to_column_dict,
+                                     encode_column)
+from ydf_amd.dataset.dataspec import DataSpecification, Semantic, Task
+from ydf_amd.metric.metric import Evaluation, evaluate_predictions
+from ydf_amd.model.forest import FlatForest
+
+
+@dataclasses.dataclass
+class ModelIOOptions:
+    file_prefix: Optional[str] = None
+
+
+def default_device() -> torch.device:
+    return torch.device("cuda") if torch.cuda.is_available() else \
+        torch.device("cpu")
+
+
+class _DeviceForest:
+    """Forest arrays resident on one device (upload once, reuse)."""
+
+    def __init__(self, forest: FlatForest, device: torch.device):
+        self.feat = torch.from_numpy(forest.feat).to(device)
+        self.thr = torch.from_numpy(forest.thr).to(device)
+        self.left = torch.from_numpy(forest.left).to(device)
+        self.roots = torch.from_numpy(forest.roots).to(device)
+
+
+class GenericModel:
+    """Base decision-forest model."""
+
+    _model_type = "GENERIC"
+
+    def __init__(self, forest: FlatForest, dataspec: DataSpecification,
+                 task: Task, label_classes: Optional[List[str]] = None,
+                 init_predictions: Optional[List[float]] = None,
+                 num_trees_per_iter: int = 1, activation: str = "identity",
+                 metadata: Optional[Dict] = None):
+        self.forest = forest
+        self.dataspec = dataspec
+        self._task = task
+        self.label_classes = label_classes
+        self.init_predictions = init_predictions or [0.0]
+        self.num_trees_per_iter = num_trees_per_iter
+        self.activation = activation
+        self.metadata = metadata or {}
+        self._dev_forest: Dict[str, _DeviceForest] = {}
+        self.training_logs = None
+
+    # ------------------------------------------------------------------
+    def task(self) -> Task:
+        return self._task
+
+    def name(self) -> str:
+        return self._model_type
+
+    def label(self) -> Optional[str]:
+        return self.dataspec.label
+
+    def label_col_idx(self) -> int:
+        for i, c in enumerate(self.dataspec.columns):
+            if c.name == self.dataspec.label:
+                return i
+        return -1
+
+    def input_feature_names(self) -> List[str]:
+        return [c.name for c in self.dataspec.feature_columns]
+
+    def num_trees(self) -> int:
+        return self.forest.n_trees
+
+    def num_nodes(self) -> int:
+        return self.forest.n_nodes
+
+    # ------------------------------------------------------------------
+    def _encode_features(self, data) -> np.ndarray:
+        """data -> feature-major [F,N] float32 matrix per the dataspec."""
+        if isinstance(data, VerticalDataset):
+            return data.X
+        cols = _to_column_dict(data)
+        specs = self.dataspec.feature_columns
+        n = len(next(iter(cols.values()))) if cols else 0
+        X = np.empty((len(specs), n), dtype=np.float32)
+        for i, spec in enumerate(specs):
+            if spec.name not in cols:
+                raise ValueError(f"missing input feature {spec.name!r}")
+            X[i] = encode_column(cols[spec.name], spec)
+        return X
+
+    def _forest_on(self, device: torch.device) -> _DeviceForest:
+        key = str(device)
+        if key not in self._dev_forest:
+            self._dev_forest[key] = _DeviceForest(self.forest, device)
+        return self._dev_forest[key]
+
+    def _n_outputs(self) -> int:
+        if (self._task == Task.CLASSIFICATION and self.label_classes
+                and len(self.label_classes) > 2):
+            return len(self.label_classes)
+        return 1
+
+    def _leaf_scale(self) -> float:
+        return 1.0
+
+    def predict_margin(self, X: torch.Tensor) -> torch.Tensor:
+        """Raw per-output forest sums/means. X [F,N] f32 on any device."""
+        df = self._forest_on(X.device)
+        C = self._n_outputs()
+        N = X.shape[1]
+        out = torch.empty((C, N), dtype=torch.float32, device=X.device)
+        T = self.forest.n_trees
+        per = T // C if C > 1 else T
+        for c in range(C):
+            ops.predict_forest(X, df.feat, df.thr, df.left, df.roots,
+                               out[c], tree_start=c, tree_step=C,
+                               n_trees=per if C > 1 else T,
+                               init=float(self.init_predictions[c]
+                                          if c < len(self.init_predictions)
+                                          else self.init_predictions[0]),
+                               scale=self._leaf_scale())
+        return out
+
+    def predict(self, data, device=None) -> np.ndarray:
+        """Predictions as numpy: binary classification -> P(class_2) [N];
+        multi-class -> [N, C]; regression/anomaly -> [N]. (Mirrors
+        ydf GenericModel.predict semantics.)"""
+        dev = torch.device(device) if device is not None else default_device()
+        X_np = self._encode_features(data)
+        X = torch.from_numpy(np.ascontiguousarray(X_np)).to(dev)
+        m = self.predict_margin(X)
+        out = self._apply_activation(m)
+        return out.cpu().numpy()
+
+    def _apply_activation(self, m: torch.Tensor) -> torch.Tensor:
+        if self.activation == "sigmoid":
+            out = torch.empty_like(m[0])
+            ops.sigmoid(m[0], out)
+            return out
+        if self.activation == "softmax":
+            return torch.softmax(m, dim=0).T.contiguous()
+        if m.shape[0] == 1:
+            return m[0]
+        return m.T.contiguous()
+
+    # ------------------------------------------------------------------
+    def evaluate(self, data, device=None) -> Evaluation:
+        cols = _to_column_dict(data) if not isinstance(data, VerticalDataset) \
+            else None
+        preds = self.predict(data, device=device)
+        if isinstance(data, VerticalDataset):
+            labels = data.label_values
+        else:
+            lname = self.dataspec.label
+            if lname is None or lname not in cols:
+                raise ValueError("dataset has no label column")
+            lspec = self.dataspec.label_column
+            if lspec.semantic == Semantic.CATEGORICAL:
+                lookup = {item: i for i, item in enumerate(lspec.vocab)}
+                labels = np.fromiter(
+                    (lookup.get(s, 0) - 1 for s in cols[lname].astype(str)),
+                    dtype=np.float32, count=len(cols[lname]))
+            else:
+                labels = np.asarray(cols[lname], dtype=np.float32)
+        n_classes = len(self.label_classes) if self.label_classes else 2
+        return evaluate_predictions(preds, labels, self._task, n_classes)
+
+    # ------------------------------------------------------------------
+    def benchmark(self, data, benchmark_duration: float = 3.0,
+                  warmup_duration: float = 0.5, batch_size: int = 0,
+                  device=None):
+        """Inference throughput benchmark (mirrors ydf model.benchmark;
+        reference cli/benchmark_inference.cc)."""
+        dev = torch.device(device) if device is not None else default_device()
+        X_np = self._encode_features(data)
+        X = torch.from_numpy(np.ascontiguousarray(X_np)).to(dev)
+        n = X.shape[1]
+
+        def run_once():
+            m = self.predict_margin(X)
+            self._apply_activation(m)
+
+        t_end = time.perf_counter() + warmup_duration
+        while time.perf_counter() < t_end:
+            run_once()
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        runs = 0
+        t0 = time.perf_counter()
+        t_end = t0 + benchmark_duration
+        while time.perf_counter() < t_end:
+            run_once()
+            runs += 1
+        if dev.type == "cuda":
+            torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        return BenchmarkResult(
+            examples_per_second=n * runs / dt,
+            num_runs=runs,
+            duration_seconds=dt,
+            batch_size=n,
+        )
+
+    # ------------------------------------------------------------------
+    def describe(self, output_format: str = "text") -> str:
+        lines = [
+            f'type: "{self._model_type}"',
+            f"task: {self._task.name}",
+            f"label: {self.dataspec.label!r}",
+            f"input features ({len(self.input_feature_names())}): "
+            + ", ".join(self.input_feature_names()[:40]),
+            f"trees: {self.num_trees()}",
+            f"nodes: {self.num_nodes()}",
+        ]
+        if self.label_classes:
+            lines.append(f"classes: {self.label_classes}")
+        return "\n".join(lines)
+
+    def __str__(self) -> str:
+        return self.describe()
+
+    # ------------------------------------------------------------------
+    def _header(self) -> dict:
+        return {
+            "framework": "ydf_amd",
+            "version": 1,
+            "model_type": self._model_type,
+            "task": self._task.name,
+            "label_classes": self.label_classes,
+            "init_predictions": [float(v) for v in self.init_predictions],
+            "num_trees_per_iter": self.num_trees_per_iter,
+            "activation": self.activation,
+            "metadata": self.metadata,
+        }
+
+    def _load_extra(self, header: dict) -> None:
+        pass
+
+    def save(self, path: str, advanced_options: ModelIOOptions = None) -> None:
+        """Model directory: header.json + dataspec.json + forest.npz + done
+        (structural analogue of the reference model dir,
+        model/model_library.cc:92-107: header.pb / data_spec.pb / nodes-* /
+        done)."""
+        os.makedirs(path, exist_ok=True)
+        with open(os.path.join(path, "header.json"), "w") as f:
+            json.dump(self._header(), f, indent=1)
+        with open(os.path.join(path, "dataspec.json"), "w") as f:
+            json.dump(self.dataspec.to_json(), f, indent=1)
+        np.savez(os.path.join(path, "forest.npz"), feat=self.forest.feat,
+                 thr=self.forest.thr, left=self.forest.left,
+                 roots=self.forest.roots)
+        with open(os.path.join(path, "done"), "w") as f:
+            f.write("")
+
+
+@dataclasses.dataclass
+class BenchmarkResult:
+    examples_per_second: float
+    num_runs: int
+    duration_seconds: float
+    batch_size: int
+
+    def __str__(self) -> str:
+        return (f"{self.examples_per_second:,.0f} examples/s "
+                f"({self.num_runs} runs over {self.duration_seconds:.2f}s)")

@@ -1,0 +1,64 @@
+"""Model persistence (reference model/model_library.h:44-58 SaveModel /
+LoadModel registry)."""
+from __future__ import annotations
+
+import json
+import os
+
+import numpy as np
+
+from ydf_amd.dataset.dataspec import DataSpecification, Task
+from ydf_amd.model.forest import FlatForest
+from ydf_amd.model.generic_model import GenericModel
+
+
+def load_model(path: str) -> GenericModel:
+    from ydf_amd.model.specialized import MODEL_CLASSES
+
+    with open(os.path.join(path, "header.json")) as f:
+        header = json.load(f)
+    with open(os.path.join(path, "dataspec.json")) as f:
+        dataspec = DataSpecification.from_json(json.load(f))
+    z = np.load(os.path.join(path, "forest.npz"))
+    forest = FlatForest(feat=z["feat"], thr=z["thr"], left=z["left"],
+                        roots=z["roots"])
+    cls = MODEL_CLASSES.get(header["model_type"], GenericModel)
+    model = cls(
+        forest=forest,
+        dataspec=dataspec,
+        task=Task[header["task"]],
+        label_classes=header.get("label_classes"),
+        init_predictions=header.get("init_predictions", [0.0]),
+        num_trees_per_iter=header.get("num_trees_per_iter", 1),
+        activation=header.get("activation", "identity"),
+        metadata=header.get("metadata"),
+    )
+    model._load_extra(header)
+    return model
+
+
+def serialize_model(model: GenericModel) -> bytes:
+    """In-memory serialization (mirrors ydf model.serialize())."""
+    import io
+    import zipfile
+
+    buf = io.BytesIO()
+    with zipfile.ZipFile(buf, "w") as zf:
+        zf.writestr("header.json", json.dumps(model._header()))
+        zf.writestr("dataspec.json", json.dumps(model.dataspec.to_json()))
+        fbuf = io.BytesIO()
+        np.savez(fbuf, feat=model.forest.feat, thr=model.forest.thr,
+                 left=model.forest.left, roots=model.forest.roots)
+        zf.writestr("forest.npz", fbuf.getvalue())
+    return buf.getvalue()
+
+
+def deserialize_model(data: bytes) -> GenericModel:
+    import io
+    import tempfile
+    import zipfile
+
+    with tempfile.TemporaryDirectory() as td:
+        with zipfile.ZipFile(io.BytesIO(data)) as zf:
+            zf.extractall(td)
+        return load_model(td)

@@ -1,0 +1,101 @@
+"""Model families: GBT / RF / Isolation Forest / CART.
+
+Capability analogues of the reference model classes
+(model/gradient_boosted_trees/gradient_boosted_trees.h:57,
+model/random_forest/random_forest.h:52,
+model/isolation_forest/isolation_forest.h:51).
+"""
+from __future__ import annotations
+
+import math
+
+import numpy as np
+import torch
+
+from ydf_amd.dataset.dataspec import Task
+from ydf_amd.model.generic_model import GenericModel
+
+
+class DecisionForestModel(GenericModel):
+    """Shared base for tree-ensemble models (mirrors
+    ydf.DecisionForestModel)."""
+
+    def get_tree(self, idx: int):
+        from ydf_amd.model import tree as tree_lib
+
+        return tree_lib.extract_tree(self.forest, idx)
+
+    def get_all_trees(self):
+        return [self.get_tree(i) for i in range(self.forest.n_trees)]
+
+    def print_tree(self, idx: int = 0, max_depth: int = 6) -> str:
+        from ydf_amd.model import tree as tree_lib
+
+        return tree_lib.format_tree(
+            self.get_tree(idx), self.dataspec, max_depth=max_depth)
+
+
+class GradientBoostedTreesModel(DecisionForestModel):
+    _model_type = "GRADIENT_BOOSTED_TREES"
+
+    def initial_predictions(self) -> np.ndarray:
+        return np.asarray(self.init_predictions, dtype=np.float32)
+
+    def validation_loss(self):
+        if self.training_logs:
+            return self.training_logs[-1].get("valid_loss")
+        return None
+
+
+class RandomForestModel(DecisionForestModel):
+    _model_type = "RANDOM_FOREST"
+
+    def _leaf_scale(self) -> float:
+        C = self._n_outputs()
+        per = self.forest.n_trees // C if C > 1 else self.forest.n_trees
+        return 1.0 / max(per, 1)
+
+
+# A CART model is a random forest with a single tree (ydf convention).
+CARTModel = RandomForestModel
+
+
+class IsolationForestModel(DecisionForestModel):
+    """Anomaly score = 2^(-E[path length] / c(n)) (reference
+    model/isolation_forest/isolation_forest.h:51; leaf values store
+    depth + c(n_leaf))."""
+
+    _model_type = "ISOLATION_FOREST"
+
+    def __init__(self, *args, num_examples_per_tree: int = 256, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.num_examples_per_tree = num_examples_per_tree
+
+    @staticmethod
+    def expected_path_length(n: float) -> float:
+        if n <= 1:
+            return 0.0
+        h = math.log(n - 1) + 0.5772156649
+        return 2.0 * h - 2.0 * (n - 1) / n
+
+    def _leaf_scale(self) -> float:
+        return 1.0 / max(self.forest.n_trees, 1)
+
+    def _apply_activation(self, m: torch.Tensor) -> torch.Tensor:
+        denom = self.expected_path_length(float(self.num_examples_per_tree))
+        return torch.exp2(-m[0] / denom)
+
+    def _header(self) -> dict:
+        h = super()._header()
+        h["num_examples_per_tree"] = self.num_examples_per_tree
+        return h
+
+    def _load_extra(self, header: dict) -> None:
+        self.num_examples_per_tree = header.get("num_examples_per_tree", 256)
+
+
+MODEL_CLASSES = {
+    c._model_type: c
+    for c in (GenericModel, GradientBoostedTreesModel, RandomForestModel,
+              IsolationForestModel)
+}

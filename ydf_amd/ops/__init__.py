@@ -1,0 +1,205 @@
+"""Torch-tensor front end for the MI355X-native decision-forest ops.
+
+Dispatches each op to the hand-written HIP/gfx950 kernel (device tensors) or
+the C++ CPU implementation (host tensors). The HIP path is mandatory on GPU:
+if the extension is missing while a CUDA device is visible, import fails
+loudly instead of silently falling back.
+"""
+from __future__ import annotations
+
+import torch
+
+try:
+    from ydf_amd import _ydf_ops as _C
+except ImportError as e:  # pragma: no cover
+    raise ImportError(
+        "ydf_amd._ydf_ops native extension not built. Run "
+        "`python tools/build_ext.py` (hipcc, gfx950). Refusing to run "
+        "without the native kernels."
+    ) from e
+
+MAX_BINS = _C.max_bins
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _chk(t: torch.Tensor, dtype, name: str) -> None:
+    assert t.dtype == dtype, f"{name}: expected {dtype}, got {t.dtype}"
+    assert t.is_contiguous(), f"{name}: must be contiguous"
+
+
+def bin_data(x: torch.Tensor, boundaries: torch.Tensor,
+             out: torch.Tensor) -> torch.Tensor:
+    """x [F,N] f32, boundaries [F,n_cuts] f32 (ascending) -> out [F,N] u8."""
+    F, N = x.shape
+    n_cuts = boundaries.shape[1]
+    _chk(x, torch.float32, "x")
+    _chk(boundaries, torch.float32, "boundaries")
+    _chk(out, torch.uint8, "out")
+    if x.is_cuda:
+        _C.gpu_bin_data(x.data_ptr(), boundaries.data_ptr(), out.data_ptr(),
+                        N, F, n_cuts, _stream())
+    else:
+        _C.cpu_bin_data(x.data_ptr(), boundaries.data_ptr(), out.data_ptr(),
+                        N, F, n_cuts)
+    return out
+
+
+def grad_hess(preds: torch.Tensor, labels: torch.Tensor, gh: torch.Tensor,
+              loss: int) -> torch.Tensor:
+    """preds/labels [N] f32 -> gh [N,2] f32 interleaved {g,h}."""
+    N = preds.numel()
+    if preds.is_cuda:
+        _C.gpu_grad_hess(preds.data_ptr(), labels.data_ptr(), gh.data_ptr(),
+                         N, loss, _stream())
+    else:
+        _C.cpu_grad_hess(preds.data_ptr(), labels.data_ptr(), gh.data_ptr(),
+                         N, loss)
+    return gh
+
+
+def grad_hess_softmax(preds: torch.Tensor, labels: torch.Tensor,
+                      gh: torch.Tensor, n_classes: int,
+                      cls: int) -> torch.Tensor:
+    """preds [C,N] f32 (logits), labels [N] f32 (class idx) -> gh for `cls`."""
+    N = labels.numel()
+    if preds.is_cuda:
+        _C.gpu_grad_hess_softmax(preds.data_ptr(), labels.data_ptr(),
+                                 gh.data_ptr(), N, n_classes, cls, _stream())
+    else:
+        _C.cpu_grad_hess_softmax(preds.data_ptr(), labels.data_ptr(),
+                                 gh.data_ptr(), N, n_classes, cls)
+    return gh
+
+
+def weighted_target(labels: torch.Tensor, weights, gh: torch.Tensor):
+    """RF/CART target: g=-w*y, h=w. weights may be None (unit)."""
+    N = labels.numel()
+    wp = weights.data_ptr() if weights is not None else 0
+    if labels.is_cuda:
+        _C.gpu_weighted_target(labels.data_ptr(), wp, gh.data_ptr(), N,
+                               _stream())
+    else:
+        _C.cpu_weighted_target(labels.data_ptr(), wp, gh.data_ptr(), N)
+    return gh
+
+
+def hist_build(bins: torch.Tensor, gh: torch.Tensor, node_ids: torch.Tensor,
+               slot_map: torch.Tensor, hist: torch.Tensor, level_base: int,
+               level_size: int, slot0: int, n_slots: int):
+    """bins [F,N] u8; hist [n_slots,F,n_bins,3] f32 (pre-zeroed, base=slot0).
+
+    slot_map [level_size] i32 maps level-relative node -> slot (-1 closed)."""
+    F, N = bins.shape
+    n_bins = hist.shape[2]
+    if bins.is_cuda:
+        _C.gpu_hist_build(bins.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
+                          slot_map.data_ptr(), hist.data_ptr(), N, F, n_bins,
+                          level_base, level_size, slot0, n_slots, _stream())
+    else:
+        _C.cpu_hist_build(bins.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
+                          slot_map.data_ptr(), hist.data_ptr(), N, F, n_bins,
+                          level_base, level_size, slot0, n_slots)
+    return hist
+
+
+def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
+               node_stats: torch.Tensor, best_gain_nf: torch.Tensor,
+               best_bin_nf: torch.Tensor, best_feat: torch.Tensor,
+               best_bin: torch.Tensor, best_gain: torch.Tensor, slot0: int,
+               n_slots: int, lambda_l2: float, min_hessian: float,
+               min_examples: int, min_gain: float, feat_mask=None):
+    F = hist.shape[1]
+    n_bins = hist.shape[2]
+    mp = feat_mask.data_ptr() if feat_mask is not None else 0
+    args = (hist.data_ptr(), abs_of_slot.data_ptr(), node_stats.data_ptr(),
+            best_gain_nf.data_ptr(), best_bin_nf.data_ptr(),
+            best_feat.data_ptr(), best_bin.data_ptr(), best_gain.data_ptr(),
+            mp, F, n_bins, slot0, n_slots, lambda_l2, min_hessian,
+            min_examples, min_gain)
+    if hist.is_cuda:
+        _C.gpu_split_scan(*args, _stream())
+    else:
+        _C.cpu_split_scan(*args)
+
+
+def update_node_ids(bins: torch.Tensor, node_ids: torch.Tensor,
+                    slot_map: torch.Tensor, best_feat: torch.Tensor,
+                    best_bin: torch.Tensor, level_base: int, level_size: int):
+    F, N = bins.shape
+    if bins.is_cuda:
+        _C.gpu_update_node_ids(bins.data_ptr(), node_ids.data_ptr(),
+                               slot_map.data_ptr(), best_feat.data_ptr(),
+                               best_bin.data_ptr(), N, level_base, level_size,
+                               _stream())
+    else:
+        _C.cpu_update_node_ids(bins.data_ptr(), node_ids.data_ptr(),
+                               slot_map.data_ptr(), best_feat.data_ptr(),
+                               best_bin.data_ptr(), N, level_base, level_size)
+
+
+def leaf_values(node_stats: torch.Tensor, out: torch.Tensor,
+                lambda_l2: float):
+    total = out.numel()
+    if node_stats.is_cuda:
+        _C.gpu_leaf_values(node_stats.data_ptr(), out.data_ptr(), total,
+                           lambda_l2, _stream())
+    else:
+        _C.cpu_leaf_values(node_stats.data_ptr(), out.data_ptr(), total,
+                           lambda_l2)
+    return out
+
+
+def update_preds(preds: torch.Tensor, node_ids: torch.Tensor,
+                 leaf_vals: torch.Tensor, shrinkage: float):
+    N = preds.numel()
+    if preds.is_cuda:
+        _C.gpu_update_preds(preds.data_ptr(), node_ids.data_ptr(),
+                            leaf_vals.data_ptr(), N, shrinkage, _stream())
+    else:
+        _C.cpu_update_preds(preds.data_ptr(), node_ids.data_ptr(),
+                            leaf_vals.data_ptr(), N, shrinkage)
+
+
+def binary_logloss(preds: torch.Tensor, labels: torch.Tensor,
+                   out2: torch.Tensor):
+    """Accumulates {loss_sum, n_correct} into out2 [2] f32 (pre-zeroed)."""
+    N = preds.numel()
+    if preds.is_cuda:
+        _C.gpu_binary_logloss(preds.data_ptr(), labels.data_ptr(),
+                              out2.data_ptr(), N, _stream())
+    else:
+        _C.cpu_binary_logloss(preds.data_ptr(), labels.data_ptr(),
+                              out2.data_ptr(), N)
+    return out2
+
+
+def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
+                   left: torch.Tensor, roots: torch.Tensor, out: torch.Tensor,
+                   tree_start: int = 0, tree_step: int = 1,
+                   n_trees: int = -1, init: float = 0.0, scale: float = 1.0):
+    """Flat-forest batch inference. X [F,N] f32, out [N] f32."""
+    F, N = X.shape
+    if n_trees < 0:
+        n_trees = roots.numel()
+    if X.is_cuda:
+        _C.gpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
+                              thr.data_ptr(), left.data_ptr(),
+                              roots.data_ptr(), tree_start, tree_step, n_trees,
+                              out.data_ptr(), init, scale, _stream())
+    else:
+        _C.cpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
+                              thr.data_ptr(), left.data_ptr(),
+                              roots.data_ptr(), tree_start, tree_step, n_trees,
+                              out.data_ptr(), init, scale)
+    return out
+
+
+def sigmoid(x: torch.Tensor, out: torch.Tensor):
+    if x.is_cuda:
+        _C.gpu_sigmoid(x.data_ptr(), out.data_ptr(), x.numel(), _stream())
+    else:
+        torch.sigmoid(x, out=out)
+    return out

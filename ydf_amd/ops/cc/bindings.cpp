@@ -1,0 +1,283 @@
+// pybind11 bindings for the MI355X-native decision-forest ops.
+// All tensor arguments are passed as raw device/host addresses
+// (torch.Tensor.data_ptr()) plus shapes; `stream` is
+// torch.cuda.current_stream().cuda_stream. This keeps the extension free of
+// torch headers (fast hipcc builds, no hipify involvement anywhere).
+#include <pybind11/pybind11.h>
+
+#include <cstdint>
+
+#include "common.h"
+
+namespace py = pybind11;
+using ydfa::SplitParams;
+
+extern "C" {
+// train_kernels.hip
+void gpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
+                  void*);
+void gpu_grad_hess(const float*, const float*, float*, int64_t, int, void*);
+void gpu_grad_hess_softmax(const float*, const float*, float*, int64_t, int,
+                           int, void*);
+void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
+                    const int32_t*, float*, int64_t, int, int, int, int, int,
+                    int, void*);
+void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
+void gpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
+                    int32_t*, int32_t*, float*, const uint8_t*, int, int, int,
+                    int, SplitParams, void*);
+void gpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
+                         const int32_t*, const int32_t*, int64_t, int, int,
+                         void*);
+void gpu_leaf_values(const float*, float*, int, float, void*);
+void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
+                      void*);
+void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
+// infer_kernels.hip
+void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
+                        const float*, const int32_t*, const int32_t*, int, int,
+                        int, float*, float, float, void*);
+void gpu_sigmoid(const float*, float*, int64_t, void*);
+// cpu_ops.cpp
+void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int);
+void cpu_grad_hess(const float*, const float*, float*, int64_t, int);
+void cpu_grad_hess_softmax(const float*, const float*, float*, int64_t, int,
+                           int);
+void cpu_hist_build(const uint8_t*, const float*, const int32_t*,
+                    const int32_t*, float*, int64_t, int, int, int, int, int,
+                    int);
+void cpu_weighted_target(const float*, const float*, float*, int64_t);
+void cpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
+                    int32_t*, int32_t*, float*, const uint8_t*, int, int, int,
+                    int, SplitParams);
+void cpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
+                         const int32_t*, const int32_t*, int64_t, int, int);
+void cpu_leaf_values(const float*, float*, int, float);
+void cpu_update_preds(float*, const int32_t*, const float*, int64_t, float);
+void cpu_binary_logloss(const float*, const float*, float*, int64_t);
+void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
+                        const float*, const int32_t*, const int32_t*, int, int,
+                        int, float*, float, float);
+}
+
+namespace {
+template <typename T>
+T* P(uintptr_t p) {
+  return reinterpret_cast<T*>(p);
+}
+SplitParams MakeSP(float lambda_l2, float min_hessian, int min_examples,
+                   float min_gain) {
+  SplitParams sp;
+  sp.lambda_l2 = lambda_l2;
+  sp.min_hessian = min_hessian;
+  sp.min_examples = min_examples;
+  sp.min_gain = min_gain;
+  return sp;
+}
+}  // namespace
+
+PYBIND11_MODULE(_ydf_ops, m) {
+  m.doc() = "MI355X-native decision forest ops (HIP/gfx950 + CPU)";
+  m.attr("max_bins") = ydfa::kMaxBins;
+  const auto nogil = py::call_guard<py::gil_scoped_release>();
+
+  // --- GPU ---
+  m.def("gpu_bin_data",
+        [](uintptr_t x, uintptr_t bnd, uintptr_t out, int64_t N, int F,
+           int n_cuts, uintptr_t stream) {
+          gpu_bin_data(P<float>(x), P<float>(bnd), P<uint8_t>(out), N, F,
+                       n_cuts, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_grad_hess",
+        [](uintptr_t preds, uintptr_t labels, uintptr_t gh, int64_t N,
+           int loss, uintptr_t stream) {
+          gpu_grad_hess(P<float>(preds), P<float>(labels), P<float>(gh), N,
+                        loss, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_grad_hess_softmax",
+        [](uintptr_t preds, uintptr_t labels, uintptr_t gh, int64_t N,
+           int n_classes, int cls, uintptr_t stream) {
+          gpu_grad_hess_softmax(P<float>(preds), P<float>(labels),
+                                P<float>(gh), N, n_classes, cls,
+                                (void*)stream);
+        },
+        nogil);
+  m.def("gpu_hist_build",
+        [](uintptr_t bins, uintptr_t gh, uintptr_t node_ids,
+           uintptr_t slot_map, uintptr_t hist, int64_t N, int F, int n_bins,
+           int level_base, int level_size, int slot0, int n_slots,
+           uintptr_t stream) {
+          gpu_hist_build(P<uint8_t>(bins), P<float>(gh), P<int32_t>(node_ids),
+                         P<int32_t>(slot_map), P<float>(hist), N, F, n_bins,
+                         level_base, level_size, slot0, n_slots,
+                         (void*)stream);
+        },
+        nogil);
+  m.def("gpu_weighted_target",
+        [](uintptr_t labels, uintptr_t weights, uintptr_t gh, int64_t N,
+           uintptr_t stream) {
+          gpu_weighted_target(P<float>(labels), P<float>(weights), P<float>(gh),
+                              N, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_split_scan",
+        [](uintptr_t hist, uintptr_t abs_of_slot, uintptr_t node_stats,
+           uintptr_t best_gain_nf, uintptr_t best_bin_nf, uintptr_t best_feat,
+           uintptr_t best_bin, uintptr_t best_gain, uintptr_t feat_mask,
+           int F, int n_bins, int slot0, int n_slots, float lambda_l2,
+           float min_hessian, int min_examples, float min_gain,
+           uintptr_t stream) {
+          gpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
+                         P<float>(node_stats), P<float>(best_gain_nf),
+                         P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
+                         P<int32_t>(best_bin), P<float>(best_gain),
+                         P<uint8_t>(feat_mask), F, n_bins, slot0, n_slots,
+                         MakeSP(lambda_l2, min_hessian, min_examples,
+                                min_gain),
+                         (void*)stream);
+        },
+        nogil);
+  m.def("gpu_update_node_ids",
+        [](uintptr_t bins, uintptr_t node_ids, uintptr_t slot_map,
+           uintptr_t best_feat, uintptr_t best_bin, int64_t N, int level_base,
+           int level_size, uintptr_t stream) {
+          gpu_update_node_ids(P<uint8_t>(bins), P<int32_t>(node_ids),
+                              P<int32_t>(slot_map), P<int32_t>(best_feat),
+                              P<int32_t>(best_bin), N, level_base, level_size,
+                              (void*)stream);
+        },
+        nogil);
+  m.def("gpu_leaf_values",
+        [](uintptr_t node_stats, uintptr_t leaf_values, int total_nodes,
+           float lambda_l2, uintptr_t stream) {
+          gpu_leaf_values(P<float>(node_stats), P<float>(leaf_values),
+                          total_nodes, lambda_l2, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_update_preds",
+        [](uintptr_t preds, uintptr_t node_ids, uintptr_t leaf_values,
+           int64_t N, float shrinkage, uintptr_t stream) {
+          gpu_update_preds(P<float>(preds), P<int32_t>(node_ids),
+                           P<float>(leaf_values), N, shrinkage,
+                           (void*)stream);
+        },
+        nogil);
+  m.def("gpu_binary_logloss",
+        [](uintptr_t preds, uintptr_t labels, uintptr_t out2, int64_t N,
+           uintptr_t stream) {
+          gpu_binary_logloss(P<float>(preds), P<float>(labels), P<float>(out2),
+                             N, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_predict_forest",
+        [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
+           uintptr_t left, uintptr_t roots, int tree_start, int tree_step,
+           int n_trees, uintptr_t out, float init, float scale,
+           uintptr_t stream) {
+          gpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
+                             P<float>(thr), P<int32_t>(left),
+                             P<int32_t>(roots), tree_start, tree_step, n_trees,
+                             P<float>(out), init, scale, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_sigmoid",
+        [](uintptr_t in, uintptr_t out, int64_t N, uintptr_t stream) {
+          gpu_sigmoid(P<float>(in), P<float>(out), N, (void*)stream);
+        },
+        nogil);
+
+  // --- CPU ---
+  m.def("cpu_bin_data",
+        [](uintptr_t x, uintptr_t bnd, uintptr_t out, int64_t N, int F,
+           int n_cuts) {
+          cpu_bin_data(P<float>(x), P<float>(bnd), P<uint8_t>(out), N, F,
+                       n_cuts);
+        },
+        nogil);
+  m.def("cpu_grad_hess",
+        [](uintptr_t preds, uintptr_t labels, uintptr_t gh, int64_t N,
+           int loss) {
+          cpu_grad_hess(P<float>(preds), P<float>(labels), P<float>(gh), N,
+                        loss);
+        },
+        nogil);
+  m.def("cpu_grad_hess_softmax",
+        [](uintptr_t preds, uintptr_t labels, uintptr_t gh, int64_t N,
+           int n_classes, int cls) {
+          cpu_grad_hess_softmax(P<float>(preds), P<float>(labels),
+                                P<float>(gh), N, n_classes, cls);
+        },
+        nogil);
+  m.def("cpu_hist_build",
+        [](uintptr_t bins, uintptr_t gh, uintptr_t node_ids,
+           uintptr_t slot_map, uintptr_t hist, int64_t N, int F, int n_bins,
+           int level_base, int level_size, int slot0, int n_slots) {
+          cpu_hist_build(P<uint8_t>(bins), P<float>(gh), P<int32_t>(node_ids),
+                         P<int32_t>(slot_map), P<float>(hist), N, F, n_bins,
+                         level_base, level_size, slot0, n_slots);
+        },
+        nogil);
+  m.def("cpu_weighted_target",
+        [](uintptr_t labels, uintptr_t weights, uintptr_t gh, int64_t N) {
+          cpu_weighted_target(P<float>(labels), P<float>(weights),
+                              P<float>(gh), N);
+        },
+        nogil);
+  m.def("cpu_split_scan",
+        [](uintptr_t hist, uintptr_t abs_of_slot, uintptr_t node_stats,
+           uintptr_t best_gain_nf, uintptr_t best_bin_nf, uintptr_t best_feat,
+           uintptr_t best_bin, uintptr_t best_gain, uintptr_t feat_mask,
+           int F, int n_bins, int slot0, int n_slots, float lambda_l2,
+           float min_hessian, int min_examples, float min_gain) {
+          cpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
+                         P<float>(node_stats), P<float>(best_gain_nf),
+                         P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
+                         P<int32_t>(best_bin), P<float>(best_gain),
+                         P<uint8_t>(feat_mask), F, n_bins, slot0, n_slots,
+                         MakeSP(lambda_l2, min_hessian, min_examples,
+                                min_gain));
+        },
+        nogil);
+  m.def("cpu_update_node_ids",
+        [](uintptr_t bins, uintptr_t node_ids, uintptr_t slot_map,
+           uintptr_t best_feat, uintptr_t best_bin, int64_t N, int level_base,
+           int level_size) {
+          cpu_update_node_ids(P<uint8_t>(bins), P<int32_t>(node_ids),
+                              P<int32_t>(slot_map), P<int32_t>(best_feat),
+                              P<int32_t>(best_bin), N, level_base,
+                              level_size);
+        },
+        nogil);
+  m.def("cpu_leaf_values",
+        [](uintptr_t node_stats, uintptr_t leaf_values, int total_nodes,
+           float lambda_l2) {
+          cpu_leaf_values(P<float>(node_stats), P<float>(leaf_values),
+                          total_nodes, lambda_l2);
+        },
+        nogil);
+  m.def("cpu_update_preds",
+        [](uintptr_t preds, uintptr_t node_ids, uintptr_t leaf_values,
+           int64_t N, float shrinkage) {
+          cpu_update_preds(P<float>(preds), P<int32_t>(node_ids),
+                           P<float>(leaf_values), N, shrinkage);
+        },
+        nogil);
+  m.def("cpu_binary_logloss",
+        [](uintptr_t preds, uintptr_t labels, uintptr_t out2, int64_t N) {
+          cpu_binary_logloss(P<float>(preds), P<float>(labels), P<float>(out2),
+                             N);
+        },
+        nogil);
+  m.def("cpu_predict_forest",
+        [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
+           uintptr_t left, uintptr_t roots, int tree_start, int tree_step,
+           int n_trees, uintptr_t out, float init, float scale) {
+          cpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
+                             P<float>(thr), P<int32_t>(left),
+                             P<int32_t>(roots), tree_start, tree_step, n_trees,
+                             P<float>(out), init, scale);
+        },
+        nogil);
+}

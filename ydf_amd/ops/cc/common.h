@@ -1,0 +1,40 @@
+// Common definitions shared by the CPU and GPU (HIP/gfx950) implementations
+// of the decision-forest training/inference ops.
+//
+// Design (MI355X-native, not a port):
+//   * Features are pre-binned to uint8 (<=256 quantile bins), stored
+//     feature-major: bins[f * N + i]. This is the GPU-resident analogue of the
+//     reference's DISCRETIZED_NUMERICAL columns + dataset_cache binned store
+//     (reference: yggdrasil_decision_forests/learner/distributed_decision_tree/
+//     dataset_cache/dataset_cache.h:15-58).
+//   * Trees are grown level-wise ("open nodes" of one depth at a time, like
+//     the reference's distributed layer-wise growth, training.h:145) over an
+//     implicit complete binary tree: node k has children 2k+1 / 2k+2.
+//   * Per-level histograms hist[node][feature][bin] = {sum_grad, sum_hess,
+//     count} are the collective unit: data-parallel ranks AllReduce this
+//     tensor (RCCL over xGMI) and then select splits redundantly.
+#pragma once
+#include <cstdint>
+
+namespace ydfa {
+
+// Fixed maximum number of bins. Runtime bin count n_bins <= kMaxBins.
+constexpr int kMaxBins = 256;
+
+// Loss ids (subset of reference model/gradient_boosted_trees/
+// gradient_boosted_trees.proto:54-80 enum).
+enum LossKind : int {
+  kLossSquaredError = 2,      // regression: g = pred - y, h = 1
+  kLossBinomial = 1,          // binary classification log-loss on logits
+  kLossMultinomial = 3,       // multi-class softmax cross-entropy
+};
+
+// Split-scan hyper-parameters (subset of the reference decision-tree proto).
+struct SplitParams {
+  float lambda_l2;        // l2_regularization
+  float min_hessian;      // min_sum_hessian_in_leaf
+  int min_examples;       // min_examples (reference default 5)
+  float min_gain;         // splits with gain <= min_gain become leaves
+};
+
+}  // namespace ydfa

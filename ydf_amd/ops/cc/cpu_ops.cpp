@@ -1,0 +1,363 @@
+// CPU implementations of the training/inference ops. Semantics match the
+// HIP kernels bit-for-bit where possible (identical tie-break rules,
+// identical gain formula); floating-point reduction order differs (sequential
+// here vs. LDS trees on GPU), which the tests cover with tolerances.
+// Capability analogue of the reference's threaded splitter
+// (learner/decision_tree/training.cc StreamProcessor jobs); parallelised
+// over features / row blocks with a persistent thread pool.
+#include <algorithm>
+#include <atomic>
+#include <cmath>
+#include <condition_variable>
+#include <cstdint>
+#include <cstring>
+#include <functional>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "common.h"
+
+namespace ydfa {
+
+// ---------------------------------------------------------------------------
+// Minimal persistent thread pool (reference analogue: utils/concurrency.h
+// ThreadPool). Static singleton, sized once from hardware_concurrency.
+// ---------------------------------------------------------------------------
+class ThreadPool {
+ public:
+  static ThreadPool& Get() {
+    static ThreadPool pool;
+    return pool;
+  }
+
+  int size() const { return (int)workers_.size(); }
+
+  // Runs fn(block_idx) for block_idx in [0, n_blocks), blocking until done.
+  void ParallelFor(int n_blocks, const std::function<void(int)>& fn) {
+    if (n_blocks <= 1 || workers_.empty()) {
+      for (int i = 0; i < n_blocks; ++i) fn(i);
+      return;
+    }
+    std::unique_lock<std::mutex> lk(m_);
+    fn_ = &fn;
+    next_.store(0);
+    remaining_.store(n_blocks);
+    n_blocks_ = n_blocks;
+    ++epoch_;
+    cv_.notify_all();
+    // Wait until all blocks ran AND every worker left the work loop — a
+    // straggler inside the loop must not observe the next epoch's counters.
+    done_cv_.wait(lk,
+                  [&] { return remaining_.load() == 0 && active_ == 0; });
+    fn_ = nullptr;
+  }
+
+ private:
+  ThreadPool() {
+    int n = (int)std::thread::hardware_concurrency();
+    if (n < 1) n = 1;
+    if (n > 64) n = 64;
+    for (int i = 0; i < n; ++i)
+      workers_.emplace_back([this] { WorkerLoop(); });
+  }
+  ~ThreadPool() {
+    {
+      std::lock_guard<std::mutex> lk(m_);
+      stop_ = true;
+      cv_.notify_all();
+    }
+    for (auto& t : workers_) t.join();
+  }
+
+  void WorkerLoop() {
+    uint64_t seen = 0;
+    for (;;) {
+      const std::function<void(int)>* fn;
+      {
+        std::unique_lock<std::mutex> lk(m_);
+        cv_.wait(lk, [&] { return stop_ || (fn_ && epoch_ != seen); });
+        if (stop_) return;
+        seen = epoch_;
+        fn = fn_;
+        ++active_;
+      }
+      for (;;) {
+        int i = next_.fetch_add(1);
+        if (i >= n_blocks_) break;
+        (*fn)(i);
+        remaining_.fetch_sub(1);
+      }
+      {
+        std::lock_guard<std::mutex> lk(m_);
+        --active_;
+        done_cv_.notify_all();
+      }
+    }
+  }
+
+  std::vector<std::thread> workers_;
+  std::mutex m_;
+  std::condition_variable cv_, done_cv_;
+  const std::function<void(int)>* fn_ = nullptr;
+  std::atomic<int> next_{0}, remaining_{0};
+  int active_ = 0;
+  int n_blocks_ = 0;
+  uint64_t epoch_ = 0;
+  bool stop_ = false;
+};
+
+extern "C" {
+
+void cpu_bin_data(const float* x, const float* boundaries, uint8_t* out,
+                  int64_t N, int F, int n_cuts) {
+  ThreadPool::Get().ParallelFor(F, [&](int f) {
+    const float* bnd = boundaries + (int64_t)f * n_cuts;
+    const float* xf = x + (int64_t)f * N;
+    uint8_t* of = out + (int64_t)f * N;
+    for (int64_t i = 0; i < N; ++i) {
+      const float v = xf[i];
+      int lo = 0, hi = n_cuts;
+      while (lo < hi) {
+        const int mid = (lo + hi) >> 1;
+        if (bnd[mid] < v) lo = mid + 1; else hi = mid;
+      }
+      of[i] = (uint8_t)lo;
+    }
+  });
+}
+
+void cpu_grad_hess(const float* preds, const float* labels, float* gh,
+                   int64_t N, int loss) {
+  const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
+                                           (int)(N / 16384) + 1));
+  const int64_t per = (N + nb - 1) / nb;
+  ThreadPool::Get().ParallelFor(nb, [&](int blk) {
+    const int64_t i0 = blk * per, i1 = std::min<int64_t>(i0 + per, N);
+    for (int64_t k = i0; k < i1; ++k) {
+      float g, h;
+      if (loss == kLossBinomial) {
+        const float p = 1.0f / (1.0f + std::exp(-preds[k]));
+        g = p - labels[k];
+        h = std::max(p * (1.0f - p), 1e-16f);
+      } else {
+        g = preds[k] - labels[k];
+        h = 1.0f;
+      }
+      gh[2 * k] = g;
+      gh[2 * k + 1] = h;
+    }
+  });
+}
+
+void cpu_grad_hess_softmax(const float* preds, const float* labels, float* gh,
+                           int64_t N, int n_classes, int cls) {
+  const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
+                                           (int)(N / 8192) + 1));
+  const int64_t per = (N + nb - 1) / nb;
+  ThreadPool::Get().ParallelFor(nb, [&](int blk) {
+    const int64_t i0 = blk * per, i1 = std::min<int64_t>(i0 + per, N);
+    for (int64_t k = i0; k < i1; ++k) {
+      float m = -1e30f;
+      for (int c = 0; c < n_classes; ++c)
+        m = std::max(m, preds[(int64_t)c * N + k]);
+      float denom = 0.f;
+      for (int c = 0; c < n_classes; ++c)
+        denom += std::exp(preds[(int64_t)c * N + k] - m);
+      const float p = std::exp(preds[(int64_t)cls * N + k] - m) / denom;
+      const float y = (labels[k] == (float)cls) ? 1.0f : 0.0f;
+      gh[2 * k] = p - y;
+      gh[2 * k + 1] = std::max(p * (1.0f - p), 1e-16f);
+    }
+  });
+}
+
+void cpu_weighted_target(const float* labels, const float* weights, float* gh,
+                         int64_t N) {
+  for (int64_t k = 0; k < N; ++k) {
+    const float w = weights ? weights[k] : 1.0f;
+    gh[2 * k] = -w * labels[k];
+    gh[2 * k + 1] = w;
+  }
+}
+
+// hist: [n_slots][F][n_bins][3], pre-zeroed by caller; hist base pointer
+// corresponds to slot0. Count accumulates examples with h != 0 (zero-weight
+// bootstrap rows are out-of-bag).
+void cpu_hist_build(const uint8_t* bins, const float* gh,
+                    const int32_t* node_ids, const int32_t* slot_map,
+                    float* hist, int64_t N, int F, int n_bins, int level_base,
+                    int level_size, int slot0, int n_slots) {
+  ThreadPool::Get().ParallelFor(F, [&](int f) {
+    const uint8_t* fb = bins + (int64_t)f * N;
+    for (int64_t i = 0; i < N; ++i) {
+      const int rel = node_ids[i] - level_base;
+      if (rel < 0 || rel >= level_size) continue;
+      const int slot = slot_map[rel] - slot0;
+      if (slot < 0 || slot >= n_slots) continue;
+      float* p = hist + (((int64_t)slot * F + f) * n_bins + fb[i]) * 3;
+      const float h = gh[2 * i + 1];
+      p[0] += gh[2 * i];
+      p[1] += h;
+      if (h != 0.f) p[2] += 1.0f;
+    }
+  });
+}
+
+void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
+                    float* node_stats, float* best_gain_nf,
+                    int32_t* best_bin_nf, int32_t* best_feat,
+                    int32_t* best_bin, float* best_gain,
+                    const uint8_t* feat_mask, int F, int n_bins, int slot0,
+                    int n_slots, SplitParams sp) {
+  ThreadPool::Get().ParallelFor(n_slots, [&](int slot) {
+    const int out = slot0 + slot;
+    const int abs_node = abs_of_slot[out];
+    // Node totals from feature 0 (every example lands in some bin).
+    const float* h0 = hist + ((int64_t)slot * F + 0) * (n_bins * 3);
+    float G = 0.f, H = 0.f, C = 0.f;
+    for (int b = 0; b < n_bins; ++b) {
+      G += h0[b * 3]; H += h0[b * 3 + 1]; C += h0[b * 3 + 2];
+    }
+    float* ns = node_stats + (int64_t)abs_node * 3;
+    ns[0] = G; ns[1] = H; ns[2] = C;
+    const float parent_term = G * G / (H + sp.lambda_l2);
+
+    float node_best_gain = -1e30f;
+    int node_best_f = -1, node_best_b = 0;
+    for (int f = 0; f < F; ++f) {
+      if (feat_mask != nullptr && !feat_mask[(int64_t)out * F + f]) {
+        best_gain_nf[(int64_t)slot * F + f] = -1e30f;
+        best_bin_nf[(int64_t)slot * F + f] = 0;
+        continue;
+      }
+      const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+      float GL = 0.f, HL = 0.f, CL = 0.f;
+      float fbest = -1e30f;
+      int fbin = 0;
+      for (int b = 0; b < n_bins - 1; ++b) {
+        GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+        const float GR = G - GL, HR = H - HL, CR = C - CL;
+        if (CL >= sp.min_examples && CR >= sp.min_examples &&
+            HL >= sp.min_hessian && HR >= sp.min_hessian) {
+          const float gain = GL * GL / (HL + sp.lambda_l2) +
+                             GR * GR / (HR + sp.lambda_l2) - parent_term;
+          if (gain > fbest) { fbest = gain; fbin = b; }
+        }
+      }
+      best_gain_nf[(int64_t)slot * F + f] = fbest;
+      best_bin_nf[(int64_t)slot * F + f] = fbin;
+      if (fbest > node_best_gain) {
+        node_best_gain = fbest;
+        node_best_f = f;
+        node_best_b = fbin;
+      }
+    }
+    if (node_best_f < 0 || node_best_gain <= sp.min_gain) {
+      best_feat[out] = -1;
+      best_bin[out] = 0;
+      best_gain[out] = 0.f;
+      return;
+    }
+    best_feat[out] = node_best_f;
+    best_bin[out] = node_best_b;
+    best_gain[out] = node_best_gain;
+    const float* hp = hist + ((int64_t)slot * F + node_best_f) * (n_bins * 3);
+    float GL = 0.f, HL = 0.f, CL = 0.f;
+    for (int b = 0; b <= node_best_b; ++b) {
+      GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+    }
+    float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
+    float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
+    nl[0] = GL; nl[1] = HL; nl[2] = CL;
+    nr[0] = G - GL; nr[1] = H - HL; nr[2] = C - CL;
+  });
+}
+
+void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
+                         const int32_t* slot_map, const int32_t* best_feat,
+                         const int32_t* best_bin, int64_t N, int level_base,
+                         int level_size) {
+  const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
+                                           (int)(N / 16384) + 1));
+  const int64_t per = (N + nb - 1) / nb;
+  ThreadPool::Get().ParallelFor(nb, [&](int blk) {
+    const int64_t i0 = blk * per, i1 = std::min<int64_t>(i0 + per, N);
+    for (int64_t k = i0; k < i1; ++k) {
+      const int nid = node_ids[k];
+      const int rel = nid - level_base;
+      if (rel < 0 || rel >= level_size) continue;
+      const int slot = slot_map[rel];
+      if (slot < 0) continue;
+      const int f = best_feat[slot];
+      if (f < 0) continue;
+      const int b = bins[(int64_t)f * N + k];
+      node_ids[k] = 2 * nid + 1 + (b > best_bin[slot] ? 1 : 0);
+    }
+  });
+}
+
+void cpu_leaf_values(const float* node_stats, float* leaf_values,
+                     int total_nodes, float lambda_l2) {
+  for (int i = 0; i < total_nodes; ++i) {
+    const float* ns = node_stats + (int64_t)i * 3;
+    leaf_values[i] = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+  }
+}
+
+void cpu_update_preds(float* preds, const int32_t* node_ids,
+                      const float* leaf_values, int64_t N, float shrinkage) {
+  const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
+                                           (int)(N / 16384) + 1));
+  const int64_t per = (N + nb - 1) / nb;
+  ThreadPool::Get().ParallelFor(nb, [&](int blk) {
+    const int64_t i0 = blk * per, i1 = std::min<int64_t>(i0 + per, N);
+    for (int64_t k = i0; k < i1; ++k) {
+      const int nid = node_ids[k];
+      if (nid >= 0) preds[k] += shrinkage * leaf_values[nid];
+    }
+  });
+}
+
+void cpu_binary_logloss(const float* preds, const float* labels, float* out2,
+                        int64_t N) {
+  double loss = 0.0, acc = 0.0;
+  for (int64_t k = 0; k < N; ++k) {
+    const float m = preds[k];
+    const float y = labels[k];
+    const float z = y > 0.5f ? -m : m;
+    loss += (z > 0.f) ? z + std::log1p(std::exp(-z)) : std::log1p(std::exp(z));
+    acc += ((m > 0.f) == (y > 0.5f)) ? 1.0 : 0.0;
+  }
+  out2[0] += (float)loss;
+  out2[1] += (float)acc;
+}
+
+void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
+                        const float* thr, const int32_t* left,
+                        const int32_t* roots, int tree_start, int tree_step,
+                        int n_trees, float* out, float init, float scale) {
+  (void)F;
+  const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
+                                           (int)(N / 1024) + 1));
+  const int64_t per = (N + nb - 1) / nb;
+  ThreadPool::Get().ParallelFor(nb, [&](int blk) {
+    const int64_t i0 = blk * per, i1 = std::min<int64_t>(i0 + per, N);
+    for (int64_t k = i0; k < i1; ++k) {
+      float acc = init;
+      for (int tt = 0; tt < n_trees; ++tt) {
+        int n = roots[tree_start + (int64_t)tt * tree_step];
+        int f = feat[n];
+        while (f >= 0) {
+          n = left[n] + (X[(int64_t)f * N + k] > thr[n] ? 1 : 0);
+          f = feat[n];
+        }
+        acc += thr[n];
+      }
+      out[k] = init + (acc - init) * scale;
+    }
+  });
+}
+
+}  // extern "C"
+}  // namespace ydfa

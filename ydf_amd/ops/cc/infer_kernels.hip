@@ -1,0 +1,118 @@
+// HIP/CDNA4 batched ensemble-inference kernels (the MI355X-native serving
+// engine; capability analogue of the reference's flat-node engines,
+// serving/decision_forest/decision_forest_serving.cc:268-344 PredictHelper,
+// redesigned example-parallel for 64-wide wavefronts with the example tile
+// staged in LDS).
+//
+// Model layout ("flat forest", built by ydf_amd/serving/flat.py):
+//   feat[n]  : i32, split feature of node n, -1 => leaf
+//   thr[n]   : f32, split threshold (x > thr -> right) or leaf value
+//   left[n]  : i32, index of left child (right = left + 1)
+//   roots[t] : i32, root node index of tree t
+// All node arrays are concatenated over trees.
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include "common.h"
+
+namespace ydfa {
+
+constexpr int kTile = 256;  // examples per block
+
+// Example tile staged in LDS: xs[f * kTile + tid]. The f-stride is a
+// multiple of 32 banks, so per-lane-group accesses with distinct tid never
+// conflict regardless of the (divergent) feature index.
+__global__ void predict_forest_lds_kernel(
+    const float* __restrict__ X, int64_t N, int F,
+    const int32_t* __restrict__ feat, const float* __restrict__ thr,
+    const int32_t* __restrict__ left, const int32_t* __restrict__ roots,
+    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
+    float init, float scale) {
+  extern __shared__ float xs[];  // [F][kTile]
+  const int64_t base = (int64_t)blockIdx.x * kTile;
+  const int tid = threadIdx.x;
+  const int64_t n_here = min((int64_t)kTile, N - base);
+  for (int idx = tid; idx < F * kTile; idx += blockDim.x) {
+    const int f = idx >> 8;     // idx / kTile
+    const int i = idx & 255;    // idx % kTile
+    xs[idx] = (i < n_here) ? X[(int64_t)f * N + base + i] : 0.f;
+  }
+  __syncthreads();
+  if (tid >= n_here) return;
+  float acc = init;
+  for (int tt = 0; tt < n_trees; ++tt) {
+    int n = roots[tree_start + (int64_t)tt * tree_step];
+    int f = feat[n];
+    while (f >= 0) {
+      n = left[n] + (xs[f * kTile + tid] > thr[n] ? 1 : 0);
+      f = feat[n];
+    }
+    acc += thr[n];
+  }
+  out[base + tid] = init + (acc - init) * scale;
+}
+
+// Fallback without the LDS tile (feature count too large to stage).
+__global__ void predict_forest_global_kernel(
+    const float* __restrict__ X, int64_t N, int F,
+    const int32_t* __restrict__ feat, const float* __restrict__ thr,
+    const int32_t* __restrict__ left, const int32_t* __restrict__ roots,
+    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
+    float init, float scale) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    float acc = init;
+    for (int tt = 0; tt < n_trees; ++tt) {
+      int n = roots[tree_start + (int64_t)tt * tree_step];
+      int f = feat[n];
+      while (f >= 0) {
+        n = left[n] + (X[(int64_t)f * N + k] > thr[n] ? 1 : 0);
+        f = feat[n];
+      }
+      acc += thr[n];
+    }
+    out[k] = init + (acc - init) * scale;
+  }
+}
+
+__global__ void sigmoid_kernel(const float* __restrict__ in,
+                               float* __restrict__ out, int64_t N) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride)
+    out[k] = 1.0f / (1.0f + __expf(-in[k]));
+}
+
+extern "C" {
+
+void gpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
+                        const float* thr, const int32_t* left,
+                        const int32_t* roots, int tree_start, int tree_step,
+                        int n_trees, float* out, float init, float scale,
+                        void* stream) {
+  const size_t lds = (size_t)F * kTile * sizeof(float);
+  if (lds <= 96 * 1024) {
+    const int grid = (int)((N + kTile - 1) / kTile);
+    hipLaunchKernelGGL(predict_forest_lds_kernel, dim3(grid), dim3(kTile), lds,
+                       (hipStream_t)stream, X, N, F, feat, thr, left, roots,
+                       tree_start, tree_step, n_trees, out, init, scale);
+  } else {
+    int grid = (int)((N + kTile - 1) / kTile);
+    if (grid > 4096) grid = 4096;
+    if (grid < 1) grid = 1;
+    hipLaunchKernelGGL(predict_forest_global_kernel, dim3(grid), dim3(kTile),
+                       0, (hipStream_t)stream, X, N, F, feat, thr, left, roots,
+                       tree_start, tree_step, n_trees, out, init, scale);
+  }
+}
+
+void gpu_sigmoid(const float* in, float* out, int64_t N, void* stream) {
+  int grid = (int)((N + 255) / 256);
+  if (grid > 2048) grid = 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(sigmoid_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, in, out, N);
+}
+
+}  // extern "C"
+}  // namespace ydfa

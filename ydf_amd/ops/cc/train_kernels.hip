@@ -1,0 +1,509 @@
+// HIP/CDNA4 (gfx950) training kernels for the MI355X-native decision-forest
+// engine. Hand-written for 64-wide wavefronts, LDS-staged histograms and
+// HBM3E-friendly access patterns — a new design, not a translation of the
+// reference's CPU splitter (SURVEY.md §2.3 maps reference hot paths to these
+// kernels).
+//
+// Level-wise growth over an implicit complete binary tree (node k ->
+// children 2k+1 / 2k+2), with a SPARSE active-node list per level: the host
+// keeps the (allreduce-deterministic) list of open nodes; `slot_map` maps a
+// node's level-relative index to its dense histogram slot (-1 = closed).
+// This keeps hist memory O(open nodes), not O(2^depth), so depth-16 random
+// forests stay feasible.
+//
+// Pipeline per tree (host loop in ydf_amd/learner/trainer.py):
+//   grad_hess / weighted_target : per-example {g,h}
+//                      (ref: loss_imp_*.cc UpdateGradients, elementwise)
+//   per level L, per slot-chunk:
+//     hist_build     : hist[slot][feat][bin] = {sum_g, sum_h, count}
+//                      (ref: splitter_scanner.h:95-185 bucket fill)
+//     [multi-GPU: RCCL AllReduce(hist) — inserted by the Python host loop]
+//     split_scan     : prefix-scan bins + deterministic argmax gain per slot
+//                      (ref: splitter_scanner.h:933-1101 ScanSplits)
+//   update_node_ids  : route examples to children (ref: training.cc:5324
+//                      SplitExamplesInPlace — here a node-id map update;
+//                      rows never move)
+//   leaf_values + update_preds
+//                      (ref: gradient_boosted_trees.cc:1576 UpdatePredictions)
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstdio>
+#include "common.h"
+
+namespace ydfa {
+
+constexpr int kBlock = 256;
+
+// ---------------------------------------------------------------------------
+// Binning: x[f*N+i] -> bin index via upper-bound binary search over the
+// feature's quantile boundaries (boundaries[f*n_cuts .. +n_cuts), ascending).
+// bin = number of cuts strictly below v, so "bin > b" <=> "v > cut[b]".
+// Boundaries staged in LDS (<= 255 floats per feature).
+// ---------------------------------------------------------------------------
+__global__ void bin_data_kernel(const float* __restrict__ x,
+                                const float* __restrict__ boundaries,
+                                uint8_t* __restrict__ out, int64_t N, int F,
+                                int n_cuts, int64_t rows_per_block) {
+  __shared__ float bnd[kMaxBins - 1];
+  const int f = blockIdx.x;
+  for (int i = threadIdx.x; i < n_cuts; i += blockDim.x)
+    bnd[i] = boundaries[(int64_t)f * n_cuts + i];
+  __syncthreads();
+  const int64_t row0 = (int64_t)blockIdx.y * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, N);
+  const float* xf = x + (int64_t)f * N;
+  uint8_t* of = out + (int64_t)f * N;
+  for (int64_t i = row0 + threadIdx.x; i < row1; i += blockDim.x) {
+    const float v = xf[i];
+    int lo = 0, hi = n_cuts;
+    while (lo < hi) {
+      const int mid = (lo + hi) >> 1;
+      if (bnd[mid] < v) lo = mid + 1; else hi = mid;
+    }
+    of[i] = (uint8_t)lo;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Gradients/hessians, written interleaved as float2 {g, h} for single-load
+// consumption by the histogram kernel.
+// ---------------------------------------------------------------------------
+__global__ void grad_hess_kernel(const float* __restrict__ preds,
+                                 const float* __restrict__ labels,
+                                 float2* __restrict__ gh, int64_t N,
+                                 int loss) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    float g, h;
+    if (loss == kLossBinomial) {
+      const float p = 1.0f / (1.0f + __expf(-preds[k]));
+      g = p - labels[k];
+      h = fmaxf(p * (1.0f - p), 1e-16f);
+    } else {  // squared error
+      g = preds[k] - labels[k];
+      h = 1.0f;
+    }
+    gh[k] = make_float2(g, h);
+  }
+}
+
+// Multi-class softmax cross-entropy: preds [C][N] (class-major), labels are
+// class indices. Writes gh for ONE class `cls` (host loops classes when
+// building the per-class trees of one iteration, matching the reference's
+// one-tree-per-class MULTINOMIAL loop, gradient_boosted_trees.cc:1539).
+__global__ void grad_hess_softmax_kernel(const float* __restrict__ preds,
+                                         const float* __restrict__ labels,
+                                         float2* __restrict__ gh, int64_t N,
+                                         int n_classes, int cls) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    float m = -1e30f;
+    for (int c = 0; c < n_classes; ++c)
+      m = fmaxf(m, preds[(int64_t)c * N + k]);
+    float denom = 0.f;
+    for (int c = 0; c < n_classes; ++c)
+      denom += __expf(preds[(int64_t)c * N + k] - m);
+    const float p = __expf(preds[(int64_t)cls * N + k] - m) / denom;
+    const float y = (labels[k] == (float)cls) ? 1.0f : 0.0f;
+    gh[k] = make_float2(p - y, fmaxf(p * (1.0f - p), 1e-16f));
+  }
+}
+
+// Random-forest / CART target: g = -w*y, h = w (leaf value -G/H = weighted
+// mean of y; gain = weighted variance reduction, which on 0/1 labels orders
+// splits like Gini). `weights` may be null (unit weights); bootstrap
+// sampling passes per-example draw counts as weights (w=0 = out-of-bag).
+__global__ void weighted_target_kernel(const float* __restrict__ labels,
+                                       const float* __restrict__ weights,
+                                       float2* __restrict__ gh, int64_t N) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    const float w = weights ? weights[k] : 1.0f;
+    gh[k] = make_float2(-w * labels[k], w);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Histogram build: one block owns (feature f, row chunk) and accumulates a
+// private LDS histogram for a CONTIGUOUS RANGE of slots [slot0, slot0+ng)
+// (ng * n_bins * 12 B <= 160 KiB => ng <= 53 at 256 bins; the host launcher
+// slices bigger levels). Examples resolve level-relative node -> slot via
+// slot_map; closed nodes (slot -1) and out-of-level examples are skipped.
+// The count accumulates examples with h != 0 so that zero-weight (out-of-
+// bag) rows don't satisfy min_examples.
+// ---------------------------------------------------------------------------
+__global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
+                                      const float2* __restrict__ gh,
+                                      const int32_t* __restrict__ node_ids,
+                                      const int32_t* __restrict__ slot_map,
+                                      float* __restrict__ hist, int64_t N,
+                                      int F, int n_bins, int level_base,
+                                      int level_size, int slot0, int n_slots,
+                                      int64_t rows_per_block) {
+  extern __shared__ float lhist[];  // [n_slots][n_bins][3]
+  const int f = blockIdx.x;
+  const int tot = n_slots * n_bins * 3;
+  for (int i = threadIdx.x; i < tot; i += blockDim.x) lhist[i] = 0.f;
+  __syncthreads();
+  const int64_t row0 = (int64_t)blockIdx.y * rows_per_block;
+  const int64_t row1 = min(row0 + rows_per_block, N);
+  const uint8_t* fb = bins + (int64_t)f * N;
+  for (int64_t i = row0 + threadIdx.x; i < row1; i += blockDim.x) {
+    const int rel = node_ids[i] - level_base;
+    if (rel < 0 || rel >= level_size) continue;
+    const int slot = slot_map[rel] - slot0;
+    if (slot < 0 || slot >= n_slots) continue;
+    const float2 v = gh[i];
+    float* p = lhist + ((slot * n_bins + (int)fb[i]) * 3);
+    atomicAdd(p, v.x);
+    atomicAdd(p + 1, v.y);
+    if (v.y != 0.f) atomicAdd(p + 2, 1.0f);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < tot; i += blockDim.x) {
+    const float v = lhist[i];
+    if (v != 0.f) {
+      const int slot = i / (n_bins * 3);
+      const int rem = i - slot * (n_bins * 3);
+      atomicAdd(&hist[((int64_t)slot * F + f) * (n_bins * 3) + rem], v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Split scan, stage A: one block per (slot, feature). Each thread owns one
+// bin; inclusive prefix sums of {g,h,c} over bins via LDS Hillis-Steele
+// (deterministic), then per-boundary gain and a deterministic block argmax.
+// Writes the per-(slot,feature) best {gain, bin}, and (f==0) node totals
+// into node_stats[abs_node]. Gain (2nd-order, reference use_hessian_gain
+// semantics): gain = GL^2/(HL+l2) + GR^2/(HR+l2) - G^2/(H+l2).
+// ---------------------------------------------------------------------------
+__global__ void split_scan_feat_kernel(const float* __restrict__ hist,
+                                       const int32_t* __restrict__ abs_of_slot,
+                                       float* __restrict__ node_stats,
+                                       float* __restrict__ best_gain_nf,
+                                       int32_t* __restrict__ best_bin_nf,
+                                       const uint8_t* __restrict__ feat_mask,
+                                       int F, int n_bins, int slot0,
+                                       SplitParams sp) {
+  const int slot = blockIdx.x;
+  const int f = blockIdx.y;
+  const int b = threadIdx.x;  // blockDim.x == n_bins (power of two <= 256)
+  __shared__ float sg[kMaxBins], sh[kMaxBins], sc[kMaxBins];
+  const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+  sg[b] = hp[b * 3];
+  sh[b] = hp[b * 3 + 1];
+  sc[b] = hp[b * 3 + 2];
+  __syncthreads();
+  for (int off = 1; off < n_bins; off <<= 1) {
+    float tg = 0.f, th = 0.f, tc = 0.f;
+    if (b >= off) { tg = sg[b - off]; th = sh[b - off]; tc = sc[b - off]; }
+    __syncthreads();
+    sg[b] += tg; sh[b] += th; sc[b] += tc;
+    __syncthreads();
+  }
+  const float G = sg[n_bins - 1], H = sh[n_bins - 1], C = sc[n_bins - 1];
+  if (f == 0 && b == 0) {
+    float* ns = node_stats + (int64_t)abs_of_slot[slot0 + slot] * 3;
+    ns[0] = G; ns[1] = H; ns[2] = C;
+  }
+  // Candidate split: left = bins [0..b], valid for b in [0, n_bins-2].
+  // feat_mask (per-slot feature sampling, reference num_candidate_attributes)
+  // disables the whole feature.
+  float gain = -1e30f;
+  if (b < n_bins - 1 &&
+      (feat_mask == nullptr || feat_mask[(slot0 + slot) * F + f])) {
+    const float GL = sg[b], HL = sh[b], CL = sc[b];
+    const float GR = G - GL, HR = H - HL, CR = C - CL;
+    if (CL >= sp.min_examples && CR >= sp.min_examples &&
+        HL >= sp.min_hessian && HR >= sp.min_hessian) {
+      gain = GL * GL / (HL + sp.lambda_l2) + GR * GR / (HR + sp.lambda_l2) -
+             G * G / (H + sp.lambda_l2);
+    }
+  }
+  // Deterministic argmax reduce: higher gain wins; ties -> smaller bin.
+  __shared__ float rg[kMaxBins];
+  __shared__ int rb[kMaxBins];
+  rg[b] = gain; rb[b] = b;
+  __syncthreads();
+  for (int off = n_bins >> 1; off > 0; off >>= 1) {
+    if (b < off) {
+      const float og = rg[b + off];
+      const int ob = rb[b + off];
+      if (og > rg[b] || (og == rg[b] && ob < rb[b])) { rg[b] = og; rb[b] = ob; }
+    }
+    __syncthreads();
+  }
+  if (b == 0) {
+    best_gain_nf[(int64_t)slot * F + f] = rg[0];
+    best_bin_nf[(int64_t)slot * F + f] = rb[0];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Split scan, stage B: one block per slot. Deterministic argmax over
+// features (ties -> smaller feature), then recompute left-child stats for
+// the winner and emit the split + child node stats. best_feat = -1 => leaf.
+// ---------------------------------------------------------------------------
+__global__ void split_select_kernel(const float* __restrict__ hist,
+                                    const int32_t* __restrict__ abs_of_slot,
+                                    const float* __restrict__ best_gain_nf,
+                                    const int32_t* __restrict__ best_bin_nf,
+                                    float* __restrict__ node_stats,
+                                    int32_t* __restrict__ best_feat,
+                                    int32_t* __restrict__ best_bin,
+                                    float* __restrict__ best_gain, int F,
+                                    int n_bins, int slot0, SplitParams sp) {
+  const int slot = blockIdx.x;
+  const int t = threadIdx.x;
+  __shared__ float rg[kBlock];
+  __shared__ int rf[kBlock];
+  float g = -1e30f;
+  int bf = -1;
+  for (int f = t; f < F; f += blockDim.x) {
+    const float fg = best_gain_nf[(int64_t)slot * F + f];
+    if (fg > g) { g = fg; bf = f; }
+  }
+  rg[t] = g; rf[t] = bf;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (t < off) {
+      const float og = rg[t + off];
+      const int of = rf[t + off];
+      if (og > rg[t] || (og == rg[t] && of >= 0 && (rf[t] < 0 || of < rf[t]))) {
+        rg[t] = og; rf[t] = of;
+      }
+    }
+    __syncthreads();
+  }
+  if (t != 0) return;
+  const float gain = rg[0];
+  const int f = rf[0];
+  const int out = slot0 + slot;
+  if (f < 0 || gain <= sp.min_gain) {
+    best_feat[out] = -1;
+    best_bin[out] = 0;
+    best_gain[out] = 0.f;
+    return;
+  }
+  const int bin = best_bin_nf[(int64_t)slot * F + f];
+  best_feat[out] = f;
+  best_bin[out] = bin;
+  best_gain[out] = gain;
+  const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+  float GL = 0.f, HL = 0.f, CL = 0.f;
+  for (int bb = 0; bb <= bin; ++bb) {
+    GL += hp[bb * 3]; HL += hp[bb * 3 + 1]; CL += hp[bb * 3 + 2];
+  }
+  const int abs_node = abs_of_slot[out];
+  const float* ns = node_stats + (int64_t)abs_node * 3;
+  float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
+  float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
+  nl[0] = GL; nl[1] = HL; nl[2] = CL;
+  nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
+}
+
+// ---------------------------------------------------------------------------
+// Route examples through this level's chosen splits: bin > split_bin ->
+// right child. Examples at closed/leaf nodes keep their node id ("parked").
+// best_feat/best_bin are indexed by SLOT (whole level, all chunks).
+// ---------------------------------------------------------------------------
+__global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
+                                       int32_t* __restrict__ node_ids,
+                                       const int32_t* __restrict__ slot_map,
+                                       const int32_t* __restrict__ best_feat,
+                                       const int32_t* __restrict__ best_bin,
+                                       int64_t N, int level_base,
+                                       int level_size) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    const int nid = node_ids[k];
+    const int rel = nid - level_base;
+    if (rel < 0 || rel >= level_size) continue;
+    const int slot = slot_map[rel];
+    if (slot < 0) continue;
+    const int f = best_feat[slot];
+    if (f < 0) continue;  // leaf: park
+    const int b = bins[(int64_t)f * N + k];
+    node_ids[k] = 2 * nid + 1 + (b > best_bin[slot]);
+  }
+}
+
+// leaf value for every materialized node: -G / (H + l2). Harmless for
+// internal nodes (examples only ever point at leaves).
+__global__ void leaf_values_kernel(const float* __restrict__ node_stats,
+                                   float* __restrict__ leaf_values,
+                                   int total_nodes, float lambda_l2) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total_nodes) return;
+  const float* ns = node_stats + (int64_t)i * 3;
+  leaf_values[i] = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+}
+
+__global__ void update_preds_kernel(float* __restrict__ preds,
+                                    const int32_t* __restrict__ node_ids,
+                                    const float* __restrict__ leaf_values,
+                                    int64_t N, float shrinkage) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    const int nid = node_ids[k];
+    if (nid >= 0) preds[k] += shrinkage * leaf_values[nid];
+  }
+}
+
+// Binary log-loss + accuracy partial sums (validation/early stopping),
+// block-reduced then atomically merged into out[0]=loss_sum, out[1]=correct.
+__global__ void binary_logloss_kernel(const float* __restrict__ preds,
+                                      const float* __restrict__ labels,
+                                      float* __restrict__ out, int64_t N) {
+  __shared__ float sl[kBlock], sa[kBlock];
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  float loss = 0.f, acc = 0.f;
+  for (int64_t k = i; k < N; k += stride) {
+    const float m = preds[k];
+    const float y = labels[k];
+    const float z = y > 0.5f ? -m : m;
+    loss += (z > 0.f) ? z + __logf(1.f + __expf(-z)) : __logf(1.f + __expf(z));
+    acc += ((m > 0.f) == (y > 0.5f)) ? 1.f : 0.f;
+  }
+  sl[threadIdx.x] = loss; sa[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sl[threadIdx.x] += sl[threadIdx.x + off];
+      sa[threadIdx.x] += sa[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    atomicAdd(&out[0], sl[0]);
+    atomicAdd(&out[1], sa[0]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers (raw pointers + explicit stream; exported via pybind).
+// ---------------------------------------------------------------------------
+static inline int row_chunks(int64_t N, int F, int max_blocks = 8192) {
+  // Enough blocks to fill 256 CUs x several waves, but bounded.
+  int per_f = (int)((max_blocks + F - 1) / F);
+  int64_t min_rows = 1024;
+  int64_t max_chunks = (N + min_rows - 1) / min_rows;
+  int chunks = (int)(max_chunks < per_f ? max_chunks : per_f);
+  return chunks < 1 ? 1 : chunks;
+}
+
+extern "C" {
+
+void gpu_bin_data(const float* x, const float* boundaries, uint8_t* out,
+                  int64_t N, int F, int n_cuts, void* stream) {
+  const int chunks = row_chunks(N, F);
+  const int64_t rpb = (N + chunks - 1) / chunks;
+  hipLaunchKernelGGL(bin_data_kernel, dim3(F, chunks), dim3(kBlock), 0,
+                     (hipStream_t)stream, x, boundaries, out, N, F, n_cuts,
+                     rpb);
+}
+
+static int elem_grid(int64_t N, int cap = 2048) {
+  int grid = (int)((N + kBlock - 1) / kBlock);
+  if (grid > cap) grid = cap;
+  if (grid < 1) grid = 1;
+  return grid;
+}
+
+void gpu_grad_hess(const float* preds, const float* labels, float* gh,
+                   int64_t N, int loss, void* stream) {
+  hipLaunchKernelGGL(grad_hess_kernel, dim3(elem_grid(N)), dim3(kBlock), 0,
+                     (hipStream_t)stream, preds, labels, (float2*)gh, N,
+                     loss);
+}
+
+void gpu_grad_hess_softmax(const float* preds, const float* labels, float* gh,
+                           int64_t N, int n_classes, int cls, void* stream) {
+  hipLaunchKernelGGL(grad_hess_softmax_kernel, dim3(elem_grid(N)),
+                     dim3(kBlock), 0, (hipStream_t)stream, preds, labels,
+                     (float2*)gh, N, n_classes, cls);
+}
+
+void gpu_weighted_target(const float* labels, const float* weights, float* gh,
+                         int64_t N, void* stream) {
+  hipLaunchKernelGGL(weighted_target_kernel, dim3(elem_grid(N)), dim3(kBlock),
+                     0, (hipStream_t)stream, labels, weights, (float2*)gh, N);
+}
+
+// hist must be zeroed by the caller. Slices [slot0, slot0+n_slots) into
+// LDS-sized groups internally; hist is the base pointer for slot0.
+void gpu_hist_build(const uint8_t* bins, const float* gh,
+                    const int32_t* node_ids, const int32_t* slot_map,
+                    float* hist, int64_t N, int F, int n_bins, int level_base,
+                    int level_size, int slot0, int n_slots, void* stream) {
+  const int max_lds_slots = (160 * 1024) / (n_bins * 3 * (int)sizeof(float));
+  const int group = n_slots < max_lds_slots ? n_slots : max_lds_slots;
+  const int chunks = row_chunks(N, F);
+  const int64_t rpb = (N + chunks - 1) / chunks;
+  for (int s0 = 0; s0 < n_slots; s0 += group) {
+    const int ng = (n_slots - s0) < group ? (n_slots - s0) : group;
+    const size_t lds = (size_t)ng * n_bins * 3 * sizeof(float);
+    hipLaunchKernelGGL(hist_build_lds_kernel, dim3(F, chunks), dim3(kBlock),
+                       lds, (hipStream_t)stream, bins, (const float2*)gh,
+                       node_ids, slot_map,
+                       hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
+                       level_base, level_size, slot0 + s0, ng, rpb);
+  }
+}
+
+void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
+                    float* node_stats, float* best_gain_nf,
+                    int32_t* best_bin_nf, int32_t* best_feat,
+                    int32_t* best_bin, float* best_gain,
+                    const uint8_t* feat_mask, int F, int n_bins, int slot0,
+                    int n_slots, SplitParams sp, void* stream) {
+  hipLaunchKernelGGL(split_scan_feat_kernel, dim3(n_slots, F), dim3(n_bins),
+                     0, (hipStream_t)stream, hist, abs_of_slot, node_stats,
+                     best_gain_nf, best_bin_nf, feat_mask, F, n_bins, slot0,
+                     sp);
+  hipLaunchKernelGGL(split_select_kernel, dim3(n_slots), dim3(kBlock), 0,
+                     (hipStream_t)stream, hist, abs_of_slot, best_gain_nf,
+                     best_bin_nf, node_stats, best_feat, best_bin, best_gain,
+                     F, n_bins, slot0, sp);
+}
+
+void gpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
+                         const int32_t* slot_map, const int32_t* best_feat,
+                         const int32_t* best_bin, int64_t N, int level_base,
+                         int level_size, void* stream) {
+  hipLaunchKernelGGL(update_node_ids_kernel, dim3(elem_grid(N, 4096)),
+                     dim3(kBlock), 0, (hipStream_t)stream, bins, node_ids,
+                     slot_map, best_feat, best_bin, N, level_base, level_size);
+}
+
+void gpu_leaf_values(const float* node_stats, float* leaf_values,
+                     int total_nodes, float lambda_l2, void* stream) {
+  const int grid = (total_nodes + kBlock - 1) / kBlock;
+  hipLaunchKernelGGL(leaf_values_kernel, dim3(grid), dim3(kBlock), 0,
+                     (hipStream_t)stream, node_stats, leaf_values, total_nodes,
+                     lambda_l2);
+}
+
+void gpu_update_preds(float* preds, const int32_t* node_ids,
+                      const float* leaf_values, int64_t N, float shrinkage,
+                      void* stream) {
+  hipLaunchKernelGGL(update_preds_kernel, dim3(elem_grid(N, 4096)),
+                     dim3(kBlock), 0, (hipStream_t)stream, preds, node_ids,
+                     leaf_values, N, shrinkage);
+}
+
+void gpu_binary_logloss(const float* preds, const float* labels, float* out2,
+                        int64_t N, void* stream) {
+  hipLaunchKernelGGL(binary_logloss_kernel, dim3(elem_grid(N)), dim3(kBlock),
+                     0, (hipStream_t)stream, preds, labels, out2, N);
+}
+
+}  // extern "C"
+}  // namespace ydfa

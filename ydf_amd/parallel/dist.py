@@ -1,0 +1,60 @@
+"""Distributed helpers: one process per GPU over RCCL/xGMI.
+
+Replaces the reference's gRPC manager/worker distribution
+(utils/distribute/, distributed_gradient_boosted_trees.cc): the only
+collective the training algorithm needs is the per-level histogram
+all-reduce (+ scalar stat reductions), done with torch.distributed
+(backend "nccl" IS RCCL on ROCm; "gloo" for CPU tests).
+"""
+from __future__ import annotations
+
+import datetime
+import os
+
+import torch
+import torch.distributed as dist
+
+
+def init_from_env(backend: str = None) -> int:
+    """Initializes torch.distributed from torchrun env vars; returns rank.
+
+    No-op (returns 0) when WORLD_SIZE is absent or 1."""
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return 0
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if backend == "nccl":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        dist.init_process_group(backend=backend,
+                                timeout=datetime.timedelta(seconds=300))
+    return dist.get_rank()
+
+
+def is_main() -> bool:
+    return (not dist.is_available()) or (not dist.is_initialized()) \
+        or dist.get_rank() == 0
+
+
+def world_size() -> int:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_world_size()
+    return 1
+
+
+def barrier() -> None:
+    if dist.is_available() and dist.is_initialized():
+        dist.barrier()
+
+
+def shard_rows(n: int, rank: int = None, world: int = None):
+    """Contiguous row shard [lo, hi) for this rank."""
+    if world is None:
+        world = world_size()
+    if rank is None:
+        rank = dist.get_rank() if world > 1 else 0
+    per = (n + world - 1) // world
+    lo = min(rank * per, n)
+    hi = min(lo + per, n)
+    return lo, hi
