@@ -1,0 +1,170 @@
+#!/usr/bin/env python3
+"""Flagship training benchmark: Higgs-shape GBT on MI355X.
+
+One "step" = one boosting iteration (one tree: gradients -> per-level LDS
+histograms [-> RCCL all-reduce] -> split scan -> routing -> leaf update)
+on synthetic Higgs-shaped data (11M rows x 28 numerical features, binary
+label), 300 trees depth 6 being the full BASELINE.json config.
+
+Usage:
+  python bench.py --gpus N --steps K --warmup W
+For N > 1 the driver launches this under torch.distributed.run with one
+rank per GPU (RCCL over xGMI); total rows stay fixed and are row-sharded
+across ranks (strong scaling).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from ydf_amd import ops  # noqa: E402
+from ydf_amd.learner import trainer as T  # noqa: E402
+from ydf_amd.parallel import dist as dist_lib  # noqa: E402
+
+N_ROWS = 11_000_000
+N_FEATURES = 28
+FULL_TREES = 300
+MAX_DEPTH = 6
+
+
+def make_higgs_shaped(rows: int, device: torch.device, seed: int = 1234):
+    """Synthetic data of the Higgs shape (11M x 28 continuous features,
+    binary label from a nonlinear margin + noise). Identical on every rank
+    (same seed), so shards are consistent with shared bin boundaries."""
+    g = torch.Generator(device=device)
+    g.manual_seed(seed)
+    X = torch.randn((N_FEATURES, rows), generator=g, device=device,
+                    dtype=torch.float32)
+    w = torch.randn((N_FEATURES,), generator=g, device=device)
+    margin = (X * w.view(-1, 1)).sum(dim=0)
+    margin += 0.8 * X[0] * X[1] - 0.6 * X[2].abs() * X[3]
+    noise = torch.randn((rows,), generator=g, device=device)
+    y = (margin + 0.5 * noise > 0).float()
+    return X, y
+
+
+def device_boundaries(X: torch.Tensor, n_cuts: int = 255):
+    """Quantile cuts from a per-feature sample, computed on device."""
+    sample = X[:, : min(X.shape[1], 1 << 18)]
+    srt, _ = torch.sort(sample, dim=1)
+    m = srt.shape[1]
+    idx = torch.linspace(0, m - 1, n_cuts + 2, device=X.device)[1:-1].long()
+    return srt[:, idx].contiguous()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=40)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--rows", type=int, default=N_ROWS,
+                    help="total rows (testing only; default = Higgs 11M)")
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+
+    rank = dist_lib.init_from_env()
+    world = dist_lib.world_size()
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    X, y = make_higgs_shaped(args.rows, device)
+    bnd = device_boundaries(X)
+    bins_full = torch.empty(X.shape, dtype=torch.uint8, device=device)
+    ops.bin_data(X, bnd, bins_full)
+    del X
+    lo, hi = dist_lib.shard_rows(args.rows, rank, world)
+    bins = bins_full[:, lo:hi].contiguous()
+    labels = y[lo:hi].contiguous()
+    del bins_full, y
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+
+    cfg = T.TrainerConfig(
+        loss=T.LOSS_BINOMIAL, num_trees=FULL_TREES, max_depth=MAX_DEPTH,
+        shrinkage=0.1, lambda_l2=0.0, min_examples=5, min_hessian=1e-3,
+        subsample=1.0, seed=1234)
+    tr = T.ForestTrainer(bins, labels, cfg)
+    N = labels.numel()
+
+    # initial predictions (log-odds of the global base rate)
+    s = torch.stack([labels.sum(), torch.tensor(float(N), device=device)])
+    tr._allreduce(s)
+    p = (s[0] / s[1]).clamp(1e-6, 1 - 1e-6)
+    init = float(torch.log(p / (1 - p)).item())
+    preds = torch.full((N,), init, dtype=torch.float32, device=device)
+
+    def step(i: int):
+        ops.grad_hess(preds, labels, tr.gh, cfg.loss)
+        tr.grow_tree(i)
+        ops.update_preds(preds, tr.node_ids, tr.leaf_vals, cfg.shrinkage)
+
+    for i in range(args.warmup):
+        step(i)
+    dist_lib.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    dist_lib.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    e = torch.tensor([elapsed], dtype=torch.float64)
+    if world > 1:
+        import torch.distributed as td
+
+        et = e.to(device) if td.get_backend() == "nccl" else e
+        td.all_reduce(et, op=td.ReduceOp.MAX)
+        e = et.cpu()
+    elapsed = float(e[0])
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    value = args.steps / elapsed  # whole-job trees/sec (ranks cooperate)
+    if rank == 0:
+        out = {
+            "metric": "gbt_higgs11m_train_trees_per_s",
+            "value": value,
+            "unit": "trees/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "GradientBoostedTrees Higgs-11Mx28 depth-6",
+                "rows": args.rows,
+                "features": N_FEATURES,
+                "num_trees_full_run": FULL_TREES,
+                "max_depth": MAX_DEPTH,
+                "n_bins": 256,
+                "global_batch": args.rows,
+                "seq_len": None,
+                "parallelism": f"dp{world} row-sharded histogram "
+                               "all-reduce (RCCL/xGMI)",
+            },
+        }
+        print(json.dumps(out), flush=True)
+        est_full = FULL_TREES * ms_per_step / 1000.0
+        print(f"# est. full {FULL_TREES}-tree train wall-clock: "
+              f"{est_full:.2f}s on {world} GPU(s)", file=sys.stderr)
+
+
+if __name__ == "__main__":
+    main()

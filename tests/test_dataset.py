@@ -1,0 +1,97 @@
+import numpy as np
+import pytest
+import torch
+
+import ydf_amd as ydf
+from ydf_amd import ops
+from ydf_amd.dataset.dataspec import (Semantic, categorical_vocab,
+                                      numerical_boundaries)
+from ydf_amd.model.forest import padded_boundaries
+
+
+def test_infer_dataspec_semantics():
+    data = {
+        "num": np.array([1.0, 2.0, 3.0]),
+        "cat": np.array(["a", "b", "a"]),
+        "int": np.array([1, 2, 3]),
+        "label": np.array(["x", "y", "x"]),
+    }
+    ds = ydf.create_vertical_dataset(data, label="label",
+                                     task=ydf.Task.CLASSIFICATION)
+    spec = ds.dataspec
+    assert spec.column("num").semantic == Semantic.NUMERICAL
+    assert spec.column("cat").semantic == Semantic.CATEGORICAL
+    assert spec.column("int").semantic == Semantic.NUMERICAL
+    assert spec.column("label").semantic == Semantic.CATEGORICAL
+    # vocab: OOV first, then by frequency
+    assert spec.column("cat").vocab[0] == "<OOD>"
+    assert spec.column("cat").vocab[1] == "a"
+    # label classes 0-based: x->0, y->1
+    assert ds.label_values.tolist() == [0.0, 1.0, 0.0]
+
+
+def test_categorical_vocab_frequency_order():
+    v = categorical_vocab(np.array(["b", "b", "b", "a", "a", "c"]))
+    assert v == ["<OOD>", "b", "a", "c"]
+
+
+def test_numerical_boundaries_dedup():
+    v = np.array([1.0] * 100 + [2.0] * 100, dtype=np.float32)
+    b = numerical_boundaries(v, max_bins=256)
+    assert len(b) <= 2
+    assert all(np.diff(b) > 0) if len(b) > 1 else True
+
+
+def test_bin_data_matches_searchsorted():
+    rng = np.random.RandomState(0)
+    x = rng.randn(3, 1000).astype(np.float32)
+    specs = ydf.create_vertical_dataset(
+        {"a": x[0], "b": x[1], "c": x[2], "label": x[0] > 0},
+        label="label").dataspec.feature_columns
+    bnd = padded_boundaries(specs)
+    xt = torch.from_numpy(np.ascontiguousarray(x))
+    bt = torch.from_numpy(bnd)
+    out = torch.empty(xt.shape, dtype=torch.uint8)
+    ops.bin_data(xt, bt, out)
+    for f in range(3):
+        expected = np.searchsorted(bnd[f], x[f], side="left")
+        # bin = number of cuts strictly below v
+        expected = np.sum(bnd[f][None, :] < x[f][:, None], axis=1)
+        np.testing.assert_array_equal(out[f].numpy(), expected)
+
+
+def test_bin_threshold_equivalence():
+    # "bin > b" must be exactly "x > cut[b]"
+    rng = np.random.RandomState(1)
+    x = rng.randn(1, 5000).astype(np.float32)
+    specs = ydf.create_vertical_dataset(
+        {"a": x[0], "label": x[0] > 0}, label="label").dataspec
+    bnd = padded_boundaries(specs.feature_columns)
+    xt = torch.from_numpy(np.ascontiguousarray(x))
+    out = torch.empty(xt.shape, dtype=torch.uint8)
+    ops.bin_data(xt, torch.from_numpy(bnd), out)
+    bins = out[0].numpy().astype(np.int64)
+    for b in (0, 5, 100, bnd.shape[1] - 1):
+        cut = bnd[0, b]
+        if not np.isfinite(cut):
+            continue
+        np.testing.assert_array_equal(bins > b, x[0] > cut)
+
+
+def test_nan_imputed_by_mean():
+    x = np.array([1.0, np.nan, 3.0], dtype=np.float32)
+    ds = ydf.create_vertical_dataset(
+        {"a": x, "label": np.array([0.0, 1.0, 0.0])}, label="label",
+        task=ydf.Task.REGRESSION)
+    assert ds.dataspec.column("a").num_nas == 1
+    np.testing.assert_allclose(ds.X[0], [1.0, 2.0, 3.0])
+
+
+def test_pandas_input():
+    pd = pytest.importorskip("pandas")
+    df = pd.DataFrame({"a": [1.0, 2.0], "b": ["x", "y"],
+                       "label": ["p", "q"]})
+    ds = ydf.create_vertical_dataset(df, label="label",
+                                     task=ydf.Task.CLASSIFICATION)
+    assert ds.n_examples == 2
+    assert ds.n_features == 2

@@ -1,0 +1,202 @@
+"""HIP kernel numerics tests: every GPU op against its C++ CPU twin and/or a
+plain torch fp32 reference. Run on an MI355X box (`pytest -m gpu`)."""
+import numpy as np
+import pytest
+import torch
+
+import ydf_amd as ydf
+from ydf_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                                  reason="no GPU")
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _check_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU available")
+
+
+@pytest.fixture(scope="module")
+def dev():
+    return torch.device("cuda:0")
+
+
+def test_native_extension_loaded():
+    import ydf_amd._ydf_ops as m
+
+    assert "ydf_amd" in m.__file__, m.__file__
+
+
+def test_bin_data_gpu_vs_cpu(dev):
+    rng = np.random.RandomState(0)
+    x = rng.randn(5, 100000).astype(np.float32)
+    bnd = np.sort(rng.randn(5, 255).astype(np.float32), axis=1)
+    xt = torch.from_numpy(x)
+    bt = torch.from_numpy(bnd)
+    out_cpu = torch.empty(xt.shape, dtype=torch.uint8)
+    ops.bin_data(xt, bt, out_cpu)
+    out_gpu = torch.empty(xt.shape, dtype=torch.uint8, device=dev)
+    ops.bin_data(xt.to(dev), bt.to(dev), out_gpu)
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(out_gpu.cpu().numpy(), out_cpu.numpy())
+
+
+def test_grad_hess_gpu_vs_torch(dev):
+    rng = np.random.RandomState(1)
+    preds = torch.from_numpy(rng.randn(50000).astype(np.float32)).to(dev)
+    labels = torch.from_numpy(
+        (rng.rand(50000) > 0.5).astype(np.float32)).to(dev)
+    gh = torch.empty((50000, 2), dtype=torch.float32, device=dev)
+    ops.grad_hess(preds, labels, gh, 1)
+    torch.cuda.synchronize()
+    p = torch.sigmoid(preds)
+    torch.testing.assert_close(gh[:, 0], p - labels, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(gh[:, 1], (p * (1 - p)).clamp_min(1e-16),
+                               rtol=1e-4, atol=1e-5)
+
+
+def test_hist_build_gpu_vs_cpu_exact_integers(dev):
+    """Integer-valued {g,h} make float sums order-independent, so GPU LDS
+    atomics must equal CPU sequential sums EXACTLY."""
+    rng = np.random.RandomState(2)
+    F, N = 6, 200000
+    bins = rng.randint(0, 256, size=(F, N)).astype(np.uint8)
+    g = rng.randint(-8, 8, N).astype(np.float32)
+    h = rng.randint(0, 4, N).astype(np.float32)
+    node_ids = (rng.randint(0, 8, N) + 7).astype(np.int32)
+    gh = torch.from_numpy(np.stack([g, h], 1).copy())
+    slot_map = torch.arange(8, dtype=torch.int32)
+    hist_c = torch.zeros((8, F, 256, 3))
+    ops.hist_build(torch.from_numpy(bins), gh, torch.from_numpy(node_ids),
+                   slot_map, hist_c, 7, 8, 0, 8)
+    hist_g = torch.zeros((8, F, 256, 3), device=dev)
+    ops.hist_build(torch.from_numpy(bins).to(dev), gh.to(dev),
+                   torch.from_numpy(node_ids).to(dev), slot_map.to(dev),
+                   hist_g, 7, 8, 0, 8)
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(hist_g.cpu().numpy(), hist_c.numpy())
+
+
+def test_split_scan_gpu_vs_cpu(dev):
+    rng = np.random.RandomState(3)
+    F, B, n_slots = 7, 256, 5
+    hist = rng.randint(0, 50, size=(n_slots, F, B, 3)).astype(np.float32)
+    ht = torch.from_numpy(hist)
+    args = dict(lambda_l2=1.0, min_hessian=0.0, min_examples=5, min_gain=0.0)
+    abs_of_slot = torch.arange(7, 7 + n_slots, dtype=torch.int32)
+
+    def run(device):
+        h = ht.to(device)
+        aos = abs_of_slot.to(device)
+        ns = torch.zeros((63, 3), device=device)
+        bg = torch.empty((n_slots, F), device=device)
+        bb = torch.empty((n_slots, F), dtype=torch.int32, device=device)
+        bf = torch.empty(n_slots, dtype=torch.int32, device=device)
+        bbin = torch.empty(n_slots, dtype=torch.int32, device=device)
+        bgain = torch.empty(n_slots, device=device)
+        ops.split_scan(h, aos, ns, bg, bb, bf, bbin, bgain, 0, n_slots,
+                       args["lambda_l2"], args["min_hessian"],
+                       args["min_examples"], args["min_gain"])
+        if device != "cpu":
+            torch.cuda.synchronize()
+        return (bf.cpu().numpy(), bbin.cpu().numpy(), bgain.cpu().numpy(),
+                ns.cpu().numpy())
+
+    f_c, b_c, g_c, ns_c = run("cpu")
+    f_g, b_g, g_g, ns_g = run(dev)
+    np.testing.assert_array_equal(f_g, f_c)
+    np.testing.assert_array_equal(b_g, b_c)
+    np.testing.assert_allclose(g_g, g_c, rtol=1e-4)
+    np.testing.assert_allclose(ns_g, ns_c, rtol=1e-4, atol=1e-3)
+
+
+def test_update_node_ids_gpu_vs_cpu(dev):
+    rng = np.random.RandomState(4)
+    F, N = 4, 300000
+    bins = rng.randint(0, 256, size=(F, N)).astype(np.uint8)
+    node_ids0 = (rng.randint(0, 4) * 0 + rng.randint(3, 7, N)).astype(
+        np.int32)
+    best_feat = rng.randint(-1, F, 4).astype(np.int32)
+    best_bin = rng.randint(0, 255, 4).astype(np.int32)
+    slot_map = torch.arange(4, dtype=torch.int32)
+    ids_c = torch.from_numpy(node_ids0.copy())
+    ops.update_node_ids(torch.from_numpy(bins), ids_c, slot_map,
+                        torch.from_numpy(best_feat),
+                        torch.from_numpy(best_bin), 3, 4)
+    ids_g = torch.from_numpy(node_ids0.copy()).to(dev)
+    ops.update_node_ids(torch.from_numpy(bins).to(dev), ids_g,
+                        slot_map.to(dev),
+                        torch.from_numpy(best_feat).to(dev),
+                        torch.from_numpy(best_bin).to(dev), 3, 4)
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(ids_g.cpu().numpy(), ids_c.numpy())
+
+
+def test_predict_forest_gpu_vs_cpu(dev, binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=30,
+                                        device="cpu").train(binary_data)
+    p_cpu = m.predict(binary_data, device="cpu")
+    p_gpu = m.predict(binary_data, device="cuda")
+    np.testing.assert_allclose(p_gpu, p_cpu, rtol=1e-5, atol=1e-6)
+
+
+def test_rf_gpu_bit_exact_vs_cpu(binary_data):
+    """Integer gradient sums: the GPU-trained forest must match the
+    CPU-trained forest exactly (same argmax, same thresholds)."""
+    kw = dict(label="label", num_trees=3, max_depth=6,
+              bootstrap_training_dataset=False, num_candidate_attributes=-1)
+    m_cpu = ydf.RandomForestLearner(device="cpu", **kw).train(binary_data)
+    m_gpu = ydf.RandomForestLearner(device="cuda", **kw).train(binary_data)
+    np.testing.assert_array_equal(m_gpu.forest.feat, m_cpu.forest.feat)
+    np.testing.assert_array_equal(m_gpu.forest.left, m_cpu.forest.left)
+    np.testing.assert_allclose(m_gpu.forest.thr, m_cpu.forest.thr, rtol=1e-6)
+
+
+def test_gbt_gpu_quality(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=100,
+                                        device="cuda").train(binary_data)
+    ev = m.evaluate(binary_data, device="cuda")
+    assert ev.accuracy > 0.93
+    assert ev.auc > 0.97
+
+
+def test_gbt_gpu_regression(regression_data):
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=100,
+        device="cuda").train(regression_data)
+    ev = m.evaluate(regression_data, device="cuda")
+    assert ev.rmse < 0.5
+
+
+def test_gbt_gpu_multiclass():
+    rng = np.random.RandomState(2)
+    n = 20000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where(x1 > 0.5, "a", np.where(x2 > 0, "b", "c"))
+    d = {"x1": x1, "x2": x2, "label": y}
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=40,
+                                        device="cuda").train(d)
+    assert m.evaluate(d, device="cuda").accuracy > 0.98
+
+
+def test_rf_gpu_feature_sampling_and_bootstrap(binary_data):
+    m = ydf.RandomForestLearner(label="label", num_trees=20, max_depth=10,
+                                device="cuda").train(binary_data)
+    assert m.evaluate(binary_data, device="cuda").accuracy > 0.9
+
+
+def test_binary_logloss_gpu(dev):
+    rng = np.random.RandomState(0)
+    preds = torch.from_numpy(rng.randn(100000).astype(np.float32)).to(dev)
+    labels = torch.from_numpy(
+        (rng.rand(100000) > 0.4).astype(np.float32)).to(dev)
+    out = torch.zeros(2, device=dev)
+    ops.binary_logloss(preds, labels, out)
+    torch.cuda.synchronize()
+    ref = torch.nn.functional.binary_cross_entropy_with_logits(
+        preds, labels, reduction="sum")
+    np.testing.assert_allclose(out[0].item(), ref.item(), rtol=1e-3)

@@ -1,0 +1,145 @@
+"""End-to-end learner quality tests (reference analogue: the
+TrainAndTestTester metric-margin checks, utils/test_utils.h:428)."""
+import numpy as np
+import pytest
+
+import ydf_amd as ydf
+
+
+def test_gbt_binary_quality(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=100).train(
+        binary_data)
+    ev = m.evaluate(binary_data)
+    assert ev.accuracy > 0.93
+    assert ev.auc > 0.97
+    assert ev.loss < 0.2
+
+
+def test_gbt_regression_quality(regression_data):
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=150).train(
+            regression_data)
+    ev = m.evaluate(regression_data)
+    assert ev.rmse < 0.35
+
+
+def test_gbt_multiclass(binary_data):
+    rng = np.random.RandomState(2)
+    n = 5000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where(x1 > 0.5, "a", np.where(x2 > 0, "b", "c"))
+    d = {"x1": x1, "x2": x2, "label": y}
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=50).train(d)
+    ev = m.evaluate(d)
+    assert ev.accuracy > 0.98
+    p = m.predict(d)
+    assert p.shape == (n, 3)
+    np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-4)
+
+
+def test_gbt_early_stopping_truncates(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=300,
+                                        validation_ratio=0.15).train(
+                                            binary_data)
+    assert m.num_trees() <= 300
+    assert m.training_logs, "validation logs missing"
+
+
+def test_gbt_no_validation(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=20,
+                                        validation_ratio=0.0).train(
+                                            binary_data)
+    assert m.num_trees() == 20
+
+
+def test_gbt_subsample(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=60,
+                                        subsample=0.5).train(binary_data)
+    ev = m.evaluate(binary_data)
+    assert ev.accuracy > 0.9
+
+
+def test_rf_binary(binary_data):
+    m = ydf.RandomForestLearner(label="label", num_trees=50,
+                                max_depth=12).train(binary_data)
+    ev = m.evaluate(binary_data)
+    assert ev.accuracy > 0.93
+    p = m.predict(binary_data)
+    assert p.min() >= 0.0 and p.max() <= 1.0
+
+
+def test_rf_regression(regression_data):
+    m = ydf.RandomForestLearner(label="label", task=ydf.Task.REGRESSION,
+                                num_trees=50).train(regression_data)
+    ev = m.evaluate(regression_data)
+    assert ev.rmse < 1.0
+
+
+def test_rf_multiclass():
+    rng = np.random.RandomState(4)
+    n = 4000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where(x1 > 0.5, "a", np.where(x2 > 0, "b", "c"))
+    d = {"x1": x1, "x2": x2, "label": y}
+    m = ydf.RandomForestLearner(label="label", num_trees=30).train(d)
+    ev = m.evaluate(d)
+    assert ev.accuracy > 0.95
+
+
+def test_cart(binary_data):
+    m = ydf.CartLearner(label="label").train(binary_data)
+    assert m.num_trees() == 1
+    ev = m.evaluate(binary_data)
+    assert ev.accuracy > 0.9
+
+
+def test_isolation_forest():
+    rng = np.random.RandomState(7)
+    inliers = rng.randn(2000, 2)
+    outliers = rng.randn(50, 2) * 0.3 + 6.0
+    X = np.concatenate([inliers, outliers])
+    d = {"a": X[:, 0].astype(np.float32), "b": X[:, 1].astype(np.float32)}
+    m = ydf.IsolationForestLearner(num_trees=100).train(d)
+    s = m.predict(d)
+    assert s.shape == (2050,)
+    # outliers must score clearly higher
+    assert s[2000:].mean() > s[:2000].mean() + 0.1
+    # AUC of anomaly detection
+    from ydf_amd.metric.metric import roc_auc
+
+    labels = np.zeros(2050, dtype=bool)
+    labels[2000:] = True
+    assert roc_auc(labels, s) > 0.95
+
+
+def test_deterministic_same_seed(binary_data):
+    m1 = ydf.GradientBoostedTreesLearner(label="label", num_trees=10,
+                                         validation_ratio=0).train(
+                                             binary_data)
+    m2 = ydf.GradientBoostedTreesLearner(label="label", num_trees=10,
+                                         validation_ratio=0).train(
+                                             binary_data)
+    np.testing.assert_array_equal(m1.forest.feat, m2.forest.feat)
+    np.testing.assert_array_equal(m1.forest.thr, m2.forest.thr)
+
+
+def test_adult_gbt_quality(adult_paths):
+    pd = pytest.importorskip("pandas")
+    tr, te = adult_paths
+    m = ydf.GradientBoostedTreesLearner(label="income").train(
+        pd.read_csv(tr))
+    ev = m.evaluate(pd.read_csv(te))
+    # reference GBT reaches ~0.873; ordinal categorical encoding costs a
+    # few points for now (CART categorical splits tracked as follow-up)
+    assert ev.accuracy > 0.825
+    assert ev.auc > 0.85
+
+
+def test_feature_subset(binary_data):
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", features=["x1", "x2"], num_trees=20).train(
+            binary_data)
+    assert m.input_feature_names() == ["x1", "x2"]
+    m.predict({"x1": binary_data["x1"], "x2": binary_data["x2"]})

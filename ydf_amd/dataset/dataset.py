@@ -154,6 +154,18 @@ class VerticalDataset:
     def feature_names(self):
         return [c.name for c in self.dataspec.feature_columns]
 
+    def shard(self, lo: int, hi: int) -> "VerticalDataset":
+        """Row shard [lo, hi) sharing this dataset's dataspec — the unit of
+        data-parallel training (each rank trains on its shard; the dataspec,
+        hence the bin boundaries, must be shared so that all-reduced
+        histograms align)."""
+        return VerticalDataset(
+            X=np.ascontiguousarray(self.X[:, lo:hi]),
+            dataspec=self.dataspec,
+            label_values=None if self.label_values is None
+            else self.label_values[lo:hi].copy(),
+        )
+
 
 def create_vertical_dataset(
     data: InputData,

@@ -164,7 +164,17 @@ def numerical_boundaries(values: np.ndarray, max_bins: int = 256,
     if v.size > max_sample:
         rng = np.random.RandomState(1234)
         v = v[rng.randint(0, v.size, max_sample)]
-    qs = np.linspace(0.0, 1.0, max_bins + 1)[1:-1]
-    cuts = np.quantile(v, qs).astype(np.float64)
-    cuts = np.unique(cuts)
+    uniq = np.unique(v)
+    if uniq.size <= 1:
+        return np.zeros((0,), dtype=np.float32)
+    if uniq.size <= max_bins:
+        # few distinct values: exact midpoints (reference
+        # DISCRETIZED_NUMERICAL behavior for low-cardinality columns)
+        cuts = (uniq[1:].astype(np.float64)
+                + uniq[:-1].astype(np.float64)) / 2.0
+    else:
+        qs = np.linspace(0.0, 1.0, max_bins + 1)[1:-1]
+        cuts = np.quantile(v, qs).astype(np.float64)
+        cuts = np.unique(cuts)
+        cuts = cuts[cuts < float(uniq[-1])]  # a cut at max separates nothing
     return cuts.astype(np.float32)
