@@ -47,7 +47,6 @@ def build(verbose: bool = True) -> Path:
         "-O3",
         "-std=c++17",
         "-fPIC",
-        "-ffast-math",
         "-fvisibility=hidden",
         f"-I{CC_DIR}",
         *pybind_includes(),

@@ -19,6 +19,7 @@ from __future__ import annotations
 
 import dataclasses
 import math
+import os
 from typing import List, Optional
 
 import numpy as np
@@ -102,6 +103,11 @@ class ForestTrainer:
         self.node_ids = torch.empty(self.N, dtype=torch.int32, device=dev)
         self.hist = torch.empty((self.max_slots, self.F, n_bins, 3),
                                 dtype=torch.float32, device=dev)
+        # Histogram-subtraction trick (sibling = parent - smaller child):
+        # previous level's histograms, indexed by its slot order.
+        self.use_hist_sub = os.environ.get("YDFA_NO_HIST_SUB", "0") != "1"
+        self.hist_prev = torch.empty_like(self.hist) if self.use_hist_sub \
+            else None
         self.node_stats = torch.zeros((self.total_nodes, 3),
                                       dtype=torch.float32, device=dev)
         self.leaf_vals = torch.empty(self.total_nodes, dtype=torch.float32,
@@ -169,6 +175,10 @@ class ForestTrainer:
                                        device=self.device)))
 
         active_abs = np.array([0], dtype=np.int64)
+        # histogram-subtraction state from the previous level
+        prev_slot_of = None   # dict abs_node -> slot in self.hist_prev
+        prev_fit = False      # prev level fully resident in hist_prev
+        count_of = {}         # abs_node -> example count (all children)
         for level in range(cfg.max_depth):
             level_base = (1 << level) - 1
             level_size = 1 << level
@@ -177,25 +187,76 @@ class ForestTrainer:
                 break
             active_abs_t = torch.from_numpy(
                 active_abs.astype(np.int32)).to(self.device)
+            rel_t = torch.from_numpy(active_abs - level_base).to(self.device)
             slot_map = torch.full((level_size,), -1, dtype=torch.int32,
                                   device=self.device)
-            slot_map[torch.from_numpy(active_abs - level_base).to(
-                self.device)] = self.arange_buf[:n_active]
+            slot_map[rel_t] = self.arange_buf[:n_active]
             feat_mask = self._feat_mask(n_active, tree_idx, level)
+
+            # Histogram subtraction (reference-free optimization; standard
+            # GBT trick): build histograms only for the SMALLER child of
+            # each split, derive the sibling as parent - smaller. Only when
+            # both this and the previous level fit un-chunked.
+            use_sub = (self.use_hist_sub and prev_fit
+                       and prev_slot_of is not None
+                       and n_active <= self.max_slots)
+            derived = []  # (slot, parent_slot, sibling_slot)
+            if use_sub:
+                active_set = {int(a): s for s, a in enumerate(active_abs)}
+                build_rel = []
+                for s, a in enumerate(active_abs):
+                    a = int(a)
+                    sib = a + 1 if (a & 1) else a - 1
+                    if sib in active_set:
+                        ca, cs = count_of.get(a, 0), count_of.get(sib, 0)
+                        bigger = ca > cs or (ca == cs and not (a & 1))
+                        if bigger:
+                            derived.append((s, prev_slot_of[(a - 1) // 2],
+                                            active_set[sib]))
+                            continue
+                    build_rel.append(a - level_base)
+                if derived:
+                    build_map = torch.full((level_size,), -1,
+                                           dtype=torch.int32,
+                                           device=self.device)
+                    br = torch.from_numpy(
+                        np.asarray(build_rel, dtype=np.int64)).to(self.device)
+                    build_map[br] = slot_map[br]
+                else:
+                    build_map = slot_map
+            else:
+                build_map = slot_map
 
             for s0 in range(0, n_active, self.max_slots):
                 ns = min(self.max_slots, n_active - s0)
                 hist_view = self.hist[:ns]
                 hist_view.zero_()
-                ops.hist_build(self.bins, self.gh, self.node_ids, slot_map,
+                ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
                                hist_view, level_base, level_size, s0, ns)
                 self._allreduce(hist_view)
+                if derived and s0 == 0:
+                    d_idx = torch.tensor([d[0] for d in derived],
+                                         dtype=torch.int64,
+                                         device=self.device)
+                    p_idx = torch.tensor([d[1] for d in derived],
+                                         dtype=torch.int64,
+                                         device=self.device)
+                    s_idx = torch.tensor([d[2] for d in derived],
+                                         dtype=torch.int64,
+                                         device=self.device)
+                    hist_view[d_idx] = self.hist_prev[p_idx] - \
+                        hist_view[s_idx]
                 ops.split_scan(hist_view, active_abs_t, self.node_stats,
                                self.bg_nf, self.bb_nf, self.best_feat,
                                self.best_bin, self.best_gain, s0, ns,
                                cfg.lambda_l2, cfg.min_hessian,
                                cfg.min_examples, cfg.min_gain,
                                feat_mask=feat_mask)
+
+            prev_fit = n_active <= self.max_slots
+            if self.use_hist_sub and prev_fit and level + 1 < cfg.max_depth:
+                self.hist_prev[:n_active].copy_(self.hist[:n_active])
+                prev_slot_of = {int(a): s for s, a in enumerate(active_abs)}
 
             # record the level's splits into the complete-tree arrays
             idx64 = torch.from_numpy(active_abs).to(self.device)
@@ -212,6 +273,7 @@ class ForestTrainer:
                 split_abs = active_abs[bf >= 0]
                 if len(split_abs) == 0:
                     active_abs = np.array([], dtype=np.int64)
+                    prev_slot_of = None
                     continue
                 children = np.concatenate([2 * split_abs + 1,
                                            2 * split_abs + 2])
@@ -220,6 +282,8 @@ class ForestTrainer:
                                   2].cpu().numpy()
                 need = max(2 * cfg.min_examples, 2)
                 active_abs = np.sort(children[ccounts >= need])
+                count_of = {int(a): float(c)
+                            for a, c in zip(children, ccounts)}
 
         ops.leaf_values(self.node_stats, self.leaf_vals, cfg.lambda_l2)
         # .copy(): on CPU .cpu().numpy() aliases the (reused) buffers
