@@ -445,8 +445,10 @@ def train_rf(trainer: ForestTrainer, log=None):
     for it in range(cfg.num_trees):
         weights = None
         if cfg.bootstrap:
-            w = trainer.rng.poisson(
-                1.0, size=N).astype(np.float32)
+            # Poisson(1) bootstrap, clipped at 15 (P < 1e-12) — the packed
+            # u64 histogram path requires per-example h <= 16
+            w = np.minimum(trainer.rng.poisson(1.0, size=N), 15).astype(
+                np.float32)
             weights = torch.from_numpy(w).to(dev)
         for c in range(C):
             if multi:

@@ -28,6 +28,7 @@
 #include <hip/hip_runtime.h>
 #include <cstdint>
 #include <cstdio>
+#include <cstdlib>
 #include "common.h"
 
 namespace ydfa {
@@ -135,16 +136,27 @@ __global__ void weighted_target_kernel(const float* __restrict__ labels,
 // The count accumulates examples with h != 0 so that zero-weight (out-of-
 // bag) rows don't satisfy min_examples.
 // ---------------------------------------------------------------------------
-// Inner-loop design (measured on MI355X: the naive branchy form serializes
-// three dependent memory round-trips per row and runs ~6x slower):
+// Inner-loop design, measured on MI355X (tools/kernel_lab.hip):
+//   * 32-bit LDS atomics are the bottleneck of the classic 3x ds_add_f32
+//     histogram (~205 G atomics/s; the LDS array sits ~100% busy), while
+//     64-bit LDS atomics (ds_add_f64 / ds_add_u64) run ~5x faster per op
+//     (~955 G/s, at the load floor). So each bin accumulates with exactly
+//     TWO 64-bit atomics: g in f64 (exact), and {h fixed-point * 2^20,
+//     count << 44} packed in one u64 — 690 Gvisits/s vs 68 for the f32
+//     form (10x). Constraint: per-example h <= 16 and rows_per_block <=
+//     2^19 keep the h field below 2^44 (trainer clips bootstrap weights).
 //   * slot_map staged in LDS when the level fits (lds_map != 0), removing a
 //     dependent global load per (row, feature) visit;
 //   * rows processed kHistUnroll at a time with UNCONDITIONAL node/gh/bin
 //     loads issued back-to-back (gh/bins are always-valid full arrays), so
-//     the wave has kHistUnroll*3 loads in flight instead of 1;
-//   * the count add is predicated arithmetically (+0 for h==0 rows) instead
-//     of by a divergent branch.
+//     the wave has kHistUnroll*3 loads in flight instead of 1 (the naive
+//     branchy form serializes three dependent round-trips per row).
+// The merge unpacks to the f32 {sum_g, sum_h, count} global tensor the
+// scan/allreduce layers consume, skipping untouched bins.
 constexpr int kHistUnroll = 4;
+constexpr float kHScale = 1048576.0f;      // 2^20
+constexpr float kHInvScale = 1.0f / 1048576.0f;
+constexpr unsigned long long kHMask = (1ull << 44) - 1;
 
 __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
                                       const float2* __restrict__ gh,
@@ -154,12 +166,18 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
                                       int F, int n_bins, int level_base,
                                       int level_size, int slot0, int n_slots,
                                       int lds_map, int64_t rows_per_block) {
-  extern __shared__ __attribute__((aligned(16))) float lhist[];
-  // carve: [n_slots*n_bins*3] f32 hist, then [level_size] i32 slot map copy
-  const int tot = n_slots * n_bins * 3;
-  int* lmap = reinterpret_cast<int*>(lhist + tot);
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // carve: [n_slots*n_bins] {f64 g, u64 h|count}, then [level_size] i32 map
+  double* lg = reinterpret_cast<double*>(smem);
+  unsigned long long* lp =
+      reinterpret_cast<unsigned long long*>(smem) + 1;  // interleaved pairs
+  const int tot = n_slots * n_bins;
+  int* lmap = reinterpret_cast<int*>(smem + (size_t)tot * 16);
   const int f = blockIdx.x;
-  for (int i = threadIdx.x; i < tot; i += blockDim.x) lhist[i] = 0.f;
+  {
+    unsigned long long* z = reinterpret_cast<unsigned long long*>(smem);
+    for (int i = threadIdx.x; i < tot * 2; i += blockDim.x) z[i] = 0ull;
+  }
   if (lds_map) {
     for (int i = threadIdx.x; i < level_size; i += blockDim.x)
       lmap[i] = slot_map[i];
@@ -188,10 +206,12 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
       if (rel < 0 || rel >= level_size) continue;
       const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
       if (slot < 0 || slot >= n_slots) continue;
-      float* p = lhist + ((slot * n_bins + (int)b[u]) * 3);
-      atomicAdd(p, v[u].x);
-      atomicAdd(p + 1, v[u].y);
-      atomicAdd(p + 2, (v[u].y != 0.f) ? 1.0f : 0.0f);
+      const int cell = 2 * (slot * n_bins + (int)b[u]);
+      atomicAdd(lg + cell, (double)v[u].x);
+      const unsigned long long hq =
+          (unsigned long long)(v[u].y * kHScale + 0.5f);
+      atomicAdd(lp + cell,
+                hq | ((unsigned long long)(v[u].y != 0.f) << 44));
     }
   }
   for (; i < row1; i += stride) {
@@ -200,19 +220,25 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
     const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
     if (slot < 0 || slot >= n_slots) continue;
     const float2 v = gh[i];
-    float* p = lhist + ((slot * n_bins + (int)fb[i]) * 3);
-    atomicAdd(p, v.x);
-    atomicAdd(p + 1, v.y);
-    atomicAdd(p + 2, (v.y != 0.f) ? 1.0f : 0.0f);
+    const int cell = 2 * (slot * n_bins + (int)fb[i]);
+    atomicAdd(lg + cell, (double)v.x);
+    const unsigned long long hq =
+        (unsigned long long)(v.y * kHScale + 0.5f);
+    atomicAdd(lp + cell, hq | ((unsigned long long)(v.y != 0.f) << 44));
   }
   __syncthreads();
   for (int k = threadIdx.x; k < tot; k += blockDim.x) {
-    const float v = lhist[k];
-    if (v != 0.f) {
-      const int slot = k / (n_bins * 3);
-      const int rem = k - slot * (n_bins * 3);
-      atomicAdd(&hist[((int64_t)slot * F + f) * (n_bins * 3) + rem], v);
-    }
+    const double g = lg[2 * k];
+    const unsigned long long pk = lp[2 * k];
+    if (pk == 0ull && g == 0.0) continue;  // untouched bin
+    const int slot = k / n_bins;
+    const int bin = k - slot * n_bins;
+    float* p = hist + ((int64_t)slot * F + f) * (n_bins * 3) + bin * 3;
+    atomicAdd(p, (float)g);
+    // double-precision unpack: one rounding to f32 (keeps integer-valued
+    // h sums exact, which the CPU/GPU equality tests rely on)
+    atomicAdd(p + 1, (float)((double)(pk & kHMask) * (double)kHInvScale));
+    atomicAdd(p + 2, (float)(pk >> 44));
   }
 }
 
@@ -461,12 +487,28 @@ __global__ void binary_logloss_kernel(const float* __restrict__ preds,
 // Host launchers (raw pointers + explicit stream; exported via pybind).
 // ---------------------------------------------------------------------------
 static inline int row_chunks(int64_t N, int F, int max_blocks = 8192) {
-  // Enough blocks to fill 256 CUs x several waves, but bounded.
+  // Enough blocks to fill 256 CUs x several waves, but bounded (every block
+  // merges its LDS histogram, so block count is also merge traffic).
   int per_f = (int)((max_blocks + F - 1) / F);
   int64_t min_rows = 1024;
   int64_t max_chunks = (N + min_rows - 1) / min_rows;
   int chunks = (int)(max_chunks < per_f ? max_chunks : per_f);
-  return chunks < 1 ? 1 : chunks;
+  if (chunks < 1) chunks = 1;
+  // count/h-field packing requires rows_per_block <= 2^19
+  const int min_chunks = (int)((N + (1 << 19) - 1) >> 19);
+  if (chunks < min_chunks) chunks = min_chunks;
+  return chunks;
+}
+
+static inline int hist_block_threads() {
+  static int cached = 0;
+  if (cached == 0) {
+    const char* e = getenv("YDFA_HIST_BLOCK");
+    int v = e ? atoi(e) : 0;
+    if (v != 256 && v != 512 && v != 1024) v = 256;
+    cached = v;
+  }
+  return cached;
 }
 
 extern "C" {
@@ -515,20 +557,21 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
                     int level_size, int slot0, int n_slots, void* stream) {
   // Stage the level's slot map in LDS when it fits comfortably (removes a
   // dependent global load per row visit); cap at 32 KiB so the histogram
-  // region keeps >= ~48 slots.
+  // region keeps >= ~32 slots at 16 B/bin.
   const size_t map_bytes_full = (size_t)level_size * sizeof(int32_t);
   const int lds_map = map_bytes_full <= 32 * 1024 ? 1 : 0;
   const size_t budget = 160 * 1024 - (lds_map ? map_bytes_full : 0);
-  int max_lds_slots = (int)(budget / (n_bins * 3 * sizeof(float)));
+  int max_lds_slots = (int)(budget / ((size_t)n_bins * 16));
   if (max_lds_slots < 1) max_lds_slots = 1;
   const int group = n_slots < max_lds_slots ? n_slots : max_lds_slots;
-  const int chunks = row_chunks(N, F);
+  const int threads = hist_block_threads();
+  const int chunks = row_chunks(N, F, 8192 * 256 / threads);
   const int64_t rpb = (N + chunks - 1) / chunks;
   for (int s0 = 0; s0 < n_slots; s0 += group) {
     const int ng = (n_slots - s0) < group ? (n_slots - s0) : group;
-    const size_t lds = (size_t)ng * n_bins * 3 * sizeof(float) +
+    const size_t lds = (size_t)ng * n_bins * 16 +
                        (lds_map ? map_bytes_full : 0);
-    hipLaunchKernelGGL(hist_build_lds_kernel, dim3(F, chunks), dim3(kBlock),
+    hipLaunchKernelGGL(hist_build_lds_kernel, dim3(F, chunks), dim3(threads),
                        lds, (hipStream_t)stream, bins, (const float2*)gh,
                        node_ids, slot_map,
                        hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
