@@ -2,6 +2,8 @@
 import numpy as np
 import torch
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from ydf_amd import ops
 
 rng = np.random.RandomState(3)

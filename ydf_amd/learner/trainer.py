@@ -108,6 +108,20 @@ class ForestTrainer:
         self.use_hist_sub = os.environ.get("YDFA_NO_HIST_SUB", "0") != "1"
         self.hist_prev = torch.empty_like(self.hist) if self.use_hist_sub \
             else None
+        # dense mode: whole levels stay device-resident while the level fits
+        # the histogram buffer (largest power of two <= max_slots, capped at
+        # 64 nodes: beyond that the per-level host sync is cheaper than
+        # scanning mostly-empty slots). CPU has no sync cost to save, so it
+        # keeps the sparse path (dense level 0 only).
+        if self.device.type == "cuda":
+            self.dense_limit = min(
+                1 << int(math.floor(math.log2(self.max_slots))), 64)
+        else:
+            self.dense_limit = 1
+        self.build_map_buf = torch.empty(self.dense_limit, dtype=torch.int32,
+                                         device=dev)
+        self.derived_buf = torch.empty(self.dense_limit, dtype=torch.uint8,
+                                       device=dev)
         self.node_stats = torch.zeros((self.total_nodes, 3),
                                       dtype=torch.float32, device=dev)
         self.leaf_vals = torch.empty(self.total_nodes, dtype=torch.float32,
@@ -174,7 +188,8 @@ class ForestTrainer:
                             torch.full((), -1, dtype=torch.int32,
                                        device=self.device)))
 
-        active_abs = np.array([0], dtype=np.int64)
+        need = max(2 * cfg.min_examples, 2)
+        active_abs = None  # host open-node list (sparse levels only)
         # histogram-subtraction state from the previous level
         prev_slot_of = None   # dict abs_node -> slot in self.hist_prev
         prev_fit = False      # prev level fully resident in hist_prev
@@ -182,7 +197,37 @@ class ForestTrainer:
         for level in range(cfg.max_depth):
             level_base = (1 << level) - 1
             level_size = 1 << level
-            n_active = len(active_abs)
+
+            if level_size <= self.dense_limit:
+                # ---- dense mode: slot == level-relative node index; all
+                # planning (build/derive/prune) happens ON DEVICE, so there
+                # is no host sync per level (grow_tree syncs once at the
+                # end, or at the dense->sparse transition).
+                prev_fit = self._dense_level(tree_idx, level, need, prev_fit)
+                if (level + 1 < cfg.max_depth
+                        and (1 << (level + 1)) > self.dense_limit):
+                    # transition: materialize the host open-node list
+                    bf = self.best_feat[:level_size].cpu().numpy()
+                    split_abs = np.nonzero(bf >= 0)[0] + level_base
+                    if len(split_abs) == 0:
+                        active_abs = np.array([], dtype=np.int64)
+                        prev_slot_of = None
+                        continue
+                    children = np.concatenate([2 * split_abs + 1,
+                                               2 * split_abs + 2])
+                    ns_view = self.node_stats.view(-1, 3)
+                    ccounts = ns_view[
+                        torch.from_numpy(children).to(self.device),
+                        2].cpu().numpy()
+                    active_abs = np.sort(children[ccounts >= need])
+                    count_of = {int(a): float(c)
+                                for a, c in zip(children, ccounts)}
+                    prev_slot_of = {level_base + r: r
+                                    for r in range(level_size)}
+                continue
+
+            # ---- sparse mode (deep levels): host-managed open-node list
+            n_active = 0 if active_abs is None else len(active_abs)
             if n_active == 0:
                 break
             active_abs_t = torch.from_numpy(
@@ -294,6 +339,49 @@ class ForestTrainer:
             counts=self.node_stats[:, 2].cpu().numpy().copy(),
             max_depth=cfg.max_depth,
         )
+
+    def _dense_level(self, tree_idx: int, level: int, need: int,
+                     prev_fit: bool) -> bool:
+        """One level with device-resident planning (no host round-trips)."""
+        cfg = self.cfg
+        level_base = (1 << level) - 1
+        level_size = 1 << level
+        identity = self.arange_buf[:level_size]
+        abs_t = identity + level_base
+        use_sub = self.use_hist_sub and prev_fit and level > 0
+        if level == 0:
+            build_map = self.arange_buf[:1]  # root always built
+            derived = None
+        else:
+            ops.plan_level(self.node_stats, self.best_feat, level_base,
+                           level_size, need, use_sub, self.build_map_buf,
+                           self.derived_buf)
+            build_map = self.build_map_buf[:level_size]
+            derived = self.derived_buf[:level_size]
+        feat_mask = self._feat_mask(level_size, tree_idx, level)
+        hist_view = self.hist[:level_size]
+        hist_view.zero_()
+        ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
+                       hist_view, level_base, level_size, 0, level_size)
+        self._allreduce(hist_view)
+        if use_sub and derived is not None:
+            ops.subtract_hist(hist_view, self.hist_prev, derived, level_size)
+        ops.split_scan(hist_view, abs_t, self.node_stats, self.bg_nf,
+                       self.bb_nf, self.best_feat, self.best_bin,
+                       self.best_gain, 0, level_size, cfg.lambda_l2,
+                       cfg.min_hessian, cfg.min_examples, cfg.min_gain,
+                       feat_mask=feat_mask)
+        fits = True
+        if self.use_hist_sub and level + 1 < cfg.max_depth:
+            self.hist_prev[:level_size].copy_(hist_view)
+        self.tree_feat[level_base:level_base + level_size] = \
+            self.best_feat[:level_size]
+        self.tree_bin[level_base:level_base + level_size] = \
+            self.best_bin[:level_size]
+        ops.update_node_ids(self.bins, self.node_ids, identity,
+                            self.best_feat, self.best_bin, level_base,
+                            level_size)
+        return fits
 
     def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor):
         """Routes arbitrary rows through the latest tree (device arrays)."""

@@ -125,6 +125,30 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
         _C.cpu_split_scan(*args)
 
 
+def plan_level(node_stats: torch.Tensor, prev_best_feat: torch.Tensor,
+               level_base: int, level_size: int, need: int, use_sub: bool,
+               build_map: torch.Tensor, derived: torch.Tensor):
+    """Dense-mode device planning: build/derive/skip per level node."""
+    args = (node_stats.data_ptr(), prev_best_feat.data_ptr(), level_base,
+            level_size, need, 1 if use_sub else 0, build_map.data_ptr(),
+            derived.data_ptr())
+    if node_stats.is_cuda:
+        _C.gpu_plan_level(*args, _stream())
+    else:
+        _C.cpu_plan_level(*args)
+
+
+def subtract_hist(hist: torch.Tensor, hist_prev: torch.Tensor,
+                  derived: torch.Tensor, level_size: int):
+    F, n_bins = hist.shape[1], hist.shape[2]
+    args = (hist.data_ptr(), hist_prev.data_ptr(), derived.data_ptr(),
+            level_size, F, n_bins)
+    if hist.is_cuda:
+        _C.gpu_subtract_hist(*args, _stream())
+    else:
+        _C.cpu_subtract_hist(*args)
+
+
 def update_node_ids(bins: torch.Tensor, node_ids: torch.Tensor,
                     slot_map: torch.Tensor, best_feat: torch.Tensor,
                     best_bin: torch.Tensor, level_base: int, level_size: int):

@@ -26,6 +26,10 @@ void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
 void gpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*, int, int, int,
                     int, SplitParams, void*);
+void gpu_plan_level(const float*, const int32_t*, int, int, int, int,
+                    int32_t*, uint8_t*, void*);
+void gpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int,
+                       void*);
 void gpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, int64_t, int, int,
                          void*);
@@ -50,6 +54,9 @@ void cpu_weighted_target(const float*, const float*, float*, int64_t);
 void cpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*, int, int, int,
                     int, SplitParams);
+void cpu_plan_level(const float*, const int32_t*, int, int, int, int,
+                    int32_t*, uint8_t*);
+void cpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int);
 void cpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, int64_t, int, int);
 void cpu_leaf_values(const float*, float*, int, float);
@@ -137,6 +144,24 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          MakeSP(lambda_l2, min_hessian, min_examples,
                                 min_gain),
                          (void*)stream);
+        },
+        nogil);
+  m.def("gpu_plan_level",
+        [](uintptr_t node_stats, uintptr_t prev_best_feat, int level_base,
+           int level_size, int need, int use_sub, uintptr_t build_map,
+           uintptr_t derived, uintptr_t stream) {
+          gpu_plan_level(P<float>(node_stats), P<int32_t>(prev_best_feat),
+                         level_base, level_size, need, use_sub,
+                         P<int32_t>(build_map), P<uint8_t>(derived),
+                         (void*)stream);
+        },
+        nogil);
+  m.def("gpu_subtract_hist",
+        [](uintptr_t hist, uintptr_t hist_prev, uintptr_t derived,
+           int level_size, int F, int n_bins, uintptr_t stream) {
+          gpu_subtract_hist(P<float>(hist), P<float>(hist_prev),
+                            P<uint8_t>(derived), level_size, F, n_bins,
+                            (void*)stream);
         },
         nogil);
   m.def("gpu_update_node_ids",
@@ -238,6 +263,22 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          P<uint8_t>(feat_mask), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
                                 min_gain));
+        },
+        nogil);
+  m.def("cpu_plan_level",
+        [](uintptr_t node_stats, uintptr_t prev_best_feat, int level_base,
+           int level_size, int need, int use_sub, uintptr_t build_map,
+           uintptr_t derived) {
+          cpu_plan_level(P<float>(node_stats), P<int32_t>(prev_best_feat),
+                         level_base, level_size, need, use_sub,
+                         P<int32_t>(build_map), P<uint8_t>(derived));
+        },
+        nogil);
+  m.def("cpu_subtract_hist",
+        [](uintptr_t hist, uintptr_t hist_prev, uintptr_t derived,
+           int level_size, int F, int n_bins) {
+          cpu_subtract_hist(P<float>(hist), P<float>(hist_prev),
+                            P<uint8_t>(derived), level_size, F, n_bins);
         },
         nogil);
   m.def("cpu_update_node_ids",

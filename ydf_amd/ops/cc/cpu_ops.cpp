@@ -219,8 +219,10 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
     for (int b = 0; b < n_bins; ++b) {
       G += h0[b * 3]; H += h0[b * 3 + 1]; C += h0[b * 3 + 2];
     }
-    float* ns = node_stats + (int64_t)abs_node * 3;
-    ns[0] = G; ns[1] = H; ns[2] = C;
+    if (C > 0.f) {  // empty dense-mode slots must not zero parent-set stats
+      float* ns = node_stats + (int64_t)abs_node * 3;
+      ns[0] = G; ns[1] = H; ns[2] = C;
+    }
     const float parent_term = G * G / (H + sp.lambda_l2);
 
     float node_best_gain = -1e30f;
@@ -271,6 +273,38 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
     float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
     nl[0] = GL; nl[1] = HL; nl[2] = CL;
     nr[0] = G - GL; nr[1] = H - HL; nr[2] = C - CL;
+  });
+}
+
+// Dense-mode planning (see plan_level_kernel in train_kernels.hip).
+void cpu_plan_level(const float* node_stats, const int32_t* prev_best_feat,
+                    int level_base, int level_size, int need, int use_sub,
+                    int32_t* build_map, uint8_t* derived) {
+  for (int rel = 0; rel < level_size; ++rel) {
+    const int a = level_base + rel;
+    const int split = prev_best_feat[rel >> 1] >= 0;
+    const float ca = node_stats[(int64_t)a * 3 + 2];
+    const float cs = node_stats[(int64_t)(level_base + (rel ^ 1)) * 3 + 2];
+    const int elig_a = split && ca >= (float)need;
+    const int elig_s = split && cs >= (float)need;
+    const int is_right = (a & 1) == 0;
+    const int dv = use_sub && elig_a && elig_s &&
+                   (ca > cs || (ca == cs && is_right));
+    derived[rel] = (uint8_t)dv;
+    build_map[rel] = (elig_a && !dv) ? rel : -1;
+  }
+}
+
+void cpu_subtract_hist(float* hist, const float* hist_prev,
+                       const uint8_t* derived, int level_size, int F,
+                       int n_bins) {
+  const int64_t cells = (int64_t)F * n_bins * 3;
+  ThreadPool::Get().ParallelFor(level_size, [&](int rel) {
+    if (!derived[rel]) return;
+    float* dst = hist + (int64_t)rel * cells;
+    const float* par = hist_prev + (int64_t)(rel >> 1) * cells;
+    const float* sib = hist + (int64_t)(rel ^ 1) * cells;
+    for (int64_t k = 0; k < cells; ++k) dst[k] = par[k] - sib[k];
   });
 }
 

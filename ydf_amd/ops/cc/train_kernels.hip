@@ -275,7 +275,9 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
     __syncthreads();
   }
   const float G = sg[n_bins - 1], H = sh[n_bins - 1], C = sc[n_bins - 1];
-  if (f == 0 && b == 0) {
+  // C > 0 guard: empty slots (dense-mode nodes that were pruned) must not
+  // zero out node stats already written by their parent's split.
+  if (f == 0 && b == 0 && C > 0.f) {
     float* ns = node_stats + (int64_t)abs_of_slot[slot0 + slot] * 3;
     ns[0] = G; ns[1] = H; ns[2] = C;
   }
@@ -373,6 +375,52 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
   float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
   nl[0] = GL; nl[1] = HL; nl[2] = CL;
   nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
+}
+
+// ---------------------------------------------------------------------------
+// Dense-mode device planning: for every node of a level (slot == rel), decide
+// build/derive/skip from the previous level's splits and the children stats
+// already on device — no host round-trip per level. derived: sibling's
+// histogram will be parent - computed sibling (histogram-subtraction trick).
+//   build_map[rel] = rel if histogram must be built else -1
+//   derived[rel]   = 1 if hist[rel] = hist_prev[rel>>1] - hist[rel^1]
+// ---------------------------------------------------------------------------
+__global__ void plan_level_kernel(const float* __restrict__ node_stats,
+                                  const int32_t* __restrict__ prev_best_feat,
+                                  int level_base, int level_size, int need,
+                                  int use_sub, int32_t* __restrict__ build_map,
+                                  uint8_t* __restrict__ derived) {
+  const int rel = blockIdx.x * blockDim.x + threadIdx.x;
+  if (rel >= level_size) return;
+  const int a = level_base + rel;
+  const int split = prev_best_feat[rel >> 1] >= 0;
+  const float ca = node_stats[(int64_t)a * 3 + 2];
+  const int sib_rel = rel ^ 1;
+  const float cs = node_stats[(int64_t)(level_base + sib_rel) * 3 + 2];
+  const int elig_a = split && ca >= (float)need;
+  const int elig_s = split && cs >= (float)need;
+  // "a" is a LEFT child iff abs index is odd
+  const int is_right = (a & 1) == 0;
+  const int dv = use_sub && elig_a && elig_s &&
+                 (ca > cs || (ca == cs && is_right));
+  derived[rel] = (uint8_t)dv;
+  build_map[rel] = (elig_a && !dv) ? rel : -1;
+}
+
+// hist[r] = hist_prev[r>>1] - hist[r^1] for derived slots (dense mode).
+__global__ void subtract_hist_kernel(float* __restrict__ hist,
+                                     const float* __restrict__ hist_prev,
+                                     const uint8_t* __restrict__ derived,
+                                     int F, int n_bins) {
+  const int rel = blockIdx.x;
+  if (!derived[rel]) return;
+  const int64_t cells = (int64_t)F * n_bins * 3;
+  float* dst = hist + (int64_t)rel * cells;
+  const float* par = hist_prev + (int64_t)(rel >> 1) * cells;
+  const float* sib = hist + (int64_t)(rel ^ 1) * cells;
+  for (int64_t k = (int64_t)blockIdx.y * blockDim.x + threadIdx.x; k < cells;
+       k += (int64_t)gridDim.y * blockDim.x)
+    dst[k] = par[k] - sib[k];
 }
 
 // ---------------------------------------------------------------------------
@@ -505,7 +553,9 @@ static inline int hist_block_threads() {
   if (cached == 0) {
     const char* e = getenv("YDFA_HIST_BLOCK");
     int v = e ? atoi(e) : 0;
-    if (v != 256 && v != 512 && v != 1024) v = 256;
+    // 1024 threads/block measured fastest on MI355X (fewer blocks => less
+    // merge traffic; 16 waves/CU still hide the streaming latency)
+    if (v != 256 && v != 512 && v != 1024) v = 1024;
     cached = v;
   }
   return cached;
@@ -593,6 +643,26 @@ void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                      (hipStream_t)stream, hist, abs_of_slot, best_gain_nf,
                      best_bin_nf, node_stats, best_feat, best_bin, best_gain,
                      F, n_bins, slot0, sp);
+}
+
+void gpu_plan_level(const float* node_stats, const int32_t* prev_best_feat,
+                    int level_base, int level_size, int need, int use_sub,
+                    int32_t* build_map, uint8_t* derived, void* stream) {
+  const int grid = (level_size + kBlock - 1) / kBlock;
+  hipLaunchKernelGGL(plan_level_kernel, dim3(grid), dim3(kBlock), 0,
+                     (hipStream_t)stream, node_stats, prev_best_feat,
+                     level_base, level_size, need, use_sub, build_map,
+                     derived);
+}
+
+void gpu_subtract_hist(float* hist, const float* hist_prev,
+                       const uint8_t* derived, int level_size, int F,
+                       int n_bins, void* stream) {
+  int gy = (int)(((int64_t)F * n_bins * 3 + kBlock - 1) / kBlock);
+  if (gy > 64) gy = 64;
+  hipLaunchKernelGGL(subtract_hist_kernel, dim3(level_size, gy), dim3(kBlock),
+                     0, (hipStream_t)stream, hist, hist_prev, derived, F,
+                     n_bins);
 }
 
 void gpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
