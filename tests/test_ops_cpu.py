@@ -80,12 +80,13 @@ def test_hist_build_skips_zero_weight_counts():
 
 
 def _brute_force_best_split(hist_nf, lam, min_ex, min_h):
-    """Reference scan: hist_nf [F, B, 3] -> (gain, feat, bin)."""
+    """Reference scan: hist_nf [F, B, 3] -> (gain, feat, bin). Totals are
+    per-feature (identical across features for any real histogram)."""
     F, B, _ = hist_nf.shape
-    G, H, C = hist_nf[0].sum(axis=0)
-    parent = G * G / (H + lam)
     best = (-np.inf, -1, 0)
     for f in range(F):
+        G, H, C = hist_nf[f].sum(axis=0)
+        parent = G * G / (H + lam)
         cg = np.cumsum(hist_nf[f, :, 0])
         ch = np.cumsum(hist_nf[f, :, 1])
         cc = np.cumsum(hist_nf[f, :, 2])

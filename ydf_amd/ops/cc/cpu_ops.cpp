@@ -223,8 +223,6 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
       float* ns = node_stats + (int64_t)abs_node * 3;
       ns[0] = G; ns[1] = H; ns[2] = C;
     }
-    const float parent_term = G * G / (H + sp.lambda_l2);
-
     float node_best_gain = -1e30f;
     int node_best_f = -1, node_best_b = 0;
     for (int f = 0; f < F; ++f) {
@@ -234,16 +232,23 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
         continue;
       }
       const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+      // Per-feature totals (matches the GPU kernel exactly; identical to
+      // the feature-0 totals for any real histogram).
+      float Gf = 0.f, Hf = 0.f, Cf = 0.f;
+      for (int b = 0; b < n_bins; ++b) {
+        Gf += hp[b * 3]; Hf += hp[b * 3 + 1]; Cf += hp[b * 3 + 2];
+      }
+      const float pterm = Gf * Gf / (Hf + sp.lambda_l2);
       float GL = 0.f, HL = 0.f, CL = 0.f;
       float fbest = -1e30f;
       int fbin = 0;
       for (int b = 0; b < n_bins - 1; ++b) {
         GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
-        const float GR = G - GL, HR = H - HL, CR = C - CL;
+        const float GR = Gf - GL, HR = Hf - HL, CR = Cf - CL;
         if (CL >= sp.min_examples && CR >= sp.min_examples &&
             HL >= sp.min_hessian && HR >= sp.min_hessian) {
           const float gain = GL * GL / (HL + sp.lambda_l2) +
-                             GR * GR / (HR + sp.lambda_l2) - parent_term;
+                             GR * GR / (HR + sp.lambda_l2) - pterm;
           if (gain > fbest) { fbest = gain; fbin = b; }
         }
       }
