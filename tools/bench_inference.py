@@ -1,0 +1,103 @@
+#!/usr/bin/env python3
+"""Serving benchmark (BASELINE config #5): batched inference of a
+1000-tree depth-6 GBT over 10M synthetic rows on 1 MI355X.
+
+Random-structure trees (as BASELINE.json specifies for the inference
+bench) + synthetic feature matrix; measures examples/sec end-to-end
+through the flat-forest HIP kernel (capability analogue of the reference's
+cli/benchmark_inference.cc + serving/decision_forest engines).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from ydf_amd import ops  # noqa: E402
+
+
+def random_forest_struct(n_trees: int, depth: int, n_features: int,
+                         rng: np.random.RandomState):
+    """Complete binary trees with random features/thresholds/leaves."""
+    n_nodes = (1 << (depth + 1)) - 1
+    n_internal = (1 << depth) - 1
+    feat = np.empty(n_trees * n_nodes, dtype=np.int32)
+    thr = np.empty(n_trees * n_nodes, dtype=np.float32)
+    left = np.zeros(n_trees * n_nodes, dtype=np.int32)
+    roots = np.arange(n_trees, dtype=np.int32) * n_nodes
+    for t in range(n_trees):
+        base = t * n_nodes
+        feat[base:base + n_internal] = rng.randint(0, n_features, n_internal)
+        feat[base + n_internal:base + n_nodes] = -1
+        thr[base:base + n_internal] = rng.randn(n_internal)
+        thr[base + n_internal:base + n_nodes] = \
+            rng.randn(n_nodes - n_internal) * 0.1
+        # BFS complete layout: left child of local node k is 2k+1
+        k = np.arange(n_internal)
+        left[base + k] = base + 2 * k + 1
+    return feat, thr, left, roots
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--rows", type=int, default=10_000_000)
+    ap.add_argument("--trees", type=int, default=1000)
+    ap.add_argument("--depth", type=int, default=6)
+    ap.add_argument("--features", type=int, default=28)
+    ap.add_argument("--runs", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--device", default=None)
+    args = ap.parse_args()
+    device = torch.device(args.device) if args.device else (
+        torch.device("cuda") if torch.cuda.is_available()
+        else torch.device("cpu"))
+
+    rng = np.random.RandomState(7)
+    feat, thr, left, roots = random_forest_struct(
+        args.trees, args.depth, args.features, rng)
+    g = torch.Generator(device=device)
+    g.manual_seed(42)
+    X = torch.randn((args.features, args.rows), generator=g, device=device)
+    featd = torch.from_numpy(feat).to(device)
+    thrd = torch.from_numpy(thr).to(device)
+    leftd = torch.from_numpy(left).to(device)
+    rootsd = torch.from_numpy(roots).to(device)
+    out = torch.empty(args.rows, dtype=torch.float32, device=device)
+    act = torch.empty_like(out)
+
+    def run():
+        ops.predict_forest(X, featd, thrd, leftd, rootsd, out)
+        ops.sigmoid(out, act)
+
+    for _ in range(args.warmup):
+        run()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.runs):
+        run()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / args.runs
+    print(json.dumps({
+        "metric": "gbt1000_d6_inference_examples_per_s",
+        "value": args.rows / dt,
+        "unit": "examples/s",
+        "rows": args.rows,
+        "trees": args.trees,
+        "depth": args.depth,
+        "features": args.features,
+        "ms_per_batch": dt * 1000,
+        "us_per_example": dt / args.rows * 1e6,
+        "device": str(device),
+        "data": "synthetic",
+    }))
+
+
+if __name__ == "__main__":
+    main()

@@ -159,10 +159,11 @@ class ForestTrainer:
         # decision_tree/training.cc num_candidate_attributes).
         rs = np.random.RandomState(
             (self.cfg.seed * 1000003 + tree_idx * 8191 + level) % (1 << 31))
+        # vectorized k-of-F sample per node: rank random keys per row
+        keys = rs.random_sample((n_active, self.F))
+        idx = np.argpartition(keys, k - 1, axis=1)[:, :k]
         mask = np.zeros((n_active, self.F), dtype=np.uint8)
-        for s in range(n_active):
-            idx = rs.choice(self.F, size=k, replace=False)
-            mask[s, idx] = 1
+        np.put_along_axis(mask, idx, 1, axis=1)
         return torch.from_numpy(mask).to(self.device)
 
     # -- one tree ---------------------------------------------------------

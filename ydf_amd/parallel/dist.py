@@ -24,6 +24,8 @@ def init_from_env(backend: str = None) -> int:
         return 0
     if not dist.is_initialized():
         if backend is None:
+            backend = os.environ.get("YDFA_DIST_BACKEND")
+        if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         if backend == "nccl":
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
