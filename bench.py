@@ -72,7 +72,10 @@ def main():
     if args.device:
         device = torch.device(args.device)
     elif torch.cuda.is_available():
-        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", "0")))
+        device = torch.device(
+            "cuda",
+            int(os.environ.get("LOCAL_RANK", "0"))
+            % torch.cuda.device_count())
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")

@@ -154,12 +154,21 @@ class ForestTrainer:
         k = self.cfg.num_candidate_features
         if k <= 0 or k >= self.F:
             return None
-        # Generated on host so every data-parallel rank draws the SAME mask
-        # (the reference samples candidate attributes per node too,
-        # decision_tree/training.cc num_candidate_attributes).
-        rs = np.random.RandomState(
-            (self.cfg.seed * 1000003 + tree_idx * 8191 + level) % (1 << 31))
-        # vectorized k-of-F sample per node: rank random keys per row
+        # Per-node k-of-F candidate sampling (reference
+        # num_candidate_attributes, decision_tree/training.cc). Seeded per
+        # (tree, level) so every data-parallel rank draws the SAME mask.
+        seed = (self.cfg.seed * 1000003 + tree_idx * 8191 + level) % (1 << 31)
+        if self.device.type == "cuda":
+            g = torch.Generator(device=self.device)
+            g.manual_seed(seed)
+            keys = torch.rand((n_active, self.F), generator=g,
+                              device=self.device)
+            idx = torch.topk(keys, k, dim=1, largest=False).indices
+            mask = torch.zeros((n_active, self.F), dtype=torch.uint8,
+                               device=self.device)
+            mask.scatter_(1, idx, 1)
+            return mask
+        rs = np.random.RandomState(seed)
         keys = rs.random_sample((n_active, self.F))
         idx = np.argpartition(keys, k - 1, axis=1)[:, :k]
         mask = np.zeros((n_active, self.F), dtype=np.uint8)
@@ -536,9 +545,15 @@ def train_rf(trainer: ForestTrainer, log=None):
         if cfg.bootstrap:
             # Poisson(1) bootstrap, clipped at 15 (P < 1e-12) — the packed
             # u64 histogram path requires per-example h <= 16
-            w = np.minimum(trainer.rng.poisson(1.0, size=N), 15).astype(
-                np.float32)
-            weights = torch.from_numpy(w).to(dev)
+            if dev.type == "cuda":
+                g = torch.Generator(device=dev)
+                g.manual_seed((cfg.seed * 31337 + it) % (1 << 31))
+                weights = torch.poisson(
+                    torch.ones(N, device=dev), generator=g).clamp_(max=15)
+            else:
+                w = np.minimum(trainer.rng.poisson(1.0, size=N), 15).astype(
+                    np.float32)
+                weights = torch.from_numpy(w).to(dev)
         for c in range(C):
             if multi:
                 if onehot is None:
