@@ -42,9 +42,32 @@ class GenericLearner:
         return torch.device("cuda") if torch.cuda.is_available() else \
             torch.device("cpu")
 
+    def _cat_feature_flags(self, ds: VerticalDataset) -> np.ndarray:
+        """bool [F]: which features use categorical set-splits (CATEGORICAL
+        semantic; categories beyond 255 share the last bin)."""
+        return np.asarray(
+            [c.semantic == Semantic.CATEGORICAL
+             for c in ds.dataspec.feature_columns], dtype=bool)
+
+    def _bin_matrix(self, ds_X: np.ndarray, cat_feats: np.ndarray,
+                    bnd: np.ndarray, device: torch.device) -> torch.Tensor:
+        """Bins numericals by quantile cuts; categorical codes pass through
+        as their own bin index (clamped to 255)."""
+        X = torch.from_numpy(np.ascontiguousarray(ds_X)).to(device)
+        bnd_t = torch.from_numpy(bnd).to(device)
+        bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
+        ops.bin_data(X, bnd_t, bins)
+        ci = np.nonzero(cat_feats)[0]
+        if ci.size:
+            idx = torch.from_numpy(ci).to(device)
+            bins[idx] = X[idx].clamp_(0, 255).to(torch.uint8)
+        del X
+        return bins
+
     def _prepare(self, data, device: torch.device):
         """Dataset -> (VerticalDataset, binned u8 [F,N] on device,
-        labels f32 [N] on device, padded boundary matrix np [F,n_cuts])."""
+        labels f32 [N] on device, padded boundary matrix np [F,n_cuts],
+        cat_flags u8 tensor or None)."""
         if isinstance(data, VerticalDataset):
             ds = data
         else:
@@ -53,16 +76,17 @@ class GenericLearner:
                 features=self.features, max_vocab_count=self.max_vocab_count,
                 min_vocab_frequency=self.min_vocab_frequency)
         bnd = padded_boundaries(ds.dataspec.feature_columns)
-        X = torch.from_numpy(np.ascontiguousarray(ds.X)).to(device)
-        bnd_t = torch.from_numpy(bnd).to(device)
-        bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
-        ops.bin_data(X, bnd_t, bins)
-        del X
+        cat_feats = self._cat_feature_flags(ds)
+        bins = self._bin_matrix(ds.X, cat_feats, bnd, device)
         labels = None
         if ds.label_values is not None:
             labels = torch.from_numpy(
                 np.ascontiguousarray(ds.label_values)).to(device)
-        return ds, bins, labels, bnd
+        cat_flags = None
+        if cat_feats.any():
+            cat_flags = torch.from_numpy(
+                cat_feats.astype(np.uint8)).to(device)
+        return ds, bins, labels, bnd, cat_flags
 
     def _label_classes(self, ds: VerticalDataset):
         if self._task != Task.CLASSIFICATION:

@@ -47,6 +47,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  num_candidate_attributes_ratio: float = -1.0,
                  use_hessian_gain: bool = True,
                  apply_link_function: bool = True,
+                 l2_categorical_regularization: float = 1.0,
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
@@ -62,13 +63,14 @@ class GradientBoostedTreesLearner(GenericLearner):
             num_candidate_attributes_ratio=num_candidate_attributes_ratio,
             use_hessian_gain=use_hessian_gain,
             apply_link_function=apply_link_function,
+            l2_categorical_regularization=l2_categorical_regularization,
         )
 
     def train(self, data, valid=None, verbose=None
               ) -> GradientBoostedTreesModel:
         hp = self.hyperparameters
         device = self._resolve_device()
-        ds, bins, labels, bnd = self._prepare(data, device)
+        ds, bins, labels, bnd, cat_flags = self._prepare(data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
         classes = self._label_classes(ds)
@@ -87,8 +89,7 @@ class GradientBoostedTreesLearner(GenericLearner):
         valid_bins = valid_labels = None
         vr = hp["validation_ratio"]
         if valid is not None:
-            vds, valid_bins, valid_labels, _ = self._prepare_valid(
-                valid, ds, device)
+            valid_bins, valid_labels = self._prepare_valid(valid, ds, device)
         elif vr > 0.0 and hp["early_stopping"] != "NONE":
             N = bins.shape[1]
             rng = np.random.RandomState(self.random_seed)
@@ -120,12 +121,15 @@ class GradientBoostedTreesLearner(GenericLearner):
                 hp["early_stopping_num_trees_look_ahead"]),
             early_stopping_initial_iteration=(
                 hp["early_stopping_initial_iteration"]),
+            cat_smooth=hp["l2_categorical_regularization"],
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
                                       valid_bins=valid_bins,
-                                      valid_labels=valid_labels)
+                                      valid_labels=valid_labels,
+                                      cat_flags=cat_flags)
         trees, init_preds, logs = trainer_lib.train_gbt(t, log=info)
-        flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"])
+        flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"],
+                                 cat_feats=self._cat_feature_flags(ds))
         C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
         activation = "identity"
         if hp["apply_link_function"]:
@@ -143,20 +147,15 @@ class GradientBoostedTreesLearner(GenericLearner):
     def _prepare_valid(self, valid, train_ds: VerticalDataset, device):
         """Bins a user-provided validation dataset with the TRAIN dataspec."""
         from ydf_amd.dataset.dataset import create_vertical_dataset
+        from ydf_amd.model.forest import padded_boundaries
 
         vds = create_vertical_dataset(valid, dataspec=train_ds.dataspec)
-        bnd = None
-        from ydf_amd.model.forest import padded_boundaries
-        from ydf_amd import ops
-
         bnd = padded_boundaries(vds.dataspec.feature_columns)
-        X = torch.from_numpy(np.ascontiguousarray(vds.X)).to(device)
-        bnd_t = torch.from_numpy(bnd).to(device)
-        bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
-        ops.bin_data(X, bnd_t, bins)
+        bins = self._bin_matrix(vds.X, self._cat_feature_flags(vds), bnd,
+                                device)
         labels = torch.from_numpy(
             np.ascontiguousarray(vds.label_values)).to(device)
-        return vds, bins, labels, bnd
+        return bins, labels
 
 
 class RandomForestLearner(GenericLearner):
@@ -212,7 +211,7 @@ class RandomForestLearner(GenericLearner):
     def train(self, data, valid=None, verbose=None) -> RandomForestModel:
         hp = self.hyperparameters
         device = self._resolve_device()
-        ds, bins, labels, bnd = self._prepare(data, device)
+        ds, bins, labels, bnd, cat_flags = self._prepare(data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
         classes = self._label_classes(ds) \
@@ -227,9 +226,11 @@ class RandomForestLearner(GenericLearner):
             bootstrap=hp["bootstrap_training_dataset"],
             num_candidate_features=self._num_candidate(F),
         )
-        t = trainer_lib.ForestTrainer(bins, labels, cfg)
+        t = trainer_lib.ForestTrainer(bins, labels, cfg,
+                                      cat_flags=cat_flags)
         trees = trainer_lib.train_rf(t, log=info)
-        flat = build_flat_forest(trees, bnd, leaf_scale=1.0)
+        flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
+                                 cat_feats=self._cat_feature_flags(ds))
         C = n_classes if (classes and n_classes > 2) else 1
         model = RandomForestModel(
             forest=flat, dataspec=ds.dataspec, task=self._task,

@@ -44,6 +44,7 @@ class TrainerConfig:
     min_hessian: float = 1e-3
     min_gain: float = 0.0
     subsample: float = 1.0
+    cat_smooth: float = 1.0      # l2_categorical_regularization
     n_classes: int = 2           # multinomial only
     seed: int = 123456
     # RF-specific
@@ -62,10 +63,11 @@ class HostTree:
     """One trained tree copied to host, complete-array form."""
 
     feat: np.ndarray        # [total_nodes] i32, -1 = leaf/unused
-    bin: np.ndarray         # [total_nodes] i32 (split bin)
+    bin: np.ndarray         # [total_nodes] i32 (split bin / sorted rank)
     leaf_value: np.ndarray  # [total_nodes] f32 (unscaled -G/(H+l2))
     counts: np.ndarray      # [total_nodes] f32
     max_depth: int
+    masks: Optional[np.ndarray] = None  # [total_nodes,4] u64 (cat splits)
 
 
 def _dist_ok() -> bool:
@@ -80,11 +82,16 @@ class ForestTrainer:
     def __init__(self, bins: torch.Tensor, labels: torch.Tensor,
                  cfg: TrainerConfig,
                  valid_bins: Optional[torch.Tensor] = None,
-                 valid_labels: Optional[torch.Tensor] = None):
+                 valid_labels: Optional[torch.Tensor] = None,
+                 cat_flags: Optional[torch.Tensor] = None):
         assert bins.dtype == torch.uint8 and bins.dim() == 2
         self.bins = bins
         self.labels = labels
         self.cfg = cfg
+        self.cat_flags = cat_flags  # u8 [F] on device; None = all numerical
+        self.has_cats = cat_flags is not None and bool(cat_flags.any())
+        if not self.has_cats:
+            self.cat_flags = None
         self.device = bins.device
         self.F, self.N = bins.shape
         self.valid_bins = valid_bins
@@ -130,6 +137,10 @@ class ForestTrainer:
                                      device=dev)
         self.tree_bin = torch.empty(self.total_nodes, dtype=torch.int32,
                                     device=dev)
+        # 256-bit "category goes right" masks, one per node (int64 bit-pattern)
+        self.tree_masks = torch.zeros((self.total_nodes, 4),
+                                      dtype=torch.int64, device=dev) \
+            if self.has_cats else None
         self.bg_nf = torch.empty((self.max_slots, self.F),
                                  dtype=torch.float32, device=dev)
         self.bb_nf = torch.empty((self.max_slots, self.F), dtype=torch.int32,
@@ -187,6 +198,8 @@ class ForestTrainer:
         self.tree_feat.fill_(-1)
         self.tree_bin.zero_()
         self.node_stats.zero_()
+        if self.tree_masks is not None:
+            self.tree_masks.zero_()
         if sample_mask is None:
             self.node_ids.zero_()
         else:
@@ -306,7 +319,9 @@ class ForestTrainer:
                                self.best_bin, self.best_gain, s0, ns,
                                cfg.lambda_l2, cfg.min_hessian,
                                cfg.min_examples, cfg.min_gain,
-                               feat_mask=feat_mask)
+                               feat_mask=feat_mask, cat_flags=self.cat_flags,
+                               masks=self.tree_masks,
+                               cat_smooth=cfg.cat_smooth)
 
             prev_fit = n_active <= self.max_slots
             if self.use_hist_sub and prev_fit and level + 1 < cfg.max_depth:
@@ -319,7 +334,8 @@ class ForestTrainer:
             self.tree_bin[idx64] = self.best_bin[:n_active]
             ops.update_node_ids(self.bins, self.node_ids, slot_map,
                                 self.best_feat, self.best_bin, level_base,
-                                level_size)
+                                level_size, cat_flags=self.cat_flags,
+                                masks=self.tree_masks)
 
             if level + 1 < cfg.max_depth:
                 # choose next level's open nodes (host sync; deterministic
@@ -348,6 +364,8 @@ class ForestTrainer:
             leaf_value=self.leaf_vals.cpu().numpy().copy(),
             counts=self.node_stats[:, 2].cpu().numpy().copy(),
             max_depth=cfg.max_depth,
+            masks=self.tree_masks.cpu().numpy().view(np.uint64).copy()
+            if self.tree_masks is not None else None,
         )
 
     def _dense_level(self, tree_idx: int, level: int, need: int,
@@ -380,7 +398,8 @@ class ForestTrainer:
                        self.bb_nf, self.best_feat, self.best_bin,
                        self.best_gain, 0, level_size, cfg.lambda_l2,
                        cfg.min_hessian, cfg.min_examples, cfg.min_gain,
-                       feat_mask=feat_mask)
+                       feat_mask=feat_mask, cat_flags=self.cat_flags,
+                       masks=self.tree_masks, cat_smooth=cfg.cat_smooth)
         fits = True
         if self.use_hist_sub and level + 1 < cfg.max_depth:
             self.hist_prev[:level_size].copy_(hist_view)
@@ -390,7 +409,8 @@ class ForestTrainer:
             self.best_bin[:level_size]
         ops.update_node_ids(self.bins, self.node_ids, identity,
                             self.best_feat, self.best_bin, level_base,
-                            level_size)
+                            level_size, cat_flags=self.cat_flags,
+                            masks=self.tree_masks)
         return fits
 
     def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor):
@@ -403,7 +423,8 @@ class ForestTrainer:
                 bins, node_ids, self.arange_buf[:level_size],
                 self.tree_feat[level_base:level_base + level_size],
                 self.tree_bin[level_base:level_base + level_size],
-                level_base, level_size)
+                level_base, level_size, cat_flags=self.cat_flags,
+                masks=self.tree_masks)
 
 
 def train_gbt(trainer: ForestTrainer, log=None):

@@ -24,6 +24,20 @@ class FlatForest:
     thr: np.ndarray    # f32 [total]
     left: np.ndarray   # i32 [total]
     roots: np.ndarray  # i32 [n_trees]
+    # categorical set-splits: cat_idx[n] = -1 (numerical/leaf) or an index
+    # into masks (256-bit "category goes right" bitmask as 4 x u64)
+    cat_idx: np.ndarray = None   # i32 [total]
+    masks: np.ndarray = None     # u64 [n_masks, 4]
+
+    def __post_init__(self):
+        if self.cat_idx is None:
+            self.cat_idx = np.full(len(self.feat), -1, dtype=np.int32)
+        if self.masks is None:
+            self.masks = np.zeros((0, 4), dtype=np.uint64)
+
+    @property
+    def has_cats(self) -> bool:
+        return len(self.masks) > 0
 
     @property
     def n_trees(self) -> int:
@@ -58,11 +72,13 @@ def _reachable(feat: np.ndarray, max_depth: int):
 
 
 def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
-                      leaf_scale: float = 1.0):
-    """Compacts a complete-array HostTree into flat (feat, thr, left) arrays.
+                      leaf_scale: float = 1.0, cat_feats=None):
+    """Compacts a complete-array HostTree into flat (feat, thr, left,
+    cat_idx, masks) arrays.
 
     boundaries: padded [F, n_cuts] cut matrix; split threshold = cut[bin]
-    (binning guarantees "bin > b" <=> "x > cut[b]").
+    (binning guarantees "bin > b" <=> "x > cut[b]"). cat_feats: bool [F]
+    marking categorical features (their conditions use tree.masks).
     """
     reach, internal = _reachable(tree.feat, tree.max_depth)
     nodes = np.nonzero(reach)[0]  # ascending = level order
@@ -74,31 +90,54 @@ def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
     lchild = np.minimum(2 * nodes + 1, total + 1)
     left = np.where(n_int, new_idx[lchild], 0).astype(np.int32)
     thr = np.empty(len(nodes), dtype=np.float32)
+    cat_idx = np.full(len(nodes), -1, dtype=np.int32)
+    masks = np.zeros((0, 4), dtype=np.uint64)
     ii = np.nonzero(n_int)[0]
     if ii.size:
-        thr[ii] = boundaries[tree.feat[nodes[ii]], tree.bin[nodes[ii]]]
+        is_cat_node = np.zeros(len(nodes), dtype=bool)
+        if cat_feats is not None and tree.masks is not None:
+            is_cat_node[ii] = cat_feats[tree.feat[nodes[ii]]]
+        ni = np.nonzero(n_int & ~is_cat_node)[0]
+        if ni.size:
+            # numerical: tree.bin is a cut index; categorical nodes store a
+            # sorted RANK there instead, so they must not index boundaries
+            thr[ni] = boundaries[tree.feat[nodes[ni]], tree.bin[nodes[ni]]]
+        ci = np.nonzero(is_cat_node)[0]
+        if ci.size:
+            thr[ci] = 0.0
+            cat_idx[ci] = np.arange(ci.size, dtype=np.int32)
+            masks = tree.masks[nodes[ci]].astype(np.uint64)
     li = np.nonzero(~n_int)[0]
     thr[li] = tree.leaf_value[nodes[li]] * leaf_scale
-    return feat, thr, left
+    return feat, thr, left, cat_idx, masks
 
 
 def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
-                      leaf_scale: float = 1.0) -> FlatForest:
-    feats, thrs, lefts, roots = [], [], [], []
+                      leaf_scale: float = 1.0, cat_feats=None) -> FlatForest:
+    feats, thrs, lefts, roots, cidxs, mask_list = [], [], [], [], [], []
     off = 0
+    mask_off = 0
     for t in trees:
-        f, th, lf = host_tree_to_flat(t, boundaries, leaf_scale)
+        f, th, lf, ci, mk = host_tree_to_flat(t, boundaries, leaf_scale,
+                                              cat_feats)
         lf = np.where(f >= 0, lf + off, 0)
+        ci = np.where(ci >= 0, ci + mask_off, -1)
         roots.append(off)
         off += len(f)
+        mask_off += len(mk)
         feats.append(f)
         thrs.append(th)
         lefts.append(lf)
+        cidxs.append(ci)
+        mask_list.append(mk)
     return FlatForest(
         feat=np.concatenate(feats) if feats else np.zeros(0, np.int32),
         thr=np.concatenate(thrs) if thrs else np.zeros(0, np.float32),
         left=np.concatenate(lefts) if lefts else np.zeros(0, np.int32),
         roots=np.asarray(roots, dtype=np.int32),
+        cat_idx=np.concatenate(cidxs) if cidxs else np.zeros(0, np.int32),
+        masks=np.concatenate(mask_list) if mask_list
+        else np.zeros((0, 4), np.uint64),
     )
 
 

@@ -43,6 +43,13 @@ class _DeviceForest:
         self.thr = torch.from_numpy(forest.thr).to(device)
         self.left = torch.from_numpy(forest.left).to(device)
         self.roots = torch.from_numpy(forest.roots).to(device)
+        if forest.has_cats:
+            self.cat_idx = torch.from_numpy(forest.cat_idx).to(device)
+            self.masks = torch.from_numpy(
+                forest.masks.view(np.int64)).to(device)
+        else:
+            self.cat_idx = None
+            self.masks = None
 
 
 class GenericModel:
@@ -136,7 +143,8 @@ class GenericModel:
                                init=float(self.init_predictions[c]
                                           if c < len(self.init_predictions)
                                           else self.init_predictions[0]),
-                               scale=self._leaf_scale())
+                               scale=self._leaf_scale(),
+                               cat_idx=df.cat_idx, masks=df.masks)
         return out
 
     def predict(self, data, device=None) -> np.ndarray:
@@ -266,7 +274,8 @@ class GenericModel:
             json.dump(self.dataspec.to_json(), f, indent=1)
         np.savez(os.path.join(path, "forest.npz"), feat=self.forest.feat,
                  thr=self.forest.thr, left=self.forest.left,
-                 roots=self.forest.roots)
+                 roots=self.forest.roots, cat_idx=self.forest.cat_idx,
+                 masks=self.forest.masks)
         with open(os.path.join(path, "done"), "w") as f:
             f.write("")
 

@@ -21,7 +21,9 @@ def load_model(path: str) -> GenericModel:
         dataspec = DataSpecification.from_json(json.load(f))
     z = np.load(os.path.join(path, "forest.npz"))
     forest = FlatForest(feat=z["feat"], thr=z["thr"], left=z["left"],
-                        roots=z["roots"])
+                        roots=z["roots"],
+                        cat_idx=z["cat_idx"] if "cat_idx" in z else None,
+                        masks=z["masks"] if "masks" in z else None)
     cls = MODEL_CLASSES.get(header["model_type"], GenericModel)
     model = cls(
         forest=forest,
@@ -48,7 +50,8 @@ def serialize_model(model: GenericModel) -> bytes:
         zf.writestr("dataspec.json", json.dumps(model.dataspec.to_json()))
         fbuf = io.BytesIO()
         np.savez(fbuf, feat=model.forest.feat, thr=model.forest.thr,
-                 left=model.forest.left, roots=model.forest.roots)
+                 left=model.forest.left, roots=model.forest.roots,
+                 cat_idx=model.forest.cat_idx, masks=model.forest.masks)
         zf.writestr("forest.npz", fbuf.getvalue())
     return buf.getvalue()
 

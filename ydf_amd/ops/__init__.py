@@ -110,15 +110,18 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
                best_bin_nf: torch.Tensor, best_feat: torch.Tensor,
                best_bin: torch.Tensor, best_gain: torch.Tensor, slot0: int,
                n_slots: int, lambda_l2: float, min_hessian: float,
-               min_examples: int, min_gain: float, feat_mask=None):
+               min_examples: int, min_gain: float, feat_mask=None,
+               cat_flags=None, masks=None, cat_smooth: float = 1.0):
     F = hist.shape[1]
     n_bins = hist.shape[2]
     mp = feat_mask.data_ptr() if feat_mask is not None else 0
+    cf = cat_flags.data_ptr() if cat_flags is not None else 0
+    mk = masks.data_ptr() if masks is not None else 0
     args = (hist.data_ptr(), abs_of_slot.data_ptr(), node_stats.data_ptr(),
             best_gain_nf.data_ptr(), best_bin_nf.data_ptr(),
             best_feat.data_ptr(), best_bin.data_ptr(), best_gain.data_ptr(),
-            mp, F, n_bins, slot0, n_slots, lambda_l2, min_hessian,
-            min_examples, min_gain)
+            mp, cf, mk, F, n_bins, slot0, n_slots, lambda_l2, min_hessian,
+            min_examples, min_gain, cat_smooth)
     if hist.is_cuda:
         _C.gpu_split_scan(*args, _stream())
     else:
@@ -151,17 +154,21 @@ def subtract_hist(hist: torch.Tensor, hist_prev: torch.Tensor,
 
 def update_node_ids(bins: torch.Tensor, node_ids: torch.Tensor,
                     slot_map: torch.Tensor, best_feat: torch.Tensor,
-                    best_bin: torch.Tensor, level_base: int, level_size: int):
+                    best_bin: torch.Tensor, level_base: int, level_size: int,
+                    cat_flags=None, masks=None):
     F, N = bins.shape
+    cf = cat_flags.data_ptr() if cat_flags is not None else 0
+    mk = masks.data_ptr() if masks is not None else 0
     if bins.is_cuda:
         _C.gpu_update_node_ids(bins.data_ptr(), node_ids.data_ptr(),
                                slot_map.data_ptr(), best_feat.data_ptr(),
-                               best_bin.data_ptr(), N, level_base, level_size,
-                               _stream())
+                               best_bin.data_ptr(), cf, mk, N, level_base,
+                               level_size, _stream())
     else:
         _C.cpu_update_node_ids(bins.data_ptr(), node_ids.data_ptr(),
                                slot_map.data_ptr(), best_feat.data_ptr(),
-                               best_bin.data_ptr(), N, level_base, level_size)
+                               best_bin.data_ptr(), cf, mk, N, level_base,
+                               level_size)
 
 
 def leaf_values(node_stats: torch.Tensor, out: torch.Tensor,
@@ -203,21 +210,24 @@ def binary_logloss(preds: torch.Tensor, labels: torch.Tensor,
 def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
                    left: torch.Tensor, roots: torch.Tensor, out: torch.Tensor,
                    tree_start: int = 0, tree_step: int = 1,
-                   n_trees: int = -1, init: float = 0.0, scale: float = 1.0):
+                   n_trees: int = -1, init: float = 0.0, scale: float = 1.0,
+                   cat_idx=None, masks=None):
     """Flat-forest batch inference. X [F,N] f32, out [N] f32."""
     F, N = X.shape
     if n_trees < 0:
         n_trees = roots.numel()
+    ci = cat_idx.data_ptr() if cat_idx is not None else 0
+    mk = masks.data_ptr() if masks is not None else 0
     if X.is_cuda:
         _C.gpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
                               thr.data_ptr(), left.data_ptr(),
-                              roots.data_ptr(), tree_start, tree_step, n_trees,
-                              out.data_ptr(), init, scale, _stream())
+                              roots.data_ptr(), ci, mk, tree_start, tree_step,
+                              n_trees, out.data_ptr(), init, scale, _stream())
     else:
         _C.cpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
                               thr.data_ptr(), left.data_ptr(),
-                              roots.data_ptr(), tree_start, tree_step, n_trees,
-                              out.data_ptr(), init, scale)
+                              roots.data_ptr(), ci, mk, tree_start, tree_step,
+                              n_trees, out.data_ptr(), init, scale)
     return out
 
 

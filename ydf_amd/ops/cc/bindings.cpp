@@ -24,14 +24,16 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
                     int, void*);
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
 void gpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
-                    int32_t*, int32_t*, float*, const uint8_t*, int, int, int,
-                    int, SplitParams, void*);
+                    int32_t*, int32_t*, float*, const uint8_t*,
+                    const uint8_t*, unsigned long long*, int, int, int, int,
+                    SplitParams, void*);
 void gpu_plan_level(const float*, const int32_t*, int, int, int, int,
                     int32_t*, uint8_t*, void*);
 void gpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int,
                        void*);
 void gpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
-                         const int32_t*, const int32_t*, int64_t, int, int,
+                         const int32_t*, const int32_t*, const uint8_t*,
+                         const unsigned long long*, int64_t, int, int,
                          void*);
 void gpu_leaf_values(const float*, float*, int, float, void*);
 void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
@@ -39,7 +41,8 @@ void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
 void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
 // infer_kernels.hip
 void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
-                        const float*, const int32_t*, const int32_t*, int, int,
+                        const float*, const int32_t*, const int32_t*,
+                        const int32_t*, const unsigned long long*, int, int,
                         int, float*, float, float, void*);
 void gpu_sigmoid(const float*, float*, int64_t, void*);
 // cpu_ops.cpp
@@ -52,18 +55,21 @@ void cpu_hist_build(const uint8_t*, const float*, const int32_t*,
                     int);
 void cpu_weighted_target(const float*, const float*, float*, int64_t);
 void cpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
-                    int32_t*, int32_t*, float*, const uint8_t*, int, int, int,
-                    int, SplitParams);
+                    int32_t*, int32_t*, float*, const uint8_t*,
+                    const uint8_t*, unsigned long long*, int, int, int, int,
+                    SplitParams);
 void cpu_plan_level(const float*, const int32_t*, int, int, int, int,
                     int32_t*, uint8_t*);
 void cpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int);
 void cpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
-                         const int32_t*, const int32_t*, int64_t, int, int);
+                         const int32_t*, const int32_t*, const uint8_t*,
+                         const unsigned long long*, int64_t, int, int);
 void cpu_leaf_values(const float*, float*, int, float);
 void cpu_update_preds(float*, const int32_t*, const float*, int64_t, float);
 void cpu_binary_logloss(const float*, const float*, float*, int64_t);
 void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
-                        const float*, const int32_t*, const int32_t*, int, int,
+                        const float*, const int32_t*, const int32_t*,
+                        const int32_t*, const unsigned long long*, int, int,
                         int, float*, float, float);
 }
 
@@ -73,12 +79,13 @@ T* P(uintptr_t p) {
   return reinterpret_cast<T*>(p);
 }
 SplitParams MakeSP(float lambda_l2, float min_hessian, int min_examples,
-                   float min_gain) {
+                   float min_gain, float cat_smooth) {
   SplitParams sp;
   sp.lambda_l2 = lambda_l2;
   sp.min_hessian = min_hessian;
   sp.min_examples = min_examples;
   sp.min_gain = min_gain;
+  sp.cat_smooth = cat_smooth;
   return sp;
 }
 }  // namespace
@@ -133,16 +140,18 @@ PYBIND11_MODULE(_ydf_ops, m) {
         [](uintptr_t hist, uintptr_t abs_of_slot, uintptr_t node_stats,
            uintptr_t best_gain_nf, uintptr_t best_bin_nf, uintptr_t best_feat,
            uintptr_t best_bin, uintptr_t best_gain, uintptr_t feat_mask,
-           int F, int n_bins, int slot0, int n_slots, float lambda_l2,
-           float min_hessian, int min_examples, float min_gain,
-           uintptr_t stream) {
+           uintptr_t cat_flags, uintptr_t masks, int F, int n_bins, int slot0,
+           int n_slots, float lambda_l2, float min_hessian, int min_examples,
+           float min_gain, float cat_smooth, uintptr_t stream) {
           gpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
                          P<float>(node_stats), P<float>(best_gain_nf),
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
                          P<int32_t>(best_bin), P<float>(best_gain),
-                         P<uint8_t>(feat_mask), F, n_bins, slot0, n_slots,
+                         P<uint8_t>(feat_mask), P<uint8_t>(cat_flags),
+                         P<unsigned long long>(masks), F, n_bins, slot0,
+                         n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
-                                min_gain),
+                                min_gain, cat_smooth),
                          (void*)stream);
         },
         nogil);
@@ -166,12 +175,14 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("gpu_update_node_ids",
         [](uintptr_t bins, uintptr_t node_ids, uintptr_t slot_map,
-           uintptr_t best_feat, uintptr_t best_bin, int64_t N, int level_base,
-           int level_size, uintptr_t stream) {
+           uintptr_t best_feat, uintptr_t best_bin, uintptr_t cat_flags,
+           uintptr_t masks, int64_t N, int level_base, int level_size,
+           uintptr_t stream) {
           gpu_update_node_ids(P<uint8_t>(bins), P<int32_t>(node_ids),
                               P<int32_t>(slot_map), P<int32_t>(best_feat),
-                              P<int32_t>(best_bin), N, level_base, level_size,
-                              (void*)stream);
+                              P<int32_t>(best_bin), P<uint8_t>(cat_flags),
+                              P<unsigned long long>(masks), N, level_base,
+                              level_size, (void*)stream);
         },
         nogil);
   m.def("gpu_leaf_values",
@@ -198,13 +209,15 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("gpu_predict_forest",
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
-           uintptr_t left, uintptr_t roots, int tree_start, int tree_step,
-           int n_trees, uintptr_t out, float init, float scale,
-           uintptr_t stream) {
+           uintptr_t left, uintptr_t roots, uintptr_t cat_idx,
+           uintptr_t masks, int tree_start, int tree_step, int n_trees,
+           uintptr_t out, float init, float scale, uintptr_t stream) {
           gpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
                              P<float>(thr), P<int32_t>(left),
-                             P<int32_t>(roots), tree_start, tree_step, n_trees,
-                             P<float>(out), init, scale, (void*)stream);
+                             P<int32_t>(roots), P<int32_t>(cat_idx),
+                             P<unsigned long long>(masks), tree_start,
+                             tree_step, n_trees, P<float>(out), init, scale,
+                             (void*)stream);
         },
         nogil);
   m.def("gpu_sigmoid",
@@ -254,15 +267,18 @@ PYBIND11_MODULE(_ydf_ops, m) {
         [](uintptr_t hist, uintptr_t abs_of_slot, uintptr_t node_stats,
            uintptr_t best_gain_nf, uintptr_t best_bin_nf, uintptr_t best_feat,
            uintptr_t best_bin, uintptr_t best_gain, uintptr_t feat_mask,
-           int F, int n_bins, int slot0, int n_slots, float lambda_l2,
-           float min_hessian, int min_examples, float min_gain) {
+           uintptr_t cat_flags, uintptr_t masks, int F, int n_bins, int slot0,
+           int n_slots, float lambda_l2, float min_hessian, int min_examples,
+           float min_gain, float cat_smooth) {
           cpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
                          P<float>(node_stats), P<float>(best_gain_nf),
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
                          P<int32_t>(best_bin), P<float>(best_gain),
-                         P<uint8_t>(feat_mask), F, n_bins, slot0, n_slots,
+                         P<uint8_t>(feat_mask), P<uint8_t>(cat_flags),
+                         P<unsigned long long>(masks), F, n_bins, slot0,
+                         n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
-                                min_gain));
+                                min_gain, cat_smooth));
         },
         nogil);
   m.def("cpu_plan_level",
@@ -283,11 +299,12 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("cpu_update_node_ids",
         [](uintptr_t bins, uintptr_t node_ids, uintptr_t slot_map,
-           uintptr_t best_feat, uintptr_t best_bin, int64_t N, int level_base,
-           int level_size) {
+           uintptr_t best_feat, uintptr_t best_bin, uintptr_t cat_flags,
+           uintptr_t masks, int64_t N, int level_base, int level_size) {
           cpu_update_node_ids(P<uint8_t>(bins), P<int32_t>(node_ids),
                               P<int32_t>(slot_map), P<int32_t>(best_feat),
-                              P<int32_t>(best_bin), N, level_base,
+                              P<int32_t>(best_bin), P<uint8_t>(cat_flags),
+                              P<unsigned long long>(masks), N, level_base,
                               level_size);
         },
         nogil);
@@ -313,12 +330,14 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("cpu_predict_forest",
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
-           uintptr_t left, uintptr_t roots, int tree_start, int tree_step,
-           int n_trees, uintptr_t out, float init, float scale) {
+           uintptr_t left, uintptr_t roots, uintptr_t cat_idx,
+           uintptr_t masks, int tree_start, int tree_step, int n_trees,
+           uintptr_t out, float init, float scale) {
           cpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
                              P<float>(thr), P<int32_t>(left),
-                             P<int32_t>(roots), tree_start, tree_step, n_trees,
-                             P<float>(out), init, scale);
+                             P<int32_t>(roots), P<int32_t>(cat_idx),
+                             P<unsigned long long>(masks), tree_start,
+                             tree_step, n_trees, P<float>(out), init, scale);
         },
         nogil);
 }

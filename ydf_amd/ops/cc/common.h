@@ -35,6 +35,8 @@ struct SplitParams {
   float min_hessian;      // min_sum_hessian_in_leaf
   int min_examples;       // min_examples (reference default 5)
   float min_gain;         // splits with gain <= min_gain become leaves
+  float cat_smooth;       // l2_categorical_regularization (category order
+                          // statistic smoothing; reference default 1.0)
 };
 
 }  // namespace ydfa

@@ -204,13 +204,32 @@ void cpu_hist_build(const uint8_t* bins, const float* gh,
   });
 }
 
+// Categories ordered ascending by (G/(H+smooth), bin); empty bins last.
+// Matches bitonic_sort_bins in train_kernels.hip.
+static void sorted_bin_order(const float* hp, int n_bins, float smooth,
+                             std::vector<int>& order) {
+  order.resize(n_bins);
+  std::vector<float> key(n_bins);
+  for (int b = 0; b < n_bins; ++b) {
+    const float c = hp[b * 3 + 2];
+    key[b] = (c > 0.f) ? hp[b * 3] / (hp[b * 3 + 1] + smooth) : 1e30f;
+    order[b] = b;
+  }
+  std::sort(order.begin(), order.end(), [&](int a, int b) {
+    if (key[a] != key[b]) return key[a] < key[b];
+    return a < b;
+  });
+}
+
 void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                     float* node_stats, float* best_gain_nf,
                     int32_t* best_bin_nf, int32_t* best_feat,
                     int32_t* best_bin, float* best_gain,
-                    const uint8_t* feat_mask, int F, int n_bins, int slot0,
+                    const uint8_t* feat_mask, const uint8_t* cat_flags,
+                    unsigned long long* masks, int F, int n_bins, int slot0,
                     int n_slots, SplitParams sp) {
   ThreadPool::Get().ParallelFor(n_slots, [&](int slot) {
+    std::vector<int> order;
     const int out = slot0 + slot;
     const int abs_node = abs_of_slot[out];
     // Node totals from feature 0 (every example lands in some bin).
@@ -232,6 +251,9 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
         continue;
       }
       const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+      const bool is_cat = cat_flags != nullptr && cat_flags[f];
+      if (is_cat) sorted_bin_order(hp, n_bins, sp.cat_smooth, order);
+      auto bin_at = [&](int b) { return is_cat ? order[b] : b; };
       // Per-feature totals (matches the GPU kernel exactly; identical to
       // the feature-0 totals for any real histogram).
       float Gf = 0.f, Hf = 0.f, Cf = 0.f;
@@ -243,7 +265,8 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
       float fbest = -1e30f;
       int fbin = 0;
       for (int b = 0; b < n_bins - 1; ++b) {
-        GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+        const int bb = bin_at(b);
+        GL += hp[bb * 3]; HL += hp[bb * 3 + 1]; CL += hp[bb * 3 + 2];
         const float GR = Gf - GL, HR = Hf - HL, CR = Cf - CL;
         if (CL >= sp.min_examples && CR >= sp.min_examples &&
             HL >= sp.min_hessian && HR >= sp.min_hessian) {
@@ -270,9 +293,26 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
     best_bin[out] = node_best_b;
     best_gain[out] = node_best_gain;
     const float* hp = hist + ((int64_t)slot * F + node_best_f) * (n_bins * 3);
+    const bool is_cat = cat_flags != nullptr && cat_flags[node_best_f];
     float GL = 0.f, HL = 0.f, CL = 0.f;
-    for (int b = 0; b <= node_best_b; ++b) {
-      GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+    if (is_cat) {
+      sorted_bin_order(hp, n_bins, sp.cat_smooth, order);
+      unsigned long long m[kMaxBins / 64] = {0};
+      for (int r = 0; r < n_bins; ++r) {
+        const int b = order[r];
+        if (r <= node_best_b) {
+          GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+        } else if (hp[b * 3 + 2] > 0.f) {
+          m[b >> 6] |= 1ull << (b & 63);  // goes RIGHT
+        }
+      }
+      if (masks != nullptr)
+        for (int w = 0; w < kMaxBins / 64; ++w)
+          masks[(int64_t)abs_node * (kMaxBins / 64) + w] = m[w];
+    } else {
+      for (int b = 0; b <= node_best_b; ++b) {
+        GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+      }
     }
     float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
     float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
@@ -315,8 +355,9 @@ void cpu_subtract_hist(float* hist, const float* hist_prev,
 
 void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
                          const int32_t* slot_map, const int32_t* best_feat,
-                         const int32_t* best_bin, int64_t N, int level_base,
-                         int level_size) {
+                         const int32_t* best_bin, const uint8_t* cat_flags,
+                         const unsigned long long* masks, int64_t N,
+                         int level_base, int level_size) {
   const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
                                            (int)(N / 16384) + 1));
   const int64_t per = (N + nb - 1) / nb;
@@ -331,7 +372,13 @@ void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
       const int f = best_feat[slot];
       if (f < 0) continue;
       const int b = bins[(int64_t)f * N + k];
-      node_ids[k] = 2 * nid + 1 + (b > best_bin[slot] ? 1 : 0);
+      int right;
+      if (cat_flags != nullptr && cat_flags[f])
+        right = (int)((masks[(int64_t)nid * (kMaxBins / 64) + (b >> 6)]
+                       >> (b & 63)) & 1ull);
+      else
+        right = b > best_bin[slot] ? 1 : 0;
+      node_ids[k] = 2 * nid + 1 + right;
     }
   });
 }
@@ -374,8 +421,10 @@ void cpu_binary_logloss(const float* preds, const float* labels, float* out2,
 
 void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
                         const float* thr, const int32_t* left,
-                        const int32_t* roots, int tree_start, int tree_step,
-                        int n_trees, float* out, float init, float scale) {
+                        const int32_t* roots, const int32_t* cat_idx,
+                        const unsigned long long* masks, int tree_start,
+                        int tree_step, int n_trees, float* out, float init,
+                        float scale) {
   (void)F;
   const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
                                            (int)(N / 1024) + 1));
@@ -388,7 +437,18 @@ void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
         int n = roots[tree_start + (int64_t)tt * tree_step];
         int f = feat[n];
         while (f >= 0) {
-          n = left[n] + (X[(int64_t)f * N + k] > thr[n] ? 1 : 0);
+          const float xv = X[(int64_t)f * N + k];
+          int right;
+          const int ci = cat_idx ? cat_idx[n] : -1;
+          if (ci >= 0) {
+            int cbin = (int)xv;
+            cbin = cbin < 0 ? 0 : (cbin > 255 ? 255 : cbin);
+            right = (int)((masks[(int64_t)ci * 4 + (cbin >> 6)]
+                           >> (cbin & 63)) & 1ull);
+          } else {
+            right = xv > thr[n] ? 1 : 0;
+          }
+          n = left[n] + right;
           f = feat[n];
         }
         acc += thr[n];

@@ -4,11 +4,15 @@
 // redesigned example-parallel for 64-wide wavefronts with the example tile
 // staged in LDS).
 //
-// Model layout ("flat forest", built by ydf_amd/serving/flat.py):
-//   feat[n]  : i32, split feature of node n, -1 => leaf
-//   thr[n]   : f32, split threshold (x > thr -> right) or leaf value
-//   left[n]  : i32, index of left child (right = left + 1)
-//   roots[t] : i32, root node index of tree t
+// Model layout ("flat forest", built by ydf_amd/model/forest.py):
+//   feat[n]     : i32, split feature of node n, -1 => leaf
+//   thr[n]      : f32, split threshold (x > thr -> right) or leaf value
+//   left[n]     : i32, index of left child (right = left + 1)
+//   roots[t]    : i32, root node index of tree t
+//   cat_idx[n]  : i32, -1 for numerical/leaf; else index into masks of a
+//                 256-bit "category goes right" bitmask (4 x u64) —
+//                 capability analogue of the reference's ContainsBitmap
+//                 conditions (model/decision_tree/decision_tree.proto:86-151)
 // All node arrays are concatenated over trees.
 #include <hip/hip_runtime.h>
 #include <cstdint>
@@ -25,8 +29,10 @@ __global__ void predict_forest_lds_kernel(
     const float* __restrict__ X, int64_t N, int F,
     const int32_t* __restrict__ feat, const float* __restrict__ thr,
     const int32_t* __restrict__ left, const int32_t* __restrict__ roots,
-    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
-    float init, float scale) {
+    const int32_t* __restrict__ cat_idx,
+    const unsigned long long* __restrict__ masks, int tree_start,
+    int tree_step, int n_trees, float* __restrict__ out, float init,
+    float scale) {
   extern __shared__ float xs[];  // [F][kTile]
   const int64_t base = (int64_t)blockIdx.x * kTile;
   const int tid = threadIdx.x;
@@ -43,7 +49,17 @@ __global__ void predict_forest_lds_kernel(
     int n = roots[tree_start + (int64_t)tt * tree_step];
     int f = feat[n];
     while (f >= 0) {
-      n = left[n] + (xs[f * kTile + tid] > thr[n] ? 1 : 0);
+      const float xv = xs[f * kTile + tid];
+      int right;
+      const int ci = cat_idx ? cat_idx[n] : -1;
+      if (ci >= 0) {
+        int c = (int)xv;
+        c = c < 0 ? 0 : (c > 255 ? 255 : c);
+        right = (int)((masks[(int64_t)ci * 4 + (c >> 6)] >> (c & 63)) & 1ull);
+      } else {
+        right = xv > thr[n] ? 1 : 0;
+      }
+      n = left[n] + right;
       f = feat[n];
     }
     acc += thr[n];
@@ -56,8 +72,10 @@ __global__ void predict_forest_global_kernel(
     const float* __restrict__ X, int64_t N, int F,
     const int32_t* __restrict__ feat, const float* __restrict__ thr,
     const int32_t* __restrict__ left, const int32_t* __restrict__ roots,
-    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
-    float init, float scale) {
+    const int32_t* __restrict__ cat_idx,
+    const unsigned long long* __restrict__ masks, int tree_start,
+    int tree_step, int n_trees, float* __restrict__ out, float init,
+    float scale) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t k = i; k < N; k += stride) {
@@ -66,7 +84,18 @@ __global__ void predict_forest_global_kernel(
       int n = roots[tree_start + (int64_t)tt * tree_step];
       int f = feat[n];
       while (f >= 0) {
-        n = left[n] + (X[(int64_t)f * N + k] > thr[n] ? 1 : 0);
+        const float xv = X[(int64_t)f * N + k];
+        int right;
+        const int ci = cat_idx ? cat_idx[n] : -1;
+        if (ci >= 0) {
+          int c = (int)xv;
+          c = c < 0 ? 0 : (c > 255 ? 255 : c);
+          right =
+              (int)((masks[(int64_t)ci * 4 + (c >> 6)] >> (c & 63)) & 1ull);
+        } else {
+          right = xv > thr[n] ? 1 : 0;
+        }
+        n = left[n] + right;
         f = feat[n];
       }
       acc += thr[n];
@@ -87,22 +116,25 @@ extern "C" {
 
 void gpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
                         const float* thr, const int32_t* left,
-                        const int32_t* roots, int tree_start, int tree_step,
-                        int n_trees, float* out, float init, float scale,
-                        void* stream) {
+                        const int32_t* roots, const int32_t* cat_idx,
+                        const unsigned long long* masks, int tree_start,
+                        int tree_step, int n_trees, float* out, float init,
+                        float scale, void* stream) {
   const size_t lds = (size_t)F * kTile * sizeof(float);
   if (lds <= 96 * 1024) {
     const int grid = (int)((N + kTile - 1) / kTile);
     hipLaunchKernelGGL(predict_forest_lds_kernel, dim3(grid), dim3(kTile), lds,
                        (hipStream_t)stream, X, N, F, feat, thr, left, roots,
-                       tree_start, tree_step, n_trees, out, init, scale);
+                       cat_idx, masks, tree_start, tree_step, n_trees, out,
+                       init, scale);
   } else {
     int grid = (int)((N + kTile - 1) / kTile);
     if (grid > 4096) grid = 4096;
     if (grid < 1) grid = 1;
     hipLaunchKernelGGL(predict_forest_global_kernel, dim3(grid), dim3(kTile),
                        0, (hipStream_t)stream, X, N, F, feat, thr, left, roots,
-                       tree_start, tree_step, n_trees, out, init, scale);
+                       cat_idx, masks, tree_start, tree_step, n_trees, out,
+                       init, scale);
   }
 }
 

@@ -278,12 +278,40 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
 // into node_stats[abs_node]. Gain (2nd-order, reference use_hessian_gain
 // semantics): gain = GL^2/(HL+l2) + GR^2/(HR+l2) - G^2/(H+l2).
 // ---------------------------------------------------------------------------
+// Bitonic sort of the n_bins (key, idx) pairs in LDS; ascending by
+// (key, idx) — deterministic under ties. Used for categorical features:
+// categories ordered by gradient statistic G/(H + smooth), then scanned
+// like a numerical feature (the reference's CART one-vs-rest ordering,
+// splitter_scanner.h:859 sorting variants; same algorithm as LightGBM's
+// categorical handling). Empty bins carry key=+inf (sort last).
+__device__ inline void bitonic_sort_bins(float* key, short* idx, int n,
+                                         int tid) {
+  for (int k = 2; k <= n; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      const int ixj = tid ^ j;
+      if (ixj > tid) {
+        const bool up = (tid & k) == 0;
+        const float a = key[tid], bval = key[ixj];
+        const short ia = idx[tid], ib = idx[ixj];
+        const bool swap = up ? (a > bval || (a == bval && ia > ib))
+                             : (a < bval || (a == bval && ia < ib));
+        if (swap) {
+          key[tid] = bval; key[ixj] = a;
+          idx[tid] = ib; idx[ixj] = ia;
+        }
+      }
+      __syncthreads();
+    }
+  }
+}
+
 __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
                                        const int32_t* __restrict__ abs_of_slot,
                                        float* __restrict__ node_stats,
                                        float* __restrict__ best_gain_nf,
                                        int32_t* __restrict__ best_bin_nf,
                                        const uint8_t* __restrict__ feat_mask,
+                                       const uint8_t* __restrict__ cat_flags,
                                        int F, int n_bins, int slot0,
                                        SplitParams sp) {
   const int slot = blockIdx.x;
@@ -291,9 +319,25 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
   const int b = threadIdx.x;  // blockDim.x == n_bins (power of two <= 256)
   __shared__ float sg[kMaxBins], sh[kMaxBins], sc[kMaxBins];
   const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
-  sg[b] = hp[b * 3];
-  sh[b] = hp[b * 3 + 1];
-  sc[b] = hp[b * 3 + 2];
+  float rg0 = hp[b * 3];
+  float rh0 = hp[b * 3 + 1];
+  float rc0 = hp[b * 3 + 2];
+  if (cat_flags != nullptr && cat_flags[f]) {
+    // categorical: sort bins by G/(H+smooth) before scanning
+    __shared__ float skey[kMaxBins];
+    __shared__ short sidx[kMaxBins];
+    skey[b] = (rc0 > 0.f) ? rg0 / (rh0 + sp.cat_smooth) : 1e30f;
+    sidx[b] = (short)b;
+    sg[b] = rg0; sh[b] = rh0; sc[b] = rc0;
+    __syncthreads();
+    bitonic_sort_bins(skey, sidx, n_bins, b);
+    const int src = sidx[b];
+    rg0 = sg[src]; rh0 = sh[src]; rc0 = sc[src];
+    __syncthreads();
+  }
+  sg[b] = rg0;
+  sh[b] = rh0;
+  sc[b] = rc0;
   __syncthreads();
   for (int off = 1; off < n_bins; off <<= 1) {
     float tg = 0.f, th = 0.f, tc = 0.f;
@@ -354,8 +398,11 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
                                     float* __restrict__ node_stats,
                                     int32_t* __restrict__ best_feat,
                                     int32_t* __restrict__ best_bin,
-                                    float* __restrict__ best_gain, int F,
-                                    int n_bins, int slot0, SplitParams sp) {
+                                    float* __restrict__ best_gain,
+                                    const uint8_t* __restrict__ cat_flags,
+                                    unsigned long long* __restrict__ masks,
+                                    int F, int n_bins, int slot0,
+                                    SplitParams sp) {
   const int slot = blockIdx.x;
   const int t = threadIdx.x;
   __shared__ float rg[kBlock];
@@ -378,31 +425,91 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
     }
     __syncthreads();
   }
-  if (t != 0) return;
-  const float gain = rg[0];
-  const int f = rf[0];
-  const int out = slot0 + slot;
-  if (f < 0 || gain <= sp.min_gain) {
-    best_feat[out] = -1;
-    best_bin[out] = 0;
-    best_gain[out] = 0.f;
+  __shared__ int w_f, w_bin, w_abs;
+  if (t == 0) {
+    const float gain = rg[0];
+    const int f = rf[0];
+    const int out = slot0 + slot;
+    if (f < 0 || gain <= sp.min_gain) {
+      best_feat[out] = -1;
+      best_bin[out] = 0;
+      best_gain[out] = 0.f;
+      w_f = -1;
+    } else {
+      const int bin = best_bin_nf[(int64_t)slot * F + f];
+      best_feat[out] = f;
+      best_bin[out] = bin;
+      best_gain[out] = gain;
+      w_f = f;
+      w_bin = bin;
+      w_abs = abs_of_slot[out];
+    }
+  }
+  __syncthreads();
+  const int f = w_f;
+  if (f < 0) return;
+  const bool is_cat = cat_flags != nullptr && cat_flags[f];
+  const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+  const int abs_node = w_abs;
+  if (!is_cat) {
+    if (t != 0) return;
+    const int bin = w_bin;
+    float GL = 0.f, HL = 0.f, CL = 0.f;
+    for (int bb = 0; bb <= bin; ++bb) {
+      GL += hp[bb * 3]; HL += hp[bb * 3 + 1]; CL += hp[bb * 3 + 2];
+    }
+    const float* ns = node_stats + (int64_t)abs_node * 3;
+    float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
+    float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
+    nl[0] = GL; nl[1] = HL; nl[2] = CL;
+    nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
     return;
   }
-  const int bin = best_bin_nf[(int64_t)slot * F + f];
-  best_feat[out] = f;
-  best_bin[out] = bin;
-  best_gain[out] = gain;
-  const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
-  float GL = 0.f, HL = 0.f, CL = 0.f;
-  for (int bb = 0; bb <= bin; ++bb) {
-    GL += hp[bb * 3]; HL += hp[bb * 3 + 1]; CL += hp[bb * 3 + 2];
+  // Categorical winner: rebuild the sorted order (same deterministic sort
+  // as stage A), emit the "goes RIGHT" category bitmask = sorted ranks >
+  // best rank with count > 0 (unseen categories default LEFT, matching
+  // the reference's negative-child policy for absent values), and the
+  // child stats from the sorted prefix.
+  __shared__ float skey[kMaxBins];
+  __shared__ short sidx[kMaxBins];
+  __shared__ float sg[kMaxBins], sh[kMaxBins], sc[kMaxBins];
+  const int b = t;  // blockDim == kBlock == kMaxBins == n_bins required
+  const float bg = hp[b * 3], bh = hp[b * 3 + 1], bc = hp[b * 3 + 2];
+  skey[b] = (bc > 0.f) ? bg / (bh + sp.cat_smooth) : 1e30f;
+  sidx[b] = (short)b;
+  sg[b] = bg; sh[b] = bh; sc[b] = bc;
+  __syncthreads();
+  bitonic_sort_bins(skey, sidx, n_bins, b);
+  const int src = sidx[b];
+  const float mg = sg[src], mh = sh[src], mc = sc[src];
+  __syncthreads();
+  // mask: ranks > w_bin with examples go right
+  __shared__ unsigned long long lmask[kMaxBins / 64];
+  if (b < n_bins / 64) lmask[b] = 0ull;
+  __syncthreads();
+  if (b > w_bin && mc > 0.f)
+    atomicOr(&lmask[src >> 6], 1ull << (src & 63));
+  // left-child stats: inclusive prefix over sorted ranks <= w_bin
+  sg[b] = (b <= w_bin) ? mg : 0.f;
+  sh[b] = (b <= w_bin) ? mh : 0.f;
+  sc[b] = (b <= w_bin) ? mc : 0.f;
+  __syncthreads();
+  for (int off = n_bins >> 1; off > 0; off >>= 1) {
+    if (b < off) {
+      sg[b] += sg[b + off]; sh[b] += sh[b + off]; sc[b] += sc[b + off];
+    }
+    __syncthreads();
   }
-  const int abs_node = abs_of_slot[out];
-  const float* ns = node_stats + (int64_t)abs_node * 3;
-  float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
-  float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
-  nl[0] = GL; nl[1] = HL; nl[2] = CL;
-  nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
+  if (b < n_bins / 64 && masks != nullptr)
+    masks[(int64_t)abs_node * (kMaxBins / 64) + b] = lmask[b];
+  if (b == 0) {
+    const float GL = sg[0], HL = sh[0], CL = sc[0];
+    const float* ns = node_stats + (int64_t)abs_node * 3;
+    float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
+    float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
+    nl[0] = GL; nl[1] = HL; nl[2] = CL;
+    nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -461,6 +568,9 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
                                        const int32_t* __restrict__ slot_map,
                                        const int32_t* __restrict__ best_feat,
                                        const int32_t* __restrict__ best_bin,
+                                       const uint8_t* __restrict__ cat_flags,
+                                       const unsigned long long* __restrict__
+                                           masks,
                                        int64_t N, int level_base,
                                        int level_size) {
   // 4-row unroll: four independent (node -> split -> bin) load chains in
@@ -489,7 +599,13 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
     for (int u = 0; u < 4; ++u) {
       if (f[u] < 0) continue;
       const int b = bins[(int64_t)f[u] * N + k + u * stride];
-      node_ids[k + u * stride] = 2 * nid[u] + 1 + (b > sbin[u]);
+      int right;
+      if (cat_flags != nullptr && cat_flags[f[u]])
+        right = (int)((masks[(int64_t)nid[u] * (kMaxBins / 64) + (b >> 6)]
+                       >> (b & 63)) & 1ull);
+      else
+        right = b > sbin[u];
+      node_ids[k + u * stride] = 2 * nid[u] + 1 + right;
     }
   }
   for (; k < N; k += stride) {
@@ -501,7 +617,13 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
     const int f = best_feat[slot];
     if (f < 0) continue;  // leaf: park
     const int b = bins[(int64_t)f * N + k];
-    node_ids[k] = 2 * nid + 1 + (b > best_bin[slot]);
+    int right;
+    if (cat_flags != nullptr && cat_flags[f])
+      right = (int)((masks[(int64_t)nid * (kMaxBins / 64) + (b >> 6)]
+                     >> (b & 63)) & 1ull);
+    else
+      right = b > best_bin[slot];
+    node_ids[k] = 2 * nid + 1 + right;
   }
 }
 
@@ -663,16 +785,17 @@ void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                     float* node_stats, float* best_gain_nf,
                     int32_t* best_bin_nf, int32_t* best_feat,
                     int32_t* best_bin, float* best_gain,
-                    const uint8_t* feat_mask, int F, int n_bins, int slot0,
+                    const uint8_t* feat_mask, const uint8_t* cat_flags,
+                    unsigned long long* masks, int F, int n_bins, int slot0,
                     int n_slots, SplitParams sp, void* stream) {
   hipLaunchKernelGGL(split_scan_feat_kernel, dim3(n_slots, F), dim3(n_bins),
                      0, (hipStream_t)stream, hist, abs_of_slot, node_stats,
-                     best_gain_nf, best_bin_nf, feat_mask, F, n_bins, slot0,
-                     sp);
+                     best_gain_nf, best_bin_nf, feat_mask, cat_flags, F,
+                     n_bins, slot0, sp);
   hipLaunchKernelGGL(split_select_kernel, dim3(n_slots), dim3(kBlock), 0,
                      (hipStream_t)stream, hist, abs_of_slot, best_gain_nf,
                      best_bin_nf, node_stats, best_feat, best_bin, best_gain,
-                     F, n_bins, slot0, sp);
+                     cat_flags, masks, F, n_bins, slot0, sp);
 }
 
 void gpu_plan_level(const float* node_stats, const int32_t* prev_best_feat,
@@ -697,11 +820,13 @@ void gpu_subtract_hist(float* hist, const float* hist_prev,
 
 void gpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
                          const int32_t* slot_map, const int32_t* best_feat,
-                         const int32_t* best_bin, int64_t N, int level_base,
-                         int level_size, void* stream) {
+                         const int32_t* best_bin, const uint8_t* cat_flags,
+                         const unsigned long long* masks, int64_t N,
+                         int level_base, int level_size, void* stream) {
   hipLaunchKernelGGL(update_node_ids_kernel, dim3(elem_grid(N, 4096)),
                      dim3(kBlock), 0, (hipStream_t)stream, bins, node_ids,
-                     slot_map, best_feat, best_bin, N, level_base, level_size);
+                     slot_map, best_feat, best_bin, cat_flags, masks, N,
+                     level_base, level_size);
 }
 
 void gpu_leaf_values(const float* node_stats, float* leaf_values,
