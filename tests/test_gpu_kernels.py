@@ -200,3 +200,35 @@ def test_binary_logloss_gpu(dev):
     ref = torch.nn.functional.binary_cross_entropy_with_logits(
         preds, labels, reduction="sum")
     np.testing.assert_allclose(out[0].item(), ref.item(), rtol=1e-3)
+
+
+def test_categorical_gpu_vs_cpu():
+    """Categorical set-splits: GPU bitonic ordering must equal CPU argsort
+    (integer-valued stats keep everything exact)."""
+    rng = np.random.RandomState(11)
+    n = 30000
+    cats = rng.randint(0, 12, n)
+    y = (cats % 3 == 0)
+    d = {"c": np.array([f"cat{v}" for v in cats]),
+         "x": rng.randint(-5, 5, n).astype(np.float32),
+         "label": np.where(y, "p", "n")}
+    kw = dict(label="label", num_trees=3, max_depth=4,
+              bootstrap_training_dataset=False, num_candidate_attributes=-1)
+    m_cpu = ydf.RandomForestLearner(device="cpu", **kw).train(d)
+    m_gpu = ydf.RandomForestLearner(device="cuda", **kw).train(d)
+    np.testing.assert_array_equal(m_gpu.forest.feat, m_cpu.forest.feat)
+    np.testing.assert_array_equal(m_gpu.forest.cat_idx, m_cpu.forest.cat_idx)
+    np.testing.assert_array_equal(m_gpu.forest.masks, m_cpu.forest.masks)
+    np.testing.assert_allclose(m_gpu.predict(d, device="cuda"),
+                               m_cpu.predict(d, device="cpu"), rtol=1e-5,
+                               atol=1e-6)
+
+
+def test_adult_gpu_quality(adult_paths):
+    pd = pytest.importorskip("pandas")
+    tr, te = adult_paths
+    m = ydf.GradientBoostedTreesLearner(label="income",
+                                        device="cuda").train(pd.read_csv(tr))
+    ev = m.evaluate(pd.read_csv(te), device="cuda")
+    assert ev.accuracy > 0.86
+    assert ev.auc > 0.92
