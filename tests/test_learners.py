@@ -179,3 +179,44 @@ def test_categorical_model_roundtrip(tmp_path):
     m2 = ydf.load_model(p)
     np.testing.assert_allclose(m.predict(d), m2.predict(d), rtol=1e-6)
     assert m.evaluate(d).accuracy > 0.99
+
+
+def test_checkpoint_resume_exact(tmp_path, binary_data):
+    """Resumed training must produce the same model as a straight run
+    (reference try_resume_training, gradient_boosted_trees.cc:1403-1443)."""
+    wd = str(tmp_path / "wd")
+    kw = dict(label="label", validation_ratio=0.1, early_stopping="NONE")
+    ydf.GradientBoostedTreesLearner(
+        num_trees=15, working_dir=wd, **kw).train(binary_data)
+    m2 = ydf.GradientBoostedTreesLearner(
+        num_trees=30, working_dir=wd, resume_training=True,
+        **kw).train(binary_data)
+    assert m2.num_trees() == 30
+    m3 = ydf.GradientBoostedTreesLearner(num_trees=30, **kw).train(
+        binary_data)
+    np.testing.assert_allclose(m2.predict(binary_data),
+                               m3.predict(binary_data), atol=1e-5)
+
+
+def test_maximum_training_duration(binary_data):
+    import time
+
+    t0 = time.time()
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=100000, validation_ratio=0,
+        maximum_training_duration_seconds=1.5).train(binary_data)
+    assert time.time() - t0 < 15
+    assert 1 <= m.num_trees() < 100000
+
+
+def test_variable_importances_and_analyze(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=20).train(
+        binary_data)
+    vi = m.variable_importances()
+    assert "SUM_SCORE" in vi and "NUM_NODES" in vi
+    # x1 carries most signal in the fixture
+    assert vi["SUM_SCORE"][0][1] == "x1"
+    an = m.analyze(binary_data)
+    assert "MEAN_DECREASE_IN_ACCURACY" in an.variable_importances
+    assert len(an.partial_dependences) == 3
+    assert an._repr_html_()

@@ -96,6 +96,19 @@ class GenericLearner:
             raise ValueError("classification label must be categorical")
         return list(lspec.vocab[1:])
 
+    @staticmethod
+    def _feature_gains(trees, names):
+        """Total split gain per feature (persisted for SUM_SCORE variable
+        importances; reference AbstractModel precomputed importances)."""
+        g = np.zeros(len(names), dtype=np.float64)
+        for t in trees:
+            if t.gain is None:
+                continue
+            valid = t.feat >= 0
+            np.add.at(g, t.feat[valid], t.gain[valid].astype(np.float64))
+        return {names[i]: float(g[i]) for i in range(len(names))
+                if g[i] > 0}
+
     def train(self, data, valid=None, verbose=None):
         raise NotImplementedError
 

@@ -48,6 +48,10 @@ class GradientBoostedTreesLearner(GenericLearner):
                  use_hessian_gain: bool = True,
                  apply_link_function: bool = True,
                  l2_categorical_regularization: float = 1.0,
+                 working_dir: Optional[str] = None,
+                 resume_training: bool = False,
+                 resume_training_snapshot_interval_seconds: float = 1800.0,
+                 maximum_training_duration_seconds: float = -1.0,
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
@@ -64,6 +68,11 @@ class GradientBoostedTreesLearner(GenericLearner):
             use_hessian_gain=use_hessian_gain,
             apply_link_function=apply_link_function,
             l2_categorical_regularization=l2_categorical_regularization,
+            working_dir=working_dir, resume_training=resume_training,
+            resume_training_snapshot_interval_seconds=(
+                resume_training_snapshot_interval_seconds),
+            maximum_training_duration_seconds=(
+                maximum_training_duration_seconds),
         )
 
     def train(self, data, valid=None, verbose=None
@@ -127,9 +136,6 @@ class GradientBoostedTreesLearner(GenericLearner):
                                       valid_bins=valid_bins,
                                       valid_labels=valid_labels,
                                       cat_flags=cat_flags)
-        trees, init_preds, logs = trainer_lib.train_gbt(t, log=info)
-        flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"],
-                                 cat_feats=self._cat_feature_flags(ds))
         C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
         activation = "identity"
         if hp["apply_link_function"]:
@@ -137,11 +143,106 @@ class GradientBoostedTreesLearner(GenericLearner):
                 activation = "sigmoid"
             elif loss == trainer_lib.LOSS_MULTINOMIAL:
                 activation = "softmax"
-        model = GradientBoostedTreesModel(
-            forest=flat, dataspec=ds.dataspec, task=self._task,
-            label_classes=classes, init_predictions=init_preds,
-            num_trees_per_iter=C, activation=activation)
+        cat_feats = self._cat_feature_flags(ds)
+        names = [c.name for c in ds.dataspec.feature_columns]
+
+        def make_model(flat, init_preds, gains):
+            return GradientBoostedTreesModel(
+                forest=flat, dataspec=ds.dataspec, task=self._task,
+                label_classes=classes, init_predictions=init_preds,
+                num_trees_per_iter=C, activation=activation,
+                metadata={"feature_gains": gains})
+
+        # checkpoint/resume (reference try_resume_training +
+        # snapshot interval, abstract_learner.proto:52-56)
+        import os as _os
+
+        partial = None
+        start_it = 0
+        resume_margins = resume_valid_margins = None
+        snap_dir = None
+        snapshot_cb = None
+        if hp.get("working_dir"):
+            snap_dir = _os.path.join(hp["working_dir"], "snapshot")
+            if hp.get("resume_training") and _os.path.exists(
+                    _os.path.join(snap_dir, "done")):
+                from ydf_amd.model.model_lib import load_model as _load
+
+                partial = _load(snap_dir)
+                start_it = partial.num_trees() // max(C, 1)
+                info(f"resuming training from snapshot at iteration "
+                     f"{start_it}")
+
+            def _snapshot(trees_so_far, iteration, init_preds_s):
+                import shutil
+                import tempfile
+
+                flat_s = build_flat_forest(trees_so_far, bnd,
+                                           leaf_scale=hp["shrinkage"],
+                                           cat_feats=cat_feats)
+                if partial is not None:
+                    from ydf_amd.model.forest import concat_forests
+
+                    flat_s = concat_forests(partial.forest, flat_s)
+                ms = make_model(flat_s, init_preds_s,
+                                self._feature_gains(trees_so_far, names))
+                tmp = snap_dir + ".tmp"
+                if _os.path.exists(tmp):
+                    shutil.rmtree(tmp)
+                ms.save(tmp)
+                if _os.path.exists(snap_dir):
+                    shutil.rmtree(snap_dir)
+                _os.replace(tmp, snap_dir)
+                info(f"snapshot written at iteration {iteration}")
+
+            snapshot_cb = _snapshot
+
+        if partial is not None:
+            import torch as _t
+
+            # exact resume: margins of the partial model over the raw
+            # training matrix, re-split with the same permutation
+            X_all = ds.X
+            perm_margins = partial.predict_margin(
+                _t.from_numpy(np.ascontiguousarray(X_all)).to(device))
+            # re-apply the train/valid permutation
+            if valid_bins is not None:
+                rngp = np.random.RandomState(self.random_seed)
+                permp = rngp.permutation(X_all.shape[1])
+                n_validp = max(1, int(X_all.shape[1]
+                                      * hp["validation_ratio"]))                     if X_all.shape[1] > 10 else 0
+                vi_idx = _t.from_numpy(permp[:n_validp].copy()).to(device)
+                ti_idx = _t.from_numpy(permp[n_validp:].copy()).to(device)
+                resume_margins = perm_margins[:, ti_idx].contiguous()
+                resume_valid_margins = perm_margins[:, vi_idx].contiguous()
+            else:
+                resume_margins = perm_margins.contiguous()
+
+        trees, init_preds, logs = trainer_lib.train_gbt(
+            t, log=info, start_iteration=start_it,
+            resume_margins=resume_margins,
+            resume_valid_margins=resume_valid_margins,
+            snapshot_cb=snapshot_cb,
+            snapshot_interval_seconds=hp.get(
+                "resume_training_snapshot_interval_seconds", 1800.0),
+            max_duration_seconds=hp.get(
+                "maximum_training_duration_seconds", -1.0))
+        flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"],
+                                 cat_feats=cat_feats)
+        gains = self._feature_gains(trees, names)
+        if partial is not None:
+            from ydf_amd.model.forest import concat_forests
+
+            flat = concat_forests(partial.forest, flat)
+            pg = (partial.metadata or {}).get("feature_gains", {})
+            for k, v in pg.items():
+                gains[k] = gains.get(k, 0.0) + v
+        model = make_model(flat, init_preds, gains)
         model.training_logs = logs
+        if snapshot_cb is not None:
+            # final state also becomes the snapshot (enables continuing
+            # with a larger num_trees later)
+            snapshot_cb(trees, hp["num_trees"], init_preds)
         return model
 
     def _prepare_valid(self, valid, train_ds: VerticalDataset, device):
@@ -235,7 +336,9 @@ class RandomForestLearner(GenericLearner):
         model = RandomForestModel(
             forest=flat, dataspec=ds.dataspec, task=self._task,
             label_classes=classes, init_predictions=[0.0] * max(C, 1),
-            num_trees_per_iter=C, activation="identity")
+            num_trees_per_iter=C, activation="identity",
+            metadata={"feature_gains": self._feature_gains(
+                trees, [c.name for c in ds.dataspec.feature_columns])})
         return model
 
 

@@ -68,6 +68,7 @@ class HostTree:
     counts: np.ndarray      # [total_nodes] f32
     max_depth: int
     masks: Optional[np.ndarray] = None  # [total_nodes,4] u64 (cat splits)
+    gain: Optional[np.ndarray] = None   # [total_nodes] f32 (split gains)
 
 
 def _dist_ok() -> bool:
@@ -137,6 +138,8 @@ class ForestTrainer:
                                      device=dev)
         self.tree_bin = torch.empty(self.total_nodes, dtype=torch.int32,
                                     device=dev)
+        self.tree_gain = torch.empty(self.total_nodes, dtype=torch.float32,
+                                     device=dev)
         # 256-bit "category goes right" masks, one per node (int64 bit-pattern)
         self.tree_masks = torch.zeros((self.total_nodes, 4),
                                       dtype=torch.int64, device=dev) \
@@ -197,6 +200,7 @@ class ForestTrainer:
         cfg = self.cfg
         self.tree_feat.fill_(-1)
         self.tree_bin.zero_()
+        self.tree_gain.zero_()
         self.node_stats.zero_()
         if self.tree_masks is not None:
             self.tree_masks.zero_()
@@ -332,6 +336,7 @@ class ForestTrainer:
             idx64 = torch.from_numpy(active_abs).to(self.device)
             self.tree_feat[idx64] = self.best_feat[:n_active]
             self.tree_bin[idx64] = self.best_bin[:n_active]
+            self.tree_gain[idx64] = self.best_gain[:n_active]
             ops.update_node_ids(self.bins, self.node_ids, slot_map,
                                 self.best_feat, self.best_bin, level_base,
                                 level_size, cat_flags=self.cat_flags,
@@ -366,6 +371,7 @@ class ForestTrainer:
             max_depth=cfg.max_depth,
             masks=self.tree_masks.cpu().numpy().view(np.uint64).copy()
             if self.tree_masks is not None else None,
+            gain=self.tree_gain.cpu().numpy().copy(),
         )
 
     def _dense_level(self, tree_idx: int, level: int, need: int,
@@ -407,6 +413,8 @@ class ForestTrainer:
             self.best_feat[:level_size]
         self.tree_bin[level_base:level_base + level_size] = \
             self.best_bin[:level_size]
+        self.tree_gain[level_base:level_base + level_size] = \
+            self.best_gain[:level_size]
         ops.update_node_ids(self.bins, self.node_ids, identity,
                             self.best_feat, self.best_bin, level_base,
                             level_size, cat_flags=self.cat_flags,
@@ -427,13 +435,25 @@ class ForestTrainer:
                 masks=self.tree_masks)
 
 
-def train_gbt(trainer: ForestTrainer, log=None):
+def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
+              resume_margins=None, resume_valid_margins=None,
+              snapshot_cb=None,
+              snapshot_interval_seconds: float = 1800.0,
+              max_duration_seconds: float = -1.0):
     """The boosting loop (reference gradient_boosted_trees.cc:1460).
 
     Returns (trees, init_preds, training_logs). For multinomial loss,
     trees are interleaved per class: tree t belongs to class t % n_classes
     (reference num_trees_per_iter semantics).
+
+    Checkpoint/resume (reference try_resume_training,
+    gradient_boosted_trees.cc:1403-1443): `snapshot_cb(trees, iteration)`
+    fires every snapshot_interval_seconds; a resumed run passes
+    start_iteration and the partial model's margins. KeyboardInterrupt
+    returns the model trained so far (reference stop_training_trigger_).
     """
+    import time as _time
+
     cfg = trainer.cfg
     dev = trainer.device
     N = trainer.N
@@ -462,6 +482,8 @@ def train_gbt(trainer: ForestTrainer, log=None):
     preds = torch.full((C, N), 0.0, dtype=torch.float32, device=dev)
     for c in range(C):
         preds[c].fill_(init_preds[c])
+    if resume_margins is not None:
+        preds.copy_(resume_margins)
 
     has_valid = trainer.valid_bins is not None
     if has_valid:
@@ -470,6 +492,8 @@ def train_gbt(trainer: ForestTrainer, log=None):
                                  device=dev)
         for c in range(C):
             valid_preds[c].fill_(init_preds[c])
+        if resume_valid_margins is not None:
+            valid_preds.copy_(resume_valid_margins)
         loss_buf = torch.zeros(2, dtype=torch.float32, device=dev)
 
     trees: List[HostTree] = []
@@ -477,14 +501,23 @@ def train_gbt(trainer: ForestTrainer, log=None):
     best_loss = math.inf
     best_num_trees = 0
     n_iters = cfg.num_trees
-    for it in range(n_iters):
+    t_start = _time.monotonic()
+    t_last_snapshot = t_start
+    interrupted = False
+    for it in range(start_iteration, n_iters):
+        if max_duration_seconds > 0 and \
+                _time.monotonic() - t_start > max_duration_seconds:
+            if log:
+                log(f"maximum_training_duration reached at iteration {it}")
+            break
         sample_mask = None
         if cfg.subsample < 1.0:
             sample_mask = (
                 torch.from_numpy(
                     trainer.rng.random_sample(N).astype(np.float32))
                 .to(dev) < cfg.subsample)
-        for c in range(C):
+        try:
+          for c in range(C):
             pc = preds[c]
             if multi:
                 ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
@@ -501,6 +534,17 @@ def train_gbt(trainer: ForestTrainer, log=None):
                                    trainer.valid_node_ids)
                 ops.update_preds(valid_preds[c], trainer.valid_node_ids,
                                  trainer.leaf_vals, cfg.shrinkage)
+        except KeyboardInterrupt:
+            if log:
+                log(f"interrupted at iteration {it}; returning partial model")
+            trees = trees[: it * C]  # drop this iteration's partial trees
+            interrupted = True
+            break
+        if snapshot_cb is not None and \
+                _time.monotonic() - t_last_snapshot >= \
+                snapshot_interval_seconds:
+            snapshot_cb(trees, it + 1, init_preds)
+            t_last_snapshot = _time.monotonic()
         if has_valid:
             vloss = _eval_loss(trainer, valid_preds, trainer.valid_labels,
                                cfg, loss_buf)
@@ -516,8 +560,11 @@ def train_gbt(trainer: ForestTrainer, log=None):
                     log(f"early stop at iteration {it + 1} "
                         f"(best={best_num_trees // C})")
                 break
-    if has_valid and cfg.early_stopping and best_num_trees > 0:
-        trees = trees[:best_num_trees]
+    if not interrupted and has_valid and cfg.early_stopping \
+            and best_num_trees > 0:
+        # trees holds only THIS run's trees; best_num_trees is global
+        keep = max(0, best_num_trees - start_iteration * C)
+        trees = trees[:keep]
     return trees, init_preds, logs
 
 

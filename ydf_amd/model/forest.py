@@ -154,3 +154,23 @@ def padded_boundaries(specs, max_bins: int = 256) -> np.ndarray:
         if b is not None and len(b):
             out[i, :min(len(b), n_cuts)] = b[:n_cuts]
     return out
+
+
+def concat_forests(a: FlatForest, b: FlatForest) -> FlatForest:
+    """Concatenates two flat forests (used by checkpoint/resume: partial
+    model + newly grown trees)."""
+    off = a.n_nodes
+    moff = len(a.masks)
+    return FlatForest(
+        feat=np.concatenate([a.feat, b.feat]),
+        thr=np.concatenate([a.thr, b.thr]),
+        left=np.concatenate([a.left,
+                             np.where(b.feat >= 0, b.left + off, 0)]),
+        roots=np.concatenate([a.roots, b.roots + off]).astype(np.int32),
+        cat_idx=np.concatenate([a.cat_idx,
+                                np.where(b.cat_idx >= 0, b.cat_idx + moff,
+                                         -1)]).astype(np.int32),
+        masks=np.concatenate([a.masks, b.masks]) if (len(a.masks)
+                                                     or len(b.masks))
+        else np.zeros((0, 4), np.uint64),
+    )

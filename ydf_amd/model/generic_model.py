@@ -192,6 +192,50 @@ class GenericModel:
         return evaluate_predictions(preds, labels, self._task, n_classes)
 
     # ------------------------------------------------------------------
+    def variable_importances(self) -> Dict:
+        """Structure-based variable importances (reference
+        AbstractModel::GetVariableImportance; SUM_SCORE requires a model
+        trained by this framework — it is persisted in the header)."""
+        from ydf_amd.utils import analysis as analysis_lib
+
+        return analysis_lib.structure_importances(self)
+
+    def analyze(self, data, sampling: float = 1.0,
+                num_bins: int = 20,
+                permutation_variable_importance_rounds: int = 1,
+                features: Optional[List[str]] = None, device=None):
+        """Full model analysis: variable importances + PDPs (mirrors ydf
+        model.analyze; reference utils/model_analysis.h:36-89)."""
+        from ydf_amd.utils import analysis as analysis_lib
+
+        labels = None
+        if self.dataspec.label is not None:
+            try:
+                cols = _to_column_dict(data) \
+                    if not isinstance(data, VerticalDataset) else None
+                if isinstance(data, VerticalDataset):
+                    labels = data.label_values
+                elif self.dataspec.label in cols:
+                    lspec = self.dataspec.label_column
+                    if lspec.semantic == Semantic.CATEGORICAL:
+                        lookup = {v: i for i, v in enumerate(lspec.vocab)}
+                        labels = np.fromiter(
+                            (lookup.get(s, 0) - 1
+                             for s in cols[self.dataspec.label].astype(str)),
+                            dtype=np.float32,
+                            count=len(cols[self.dataspec.label]))
+                    else:
+                        labels = np.asarray(cols[self.dataspec.label],
+                                            dtype=np.float32)
+            except Exception:
+                labels = None
+        return analysis_lib.analyze(
+            self, data, labels=labels,
+            permutation_variable_importance=(
+                permutation_variable_importance_rounds > 0),
+            features=features, num_grid_points=num_bins, device=device)
+
+    # ------------------------------------------------------------------
     def benchmark(self, data, benchmark_duration: float = 3.0,
                   warmup_duration: float = 0.5, batch_size: int = 0,
                   device=None):

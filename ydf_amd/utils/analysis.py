@@ -1,0 +1,201 @@
+"""Model analysis: variable importances, partial dependence, permutation
+importance.
+
+Capability analogue of the reference's utils/model_analysis.{h,cc} (Analyse
+PDP/CEP reports), utils/partial_dependence_plot.*, and
+utils/feature_importance.* (permutation importances,
+random_forest.cc:1411-1477).
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ydf_amd.dataset.dataspec import Semantic, Task
+
+
+@dataclasses.dataclass
+class PartialDependence:
+    feature: str
+    grid: np.ndarray           # grid values (numerical) or category ids
+    mean_prediction: np.ndarray
+    is_categorical: bool = False
+    categories: Optional[List[str]] = None
+
+
+@dataclasses.dataclass
+class Analysis:
+    """Analysis report (mirrors ydf model.analyze() output)."""
+
+    variable_importances: Dict[str, List[Tuple[float, str]]]
+    partial_dependences: List[PartialDependence]
+
+    def to_text(self) -> str:
+        out = []
+        for name, vi in self.variable_importances.items():
+            out.append(f"Variable importance: {name}")
+            for rank, (score, feat) in enumerate(vi, 1):
+                out.append(f"  {rank:3d}. {feat:30s} {score:.6g}")
+            out.append("")
+        for pd in self.partial_dependences[:10]:
+            out.append(f"PDP {pd.feature}: "
+                       + " ".join(f"{v:.3g}" for v in
+                                  pd.mean_prediction[:8]))
+        return "\n".join(out)
+
+    def __str__(self) -> str:
+        return self.to_text()
+
+    def _repr_html_(self) -> str:
+        rows = []
+        for name, vi in self.variable_importances.items():
+            body = "".join(
+                f"<tr><td>{f}</td><td>{s:.6g}</td></tr>" for s, f in vi)
+            rows.append(f"<h3>{name}</h3><table><tr><th>feature</th>"
+                        f"<th>score</th></tr>{body}</table>")
+        return "".join(rows)
+
+
+def structure_importances(model) -> Dict[str, List[Tuple[float, str]]]:
+    """Importances from the forest structure: number of nodes per feature
+    and number-of-times-root (reference NUM_NODES / NUM_AS_ROOT)."""
+    names = model.input_feature_names()
+    forest = model.forest
+    num_nodes = np.zeros(len(names), dtype=np.int64)
+    used = forest.feat[forest.feat >= 0]
+    np.add.at(num_nodes, used, 1)
+    num_root = np.zeros(len(names), dtype=np.int64)
+    root_feats = forest.feat[forest.roots]
+    np.add.at(num_root, root_feats[root_feats >= 0], 1)
+
+    def ranked(scores):
+        order = np.argsort(-scores, kind="stable")
+        return [(float(scores[i]), names[i]) for i in order if scores[i] > 0]
+
+    out = {
+        "NUM_NODES": ranked(num_nodes.astype(np.float64)),
+        "NUM_AS_ROOT": ranked(num_root.astype(np.float64)),
+    }
+    gains = model.metadata.get("feature_gains") if model.metadata else None
+    if gains:
+        out["SUM_SCORE"] = sorted(
+            ((float(v), k) for k, v in gains.items() if v > 0), reverse=True)
+    return out
+
+
+def permutation_importances(model, data, labels: np.ndarray,
+                            num_repetitions: int = 1,
+                            seed: int = 1234,
+                            device=None) -> List[Tuple[float, str]]:
+    """Mean metric drop when a feature column is shuffled (reference
+    MEAN_DECREASE_IN_ACCURACY / permutation variable importances)."""
+    from ydf_amd.metric.metric import accuracy, rmse
+
+    X = model._encode_features(data).copy()
+    names = model.input_feature_names()
+    rng = np.random.RandomState(seed)
+
+    def score(Xm):
+        import torch
+
+        dev = (torch.device(device) if device is not None
+               else (torch.device("cuda") if torch.cuda.is_available()
+                     else torch.device("cpu")))
+        Xt = torch.from_numpy(np.ascontiguousarray(Xm)).to(dev)
+        m = model.predict_margin(Xt)
+        p = model._apply_activation(m).cpu().numpy()
+        if model.task() == Task.CLASSIFICATION:
+            if p.ndim == 1:
+                return accuracy(labels.astype(np.int64),
+                                (p >= 0.5).astype(np.int64))
+            return accuracy(labels.astype(np.int64), p.argmax(axis=1))
+        return -rmse(labels, p)
+
+    base = score(X)
+    drops = []
+    for fi, name in enumerate(names):
+        drop = 0.0
+        saved = X[fi].copy()
+        for _ in range(num_repetitions):
+            X[fi] = saved[rng.permutation(X.shape[1])]
+            drop += base - score(X)
+        X[fi] = saved
+        drops.append((drop / num_repetitions, name))
+    drops.sort(reverse=True)
+    return drops
+
+
+def partial_dependences(model, data, features: Optional[List[str]] = None,
+                        num_grid_points: int = 20, max_examples: int = 5000,
+                        device=None) -> List[PartialDependence]:
+    import torch
+
+    X = model._encode_features(data)
+    if X.shape[1] > max_examples:
+        idx = np.random.RandomState(0).choice(X.shape[1], max_examples,
+                                              replace=False)
+        X = X[:, idx]
+    X = X.copy()
+    specs = model.dataspec.feature_columns
+    names = model.input_feature_names()
+    wanted = set(features) if features else None
+    dev = (torch.device(device) if device is not None
+           else (torch.device("cuda") if torch.cuda.is_available()
+                 else torch.device("cpu")))
+
+    def mean_pred(Xm):
+        Xt = torch.from_numpy(np.ascontiguousarray(Xm)).to(dev)
+        p = model._apply_activation(model.predict_margin(Xt)).cpu().numpy()
+        if p.ndim == 2:  # multi-class: track P(class index 1) like binary
+            p = p[:, min(1, p.shape[1] - 1)]
+        return float(p.mean())
+
+    out = []
+    for fi, spec in enumerate(specs):
+        if wanted is not None and spec.name not in wanted:
+            continue
+        saved = X[fi].copy()
+        if spec.semantic == Semantic.CATEGORICAL:
+            cats = list(range(min(spec.vocab_size, 16)))
+            means = []
+            for c in cats:
+                X[fi] = float(c)
+                means.append(mean_pred(X))
+            out.append(PartialDependence(
+                feature=spec.name, grid=np.asarray(cats, dtype=np.float32),
+                mean_prediction=np.asarray(means, dtype=np.float32),
+                is_categorical=True,
+                categories=[spec.vocab[c] for c in cats]))
+        else:
+            qs = np.linspace(0.02, 0.98, num_grid_points)
+            grid = np.quantile(saved, qs).astype(np.float32)
+            means = []
+            for v in grid:
+                X[fi] = v
+                means.append(mean_pred(X))
+            out.append(PartialDependence(
+                feature=spec.name, grid=grid,
+                mean_prediction=np.asarray(means, dtype=np.float32)))
+        X[fi] = saved
+    return out
+
+
+def analyze(model, data, labels: Optional[np.ndarray] = None,
+            permutation_variable_importance: bool = True,
+            partial_dependence: bool = True,
+            features: Optional[List[str]] = None,
+            num_grid_points: int = 20, device=None) -> Analysis:
+    vi = structure_importances(model)
+    if permutation_variable_importance and labels is not None:
+        key = ("MEAN_DECREASE_IN_ACCURACY"
+               if model.task() == Task.CLASSIFICATION
+               else "MEAN_INCREASE_IN_RMSE")
+        vi[key] = permutation_importances(model, data, labels, device=device)
+    pdps = []
+    if partial_dependence:
+        pdps = partial_dependences(model, data, features=features,
+                                   num_grid_points=num_grid_points,
+                                   device=device)
+    return Analysis(variable_importances=vi, partial_dependences=pdps)
