@@ -224,3 +224,32 @@ def test_variable_importances_and_analyze(binary_data):
     assert "MEAN_DECREASE_IN_ACCURACY" in an.variable_importances
     assert len(an.partial_dependences) == 3
     assert an._repr_html_()
+
+
+def test_example_weights(binary_data):
+    rng = np.random.RandomState(3)
+    d = dict(binary_data)
+    d["w"] = (rng.rand(len(d["x1"])) + 0.5).astype(np.float32)
+    m = ydf.GradientBoostedTreesLearner(label="label", weights="w",
+                                        num_trees=30).train(d)
+    assert "w" not in m.input_feature_names()
+    assert m.evaluate(binary_data).accuracy > 0.9
+
+
+def test_random_search_tuner(binary_data):
+    tuner = ydf.RandomSearchTuner(num_trials=3)
+    tuner.choice("shrinkage", [0.05, 0.2])
+    tuner.choice("max_depth", [3, 5])
+    m = ydf.GradientBoostedTreesLearner(label="label", tuner=tuner,
+                                        num_trees=20).train(binary_data)
+    assert m.tuner_logs is not None
+    assert len(m.tuner_logs.trials) == 3
+    assert m.tuner_logs.best_trial.score >= max(
+        t.score for t in m.tuner_logs.trials) - 1e-9
+
+
+def test_cross_validation(binary_data):
+    ev = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15).cross_validation(binary_data, folds=3)
+    assert ev.accuracy > 0.9
+    assert ev.num_examples == len(binary_data["x1"])

@@ -38,6 +38,13 @@ from ydf_amd.model.model_lib import (
 )
 from ydf_amd.model import tree
 
+# Tuner
+from ydf_amd.learner.tuner import (
+    OptimizerLogs,
+    RandomSearchTuner,
+    VizierTuner,
+)
+
 # Dataset
 from ydf_amd.dataset.dataset import VerticalDataset, create_vertical_dataset
 from ydf_amd.dataset.dataspec import (

@@ -19,12 +19,15 @@ class GenericLearner:
 
     def __init__(self, label: Optional[str], task: Task = Task.CLASSIFICATION,
                  features: Optional[Sequence[Union[str, Column]]] = None,
+                 weights: Optional[str] = None, tuner=None,
                  max_vocab_count: int = 2000, min_vocab_frequency: int = 1,
                  random_seed: int = 123456, device=None,
                  num_threads: Optional[int] = None):
         self.label = label
         self._task = task
         self.features = features
+        self.weights_col = weights
+        self.tuner = tuner
         self.max_vocab_count = max_vocab_count
         self.min_vocab_frequency = min_vocab_frequency
         self.random_seed = random_seed
@@ -68,13 +71,27 @@ class GenericLearner:
         """Dataset -> (VerticalDataset, binned u8 [F,N] on device,
         labels f32 [N] on device, padded boundary matrix np [F,n_cuts],
         cat_flags u8 tensor or None)."""
+        weights_np = None
         if isinstance(data, VerticalDataset):
             ds = data
         else:
+            from ydf_amd.dataset.dataset import _to_column_dict
+
+            features = self.features
+            cols = _to_column_dict(data)
+            if self.weights_col is not None and features is None:
+                features = [c for c in cols
+                            if c not in (self.label, self.weights_col)]
             ds = create_vertical_dataset(
-                data, label=self.label, task=self._task,
-                features=self.features, max_vocab_count=self.max_vocab_count,
+                cols, label=self.label, task=self._task,
+                features=features, max_vocab_count=self.max_vocab_count,
                 min_vocab_frequency=self.min_vocab_frequency)
+            if self.weights_col is not None:
+                if self.weights_col not in cols:
+                    raise ValueError(
+                        f"weights column {self.weights_col!r} missing")
+                weights_np = np.asarray(cols[self.weights_col],
+                                        dtype=np.float32)
         bnd = padded_boundaries(ds.dataspec.feature_columns)
         cat_feats = self._cat_feature_flags(ds)
         bins = self._bin_matrix(ds.X, cat_feats, bnd, device)
@@ -86,7 +103,17 @@ class GenericLearner:
         if cat_feats.any():
             cat_flags = torch.from_numpy(
                 cat_feats.astype(np.uint8)).to(device)
-        return ds, bins, labels, bnd, cat_flags
+        weights = None
+        if weights_np is not None:
+            # packed-u64 histogram path needs per-example h <= 16: scale
+            # down uniformly if needed (leaf values are scale-invariant at
+            # the default l2=0; documented in README)
+            mx = float(weights_np.max()) if len(weights_np) else 1.0
+            if mx > 8.0:
+                weights_np = weights_np * (8.0 / mx)
+            weights = torch.from_numpy(
+                np.ascontiguousarray(weights_np)).to(device)
+        return ds, bins, labels, bnd, cat_flags, weights
 
     def _label_classes(self, ds: VerticalDataset):
         if self._task != Task.CLASSIFICATION:
@@ -111,6 +138,80 @@ class GenericLearner:
 
     def train(self, data, valid=None, verbose=None):
         raise NotImplementedError
+
+    def cross_validation(self, data, folds: int = 10, seed: int = 1234):
+        """K-fold cross validation (reference utils/fold_generator +
+        AbstractLearner evaluation, abstract_learner.h:270). Returns the
+        pooled out-of-fold Evaluation."""
+        from ydf_amd.dataset.dataset import _to_column_dict
+        from ydf_amd.metric.metric import evaluate_predictions
+
+        cols = _to_column_dict(data)
+        n = len(next(iter(cols.values())))
+        rng = np.random.RandomState(seed)
+        perm = rng.permutation(n)
+        all_preds = None
+        all_labels = np.empty(n, dtype=np.float32)
+        for k in range(folds):
+            test_idx = perm[k::folds]
+            train_idx = np.setdiff1d(perm, test_idx)
+            tr = {c: v[train_idx] for c, v in cols.items()}
+            te = {c: v[test_idx] for c, v in cols.items()}
+            model = self.train(tr)
+            p = model.predict(te)
+            if all_preds is None:
+                all_preds = np.empty((n,) + p.shape[1:], dtype=np.float32)
+            all_preds[test_idx] = p
+            lspec = model.dataspec.label_column
+            from ydf_amd.dataset.dataspec import Semantic
+
+            if lspec.semantic == Semantic.CATEGORICAL:
+                lookup = {v: i for i, v in enumerate(lspec.vocab)}
+                all_labels[test_idx] = [
+                    lookup.get(s, 0) - 1
+                    for s in te[self.label].astype(str)]
+            else:
+                all_labels[test_idx] = te[self.label]
+        n_classes = 2
+        if model.label_classes:
+            n_classes = len(model.label_classes)
+        return evaluate_predictions(all_preds, all_labels, self._task,
+                                    n_classes)
+
+    def _train_with_tuner(self, data, valid=None):
+        """Random-search trials; keeps the best model by validation loss
+        (reference hyperparameters_optimizer.cc random trials)."""
+        from ydf_amd.learner.tuner import OptimizerLogs, TrialLog
+
+        rng = np.random.RandomState(self.tuner.seed)
+        tuner = self.tuner
+        best = None
+        logs = []
+        base_hp = dict(self.hyperparameters)
+        self.tuner = None  # avoid recursion
+        try:
+            for trial in range(tuner.num_trials):
+                hp = tuner.sample(rng)
+                self.hyperparameters = dict(base_hp)
+                self.hyperparameters.update(
+                    {k: v for k, v in hp.items()
+                     if k in self.hyperparameters})
+                if self.hyperparameters.get("validation_ratio", 0) == 0:
+                    self.hyperparameters["validation_ratio"] = 0.1
+                model = self.train(data, valid=valid)
+                vloss = None
+                if model.training_logs:
+                    vloss = model.training_logs[-1].get("valid_loss")
+                score = -(vloss if vloss is not None else float("inf"))
+                logs.append(TrialLog(hyperparameters=hp, score=score))
+                if best is None or score > best[0]:
+                    best = (score, model)
+        finally:
+            self.tuner = tuner
+            self.hyperparameters = base_hp
+        model = best[1]
+        model.tuner_logs = OptimizerLogs(trials=logs)
+        return model
 
     def validate_hyperparameters(self) -> None:
         pass

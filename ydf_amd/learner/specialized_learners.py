@@ -77,9 +77,12 @@ class GradientBoostedTreesLearner(GenericLearner):
 
     def train(self, data, valid=None, verbose=None
               ) -> GradientBoostedTreesModel:
+        if self.tuner is not None:
+            return self._train_with_tuner(data, valid=valid)
         hp = self.hyperparameters
         device = self._resolve_device()
-        ds, bins, labels, bnd, cat_flags = self._prepare(data, device)
+        ds, bins, labels, bnd, cat_flags, weights = self._prepare(
+            data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
         classes = self._label_classes(ds)
@@ -111,6 +114,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                 valid_labels = labels[vi].contiguous()
                 bins = bins[:, ti].contiguous()
                 labels = labels[ti].contiguous()
+                if weights is not None:
+                    weights = weights[ti].contiguous()
 
         F = bins.shape[0]
         ncand = 0
@@ -135,7 +140,7 @@ class GradientBoostedTreesLearner(GenericLearner):
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
                                       valid_bins=valid_bins,
                                       valid_labels=valid_labels,
-                                      cat_flags=cat_flags)
+                                      cat_flags=cat_flags, weights=weights)
         C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
         activation = "identity"
         if hp["apply_link_function"]:
@@ -310,9 +315,12 @@ class RandomForestLearner(GenericLearner):
         return max(1, F // 3)
 
     def train(self, data, valid=None, verbose=None) -> RandomForestModel:
+        if self.tuner is not None:
+            return self._train_with_tuner(data, valid=valid)
         hp = self.hyperparameters
         device = self._resolve_device()
-        ds, bins, labels, bnd, cat_flags = self._prepare(data, device)
+        ds, bins, labels, bnd, cat_flags, weights = self._prepare(
+            data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
         classes = self._label_classes(ds) \
@@ -328,7 +336,7 @@ class RandomForestLearner(GenericLearner):
             num_candidate_features=self._num_candidate(F),
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
-                                      cat_flags=cat_flags)
+                                      cat_flags=cat_flags, weights=weights)
         trees = trainer_lib.train_rf(t, log=info)
         flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
                                  cat_feats=self._cat_feature_flags(ds))

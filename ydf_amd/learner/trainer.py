@@ -84,11 +84,13 @@ class ForestTrainer:
                  cfg: TrainerConfig,
                  valid_bins: Optional[torch.Tensor] = None,
                  valid_labels: Optional[torch.Tensor] = None,
-                 cat_flags: Optional[torch.Tensor] = None):
+                 cat_flags: Optional[torch.Tensor] = None,
+                 weights: Optional[torch.Tensor] = None):
         assert bins.dtype == torch.uint8 and bins.dim() == 2
         self.bins = bins
         self.labels = labels
         self.cfg = cfg
+        self.weights = weights      # f32 [N] example weights (or None)
         self.cat_flags = cat_flags  # u8 [F] on device; None = all numerical
         self.has_cats = cat_flags is not None and bool(cat_flags.any())
         if not self.has_cats:
@@ -523,6 +525,10 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
             else:
                 ops.grad_hess(pc, y, trainer.gh, cfg.loss)
+            if trainer.weights is not None:
+                # weighted loss: g,h scale linearly with the example weight
+                # (reference dataset/weight.h GetWeights path)
+                trainer.gh.mul_(trainer.weights.view(-1, 1))
             tree = trainer.grow_tree(it * C + c, sample_mask)
             trees.append(tree)
             if sample_mask is not None:
@@ -622,6 +628,10 @@ def train_rf(trainer: ForestTrainer, log=None):
                 w = np.minimum(trainer.rng.poisson(1.0, size=N), 15).astype(
                     np.float32)
                 weights = torch.from_numpy(w).to(dev)
+        if trainer.weights is not None:
+            # user example weights compose with the bootstrap draw counts
+            weights = trainer.weights if weights is None \
+                else (weights * trainer.weights).clamp_(max=15)
         for c in range(C):
             if multi:
                 if onehot is None:
