@@ -1,0 +1,64 @@
+"""CLI smoke tests (reference cli/ layer)."""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope="module")
+def csv_path(tmp_path_factory):
+    pd = pytest.importorskip("pandas")
+    rng = np.random.RandomState(0)
+    n = 2000
+    df = pd.DataFrame({
+        "a": rng.randn(n), "b": rng.randn(n),
+        "label": np.where(rng.randn(n) + 2 * rng.randn(n) > 0, "p", "n"),
+    })
+    p = tmp_path_factory.mktemp("cli") / "data.csv"
+    df.to_csv(p, index=False)
+    return str(p)
+
+
+def run_cli(tool, *args):
+    return subprocess.run(
+        [sys.executable, "-m", f"ydf_amd.cli.{tool}", *args],
+        capture_output=True, text=True, cwd=REPO, timeout=300)
+
+
+def test_train_predict_evaluate_show(csv_path, tmp_path):
+    model_dir = str(tmp_path / "model")
+    r = run_cli("train", "--dataset", f"csv:{csv_path}", "--output",
+                model_dir, "--label", "label", "--hparams",
+                '{"num_trees": 10}')
+    assert r.returncode == 0, r.stderr
+    assert os.path.exists(os.path.join(model_dir, "done"))
+
+    out_csv = str(tmp_path / "preds.csv")
+    r = run_cli("predict", "--model", model_dir, "--dataset", csv_path,
+                "--output", out_csv)
+    assert r.returncode == 0, r.stderr
+    assert os.path.exists(out_csv)
+
+    r = run_cli("evaluate", "--model", model_dir, "--dataset", csv_path)
+    assert r.returncode == 0, r.stderr
+    assert "accuracy" in r.stdout
+
+    r = run_cli("show_model", "--model", model_dir, "--full_definition")
+    assert r.returncode == 0, r.stderr
+    assert "GRADIENT_BOOSTED_TREES" in r.stdout
+
+    r = run_cli("show_dataspec", "--model", model_dir)
+    assert r.returncode == 0, r.stderr
+
+    r = run_cli("infer_dataspec", "--dataset", csv_path, "--label", "label")
+    assert r.returncode == 0, r.stderr
+    assert "NUMERICAL" in r.stdout
+
+    r = run_cli("compute_variable_importances", "--model", model_dir,
+                "--dataset", csv_path)
+    assert r.returncode == 0, r.stderr
+    assert "SUM_SCORE" in r.stdout
