@@ -253,3 +253,46 @@ def test_cross_validation(binary_data):
         label="label", num_trees=15).cross_validation(binary_data, folds=3)
     assert ev.accuracy > 0.9
     assert ev.num_examples == len(binary_data["x1"])
+
+
+def test_goss_sampling(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=40,
+                                        sampling_method="GOSS").train(
+                                            binary_data)
+    assert m.evaluate(binary_data).accuracy > 0.9
+
+
+def test_rf_oob_and_winner_take_all(binary_data):
+    m = ydf.RandomForestLearner(label="label", num_trees=30,
+                                max_depth=10).train(binary_data)
+    ev = m.self_evaluation()
+    assert ev is not None and ev.accuracy > 0.9
+    # WTA leaves are votes in {0,1}; mean over trees stays in [0,1]
+    p = m.predict(binary_data)
+    assert p.min() >= 0 and p.max() <= 1
+
+
+def test_custom_regression_loss(regression_data):
+    def gh(labels, preds):
+        r = preds - labels
+        return np.clip(r, -1, 1), np.ones_like(r)
+
+    L = ydf.RegressionLoss(
+        gradient_and_hessian=gh,
+        initial_predictions=lambda y, w: float(np.median(y)),
+        loss=lambda y, p, w: float(np.abs(p - y).mean()))
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, loss=L, num_trees=60,
+        shrinkage=0.3).train(regression_data)
+    assert m.evaluate(regression_data).rmse < 1.0
+
+
+def test_custom_binary_loss(binary_data):
+    def gh(labels, preds):
+        p = 1.0 / (1.0 + np.exp(-preds))
+        return p - labels, np.maximum(p * (1 - p), 1e-6)
+
+    L = ydf.BinaryClassificationLoss(gradient_and_hessian=gh)
+    m = ydf.GradientBoostedTreesLearner(label="label", loss=L,
+                                        num_trees=40).train(binary_data)
+    assert m.evaluate(binary_data).accuracy > 0.9

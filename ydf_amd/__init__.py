@@ -38,6 +38,14 @@ from ydf_amd.model.model_lib import (
 )
 from ydf_amd.model import tree
 
+# Custom losses
+from ydf_amd.learner.custom_loss import (
+    Activation,
+    BinaryClassificationLoss,
+    MultiClassificationLoss,
+    RegressionLoss,
+)
+
 # Tuner
 from ydf_amd.learner.tuner import (
     OptimizerLogs,

@@ -38,6 +38,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                  features: Optional[Sequence[Union[str, Column]]] = None,
                  num_trees: int = 300, max_depth: int = 6,
                  shrinkage: float = 0.1, subsample: float = 1.0,
+                 sampling_method: str = "RANDOM",
+                 goss_alpha: float = 0.2, goss_beta: float = 0.1,
                  min_examples: int = 5, l2_regularization: float = 0.0,
                  min_sum_hessian_in_leaf: float = 1e-3,
                  validation_ratio: float = 0.1,
@@ -48,6 +50,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  use_hessian_gain: bool = True,
                  apply_link_function: bool = True,
                  l2_categorical_regularization: float = 1.0,
+                 loss: str = "DEFAULT",
                  working_dir: Optional[str] = None,
                  resume_training: bool = False,
                  resume_training_snapshot_interval_seconds: float = 1800.0,
@@ -57,7 +60,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                          random_seed=random_seed, **kwargs)
         self.hyperparameters = dict(
             num_trees=num_trees, max_depth=max_depth, shrinkage=shrinkage,
-            subsample=subsample, min_examples=min_examples,
+            subsample=subsample, sampling_method=sampling_method,
+            goss_alpha=goss_alpha, goss_beta=goss_beta,
+            min_examples=min_examples,
             l2_regularization=l2_regularization,
             min_sum_hessian_in_leaf=min_sum_hessian_in_leaf,
             validation_ratio=validation_ratio, early_stopping=early_stopping,
@@ -68,6 +73,7 @@ class GradientBoostedTreesLearner(GenericLearner):
             use_hessian_gain=use_hessian_gain,
             apply_link_function=apply_link_function,
             l2_categorical_regularization=l2_categorical_regularization,
+            loss=loss,
             working_dir=working_dir, resume_training=resume_training,
             resume_training_snapshot_interval_seconds=(
                 resume_training_snapshot_interval_seconds),
@@ -87,6 +93,10 @@ class GradientBoostedTreesLearner(GenericLearner):
             raise ValueError(f"label column {self.label!r} missing")
         classes = self._label_classes(ds)
         n_classes = len(classes) if classes else 2
+        custom_loss = None
+        if hp.get("loss") is not None and hp.get("loss") != "DEFAULT" \
+                and not isinstance(hp.get("loss"), str):
+            custom_loss = hp["loss"]
         if self._task == Task.CLASSIFICATION:
             loss = (trainer_lib.LOSS_MULTINOMIAL if n_classes > 2
                     else trainer_lib.LOSS_BINOMIAL)
@@ -127,7 +137,11 @@ class GradientBoostedTreesLearner(GenericLearner):
             shrinkage=hp["shrinkage"], lambda_l2=hp["l2_regularization"],
             min_examples=hp["min_examples"],
             min_hessian=hp["min_sum_hessian_in_leaf"],
-            subsample=hp["subsample"], n_classes=n_classes,
+            subsample=hp["subsample"],
+            sampling_method=hp.get("sampling_method", "RANDOM"),
+            goss_alpha=hp.get("goss_alpha", 0.2),
+            goss_beta=hp.get("goss_beta", 0.1),
+            n_classes=n_classes,
             seed=self.random_seed, num_candidate_features=ncand,
             early_stopping=(hp["early_stopping"] != "NONE"
                             and valid_bins is not None),
@@ -143,7 +157,10 @@ class GradientBoostedTreesLearner(GenericLearner):
                                       cat_flags=cat_flags, weights=weights)
         C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
         activation = "identity"
-        if hp["apply_link_function"]:
+        if custom_loss is not None:
+            activation = custom_loss.activation.value \
+                if hp["apply_link_function"] else "identity"
+        elif hp["apply_link_function"]:
             if loss == trainer_lib.LOSS_BINOMIAL:
                 activation = "sigmoid"
             elif loss == trainer_lib.LOSS_MULTINOMIAL:
@@ -227,6 +244,7 @@ class GradientBoostedTreesLearner(GenericLearner):
             t, log=info, start_iteration=start_it,
             resume_margins=resume_margins,
             resume_valid_margins=resume_valid_margins,
+            custom_loss=custom_loss,
             snapshot_cb=snapshot_cb,
             snapshot_interval_seconds=hp.get(
                 "resume_training_snapshot_interval_seconds", 1800.0),
@@ -283,8 +301,8 @@ class RandomForestLearner(GenericLearner):
                  bootstrap_size_ratio: float = 1.0,
                  num_candidate_attributes: int = 0,
                  num_candidate_attributes_ratio: float = -1.0,
-                 winner_take_all: bool = False,
-                 compute_oob_performances: bool = False,
+                 winner_take_all: bool = True,
+                 compute_oob_performances: bool = True,
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
@@ -337,7 +355,34 @@ class RandomForestLearner(GenericLearner):
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
                                       cat_flags=cat_flags, weights=weights)
-        trees = trainer_lib.train_rf(t, log=info)
+        compute_oob = (hp["compute_oob_performances"]
+                       and hp["bootstrap_training_dataset"])
+        result = trainer_lib.train_rf(t, log=info, compute_oob=compute_oob)
+        oob_eval = None
+        if compute_oob:
+            trees, oob_sum, oob_cnt = result
+            from ydf_amd.metric.metric import evaluate_predictions
+
+            cnt = oob_cnt.cpu().numpy()
+            covered = cnt > 0
+            if covered.any():
+                preds = (oob_sum.cpu().numpy()[:, covered]
+                         / cnt[covered]).T
+                if preds.shape[1] == 1:
+                    preds = preds[:, 0]
+                y = labels.cpu().numpy()[covered]
+                oob_eval = evaluate_predictions(
+                    preds, y, self._task, n_classes)
+        else:
+            trees = result
+        # winner_take_all (reference RF default): each tree votes its
+        # majority class; for binary trees this is a leaf-value threshold,
+        # so the vote transform happens at model-build time
+        wta = (hp["winner_take_all"]
+               and self._task == Task.CLASSIFICATION and n_classes == 2)
+        if wta:
+            for tr in trees:
+                tr.leaf_value = (tr.leaf_value > 0.5).astype(np.float32)
         flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
                                  cat_feats=self._cat_feature_flags(ds))
         C = n_classes if (classes and n_classes > 2) else 1
@@ -346,7 +391,9 @@ class RandomForestLearner(GenericLearner):
             label_classes=classes, init_predictions=[0.0] * max(C, 1),
             num_trees_per_iter=C, activation="identity",
             metadata={"feature_gains": self._feature_gains(
-                trees, [c.name for c in ds.dataspec.feature_columns])})
+                trees, [c.name for c in ds.dataspec.feature_columns]),
+                "winner_take_all": wta})
+        model._self_evaluation = oob_eval
         return model
 
 

@@ -44,6 +44,9 @@ class TrainerConfig:
     min_hessian: float = 1e-3
     min_gain: float = 0.0
     subsample: float = 1.0
+    sampling_method: str = "RANDOM"   # RANDOM | GOSS
+    goss_alpha: float = 0.2
+    goss_beta: float = 0.1
     cat_smooth: float = 1.0      # l2_categorical_regularization
     n_classes: int = 2           # multinomial only
     seed: int = 123456
@@ -441,7 +444,8 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
               resume_margins=None, resume_valid_margins=None,
               snapshot_cb=None,
               snapshot_interval_seconds: float = 1800.0,
-              max_duration_seconds: float = -1.0):
+              max_duration_seconds: float = -1.0,
+              custom_loss=None):
     """The boosting loop (reference gradient_boosted_trees.cc:1460).
 
     Returns (trees, init_preds, training_logs). For multinomial loss,
@@ -466,7 +470,16 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
     # initial predictions (reference loss->InitialPredictions,
     # gradient_boosted_trees.cc:1329)
     counts = torch.tensor([float(N)], device=dev)
-    if cfg.loss == LOSS_BINOMIAL:
+    if custom_loss is not None:
+        from ydf_amd.learner.custom_loss import default_initial_predictions
+
+        if trainer.distributed:
+            raise NotImplementedError(
+                "custom losses are not supported with multi-process "
+                "training (host callbacks)")
+        init = default_initial_predictions(custom_loss, y.cpu().numpy())
+        init_preds = [init] * C
+    elif cfg.loss == LOSS_BINOMIAL:
         s = torch.stack([y.sum(), counts[0]])
         trainer._allreduce(s)
         p = (s[0] / s[1]).clamp(1e-6, 1 - 1e-6)
@@ -513,15 +526,29 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 log(f"maximum_training_duration reached at iteration {it}")
             break
         sample_mask = None
-        if cfg.subsample < 1.0:
+        if cfg.sampling_method != "GOSS" and cfg.subsample < 1.0:
             sample_mask = (
                 torch.from_numpy(
                     trainer.rng.random_sample(N).astype(np.float32))
                 .to(dev) < cfg.subsample)
+        custom_gh = None
+        if custom_loss is not None:
+            y_np = y.cpu().numpy()
+            p_np = (preds[0] if C == 1 else preds).cpu().numpy()
+            g_np, h_np = custom_loss.gradient_and_hessian(y_np, p_np)
+            custom_gh = (np.asarray(g_np, dtype=np.float32),
+                         np.asarray(h_np, dtype=np.float32))
         try:
           for c in range(C):
             pc = preds[c]
-            if multi:
+            if custom_gh is not None:
+                g_np, h_np = custom_gh
+                gc = g_np[c] if g_np.ndim == 2 else g_np
+                hc = h_np[c] if h_np.ndim == 2 else h_np
+                trainer.gh.copy_(torch.from_numpy(
+                    np.stack([gc, np.clip(hc, 1e-16, 16.0)],
+                             axis=1)).to(dev))
+            elif multi:
                 ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
             else:
                 ops.grad_hess(pc, y, trainer.gh, cfg.loss)
@@ -529,6 +556,26 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 # weighted loss: g,h scale linearly with the example weight
                 # (reference dataset/weight.h GetWeights path)
                 trainer.gh.mul_(trainer.weights.view(-1, 1))
+            if cfg.sampling_method == "GOSS":
+                # Gradient-based one-side sampling (reference
+                # gradient_boosted_trees.cc:1488-1522 GOSS): keep the top
+                # alpha fraction by |g|, sample beta of the rest with
+                # (1-alpha)/beta amplification
+                absg = trainer.gh[:, 0].abs()
+                k = max(1, int(cfg.goss_alpha * N))
+                thr = torch.kthvalue(
+                    absg, max(1, N - k)).values if N > 1 else absg.min()
+                rnd = torch.from_numpy(
+                    trainer.rng.random_sample(N).astype(np.float32)).to(dev)
+                big = absg >= thr
+                small_kept = (~big) & (rnd < cfg.goss_beta)
+                sample_mask = big | small_kept
+                amp = (1.0 - cfg.goss_alpha) / max(cfg.goss_beta, 1e-9)
+                scale = torch.where(
+                    small_kept,
+                    torch.full((), amp, device=dev),
+                    torch.ones((), device=dev))
+                trainer.gh.mul_(scale.view(-1, 1))
             tree = trainer.grow_tree(it * C + c, sample_mask)
             trees.append(tree)
             if sample_mask is not None:
@@ -552,8 +599,17 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             snapshot_cb(trees, it + 1, init_preds)
             t_last_snapshot = _time.monotonic()
         if has_valid:
-            vloss = _eval_loss(trainer, valid_preds, trainer.valid_labels,
-                               cfg, loss_buf)
+            if custom_loss is not None and custom_loss.loss is not None:
+                vl_np = (valid_preds[0] if C == 1
+                         else valid_preds).cpu().numpy()
+                vy_np = trainer.valid_labels.cpu().numpy()
+                vloss = float(custom_loss.loss(
+                    vy_np, vl_np, np.ones_like(vy_np)))
+            elif custom_loss is not None:
+                vloss = float("nan")
+            else:
+                vloss = _eval_loss(trainer, valid_preds,
+                                   trainer.valid_labels, cfg, loss_buf)
             logs.append({"iteration": it + 1, "valid_loss": vloss})
             if vloss < best_loss:
                 best_loss = vloss
@@ -599,13 +655,18 @@ def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
     return float((s[0] / s[1]).item())
 
 
-def train_rf(trainer: ForestTrainer, log=None):
+def train_rf(trainer: ForestTrainer, log=None,
+             compute_oob: bool = False):
     """Random-forest bagging loop (reference random_forest.cc:917).
 
     Binary classification / regression: target mean leaves. Multi-class:
     one tree per class per iteration (probability forest). Bootstrap is
     Poisson(1)-approximated (documented deviation from the reference's exact
     multinomial resampling; same expectation).
+
+    With compute_oob, accumulates out-of-bag predictions (rows whose
+    bootstrap weight was 0 for a tree; reference OOB evaluations,
+    random_forest.cc:557) and returns (trees, oob_pred [C,N], oob_cnt [N]).
     """
     cfg = trainer.cfg
     dev = trainer.device
@@ -614,6 +675,10 @@ def train_rf(trainer: ForestTrainer, log=None):
     C = cfg.n_classes if multi else 1
     trees: List[HostTree] = []
     onehot = None
+    oob_sum = oob_cnt = None
+    if compute_oob and cfg.bootstrap:
+        oob_sum = torch.zeros((C, N), dtype=torch.float32, device=dev)
+        oob_cnt = torch.zeros(N, dtype=torch.float32, device=dev)
     for it in range(cfg.num_trees):
         weights = None
         if cfg.bootstrap:
@@ -642,6 +707,16 @@ def train_rf(trainer: ForestTrainer, log=None):
                 ops.weighted_target(trainer.labels, weights, trainer.gh)
             tree = trainer.grow_tree(it * C + c)
             trees.append(tree)
+            if oob_sum is not None:
+                # node_ids already hold every row's leaf (zero-weight rows
+                # route but do not contribute to histograms)
+                lv = trainer.leaf_vals[trainer.node_ids.long()]
+                oob = (weights == 0).float()
+                oob_sum[c].add_(lv * oob)
+                if c == 0:
+                    oob_cnt.add_(oob)
         if log and (it + 1) % 100 == 0:
             log(f"trained {it + 1}/{cfg.num_trees} trees")
+    if oob_sum is not None:
+        return trees, oob_sum, oob_cnt
     return trees

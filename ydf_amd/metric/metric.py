@@ -61,10 +61,12 @@ def log_loss(labels: np.ndarray, probs: np.ndarray,
              eps: float = 1e-12) -> float:
     """labels: int class idx; probs: [N] (binary, P(class1)) or [N,C]."""
     if probs.ndim == 1:
-        p = np.clip(probs, eps, 1 - eps)
+        # float64 first: clipping float32 probs against 1-1e-12 rounds to 1.0
+        p = np.clip(probs.astype(np.float64), eps, 1 - eps)
         y = labels.astype(np.float64)
         return float(-(y * np.log(p) + (1 - y) * np.log(1 - p)).mean())
-    p = np.clip(probs[np.arange(len(labels)), labels.astype(int)], eps, 1.0)
+    p = np.clip(probs[np.arange(len(labels)),
+                      labels.astype(int)].astype(np.float64), eps, 1.0)
     return float(-np.log(p).mean())
 
 

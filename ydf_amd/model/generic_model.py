@@ -72,6 +72,8 @@ class GenericModel:
         self.metadata = metadata or {}
         self._dev_forest: Dict[str, _DeviceForest] = {}
         self.training_logs = None
+        self.tuner_logs = None
+        self._self_evaluation = None
 
     # ------------------------------------------------------------------
     def task(self) -> Task:
@@ -97,6 +99,18 @@ class GenericModel:
 
     def num_nodes(self) -> int:
         return self.forest.n_nodes
+
+    def self_evaluation(self):
+        """Out-of-bag evaluation (RF) or last validation evaluation (GBT);
+        mirrors ydf model.self_evaluation (reference OOB evaluations,
+        random_forest.cc:557 / GBT validation logs)."""
+        if self._self_evaluation is not None:
+            return self._self_evaluation
+        if self.training_logs:
+            from ydf_amd.metric.metric import Evaluation
+
+            return Evaluation(loss=self.training_logs[-1].get("valid_loss"))
+        return None
 
     # ------------------------------------------------------------------
     def _encode_features(self, data) -> np.ndarray:
