@@ -45,24 +45,38 @@ __global__ void predict_forest_lds_kernel(
   __syncthreads();
   if (tid >= n_here) return;
   float acc = init;
-  for (int tt = 0; tt < n_trees; ++tt) {
-    int n = roots[tree_start + (int64_t)tt * tree_step];
-    int f = feat[n];
-    while (f >= 0) {
-      const float xv = xs[f * kTile + tid];
-      int right;
-      const int ci = cat_idx ? cat_idx[n] : -1;
-      if (ci >= 0) {
-        int c = (int)xv;
-        c = c < 0 ? 0 : (c > 255 ? 255 : c);
-        right = (int)((masks[(int64_t)ci * 4 + (c >> 6)] >> (c & 63)) & 1ull);
-      } else {
-        right = xv > thr[n] ? 1 : 0;
+  if (cat_idx == nullptr) {
+    // pure-numerical fast path: tight compare-and-descend loop
+    for (int tt = 0; tt < n_trees; ++tt) {
+      int n = roots[tree_start + (int64_t)tt * tree_step];
+      int f = feat[n];
+      while (f >= 0) {
+        n = left[n] + (xs[f * kTile + tid] > thr[n] ? 1 : 0);
+        f = feat[n];
       }
-      n = left[n] + right;
-      f = feat[n];
+      acc += thr[n];
     }
-    acc += thr[n];
+  } else {
+    for (int tt = 0; tt < n_trees; ++tt) {
+      int n = roots[tree_start + (int64_t)tt * tree_step];
+      int f = feat[n];
+      while (f >= 0) {
+        const float xv = xs[f * kTile + tid];
+        int right;
+        const int ci = cat_idx[n];
+        if (ci >= 0) {
+          int c = (int)xv;
+          c = c < 0 ? 0 : (c > 255 ? 255 : c);
+          right =
+              (int)((masks[(int64_t)ci * 4 + (c >> 6)] >> (c & 63)) & 1ull);
+        } else {
+          right = xv > thr[n] ? 1 : 0;
+        }
+        n = left[n] + right;
+        f = feat[n];
+      }
+      acc += thr[n];
+    }
   }
   out[base + tid] = init + (acc - init) * scale;
 }
@@ -80,25 +94,37 @@ __global__ void predict_forest_global_kernel(
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t k = i; k < N; k += stride) {
     float acc = init;
-    for (int tt = 0; tt < n_trees; ++tt) {
-      int n = roots[tree_start + (int64_t)tt * tree_step];
-      int f = feat[n];
-      while (f >= 0) {
-        const float xv = X[(int64_t)f * N + k];
-        int right;
-        const int ci = cat_idx ? cat_idx[n] : -1;
-        if (ci >= 0) {
-          int c = (int)xv;
-          c = c < 0 ? 0 : (c > 255 ? 255 : c);
-          right =
-              (int)((masks[(int64_t)ci * 4 + (c >> 6)] >> (c & 63)) & 1ull);
-        } else {
-          right = xv > thr[n] ? 1 : 0;
+    if (cat_idx == nullptr) {
+      for (int tt = 0; tt < n_trees; ++tt) {
+        int n = roots[tree_start + (int64_t)tt * tree_step];
+        int f = feat[n];
+        while (f >= 0) {
+          n = left[n] + (X[(int64_t)f * N + k] > thr[n] ? 1 : 0);
+          f = feat[n];
         }
-        n = left[n] + right;
-        f = feat[n];
+        acc += thr[n];
       }
-      acc += thr[n];
+    } else {
+      for (int tt = 0; tt < n_trees; ++tt) {
+        int n = roots[tree_start + (int64_t)tt * tree_step];
+        int f = feat[n];
+        while (f >= 0) {
+          const float xv = X[(int64_t)f * N + k];
+          int right;
+          const int ci = cat_idx[n];
+          if (ci >= 0) {
+            int c = (int)xv;
+            c = c < 0 ? 0 : (c > 255 ? 255 : c);
+            right = (int)((masks[(int64_t)ci * 4 + (c >> 6)]
+                           >> (c & 63)) & 1ull);
+          } else {
+            right = xv > thr[n] ? 1 : 0;
+          }
+          n = left[n] + right;
+          f = feat[n];
+        }
+        acc += thr[n];
+      }
     }
     out[k] = init + (acc - init) * scale;
   }
