@@ -401,7 +401,8 @@ class ForestTrainer:
         hist_view = self.hist[:level_size]
         hist_view.zero_()
         ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
-                       hist_view, level_base, level_size, 0, level_size)
+                       hist_view, level_base, level_size, 0, level_size,
+                       filtered_hint=use_sub)
         self._allreduce(hist_view)
         if use_sub and derived is not None:
             ops.subtract_hist(hist_view, self.hist_prev, derived, level_size)

@@ -166,6 +166,7 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
                                       int F, int n_bins, int level_base,
                                       int level_size, int slot0, int n_slots,
                                       int lds_map, int filtered,
+                                      int n_chunks, int swizzle,
                                       int64_t rows_per_block) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   // carve: [n_slots*n_bins] {f64 g, u64 h|count}, then [level_size] i32 map
@@ -174,7 +175,23 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
       reinterpret_cast<unsigned long long*>(smem) + 1;  // interleaved pairs
   const int tot = n_slots * n_bins;
   int* lmap = reinterpret_cast<int*>(smem + (size_t)tot * 16);
-  const int f = blockIdx.x;
+  // XCD-aware mapping (cdna_hip_programming.md T1): the dispatcher places
+  // block b on XCD b%8, so id%8 selects the row chunk and consecutive
+  // ids on one XCD sweep the FEATURES of that chunk — the chunk's gh /
+  // node_ids stay hot in that XCD's private L2 across all F feature
+  // passes (heuristic placement: affects speed only, never correctness).
+  int f, chunk;
+  if (swizzle) {
+    const int id = blockIdx.x;
+    const int xcd = id & 7;
+    const int sid = id >> 3;
+    chunk = xcd + 8 * (sid / F);
+    f = sid - (sid / F) * F;
+    if (chunk >= n_chunks) return;
+  } else {
+    f = blockIdx.x;
+    chunk = blockIdx.y;
+  }
   {
     unsigned long long* z = reinterpret_cast<unsigned long long*>(smem);
     for (int i = threadIdx.x; i < tot * 2; i += blockDim.x) z[i] = 0ull;
@@ -184,7 +201,7 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
       lmap[i] = slot_map[i];
   }
   __syncthreads();
-  const int64_t row0 = (int64_t)blockIdx.y * rows_per_block;
+  const int64_t row0 = (int64_t)chunk * rows_per_block;
   const int64_t row1 = min(row0 + rows_per_block, N);
   const uint8_t* fb = bins + (int64_t)f * N;
   const int64_t stride = blockDim.x;
@@ -754,7 +771,8 @@ void gpu_weighted_target(const float* labels, const float* weights, float* gh,
 void gpu_hist_build(const uint8_t* bins, const float* gh,
                     const int32_t* node_ids, const int32_t* slot_map,
                     float* hist, int64_t N, int F, int n_bins, int level_base,
-                    int level_size, int slot0, int n_slots, void* stream) {
+                    int level_size, int slot0, int n_slots,
+                    int filtered_hint, void* stream) {
   // Stage the level's slot map in LDS when it fits comfortably (removes a
   // dependent global load per row visit); cap at 32 KiB so the histogram
   // region keeps >= ~32 slots at 16 B/bin.
@@ -765,19 +783,42 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   if (max_lds_slots < 1) max_lds_slots = 1;
   const int group = n_slots < max_lds_slots ? n_slots : max_lds_slots;
   const int threads = hist_block_threads();
-  const int chunks = row_chunks(N, F, 8192 * 256 / threads);
+  int chunks = row_chunks(N, F, 8192 * 256 / threads);
+  // keep one chunk's gh (8 B/row) within a single XCD's 4 MiB L2 so the
+  // swizzled feature sweep re-reads it from L2, not HBM
+  {
+    const int64_t max_rpb = (3 * 1024 * 1024) / 8;
+    const int min_chunks = (int)((N + max_rpb - 1) / max_rpb);
+    if (chunks < min_chunks) chunks = min_chunks;
+  }
+  static int swizzle = -1;
+  if (swizzle < 0) {
+    const char* e = getenv("YDFA_HIST_SWIZZLE");
+    swizzle = e ? atoi(e) : 1;
+  }
   const int64_t rpb = (N + chunks - 1) / chunks;
+  const int chunk_groups = (chunks + 7) / 8;
+  const int grid_flat = chunk_groups * 8 * F;
   for (int s0 = 0; s0 < n_slots; s0 += group) {
     const int ng = (n_slots - s0) < group ? (n_slots - s0) : group;
     const size_t lds = (size_t)ng * n_bins * 16 +
                        (lds_map ? map_bytes_full : 0);
-    const int filtered = (n_slots > group) ? 1 : 0;
-    hipLaunchKernelGGL(hist_build_lds_kernel, dim3(F, chunks), dim3(threads),
-                       lds, (hipStream_t)stream, bins, (const float2*)gh,
-                       node_ids, slot_map,
-                       hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
-                       level_base, level_size, slot0 + s0, ng, lds_map,
-                       filtered, rpb);
+    const int filtered = (n_slots > group || filtered_hint) ? 1 : 0;
+    if (swizzle) {
+      hipLaunchKernelGGL(hist_build_lds_kernel, dim3(grid_flat),
+                         dim3(threads), lds, (hipStream_t)stream, bins,
+                         (const float2*)gh, node_ids, slot_map,
+                         hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
+                         level_base, level_size, slot0 + s0, ng, lds_map,
+                         filtered, chunks, 1, rpb);
+    } else {
+      hipLaunchKernelGGL(hist_build_lds_kernel, dim3(F, chunks),
+                         dim3(threads), lds, (hipStream_t)stream, bins,
+                         (const float2*)gh, node_ids, slot_map,
+                         hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
+                         level_base, level_size, slot0 + s0, ng, lds_map,
+                         filtered, chunks, 0, rpb);
+    }
   }
 }
 
