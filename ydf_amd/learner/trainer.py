@@ -402,7 +402,8 @@ class ForestTrainer:
         hist_view.zero_()
         ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
                        hist_view, level_base, level_size, 0, level_size,
-                       filtered_hint=use_sub)
+                       filtered_hint=use_sub and os.environ.get(
+                           "YDFA_HIST_FILTER_SUB", "0") == "1")
         self._allreduce(hist_view)
         if use_sub and derived is not None:
             ops.subtract_hist(hist_view, self.hist_prev, derived, level_size)

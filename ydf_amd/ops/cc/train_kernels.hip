@@ -787,7 +787,12 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   // keep one chunk's gh (8 B/row) within a single XCD's 4 MiB L2 so the
   // swizzled feature sweep re-reads it from L2, not HBM
   {
-    const int64_t max_rpb = (3 * 1024 * 1024) / 8;
+    static int64_t max_rpb = -1;
+    if (max_rpb < 0) {
+      const char* e = getenv("YDFA_HIST_CHUNK_KB");  // gh bytes per chunk
+      max_rpb = (e ? atoi(e) : 3072) * 1024 / 8;
+      if (max_rpb <= 0) max_rpb = 1 << 30;
+    }
     const int min_chunks = (int)((N + max_rpb - 1) / max_rpb);
     if (chunks < min_chunks) chunks = min_chunks;
   }
