@@ -165,33 +165,37 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
                                       float* __restrict__ hist, int64_t N,
                                       int F, int n_bins, int level_base,
                                       int level_size, int slot0, int n_slots,
-                                      int lds_map, int filtered,
-                                      int n_chunks, int swizzle,
-                                      int64_t rows_per_block) {
+                                      int lds_map, int filtered, int fpb,
+                                      int n_fgroups, int n_chunks,
+                                      int swizzle, int64_t rows_per_block) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  // carve: [n_slots*n_bins] {f64 g, u64 h|count}, then [level_size] i32 map
+  // carve: [fpb][n_slots][n_bins] {f64 g, u64 h|count}, then the slot map.
+  // fpb > 1 ("multi-feature blocks") amortizes the gh/node_ids streams:
+  // one pass over the rows feeds fpb features' histograms.
   double* lg = reinterpret_cast<double*>(smem);
   unsigned long long* lp =
       reinterpret_cast<unsigned long long*>(smem) + 1;  // interleaved pairs
-  const int tot = n_slots * n_bins;
+  const int tot = fpb * n_slots * n_bins;
   int* lmap = reinterpret_cast<int*>(smem + (size_t)tot * 16);
   // XCD-aware mapping (cdna_hip_programming.md T1): the dispatcher places
-  // block b on XCD b%8, so id%8 selects the row chunk and consecutive
-  // ids on one XCD sweep the FEATURES of that chunk — the chunk's gh /
-  // node_ids stay hot in that XCD's private L2 across all F feature
-  // passes (heuristic placement: affects speed only, never correctness).
-  int f, chunk;
+  // block b on XCD b%8, so id%8 selects the row chunk and consecutive ids
+  // on one XCD sweep the feature groups of that chunk — the chunk's gh and
+  // node_ids stay hot in that XCD's private L2 (placement heuristic:
+  // affects speed only, never correctness).
+  int fgroup, chunk;
   if (swizzle) {
     const int id = blockIdx.x;
     const int xcd = id & 7;
     const int sid = id >> 3;
-    chunk = xcd + 8 * (sid / F);
-    f = sid - (sid / F) * F;
+    chunk = xcd + 8 * (sid / n_fgroups);
+    fgroup = sid - (sid / n_fgroups) * n_fgroups;
     if (chunk >= n_chunks) return;
   } else {
-    f = blockIdx.x;
+    fgroup = blockIdx.x;
     chunk = blockIdx.y;
   }
+  const int f0 = fgroup * fpb;
+  const int nf = (F - f0) < fpb ? (F - f0) : fpb;
   {
     unsigned long long* z = reinterpret_cast<unsigned long long*>(smem);
     for (int i = threadIdx.x; i < tot * 2; i += blockDim.x) z[i] = 0ull;
@@ -203,7 +207,6 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
   __syncthreads();
   const int64_t row0 = (int64_t)chunk * rows_per_block;
   const int64_t row1 = min(row0 + rows_per_block, N);
-  const uint8_t* fb = bins + (int64_t)f * N;
   const int64_t stride = blockDim.x;
   int64_t i = row0 + threadIdx.x;
   const int64_t bulk_end = row1 - (kHistUnroll - 1) * stride;
@@ -224,61 +227,93 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
         const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
         if (slot < 0 || slot >= n_slots) continue;
         const float2 v = gh[i + u * stride];
-        const int cell = 2 * (slot * n_bins + (int)fb[i + u * stride]);
-        atomicAdd(lg + cell, (double)v.x);
         const unsigned long long hq =
             (unsigned long long)(v.y * kHScale + 0.5f);
-        atomicAdd(lp + cell,
-                  hq | ((unsigned long long)(v.y != 0.f) << 44));
+        const unsigned long long pk =
+            hq | ((unsigned long long)(v.y != 0.f) << 44);
+        for (int j = 0; j < nf; ++j) {
+          const int b = bins[(int64_t)(f0 + j) * N + i + u * stride];
+          const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+          atomicAdd(lg + cell, (double)v.x);
+          atomicAdd(lp + cell, pk);
+        }
+      }
+    }
+    for (; i < row1; i += stride) {
+      const int rel = node_ids[i] - level_base;
+      if (rel < 0 || rel >= level_size) continue;
+      const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+      if (slot < 0 || slot >= n_slots) continue;
+      const float2 v = gh[i];
+      const unsigned long long hq =
+          (unsigned long long)(v.y * kHScale + 0.5f);
+      const unsigned long long pk =
+          hq | ((unsigned long long)(v.y != 0.f) << 44);
+      for (int j = 0; j < nf; ++j) {
+        const int b = bins[(int64_t)(f0 + j) * N + i];
+        const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+        atomicAdd(lg + cell, (double)v.x);
+        atomicAdd(lp + cell, pk);
       }
     }
   } else {
     for (; i < bulk_end; i += kHistUnroll * stride) {
       int nid[kHistUnroll];
       float2 v[kHistUnroll];
-      uint8_t b[kHistUnroll];
 #pragma unroll
       for (int u = 0; u < kHistUnroll; ++u)
         nid[u] = node_ids[i + u * stride];
 #pragma unroll
       for (int u = 0; u < kHistUnroll; ++u) v[u] = gh[i + u * stride];
+      for (int j = 0; j < nf; ++j) {
+        const uint8_t* fb = bins + (int64_t)(f0 + j) * N;
+        uint8_t b[kHistUnroll];
 #pragma unroll
-      for (int u = 0; u < kHistUnroll; ++u) b[u] = fb[i + u * stride];
+        for (int u = 0; u < kHistUnroll; ++u) b[u] = fb[i + u * stride];
 #pragma unroll
-      for (int u = 0; u < kHistUnroll; ++u) {
-        const int rel = nid[u] - level_base;
-        if (rel < 0 || rel >= level_size) continue;
-        const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
-        if (slot < 0 || slot >= n_slots) continue;
-        const int cell = 2 * (slot * n_bins + (int)b[u]);
-        atomicAdd(lg + cell, (double)v[u].x);
-        const unsigned long long hq =
-            (unsigned long long)(v[u].y * kHScale + 0.5f);
-        atomicAdd(lp + cell,
-                  hq | ((unsigned long long)(v[u].y != 0.f) << 44));
+        for (int u = 0; u < kHistUnroll; ++u) {
+          const int rel = nid[u] - level_base;
+          if (rel < 0 || rel >= level_size) continue;
+          const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+          if (slot < 0 || slot >= n_slots) continue;
+          const int cell = 2 * (((j * n_slots) + slot) * n_bins + (int)b[u]);
+          atomicAdd(lg + cell, (double)v[u].x);
+          const unsigned long long hq =
+              (unsigned long long)(v[u].y * kHScale + 0.5f);
+          atomicAdd(lp + cell,
+                    hq | ((unsigned long long)(v[u].y != 0.f) << 44));
+        }
       }
     }
-  }
-  for (; i < row1; i += stride) {
-    const int rel = node_ids[i] - level_base;
-    if (rel < 0 || rel >= level_size) continue;
-    const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
-    if (slot < 0 || slot >= n_slots) continue;
-    const float2 v = gh[i];
-    const int cell = 2 * (slot * n_bins + (int)fb[i]);
-    atomicAdd(lg + cell, (double)v.x);
-    const unsigned long long hq =
-        (unsigned long long)(v.y * kHScale + 0.5f);
-    atomicAdd(lp + cell, hq | ((unsigned long long)(v.y != 0.f) << 44));
+    for (; i < row1; i += stride) {
+      const int rel = node_ids[i] - level_base;
+      if (rel < 0 || rel >= level_size) continue;
+      const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+      if (slot < 0 || slot >= n_slots) continue;
+      const float2 v = gh[i];
+      const unsigned long long hq =
+          (unsigned long long)(v.y * kHScale + 0.5f);
+      const unsigned long long pk =
+          hq | ((unsigned long long)(v.y != 0.f) << 44);
+      for (int j = 0; j < nf; ++j) {
+        const int b = bins[(int64_t)(f0 + j) * N + i];
+        const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+        atomicAdd(lg + cell, (double)v.x);
+        atomicAdd(lp + cell, pk);
+      }
+    }
   }
   __syncthreads();
   for (int k = threadIdx.x; k < tot; k += blockDim.x) {
     const double g = lg[2 * k];
     const unsigned long long pk = lp[2 * k];
     if (pk == 0ull && g == 0.0) continue;  // untouched bin
-    const int slot = k / n_bins;
-    const int bin = k - slot * n_bins;
-    float* p = hist + ((int64_t)slot * F + f) * (n_bins * 3) + bin * 3;
+    const int j = k / (n_slots * n_bins);
+    const int rem = k - j * (n_slots * n_bins);
+    const int slot = rem / n_bins;
+    const int bin = rem - slot * n_bins;
+    if (f0 + j >= F) continue;
+    float* p = hist + ((int64_t)slot * F + (f0 + j)) * (n_bins * 3) + bin * 3;
     atomicAdd(p, (float)g);
     // double-precision unpack: one rounding to f32 (keeps integer-valued
     // h sums exact, which the CPU/GPU equality tests rely on)
@@ -782,10 +817,24 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   int max_lds_slots = (int)(budget / ((size_t)n_bins * 16));
   if (max_lds_slots < 1) max_lds_slots = 1;
   const int group = n_slots < max_lds_slots ? n_slots : max_lds_slots;
+  // multi-feature blocks: when the whole level fits with room to spare,
+  // each block histograms fpb features from ONE pass over the rows
+  static int max_fpb = -1;
+  if (max_fpb < 0) {
+    const char* e = getenv("YDFA_HIST_FPB");
+    max_fpb = e ? atoi(e) : 8;
+    if (max_fpb < 1) max_fpb = 1;
+  }
+  int fpb = 1;
+  if (group == n_slots) {
+    fpb = (int)(budget / ((size_t)n_slots * n_bins * 16));
+    if (fpb > max_fpb) fpb = max_fpb;
+    if (fpb > F) fpb = F;
+    if (fpb < 1) fpb = 1;
+  }
+  const int n_fgroups = (F + fpb - 1) / fpb;
   const int threads = hist_block_threads();
-  int chunks = row_chunks(N, F, 8192 * 256 / threads);
-  // keep one chunk's gh (8 B/row) within a single XCD's 4 MiB L2 so the
-  // swizzled feature sweep re-reads it from L2, not HBM
+  int chunks = row_chunks(N, n_fgroups, 8192 * 256 / threads);
   {
     static int64_t max_rpb = -1;
     if (max_rpb < 0) {
@@ -802,27 +851,28 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
     swizzle = e ? atoi(e) : 1;
   }
   const int64_t rpb = (N + chunks - 1) / chunks;
-  const int chunk_groups = (chunks + 7) / 8;
-  const int grid_flat = chunk_groups * 8 * F;
   for (int s0 = 0; s0 < n_slots; s0 += group) {
     const int ng = (n_slots - s0) < group ? (n_slots - s0) : group;
-    const size_t lds = (size_t)ng * n_bins * 16 +
+    const int this_fpb = (ng == n_slots) ? fpb : 1;
+    const int this_nfg = (F + this_fpb - 1) / this_fpb;
+    const size_t lds = (size_t)this_fpb * ng * n_bins * 16 +
                        (lds_map ? map_bytes_full : 0);
     const int filtered = (n_slots > group || filtered_hint) ? 1 : 0;
     if (swizzle) {
+      const int grid_flat = ((chunks + 7) / 8) * 8 * this_nfg;
       hipLaunchKernelGGL(hist_build_lds_kernel, dim3(grid_flat),
                          dim3(threads), lds, (hipStream_t)stream, bins,
                          (const float2*)gh, node_ids, slot_map,
                          hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
                          level_base, level_size, slot0 + s0, ng, lds_map,
-                         filtered, chunks, 1, rpb);
+                         filtered, this_fpb, this_nfg, chunks, 1, rpb);
     } else {
-      hipLaunchKernelGGL(hist_build_lds_kernel, dim3(F, chunks),
+      hipLaunchKernelGGL(hist_build_lds_kernel, dim3(this_nfg, chunks),
                          dim3(threads), lds, (hipStream_t)stream, bins,
                          (const float2*)gh, node_ids, slot_map,
                          hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
                          level_base, level_size, slot0 + s0, ng, lds_map,
-                         filtered, chunks, 0, rpb);
+                         filtered, this_fpb, this_nfg, chunks, 0, rpb);
     }
   }
 }
