@@ -822,7 +822,7 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   static int max_fpb = -1;
   if (max_fpb < 0) {
     const char* e = getenv("YDFA_HIST_FPB");
-    max_fpb = e ? atoi(e) : 8;
+    max_fpb = e ? atoi(e) : 1;
     if (max_fpb < 1) max_fpb = 1;
   }
   int fpb = 1;
