@@ -296,3 +296,39 @@ def test_custom_binary_loss(binary_data):
     m = ydf.GradientBoostedTreesLearner(label="label", loss=L,
                                         num_trees=40).train(binary_data)
     assert m.evaluate(binary_data).accuracy > 0.9
+
+
+def _ranking_data(seed=0, Q=500, M=12):
+    rng = np.random.RandomState(seed)
+    n = Q * M
+    group = np.repeat(np.arange(Q), M)
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    true = 2 * x1 + x2
+    rel = np.zeros(n, dtype=np.float32)
+    for q in range(Q):
+        order = np.argsort(np.argsort(-true[q * M:(q + 1) * M]))
+        rel[q * M:(q + 1) * M] = np.clip(4 - order // 3, 0, 4)
+    return {"x1": x1, "x2": x2, "g": group, "label": rel}
+
+
+def test_ranking_lambdamart_ndcg():
+    d = _ranking_data()
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.RANKING, ranking_group="g",
+        num_trees=50).train(d)
+    ev = m.evaluate(d)
+    assert ev.ndcg is not None and ev.ndcg > 0.95
+    assert "g" not in m.input_feature_names()
+
+
+def test_ranking_model_roundtrip(tmp_path):
+    d = _ranking_data(Q=100)
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.RANKING, ranking_group="g",
+        num_trees=10, validation_ratio=0).train(d)
+    p = str(tmp_path / "rankm")
+    m.save(p)
+    m2 = ydf.load_model(p)
+    np.testing.assert_allclose(m.predict(d), m2.predict(d), rtol=1e-6)
+    assert m2.evaluate(d).ndcg is not None

@@ -36,6 +36,8 @@ class GradientBoostedTreesLearner(GenericLearner):
 
     def __init__(self, label: str, task: Task = Task.CLASSIFICATION,
                  features: Optional[Sequence[Union[str, Column]]] = None,
+                 ranking_group: Optional[str] = None,
+                 ndcg_truncation: int = 5,
                  num_trees: int = 300, max_depth: int = 6,
                  shrinkage: float = 0.1, subsample: float = 1.0,
                  sampling_method: str = "RANDOM",
@@ -58,6 +60,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
+        self.ranking_group = ranking_group
+        self.ndcg_truncation = ndcg_truncation
         self.hyperparameters = dict(
             num_trees=num_trees, max_depth=max_depth, shrinkage=shrinkage,
             subsample=subsample, sampling_method=sampling_method,
@@ -87,6 +91,21 @@ class GradientBoostedTreesLearner(GenericLearner):
             return self._train_with_tuner(data, valid=valid)
         hp = self.hyperparameters
         device = self._resolve_device()
+        group_ids = None
+        if self._task == Task.RANKING:
+            if self.ranking_group is None:
+                raise ValueError("task=RANKING needs ranking_group=")
+            # sort rows by group so queries are contiguous, drop the group
+            # column from the features
+            from ydf_amd.dataset.dataset import _to_column_dict
+
+            cols = _to_column_dict(data)
+            gvals = np.asarray(cols[self.ranking_group])
+            order = np.argsort(gvals, kind="stable")
+            data = {k: np.asarray(v)[order] for k, v in cols.items()}
+            group_ids = data.pop(self.ranking_group)
+            if self.features is None:
+                self.features = [k for k in data if k != self.label]
         ds, bins, labels, bnd, cat_flags, weights = self._prepare(
             data, device)
         if labels is None:
@@ -102,6 +121,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                     else trainer_lib.LOSS_BINOMIAL)
         elif self._task == Task.REGRESSION:
             loss = trainer_lib.LOSS_SQUARED_ERROR
+        elif self._task == Task.RANKING:
+            loss = trainer_lib.LOSS_LAMBDA_MART_NDCG
         else:
             raise NotImplementedError(
                 f"GBT task {self._task} not yet supported")
@@ -109,8 +130,35 @@ class GradientBoostedTreesLearner(GenericLearner):
         # validation split (reference validation_set_ratio,
         # gradient_boosted_trees.cc:1243)
         valid_bins = valid_labels = None
+        ranking = valid_ranking = None
         vr = hp["validation_ratio"]
-        if valid is not None:
+        if self._task == Task.RANKING:
+            from ydf_amd.learner.ranking import RankingLambdas
+
+            # group-aware validation split: whole queries go to validation
+            if vr > 0.0 and hp["early_stopping"] != "NONE":
+                uniq, starts = np.unique(group_ids, return_index=True)
+                rng = np.random.RandomState(self.random_seed)
+                vmask_g = rng.random_sample(len(uniq)) < vr
+                gidx = np.searchsorted(np.sort(starts),
+                                       np.arange(len(group_ids)),
+                                       side="right") - 1
+                vmask = vmask_g[gidx]
+                vi = torch.from_numpy(np.nonzero(vmask)[0]).to(device)
+                ti = torch.from_numpy(np.nonzero(~vmask)[0]).to(device)
+                valid_bins = bins[:, vi].contiguous()
+                valid_labels = labels[vi].contiguous()
+                bins = bins[:, ti].contiguous()
+                labels = labels[ti].contiguous()
+                valid_ranking = RankingLambdas(
+                    group_ids[vmask], group_ids[vmask] * 0
+                    + valid_labels.cpu().numpy(), device,
+                    truncation=self.ndcg_truncation)
+                group_ids = group_ids[~vmask]
+            ranking = RankingLambdas(group_ids, labels.cpu().numpy(),
+                                     device,
+                                     truncation=self.ndcg_truncation)
+        elif valid is not None:
             valid_bins, valid_labels = self._prepare_valid(valid, ds, device)
         elif vr > 0.0 and hp["early_stopping"] != "NONE":
             N = bins.shape[1]
@@ -244,7 +292,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             t, log=info, start_iteration=start_it,
             resume_margins=resume_margins,
             resume_valid_margins=resume_valid_margins,
-            custom_loss=custom_loss,
+            custom_loss=custom_loss, ranking=ranking,
+            valid_ranking=valid_ranking,
             snapshot_cb=snapshot_cb,
             snapshot_interval_seconds=hp.get(
                 "resume_training_snapshot_interval_seconds", 1800.0),

@@ -30,6 +30,7 @@ from ydf_amd import ops
 LOSS_SQUARED_ERROR = 2
 LOSS_BINOMIAL = 1
 LOSS_MULTINOMIAL = 3
+LOSS_LAMBDA_MART_NDCG = 9
 LOSS_RF = 100  # weighted-target mode (RF/CART): not a GBT loss
 
 
@@ -447,7 +448,7 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
               snapshot_cb=None,
               snapshot_interval_seconds: float = 1800.0,
               max_duration_seconds: float = -1.0,
-              custom_loss=None):
+              custom_loss=None, ranking=None, valid_ranking=None):
     """The boosting loop (reference gradient_boosted_trees.cc:1460).
 
     Returns (trees, init_preds, training_logs). For multinomial loss,
@@ -481,6 +482,13 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 "training (host callbacks)")
         init = default_initial_predictions(custom_loss, y.cpu().numpy())
         init_preds = [init] * C
+    elif cfg.loss == LOSS_LAMBDA_MART_NDCG:
+        if trainer.distributed:
+            raise NotImplementedError(
+                "ranking is single-process for now (groups are not "
+                "row-shardable without group-aware sharding)")
+        init = 0.0
+        init_preds = [0.0]
     elif cfg.loss == LOSS_BINOMIAL:
         s = torch.stack([y.sum(), counts[0]])
         trainer._allreduce(s)
@@ -550,6 +558,9 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 trainer.gh.copy_(torch.from_numpy(
                     np.stack([gc, np.clip(hc, 1e-16, 16.0)],
                              axis=1)).to(dev))
+            elif cfg.loss == LOSS_LAMBDA_MART_NDCG:
+                lg, lh = ranking.lambdas(pc)
+                trainer.gh.copy_(torch.stack([lg, lh], dim=1))
             elif multi:
                 ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
             else:
@@ -601,7 +612,10 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             snapshot_cb(trees, it + 1, init_preds)
             t_last_snapshot = _time.monotonic()
         if has_valid:
-            if custom_loss is not None and custom_loss.loss is not None:
+            if cfg.loss == LOSS_LAMBDA_MART_NDCG:
+                vloss = -valid_ranking.ndcg(valid_preds[0]) \
+                    if valid_ranking is not None else float("nan")
+            elif custom_loss is not None and custom_loss.loss is not None:
                 vl_np = (valid_preds[0] if C == 1
                          else valid_preds).cpu().numpy()
                 vy_np = trainer.valid_labels.cpu().numpy()

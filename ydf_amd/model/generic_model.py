@@ -203,7 +203,18 @@ class GenericModel:
             else:
                 labels = np.asarray(cols[lname], dtype=np.float32)
         n_classes = len(self.label_classes) if self.label_classes else 2
-        return evaluate_predictions(preds, labels, self._task, n_classes)
+        ev = evaluate_predictions(preds, labels, self._task, n_classes)
+        if self._task == Task.RANKING:
+            gcol = (self.metadata or {}).get("ranking_group")
+            if gcol and cols is not None and gcol in cols:
+                from ydf_amd.metric.metric import ndcg as ndcg_fn
+
+                ev.ndcg = ndcg_fn(
+                    labels, preds, np.asarray(cols[gcol]),
+                    truncation=(self.metadata or {}).get(
+                        "ndcg_truncation", 5))
+                ev.loss = -ev.ndcg
+        return ev
 
     # ------------------------------------------------------------------
     def variable_importances(self) -> Dict:
