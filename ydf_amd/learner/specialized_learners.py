@@ -221,7 +221,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                 forest=flat, dataspec=ds.dataspec, task=self._task,
                 label_classes=classes, init_predictions=init_preds,
                 num_trees_per_iter=C, activation=activation,
-                metadata={"feature_gains": gains})
+                metadata={"feature_gains": gains,
+                          "ranking_group": self.ranking_group,
+                          "ndcg_truncation": self.ndcg_truncation})
 
         # checkpoint/resume (reference try_resume_training +
         # snapshot interval, abstract_learner.proto:52-56)
