@@ -332,3 +332,29 @@ def test_ranking_model_roundtrip(tmp_path):
     m2 = ydf.load_model(p)
     np.testing.assert_allclose(m.predict(d), m2.predict(d), rtol=1e-6)
     assert m2.evaluate(d).ndcg is not None
+
+
+def test_tree_shap_efficiency(regression_data):
+    """Sum of SHAP values + bias must equal the prediction (the Shapley
+    efficiency property; reference utils/shap.h TreeSHAP)."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=25,
+        validation_ratio=0).train(regression_data)
+    shap = m.predict_shap(regression_data)
+    phi = np.stack([shap[k] for k in m.input_feature_names()], axis=1)
+    total = phi.sum(axis=1) + shap["__BIAS__"]
+    np.testing.assert_allclose(total, m.predict(regression_data), atol=1e-4)
+
+
+def test_tree_shap_binary_margin(binary_data):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=20).train(
+        binary_data)
+    shap = m.predict_shap(binary_data)
+    phi = np.stack([shap[k] for k in m.input_feature_names()], axis=1)
+    margin = phi.sum(axis=1) + shap["__BIAS__"]
+    prob = 1.0 / (1.0 + np.exp(-margin))
+    np.testing.assert_allclose(prob, m.predict(binary_data), atol=1e-4)
+    # x1 dominates the fixture
+    means = {k: np.abs(v).mean() for k, v in shap.items()
+             if k != "__BIAS__"}
+    assert max(means, key=means.get) == "x1"

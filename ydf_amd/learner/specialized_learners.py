@@ -505,12 +505,13 @@ class IsolationForestLearner(GenericLearner):
             max_depth = max(1, int(math.ceil(math.log2(max(sub, 2)))))
 
         c = IsolationForestModel.expected_path_length
-        feats, thrs, lefts, roots = [], [], [], []
+        feats, thrs, lefts, roots, covers = [], [], [], [], []
 
         def new_node() -> int:
             feats.append(-1)
             thrs.append(0.0)
             lefts.append(0)
+            covers.append(0.0)
             return len(feats) - 1
 
         def build(root_rows: np.ndarray) -> None:
@@ -521,6 +522,7 @@ class IsolationForestLearner(GenericLearner):
             while stack:
                 my, rows, depth = stack.pop()
                 n = len(rows)
+                covers[my] = float(n)
                 split = None
                 if depth < max_depth and n > 1:
                     # random feature with a non-constant range, uniform cut
@@ -552,7 +554,8 @@ class IsolationForestLearner(GenericLearner):
         flat = FlatForest(feat=np.asarray(feats, np.int32),
                           thr=np.asarray(thrs, np.float32),
                           left=np.asarray(lefts, np.int32),
-                          roots=np.asarray(roots, np.int32))
+                          roots=np.asarray(roots, np.int32),
+                          cover=np.asarray(covers, np.float32))
         return IsolationForestModel(
             forest=flat, dataspec=ds.dataspec, task=self._task,
             label_classes=None, init_predictions=[0.0],

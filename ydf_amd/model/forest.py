@@ -28,12 +28,16 @@ class FlatForest:
     # into masks (256-bit "category goes right" bitmask as 4 x u64)
     cat_idx: np.ndarray = None   # i32 [total]
     masks: np.ndarray = None     # u64 [n_masks, 4]
+    # training cover (weighted example count) per node; used by TreeSHAP
+    cover: np.ndarray = None     # f32 [total]
 
     def __post_init__(self):
         if self.cat_idx is None:
             self.cat_idx = np.full(len(self.feat), -1, dtype=np.int32)
         if self.masks is None:
             self.masks = np.zeros((0, 4), dtype=np.uint64)
+        if self.cover is None:
+            self.cover = np.zeros(len(self.feat), dtype=np.float32)
 
     @property
     def has_cats(self) -> bool:
@@ -109,17 +113,19 @@ def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
             masks = tree.masks[nodes[ci]].astype(np.uint64)
     li = np.nonzero(~n_int)[0]
     thr[li] = tree.leaf_value[nodes[li]] * leaf_scale
-    return feat, thr, left, cat_idx, masks
+    cover = tree.counts[nodes].astype(np.float32)
+    return feat, thr, left, cat_idx, masks, cover
 
 
 def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
                       leaf_scale: float = 1.0, cat_feats=None) -> FlatForest:
-    feats, thrs, lefts, roots, cidxs, mask_list = [], [], [], [], [], []
+    feats, thrs, lefts, roots, cidxs, mask_list, covers = \
+        [], [], [], [], [], [], []
     off = 0
     mask_off = 0
     for t in trees:
-        f, th, lf, ci, mk = host_tree_to_flat(t, boundaries, leaf_scale,
-                                              cat_feats)
+        f, th, lf, ci, mk, cv = host_tree_to_flat(t, boundaries, leaf_scale,
+                                                  cat_feats)
         lf = np.where(f >= 0, lf + off, 0)
         ci = np.where(ci >= 0, ci + mask_off, -1)
         roots.append(off)
@@ -130,6 +136,7 @@ def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
         lefts.append(lf)
         cidxs.append(ci)
         mask_list.append(mk)
+        covers.append(cv)
     return FlatForest(
         feat=np.concatenate(feats) if feats else np.zeros(0, np.int32),
         thr=np.concatenate(thrs) if thrs else np.zeros(0, np.float32),
@@ -138,6 +145,8 @@ def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
         cat_idx=np.concatenate(cidxs) if cidxs else np.zeros(0, np.int32),
         masks=np.concatenate(mask_list) if mask_list
         else np.zeros((0, 4), np.uint64),
+        cover=np.concatenate(covers) if covers
+        else np.zeros(0, np.float32),
     )
 
 
@@ -173,4 +182,5 @@ def concat_forests(a: FlatForest, b: FlatForest) -> FlatForest:
         masks=np.concatenate([a.masks, b.masks]) if (len(a.masks)
                                                      or len(b.masks))
         else np.zeros((0, 4), np.uint64),
+        cover=np.concatenate([a.cover, b.cover]),
     )

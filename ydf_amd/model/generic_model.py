@@ -217,6 +217,47 @@ class GenericModel:
         return ev
 
     # ------------------------------------------------------------------
+    def predict_shap(self, data) -> Dict[str, np.ndarray]:
+        """Path-dependent TreeSHAP values (reference utils/shap.h:83;
+        mirrors ydf model.predict_shap). Returns {feature: phi [N]} plus
+        the expected value under "__BIAS__"; margins (pre-activation)
+        for classification. CPU implementation."""
+        from ydf_amd import ops as _ops
+        from ydf_amd._ydf_ops import (cpu_forest_expected_value,
+                                      cpu_tree_shap)
+
+        if self._n_outputs() > 1:
+            raise NotImplementedError(
+                "predict_shap supports single-output models for now")
+        if not (self.forest.cover > 0).any():
+            raise ValueError("model has no node covers (old format?)")
+        X = np.ascontiguousarray(self._encode_features(data))
+        F, N = X.shape
+        phi = np.zeros((N, F + 1), dtype=np.float32)
+        f = self.forest
+        feat = np.ascontiguousarray(f.feat)
+        thr = np.ascontiguousarray(f.thr)
+        left = np.ascontiguousarray(f.left)
+        cover = np.ascontiguousarray(f.cover)
+        roots = np.ascontiguousarray(f.roots)
+        cat_idx = np.ascontiguousarray(f.cat_idx)
+        masks = np.ascontiguousarray(f.masks)
+        scale = self._leaf_scale()
+        cpu_tree_shap(X.ctypes.data, N, F, feat.ctypes.data, thr.ctypes.data,
+                      left.ctypes.data,
+                      cat_idx.ctypes.data if f.has_cats else 0,
+                      masks.ctypes.data if f.has_cats else 0,
+                      cover.ctypes.data, roots.ctypes.data, 0, 1,
+                      f.n_trees, scale, 0.0, phi.ctypes.data)
+        ev = cpu_forest_expected_value(
+            feat.ctypes.data, thr.ctypes.data, left.ctypes.data,
+            cover.ctypes.data, roots.ctypes.data, 0, 1, f.n_trees, scale)
+        phi[:, F] = float(self.init_predictions[0]) + ev
+        out = {name: phi[:, i]
+               for i, name in enumerate(self.input_feature_names())}
+        out["__BIAS__"] = phi[:, F]
+        return out
+
     def variable_importances(self) -> Dict:
         """Structure-based variable importances (reference
         AbstractModel::GetVariableImportance; SUM_SCORE requires a model
@@ -344,7 +385,7 @@ class GenericModel:
         np.savez(os.path.join(path, "forest.npz"), feat=self.forest.feat,
                  thr=self.forest.thr, left=self.forest.left,
                  roots=self.forest.roots, cat_idx=self.forest.cat_idx,
-                 masks=self.forest.masks)
+                 masks=self.forest.masks, cover=self.forest.cover)
         with open(os.path.join(path, "done"), "w") as f:
             f.write("")
 

@@ -23,7 +23,8 @@ def load_model(path: str) -> GenericModel:
     forest = FlatForest(feat=z["feat"], thr=z["thr"], left=z["left"],
                         roots=z["roots"],
                         cat_idx=z["cat_idx"] if "cat_idx" in z else None,
-                        masks=z["masks"] if "masks" in z else None)
+                        masks=z["masks"] if "masks" in z else None,
+                        cover=z["cover"] if "cover" in z else None)
     cls = MODEL_CLASSES.get(header["model_type"], GenericModel)
     model = cls(
         forest=forest,
@@ -51,7 +52,8 @@ def serialize_model(model: GenericModel) -> bytes:
         fbuf = io.BytesIO()
         np.savez(fbuf, feat=model.forest.feat, thr=model.forest.thr,
                  left=model.forest.left, roots=model.forest.roots,
-                 cat_idx=model.forest.cat_idx, masks=model.forest.masks)
+                 cat_idx=model.forest.cat_idx, masks=model.forest.masks,
+                 cover=model.forest.cover)
         zf.writestr("forest.npz", fbuf.getvalue())
     return buf.getvalue()
 

@@ -71,6 +71,13 @@ void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const float*, const int32_t*, const int32_t*,
                         const int32_t*, const unsigned long long*, int, int,
                         int, float*, float, float);
+void cpu_tree_shap(const float*, int64_t, int, const int32_t*, const float*,
+                   const int32_t*, const int32_t*, const unsigned long long*,
+                   const float*, const int32_t*, int, int, int, float, float,
+                   float*);
+void cpu_forest_expected_value(const int32_t*, const float*, const int32_t*,
+                               const float*, const int32_t*, int, int, int,
+                               float, double*);
 }
 
 namespace {
@@ -328,6 +335,29 @@ PYBIND11_MODULE(_ydf_ops, m) {
                              N);
         },
         nogil);
+  m.def("cpu_tree_shap",
+        [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
+           uintptr_t left, uintptr_t cat_idx, uintptr_t masks,
+           uintptr_t cover, uintptr_t roots, int tree_start, int tree_step,
+           int n_trees, float scale, float init, uintptr_t phi_out) {
+          cpu_tree_shap(P<float>(X), N, F, P<int32_t>(feat), P<float>(thr),
+                        P<int32_t>(left), P<int32_t>(cat_idx),
+                        P<unsigned long long>(masks), P<float>(cover),
+                        P<int32_t>(roots), tree_start, tree_step, n_trees,
+                        scale, init, P<float>(phi_out));
+        },
+        nogil);
+  m.def("cpu_forest_expected_value",
+        [](uintptr_t feat, uintptr_t thr, uintptr_t left, uintptr_t cover,
+           uintptr_t roots, int tree_start, int tree_step, int n_trees,
+           float scale) {
+          double out = 0.0;
+          cpu_forest_expected_value(P<int32_t>(feat), P<float>(thr),
+                                    P<int32_t>(left), P<float>(cover),
+                                    P<int32_t>(roots), tree_start, tree_step,
+                                    n_trees, scale, &out);
+          return out;
+        });
   m.def("cpu_predict_forest",
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
            uintptr_t left, uintptr_t roots, uintptr_t cat_idx,

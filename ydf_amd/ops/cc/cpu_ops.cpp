@@ -458,5 +458,193 @@ void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
   });
 }
 
+// ---------------------------------------------------------------------------
+// TreeSHAP (Lundberg et al. 2018, consistent with reference utils/shap.h:83):
+// path-dependent Shapley values for one flat-forest tree ensemble.
+// phi layout: [N][F+1], last column = bias (expected value).
+// ---------------------------------------------------------------------------
+namespace {
+
+struct PathElem {
+  int feature;
+  float zero_frac;
+  float one_frac;
+  float pweight;
+};
+
+void ShapExtend(PathElem* path, int depth, float pz, float po, int pi) {
+  path[depth].feature = pi;
+  path[depth].zero_frac = pz;
+  path[depth].one_frac = po;
+  path[depth].pweight = depth == 0 ? 1.0f : 0.0f;
+  for (int i = depth - 1; i >= 0; --i) {
+    path[i + 1].pweight += po * path[i].pweight * (i + 1) / (float)(depth + 1);
+    path[i].pweight = pz * path[i].pweight * (depth - i) / (float)(depth + 1);
+  }
+}
+
+void ShapUnwind(PathElem* path, int depth, int idx) {
+  const float po = path[idx].one_frac;
+  const float pz = path[idx].zero_frac;
+  float next = path[depth].pweight;
+  for (int i = depth - 1; i >= 0; --i) {
+    if (po != 0.f) {
+      const float tmp = path[i].pweight;
+      path[i].pweight = next * (depth + 1) / ((i + 1) * po);
+      next = tmp - path[i].pweight * pz * (depth - i) / (float)(depth + 1);
+    } else {
+      path[i].pweight =
+          path[i].pweight * (depth + 1) / ((float)(depth - i) * pz);
+    }
+  }
+  for (int i = idx; i < depth; ++i) {
+    path[i].feature = path[i + 1].feature;
+    path[i].zero_frac = path[i + 1].zero_frac;
+    path[i].one_frac = path[i + 1].one_frac;
+  }
+}
+
+float ShapUnwoundSum(const PathElem* path, int depth, int idx) {
+  const float po = path[idx].one_frac;
+  const float pz = path[idx].zero_frac;
+  float total = 0.f;
+  float next = path[depth].pweight;
+  for (int i = depth - 1; i >= 0; --i) {
+    if (po != 0.f) {
+      const float tmp = next * (depth + 1) / ((i + 1) * po);
+      total += tmp;
+      next = path[i].pweight - tmp * pz * (depth - i) / (float)(depth + 1);
+    } else {
+      total += path[i].pweight / (pz * (depth - i) / (float)(depth + 1));
+    }
+  }
+  return total;
+}
+
+struct ShapCtx {
+  const float* X;
+  int64_t N;
+  const int32_t* feat;
+  const float* thr;
+  const int32_t* left;
+  const int32_t* cat_idx;
+  const unsigned long long* masks;
+  const float* cover;
+  double* phi;  // [F+1]
+  int64_t row;
+  float scale;
+};
+
+int ShapGoesRight(const ShapCtx& c, int node) {
+  const float xv = c.X[(int64_t)c.feat[node] * c.N + c.row];
+  const int ci = c.cat_idx ? c.cat_idx[node] : -1;
+  if (ci >= 0) {
+    int cb = (int)xv;
+    cb = cb < 0 ? 0 : (cb > 255 ? 255 : cb);
+    return (int)((c.masks[(int64_t)ci * 4 + (cb >> 6)] >> (cb & 63)) & 1ull);
+  }
+  return xv > c.thr[node] ? 1 : 0;
+}
+
+void ShapRecurse(const ShapCtx& c, int node, PathElem* parent_path,
+                 int depth, float pz, float po, int pi) {
+  PathElem path[64];
+  for (int i = 0; i < depth; ++i) path[i] = parent_path[i];
+  ShapExtend(path, depth, pz, po, pi);
+  if (c.feat[node] < 0) {  // leaf
+    for (int i = 1; i <= depth; ++i) {
+      const float w = ShapUnwoundSum(path, depth, i);
+      c.phi[path[i].feature] +=
+          w * (path[i].one_frac - path[i].zero_frac) * c.thr[node] * c.scale;
+    }
+    return;
+  }
+  const int hot = c.left[node] + ShapGoesRight(c, node);
+  const int cold = c.left[node] + (1 - ShapGoesRight(c, node));
+  const float cover_n = c.cover[node] > 0.f ? c.cover[node] : 1.f;
+  float hot_z = c.cover[hot] / cover_n;
+  float cold_z = c.cover[cold] / cover_n;
+  float incoming_z = 1.f, incoming_o = 1.f;
+  int path_idx = -1;
+  for (int i = 1; i <= depth; ++i) {
+    if (path[i].feature == c.feat[node]) { path_idx = i; break; }
+  }
+  int new_depth = depth;
+  if (path_idx >= 0) {
+    incoming_z = path[path_idx].zero_frac;
+    incoming_o = path[path_idx].one_frac;
+    ShapUnwind(path, depth, path_idx);
+    new_depth = depth - 1;
+  }
+  ShapRecurse(c, hot, path, new_depth + 1, hot_z * incoming_z, incoming_o,
+              c.feat[node]);
+  ShapRecurse(c, cold, path, new_depth + 1, cold_z * incoming_z, 0.f,
+              c.feat[node]);
+}
+
+}  // namespace
+
+extern "C" void cpu_tree_shap(const float* X, int64_t N, int F,
+                              const int32_t* feat, const float* thr,
+                              const int32_t* left, const int32_t* cat_idx,
+                              const unsigned long long* masks,
+                              const float* cover, const int32_t* roots,
+                              int tree_start, int tree_step, int n_trees,
+                              float scale, float init, float* phi_out) {
+  ThreadPool::Get().ParallelFor(
+      std::max(1, std::min<int>(ThreadPool::Get().size() * 4,
+                                (int)((N + 63) / 64))),
+      [&](int blk) {
+        const int nb = std::max(1, std::min<int>(
+            ThreadPool::Get().size() * 4, (int)((N + 63) / 64)));
+        const int64_t per = (N + nb - 1) / nb;
+        const int64_t i0 = blk * per, i1 = std::min<int64_t>(i0 + per, N);
+        std::vector<double> phi(F + 1);
+        for (int64_t row = i0; row < i1; ++row) {
+          std::fill(phi.begin(), phi.end(), 0.0);
+          for (int tt = 0; tt < n_trees; ++tt) {
+            const int root = roots[tree_start + (int64_t)tt * tree_step];
+            ShapCtx c{X,    N,     feat, thr, left, cat_idx, masks,
+                      cover, phi.data(), row, scale};
+            PathElem dummy[1];
+            ShapRecurse(c, root, dummy, 0, 1.f, 1.f, -1);
+          }
+          // bias column filled by the Python wrapper (init + E[forest])
+          phi[F] = 0.0;
+          (void)init;
+          float* out = phi_out + (int64_t)row * (F + 1);
+          for (int k = 0; k <= F; ++k) out[k] = (float)phi[k];
+        }
+      });
+}
+
+extern "C" void cpu_forest_expected_value(
+    const int32_t* feat, const float* thr, const int32_t* left,
+    const float* cover, const int32_t* roots, int tree_start, int tree_step,
+    int n_trees, float scale, double* out) {
+  double total = 0.0;
+  for (int tt = 0; tt < n_trees; ++tt) {
+    // iterative cover-weighted mean of leaf values
+    const int root = roots[tree_start + (int64_t)tt * tree_step];
+    struct Item { int node; double w; };
+    std::vector<Item> stack{{root, 1.0}};
+    double ev = 0.0;
+    while (!stack.empty()) {
+      Item it = stack.back();
+      stack.pop_back();
+      if (feat[it.node] < 0) {
+        ev += it.w * thr[it.node];
+        continue;
+      }
+      const int l = left[it.node];
+      const double cn = cover[it.node] > 0 ? cover[it.node] : 1.0;
+      stack.push_back({l, it.w * cover[l] / cn});
+      stack.push_back({l + 1, it.w * cover[l + 1] / cn});
+    }
+    total += ev * scale;
+  }
+  *out = total;
+}
+
 }  // extern "C"
 }  // namespace ydfa
