@@ -115,6 +115,9 @@ class ForestTrainer:
         dev = self.device
         self.gh = torch.empty((self.N, 2), dtype=torch.float32, device=dev)
         self.node_ids = torch.empty(self.N, dtype=torch.int32, device=dev)
+        # per-row slot-group id scratch for multi-group (deep) levels
+        self.grp_buf = torch.empty(self.N, dtype=torch.uint8, device=dev) \
+            if self.device.type == "cuda" else None
         self.hist = torch.empty((self.max_slots, self.F, n_bins, 3),
                                 dtype=torch.float32, device=dev)
         # Histogram-subtraction trick (sibling = parent - smaller child):
@@ -310,7 +313,8 @@ class ForestTrainer:
                 hist_view = self.hist[:ns]
                 hist_view.zero_()
                 ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
-                               hist_view, level_base, level_size, s0, ns)
+                               hist_view, level_base, level_size, s0, ns,
+                               grp_scratch=self.grp_buf)
                 self._allreduce(hist_view)
                 if derived and s0 == 0:
                     d_idx = torch.tensor([d[0] for d in derived],

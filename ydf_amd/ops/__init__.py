@@ -89,7 +89,7 @@ def weighted_target(labels: torch.Tensor, weights, gh: torch.Tensor):
 def hist_build(bins: torch.Tensor, gh: torch.Tensor, node_ids: torch.Tensor,
                slot_map: torch.Tensor, hist: torch.Tensor, level_base: int,
                level_size: int, slot0: int, n_slots: int,
-               filtered_hint: bool = False):
+               filtered_hint: bool = False, grp_scratch=None):
     """bins [F,N] u8; hist [n_slots,F,n_bins,3] f32 (pre-zeroed, base=slot0).
 
     slot_map [level_size] i32 maps level-relative node -> slot (-1 closed).
@@ -101,7 +101,9 @@ def hist_build(bins: torch.Tensor, gh: torch.Tensor, node_ids: torch.Tensor,
         _C.gpu_hist_build(bins.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
                           slot_map.data_ptr(), hist.data_ptr(), N, F, n_bins,
                           level_base, level_size, slot0, n_slots,
-                          1 if filtered_hint else 0, _stream())
+                          1 if filtered_hint else 0,
+                          grp_scratch.data_ptr() if grp_scratch is not None
+                          else 0, _stream())
     else:
         _C.cpu_hist_build(bins.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
                           slot_map.data_ptr(), hist.data_ptr(), N, F, n_bins,

@@ -165,6 +165,8 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
                                       float* __restrict__ hist, int64_t N,
                                       int F, int n_bins, int level_base,
                                       int level_size, int slot0, int n_slots,
+                                      const uint8_t* __restrict__ row_grp,
+                                      int n_slots_group,
                                       int lds_map, int filtered, int fpb,
                                       int n_fgroups, int n_chunks,
                                       int swizzle, int64_t rows_per_block) {
@@ -211,7 +213,51 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
   int64_t i = row0 + threadIdx.x;
   const int64_t bulk_end = row1 - (kHistUnroll - 1) * stride;
 
-  if (filtered) {
+  if (filtered && row_grp != nullptr) {
+    // Cheapest multi-group pass: 1-byte group id per row; only matching
+    // rows touch node_ids/gh/bins.
+    const uint8_t my_grp = (uint8_t)(slot0 / n_slots_group);
+    for (; i < bulk_end; i += kHistUnroll * stride) {
+      uint8_t gb[kHistUnroll];
+#pragma unroll
+      for (int u = 0; u < kHistUnroll; ++u) gb[u] = row_grp[i + u * stride];
+#pragma unroll
+      for (int u = 0; u < kHistUnroll; ++u) {
+        if (gb[u] != my_grp) continue;
+        const int rel = node_ids[i + u * stride] - level_base;
+        const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+        if (slot < 0 || slot >= n_slots) continue;
+        const float2 v = gh[i + u * stride];
+        const unsigned long long hq =
+            (unsigned long long)(v.y * kHScale + 0.5f);
+        const unsigned long long pk =
+            hq | ((unsigned long long)(v.y != 0.f) << 44);
+        for (int j = 0; j < nf; ++j) {
+          const int b = bins[(int64_t)(f0 + j) * N + i + u * stride];
+          const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+          atomicAdd(lg + cell, (double)v.x);
+          atomicAdd(lp + cell, pk);
+        }
+      }
+    }
+    for (; i < row1; i += stride) {
+      if (row_grp[i] != my_grp) continue;
+      const int rel = node_ids[i] - level_base;
+      const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+      if (slot < 0 || slot >= n_slots) continue;
+      const float2 v = gh[i];
+      const unsigned long long hq =
+          (unsigned long long)(v.y * kHScale + 0.5f);
+      const unsigned long long pk =
+          hq | ((unsigned long long)(v.y != 0.f) << 44);
+      for (int j = 0; j < nf; ++j) {
+        const int b = bins[(int64_t)(f0 + j) * N + i];
+        const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+        atomicAdd(lg + cell, (double)v.x);
+        atomicAdd(lp + cell, pk);
+      }
+    }
+  } else if (filtered) {
     // Multi-slot-group launch: most rows belong to ANOTHER group, so load
     // only node_ids eagerly and fetch gh/bins under the match predicate —
     // a group pass then streams ~4B/row instead of 13B/row.
@@ -319,6 +365,27 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
     // h sums exact, which the CPU/GPU equality tests rely on)
     atomicAdd(p + 1, (float)((double)(pk & kHMask) * (double)kHInvScale));
     atomicAdd(p + 2, (float)(pk >> 44));
+  }
+}
+
+// Precomputes the per-row slot-GROUP id (u8) for multi-group levels so each
+// group pass streams 1 B/row instead of node_id + slot-map lookups.
+// 255 = row not in any open slot of this level.
+__global__ void row_group_kernel(const int32_t* __restrict__ node_ids,
+                                 const int32_t* __restrict__ slot_map,
+                                 uint8_t* __restrict__ grp, int64_t N,
+                                 int level_base, int level_size,
+                                 int group_size) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < N; k += stride) {
+    const int rel = node_ids[k] - level_base;
+    int g = 255;
+    if (rel >= 0 && rel < level_size) {
+      const int slot = slot_map[rel];
+      if (slot >= 0) g = slot / group_size;
+    }
+    grp[k] = (uint8_t)g;
   }
 }
 
@@ -807,7 +874,7 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
                     const int32_t* node_ids, const int32_t* slot_map,
                     float* hist, int64_t N, int F, int n_bins, int level_base,
                     int level_size, int slot0, int n_slots,
-                    int filtered_hint, void* stream) {
+                    int filtered_hint, uint8_t* grp_scratch, void* stream) {
   // Stage the level's slot map in LDS when it fits comfortably (removes a
   // dependent global load per row visit); cap at 32 KiB so the histogram
   // region keeps >= ~32 slots at 16 B/bin.
@@ -851,6 +918,16 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
     swizzle = e ? atoi(e) : 1;
   }
   const int64_t rpb = (N + chunks - 1) / chunks;
+  // the u8 prefilter is only valid when every launch covers whole groups
+  // (slot0 must be group-aligned; the trainer's outer chunking may not be)
+  const int use_grp = (grp_scratch != nullptr && n_slots > group &&
+                       group <= 255 && (slot0 % group) == 0) ? 1 : 0;
+  if (use_grp) {
+    hipLaunchKernelGGL(row_group_kernel, dim3(elem_grid(N, 4096)),
+                       dim3(kBlock), 0, (hipStream_t)stream, node_ids,
+                       slot_map, grp_scratch, N, level_base, level_size,
+                       group);
+  }
   for (int s0 = 0; s0 < n_slots; s0 += group) {
     const int ng = (n_slots - s0) < group ? (n_slots - s0) : group;
     const int this_fpb = (ng == n_slots) ? fpb : 1;
@@ -858,21 +935,24 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
     const size_t lds = (size_t)this_fpb * ng * n_bins * 16 +
                        (lds_map ? map_bytes_full : 0);
     const int filtered = (n_slots > group || filtered_hint) ? 1 : 0;
+    const uint8_t* rg = use_grp ? grp_scratch : nullptr;
     if (swizzle) {
       const int grid_flat = ((chunks + 7) / 8) * 8 * this_nfg;
       hipLaunchKernelGGL(hist_build_lds_kernel, dim3(grid_flat),
                          dim3(threads), lds, (hipStream_t)stream, bins,
                          (const float2*)gh, node_ids, slot_map,
                          hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
-                         level_base, level_size, slot0 + s0, ng, lds_map,
-                         filtered, this_fpb, this_nfg, chunks, 1, rpb);
+                         level_base, level_size, slot0 + s0, ng, rg, group,
+                         lds_map, filtered, this_fpb, this_nfg, chunks, 1,
+                         rpb);
     } else {
       hipLaunchKernelGGL(hist_build_lds_kernel, dim3(this_nfg, chunks),
                          dim3(threads), lds, (hipStream_t)stream, bins,
                          (const float2*)gh, node_ids, slot_map,
                          hist + (int64_t)s0 * F * n_bins * 3, N, F, n_bins,
-                         level_base, level_size, slot0 + s0, ng, lds_map,
-                         filtered, this_fpb, this_nfg, chunks, 0, rpb);
+                         level_base, level_size, slot0 + s0, ng, rg, group,
+                         lds_map, filtered, this_fpb, this_nfg, chunks, 0,
+                         rpb);
     }
   }
 }
