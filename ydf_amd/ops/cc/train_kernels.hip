@@ -919,9 +919,17 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   }
   const int64_t rpb = (N + chunks - 1) / chunks;
   // the u8 prefilter is only valid when every launch covers whole groups
-  // (slot0 must be group-aligned; the trainer's outer chunking may not be)
-  const int use_grp = (grp_scratch != nullptr && n_slots > group &&
-                       group <= 255 && (slot0 % group) == 0) ? 1 : 0;
+  // (slot0 must be group-aligned). Measured SLOWER than the node-id path
+  // on the 1Mx500 depth-16 RF config (deep levels are issue-bound, not
+  // bytes-bound), so it is opt-in via YDFA_HIST_GRP=1.
+  static int grp_enabled = -1;
+  if (grp_enabled < 0) {
+    const char* e = getenv("YDFA_HIST_GRP");
+    grp_enabled = e ? atoi(e) : 0;
+  }
+  const int use_grp = (grp_enabled && grp_scratch != nullptr &&
+                       n_slots > group && group <= 255 &&
+                       (slot0 % group) == 0) ? 1 : 0;
   if (use_grp) {
     hipLaunchKernelGGL(row_group_kernel, dim3(elem_grid(N, 4096)),
                        dim3(kBlock), 0, (hipStream_t)stream, node_ids,
