@@ -167,6 +167,17 @@ def main():
         est_full = FULL_TREES * ms_per_step / 1000.0
         print(f"# est. full {FULL_TREES}-tree train wall-clock: "
               f"{est_full:.2f}s on {world} GPU(s)", file=sys.stderr)
+    # sanity: the timed steps did real boosting — training accuracy/loss
+    # after warmup+steps trees must beat the base rate
+    lb = torch.zeros(2, dtype=torch.float32, device=device)
+    ops.binary_logloss(preds, labels, lb)
+    s2 = torch.stack([lb[0], lb[1],
+                      torch.tensor(float(N), device=device)])
+    tr._allreduce(s2)
+    if rank == 0:
+        print(f"# train logloss={float(s2[0] / s2[2]):.4f} "
+              f"accuracy={float(s2[1] / s2[2]):.4f} after "
+              f"{args.warmup + args.steps} trees", file=sys.stderr)
 
 
 if __name__ == "__main__":
