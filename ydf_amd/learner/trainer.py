@@ -206,6 +206,13 @@ class ForestTrainer:
         On exit self.node_ids holds the final (sampled-rows) assignment and
         self.leaf_vals the per-node values; callers apply update_preds.
         """
+        self.grow_tree_device(tree_idx, sample_mask)
+        return self.extract_host_tree()
+
+    def grow_tree_device(self, tree_idx: int,
+                         sample_mask: Optional[torch.Tensor] = None) -> None:
+        """Device-side tree growth (no host copies; hipGraph-capturable
+        when every level runs in dense mode)."""
         cfg = self.cfg
         self.tree_feat.fill_(-1)
         self.tree_bin.zero_()
@@ -372,6 +379,9 @@ class ForestTrainer:
                             for a, c in zip(children, ccounts)}
 
         ops.leaf_values(self.node_stats, self.leaf_vals, cfg.lambda_l2)
+
+    def extract_host_tree(self) -> HostTree:
+        cfg = self.cfg
         # .copy(): on CPU .cpu().numpy() aliases the (reused) buffers
         return HostTree(
             feat=self.tree_feat.cpu().numpy().copy(),
@@ -432,6 +442,22 @@ class ForestTrainer:
                             level_size, cat_flags=self.cat_flags,
                             masks=self.tree_masks)
         return fits
+
+    def capture_step_graph(self, preds: torch.Tensor, labels: torch.Tensor,
+                           shrinkage: float):
+        """Captures one boosting step (gradients -> dense tree growth ->
+        prediction update) as a hipGraph; replay + extract_host_tree()
+        per tree. Requires: CUDA device, single process, all levels dense,
+        no feature sampling / subsampling."""
+        assert self.device.type == "cuda" and not self.distributed
+        assert (1 << (self.cfg.max_depth - 1)) <= self.dense_limit
+        assert self.cfg.num_candidate_features <= 0
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            ops.grad_hess(preds, labels, self.gh, self.cfg.loss)
+            self.grow_tree_device(0, None)
+            ops.update_preds(preds, self.node_ids, self.leaf_vals, shrinkage)
+        return g
 
     def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor):
         """Routes arbitrary rows through the latest tree (device arrays)."""
