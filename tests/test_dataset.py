@@ -95,3 +95,21 @@ def test_pandas_input():
                                      task=ydf.Task.CLASSIFICATION)
     assert ds.n_examples == 2
     assert ds.n_features == 2
+
+
+def test_pav_calibration():
+    from ydf_amd.utils.calibration import fit_pav
+
+    rng = np.random.RandomState(0)
+    s = rng.randn(3000)
+    p_true = 1 / (1 + np.exp(-2 * s))
+    y = (rng.rand(3000) < p_true).astype(float)
+    cal = fit_pav(s, y)
+    # monotone non-decreasing
+    assert (np.diff(cal.values) >= -1e-12).all()
+    phat = cal.apply(s)
+    # calibrated probabilities track the true ones
+    assert np.abs(phat - p_true).mean() < 0.08
+    # round-trip
+    cal2 = type(cal).from_json(cal.to_json())
+    np.testing.assert_allclose(cal2.apply(s), phat)
