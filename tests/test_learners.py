@@ -358,3 +358,23 @@ def test_tree_shap_binary_margin(binary_data):
     means = {k: np.abs(v).mean() for k, v in shap.items()
              if k != "__BIAS__"}
     assert max(means, key=means.get) == "x1"
+
+
+def test_monotonic_constraint():
+    """Predictions must be non-decreasing in a monotonic(+1) feature
+    (reference monotonic constraints, decision_tree.proto)."""
+    rng = np.random.RandomState(0)
+    n = 6000
+    x = rng.randn(n).astype(np.float32)
+    z = rng.randn(n).astype(np.float32)
+    y = (x + 0.3 * np.sin(5 * x) + z).astype(np.float32)
+    d = {"x": x, "z": z, "label": y}
+    feats = [ydf.Feature("x", monotonic=ydf.Monotonic.INCREASING),
+             ydf.Feature("z")]
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, features=feats,
+        num_trees=60, validation_ratio=0).train(d)
+    grid = np.linspace(-3, 3, 300).astype(np.float32)
+    p = m.predict({"x": grid, "z": np.zeros_like(grid)})
+    assert np.diff(p).min() >= -1e-6
+    assert m.evaluate(d).rmse < 0.5

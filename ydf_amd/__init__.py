@@ -64,6 +64,13 @@ from ydf_amd.dataset.dataspec import (
 
 Feature = Column
 
+
+class Monotonic:
+    """Monotonic direction constants (mirrors ydf.Monotonic)."""
+
+    INCREASING = 1
+    DECREASING = -1
+
 # Metric
 from ydf_amd.metric.metric import Evaluation, evaluate_predictions
 

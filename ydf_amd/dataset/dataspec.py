@@ -43,6 +43,9 @@ class Column:
 
     name: str
     semantic: Optional[Semantic] = None
+    # monotonic constraint direction: +1 increasing, -1 decreasing, 0 none
+    # (mirrors ydf.Feature(monotonic=...))
+    monotonic: int = 0
 
 
 # Out-of-vocabulary item: index 0 of every categorical vocabulary

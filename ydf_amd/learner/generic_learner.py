@@ -103,6 +103,16 @@ class GenericLearner:
         if cat_feats.any():
             cat_flags = torch.from_numpy(
                 cat_feats.astype(np.uint8)).to(device)
+        mono = None
+        if self.features is not None:
+            dirs = {f.name: int(getattr(f, "monotonic", 0) or 0)
+                    for f in self.features if isinstance(f, Column)}
+            if any(dirs.values()):
+                arr = np.zeros(len(ds.dataspec.feature_columns),
+                               dtype=np.int8)
+                for i, c in enumerate(ds.dataspec.feature_columns):
+                    arr[i] = dirs.get(c.name, 0)
+                mono = torch.from_numpy(arr).to(device)
         weights = None
         if weights_np is not None:
             # packed-u64 histogram path needs per-example h <= 16: scale
@@ -113,7 +123,7 @@ class GenericLearner:
                 weights_np = weights_np * (8.0 / mx)
             weights = torch.from_numpy(
                 np.ascontiguousarray(weights_np)).to(device)
-        return ds, bins, labels, bnd, cat_flags, weights
+        return ds, bins, labels, bnd, cat_flags, weights, mono
 
     def _label_classes(self, ds: VerticalDataset):
         if self._task != Task.CLASSIFICATION:

@@ -106,7 +106,7 @@ class GradientBoostedTreesLearner(GenericLearner):
             group_ids = data.pop(self.ranking_group)
             if self.features is None:
                 self.features = [k for k in data if k != self.label]
-        ds, bins, labels, bnd, cat_flags, weights = self._prepare(
+        ds, bins, labels, bnd, cat_flags, weights, mono = self._prepare(
             data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
@@ -202,7 +202,8 @@ class GradientBoostedTreesLearner(GenericLearner):
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
                                       valid_bins=valid_bins,
                                       valid_labels=valid_labels,
-                                      cat_flags=cat_flags, weights=weights)
+                                      cat_flags=cat_flags, weights=weights,
+                                      mono=mono)
         C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
         activation = "identity"
         if custom_loss is not None:
@@ -388,7 +389,7 @@ class RandomForestLearner(GenericLearner):
             return self._train_with_tuner(data, valid=valid)
         hp = self.hyperparameters
         device = self._resolve_device()
-        ds, bins, labels, bnd, cat_flags, weights = self._prepare(
+        ds, bins, labels, bnd, cat_flags, weights, mono = self._prepare(
             data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
@@ -405,7 +406,8 @@ class RandomForestLearner(GenericLearner):
             num_candidate_features=self._num_candidate(F),
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
-                                      cat_flags=cat_flags, weights=weights)
+                                      cat_flags=cat_flags, weights=weights,
+                                      mono=mono)
         compute_oob = (hp["compute_oob_performances"]
                        and hp["bootstrap_training_dataset"])
         result = trainer_lib.train_rf(t, log=info, compute_oob=compute_oob)

@@ -89,12 +89,14 @@ class ForestTrainer:
                  valid_bins: Optional[torch.Tensor] = None,
                  valid_labels: Optional[torch.Tensor] = None,
                  cat_flags: Optional[torch.Tensor] = None,
-                 weights: Optional[torch.Tensor] = None):
+                 weights: Optional[torch.Tensor] = None,
+                 mono: Optional[torch.Tensor] = None):
         assert bins.dtype == torch.uint8 and bins.dim() == 2
         self.bins = bins
         self.labels = labels
         self.cfg = cfg
         self.weights = weights      # f32 [N] example weights (or None)
+        self.mono = mono            # i8 [F] monotonic dirs (or None)
         self.cat_flags = cat_flags  # u8 [F] on device; None = all numerical
         self.has_cats = cat_flags is not None and bool(cat_flags.any())
         if not self.has_cats:
@@ -143,6 +145,13 @@ class ForestTrainer:
                                       dtype=torch.float32, device=dev)
         self.leaf_vals = torch.empty(self.total_nodes, dtype=torch.float32,
                                      device=dev)
+        # per-node [lo, hi] leaf-value bounds for monotonic constraints
+        self.node_bounds = None
+        if mono is not None and bool((mono != 0).any()):
+            self.node_bounds = torch.empty((self.total_nodes, 2),
+                                           dtype=torch.float32, device=dev)
+        else:
+            self.mono = None
         self.tree_feat = torch.empty(self.total_nodes, dtype=torch.int32,
                                      device=dev)
         self.tree_bin = torch.empty(self.total_nodes, dtype=torch.int32,
@@ -220,6 +229,9 @@ class ForestTrainer:
         self.node_stats.zero_()
         if self.tree_masks is not None:
             self.tree_masks.zero_()
+        if self.node_bounds is not None:
+            self.node_bounds[:, 0] = float("-inf")
+            self.node_bounds[:, 1] = float("inf")
         if sample_mask is None:
             self.node_ids.zero_()
         else:
@@ -342,7 +354,8 @@ class ForestTrainer:
                                cfg.min_examples, cfg.min_gain,
                                feat_mask=feat_mask, cat_flags=self.cat_flags,
                                masks=self.tree_masks,
-                               cat_smooth=cfg.cat_smooth)
+                               cat_smooth=cfg.cat_smooth, mono=self.mono,
+                               node_bounds=self.node_bounds)
 
             prev_fit = n_active <= self.max_slots
             if self.use_hist_sub and prev_fit and level + 1 < cfg.max_depth:
@@ -378,7 +391,8 @@ class ForestTrainer:
                 count_of = {int(a): float(c)
                             for a, c in zip(children, ccounts)}
 
-        ops.leaf_values(self.node_stats, self.leaf_vals, cfg.lambda_l2)
+        ops.leaf_values(self.node_stats, self.leaf_vals, cfg.lambda_l2,
+                        node_bounds=self.node_bounds)
 
     def extract_host_tree(self) -> HostTree:
         cfg = self.cfg
@@ -427,7 +441,8 @@ class ForestTrainer:
                        self.best_gain, 0, level_size, cfg.lambda_l2,
                        cfg.min_hessian, cfg.min_examples, cfg.min_gain,
                        feat_mask=feat_mask, cat_flags=self.cat_flags,
-                       masks=self.tree_masks, cat_smooth=cfg.cat_smooth)
+                       masks=self.tree_masks, cat_smooth=cfg.cat_smooth,
+                       mono=self.mono, node_bounds=self.node_bounds)
         fits = True
         if self.use_hist_sub and level + 1 < cfg.max_depth:
             self.hist_prev[:level_size].copy_(hist_view)

@@ -117,17 +117,20 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
                best_bin: torch.Tensor, best_gain: torch.Tensor, slot0: int,
                n_slots: int, lambda_l2: float, min_hessian: float,
                min_examples: int, min_gain: float, feat_mask=None,
-               cat_flags=None, masks=None, cat_smooth: float = 1.0):
+               cat_flags=None, masks=None, cat_smooth: float = 1.0,
+               mono=None, node_bounds=None):
     F = hist.shape[1]
     n_bins = hist.shape[2]
     mp = feat_mask.data_ptr() if feat_mask is not None else 0
     cf = cat_flags.data_ptr() if cat_flags is not None else 0
     mk = masks.data_ptr() if masks is not None else 0
+    mn = mono.data_ptr() if mono is not None else 0
+    nb = node_bounds.data_ptr() if node_bounds is not None else 0
     args = (hist.data_ptr(), abs_of_slot.data_ptr(), node_stats.data_ptr(),
             best_gain_nf.data_ptr(), best_bin_nf.data_ptr(),
             best_feat.data_ptr(), best_bin.data_ptr(), best_gain.data_ptr(),
-            mp, cf, mk, F, n_bins, slot0, n_slots, lambda_l2, min_hessian,
-            min_examples, min_gain, cat_smooth)
+            mp, cf, mk, mn, nb, F, n_bins, slot0, n_slots, lambda_l2,
+            min_hessian, min_examples, min_gain, cat_smooth)
     if hist.is_cuda:
         _C.gpu_split_scan(*args, _stream())
     else:
@@ -178,13 +181,14 @@ def update_node_ids(bins: torch.Tensor, node_ids: torch.Tensor,
 
 
 def leaf_values(node_stats: torch.Tensor, out: torch.Tensor,
-                lambda_l2: float):
+                lambda_l2: float, node_bounds=None):
     total = out.numel()
+    nb = node_bounds.data_ptr() if node_bounds is not None else 0
     if node_stats.is_cuda:
-        _C.gpu_leaf_values(node_stats.data_ptr(), out.data_ptr(), total,
+        _C.gpu_leaf_values(node_stats.data_ptr(), nb, out.data_ptr(), total,
                            lambda_l2, _stream())
     else:
-        _C.cpu_leaf_values(node_stats.data_ptr(), out.data_ptr(), total,
+        _C.cpu_leaf_values(node_stats.data_ptr(), nb, out.data_ptr(), total,
                            lambda_l2)
     return out
 

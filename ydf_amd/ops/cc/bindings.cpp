@@ -25,8 +25,8 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
 void gpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*,
-                    const uint8_t*, unsigned long long*, int, int, int, int,
-                    SplitParams, void*);
+                    const uint8_t*, unsigned long long*, const int8_t*,
+                    float*, int, int, int, int, SplitParams, void*);
 void gpu_plan_level(const float*, const int32_t*, int, int, int, int,
                     int32_t*, uint8_t*, void*);
 void gpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int,
@@ -35,7 +35,7 @@ void gpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, const uint8_t*,
                          const unsigned long long*, int64_t, int, int,
                          void*);
-void gpu_leaf_values(const float*, float*, int, float, void*);
+void gpu_leaf_values(const float*, const float*, float*, int, float, void*);
 void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
                       void*);
 void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
@@ -56,15 +56,15 @@ void cpu_hist_build(const uint8_t*, const float*, const int32_t*,
 void cpu_weighted_target(const float*, const float*, float*, int64_t);
 void cpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*,
-                    const uint8_t*, unsigned long long*, int, int, int, int,
-                    SplitParams);
+                    const uint8_t*, unsigned long long*, const int8_t*,
+                    float*, int, int, int, int, SplitParams);
 void cpu_plan_level(const float*, const int32_t*, int, int, int, int,
                     int32_t*, uint8_t*);
 void cpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int);
 void cpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, const uint8_t*,
                          const unsigned long long*, int64_t, int, int);
-void cpu_leaf_values(const float*, float*, int, float);
+void cpu_leaf_values(const float*, const float*, float*, int, float);
 void cpu_update_preds(float*, const int32_t*, const float*, int64_t, float);
 void cpu_binary_logloss(const float*, const float*, float*, int64_t);
 void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
@@ -148,7 +148,8 @@ PYBIND11_MODULE(_ydf_ops, m) {
         [](uintptr_t hist, uintptr_t abs_of_slot, uintptr_t node_stats,
            uintptr_t best_gain_nf, uintptr_t best_bin_nf, uintptr_t best_feat,
            uintptr_t best_bin, uintptr_t best_gain, uintptr_t feat_mask,
-           uintptr_t cat_flags, uintptr_t masks, int F, int n_bins, int slot0,
+           uintptr_t cat_flags, uintptr_t masks, uintptr_t mono,
+           uintptr_t node_bounds, int F, int n_bins, int slot0,
            int n_slots, float lambda_l2, float min_hessian, int min_examples,
            float min_gain, float cat_smooth, uintptr_t stream) {
           gpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
@@ -156,8 +157,8 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
                          P<int32_t>(best_bin), P<float>(best_gain),
                          P<uint8_t>(feat_mask), P<uint8_t>(cat_flags),
-                         P<unsigned long long>(masks), F, n_bins, slot0,
-                         n_slots,
+                         P<unsigned long long>(masks), P<int8_t>(mono),
+                         P<float>(node_bounds), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
                                 min_gain, cat_smooth),
                          (void*)stream);
@@ -194,10 +195,11 @@ PYBIND11_MODULE(_ydf_ops, m) {
         },
         nogil);
   m.def("gpu_leaf_values",
-        [](uintptr_t node_stats, uintptr_t leaf_values, int total_nodes,
-           float lambda_l2, uintptr_t stream) {
-          gpu_leaf_values(P<float>(node_stats), P<float>(leaf_values),
-                          total_nodes, lambda_l2, (void*)stream);
+        [](uintptr_t node_stats, uintptr_t node_bounds, uintptr_t leaf_values,
+           int total_nodes, float lambda_l2, uintptr_t stream) {
+          gpu_leaf_values(P<float>(node_stats), P<float>(node_bounds),
+                          P<float>(leaf_values), total_nodes, lambda_l2,
+                          (void*)stream);
         },
         nogil);
   m.def("gpu_update_preds",
@@ -275,7 +277,8 @@ PYBIND11_MODULE(_ydf_ops, m) {
         [](uintptr_t hist, uintptr_t abs_of_slot, uintptr_t node_stats,
            uintptr_t best_gain_nf, uintptr_t best_bin_nf, uintptr_t best_feat,
            uintptr_t best_bin, uintptr_t best_gain, uintptr_t feat_mask,
-           uintptr_t cat_flags, uintptr_t masks, int F, int n_bins, int slot0,
+           uintptr_t cat_flags, uintptr_t masks, uintptr_t mono,
+           uintptr_t node_bounds, int F, int n_bins, int slot0,
            int n_slots, float lambda_l2, float min_hessian, int min_examples,
            float min_gain, float cat_smooth) {
           cpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
@@ -283,8 +286,8 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
                          P<int32_t>(best_bin), P<float>(best_gain),
                          P<uint8_t>(feat_mask), P<uint8_t>(cat_flags),
-                         P<unsigned long long>(masks), F, n_bins, slot0,
-                         n_slots,
+                         P<unsigned long long>(masks), P<int8_t>(mono),
+                         P<float>(node_bounds), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
                                 min_gain, cat_smooth));
         },
@@ -317,10 +320,10 @@ PYBIND11_MODULE(_ydf_ops, m) {
         },
         nogil);
   m.def("cpu_leaf_values",
-        [](uintptr_t node_stats, uintptr_t leaf_values, int total_nodes,
-           float lambda_l2) {
-          cpu_leaf_values(P<float>(node_stats), P<float>(leaf_values),
-                          total_nodes, lambda_l2);
+        [](uintptr_t node_stats, uintptr_t node_bounds, uintptr_t leaf_values,
+           int total_nodes, float lambda_l2) {
+          cpu_leaf_values(P<float>(node_stats), P<float>(node_bounds),
+                          P<float>(leaf_values), total_nodes, lambda_l2);
         },
         nogil);
   m.def("cpu_update_preds",

@@ -226,7 +226,8 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                     int32_t* best_bin_nf, int32_t* best_feat,
                     int32_t* best_bin, float* best_gain,
                     const uint8_t* feat_mask, const uint8_t* cat_flags,
-                    unsigned long long* masks, int F, int n_bins, int slot0,
+                    unsigned long long* masks, const int8_t* mono,
+                    float* node_bounds, int F, int n_bins, int slot0,
                     int n_slots, SplitParams sp) {
   ThreadPool::Get().ParallelFor(n_slots, [&](int slot) {
     std::vector<int> order;
@@ -270,9 +271,15 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
         const float GR = Gf - GL, HR = Hf - HL, CR = Cf - CL;
         if (CL >= sp.min_examples && CR >= sp.min_examples &&
             HL >= sp.min_hessian && HR >= sp.min_hessian) {
+          bool okm = true;
+          if (mono != nullptr && mono[f] != 0) {
+            const float wl = -GL / (HL + sp.lambda_l2);
+            const float wr = -GR / (HR + sp.lambda_l2);
+            okm = (mono[f] > 0) ? (wl <= wr) : (wl >= wr);
+          }
           const float gain = GL * GL / (HL + sp.lambda_l2) +
                              GR * GR / (HR + sp.lambda_l2) - pterm;
-          if (gain > fbest) { fbest = gain; fbin = b; }
+          if (okm && gain > fbest) { fbest = gain; fbin = b; }
         }
       }
       best_gain_nf[(int64_t)slot * F + f] = fbest;
@@ -318,6 +325,24 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
     float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
     nl[0] = GL; nl[1] = HL; nl[2] = CL;
     nr[0] = G - GL; nr[1] = H - HL; nr[2] = C - CL;
+    if (node_bounds != nullptr) {
+      const float lo = node_bounds[2 * abs_node];
+      const float hi = node_bounds[2 * abs_node + 1];
+      float llo = lo, lhi = hi, rlo = lo, rhi = hi;
+      if (!is_cat && mono != nullptr && mono[node_best_f] != 0) {
+        const float wl =
+            std::min(std::max(-GL / (HL + sp.lambda_l2), lo), hi);
+        const float wr = std::min(
+            std::max(-(G - GL) / (H - HL + sp.lambda_l2), lo), hi);
+        const float mid = 0.5f * (wl + wr);
+        if (mono[node_best_f] > 0) { lhi = mid; rlo = mid; }
+        else { llo = mid; rhi = mid; }
+      }
+      node_bounds[2 * (2 * abs_node + 1)] = llo;
+      node_bounds[2 * (2 * abs_node + 1) + 1] = lhi;
+      node_bounds[2 * (2 * abs_node + 2)] = rlo;
+      node_bounds[2 * (2 * abs_node + 2) + 1] = rhi;
+    }
   });
 }
 
@@ -383,11 +408,14 @@ void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
   });
 }
 
-void cpu_leaf_values(const float* node_stats, float* leaf_values,
-                     int total_nodes, float lambda_l2) {
+void cpu_leaf_values(const float* node_stats, const float* node_bounds,
+                     float* leaf_values, int total_nodes, float lambda_l2) {
   for (int i = 0; i < total_nodes; ++i) {
     const float* ns = node_stats + (int64_t)i * 3;
-    leaf_values[i] = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+    float v = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+    if (node_bounds != nullptr)
+      v = std::min(std::max(v, node_bounds[2 * i]), node_bounds[2 * i + 1]);
+    leaf_values[i] = v;
   }
 }
 

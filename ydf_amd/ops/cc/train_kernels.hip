@@ -431,6 +431,7 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
                                        int32_t* __restrict__ best_bin_nf,
                                        const uint8_t* __restrict__ feat_mask,
                                        const uint8_t* __restrict__ cat_flags,
+                                       const int8_t* __restrict__ mono,
                                        int F, int n_bins, int slot0,
                                        SplitParams sp) {
   const int slot = blockIdx.x;
@@ -482,8 +483,18 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
     const float GR = G - GL, HR = H - HL, CR = C - CL;
     if (CL >= sp.min_examples && CR >= sp.min_examples &&
         HL >= sp.min_hessian && HR >= sp.min_hessian) {
-      gain = GL * GL / (HL + sp.lambda_l2) + GR * GR / (HR + sp.lambda_l2) -
-             G * G / (H + sp.lambda_l2);
+      bool ok = true;
+      if (mono != nullptr && mono[f] != 0) {
+        // monotonic constraint (reference monotonic_constraints;
+        // XGBoost-style): reject splits whose child values violate the
+        // declared direction
+        const float wl = -GL / (HL + sp.lambda_l2);
+        const float wr = -GR / (HR + sp.lambda_l2);
+        ok = (mono[f] > 0) ? (wl <= wr) : (wl >= wr);
+      }
+      if (ok)
+        gain = GL * GL / (HL + sp.lambda_l2) + GR * GR / (HR + sp.lambda_l2) -
+               G * G / (H + sp.lambda_l2);
     }
   }
   // Deterministic argmax reduce: higher gain wins; ties -> smaller bin.
@@ -520,6 +531,8 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
                                     float* __restrict__ best_gain,
                                     const uint8_t* __restrict__ cat_flags,
                                     unsigned long long* __restrict__ masks,
+                                    const int8_t* __restrict__ mono,
+                                    float* __restrict__ node_bounds,
                                     int F, int n_bins, int slot0,
                                     SplitParams sp) {
   const int slot = blockIdx.x;
@@ -582,6 +595,24 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
     float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
     nl[0] = GL; nl[1] = HL; nl[2] = CL;
     nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
+    if (node_bounds != nullptr) {
+      const float lo = node_bounds[2 * abs_node];
+      const float hi = node_bounds[2 * abs_node + 1];
+      float llo = lo, lhi = hi, rlo = lo, rhi = hi;
+      if (mono != nullptr && mono[f] != 0) {
+        const float wl =
+            fminf(fmaxf(-GL / (HL + sp.lambda_l2), lo), hi);
+        const float wr = fminf(
+            fmaxf(-(ns[0] - GL) / (ns[1] - HL + sp.lambda_l2), lo), hi);
+        const float mid = 0.5f * (wl + wr);
+        if (mono[f] > 0) { lhi = mid; rlo = mid; }
+        else { llo = mid; rhi = mid; }
+      }
+      node_bounds[2 * (2 * abs_node + 1)] = llo;
+      node_bounds[2 * (2 * abs_node + 1) + 1] = lhi;
+      node_bounds[2 * (2 * abs_node + 2)] = rlo;
+      node_bounds[2 * (2 * abs_node + 2) + 1] = rhi;
+    }
     return;
   }
   // Categorical winner: rebuild the sorted order (same deterministic sort
@@ -628,6 +659,14 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
     float* nr = node_stats + (int64_t)(2 * abs_node + 2) * 3;
     nl[0] = GL; nl[1] = HL; nl[2] = CL;
     nr[0] = ns[0] - GL; nr[1] = ns[1] - HL; nr[2] = ns[2] - CL;
+    if (node_bounds != nullptr) {  // categorical: children inherit bounds
+      node_bounds[2 * (2 * abs_node + 1)] = node_bounds[2 * abs_node];
+      node_bounds[2 * (2 * abs_node + 1) + 1] =
+          node_bounds[2 * abs_node + 1];
+      node_bounds[2 * (2 * abs_node + 2)] = node_bounds[2 * abs_node];
+      node_bounds[2 * (2 * abs_node + 2) + 1] =
+          node_bounds[2 * abs_node + 1];
+    }
   }
 }
 
@@ -749,12 +788,16 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
 // leaf value for every materialized node: -G / (H + l2). Harmless for
 // internal nodes (examples only ever point at leaves).
 __global__ void leaf_values_kernel(const float* __restrict__ node_stats,
+                                   const float* __restrict__ node_bounds,
                                    float* __restrict__ leaf_values,
                                    int total_nodes, float lambda_l2) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= total_nodes) return;
   const float* ns = node_stats + (int64_t)i * 3;
-  leaf_values[i] = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+  float v = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+  if (node_bounds != nullptr)
+    v = fminf(fmaxf(v, node_bounds[2 * i]), node_bounds[2 * i + 1]);
+  leaf_values[i] = v;
 }
 
 __global__ void update_preds_kernel(float* __restrict__ preds,
@@ -970,16 +1013,18 @@ void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                     int32_t* best_bin_nf, int32_t* best_feat,
                     int32_t* best_bin, float* best_gain,
                     const uint8_t* feat_mask, const uint8_t* cat_flags,
-                    unsigned long long* masks, int F, int n_bins, int slot0,
+                    unsigned long long* masks, const int8_t* mono,
+                    float* node_bounds, int F, int n_bins, int slot0,
                     int n_slots, SplitParams sp, void* stream) {
   hipLaunchKernelGGL(split_scan_feat_kernel, dim3(n_slots, F), dim3(n_bins),
                      0, (hipStream_t)stream, hist, abs_of_slot, node_stats,
-                     best_gain_nf, best_bin_nf, feat_mask, cat_flags, F,
-                     n_bins, slot0, sp);
+                     best_gain_nf, best_bin_nf, feat_mask, cat_flags, mono,
+                     F, n_bins, slot0, sp);
   hipLaunchKernelGGL(split_select_kernel, dim3(n_slots), dim3(kBlock), 0,
                      (hipStream_t)stream, hist, abs_of_slot, best_gain_nf,
                      best_bin_nf, node_stats, best_feat, best_bin, best_gain,
-                     cat_flags, masks, F, n_bins, slot0, sp);
+                     cat_flags, masks, mono, node_bounds, F, n_bins, slot0,
+                     sp);
 }
 
 void gpu_plan_level(const float* node_stats, const int32_t* prev_best_feat,
@@ -1013,12 +1058,13 @@ void gpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
                      level_base, level_size);
 }
 
-void gpu_leaf_values(const float* node_stats, float* leaf_values,
-                     int total_nodes, float lambda_l2, void* stream) {
+void gpu_leaf_values(const float* node_stats, const float* node_bounds,
+                     float* leaf_values, int total_nodes, float lambda_l2,
+                     void* stream) {
   const int grid = (total_nodes + kBlock - 1) / kBlock;
   hipLaunchKernelGGL(leaf_values_kernel, dim3(grid), dim3(kBlock), 0,
-                     (hipStream_t)stream, node_stats, leaf_values, total_nodes,
-                     lambda_l2);
+                     (hipStream_t)stream, node_stats, node_bounds,
+                     leaf_values, total_nodes, lambda_l2);
 }
 
 void gpu_update_preds(float* preds, const int32_t* node_ids,
