@@ -1,0 +1,73 @@
+"""Golden-model import tests: load reference YDF models and reproduce the
+reference's own prediction files (reference analogue: ExpectEqualPredictions
+golden checks, utils/test_utils.h:258-297)."""
+import os
+import sys
+
+import numpy as np
+import pytest
+
+import ydf_amd as ydf
+
+BASE = "/root/reference/yggdrasil_decision_forests/test_data"
+
+pytestmark = pytest.mark.skipif(not os.path.exists(BASE),
+                                reason="reference test_data not available")
+sys.setrecursionlimit(100000)
+
+
+def _check(model_name, data_csv, pred_csv, col, tol):
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/{model_name}")
+    te = pd.read_csv(f"{BASE}/dataset/{data_csv}")
+    p = m.predict(te, device="cpu")
+    g = pd.read_csv(f"{BASE}/prediction/{pred_csv}")[col].values
+    assert np.abs(p - g).max() < tol, np.abs(p - g).max()
+    return m
+
+
+def test_golden_gbt_adult():
+    m = _check("adult_binary_class_gbdt", "adult_test.csv",
+               "adult_test_binary_class_gbdt.csv", ">50K", 1e-5)
+    assert m.num_trees() == 68
+
+
+def test_golden_gbt_abalone_regression():
+    _check("abalone_regression_gbdt", "abalone.csv",
+           "abalone_regression_gbdt.csv", "Rings", 1e-3)
+
+
+def test_rf_small_models_load():
+    for name in ("adult_binary_class_rf_wta_small",
+                 "adult_binary_class_rf_nwta_small"):
+        m = ydf.load_ydf_model(f"{BASE}/model/{name}")
+        assert m.num_trees() == 10
+        pd = pytest.importorskip("pandas")
+        p = m.predict(pd.read_csv(f"{BASE}/dataset/adult_test.csv"),
+                      device="cpu")
+        assert 0.0 <= p.min() and p.max() <= 1.0
+        assert 0.15 < p.mean() < 0.35  # base rate ~0.24
+
+
+def test_multiclass_gbt_loads():
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/iris_multi_class_gbdt")
+    p = m.predict(pd.read_csv(f"{BASE}/dataset/iris.csv"), device="cpu")
+    assert p.shape[1] == 3
+    np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-4)
+    # the model should classify its own training data well
+    labels = pd.read_csv(f"{BASE}/dataset/iris.csv")["class"].values
+    pred_cls = np.asarray(m.label_classes)[p.argmax(axis=1)]
+    assert (pred_cls == labels).mean() > 0.95
+
+
+def test_imported_model_reserialization(tmp_path):
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/adult_binary_class_gbdt")
+    p1 = m.predict(pd.read_csv(f"{BASE}/dataset/adult_test.csv"),
+                   device="cpu")
+    m.save(str(tmp_path / "m"))
+    m2 = ydf.load_model(str(tmp_path / "m"))
+    p2 = m2.predict(pd.read_csv(f"{BASE}/dataset/adult_test.csv"),
+                    device="cpu")
+    np.testing.assert_allclose(p1, p2, atol=1e-6)

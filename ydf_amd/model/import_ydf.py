@@ -1,0 +1,415 @@
+"""Imports reference Yggdrasil Decision Forests models (read-only).
+
+Decodes the reference's on-disk model directory (model/model_library.cc:
+92-107: header.pb + data_spec.pb + <prefix>gradient_boosted_trees_header.pb /
+random_forest_header.pb + nodes-XXXXX-of-YYYYY + done) with a hand-rolled
+protobuf *wire-format* reader — no protoc, no schema files; the field
+numbers below are documented against the reference .proto sources.
+
+Supported: GBT (binary/multiclass/regression) and RF models with numerical
+(Higher/DiscretizedHigher), categorical (ContainsBitmap/ContainsVector) and
+boolean (TrueValue) conditions. Oblique and vector-sequence conditions are
+not supported (ROADMAP).
+"""
+from __future__ import annotations
+
+import gzip
+import os
+import struct
+from typing import Dict, Iterator, List, Optional, Tuple
+
+import numpy as np
+
+from ydf_amd.dataset.dataspec import (ColumnSpec, DataSpecification,
+                                      OOV_ITEM, Semantic, Task)
+from ydf_amd.model.forest import FlatForest
+from ydf_amd.model.specialized import (GradientBoostedTreesModel,
+                                       RandomForestModel)
+
+
+# ---------------------------------------------------------------------------
+# Minimal protobuf wire-format reader
+# ---------------------------------------------------------------------------
+class Wire:
+    def __init__(self, data: bytes):
+        self.d = data
+        self.p = 0
+        self.n = len(data)
+
+    def varint(self) -> int:
+        r = 0
+        s = 0
+        while True:
+            b = self.d[self.p]
+            self.p += 1
+            r |= (b & 0x7F) << s
+            if not b & 0x80:
+                return r
+            s += 7
+
+    def fields(self) -> Iterator[Tuple[int, int, object]]:
+        """Yields (field_number, wire_type, value)."""
+        while self.p < self.n:
+            tag = self.varint()
+            fn, wt = tag >> 3, tag & 7
+            if wt == 0:
+                yield fn, wt, self.varint()
+            elif wt == 1:
+                v = struct.unpack_from("<d", self.d, self.p)[0]
+                self.p += 8
+                yield fn, wt, v
+            elif wt == 2:
+                ln = self.varint()
+                yield fn, wt, self.d[self.p:self.p + ln]
+                self.p += ln
+            elif wt == 5:
+                v = struct.unpack_from("<f", self.d, self.p)[0]
+                self.p += 4
+                yield fn, wt, v
+            else:
+                raise ValueError(f"unsupported wire type {wt}")
+
+
+def _msg(data: bytes) -> Dict[int, list]:
+    out: Dict[int, list] = {}
+    for fn, _, v in Wire(data).fields():
+        out.setdefault(fn, []).append(v)
+    return out
+
+
+def _f32(v) -> float:
+    if isinstance(v, float):
+        return v
+    return struct.unpack("<f", struct.pack("<I", v & 0xFFFFFFFF))[0]
+
+
+def _packed_varints(b: bytes) -> List[int]:
+    w = Wire(b)
+    out = []
+    while w.p < w.n:
+        out.append(w.varint())
+    return out
+
+
+def _packed_floats(b: bytes) -> List[float]:
+    return list(struct.unpack(f"<{len(b) // 4}f", b))
+
+
+def _packed_doubles(b: bytes) -> List[float]:
+    return list(struct.unpack(f"<{len(b) // 8}d", b))
+
+
+# ---------------------------------------------------------------------------
+# Blob sequence (reference utils/blob_sequence.h:125-150): "BS" magic,
+# u16 version, u8 compression, 3 reserved bytes; then records of
+# u32-LE length + payload.
+# ---------------------------------------------------------------------------
+def read_blob_sequence(path: str) -> Iterator[bytes]:
+    with open(path, "rb") as f:
+        data = f.read()
+    if data[:2] != b"BS":
+        raise ValueError(f"{path}: not a blob sequence")
+    version, compression = struct.unpack_from("<HB", data, 2)
+    p = 8
+    if compression == 1:
+        data = data[:8] + gzip.decompress(data[8:])
+    while p + 4 <= len(data):
+        (ln,) = struct.unpack_from("<I", data, p)
+        p += 4
+        yield data[p:p + ln]
+        p += ln
+
+
+# ---------------------------------------------------------------------------
+# data_spec.pb (reference dataset/data_spec.proto)
+# ---------------------------------------------------------------------------
+_COLTYPE = {1: Semantic.NUMERICAL, 4: Semantic.CATEGORICAL,
+            7: Semantic.BOOLEAN, 9: Semantic.DISCRETIZED_NUMERICAL,
+            10: Semantic.HASH, 5: Semantic.CATEGORICAL_SET}
+
+
+def parse_data_spec(raw: bytes):
+    """Returns (columns, boundaries_by_col). Column fields: type=1, name=2,
+    numerical=5 {mean=1,min=2,max=3}, categorical=6 {number_of_unique=2,
+    is_already_integerized=5, items map=7 {key=1,value=2{index=1,count=2}}},
+    discretized_numerical=8 {boundaries=1 packed f32}, boolean=9."""
+    spec = _msg(raw)
+    columns = []
+    disc_bounds = {}
+    for ci, colb in enumerate(spec.get(1, [])):
+        col = _msg(colb)
+        ctype = col.get(1, [1])[0]
+        name = col.get(2, [b""])[0].decode()
+        sem = _COLTYPE.get(ctype, Semantic.NUMERICAL)
+        mean = 0.0
+        vocab = None
+        if 5 in col:  # numerical spec
+            num = _msg(col[5][0])
+            mean = _f32(num.get(1, [0.0])[0]) if 1 in num else 0.0
+        if sem in (Semantic.CATEGORICAL, Semantic.CATEGORICAL_SET) \
+                and 6 in col:
+            cat = _msg(col[6][0])
+            n_unique = cat.get(2, [0])[0]
+            integerized = bool(cat.get(5, [0])[0])
+            if integerized:
+                vocab = [str(i) for i in range(n_unique)]
+                vocab[0] = OOV_ITEM
+            else:
+                vocab = [OOV_ITEM] * max(n_unique, 1)
+                for entry in cat.get(7, []):
+                    e = _msg(entry)
+                    key = e.get(1, [b""])[0].decode()
+                    val = _msg(e.get(2, [b""])[0])
+                    idx = val.get(1, [0])[0]
+                    if idx < len(vocab):
+                        vocab[idx] = key
+        if sem == Semantic.DISCRETIZED_NUMERICAL and 8 in col:
+            dn = _msg(col[8][0])
+            if 1 in dn:
+                disc_bounds[ci] = np.asarray(_packed_floats(dn[1][0]),
+                                             dtype=np.float32)
+        columns.append(ColumnSpec(
+            name=name,
+            semantic=(Semantic.NUMERICAL
+                      if sem == Semantic.DISCRETIZED_NUMERICAL else sem),
+            vocab=vocab, mean=mean))
+    return columns, disc_bounds
+
+
+# ---------------------------------------------------------------------------
+# Node records (reference model/decision_tree/decision_tree.proto): Node
+# {classifier=1 {top_value=1, distribution=2 {counts=1 packed double,
+# sum=2}}, regressor=2 {top_value=1}, condition=3 NodeCondition
+# {na_value=1, attribute=2, condition=3 Condition{higher=2{threshold=1},
+# true_value=3, contains=4{elements=1 packed}, contains_bitmap=5{bytes=1},
+# discretized_higher=6{threshold=1}}},
+# num_training_examples_with_weight=5}.
+# Written pre-order, NEGATIVE child first (decision_tree.cc:580-585).
+# ---------------------------------------------------------------------------
+class _NodeRec:
+    __slots__ = ("is_leaf", "attr", "thr", "mask", "value", "cover",
+                 "na_value")
+
+
+def parse_node(raw: bytes, disc_bounds, n_classes: int,
+               binary_class_idx: int = 2) -> _NodeRec:
+    node = _msg(raw)
+    r = _NodeRec()
+    r.is_leaf = 3 not in node
+    r.attr = -1
+    r.thr = 0.0
+    r.mask = None
+    r.value = 0.0
+    r.cover = 0.0
+    r.na_value = False
+    if 5 in node:
+        pass
+    if 1 in node:  # classifier output
+        cls = _msg(node[1][0])
+        if 2 in cls:
+            dist = _msg(cls[2][0])
+            counts = _packed_doubles(dist.get(1, [b""])[0]) \
+                if dist.get(1) else []
+            total = dist.get(2, [0.0])[0]
+            if n_classes == 2:
+                c = counts[binary_class_idx] if len(counts) > \
+                    binary_class_idx else 0.0
+                r.value = c / total if total else 0.0
+            else:
+                r.value = [c / total if total else 0.0
+                           for c in counts[1:1 + n_classes]]
+        else:
+            r.value = float(cls.get(1, [0])[0])
+    elif 2 in node:  # regressor output
+        reg = _msg(node[2][0])
+        r.value = _f32(reg.get(1, [0.0])[0]) if 1 in reg else 0.0
+    if not r.is_leaf:
+        cond = _msg(node[3][0])
+        r.na_value = bool(cond.get(1, [0])[0])
+        r.attr = cond.get(2, [0])[0]
+        inner = _msg(cond.get(3, [b""])[0])
+        if 4 in cond:
+            r.cover = float(cond[4][0])
+        if 2 in inner:  # Higher: value >= threshold
+            thr = _msg(inner[2][0])
+            t = _f32(thr.get(1, [0.0])[0]) if 1 in thr else 0.0
+            # our kernels test strict >; x >= t  <=>  x > nextafter(t, -inf)
+            r.thr = float(np.nextafter(np.float32(t), np.float32("-inf")))
+        elif 6 in inner:  # DiscretizedHigher: bin_index >= threshold
+            thr = _msg(inner[6][0])
+            t = int(thr.get(1, [0])[0])
+            b = disc_bounds.get(r.attr)
+            if b is None or t - 1 >= len(b) or t < 1:
+                raise ValueError("discretized condition without boundaries")
+            # bin(v) >= t  <=>  v >= boundaries[t-1]
+            r.thr = float(np.nextafter(np.float32(b[t - 1]),
+                                       np.float32("-inf")))
+        elif 3 in inner:  # TrueValue (boolean): v == True -> positive
+            r.thr = 0.5
+        elif 4 in inner:  # ContainsVector
+            els = _msg(inner[4][0])
+            m = np.zeros(4, dtype=np.uint64)
+            for e in (_packed_varints(els[1][0]) if els.get(1) else []):
+                if e < 256:
+                    m[e >> 6] |= np.uint64(1 << (e & 63))
+            r.mask = m
+        elif 5 in inner:  # ContainsBitmap
+            bm = _msg(inner[5][0]).get(1, [b""])[0]
+            m = np.zeros(32, dtype=np.uint8)
+            m[:min(len(bm), 32)] = np.frombuffer(bm[:32], dtype=np.uint8)
+            r.mask = m.view(np.uint64)
+        else:
+            raise ValueError(
+                f"unsupported condition type (fields {list(inner)})")
+    return r
+
+
+# ---------------------------------------------------------------------------
+# Model assembly
+# ---------------------------------------------------------------------------
+def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
+                value_scale: float = 1.0, wta: bool = False):
+    shards = sorted(p for p in os.listdir(model_dir)
+                    if p.startswith(prefix + "nodes-"))
+    records: List[bytes] = []
+    for s in shards:
+        records.extend(read_blob_sequence(os.path.join(model_dir, s)))
+    feats, thrs, lefts, roots, cidx, masks, covers = \
+        [], [], [], [], [], [], []
+    pos = 0
+
+    def new_slot():
+        feats.append(-1)
+        thrs.append(0.0)
+        lefts.append(0)
+        cidx.append(-1)
+        covers.append(0.0)
+        return len(feats) - 1
+
+    def fill_node(idx):
+        # consumes the next record into slot idx; children are allocated as
+        # an ADJACENT pair (our flat layout needs right == left + 1, which
+        # the on-disk DFS pre-order does not give for free)
+        nonlocal pos
+        rec = parse_node(records[pos], disc_bounds, n_classes)
+        pos += 1
+        covers[idx] = rec.cover
+        if rec.is_leaf:
+            if wta and not isinstance(rec.value, list):
+                # winner-take-all: the tree votes its majority class
+                # (binary: leaf P(class2) > 0.5 -> vote 1)
+                thrs[idx] = 1.0 if rec.value > 0.5 else 0.0
+            else:
+                thrs[idx] = float(rec.value) * value_scale \
+                    if not isinstance(rec.value, list) else 0.0
+            return
+        feats[idx] = rec.attr
+        if rec.mask is not None:
+            cidx[idx] = len(masks)
+            masks.append(rec.mask)
+        else:
+            thrs[idx] = rec.thr
+        li = new_slot()
+        new_slot()
+        lefts[idx] = li
+        fill_node(li)        # negative child first on disk
+        fill_node(li + 1)
+
+    while pos < len(records):
+        root = new_slot()
+        roots.append(root)
+        fill_node(root)
+    return FlatForest(
+        feat=np.asarray(feats, np.int32), thr=np.asarray(thrs, np.float32),
+        left=np.asarray(lefts, np.int32), roots=np.asarray(roots, np.int32),
+        cat_idx=np.asarray(cidx, np.int32),
+        masks=np.stack(masks).astype(np.uint64) if masks
+        else np.zeros((0, 4), np.uint64),
+        cover=np.asarray(covers, np.float32))
+
+
+def load_ydf_model(path: str, file_prefix: str = ""):
+    """Loads a reference YDF model directory as a ydf_amd model."""
+    with open(os.path.join(path, file_prefix + "header.pb"), "rb") as f:
+        header = _msg(f.read())
+    with open(os.path.join(path, file_prefix + "data_spec.pb"), "rb") as f:
+        columns, disc_bounds = parse_data_spec(f.read())
+    task = Task(header.get(2, [1])[0]) if header.get(2, [1])[0] in (
+        1, 2, 3, 6) else Task.CLASSIFICATION
+    label_idx = header.get(3, [len(columns) - 1])[0]
+    input_features = []
+    for v in header.get(5, []):
+        if isinstance(v, bytes):
+            input_features.extend(_packed_varints(v))
+        else:
+            input_features.append(v)
+    label_name = columns[label_idx].name
+    # order dataspec features like the model's input_features
+    feat_cols = [columns[i] for i in input_features]
+    dataspec = DataSpecification(columns=feat_cols + [columns[label_idx]],
+                                 label=label_name)
+    # remap attribute indices (original column idx -> dense feature idx)
+    remap = {ci: i for i, ci in enumerate(input_features)}
+
+    gbt_hdr_path = os.path.join(
+        path, file_prefix + "gradient_boosted_trees_header.pb")
+    rf_hdr_path = os.path.join(path, file_prefix + "random_forest_header.pb")
+    label_vocab = columns[label_idx].vocab
+    classes = list(label_vocab[1:]) if label_vocab else None
+    n_classes = len(classes) if classes else 2
+
+    if os.path.exists(gbt_hdr_path):
+        # gbt header (model/gradient_boosted_trees/gradient_boosted_trees.
+        # proto): num_trees=2, loss=3, initial_predictions=4 (repeated f32),
+        # num_trees_per_iter=5
+        with open(gbt_hdr_path, "rb") as f:
+            gh = _msg(f.read())
+        inits = [(_f32(v) if not isinstance(v, bytes) else None)
+                 for v in gh.get(4, [])]
+        init_preds = []
+        for v in gh.get(4, []):
+            if isinstance(v, bytes):
+                init_preds.extend(_packed_floats(v))
+            else:
+                init_preds.append(_f32(v))
+        if not init_preds:
+            init_preds = [0.0]
+        loss = gh.get(3, [0])[0]
+        forest = _read_trees(path, file_prefix, disc_bounds, n_classes)
+        ntpi = gh.get(5, [1])[0]
+        activation = "identity"
+        if task == Task.CLASSIFICATION:
+            activation = "softmax" if ntpi > 1 else "sigmoid"
+        model = GradientBoostedTreesModel(
+            forest=_remap_forest(forest, remap), dataspec=dataspec,
+            task=task, label_classes=classes, init_predictions=init_preds,
+            num_trees_per_iter=ntpi, activation=activation,
+            metadata={"imported_from": "yggdrasil-decision-forests",
+                      "loss": int(loss)})
+        return model
+    if os.path.exists(rf_hdr_path):
+        # rf header (model/random_forest/random_forest.proto): num_trees=2,
+        # winner_take_all_inference=3 (default true)
+        with open(rf_hdr_path, "rb") as f:
+            rh = _msg(f.read())
+        wta = bool(rh.get(3, [1])[0])
+        forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
+                             wta=wta and task == Task.CLASSIFICATION)
+        model = RandomForestModel(
+            forest=_remap_forest(forest, remap), dataspec=dataspec,
+            task=task, label_classes=classes,
+            init_predictions=[0.0],
+            num_trees_per_iter=1, activation="identity",
+            metadata={"imported_from": "yggdrasil-decision-forests",
+                      "winner_take_all": wta})
+        return model
+    raise ValueError(f"unsupported or missing model header in {path}")
+
+
+def _remap_forest(forest: FlatForest, remap: Dict[int, int]) -> FlatForest:
+    feat = forest.feat.copy()
+    for i in range(len(feat)):
+        if feat[i] >= 0:
+            feat[i] = remap.get(int(feat[i]), 0)
+    forest.feat = feat
+    return forest
