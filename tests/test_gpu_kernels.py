@@ -232,3 +232,14 @@ def test_adult_gpu_quality(adult_paths):
     ev = m.evaluate(pd.read_csv(te), device="cuda")
     assert ev.accuracy > 0.86
     assert ev.auc > 0.92
+
+
+def test_monotonic_gpu(regression_data):
+    feats = [ydf.Feature("x1", monotonic=1), ydf.Feature("x2")]
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, features=feats,
+        num_trees=40, validation_ratio=0, device="cuda").train(
+            regression_data)
+    grid = np.linspace(-3, 3, 200).astype(np.float32)
+    p = m.predict({"x1": grid, "x2": np.zeros_like(grid)}, device="cuda")
+    assert np.diff(p).min() >= -1e-6

@@ -71,3 +71,34 @@ def test_imported_model_reserialization(tmp_path):
     p2 = m2.predict(pd.read_csv(f"{BASE}/dataset/adult_test.csv"),
                     device="cpu")
     np.testing.assert_allclose(p1, p2, atol=1e-6)
+
+
+def test_export_roundtrip(tmp_path, binary_data):
+    """Our GBT model written in the reference format and read back through
+    the importer must predict identically."""
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=15).train(
+        binary_data)
+    p = str(tmp_path / "ydf_format")
+    ydf.export_ydf_model(m, p)
+    assert os.path.exists(os.path.join(p, "done"))
+    m2 = ydf.load_ydf_model(p)
+    np.testing.assert_allclose(m.predict(binary_data, device="cpu"),
+                               m2.predict(binary_data, device="cpu"),
+                               atol=2e-6)
+
+
+def test_export_roundtrip_categorical(tmp_path):
+    rng = np.random.RandomState(3)
+    n = 4000
+    cats = rng.randint(0, 8, n)
+    d = {"c": np.array([f"v{v}" for v in cats]),
+         "x": rng.randn(n).astype(np.float32),
+         "label": np.where((cats % 3 == 0) ^ (rng.rand(n) < 0.1),
+                           "p", "n")}
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=10,
+                                        validation_ratio=0).train(d)
+    p = str(tmp_path / "ydf_cat")
+    ydf.export_ydf_model(m, p)
+    m2 = ydf.load_ydf_model(p)
+    np.testing.assert_allclose(m.predict(d, device="cpu"),
+                               m2.predict(d, device="cpu"), atol=2e-6)

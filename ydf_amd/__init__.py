@@ -37,6 +37,7 @@ from ydf_amd.model.model_lib import (
     serialize_model,
 )
 from ydf_amd.model.import_ydf import load_ydf_model
+from ydf_amd.model.export_ydf import export_ydf_model
 
 # alias mirroring ydf.from_tensorflow_decision_forests-style importers
 from_yggdrasil_model = load_ydf_model
