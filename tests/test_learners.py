@@ -378,3 +378,49 @@ def test_monotonic_constraint():
     p = m.predict({"x": grid, "z": np.zeros_like(grid)})
     assert np.diff(p).min() >= -1e-6
     assert m.evaluate(d).rmse < 0.5
+
+
+def test_cart_pruning_reduces_overfit():
+    rng = np.random.RandomState(0)
+    n = 8000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where(x1 > 0, "p", "n")
+    flip = rng.rand(n) < 0.25
+    y = np.where(flip, np.where(y == "p", "n", "p"), y)
+    d = {"x1": x1, "x2": x2, "label": y}
+    m_pruned = ydf.CartLearner(label="label").train(d)
+    m_raw = ydf.CartLearner(label="label", validation_ratio=0).train(d)
+    assert m_pruned.num_nodes() < m_raw.num_nodes() / 2
+    x1t = rng.randn(4000).astype(np.float32)
+    dt = {"x1": x1t, "x2": rng.randn(4000).astype(np.float32),
+          "label": np.where(x1t > 0, "p", "n")}
+    assert m_pruned.evaluate(dt).accuracy > m_raw.evaluate(dt).accuracy
+
+
+def test_poisson_loss():
+    rng = np.random.RandomState(0)
+    n = 6000
+    x = rng.randn(n).astype(np.float32)
+    lam = np.exp(0.5 + x)
+    y = rng.poisson(lam).astype(np.float32)
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, loss="POISSON",
+        num_trees=60).train({"x": x, "label": y})
+    p = m.predict({"x": x})
+    assert np.corrcoef(p, lam)[0, 1] > 0.95
+    assert p.min() >= 0  # exp link
+
+
+def test_mae_loss_robust_to_outliers():
+    rng = np.random.RandomState(0)
+    n = 6000
+    x = rng.randn(n).astype(np.float32)
+    y = (2 * x).astype(np.float32)
+    y[:150] += 100.0
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, loss="MEAN_AVERAGE_ERROR",
+        num_trees=150, shrinkage=0.3, validation_ratio=0).train(
+            {"x": x, "label": y})
+    p = m.predict({"x": x})
+    assert np.median(np.abs(p - 2 * x)) < 0.5

@@ -120,7 +120,13 @@ class GradientBoostedTreesLearner(GenericLearner):
             loss = (trainer_lib.LOSS_MULTINOMIAL if n_classes > 2
                     else trainer_lib.LOSS_BINOMIAL)
         elif self._task == Task.REGRESSION:
-            loss = trainer_lib.LOSS_SQUARED_ERROR
+            named = hp.get("loss") if isinstance(hp.get("loss"), str) \
+                else "DEFAULT"
+            loss = {"DEFAULT": trainer_lib.LOSS_SQUARED_ERROR,
+                    "SQUARED_ERROR": trainer_lib.LOSS_SQUARED_ERROR,
+                    "POISSON": trainer_lib.LOSS_POISSON,
+                    "MEAN_AVERAGE_ERROR": trainer_lib.LOSS_MAE,
+                    }.get(named, trainer_lib.LOSS_SQUARED_ERROR)
         elif self._task == Task.RANKING:
             loss = trainer_lib.LOSS_LAMBDA_MART_NDCG
         else:
@@ -214,6 +220,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                 activation = "sigmoid"
             elif loss == trainer_lib.LOSS_MULTINOMIAL:
                 activation = "softmax"
+            elif loss == trainer_lib.LOSS_POISSON:
+                activation = "exp"
         cat_feats = self._cat_feature_flags(ds)
         names = [c.name for c in ds.dataspec.feature_columns]
 
@@ -431,6 +439,8 @@ class RandomForestLearner(GenericLearner):
         # winner_take_all (reference RF default): each tree votes its
         # majority class; for binary trees this is a leaf-value threshold,
         # so the vote transform happens at model-build time
+        self._last_trees = trees
+        self._last_ds = ds
         wta = (hp["winner_take_all"]
                and self._task == Task.CLASSIFICATION and n_classes == 2)
         if wta:
@@ -451,8 +461,8 @@ class RandomForestLearner(GenericLearner):
 
 
 class CartLearner(RandomForestLearner):
-    """CART: a single tree, no bagging, all features as candidates
-    (reference learner/cart/cart.h:44; validation-set pruning TODO)."""
+    """CART: a single tree, no bagging, all features as candidates, with
+    validation-set pruning (reference learner/cart/cart.h:44)."""
 
     def __init__(self, label: Optional[str] = None,
                  task: Task = Task.CLASSIFICATION,
@@ -461,9 +471,48 @@ class CartLearner(RandomForestLearner):
         kwargs.setdefault("num_trees", 1)
         kwargs.setdefault("bootstrap_training_dataset", False)
         kwargs.setdefault("num_candidate_attributes", -1)
+        kwargs.setdefault("compute_oob_performances", False)
         super().__init__(label=label, task=task, max_depth=max_depth,
                          min_examples=min_examples, **kwargs)
         self.hyperparameters["validation_ratio"] = validation_ratio
+
+    def train(self, data, valid=None, verbose=None):
+        vr = self.hyperparameters.get("validation_ratio", 0.1)
+        from ydf_amd.dataset.dataset import _to_column_dict
+
+        if vr <= 0 or self._task not in (Task.CLASSIFICATION,
+                                         Task.REGRESSION):
+            return super().train(data, valid=valid, verbose=verbose)
+        cols = _to_column_dict(data)
+        n = len(next(iter(cols.values())))
+        rng = np.random.RandomState(self.random_seed)
+        perm = rng.permutation(n)
+        n_valid = int(n * vr)
+        if n_valid < 10:
+            return super().train(data, valid=valid, verbose=verbose)
+        vidx, tidx = perm[:n_valid], perm[n_valid:]
+        train_cols = {k: np.asarray(v)[tidx] for k, v in cols.items()}
+        model = super().train(train_cols, verbose=verbose)
+        # prune with the held-out rows (reference cart.cc pruning)
+        from ydf_amd.learner.pruning import prune_tree
+        from ydf_amd.model.forest import build_flat_forest, \
+            padded_boundaries
+
+        tree = self._last_trees[0]
+        ds = self._last_ds
+        bnd = padded_boundaries(ds.dataspec.feature_columns)
+        cat_feats = self._cat_feature_flags(ds)
+        valid_cols = {k: np.asarray(v)[vidx] for k, v in cols.items()}
+        from ydf_amd.dataset.dataset import create_vertical_dataset
+
+        vds = create_vertical_dataset(valid_cols, dataspec=ds.dataspec)
+        pruned = prune_tree(tree, vds.X, vds.label_values, bnd, cat_feats,
+                            self._task)
+        info(f"CART pruning removed {pruned} nodes")
+        model.forest = build_flat_forest([tree], bnd, leaf_scale=1.0,
+                                         cat_feats=cat_feats)
+        model._dev_forest = {}
+        return model
 
 
 class IsolationForestLearner(GenericLearner):

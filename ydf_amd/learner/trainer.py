@@ -30,6 +30,8 @@ from ydf_amd import ops
 LOSS_SQUARED_ERROR = 2
 LOSS_BINOMIAL = 1
 LOSS_MULTINOMIAL = 3
+LOSS_POISSON = 7
+LOSS_MAE = 8
 LOSS_LAMBDA_MART_NDCG = 9
 LOSS_RF = 100  # weighted-target mode (RF/CART): not a GBT loss
 
@@ -545,6 +547,14 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
         trainer._allreduce(s)
         init = float((s[0] / s[1]).item())
         init_preds = [init]
+    elif cfg.loss == LOSS_POISSON:
+        s = torch.stack([y.sum(), counts[0]])
+        trainer._allreduce(s)
+        init = float(torch.log((s[0] / s[1]).clamp(min=1e-9)).item())
+        init_preds = [init]
+    elif cfg.loss == LOSS_MAE:
+        init = float(y.median().item())
+        init_preds = [init]
     else:  # multinomial: zeros
         init = 0.0
         init_preds = [0.0] * C
@@ -699,8 +709,14 @@ def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
         s = torch.cat([loss_buf[:1], n])
         trainer._allreduce(s)
         return float((s[0] / s[1]).item())
-    if cfg.loss == LOSS_SQUARED_ERROR:
-        s = torch.stack([((preds[0] - labels) ** 2).sum(),
+    if cfg.loss in (LOSS_SQUARED_ERROR, LOSS_POISSON, LOSS_MAE):
+        if cfg.loss == LOSS_MAE:
+            per = (preds[0] - labels).abs()
+        elif cfg.loss == LOSS_POISSON:
+            per = preds[0].clamp(max=15).exp() - labels * preds[0]
+        else:
+            per = (preds[0] - labels) ** 2
+        s = torch.stack([per.sum(),
                          torch.tensor(float(labels.numel()),
                                       device=preds.device)])
         trainer._allreduce(s)

@@ -179,6 +179,8 @@ class GenericModel:
             return out
         if self.activation == "softmax":
             return torch.softmax(m, dim=0).T.contiguous()
+        if self.activation == "exp":  # Poisson log link
+            return torch.exp(m[0].clamp(max=30))
         if m.shape[0] == 1:
             return m[0]
         return m.T.contiguous()

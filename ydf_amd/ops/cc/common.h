@@ -27,6 +27,8 @@ enum LossKind : int {
   kLossSquaredError = 2,      // regression: g = pred - y, h = 1
   kLossBinomial = 1,          // binary classification log-loss on logits
   kLossMultinomial = 3,       // multi-class softmax cross-entropy
+  kLossPoisson = 7,           // log-link Poisson: g = exp(p) - y, h = exp(p)
+  kLossMAE = 8,               // mean absolute error: g = sign(p - y), h = 1
 };
 
 // Split-scan hyper-parameters (subset of the reference decision-tree proto).

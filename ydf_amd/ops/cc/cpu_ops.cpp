@@ -140,6 +140,13 @@ void cpu_grad_hess(const float* preds, const float* labels, float* gh,
         const float p = 1.0f / (1.0f + std::exp(-preds[k]));
         g = p - labels[k];
         h = std::max(p * (1.0f - p), 1e-16f);
+      } else if (loss == kLossPoisson) {
+        const float ep = std::exp(std::min(preds[k], 15.0f));
+        g = ep - labels[k];
+        h = std::max(ep, 1e-6f);
+      } else if (loss == kLossMAE) {
+        g = (preds[k] > labels[k]) ? 1.0f : -1.0f;
+        h = 1.0f;
       } else {
         g = preds[k] - labels[k];
         h = 1.0f;

@@ -81,6 +81,13 @@ __global__ void grad_hess_kernel(const float* __restrict__ preds,
       const float p = 1.0f / (1.0f + __expf(-preds[k]));
       g = p - labels[k];
       h = fmaxf(p * (1.0f - p), 1e-16f);
+    } else if (loss == kLossPoisson) {
+      const float ep = __expf(fminf(preds[k], 15.0f));
+      g = ep - labels[k];
+      h = fmaxf(ep, 1e-6f);
+    } else if (loss == kLossMAE) {
+      g = (preds[k] > labels[k]) ? 1.0f : -1.0f;
+      h = 1.0f;
     } else {  // squared error
       g = preds[k] - labels[k];
       h = 1.0f;
