@@ -26,12 +26,45 @@ from ydf_amd.dataset.dataspec import (
 InputData = Union[dict, "pandas.DataFrame", str]  # noqa: F821
 
 
+def expand_sharded_paths(path: str):
+    """Typed/sharded dataset paths (reference dataset/formats.proto:23-31 +
+    utils/sharded_io.h ExpandInputShards): supports "csv:" prefixes,
+    "path@N" shard counts, comma lists and globs."""
+    import glob as globlib
+
+    fmt = "csv"
+    if ":" in path and path.split(":", 1)[0] in ("csv", "tfrecord", "avro"):
+        fmt, path = path.split(":", 1)
+    out = []
+    for part in path.split(","):
+        if "@" in part:
+            base, n = part.rsplit("@", 1)
+            n = int(n)
+            stem, dot, ext = base.rpartition(".")
+            if dot:
+                out.extend(f"{stem}-{i:05d}-of-{n:05d}.{ext}"
+                           for i in range(n))
+            else:
+                out.extend(f"{base}-{i:05d}-of-{n:05d}" for i in range(n))
+        elif any(ch in part for ch in "*?["):
+            out.extend(sorted(globlib.glob(part)))
+        else:
+            out.append(part)
+    return fmt, out
+
+
 def _to_column_dict(data: InputData) -> Dict[str, np.ndarray]:
     if isinstance(data, str):
-        path = data[4:] if data.startswith("csv:") else data
+        fmt, paths = expand_sharded_paths(data)
+        if fmt != "csv":
+            raise NotImplementedError(
+                f"dataset format {fmt!r} not supported yet (ROADMAP)")
         import pandas as pd
 
-        return _to_column_dict(pd.read_csv(path))
+        frames = [pd.read_csv(p) for p in paths]
+        df = frames[0] if len(frames) == 1 else pd.concat(
+            frames, ignore_index=True)
+        return _to_column_dict(df)
     if isinstance(data, dict):
         return {k: np.asarray(v) for k, v in data.items()}
     # pandas DataFrame (duck-typed to avoid a hard dependency)

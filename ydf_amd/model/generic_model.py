@@ -341,6 +341,8 @@ class GenericModel:
 
     # ------------------------------------------------------------------
     def describe(self, output_format: str = "text") -> str:
+        """Model card (reference model/describe.{h,cc}; ydf
+        model.describe()). output_format: "text" or "html"."""
         lines = [
             f'type: "{self._model_type}"',
             f"task: {self._task.name}",
@@ -352,7 +354,56 @@ class GenericModel:
         ]
         if self.label_classes:
             lines.append(f"classes: {self.label_classes}")
+        if self.num_trees():
+            sizes = np.diff(np.append(self.forest.roots,
+                                      self.forest.n_nodes))
+            lines.append(f"nodes per tree: min={int(sizes.min())} "
+                         f"mean={float(sizes.mean()):.1f} "
+                         f"max={int(sizes.max())}")
+        vi = self.variable_importances()
+        top = (vi.get("SUM_SCORE") or vi.get("NUM_NODES") or [])[:8]
+        if top:
+            lines.append("top features (" +
+                         ("SUM_SCORE" if "SUM_SCORE" in vi else "NUM_NODES")
+                         + "): " + ", ".join(f"{n} ({s:.3g})"
+                                             for s, n in top))
+        if self.training_logs:
+            last = self.training_logs[-1]
+            lines.append(f"training: {len(self.training_logs)} iterations, "
+                         f"final validation loss "
+                         f"{last.get('valid_loss'):.6g}")
+        ev = self.self_evaluation()
+        if ev is not None and getattr(ev, "accuracy", None) is not None:
+            lines.append(f"self evaluation (OOB/validation): "
+                         f"accuracy={ev.accuracy:.4f}")
+        if output_format == "html":
+            rows = "".join(f"<tr><td>{ln.split(':', 1)[0]}</td>"
+                           f"<td>{ln.split(':', 1)[1] if ':' in ln else ''}"
+                           "</td></tr>" for ln in lines)
+            html = f"<h2>{self._model_type}</h2><table>{rows}</table>"
+            if self.training_logs:
+                pts = [(d['iteration'], d.get('valid_loss'))
+                       for d in self.training_logs
+                       if d.get('valid_loss') is not None]
+                if pts:
+                    xs = [p[0] for p in pts]
+                    ys = [p[1] for p in pts]
+                    w, h = 480, 160
+                    ymin, ymax = min(ys), max(ys)
+                    yr = (ymax - ymin) or 1.0
+                    poly = " ".join(
+                        f"{10 + (x - xs[0]) / max(xs[-1] - xs[0], 1) * (w - 20):.1f},"
+                        f"{h - 10 - (y - ymin) / yr * (h - 20):.1f}"
+                        for x, y in pts)
+                    html += (f'<h3>validation loss</h3>'
+                             f'<svg width="{w}" height="{h}">'
+                             f'<polyline fill="none" stroke="steelblue" '
+                             f'points="{poly}"/></svg>')
+            return html
         return "\n".join(lines)
+
+    def _repr_html_(self) -> str:
+        return self.describe(output_format="html")
 
     def __str__(self) -> str:
         return self.describe()
