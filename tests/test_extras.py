@@ -1,0 +1,58 @@
+"""Multitasker, feature selection, embed codegen, calibration extras."""
+import ctypes
+import os
+import subprocess
+import tempfile
+
+import numpy as np
+import pytest
+
+import ydf_amd as ydf
+
+
+def test_multitasker(binary_data, regression_data):
+    d = dict(binary_data)
+    d["reg_label"] = (2 * d["x1"] + d["x2"]).astype(np.float32)
+    learner = ydf.MultitaskerLearner(
+        tasks=[ydf.MultitaskItem(label="label",
+                                 task=ydf.Task.CLASSIFICATION),
+               ydf.MultitaskItem(label="reg_label",
+                                 task=ydf.Task.REGRESSION)],
+        num_trees=15)
+    mm = learner.train(d)
+    preds = mm.predict(d)
+    assert set(preds) == {"label", "reg_label"}
+    evs = mm.evaluate(d)
+    assert evs["label"].accuracy > 0.85
+    assert evs["reg_label"].rmse < 1.0
+
+
+def test_backward_feature_selection(binary_data):
+    sel = ydf.BackwardSelectionFeatureSelector(objective_metric="accuracy")
+    learner = ydf.GradientBoostedTreesLearner(label="label", num_trees=15)
+    logs = sel.run(learner, binary_data, binary_data)
+    # x3 is pure noise in the fixture; x1/x2 carry the signal
+    assert "x1" in logs.selected_features
+    assert len(logs.iterations) >= 2
+
+
+def test_embed_cpp_codegen(binary_data, tmp_path):
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=10,
+                                        validation_ratio=0).train(
+                                            binary_data)
+    src = ydf.to_cpp(m, "gen")
+    cpp = tmp_path / "m.cpp"
+    cpp.write_text(src + '\nextern "C" float gen_predict_c(const float* f)'
+                   '{return gen_predict(f);}\n')
+    so = str(tmp_path / "m.so")
+    subprocess.run(["g++", "-O2", "-shared", "-fPIC", str(cpp), "-o", so],
+                   check=True)
+    lib = ctypes.CDLL(so)
+    lib.gen_predict_c.restype = ctypes.c_float
+    X = m._encode_features(binary_data)
+    ref = m.predict(binary_data, device="cpu")
+    for i in range(0, X.shape[1], 509):
+        row = np.ascontiguousarray(X[:, i])
+        p = lib.gen_predict_c(
+            row.ctypes.data_as(ctypes.POINTER(ctypes.c_float)))
+        assert abs(p - ref[i]) < 1e-5

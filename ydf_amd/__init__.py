@@ -43,6 +43,14 @@ from ydf_amd.model.export_ydf import export_ydf_model
 from_yggdrasil_model = load_ydf_model
 from ydf_amd.model import tree
 
+# Model export / serving extras
+from ydf_amd.serving.embed import to_cpp
+from ydf_amd.learner.extras import (
+    BackwardSelectionFeatureSelector,
+    MultitaskerLearner,
+    MultitaskItem,
+)
+
 # Custom losses
 from ydf_amd.learner.custom_loss import (
     Activation,
