@@ -329,13 +329,51 @@ class ForestTrainer:
             else:
                 build_map = slot_map
 
+            # row partitioning: when the level needs many LDS slot groups,
+            # sort active rows by slot once so each group pass sweeps only
+            # its own contiguous row range (vs rescanning the full table
+            # per group). Group boundaries come from local per-slot counts.
+            lds_group = max(1, (160 * 1024 - min(4 * level_size, 32768))
+                            // (ops.MAX_BINS * 16))
+            use_partition = (self.device.type == "cuda"
+                             and n_active > 2 * lds_group)
+            row_order = None
+            if use_partition:
+                # build_map (not slot_map): derived (histogram-subtraction)
+                # slots must not be built, so their rows sort to the end
+                rel_all = self.node_ids - level_base
+                keys = torch.where(
+                    (rel_all >= 0) & (rel_all < level_size),
+                    build_map[rel_all.clamp(0, level_size - 1)],
+                    torch.full((), -1, dtype=torch.int32,
+                               device=self.device))
+                keys = torch.where(keys >= 0, keys,
+                                   torch.full((), n_active,
+                                              dtype=torch.int32,
+                                              device=self.device))
+                row_order = torch.argsort(keys, stable=True).to(torch.int32)
+                slot_counts = torch.bincount(
+                    keys.long(), minlength=n_active + 1)[:n_active]
+                offs = np.zeros(n_active + 1, dtype=np.int64)
+                np.cumsum(slot_counts.cpu().numpy(), out=offs[1:])
+
             for s0 in range(0, n_active, self.max_slots):
                 ns = min(self.max_slots, n_active - s0)
                 hist_view = self.hist[:ns]
                 hist_view.zero_()
-                ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
-                               hist_view, level_base, level_size, s0, ns,
-                               grp_scratch=self.grp_buf)
+                if use_partition:
+                    for g0 in range(s0, s0 + ns, lds_group):
+                        g1 = min(g0 + lds_group, s0 + ns)
+                        ops.hist_build_gathered(
+                            self.bins, self.gh, self.node_ids, build_map,
+                            row_order, self.hist[g0 - s0:g1 - s0],
+                            level_base, level_size, g0, g1 - g0,
+                            int(offs[g0]), int(offs[g1]))
+                else:
+                    ops.hist_build(self.bins, self.gh, self.node_ids,
+                                   build_map, hist_view, level_base,
+                                   level_size, s0, ns,
+                                   grp_scratch=self.grp_buf)
                 self._allreduce(hist_view)
                 if derived and s0 == 0:
                     d_idx = torch.tensor([d[0] for d in derived],

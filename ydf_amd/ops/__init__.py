@@ -137,6 +137,21 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
         _C.cpu_split_scan(*args)
 
 
+def hist_build_gathered(bins, gh, node_ids, slot_map, row_order, hist,
+                        level_base, level_size, slot0, n_slots, row_lo,
+                        row_hi):
+    """Deep-level histogram build over a contiguous row-partitioned range
+    of `row_order` (GPU only)."""
+    F, N = bins.shape
+    n_bins = hist.shape[2]
+    _C.gpu_hist_build_gathered(
+        bins.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
+        slot_map.data_ptr(), row_order.data_ptr(), hist.data_ptr(), N, F,
+        n_bins, level_base, level_size, slot0, n_slots, row_lo, row_hi,
+        _stream())
+    return hist
+
+
 def plan_level(node_stats: torch.Tensor, prev_best_feat: torch.Tensor,
                level_base: int, level_size: int, need: int, use_sub: bool,
                build_map: torch.Tensor, derived: torch.Tensor):

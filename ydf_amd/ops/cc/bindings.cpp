@@ -23,6 +23,10 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
                     const int32_t*, float*, int64_t, int, int, int, int, int,
                     int, int, uint8_t*, void*);
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
+void gpu_hist_build_gathered(const uint8_t*, const float*, const int32_t*,
+                             const int32_t*, const int32_t*, float*, int64_t,
+                             int, int, int, int, int, int, int64_t, int64_t,
+                             void*);
 void gpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*,
                     const uint8_t*, unsigned long long*, const int8_t*,
@@ -135,6 +139,19 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          level_base, level_size, slot0, n_slots,
                          filtered_hint, P<uint8_t>(grp_scratch),
                          (void*)stream);
+        },
+        nogil);
+  m.def("gpu_hist_build_gathered",
+        [](uintptr_t bins, uintptr_t gh, uintptr_t node_ids,
+           uintptr_t slot_map, uintptr_t row_order, uintptr_t hist,
+           int64_t N, int F, int n_bins, int level_base, int level_size,
+           int slot0, int n_slots, int64_t row_lo, int64_t row_hi,
+           uintptr_t stream) {
+          gpu_hist_build_gathered(
+              P<uint8_t>(bins), P<float>(gh), P<int32_t>(node_ids),
+              P<int32_t>(slot_map), P<int32_t>(row_order), P<float>(hist),
+              N, F, n_bins, level_base, level_size, slot0, n_slots, row_lo,
+              row_hi, (void*)stream);
         },
         nogil);
   m.def("gpu_weighted_target",

@@ -396,6 +396,93 @@ __global__ void row_group_kernel(const int32_t* __restrict__ node_ids,
   }
 }
 
+// Gathered histogram build for DEEP levels (row partitioning): rows are
+// pre-sorted by slot (torch stable argsort in the trainer), so one slot
+// GROUP's rows form a contiguous range [row_lo, row_hi) of `row_order`.
+// A group pass then touches only its own rows — at 1000+ open nodes this
+// replaces ~30 full-table scans per level with one partitioned sweep
+// (gh/node_ids are L2-resident; the scattered bins gathers are the cost).
+// ---------------------------------------------------------------------------
+__global__ void hist_build_gathered_kernel(
+    const uint8_t* __restrict__ bins, const float2* __restrict__ gh,
+    const int32_t* __restrict__ node_ids,
+    const int32_t* __restrict__ slot_map,
+    const int32_t* __restrict__ row_order, float* __restrict__ hist,
+    int64_t N, int F, int n_bins, int level_base, int level_size, int slot0,
+    int n_slots, int lds_map, int64_t row_lo, int64_t row_hi,
+    int64_t rows_per_block) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double* lg = reinterpret_cast<double*>(smem);
+  unsigned long long* lp =
+      reinterpret_cast<unsigned long long*>(smem) + 1;
+  const int tot = n_slots * n_bins;
+  int* lmap = reinterpret_cast<int*>(smem + (size_t)tot * 16);
+  const int f = blockIdx.x;
+  {
+    unsigned long long* z = reinterpret_cast<unsigned long long*>(smem);
+    for (int i = threadIdx.x; i < tot * 2; i += blockDim.x) z[i] = 0ull;
+  }
+  if (lds_map) {
+    for (int i = threadIdx.x; i < level_size; i += blockDim.x)
+      lmap[i] = slot_map[i];
+  }
+  __syncthreads();
+  const uint8_t* fb = bins + (int64_t)f * N;
+  const int64_t j0 = row_lo + (int64_t)blockIdx.y * rows_per_block;
+  const int64_t j1 = min(j0 + rows_per_block, row_hi);
+  const int64_t stride = blockDim.x;
+  int64_t j = j0 + threadIdx.x;
+  const int64_t bulk_end = j1 - (kHistUnroll - 1) * stride;
+  for (; j < bulk_end; j += kHistUnroll * stride) {
+    int rows[kHistUnroll];
+    int nid[kHistUnroll];
+#pragma unroll
+    for (int u = 0; u < kHistUnroll; ++u)
+      rows[u] = row_order[j + u * stride];
+#pragma unroll
+    for (int u = 0; u < kHistUnroll; ++u) nid[u] = node_ids[rows[u]];
+#pragma unroll
+    for (int u = 0; u < kHistUnroll; ++u) {
+      const int rel = nid[u] - level_base;
+      if (rel < 0 || rel >= level_size) continue;
+      const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+      if (slot < 0 || slot >= n_slots) continue;
+      const float2 v = gh[rows[u]];
+      const int cell = 2 * (slot * n_bins + (int)fb[rows[u]]);
+      atomicAdd(lg + cell, (double)v.x);
+      const unsigned long long hq =
+          (unsigned long long)(v.y * kHScale + 0.5f);
+      atomicAdd(lp + cell,
+                hq | ((unsigned long long)(v.y != 0.f) << 44));
+    }
+  }
+  for (; j < j1; j += stride) {
+    const int row = row_order[j];
+    const int rel = node_ids[row] - level_base;
+    if (rel < 0 || rel >= level_size) continue;
+    const int slot = (lds_map ? lmap[rel] : slot_map[rel]) - slot0;
+    if (slot < 0 || slot >= n_slots) continue;
+    const float2 v = gh[row];
+    const int cell = 2 * (slot * n_bins + (int)fb[row]);
+    atomicAdd(lg + cell, (double)v.x);
+    const unsigned long long hq =
+        (unsigned long long)(v.y * kHScale + 0.5f);
+    atomicAdd(lp + cell, hq | ((unsigned long long)(v.y != 0.f) << 44));
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < tot; k += blockDim.x) {
+    const double g = lg[2 * k];
+    const unsigned long long pk = lp[2 * k];
+    if (pk == 0ull && g == 0.0) continue;
+    const int slot = k / n_bins;
+    const int bin = k - slot * n_bins;
+    float* p = hist + ((int64_t)slot * F + f) * (n_bins * 3) + bin * 3;
+    atomicAdd(p, (float)g);
+    atomicAdd(p + 1, (float)((double)(pk & kHMask) * (double)kHInvScale));
+    atomicAdd(p + 2, (float)(pk >> 44));
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Split scan, stage A: one block per (slot, feature). Each thread owns one
 // bin; inclusive prefix sums of {g,h,c} over bins via LDS Hillis-Steele
@@ -1013,6 +1100,32 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
                          rpb);
     }
   }
+}
+
+void gpu_hist_build_gathered(const uint8_t* bins, const float* gh,
+                             const int32_t* node_ids,
+                             const int32_t* slot_map,
+                             const int32_t* row_order, float* hist,
+                             int64_t N, int F, int n_bins, int level_base,
+                             int level_size, int slot0, int n_slots,
+                             int64_t row_lo, int64_t row_hi, void* stream) {
+  const size_t map_bytes_full = (size_t)level_size * sizeof(int32_t);
+  const int lds_map = map_bytes_full <= 32 * 1024 ? 1 : 0;
+  const size_t lds = (size_t)n_slots * n_bins * 16 +
+                     (lds_map ? map_bytes_full : 0);
+  const int64_t rows = row_hi - row_lo;
+  if (rows <= 0) return;
+  int chunks = (int)((2048 + F - 1) / F);
+  const int64_t min_rows = 512;
+  if (rows / chunks < min_rows)
+    chunks = (int)((rows + min_rows - 1) / min_rows);
+  if (chunks < 1) chunks = 1;
+  const int64_t rpb = (rows + chunks - 1) / chunks;
+  hipLaunchKernelGGL(hist_build_gathered_kernel, dim3(F, chunks),
+                     dim3(kBlock), lds, (hipStream_t)stream, bins,
+                     (const float2*)gh, node_ids, slot_map, row_order, hist,
+                     N, F, n_bins, level_base, level_size, slot0, n_slots,
+                     lds_map, row_lo, row_hi, rpb);
 }
 
 void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
