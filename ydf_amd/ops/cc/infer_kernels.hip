@@ -46,8 +46,34 @@ __global__ void predict_forest_lds_kernel(
   if (tid >= n_here) return;
   float acc = init;
   if (cat_idx == nullptr) {
-    // pure-numerical fast path: tight compare-and-descend loop
-    for (int tt = 0; tt < n_trees; ++tt) {
+    // pure-numerical fast path. Four trees walk in parallel per thread:
+    // a single walk is a chain of DEPENDENT L2 node loads (~200 cy each),
+    // so interleaving 4 independent chains hides most of that latency.
+    int tt = 0;
+    for (; tt + 4 <= n_trees; tt += 4) {
+      int n[4], fx[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        n[u] = roots[tree_start + (int64_t)(tt + u) * tree_step];
+        fx[u] = feat[n[u]];
+      }
+      bool done = false;
+      while (!done) {
+        done = true;
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (fx[u] >= 0) {
+            n[u] = left[n[u]] + (xs[fx[u] * kTile + tid] > thr[n[u]] ? 1
+                                                                     : 0);
+            fx[u] = feat[n[u]];
+            done &= fx[u] < 0;
+          }
+        }
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) acc += thr[n[u]];
+    }
+    for (; tt < n_trees; ++tt) {
       int n = roots[tree_start + (int64_t)tt * tree_step];
       int f = feat[n];
       while (f >= 0) {
