@@ -50,6 +50,12 @@ class _DeviceForest:
         else:
             self.cat_idx = None
             self.masks = None
+        self.packed = None
+        if device.type == "cuda":
+            ci = self.cat_idx if self.cat_idx is not None else torch.full(
+                (self.feat.numel(),), -1, dtype=torch.int32, device=device)
+            self.packed = ops.pack_forest_nodes(self.feat, self.thr,
+                                                self.left, ci)
 
 
 class GenericModel:
@@ -158,7 +164,8 @@ class GenericModel:
                                           if c < len(self.init_predictions)
                                           else self.init_predictions[0]),
                                scale=self._leaf_scale(),
-                               cat_idx=df.cat_idx, masks=df.masks)
+                               cat_idx=df.cat_idx, masks=df.masks,
+                               packed=df.packed)
         return out
 
     def predict(self, data, device=None) -> np.ndarray:

@@ -232,23 +232,43 @@ def binary_logloss(preds: torch.Tensor, labels: torch.Tensor,
     return out2
 
 
+def pack_forest_nodes(feat: torch.Tensor, thr: torch.Tensor,
+                      left: torch.Tensor, cat_idx: torch.Tensor
+                      ) -> torch.Tensor:
+    """Interleaves node arrays into 16-B {feat, thr, left, cat_idx}
+    structs (one gather per visit on GPU)."""
+    packed = torch.empty((feat.numel(), 4), dtype=torch.int32,
+                         device=feat.device)
+    packed[:, 0] = feat
+    packed[:, 1] = thr.view(torch.int32)
+    packed[:, 2] = left
+    packed[:, 3] = cat_idx
+    return packed
+
+
 def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
                    left: torch.Tensor, roots: torch.Tensor, out: torch.Tensor,
                    tree_start: int = 0, tree_step: int = 1,
                    n_trees: int = -1, init: float = 0.0, scale: float = 1.0,
-                   cat_idx=None, masks=None):
-    """Flat-forest batch inference. X [F,N] f32, out [N] f32."""
+                   cat_idx=None, masks=None, packed=None):
+    """Flat-forest batch inference. X [F,N] f32, out [N] f32. On GPU,
+    pass `packed` (pack_forest_nodes output) for the fast path."""
     F, N = X.shape
     if n_trees < 0:
         n_trees = roots.numel()
-    ci = cat_idx.data_ptr() if cat_idx is not None else 0
     mk = masks.data_ptr() if masks is not None else 0
     if X.is_cuda:
-        _C.gpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
-                              thr.data_ptr(), left.data_ptr(),
-                              roots.data_ptr(), ci, mk, tree_start, tree_step,
-                              n_trees, out.data_ptr(), init, scale, _stream())
+        if packed is None:
+            ci = cat_idx if cat_idx is not None else torch.full(
+                (feat.numel(),), -1, dtype=torch.int32, device=X.device)
+            packed = pack_forest_nodes(feat, thr, left, ci)
+        _C.gpu_predict_forest(X.data_ptr(), N, F, packed.data_ptr(),
+                              roots.data_ptr(), mk,
+                              1 if cat_idx is not None else 0, tree_start,
+                              tree_step, n_trees, out.data_ptr(), init,
+                              scale, _stream())
     else:
+        ci = cat_idx.data_ptr() if cat_idx is not None else 0
         _C.cpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
                               thr.data_ptr(), left.data_ptr(),
                               roots.data_ptr(), ci, mk, tree_start, tree_step,

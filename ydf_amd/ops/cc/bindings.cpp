@@ -45,9 +45,8 @@ void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
 void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
 // infer_kernels.hip
 void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
-                        const float*, const int32_t*, const int32_t*,
                         const int32_t*, const unsigned long long*, int, int,
-                        int, float*, float, float, void*);
+                        int, int, float*, float, float, void*);
 void gpu_sigmoid(const float*, float*, int64_t, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int);
@@ -235,16 +234,15 @@ PYBIND11_MODULE(_ydf_ops, m) {
         },
         nogil);
   m.def("gpu_predict_forest",
-        [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
-           uintptr_t left, uintptr_t roots, uintptr_t cat_idx,
-           uintptr_t masks, int tree_start, int tree_step, int n_trees,
-           uintptr_t out, float init, float scale, uintptr_t stream) {
-          gpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
-                             P<float>(thr), P<int32_t>(left),
-                             P<int32_t>(roots), P<int32_t>(cat_idx),
-                             P<unsigned long long>(masks), tree_start,
-                             tree_step, n_trees, P<float>(out), init, scale,
-                             (void*)stream);
+        [](uintptr_t X, int64_t N, int F, uintptr_t packed_nodes,
+           uintptr_t roots, uintptr_t masks, int has_cats, int tree_start,
+           int tree_step, int n_trees, uintptr_t out, float init,
+           float scale, uintptr_t stream) {
+          gpu_predict_forest(P<float>(X), N, F, P<int32_t>(packed_nodes),
+                             P<int32_t>(roots),
+                             P<unsigned long long>(masks), has_cats,
+                             tree_start, tree_step, n_trees, P<float>(out),
+                             init, scale, (void*)stream);
         },
         nogil);
   m.def("gpu_sigmoid",

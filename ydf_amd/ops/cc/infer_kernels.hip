@@ -22,17 +22,26 @@ namespace ydfa {
 
 constexpr int kTile = 256;  // examples per block
 
+// Packed node: one 16-B gather per visit instead of three scattered loads
+// (feat/thr/left in separate arrays tripled the divergent-gather count,
+// which is what bounds this kernel).
+struct PackedNode {
+  int32_t feat;     // -1 = leaf
+  float thr;        // threshold or leaf value
+  int32_t left;     // left child (right = left + 1)
+  int32_t cat_idx;  // -1 = numerical; else mask index
+};
+
 // Example tile staged in LDS: xs[f * kTile + tid]. The f-stride is a
 // multiple of 32 banks, so per-lane-group accesses with distinct tid never
 // conflict regardless of the (divergent) feature index.
 __global__ void predict_forest_lds_kernel(
     const float* __restrict__ X, int64_t N, int F,
-    const int32_t* __restrict__ feat, const float* __restrict__ thr,
-    const int32_t* __restrict__ left, const int32_t* __restrict__ roots,
-    const int32_t* __restrict__ cat_idx,
-    const unsigned long long* __restrict__ masks, int tree_start,
-    int tree_step, int n_trees, float* __restrict__ out, float init,
-    float scale) {
+    const PackedNode* __restrict__ nodes,
+    const int32_t* __restrict__ roots,
+    const unsigned long long* __restrict__ masks, int has_cats,
+    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
+    float init, float scale) {
   extern __shared__ float xs[];  // [F][kTile]
   const int64_t base = (int64_t)blockIdx.x * kTile;
   const int tid = threadIdx.x;
@@ -45,63 +54,55 @@ __global__ void predict_forest_lds_kernel(
   __syncthreads();
   if (tid >= n_here) return;
   float acc = init;
-  if (cat_idx == nullptr) {
-    // pure-numerical fast path. Four trees walk in parallel per thread:
-    // a single walk is a chain of DEPENDENT L2 node loads (~200 cy each),
-    // so interleaving 4 independent chains hides most of that latency.
+  if (!has_cats) {
+    // pure-numerical fast path: 4 trees walk in parallel per thread (a
+    // single walk is a chain of DEPENDENT L2 gathers), one 16-B packed
+    // node load per step.
     int tt = 0;
     for (; tt + 4 <= n_trees; tt += 4) {
-      int n[4], fx[4];
+      PackedNode nd[4];
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        n[u] = roots[tree_start + (int64_t)(tt + u) * tree_step];
-        fx[u] = feat[n[u]];
-      }
+      for (int u = 0; u < 4; ++u)
+        nd[u] = nodes[roots[tree_start + (int64_t)(tt + u) * tree_step]];
       bool done = false;
       while (!done) {
         done = true;
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
-          if (fx[u] >= 0) {
-            n[u] = left[n[u]] + (xs[fx[u] * kTile + tid] > thr[n[u]] ? 1
-                                                                     : 0);
-            fx[u] = feat[n[u]];
-            done &= fx[u] < 0;
+          if (nd[u].feat >= 0) {
+            const int nx = nd[u].left +
+                (xs[nd[u].feat * kTile + tid] > nd[u].thr ? 1 : 0);
+            nd[u] = nodes[nx];
+            done &= nd[u].feat < 0;
           }
         }
       }
 #pragma unroll
-      for (int u = 0; u < 4; ++u) acc += thr[n[u]];
+      for (int u = 0; u < 4; ++u) acc += nd[u].thr;
     }
     for (; tt < n_trees; ++tt) {
-      int n = roots[tree_start + (int64_t)tt * tree_step];
-      int f = feat[n];
-      while (f >= 0) {
-        n = left[n] + (xs[f * kTile + tid] > thr[n] ? 1 : 0);
-        f = feat[n];
-      }
-      acc += thr[n];
+      PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
+      while (nd.feat >= 0)
+        nd = nodes[nd.left + (xs[nd.feat * kTile + tid] > nd.thr ? 1 : 0)];
+      acc += nd.thr;
     }
   } else {
     for (int tt = 0; tt < n_trees; ++tt) {
-      int n = roots[tree_start + (int64_t)tt * tree_step];
-      int f = feat[n];
-      while (f >= 0) {
-        const float xv = xs[f * kTile + tid];
+      PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
+      while (nd.feat >= 0) {
+        const float xv = xs[nd.feat * kTile + tid];
         int right;
-        const int ci = cat_idx[n];
-        if (ci >= 0) {
+        if (nd.cat_idx >= 0) {
           int c = (int)xv;
           c = c < 0 ? 0 : (c > 255 ? 255 : c);
-          right =
-              (int)((masks[(int64_t)ci * 4 + (c >> 6)] >> (c & 63)) & 1ull);
+          right = (int)((masks[(int64_t)nd.cat_idx * 4 + (c >> 6)]
+                         >> (c & 63)) & 1ull);
         } else {
-          right = xv > thr[n] ? 1 : 0;
+          right = xv > nd.thr ? 1 : 0;
         }
-        n = left[n] + right;
-        f = feat[n];
+        nd = nodes[nd.left + right];
       }
-      acc += thr[n];
+      acc += nd.thr;
     }
   }
   out[base + tid] = init + (acc - init) * scale;
@@ -110,47 +111,31 @@ __global__ void predict_forest_lds_kernel(
 // Fallback without the LDS tile (feature count too large to stage).
 __global__ void predict_forest_global_kernel(
     const float* __restrict__ X, int64_t N, int F,
-    const int32_t* __restrict__ feat, const float* __restrict__ thr,
-    const int32_t* __restrict__ left, const int32_t* __restrict__ roots,
-    const int32_t* __restrict__ cat_idx,
-    const unsigned long long* __restrict__ masks, int tree_start,
-    int tree_step, int n_trees, float* __restrict__ out, float init,
-    float scale) {
+    const PackedNode* __restrict__ nodes,
+    const int32_t* __restrict__ roots,
+    const unsigned long long* __restrict__ masks, int has_cats,
+    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
+    float init, float scale) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t k = i; k < N; k += stride) {
     float acc = init;
-    if (cat_idx == nullptr) {
-      for (int tt = 0; tt < n_trees; ++tt) {
-        int n = roots[tree_start + (int64_t)tt * tree_step];
-        int f = feat[n];
-        while (f >= 0) {
-          n = left[n] + (X[(int64_t)f * N + k] > thr[n] ? 1 : 0);
-          f = feat[n];
+    for (int tt = 0; tt < n_trees; ++tt) {
+      PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
+      while (nd.feat >= 0) {
+        const float xv = X[(int64_t)nd.feat * N + k];
+        int right;
+        if (has_cats && nd.cat_idx >= 0) {
+          int c = (int)xv;
+          c = c < 0 ? 0 : (c > 255 ? 255 : c);
+          right = (int)((masks[(int64_t)nd.cat_idx * 4 + (c >> 6)]
+                         >> (c & 63)) & 1ull);
+        } else {
+          right = xv > nd.thr ? 1 : 0;
         }
-        acc += thr[n];
+        nd = nodes[nd.left + right];
       }
-    } else {
-      for (int tt = 0; tt < n_trees; ++tt) {
-        int n = roots[tree_start + (int64_t)tt * tree_step];
-        int f = feat[n];
-        while (f >= 0) {
-          const float xv = X[(int64_t)f * N + k];
-          int right;
-          const int ci = cat_idx[n];
-          if (ci >= 0) {
-            int c = (int)xv;
-            c = c < 0 ? 0 : (c > 255 ? 255 : c);
-            right = (int)((masks[(int64_t)ci * 4 + (c >> 6)]
-                           >> (c & 63)) & 1ull);
-          } else {
-            right = xv > thr[n] ? 1 : 0;
-          }
-          n = left[n] + right;
-          f = feat[n];
-        }
-        acc += thr[n];
-      }
+      acc += nd.thr;
     }
     out[k] = init + (acc - init) * scale;
   }
@@ -166,27 +151,28 @@ __global__ void sigmoid_kernel(const float* __restrict__ in,
 
 extern "C" {
 
-void gpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
-                        const float* thr, const int32_t* left,
-                        const int32_t* roots, const int32_t* cat_idx,
-                        const unsigned long long* masks, int tree_start,
-                        int tree_step, int n_trees, float* out, float init,
-                        float scale, void* stream) {
+void gpu_predict_forest(const float* X, int64_t N, int F,
+                        const int32_t* packed_nodes, const int32_t* roots,
+                        const unsigned long long* masks, int has_cats,
+                        int tree_start, int tree_step, int n_trees,
+                        float* out, float init, float scale, void* stream) {
+  const PackedNode* nodes =
+      reinterpret_cast<const PackedNode*>(packed_nodes);
   const size_t lds = (size_t)F * kTile * sizeof(float);
   if (lds <= 96 * 1024) {
     const int grid = (int)((N + kTile - 1) / kTile);
     hipLaunchKernelGGL(predict_forest_lds_kernel, dim3(grid), dim3(kTile), lds,
-                       (hipStream_t)stream, X, N, F, feat, thr, left, roots,
-                       cat_idx, masks, tree_start, tree_step, n_trees, out,
-                       init, scale);
+                       (hipStream_t)stream, X, N, F, nodes, roots, masks,
+                       has_cats, tree_start, tree_step, n_trees, out, init,
+                       scale);
   } else {
     int grid = (int)((N + kTile - 1) / kTile);
     if (grid > 4096) grid = 4096;
     if (grid < 1) grid = 1;
     hipLaunchKernelGGL(predict_forest_global_kernel, dim3(grid), dim3(kTile),
-                       0, (hipStream_t)stream, X, N, F, feat, thr, left, roots,
-                       cat_idx, masks, tree_start, tree_step, n_trees, out,
-                       init, scale);
+                       0, (hipStream_t)stream, X, N, F, nodes, roots, masks,
+                       has_cats, tree_start, tree_step, n_trees, out, init,
+                       scale);
   }
 }
 
