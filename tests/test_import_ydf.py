@@ -102,3 +102,57 @@ def test_export_roundtrip_categorical(tmp_path):
     m2 = ydf.load_ydf_model(p)
     np.testing.assert_allclose(m.predict(d, device="cpu"),
                                m2.predict(d, device="cpu"), atol=2e-6)
+
+
+def test_oblique_gbt_import_quality(tmp_path):
+    """Oblique (sparse linear projection) conditions: the reference's
+    adult_binary_class_gbdt_oblique model must load and keep its quality
+    (no golden prediction file exists for this model)."""
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/adult_binary_class_gbdt_oblique")
+    f = m.forest
+    assert (f.cat_idx <= -2).sum() > 0
+    assert len(f.obl_ranges) == (f.cat_idx <= -2).sum()
+    te = pd.read_csv(f"{BASE}/dataset/adult_test.csv")
+    ev = m.evaluate(te, device="cpu")
+    assert ev.accuracy > 0.86
+    assert ev.auc > 0.92
+    # save/load round-trip preserves oblique predictions exactly
+    p1 = m.predict(te, device="cpu")
+    m.save(str(tmp_path / "obl"))
+    m2 = ydf.load_model(str(tmp_path / "obl"))
+    np.testing.assert_array_equal(p1, m2.predict(te, device="cpu"))
+    # serialize/deserialize too
+    m3 = ydf.deserialize_model(ydf.serialize_model(m))
+    np.testing.assert_array_equal(p1, m3.predict(te, device="cpu"))
+
+
+def test_oblique_embed_cpp(tmp_path):
+    """C++ codegen of an oblique model compiles and matches predictions."""
+    import subprocess
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/adult_binary_class_gbdt_oblique")
+    te = pd.read_csv(f"{BASE}/dataset/adult_test.csv").head(200)
+    src = ydf.to_cpp(m, "obl")
+    X = m._encode_features(te)
+    F, N = X.shape
+    main = (
+        "#include <cstdio>\n" + src +
+        "int main() { static float x[%d];\n" % F +
+        "  FILE* f = fopen(\"x.bin\", \"rb\");\n"
+        "  for (int i = 0; i < %d; ++i) {\n" % N +
+        "    if (fread(x, sizeof(float), %d, f) != %d) return 1;\n"
+        % (F, F) +
+        "    printf(\"%.9g\\n\", obl_predict(x));\n"
+        "  }\n  return 0;\n}\n")
+    cpp = tmp_path / "m.cpp"
+    cpp.write_text(main)
+    exe = tmp_path / "m"
+    subprocess.run(["g++", "-O1", "-o", str(exe), str(cpp)], check=True)
+    (tmp_path / "x.bin").write_bytes(
+        np.ascontiguousarray(X.T, dtype=np.float32).tobytes())
+    out = subprocess.run([str(exe)], cwd=tmp_path, capture_output=True,
+                         text=True, check=True)
+    got = np.array([float(v) for v in out.stdout.split()])
+    want = m.predict(te, device="cpu")
+    np.testing.assert_allclose(got, want, atol=2e-5)

@@ -30,6 +30,13 @@ class FlatForest:
     masks: np.ndarray = None     # u64 [n_masks, 4]
     # training cover (weighted example count) per node; used by TreeSHAP
     cover: np.ndarray = None     # f32 [total]
+    # oblique (sparse linear) conditions: a node with cat_idx <= -2 uses
+    # oblique record -(cat_idx+2): sum_k obl_w[s+k]*x[obl_attr[s+k]] > thr
+    # with (s, n) = obl_ranges[record] (reference Condition.Oblique,
+    # decision_tree.proto:114-131)
+    obl_ranges: np.ndarray = None  # i32 [n_obl, 2] (start, count)
+    obl_attr: np.ndarray = None    # i32 [total_terms]
+    obl_w: np.ndarray = None       # f32 [total_terms]
 
     def __post_init__(self):
         if self.cat_idx is None:
@@ -38,10 +45,16 @@ class FlatForest:
             self.masks = np.zeros((0, 4), dtype=np.uint64)
         if self.cover is None:
             self.cover = np.zeros(len(self.feat), dtype=np.float32)
+        if self.obl_ranges is None:
+            self.obl_ranges = np.zeros((0, 2), dtype=np.int32)
+        if self.obl_attr is None:
+            self.obl_attr = np.zeros(0, dtype=np.int32)
+        if self.obl_w is None:
+            self.obl_w = np.zeros(0, dtype=np.float32)
 
     @property
     def has_cats(self) -> bool:
-        return len(self.masks) > 0
+        return len(self.masks) > 0 or len(self.obl_ranges) > 0
 
     @property
     def n_trees(self) -> int:
@@ -170,17 +183,26 @@ def concat_forests(a: FlatForest, b: FlatForest) -> FlatForest:
     model + newly grown trees)."""
     off = a.n_nodes
     moff = len(a.masks)
+    ooff = len(a.obl_ranges)
+    b_cat = np.where(b.cat_idx >= 0, b.cat_idx + moff, b.cat_idx)
+    # oblique markers (<= -2) shift by the number of a's oblique records
+    b_cat = np.where(b_cat <= -2, b_cat - ooff, b_cat)
+    b_rng = b.obl_ranges.copy()
+    if len(b_rng):
+        b_rng[:, 0] += len(a.obl_attr)
     return FlatForest(
         feat=np.concatenate([a.feat, b.feat]),
         thr=np.concatenate([a.thr, b.thr]),
         left=np.concatenate([a.left,
                              np.where(b.feat >= 0, b.left + off, 0)]),
         roots=np.concatenate([a.roots, b.roots + off]).astype(np.int32),
-        cat_idx=np.concatenate([a.cat_idx,
-                                np.where(b.cat_idx >= 0, b.cat_idx + moff,
-                                         -1)]).astype(np.int32),
+        cat_idx=np.concatenate([a.cat_idx, b_cat]).astype(np.int32),
         masks=np.concatenate([a.masks, b.masks]) if (len(a.masks)
                                                      or len(b.masks))
         else np.zeros((0, 4), np.uint64),
         cover=np.concatenate([a.cover, b.cover]),
+        obl_ranges=np.concatenate([a.obl_ranges, b_rng]).astype(np.int32)
+        if (ooff or len(b_rng)) else None,
+        obl_attr=np.concatenate([a.obl_attr, b.obl_attr]).astype(np.int32),
+        obl_w=np.concatenate([a.obl_w, b.obl_w]).astype(np.float32),
     )

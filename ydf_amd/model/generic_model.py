@@ -50,6 +50,13 @@ class _DeviceForest:
         else:
             self.cat_idx = None
             self.masks = None
+        if len(forest.obl_ranges):
+            self.obl_ranges = torch.from_numpy(
+                np.ascontiguousarray(forest.obl_ranges)).to(device)
+            self.obl_attr = torch.from_numpy(forest.obl_attr).to(device)
+            self.obl_w = torch.from_numpy(forest.obl_w).to(device)
+        else:
+            self.obl_ranges = self.obl_attr = self.obl_w = None
         self.packed = None
         if device.type == "cuda":
             ci = self.cat_idx if self.cat_idx is not None else torch.full(
@@ -165,7 +172,8 @@ class GenericModel:
                                           else self.init_predictions[0]),
                                scale=self._leaf_scale(),
                                cat_idx=df.cat_idx, masks=df.masks,
-                               packed=df.packed)
+                               packed=df.packed, obl_ranges=df.obl_ranges,
+                               obl_attr=df.obl_attr, obl_w=df.obl_w)
         return out
 
     def predict(self, data, device=None) -> np.ndarray:
@@ -252,10 +260,15 @@ class GenericModel:
         cat_idx = np.ascontiguousarray(f.cat_idx)
         masks = np.ascontiguousarray(f.masks)
         scale = self._leaf_scale()
+        obl_ranges = np.ascontiguousarray(f.obl_ranges)
+        obl_attr = np.ascontiguousarray(f.obl_attr)
+        obl_w = np.ascontiguousarray(f.obl_w)
         cpu_tree_shap(X.ctypes.data, N, F, feat.ctypes.data, thr.ctypes.data,
                       left.ctypes.data,
                       cat_idx.ctypes.data if f.has_cats else 0,
                       masks.ctypes.data if f.has_cats else 0,
+                      obl_ranges.ctypes.data, obl_attr.ctypes.data,
+                      obl_w.ctypes.data,
                       cover.ctypes.data, roots.ctypes.data, 0, 1,
                       f.n_trees, scale, 0.0, phi.ctypes.data)
         ev = cpu_forest_expected_value(
@@ -445,7 +458,9 @@ class GenericModel:
         np.savez(os.path.join(path, "forest.npz"), feat=self.forest.feat,
                  thr=self.forest.thr, left=self.forest.left,
                  roots=self.forest.roots, cat_idx=self.forest.cat_idx,
-                 masks=self.forest.masks, cover=self.forest.cover)
+                 masks=self.forest.masks, cover=self.forest.cover,
+                 obl_ranges=self.forest.obl_ranges,
+                 obl_attr=self.forest.obl_attr, obl_w=self.forest.obl_w)
         with open(os.path.join(path, "done"), "w") as f:
             f.write("")
 

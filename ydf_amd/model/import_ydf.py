@@ -8,8 +8,8 @@ numbers below are documented against the reference .proto sources.
 
 Supported: GBT (binary/multiclass/regression) and RF models with numerical
 (Higher/DiscretizedHigher), categorical (ContainsBitmap/ContainsVector) and
-boolean (TrueValue) conditions. Oblique and vector-sequence conditions are
-not supported (ROADMAP).
+boolean (TrueValue) and oblique (sparse linear projection) conditions.
+Vector-sequence conditions are not supported (ROADMAP).
 """
 from __future__ import annotations
 
@@ -188,7 +188,7 @@ def parse_data_spec(raw: bytes):
 # ---------------------------------------------------------------------------
 class _NodeRec:
     __slots__ = ("is_leaf", "attr", "thr", "mask", "value", "cover",
-                 "na_value")
+                 "na_value", "obl")
 
 
 def parse_node(raw: bytes, disc_bounds, n_classes: int,
@@ -202,6 +202,7 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
     r.value = 0.0
     r.cover = 0.0
     r.na_value = False
+    r.obl = None
     if 5 in node:
         pass
     if 1 in node:  # classifier output
@@ -253,6 +254,16 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
                 if e < 256:
                     m[e >> 6] |= np.uint64(1 << (e & 63))
             r.mask = m
+        elif 7 in inner:  # Oblique: sum_i w_i * x[a_i] >= threshold
+            # (decision_tree.proto:114-131: attributes=1 packed,
+            # weights=2 packed f32, threshold=3)
+            ob = _msg(inner[7][0])
+            attrs = _packed_varints(ob[1][0]) if ob.get(1) else []
+            ws = _packed_floats(ob[2][0]) if ob.get(2) else []
+            t = _f32(ob.get(3, [0.0])[0]) if 3 in ob else 0.0
+            r.obl = (attrs, ws)
+            # our kernels test strict >; dot >= t  <=>  dot > nextafter down
+            r.thr = float(np.nextafter(np.float32(t), np.float32("-inf")))
         elif 5 in inner:  # ContainsBitmap
             bm = _msg(inner[5][0]).get(1, [b""])[0]
             m = np.zeros(32, dtype=np.uint8)
@@ -276,6 +287,7 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         records.extend(read_blob_sequence(os.path.join(model_dir, s)))
     feats, thrs, lefts, roots, cidx, masks, covers = \
         [], [], [], [], [], [], []
+    obl_ranges, obl_attr, obl_w = [], [], []
     pos = 0
 
     def new_slot():
@@ -307,6 +319,14 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         if rec.mask is not None:
             cidx[idx] = len(masks)
             masks.append(rec.mask)
+        elif rec.obl is not None:
+            attrs, ws = rec.obl
+            feats[idx] = attrs[0] if attrs else 0
+            cidx[idx] = -(2 + len(obl_ranges))
+            obl_ranges.append((len(obl_attr), len(attrs)))
+            obl_attr.extend(attrs)
+            obl_w.extend(ws)
+            thrs[idx] = rec.thr
         else:
             thrs[idx] = rec.thr
         li = new_slot()
@@ -325,7 +345,10 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         cat_idx=np.asarray(cidx, np.int32),
         masks=np.stack(masks).astype(np.uint64) if masks
         else np.zeros((0, 4), np.uint64),
-        cover=np.asarray(covers, np.float32))
+        cover=np.asarray(covers, np.float32),
+        obl_ranges=np.asarray(obl_ranges, np.int32).reshape(-1, 2),
+        obl_attr=np.asarray(obl_attr, np.int32),
+        obl_w=np.asarray(obl_w, np.float32))
 
 
 def load_ydf_model(path: str, file_prefix: str = ""):
@@ -412,4 +435,9 @@ def _remap_forest(forest: FlatForest, remap: Dict[int, int]) -> FlatForest:
         if feat[i] >= 0:
             feat[i] = remap.get(int(feat[i]), 0)
     forest.feat = feat
+    if len(forest.obl_attr):
+        oa = forest.obl_attr.copy()
+        for i in range(len(oa)):
+            oa[i] = remap.get(int(oa[i]), 0)
+        forest.obl_attr = oa
     return forest

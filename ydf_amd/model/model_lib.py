@@ -24,7 +24,11 @@ def load_model(path: str) -> GenericModel:
                         roots=z["roots"],
                         cat_idx=z["cat_idx"] if "cat_idx" in z else None,
                         masks=z["masks"] if "masks" in z else None,
-                        cover=z["cover"] if "cover" in z else None)
+                        cover=z["cover"] if "cover" in z else None,
+                        obl_ranges=z["obl_ranges"] if "obl_ranges" in z
+                        else None,
+                        obl_attr=z["obl_attr"] if "obl_attr" in z else None,
+                        obl_w=z["obl_w"] if "obl_w" in z else None)
     cls = MODEL_CLASSES.get(header["model_type"], GenericModel)
     model = cls(
         forest=forest,
@@ -53,7 +57,9 @@ def serialize_model(model: GenericModel) -> bytes:
         np.savez(fbuf, feat=model.forest.feat, thr=model.forest.thr,
                  left=model.forest.left, roots=model.forest.roots,
                  cat_idx=model.forest.cat_idx, masks=model.forest.masks,
-                 cover=model.forest.cover)
+                 cover=model.forest.cover,
+                 obl_ranges=model.forest.obl_ranges,
+                 obl_attr=model.forest.obl_attr, obl_w=model.forest.obl_w)
         zf.writestr("forest.npz", fbuf.getvalue())
     return buf.getvalue()
 

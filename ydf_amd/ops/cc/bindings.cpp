@@ -45,8 +45,9 @@ void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
 void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
 // infer_kernels.hip
 void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
-                        const int32_t*, const unsigned long long*, int, int,
-                        int, int, float*, float, float, void*);
+                        const int32_t*, const unsigned long long*,
+                        const int32_t*, const int32_t*, const float*, int,
+                        int, int, int, float*, float, float, void*);
 void gpu_sigmoid(const float*, float*, int64_t, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int);
@@ -72,10 +73,12 @@ void cpu_update_preds(float*, const int32_t*, const float*, int64_t, float);
 void cpu_binary_logloss(const float*, const float*, float*, int64_t);
 void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const float*, const int32_t*, const int32_t*,
-                        const int32_t*, const unsigned long long*, int, int,
-                        int, float*, float, float);
+                        const int32_t*, const unsigned long long*,
+                        const int32_t*, const int32_t*, const float*, int,
+                        int, int, float*, float, float);
 void cpu_tree_shap(const float*, int64_t, int, const int32_t*, const float*,
                    const int32_t*, const int32_t*, const unsigned long long*,
+                   const int32_t*, const int32_t*, const float*,
                    const float*, const int32_t*, int, int, int, float, float,
                    float*);
 void cpu_forest_expected_value(const int32_t*, const float*, const int32_t*,
@@ -235,12 +238,15 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("gpu_predict_forest",
         [](uintptr_t X, int64_t N, int F, uintptr_t packed_nodes,
-           uintptr_t roots, uintptr_t masks, int has_cats, int tree_start,
+           uintptr_t roots, uintptr_t masks, uintptr_t obl_ranges,
+           uintptr_t obl_attr, uintptr_t obl_w, int has_cats, int tree_start,
            int tree_step, int n_trees, uintptr_t out, float init,
            float scale, uintptr_t stream) {
           gpu_predict_forest(P<float>(X), N, F, P<int32_t>(packed_nodes),
                              P<int32_t>(roots),
-                             P<unsigned long long>(masks), has_cats,
+                             P<unsigned long long>(masks),
+                             P<int32_t>(obl_ranges), P<int32_t>(obl_attr),
+                             P<float>(obl_w), has_cats,
                              tree_start, tree_step, n_trees, P<float>(out),
                              init, scale, (void*)stream);
         },
@@ -357,11 +363,14 @@ PYBIND11_MODULE(_ydf_ops, m) {
   m.def("cpu_tree_shap",
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
            uintptr_t left, uintptr_t cat_idx, uintptr_t masks,
+           uintptr_t obl_ranges, uintptr_t obl_attr, uintptr_t obl_w,
            uintptr_t cover, uintptr_t roots, int tree_start, int tree_step,
            int n_trees, float scale, float init, uintptr_t phi_out) {
           cpu_tree_shap(P<float>(X), N, F, P<int32_t>(feat), P<float>(thr),
                         P<int32_t>(left), P<int32_t>(cat_idx),
-                        P<unsigned long long>(masks), P<float>(cover),
+                        P<unsigned long long>(masks), P<int32_t>(obl_ranges),
+                        P<int32_t>(obl_attr), P<float>(obl_w),
+                        P<float>(cover),
                         P<int32_t>(roots), tree_start, tree_step, n_trees,
                         scale, init, P<float>(phi_out));
         },
@@ -380,12 +389,15 @@ PYBIND11_MODULE(_ydf_ops, m) {
   m.def("cpu_predict_forest",
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
            uintptr_t left, uintptr_t roots, uintptr_t cat_idx,
-           uintptr_t masks, int tree_start, int tree_step, int n_trees,
+           uintptr_t masks, uintptr_t obl_ranges, uintptr_t obl_attr,
+           uintptr_t obl_w, int tree_start, int tree_step, int n_trees,
            uintptr_t out, float init, float scale) {
           cpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
                              P<float>(thr), P<int32_t>(left),
                              P<int32_t>(roots), P<int32_t>(cat_idx),
-                             P<unsigned long long>(masks), tree_start,
+                             P<unsigned long long>(masks),
+                             P<int32_t>(obl_ranges), P<int32_t>(obl_attr),
+                             P<float>(obl_w), tree_start,
                              tree_step, n_trees, P<float>(out), init, scale);
         },
         nogil);

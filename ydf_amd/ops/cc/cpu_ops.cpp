@@ -457,7 +457,9 @@ void cpu_binary_logloss(const float* preds, const float* labels, float* out2,
 void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
                         const float* thr, const int32_t* left,
                         const int32_t* roots, const int32_t* cat_idx,
-                        const unsigned long long* masks, int tree_start,
+                        const unsigned long long* masks,
+                        const int32_t* obl_ranges, const int32_t* obl_attr,
+                        const float* obl_w, int tree_start,
                         int tree_step, int n_trees, float* out, float init,
                         float scale) {
   (void)F;
@@ -472,16 +474,22 @@ void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
         int n = roots[tree_start + (int64_t)tt * tree_step];
         int f = feat[n];
         while (f >= 0) {
-          const float xv = X[(int64_t)f * N + k];
           int right;
           const int ci = cat_idx ? cat_idx[n] : -1;
           if (ci >= 0) {
-            int cbin = (int)xv;
+            int cbin = (int)X[(int64_t)f * N + k];
             cbin = cbin < 0 ? 0 : (cbin > 255 ? 255 : cbin);
             right = (int)((masks[(int64_t)ci * 4 + (cbin >> 6)]
                            >> (cbin & 63)) & 1ull);
+          } else if (ci <= -2) {  // oblique sparse projection
+            const int oi = -(ci + 2);
+            const int s0 = obl_ranges[2 * oi], nn = obl_ranges[2 * oi + 1];
+            float dot = 0.f;
+            for (int q = 0; q < nn; ++q)
+              dot += obl_w[s0 + q] * X[(int64_t)obl_attr[s0 + q] * N + k];
+            right = dot > thr[n] ? 1 : 0;
           } else {
-            right = xv > thr[n] ? 1 : 0;
+            right = X[(int64_t)f * N + k] > thr[n] ? 1 : 0;
           }
           n = left[n] + right;
           f = feat[n];
@@ -564,6 +572,9 @@ struct ShapCtx {
   const int32_t* left;
   const int32_t* cat_idx;
   const unsigned long long* masks;
+  const int32_t* obl_ranges;
+  const int32_t* obl_attr;
+  const float* obl_w;
   const float* cover;
   double* phi;  // [F+1]
   int64_t row;
@@ -571,14 +582,21 @@ struct ShapCtx {
 };
 
 int ShapGoesRight(const ShapCtx& c, int node) {
-  const float xv = c.X[(int64_t)c.feat[node] * c.N + c.row];
   const int ci = c.cat_idx ? c.cat_idx[node] : -1;
   if (ci >= 0) {
-    int cb = (int)xv;
+    int cb = (int)c.X[(int64_t)c.feat[node] * c.N + c.row];
     cb = cb < 0 ? 0 : (cb > 255 ? 255 : cb);
     return (int)((c.masks[(int64_t)ci * 4 + (cb >> 6)] >> (cb & 63)) & 1ull);
   }
-  return xv > c.thr[node] ? 1 : 0;
+  if (ci <= -2) {  // oblique: credit goes to feat[node] (first attribute)
+    const int oi = -(ci + 2);
+    const int s0 = c.obl_ranges[2 * oi], nn = c.obl_ranges[2 * oi + 1];
+    float dot = 0.f;
+    for (int q = 0; q < nn; ++q)
+      dot += c.obl_w[s0 + q] * c.X[(int64_t)c.obl_attr[s0 + q] * c.N + c.row];
+    return dot > c.thr[node] ? 1 : 0;
+  }
+  return c.X[(int64_t)c.feat[node] * c.N + c.row] > c.thr[node] ? 1 : 0;
 }
 
 void ShapRecurse(const ShapCtx& c, int node, PathElem* parent_path,
@@ -623,6 +641,8 @@ extern "C" void cpu_tree_shap(const float* X, int64_t N, int F,
                               const int32_t* feat, const float* thr,
                               const int32_t* left, const int32_t* cat_idx,
                               const unsigned long long* masks,
+                              const int32_t* obl_ranges,
+                              const int32_t* obl_attr, const float* obl_w,
                               const float* cover, const int32_t* roots,
                               int tree_start, int tree_step, int n_trees,
                               float scale, float init, float* phi_out) {
@@ -639,8 +659,9 @@ extern "C" void cpu_tree_shap(const float* X, int64_t N, int F,
           std::fill(phi.begin(), phi.end(), 0.0);
           for (int tt = 0; tt < n_trees; ++tt) {
             const int root = roots[tree_start + (int64_t)tt * tree_step];
-            ShapCtx c{X,    N,     feat, thr, left, cat_idx, masks,
-                      cover, phi.data(), row, scale};
+            ShapCtx c{X,          N,        feat,  thr,   left,
+                      cat_idx,    masks,    obl_ranges, obl_attr, obl_w,
+                      cover,      phi.data(), row,  scale};
             PathElem dummy[1];
             ShapRecurse(c, root, dummy, 0, 1.f, 1.f, -1);
           }

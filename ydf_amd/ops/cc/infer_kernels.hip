@@ -39,7 +39,10 @@ __global__ void predict_forest_lds_kernel(
     const float* __restrict__ X, int64_t N, int F,
     const PackedNode* __restrict__ nodes,
     const int32_t* __restrict__ roots,
-    const unsigned long long* __restrict__ masks, int has_cats,
+    const unsigned long long* __restrict__ masks,
+    const int32_t* __restrict__ obl_ranges,
+    const int32_t* __restrict__ obl_attr,
+    const float* __restrict__ obl_w, int has_cats,
     int tree_start, int tree_step, int n_trees, float* __restrict__ out,
     float init, float scale) {
   extern __shared__ float xs[];  // [F][kTile]
@@ -90,15 +93,23 @@ __global__ void predict_forest_lds_kernel(
     for (int tt = 0; tt < n_trees; ++tt) {
       PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
       while (nd.feat >= 0) {
-        const float xv = xs[nd.feat * kTile + tid];
         int right;
         if (nd.cat_idx >= 0) {
+          const float xv = xs[nd.feat * kTile + tid];
           int c = (int)xv;
           c = c < 0 ? 0 : (c > 255 ? 255 : c);
           right = (int)((masks[(int64_t)nd.cat_idx * 4 + (c >> 6)]
                          >> (c & 63)) & 1ull);
+        } else if (nd.cat_idx <= -2) {  // oblique: sparse dot > thr
+          const int oi = -(nd.cat_idx + 2);
+          const int s0 = obl_ranges[2 * oi];
+          const int nn = obl_ranges[2 * oi + 1];
+          float dot = 0.f;
+          for (int k = 0; k < nn; ++k)
+            dot += obl_w[s0 + k] * xs[obl_attr[s0 + k] * kTile + tid];
+          right = dot > nd.thr ? 1 : 0;
         } else {
-          right = xv > nd.thr ? 1 : 0;
+          right = xs[nd.feat * kTile + tid] > nd.thr ? 1 : 0;
         }
         nd = nodes[nd.left + right];
       }
@@ -113,7 +124,10 @@ __global__ void predict_forest_global_kernel(
     const float* __restrict__ X, int64_t N, int F,
     const PackedNode* __restrict__ nodes,
     const int32_t* __restrict__ roots,
-    const unsigned long long* __restrict__ masks, int has_cats,
+    const unsigned long long* __restrict__ masks,
+    const int32_t* __restrict__ obl_ranges,
+    const int32_t* __restrict__ obl_attr,
+    const float* __restrict__ obl_w, int has_cats,
     int tree_start, int tree_step, int n_trees, float* __restrict__ out,
     float init, float scale) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -123,15 +137,23 @@ __global__ void predict_forest_global_kernel(
     for (int tt = 0; tt < n_trees; ++tt) {
       PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
       while (nd.feat >= 0) {
-        const float xv = X[(int64_t)nd.feat * N + k];
         int right;
         if (has_cats && nd.cat_idx >= 0) {
+          const float xv = X[(int64_t)nd.feat * N + k];
           int c = (int)xv;
           c = c < 0 ? 0 : (c > 255 ? 255 : c);
           right = (int)((masks[(int64_t)nd.cat_idx * 4 + (c >> 6)]
                          >> (c & 63)) & 1ull);
+        } else if (has_cats && nd.cat_idx <= -2) {
+          const int oi = -(nd.cat_idx + 2);
+          const int s0 = obl_ranges[2 * oi];
+          const int nn = obl_ranges[2 * oi + 1];
+          float dot = 0.f;
+          for (int kk = 0; kk < nn; ++kk)
+            dot += obl_w[s0 + kk] * X[(int64_t)obl_attr[s0 + kk] * N + k];
+          right = dot > nd.thr ? 1 : 0;
         } else {
-          right = xv > nd.thr ? 1 : 0;
+          right = X[(int64_t)nd.feat * N + k] > nd.thr ? 1 : 0;
         }
         nd = nodes[nd.left + right];
       }
@@ -153,9 +175,11 @@ extern "C" {
 
 void gpu_predict_forest(const float* X, int64_t N, int F,
                         const int32_t* packed_nodes, const int32_t* roots,
-                        const unsigned long long* masks, int has_cats,
-                        int tree_start, int tree_step, int n_trees,
-                        float* out, float init, float scale, void* stream) {
+                        const unsigned long long* masks,
+                        const int32_t* obl_ranges, const int32_t* obl_attr,
+                        const float* obl_w, int has_cats, int tree_start,
+                        int tree_step, int n_trees, float* out, float init,
+                        float scale, void* stream) {
   const PackedNode* nodes =
       reinterpret_cast<const PackedNode*>(packed_nodes);
   const size_t lds = (size_t)F * kTile * sizeof(float);
@@ -163,16 +187,16 @@ void gpu_predict_forest(const float* X, int64_t N, int F,
     const int grid = (int)((N + kTile - 1) / kTile);
     hipLaunchKernelGGL(predict_forest_lds_kernel, dim3(grid), dim3(kTile), lds,
                        (hipStream_t)stream, X, N, F, nodes, roots, masks,
-                       has_cats, tree_start, tree_step, n_trees, out, init,
-                       scale);
+                       obl_ranges, obl_attr, obl_w, has_cats, tree_start,
+                       tree_step, n_trees, out, init, scale);
   } else {
     int grid = (int)((N + kTile - 1) / kTile);
     if (grid > 4096) grid = 4096;
     if (grid < 1) grid = 1;
     hipLaunchKernelGGL(predict_forest_global_kernel, dim3(grid), dim3(kTile),
                        0, (hipStream_t)stream, X, N, F, nodes, roots, masks,
-                       has_cats, tree_start, tree_step, n_trees, out, init,
-                       scale);
+                       obl_ranges, obl_attr, obl_w, has_cats, tree_start,
+                       tree_step, n_trees, out, init, scale);
   }
 }
 
