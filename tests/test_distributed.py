@@ -127,3 +127,45 @@ def test_gbt_distributed_quality(tmp_path):
         got = pickle.load(f)
     assert got["auc"] > 0.97, got
     assert got["n_trees"] >= 10
+
+
+def _worker_oblique(rank, world, port, out_path):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from ydf_amd.parallel.dist import shard_rows
+
+        data = _make_data()
+        ds = ydf.create_vertical_dataset(data, label="label",
+                                         task=ydf.Task.CLASSIFICATION)
+        lo, hi = shard_rows(ds.n_examples, rank, world)
+        m = ydf.GradientBoostedTreesLearner(
+            label="label", num_trees=8, max_depth=4, validation_ratio=0,
+            split_axis="SPARSE_OBLIQUE",
+            device="cpu").train(ds.shard(lo, hi))
+        # EVERY rank writes its forest: ranks must agree bit-for-bit
+        with open(f"{out_path}.{rank}", "wb") as f:
+            pickle.dump({"feat": m.forest.feat, "thr": m.forest.thr,
+                         "obl_w": m.forest.obl_w,
+                         "obl_attr": m.forest.obl_attr}, f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_oblique_distributed_rank_consistent(tmp_path):
+    """Oblique projections are seeded per (tree, level) and bin cuts are
+    broadcast from rank 0, so all ranks must build the same model."""
+    out = str(tmp_path / "obl.pkl")
+    _spawn(_worker_oblique, out)
+    with open(out + ".0", "rb") as f:
+        a = pickle.load(f)
+    with open(out + ".1", "rb") as f:
+        b = pickle.load(f)
+    np.testing.assert_array_equal(a["feat"], b["feat"])
+    np.testing.assert_array_equal(a["thr"], b["thr"])
+    np.testing.assert_array_equal(a["obl_attr"], b["obl_attr"])
+    np.testing.assert_array_equal(a["obl_w"], b["obl_w"])
+    assert len(a["obl_attr"]) > 0

@@ -424,3 +424,78 @@ def test_mae_loss_robust_to_outliers():
             {"x": x, "label": y})
     p = m.predict({"x": x})
     assert np.median(np.abs(p - 2 * x)) < 0.5
+
+
+def _rotated_data(n=8000, seed=0):
+    rng = np.random.RandomState(seed)
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    return {"x1": x1, "x2": x2,
+            "label": np.where(x1 + x2 > 0, "p", "n")}
+
+
+def test_oblique_gbt_beats_axis_aligned(tmp_path):
+    """A diagonal decision boundary: oblique splits must (a) appear in the
+    model and (b) beat axis-aligned trees of the same budget (reference
+    SparseObliqueSplit behavior)."""
+    tr = _rotated_data(8000, 0)
+    te = _rotated_data(2000, 1)
+    kw = dict(label="label", num_trees=20, max_depth=3, validation_ratio=0)
+    acc_ax = ydf.GradientBoostedTreesLearner(**kw).train(tr).evaluate(
+        te).accuracy
+    m = ydf.GradientBoostedTreesLearner(
+        split_axis="SPARSE_OBLIQUE", **kw).train(tr)
+    acc_ob = m.evaluate(te).accuracy
+    assert (m.forest.cat_idx <= -2).sum() > 0
+    assert acc_ob > acc_ax
+    assert acc_ob > 0.99
+    # persistence round-trip keeps oblique predictions exactly
+    p1 = m.predict(te, device="cpu")
+    m.save(str(tmp_path / "obl"))
+    m2 = ydf.load_model(str(tmp_path / "obl"))
+    np.testing.assert_array_equal(p1, m2.predict(te, device="cpu"))
+
+
+def test_oblique_gbt_validation_and_subsample():
+    """Oblique + internal validation split (routes the valid rows through
+    per-level replayed projections) + row subsampling (routes
+    out-of-sample rows)."""
+    tr = _rotated_data(6000, 2)
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15, max_depth=3, validation_ratio=0.15,
+        subsample=0.6, split_axis="SPARSE_OBLIQUE").train(tr)
+    assert m.evaluate(_rotated_data(2000, 3)).accuracy > 0.98
+    assert m.training_logs  # validation loss was tracked
+
+
+def test_oblique_rf():
+    tr = _rotated_data(6000, 4)
+    m = ydf.RandomForestLearner(
+        label="label", num_trees=10, max_depth=8,
+        split_axis="SPARSE_OBLIQUE",
+        compute_oob_performances=False).train(tr)
+    assert (m.forest.cat_idx <= -2).sum() > 0
+    assert m.evaluate(_rotated_data(2000, 5)).accuracy > 0.97
+
+
+def test_oblique_with_categorical_features():
+    """Categorical features keep set-splits; projections draw only from
+    the numerical columns."""
+    rng = np.random.RandomState(7)
+    n = 6000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    c = rng.randint(0, 6, n)
+    y = (x1 + x2 > 0) ^ (c % 3 == 0)
+    d = {"x1": x1, "x2": x2, "c": np.array([f"v{v}" for v in c]),
+         "label": np.where(y, "p", "n")}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=25, max_depth=4, validation_ratio=0,
+        split_axis="SPARSE_OBLIQUE").train(d)
+    assert m.evaluate(d).accuracy > 0.95
+    f = m.forest
+    # oblique conditions never reference the categorical column
+    ci = f.dataspec_cat_col if hasattr(f, "dataspec_cat_col") else None
+    cat_pos = [i for i, cspec in enumerate(m.dataspec.feature_columns)
+               if cspec.name == "c"][0]
+    assert not np.any(f.obl_attr == cat_pos)

@@ -52,6 +52,30 @@ class GenericLearner:
             [c.semantic == Semantic.CATEGORICAL
              for c in ds.dataspec.feature_columns], dtype=bool)
 
+    def _oblique_cfg(self, F: int, cat_flags) -> Dict:
+        """TrainerConfig kwargs for sparse-oblique splits (reference
+        SparseObliqueSplit defaults, decision_tree.proto:173-296):
+        P = clamp(ceil(n_numerical ^ exponent), 1, max)."""
+        hp = self.hyperparameters
+        if hp.get("split_axis", "AXIS_ALIGNED") != "SPARSE_OBLIQUE":
+            return {}
+        import math as _math
+
+        n_num = F - int(cat_flags.sum().item()) \
+            if cat_flags is not None else F
+        if n_num <= 0:
+            return {}
+        P = int(min(hp.get("sparse_oblique_max_num_projections", 6000),
+                    max(1, _math.ceil(n_num ** hp.get(
+                        "sparse_oblique_num_projections_exponent", 2.0)))))
+        return dict(
+            oblique_projections=P,
+            oblique_density=hp.get(
+                "sparse_oblique_projection_density_factor", 2.0),
+            oblique_weights=hp.get("sparse_oblique_weights", "BINARY"),
+            oblique_norm=hp.get("sparse_oblique_normalization", "NONE"),
+        )
+
     def _bin_matrix(self, ds_X: np.ndarray, cat_feats: np.ndarray,
                     bnd: np.ndarray, device: torch.device) -> torch.Tensor:
         """Bins numericals by quantile cuts; categorical codes pass through

@@ -52,6 +52,12 @@ class GradientBoostedTreesLearner(GenericLearner):
                  use_hessian_gain: bool = True,
                  apply_link_function: bool = True,
                  l2_categorical_regularization: float = 1.0,
+                 split_axis: str = "AXIS_ALIGNED",
+                 sparse_oblique_num_projections_exponent: float = 2.0,
+                 sparse_oblique_max_num_projections: int = 6000,
+                 sparse_oblique_projection_density_factor: float = 2.0,
+                 sparse_oblique_normalization: str = "NONE",
+                 sparse_oblique_weights: str = "BINARY",
                  loss: str = "DEFAULT",
                  working_dir: Optional[str] = None,
                  resume_training: bool = False,
@@ -77,6 +83,15 @@ class GradientBoostedTreesLearner(GenericLearner):
             use_hessian_gain=use_hessian_gain,
             apply_link_function=apply_link_function,
             l2_categorical_regularization=l2_categorical_regularization,
+            split_axis=split_axis,
+            sparse_oblique_num_projections_exponent=(
+                sparse_oblique_num_projections_exponent),
+            sparse_oblique_max_num_projections=(
+                sparse_oblique_max_num_projections),
+            sparse_oblique_projection_density_factor=(
+                sparse_oblique_projection_density_factor),
+            sparse_oblique_normalization=sparse_oblique_normalization,
+            sparse_oblique_weights=sparse_oblique_weights,
             loss=loss,
             working_dir=working_dir, resume_training=resume_training,
             resume_training_snapshot_interval_seconds=(
@@ -110,6 +125,11 @@ class GradientBoostedTreesLearner(GenericLearner):
             data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
+        obl = self._oblique_cfg(bins.shape[0], cat_flags)
+        raw_t = valid_raw_t = None
+        if obl:
+            raw_t = torch.from_numpy(
+                np.ascontiguousarray(ds.X)).to(device)
         classes = self._label_classes(ds)
         n_classes = len(classes) if classes else 2
         custom_loss = None
@@ -156,6 +176,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                 valid_labels = labels[vi].contiguous()
                 bins = bins[:, ti].contiguous()
                 labels = labels[ti].contiguous()
+                if raw_t is not None:
+                    valid_raw_t = raw_t[:, vi].contiguous()
+                    raw_t = raw_t[:, ti].contiguous()
                 valid_ranking = RankingLambdas(
                     group_ids[vmask], group_ids[vmask] * 0
                     + valid_labels.cpu().numpy(), device,
@@ -165,7 +188,11 @@ class GradientBoostedTreesLearner(GenericLearner):
                                      device,
                                      truncation=self.ndcg_truncation)
         elif valid is not None:
-            valid_bins, valid_labels = self._prepare_valid(valid, ds, device)
+            valid_bins, valid_labels, vds = self._prepare_valid(
+                valid, ds, device)
+            if raw_t is not None:
+                valid_raw_t = torch.from_numpy(
+                    np.ascontiguousarray(vds.X)).to(device)
         elif vr > 0.0 and hp["early_stopping"] != "NONE":
             N = bins.shape[1]
             rng = np.random.RandomState(self.random_seed)
@@ -180,6 +207,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                 labels = labels[ti].contiguous()
                 if weights is not None:
                     weights = weights[ti].contiguous()
+                if raw_t is not None:
+                    valid_raw_t = raw_t[:, vi].contiguous()
+                    raw_t = raw_t[:, ti].contiguous()
 
         F = bins.shape[0]
         ncand = 0
@@ -204,12 +234,14 @@ class GradientBoostedTreesLearner(GenericLearner):
             early_stopping_initial_iteration=(
                 hp["early_stopping_initial_iteration"]),
             cat_smooth=hp["l2_categorical_regularization"],
+            **obl,
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
                                       valid_bins=valid_bins,
                                       valid_labels=valid_labels,
                                       cat_flags=cat_flags, weights=weights,
-                                      mono=mono)
+                                      mono=mono, raw=raw_t,
+                                      valid_raw=valid_raw_t)
         C = n_classes if loss == trainer_lib.LOSS_MULTINOMIAL else 1
         activation = "identity"
         if custom_loss is not None:
@@ -339,7 +371,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                                 device)
         labels = torch.from_numpy(
             np.ascontiguousarray(vds.label_values)).to(device)
-        return bins, labels
+        return bins, labels, vds
 
 
 class RandomForestLearner(GenericLearner):
@@ -363,6 +395,12 @@ class RandomForestLearner(GenericLearner):
                  num_candidate_attributes_ratio: float = -1.0,
                  winner_take_all: bool = True,
                  compute_oob_performances: bool = True,
+                 split_axis: str = "AXIS_ALIGNED",
+                 sparse_oblique_num_projections_exponent: float = 2.0,
+                 sparse_oblique_max_num_projections: int = 6000,
+                 sparse_oblique_projection_density_factor: float = 2.0,
+                 sparse_oblique_normalization: str = "NONE",
+                 sparse_oblique_weights: str = "BINARY",
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
@@ -375,6 +413,15 @@ class RandomForestLearner(GenericLearner):
             num_candidate_attributes_ratio=num_candidate_attributes_ratio,
             winner_take_all=winner_take_all,
             compute_oob_performances=compute_oob_performances,
+            split_axis=split_axis,
+            sparse_oblique_num_projections_exponent=(
+                sparse_oblique_num_projections_exponent),
+            sparse_oblique_max_num_projections=(
+                sparse_oblique_max_num_projections),
+            sparse_oblique_projection_density_factor=(
+                sparse_oblique_projection_density_factor),
+            sparse_oblique_normalization=sparse_oblique_normalization,
+            sparse_oblique_weights=sparse_oblique_weights,
         )
 
     def _num_candidate(self, F: int) -> int:
@@ -401,21 +448,38 @@ class RandomForestLearner(GenericLearner):
             data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
+        obl = self._oblique_cfg(bins.shape[0], cat_flags)
+        raw_t = valid_raw_t = None
+        if obl:
+            raw_t = torch.from_numpy(
+                np.ascontiguousarray(ds.X)).to(device)
         classes = self._label_classes(ds) \
             if self._task == Task.CLASSIFICATION else None
         n_classes = len(classes) if classes else 2
         F = bins.shape[0]
+        obl = self._oblique_cfg(F, cat_flags)
+        raw_t = None
+        if obl:
+            raw_t = torch.from_numpy(np.ascontiguousarray(ds.X)).to(device)
+        # with oblique projections the candidate pool is F + P virtual
+        # features; the reference evaluates every projection, so feature
+        # sampling is disabled unless explicitly requested
+        ncand = self._num_candidate(F)
+        if obl and hp["num_candidate_attributes"] == 0 \
+                and hp["num_candidate_attributes_ratio"] <= 0:
+            ncand = 0
         cfg = trainer_lib.TrainerConfig(
             loss=trainer_lib.LOSS_RF, num_trees=hp["num_trees"],
             max_depth=hp["max_depth"], shrinkage=1.0, lambda_l2=0.0,
             min_examples=hp["min_examples"], min_hessian=0.0,
             n_classes=n_classes, seed=self.random_seed,
             bootstrap=hp["bootstrap_training_dataset"],
-            num_candidate_features=self._num_candidate(F),
+            num_candidate_features=ncand,
+            **obl,
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
                                       cat_flags=cat_flags, weights=weights,
-                                      mono=mono)
+                                      mono=mono, raw=raw_t)
         compute_oob = (hp["compute_oob_performances"]
                        and hp["bootstrap_training_dataset"])
         result = trainer_lib.train_rf(t, log=info, compute_oob=compute_oob)

@@ -62,6 +62,16 @@ class TrainerConfig:
     early_stopping_initial_iteration: int = 10
     # device memory budget for the per-level histogram buffer
     hist_budget_bytes: int = 1 << 31
+    # sparse-oblique splits (reference SparseObliqueSplit,
+    # learner/decision_tree/decision_tree.proto:173): P random sparse
+    # projections are sampled PER LEVEL, projected via GEMM (rocBLAS/MFMA),
+    # quantile-binned and appended as virtual features [F, F+P) to the
+    # histogram/scan machinery; a winning virtual feature becomes an
+    # oblique node. 0 = axis-aligned only.
+    oblique_projections: int = 0
+    oblique_density: float = 2.0      # expected nonzeros per projection
+    oblique_weights: str = "BINARY"   # BINARY | CONTINUOUS
+    oblique_norm: str = "NONE"        # NONE | STANDARD_DEVIATION | MIN_MAX
 
 
 @dataclasses.dataclass
@@ -75,6 +85,8 @@ class HostTree:
     max_depth: int
     masks: Optional[np.ndarray] = None  # [total_nodes,4] u64 (cat splits)
     gain: Optional[np.ndarray] = None   # [total_nodes] f32 (split gains)
+    # oblique nodes: {node_idx: (attrs i32[], weights f32[], threshold)}
+    oblique: Optional[dict] = None
 
 
 def _dist_ok() -> bool:
@@ -92,7 +104,9 @@ class ForestTrainer:
                  valid_labels: Optional[torch.Tensor] = None,
                  cat_flags: Optional[torch.Tensor] = None,
                  weights: Optional[torch.Tensor] = None,
-                 mono: Optional[torch.Tensor] = None):
+                 mono: Optional[torch.Tensor] = None,
+                 raw: Optional[torch.Tensor] = None,
+                 valid_raw: Optional[torch.Tensor] = None):
         assert bins.dtype == torch.uint8 and bins.dim() == 2
         self.bins = bins
         self.labels = labels
@@ -108,6 +122,65 @@ class ForestTrainer:
         self.valid_bins = valid_bins
         self.valid_labels = valid_labels
         self.distributed = _dist_ok()
+
+        # ---- sparse-oblique setup: virtual projection features ----------
+        self.P = cfg.oblique_projections
+        self.base_F = self.F
+        self.raw = raw
+        self.valid_raw = valid_raw
+        if self.P > 0:
+            assert raw is not None, \
+                "oblique training needs the raw feature matrix (raw=)"
+            dev = self.device
+            FT = self.F + self.P
+            bins_ext = torch.zeros((FT, self.N), dtype=torch.uint8,
+                                   device=dev)
+            bins_ext[:self.F] = bins
+            self.bins = bins_ext
+            if self.cat_flags is not None:
+                num_idx = np.nonzero(
+                    self.cat_flags.cpu().numpy() == 0)[0]
+                cat_ext = torch.zeros(FT, dtype=torch.uint8, device=dev)
+                cat_ext[:self.F] = self.cat_flags
+                self.cat_flags = cat_ext
+            else:
+                num_idx = np.arange(self.F)
+            self.num_feat_idx = num_idx.astype(np.int64)
+            if self.mono is not None:
+                mono_ext = torch.zeros(FT, dtype=torch.int8, device=dev)
+                mono_ext[:self.F] = self.mono
+                self.mono = mono_ext
+            if valid_bins is not None:
+                assert valid_raw is not None, \
+                    "oblique + validation needs valid_raw="
+                vext = torch.zeros((FT, valid_bins.shape[1]),
+                                   dtype=torch.uint8, device=dev)
+                vext[:self.F] = valid_bins
+                self.valid_bins = vext
+            # normalization is a per-feature weight scale (centering only
+            # shifts the projected values, which the learned threshold
+            # absorbs — reference sparse_oblique_normalization)
+            sub = raw[:, ::max(1, self.N // 65536)]
+            if cfg.oblique_norm == "STANDARD_DEVIATION":
+                sc = 1.0 / sub.std(dim=1).clamp(min=1e-12)
+            elif cfg.oblique_norm == "MIN_MAX":
+                sc = 1.0 / (sub.max(dim=1).values
+                            - sub.min(dim=1).values).clamp(min=1e-12)
+            else:
+                sc = torch.ones(self.F, device=dev)
+            if self.distributed:
+                torch.distributed.broadcast(sc, src=0)
+            self.proj_scale = sc.cpu().numpy().astype(np.float32)
+            self.Z = torch.empty((self.P, self.N), dtype=torch.float32,
+                                 device=dev)
+            self._q_levels = torch.linspace(0.0, 1.0, ops.MAX_BINS + 1,
+                                            device=dev)[1:-1]
+            d = cfg.max_depth
+            self._W_lv = [None] * d       # device [P, base_F] f32
+            self._W_host_lv = [None] * d  # host copies for extraction
+            self._cuts_lv = [None] * d    # device [P, MAX_BINS-1] f32
+            self._cuts_host_lv = [None] * d
+            self.F = FT
 
         d = cfg.max_depth
         self.total_nodes = (1 << (d + 1)) - 1
@@ -126,7 +199,8 @@ class ForestTrainer:
                                 dtype=torch.float32, device=dev)
         # Histogram-subtraction trick (sibling = parent - smaller child):
         # previous level's histograms, indexed by its slot order.
-        self.use_hist_sub = os.environ.get("YDFA_NO_HIST_SUB", "0") != "1"
+        self.use_hist_sub = os.environ.get("YDFA_NO_HIST_SUB", "0") != "1" \
+            and self.P == 0
         self.hist_prev = torch.empty_like(self.hist) if self.use_hist_sub \
             else None
         # dense mode: whole levels stay device-resident while the level fits
@@ -209,6 +283,56 @@ class ForestTrainer:
         np.put_along_axis(mask, idx, 1, axis=1)
         return torch.from_numpy(mask).to(self.device)
 
+    # -- sparse-oblique projections ---------------------------------------
+    def _sample_projections(self, tree_idx: int, level: int) -> None:
+        """Samples P sparse projections for this (tree, level), projects
+        the training rows (GEMM -> MFMA on GPU), quantile-bins the
+        projected values into the virtual-feature rows of self.bins.
+        Seeded per (tree, level): every data-parallel rank draws the SAME
+        projections; bin cuts are broadcast from rank 0."""
+        cfg = self.cfg
+        P, bF = self.P, self.base_F
+        nf = len(self.num_feat_idx)
+        seed = (cfg.seed * 2654435761 + tree_idx * 97561 + level * 131
+                + 17) % (1 << 31)
+        rs = np.random.RandomState(seed)
+        dens = min(1.0, cfg.oblique_density / max(nf, 1))
+        sel = rs.random_sample((P, nf)) < dens
+        empty = ~sel.any(axis=1)
+        if empty.any():
+            sel[np.nonzero(empty)[0], rs.randint(0, nf, empty.sum())] = True
+        if cfg.oblique_weights == "CONTINUOUS":
+            w = rs.uniform(-1.0, 1.0, size=(P, nf)).astype(np.float32)
+        else:  # BINARY (reference default)
+            w = (rs.randint(0, 2, size=(P, nf)) * 2 - 1).astype(np.float32)
+        w *= sel
+        w *= self.proj_scale[self.num_feat_idx][None, :]
+        W = np.zeros((P, bF), dtype=np.float32)
+        W[:, self.num_feat_idx] = w
+        self._W_host_lv[level] = W
+        Wd = torch.from_numpy(W).to(self.device)
+        self._W_lv[level] = Wd
+        torch.matmul(Wd, self.raw, out=self.Z)
+        sub = self.Z[:, ::max(1, self.N // 65536)]
+        cuts = torch.quantile(sub.float(), self._q_levels, dim=1) \
+            .T.contiguous()
+        if self.distributed:
+            torch.distributed.broadcast(cuts, src=0)
+        self._cuts_lv[level] = cuts
+        self._cuts_host_lv[level] = None  # fetched lazily at extraction
+        ops.bin_data(self.Z, cuts, self.bins[bF:])
+
+    def _apply_projections(self, level: int, raw: torch.Tensor,
+                           bins: torch.Tensor) -> None:
+        """Replays level `level`'s stored projections onto other rows
+        (validation / out-of-sample routing)."""
+        Wd = self._W_lv[level]
+        if Wd is None:
+            return
+        Z = torch.matmul(Wd, raw)
+        ops.bin_data(Z.contiguous(), self._cuts_lv[level],
+                     bins[self.base_F:])
+
     # -- one tree ---------------------------------------------------------
     def grow_tree(self, tree_idx: int,
                   sample_mask: Optional[torch.Tensor] = None) -> HostTree:
@@ -254,6 +378,8 @@ class ForestTrainer:
         for level in range(cfg.max_depth):
             level_base = (1 << level) - 1
             level_size = 1 << level
+            if self.P > 0:
+                self._sample_projections(tree_idx, level)
 
             if level_size <= self.dense_limit:
                 # ---- dense mode: slot == level-relative node index; all
@@ -437,15 +563,33 @@ class ForestTrainer:
     def extract_host_tree(self) -> HostTree:
         cfg = self.cfg
         # .copy(): on CPU .cpu().numpy() aliases the (reused) buffers
+        feat = self.tree_feat.cpu().numpy().copy()
+        bins = self.tree_bin.cpu().numpy().copy()
+        oblique = None
+        if self.P > 0:
+            oblique = {}
+            for n in np.nonzero(feat >= self.base_F)[0]:
+                level = int(n + 1).bit_length() - 1
+                p = int(feat[n]) - self.base_F
+                if self._cuts_host_lv[level] is None:
+                    self._cuts_host_lv[level] = \
+                        self._cuts_lv[level].cpu().numpy()
+                row = self._W_host_lv[level][p]
+                attrs = np.nonzero(row)[0].astype(np.int32)
+                thr = float(self._cuts_host_lv[level][p, int(bins[n])])
+                oblique[int(n)] = (attrs, row[attrs].astype(np.float32),
+                                   thr)
+                feat[n] = int(attrs[0])
         return HostTree(
-            feat=self.tree_feat.cpu().numpy().copy(),
-            bin=self.tree_bin.cpu().numpy().copy(),
+            feat=feat,
+            bin=bins,
             leaf_value=self.leaf_vals.cpu().numpy().copy(),
             counts=self.node_stats[:, 2].cpu().numpy().copy(),
             max_depth=cfg.max_depth,
             masks=self.tree_masks.cpu().numpy().view(np.uint64).copy()
             if self.tree_masks is not None else None,
             gain=self.tree_gain.cpu().numpy().copy(),
+            oblique=oblique,
         )
 
     def _dense_level(self, tree_idx: int, level: int, need: int,
@@ -507,6 +651,7 @@ class ForestTrainer:
         assert self.device.type == "cuda" and not self.distributed
         assert (1 << (self.cfg.max_depth - 1)) <= self.dense_limit
         assert self.cfg.num_candidate_features <= 0
+        assert self.cfg.oblique_projections == 0
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             ops.grad_hess(preds, labels, self.gh, self.cfg.loss)
@@ -514,12 +659,21 @@ class ForestTrainer:
             ops.update_preds(preds, self.node_ids, self.leaf_vals, shrinkage)
         return g
 
-    def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor):
-        """Routes arbitrary rows through the latest tree (device arrays)."""
+    def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor,
+                   raw: Optional[torch.Tensor] = None):
+        """Routes arbitrary rows through the latest tree (device arrays).
+        With oblique training, per-level projections are replayed onto the
+        given rows (raw defaults to the training/validation matrix matching
+        `bins`)."""
         node_ids.zero_()
         for level in range(self.cfg.max_depth):
             level_base = (1 << level) - 1
             level_size = 1 << level
+            if self.P > 0:
+                if raw is None:
+                    raw = self.raw if bins.data_ptr() == \
+                        self.bins.data_ptr() else self.valid_raw
+                self._apply_projections(level, raw, bins)
             ops.update_node_ids(
                 bins, node_ids, self.arange_buf[:level_size],
                 self.tree_feat[level_base:level_base + level_size],

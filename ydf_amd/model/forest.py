@@ -109,12 +109,20 @@ def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
     thr = np.empty(len(nodes), dtype=np.float32)
     cat_idx = np.full(len(nodes), -1, dtype=np.int32)
     masks = np.zeros((0, 4), dtype=np.uint64)
+    obl_ranges, obl_attr, obl_w = [], [], []
+    is_obl_node = np.zeros(len(nodes), dtype=bool)
+    if tree.oblique:
+        for a in tree.oblique:
+            pos = new_idx[a]
+            if pos >= 0 and internal[a]:
+                is_obl_node[pos] = True
     ii = np.nonzero(n_int)[0]
     if ii.size:
         is_cat_node = np.zeros(len(nodes), dtype=bool)
         if cat_feats is not None and tree.masks is not None:
             is_cat_node[ii] = cat_feats[tree.feat[nodes[ii]]]
-        ni = np.nonzero(n_int & ~is_cat_node)[0]
+            is_cat_node &= ~is_obl_node
+        ni = np.nonzero(n_int & ~is_cat_node & ~is_obl_node)[0]
         if ni.size:
             # numerical: tree.bin is a cut index; categorical nodes store a
             # sorted RANK there instead, so they must not index boundaries
@@ -124,32 +132,53 @@ def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
             thr[ci] = 0.0
             cat_idx[ci] = np.arange(ci.size, dtype=np.int32)
             masks = tree.masks[nodes[ci]].astype(np.uint64)
+        for pos in np.nonzero(is_obl_node)[0]:
+            attrs, ws, t = tree.oblique[int(nodes[pos])]
+            cat_idx[pos] = -(2 + len(obl_ranges))
+            obl_ranges.append((len(obl_attr), len(attrs)))
+            obl_attr.extend(int(a) for a in attrs)
+            obl_w.extend(float(w) for w in ws)
+            thr[pos] = t
     li = np.nonzero(~n_int)[0]
     thr[li] = tree.leaf_value[nodes[li]] * leaf_scale
     cover = tree.counts[nodes].astype(np.float32)
-    return feat, thr, left, cat_idx, masks, cover
+    obl = (np.asarray(obl_ranges, np.int32).reshape(-1, 2),
+           np.asarray(obl_attr, np.int32), np.asarray(obl_w, np.float32))
+    return feat, thr, left, cat_idx, masks, cover, obl
 
 
 def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
                       leaf_scale: float = 1.0, cat_feats=None) -> FlatForest:
     feats, thrs, lefts, roots, cidxs, mask_list, covers = \
         [], [], [], [], [], [], []
+    rng_list, attr_list, w_list = [], [], []
     off = 0
     mask_off = 0
+    obl_off = 0
+    term_off = 0
     for t in trees:
-        f, th, lf, ci, mk, cv = host_tree_to_flat(t, boundaries, leaf_scale,
-                                                  cat_feats)
+        f, th, lf, ci, mk, cv, (orng, oat, ow) = host_tree_to_flat(
+            t, boundaries, leaf_scale, cat_feats)
         lf = np.where(f >= 0, lf + off, 0)
-        ci = np.where(ci >= 0, ci + mask_off, -1)
+        ci = np.where(ci >= 0, ci + mask_off, ci)
+        ci = np.where(ci <= -2, ci - obl_off, ci)
         roots.append(off)
         off += len(f)
         mask_off += len(mk)
+        if len(orng):
+            orng = orng.copy()
+            orng[:, 0] += term_off
+        obl_off += len(orng)
+        term_off += len(oat)
         feats.append(f)
         thrs.append(th)
         lefts.append(lf)
         cidxs.append(ci)
         mask_list.append(mk)
         covers.append(cv)
+        rng_list.append(orng)
+        attr_list.append(oat)
+        w_list.append(ow)
     return FlatForest(
         feat=np.concatenate(feats) if feats else np.zeros(0, np.int32),
         thr=np.concatenate(thrs) if thrs else np.zeros(0, np.float32),
@@ -160,6 +189,11 @@ def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
         else np.zeros((0, 4), np.uint64),
         cover=np.concatenate(covers) if covers
         else np.zeros(0, np.float32),
+        obl_ranges=(np.concatenate(rng_list).astype(np.int32)
+                    if obl_off else None),
+        obl_attr=np.concatenate(attr_list).astype(np.int32)
+        if attr_list else None,
+        obl_w=np.concatenate(w_list).astype(np.float32) if w_list else None,
     )
 
 
