@@ -499,3 +499,62 @@ def test_oblique_with_categorical_features():
     cat_pos = [i for i, cspec in enumerate(m.dataspec.feature_columns)
                if cspec.name == "c"][0]
     assert not np.any(f.obl_attr == cat_pos)
+
+
+def test_focal_loss_binary():
+    """BINARY_FOCAL_LOSS (reference loss_imp_binary_focal.cc): trains a
+    usable classifier on an imbalanced problem; sigmoid activation."""
+    rng = np.random.RandomState(0)
+    n = 8000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where((2 * x1 - x2 + 0.3 * rng.randn(n)) > 1.8, "pos", "neg")
+    d = {"x1": x1, "x2": x2, "label": y}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=40, loss="BINARY_FOCAL_LOSS",
+        focal_loss_alpha=0.75, validation_ratio=0.1).train(d)
+    ev = m.evaluate(d)
+    assert ev.auc > 0.99
+    p = m.predict(d)
+    assert 0.0 <= p.min() and p.max() <= 1.0  # sigmoid applied
+    with pytest.raises(ValueError):
+        ydf.GradientBoostedTreesLearner(
+            label="label", loss="BINARY_FOCAL_LOSS").train(
+                {"x": np.arange(30, dtype=np.float32),
+                 "label": np.array(["a", "b", "c"] * 10)})
+
+
+def test_xe_ndcg_ranking():
+    """XE_NDCG_MART (reference loss_imp_cross_entropy_ndcg.cc)."""
+    rng = np.random.RandomState(1)
+    q = np.repeat(np.arange(200), 10)
+    rel = rng.randint(0, 5, 2000).astype(np.float32)
+    d = {"q": q, "rel": rel,
+         "f1": (rel + rng.randn(2000)).astype(np.float32),
+         "f2": rng.randn(2000).astype(np.float32)}
+    m = ydf.GradientBoostedTreesLearner(
+        label="rel", ranking_group="q", task=ydf.Task.RANKING,
+        num_trees=30, loss="XE_NDCG_MART").train(d)
+    assert m.evaluate(d).ndcg > 0.9
+
+
+def test_dart():
+    """DART forest extraction: dropout + retroactive rescale. With one
+    tree no dropout can occur, so DART == MART exactly; with many trees
+    the model stays accurate."""
+    rng = np.random.RandomState(2)
+    n = 6000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2,
+         "label": np.where(2 * x1 - x2 + 0.5 * x1 * x2 > 0, "a", "b")}
+    kw = dict(label="label", validation_ratio=0)
+    m1 = ydf.GradientBoostedTreesLearner(
+        num_trees=1, forest_extraction="DART", dart_dropout=0.5, **kw
+    ).train(d)
+    m2 = ydf.GradientBoostedTreesLearner(num_trees=1, **kw).train(d)
+    np.testing.assert_allclose(m1.predict(d), m2.predict(d), atol=1e-6)
+    m = ydf.GradientBoostedTreesLearner(
+        num_trees=60, forest_extraction="DART", dart_dropout=0.1,
+        validation_ratio=0.1, label="label").train(d)
+    assert m.evaluate(d).accuracy > 0.98

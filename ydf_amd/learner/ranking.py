@@ -46,6 +46,7 @@ class RankingLambdas:
         sorted_gains, _ = torch.sort(self.gains, dim=1, descending=True)
         disc = self._discounts(device)
         self.idcg = (sorted_gains[:, : self.M] * disc).sum(dim=1).clamp(1e-9)
+        self.sizes = self.valid.sum(dim=1)
         self.N = len(group_ids)
         self.device = device
 
@@ -89,6 +90,46 @@ class RankingLambdas:
         h = torch.zeros(self.N, dtype=torch.float32, device=self.device)
         flat_idx = self.safe_idx[self.valid]
         g[flat_idx] = g_mat[self.valid]
+        h[flat_idx] = h_mat[self.valid].clamp(1e-6, 16.0)
+        return g, h
+
+    def xe_ndcg(self, scores: torch.Tensor, seed: int):
+        """Cross-entropy NDCG gradients (reference
+        loss_imp_cross_entropy_ndcg.cc:104-180): softmax the per-group
+        scores, target distribution (2^rel - gamma)/sum with gamma ~
+        U(0,1) resampled per iteration, third-order Newton correction."""
+        S = torch.where(self.valid, scores[self.safe_idx],
+                        torch.full((), -1e30, device=self.device))
+        p = torch.softmax(S, dim=1).clamp(1e-5, 0.99999)
+        p = torch.where(self.valid, p, torch.zeros_like(p))
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed % (1 << 62))
+        gamma = torch.rand((self.Q, self.M), generator=gen,
+                           device=self.device)
+        y = torch.where(self.valid, 2.0 ** self.rel - gamma,
+                        torch.zeros_like(gamma))
+        denom = y.sum(dim=1, keepdim=True)
+        ok = ((denom != 0) & (self.sizes.unsqueeze(1) > 1)).float()
+        denom = torch.where(denom == 0, torch.ones_like(denom), denom)
+        t1 = -y / denom + p
+        g_mat = -t1
+        one_m_p = (1.0 - p).clamp(min=1e-5)
+        l1 = t1 / one_m_p
+        s1 = l1.sum(dim=1, keepdim=True)
+        t2 = p * (s1 - l1)
+        g_mat = g_mat - t2
+        l2 = t2 / one_m_p
+        s2 = l2.sum(dim=1, keepdim=True)
+        g_mat = g_mat - p * (s2 - l2)
+        h_mat = p * (1.0 - p)
+        g_mat = g_mat * ok
+        h_mat = h_mat * ok
+        # reference fits trees to -gradient; our convention is g = dL/dm,
+        # and the reference's gradient_data is the DESCENT direction
+        g = torch.zeros(self.N, dtype=torch.float32, device=self.device)
+        h = torch.zeros(self.N, dtype=torch.float32, device=self.device)
+        flat_idx = self.safe_idx[self.valid]
+        g[flat_idx] = -g_mat[self.valid]
         h[flat_idx] = h_mat[self.valid].clamp(1e-6, 16.0)
         return g, h
 

@@ -52,6 +52,10 @@ class GradientBoostedTreesLearner(GenericLearner):
                  use_hessian_gain: bool = True,
                  apply_link_function: bool = True,
                  l2_categorical_regularization: float = 1.0,
+                 focal_loss_alpha: float = 0.5,
+                 focal_loss_gamma: float = 2.0,
+                 forest_extraction: str = "MART",
+                 dart_dropout: float = 0.01,
                  split_axis: str = "AXIS_ALIGNED",
                  sparse_oblique_num_projections_exponent: float = 2.0,
                  sparse_oblique_max_num_projections: int = 6000,
@@ -83,6 +87,10 @@ class GradientBoostedTreesLearner(GenericLearner):
             use_hessian_gain=use_hessian_gain,
             apply_link_function=apply_link_function,
             l2_categorical_regularization=l2_categorical_regularization,
+            focal_loss_alpha=focal_loss_alpha,
+            focal_loss_gamma=focal_loss_gamma,
+            forest_extraction=forest_extraction,
+            dart_dropout=dart_dropout,
             split_axis=split_axis,
             sparse_oblique_num_projections_exponent=(
                 sparse_oblique_num_projections_exponent),
@@ -137,8 +145,16 @@ class GradientBoostedTreesLearner(GenericLearner):
                 and not isinstance(hp.get("loss"), str):
             custom_loss = hp["loss"]
         if self._task == Task.CLASSIFICATION:
-            loss = (trainer_lib.LOSS_MULTINOMIAL if n_classes > 2
-                    else trainer_lib.LOSS_BINOMIAL)
+            named = hp.get("loss") if isinstance(hp.get("loss"), str) \
+                else "DEFAULT"
+            if named == "BINARY_FOCAL_LOSS":
+                if n_classes > 2:
+                    raise ValueError(
+                        "BINARY_FOCAL_LOSS needs a binary label")
+                loss = trainer_lib.LOSS_FOCAL
+            else:
+                loss = (trainer_lib.LOSS_MULTINOMIAL if n_classes > 2
+                        else trainer_lib.LOSS_BINOMIAL)
         elif self._task == Task.REGRESSION:
             named = hp.get("loss") if isinstance(hp.get("loss"), str) \
                 else "DEFAULT"
@@ -148,7 +164,11 @@ class GradientBoostedTreesLearner(GenericLearner):
                     "MEAN_AVERAGE_ERROR": trainer_lib.LOSS_MAE,
                     }.get(named, trainer_lib.LOSS_SQUARED_ERROR)
         elif self._task == Task.RANKING:
-            loss = trainer_lib.LOSS_LAMBDA_MART_NDCG
+            named = hp.get("loss") if isinstance(hp.get("loss"), str) \
+                else "DEFAULT"
+            loss = (trainer_lib.LOSS_XE_NDCG
+                    if named in ("XE_NDCG", "XE_NDCG_MART")
+                    else trainer_lib.LOSS_LAMBDA_MART_NDCG)
         else:
             raise NotImplementedError(
                 f"GBT task {self._task} not yet supported")
@@ -234,6 +254,10 @@ class GradientBoostedTreesLearner(GenericLearner):
             early_stopping_initial_iteration=(
                 hp["early_stopping_initial_iteration"]),
             cat_smooth=hp["l2_categorical_regularization"],
+            focal_gamma=hp.get("focal_loss_gamma", 2.0),
+            focal_alpha=hp.get("focal_loss_alpha", 0.5),
+            dart_dropout=(hp.get("dart_dropout", 0.01)
+                          if hp.get("forest_extraction") == "DART" else 0.0),
             **obl,
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
@@ -248,7 +272,7 @@ class GradientBoostedTreesLearner(GenericLearner):
             activation = custom_loss.activation.value \
                 if hp["apply_link_function"] else "identity"
         elif hp["apply_link_function"]:
-            if loss == trainer_lib.LOSS_BINOMIAL:
+            if loss in (trainer_lib.LOSS_BINOMIAL, trainer_lib.LOSS_FOCAL):
                 activation = "sigmoid"
             elif loss == trainer_lib.LOSS_MULTINOMIAL:
                 activation = "softmax"
@@ -342,8 +366,15 @@ class GradientBoostedTreesLearner(GenericLearner):
                 "resume_training_snapshot_interval_seconds", 1800.0),
             max_duration_seconds=hp.get(
                 "maximum_training_duration_seconds", -1.0))
-        flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"],
-                                 cat_feats=cat_feats)
+        if hp.get("forest_extraction") == "DART":
+            # HostTree.scale holds each tree's final absolute leaf scale
+            for tr in trees:
+                tr.leaf_value = tr.leaf_value * tr.scale
+            flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
+                                     cat_feats=cat_feats)
+        else:
+            flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"],
+                                     cat_feats=cat_feats)
         gains = self._feature_gains(trees, names)
         if partial is not None:
             from ydf_amd.model.forest import concat_forests

@@ -33,6 +33,8 @@ LOSS_MULTINOMIAL = 3
 LOSS_POISSON = 7
 LOSS_MAE = 8
 LOSS_LAMBDA_MART_NDCG = 9
+LOSS_FOCAL = 11            # binary focal loss (Lin et al. 2017)
+LOSS_XE_NDCG = 12          # cross-entropy NDCG (Bruch et al. 2020)
 LOSS_RF = 100  # weighted-target mode (RF/CART): not a GBT loss
 
 
@@ -52,6 +54,12 @@ class TrainerConfig:
     goss_beta: float = 0.1
     cat_smooth: float = 1.0      # l2_categorical_regularization
     n_classes: int = 2           # multinomial only
+    focal_gamma: float = 2.0     # focal loss misprediction exponent
+    focal_alpha: float = 0.5     # focal loss positive-class weight
+    # DART (Rashmi & Gilad-Bachrach 2015; reference forest_extraction=DART,
+    # gradient_boosted_trees.h:338): per-iteration dropout of existing
+    # trees with retroactive 1/(k+1) rescaling. 0 = plain MART.
+    dart_dropout: float = 0.0
     seed: int = 123456
     # RF-specific
     bootstrap: bool = False
@@ -87,6 +95,8 @@ class HostTree:
     gain: Optional[np.ndarray] = None   # [total_nodes] f32 (split gains)
     # oblique nodes: {node_idx: (attrs i32[], weights f32[], threshold)}
     oblique: Optional[dict] = None
+    # DART: final absolute leaf scale (bakes shrinkage + dropout rescales)
+    scale: float = 1.0
 
 
 def _dist_ok() -> bool:
@@ -659,6 +669,24 @@ class ForestTrainer:
             ops.update_preds(preds, self.node_ids, self.leaf_vals, shrinkage)
         return g
 
+    def route_tree(self, bins: torch.Tensor, node_ids: torch.Tensor,
+                   feat_t: torch.Tensor, bin_t: torch.Tensor,
+                   masks_t: Optional[torch.Tensor]):
+        """Routes rows through a PREVIOUSLY extracted tree given its
+        device arrays (DART dropout re-evaluation). Oblique trees are not
+        supported here (projections are per-level ephemeral)."""
+        node_ids.zero_()
+        for level in range(self.cfg.max_depth):
+            level_base = (1 << level) - 1
+            level_size = 1 << level
+            ops.update_node_ids(
+                bins, node_ids, self.arange_buf[:level_size],
+                feat_t[level_base:level_base + level_size],
+                bin_t[level_base:level_base + level_size],
+                level_base, level_size, cat_flags=self.cat_flags,
+                masks=masks_t)
+        return node_ids
+
     def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor,
                    raw: Optional[torch.Tensor] = None):
         """Routes arbitrary rows through the latest tree (device arrays).
@@ -721,14 +749,14 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 "training (host callbacks)")
         init = default_initial_predictions(custom_loss, y.cpu().numpy())
         init_preds = [init] * C
-    elif cfg.loss == LOSS_LAMBDA_MART_NDCG:
+    elif cfg.loss in (LOSS_LAMBDA_MART_NDCG, LOSS_XE_NDCG):
         if trainer.distributed:
             raise NotImplementedError(
                 "ranking is single-process for now (groups are not "
                 "row-shardable without group-aware sharding)")
         init = 0.0
         init_preds = [0.0]
-    elif cfg.loss == LOSS_BINOMIAL:
+    elif cfg.loss in (LOSS_BINOMIAL, LOSS_FOCAL):
         s = torch.stack([y.sum(), counts[0]])
         trainer._allreduce(s)
         p = (s[0] / s[1]).clamp(1e-6, 1 - 1e-6)
@@ -769,6 +797,24 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
         loss_buf = torch.zeros(2, dtype=torch.float32, device=dev)
 
     trees: List[HostTree] = []
+    dart = cfg.dart_dropout > 0.0
+    if dart:
+        if C > 1:
+            raise NotImplementedError("DART supports single-output losses")
+        if cfg.oblique_projections > 0:
+            raise NotImplementedError("DART + oblique is not supported "
+                                      "(per-level projections)")
+        if snapshot_cb is not None or start_iteration:
+            raise NotImplementedError(
+                "DART + checkpoint/resume is not supported (tree scales "
+                "change retroactively)")
+        dart_rec = []   # per tree: (feat_dev, bin_dev, masks_dev, leaf_dev)
+        dart_scale = []  # per tree: current absolute leaf scale
+        dart_rng = np.random.RandomState(cfg.seed ^ 0x5bd1e995)
+        dart_ids = torch.empty_like(trainer.node_ids)
+        dart_valid_ids = torch.empty(
+            trainer.valid_bins.shape[1], dtype=torch.int32,
+            device=dev) if has_valid else None
     logs = []
     best_loss = math.inf
     best_num_trees = 0
@@ -795,6 +841,21 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             g_np, h_np = custom_loss.gradient_and_hessian(y_np, p_np)
             custom_gh = (np.asarray(g_np, dtype=np.float32),
                          np.asarray(h_np, dtype=np.float32))
+        dropped = []
+        if dart and dart_rec:
+            dmask = dart_rng.random_sample(len(dart_rec)) < cfg.dart_dropout
+            if not dmask.any():
+                dmask[dart_rng.randint(len(dart_rec))] = True
+            dropped = list(np.nonzero(dmask)[0])
+            for ti in dropped:
+                ft, bt, mt, lv = dart_rec[ti]
+                trainer.route_tree(trainer.bins, dart_ids, ft, bt, mt)
+                preds[0].sub_(lv[dart_ids.long()], alpha=dart_scale[ti])
+                if has_valid:
+                    trainer.route_tree(trainer.valid_bins, dart_valid_ids,
+                                       ft, bt, mt)
+                    valid_preds[0].sub_(lv[dart_valid_ids.long()],
+                                        alpha=dart_scale[ti])
         try:
           for c in range(C):
             pc = preds[c]
@@ -808,6 +869,13 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             elif cfg.loss == LOSS_LAMBDA_MART_NDCG:
                 lg, lh = ranking.lambdas(pc)
                 trainer.gh.copy_(torch.stack([lg, lh], dim=1))
+            elif cfg.loss == LOSS_XE_NDCG:
+                lg, lh = ranking.xe_ndcg(pc, cfg.seed + it)
+                trainer.gh.copy_(torch.stack([lg, lh], dim=1))
+            elif cfg.loss == LOSS_FOCAL:
+                fg, fh = _focal_grad_hess(pc, y, cfg.focal_gamma,
+                                          cfg.focal_alpha)
+                trainer.gh.copy_(torch.stack([fg, fh], dim=1))
             elif multi:
                 ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
             else:
@@ -840,13 +908,36 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             trees.append(tree)
             if sample_mask is not None:
                 trainer.route_rows(trainer.bins, trainer.node_ids)
+            step_scale = cfg.shrinkage
+            if dart:
+                k = len(dropped)
+                step_scale = cfg.shrinkage / (k + 1)
+                dart_rec.append((
+                    trainer.tree_feat.clone(), trainer.tree_bin.clone(),
+                    trainer.tree_masks.clone()
+                    if trainer.tree_masks is not None else None,
+                    trainer.leaf_vals.clone()))
+                dart_scale.append(step_scale)
             ops.update_preds(pc, trainer.node_ids, trainer.leaf_vals,
-                             cfg.shrinkage)
+                             step_scale)
             if has_valid:
                 trainer.route_rows(trainer.valid_bins,
                                    trainer.valid_node_ids)
                 ops.update_preds(valid_preds[c], trainer.valid_node_ids,
-                                 trainer.leaf_vals, cfg.shrinkage)
+                                 trainer.leaf_vals, step_scale)
+            if dart and dropped:
+                k = len(dropped)
+                for ti in dropped:
+                    ft, bt, mt, lv = dart_rec[ti]
+                    new_scale = dart_scale[ti] * k / (k + 1)
+                    dart_scale[ti] = new_scale
+                    trainer.route_tree(trainer.bins, dart_ids, ft, bt, mt)
+                    preds[0].add_(lv[dart_ids.long()], alpha=new_scale)
+                    if has_valid:
+                        trainer.route_tree(trainer.valid_bins,
+                                           dart_valid_ids, ft, bt, mt)
+                        valid_preds[0].add_(lv[dart_valid_ids.long()],
+                                            alpha=new_scale)
         except KeyboardInterrupt:
             if log:
                 log(f"interrupted at iteration {it}; returning partial model")
@@ -859,7 +950,7 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             snapshot_cb(trees, it + 1, init_preds)
             t_last_snapshot = _time.monotonic()
         if has_valid:
-            if cfg.loss == LOSS_LAMBDA_MART_NDCG:
+            if cfg.loss in (LOSS_LAMBDA_MART_NDCG, LOSS_XE_NDCG):
                 vloss = -valid_ranking.ndcg(valid_preds[0]) \
                     if valid_ranking is not None else float("nan")
             elif custom_loss is not None and custom_loss.loss is not None:
@@ -890,10 +981,49 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
         # trees holds only THIS run's trees; best_num_trees is global
         keep = max(0, best_num_trees - start_iteration * C)
         trees = trees[:keep]
+    if dart:
+        # bake each tree's final absolute scale into HostTree.scale
+        # (the learner builds the flat forest with leaf_scale=1.0)
+        for t, sc in zip(trees, dart_scale):
+            t.scale = float(sc)
     return trees, init_preds, logs
 
 
+def _focal_grad_hess(m: torch.Tensor, y: torch.Tensor, gamma: float,
+                     alpha: float):
+    """Binary focal loss FL = -a_t (1-p_t)^g log(p_t) (reference
+    loss_imp_binary_focal.cc; gradients w.r.t. the margin, our g = dL/dm
+    convention). Vectorized torch — runs on the training device."""
+    s = 2.0 * y - 1.0                      # ±1
+    mp = s * m
+    log_pt = torch.nn.functional.logsigmoid(mp)
+    pt = torch.sigmoid(mp).clamp(1e-7, 1 - 1e-7)
+    at = alpha * y + (1.0 - alpha) * (1.0 - y)
+    u = (1.0 - pt) ** gamma
+    v = gamma * pt * log_pt - (1.0 - pt)
+    g = s * at * u * v
+    h = at * u * (-gamma * pt * v
+                  + pt * (1.0 - pt) * (gamma * log_pt + gamma + 1.0))
+    return g, h.clamp(1e-6, 16.0)
+
+
+def _focal_loss_value(m, y, gamma, alpha):
+    s = 2.0 * y - 1.0
+    log_pt = torch.nn.functional.logsigmoid(s * m)
+    pt = torch.sigmoid(s * m)
+    at = alpha * y + (1.0 - alpha) * (1.0 - y)
+    return -(at * (1.0 - pt) ** gamma * log_pt)
+
+
 def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
+    if cfg.loss == LOSS_FOCAL:
+        per = _focal_loss_value(preds[0], labels, cfg.focal_gamma,
+                                cfg.focal_alpha)
+        s = torch.stack([per.sum(),
+                         torch.tensor(float(labels.numel()),
+                                      device=preds.device)])
+        trainer._allreduce(s)
+        return float((s[0] / s[1]).item())
     if cfg.loss == LOSS_BINOMIAL:
         loss_buf.zero_()
         ops.binary_logloss(preds[0], labels, loss_buf)
