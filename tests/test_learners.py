@@ -558,3 +558,44 @@ def test_dart():
         num_trees=60, forest_extraction="DART", dart_dropout=0.1,
         validation_ratio=0.1, label="label").train(d)
     assert m.evaluate(d).accuracy > 0.98
+
+
+def test_rf_honest_trees():
+    """Honest trees (reference Honest message, decision_tree.proto): leaf
+    values from held-out rows; quality stays competitive."""
+    rng = np.random.RandomState(3)
+    n = 4000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2,
+         "label": np.where(2 * x1 - x2 + 0.3 * rng.randn(n) > 0, "a", "b")}
+    m = ydf.RandomForestLearner(label="label", num_trees=25, max_depth=8,
+                                honest=True).train(d)
+    assert m.evaluate(d).accuracy > 0.93
+    # fixed separation reuses one honest split across trees
+    m2 = ydf.RandomForestLearner(label="label", num_trees=10, max_depth=6,
+                                 honest=True,
+                                 honest_fixed_separation=True).train(d)
+    assert m2.evaluate(d).accuracy > 0.9
+
+
+def test_rf_oob_permutation_importances():
+    """OOB permutation variable importances (reference
+    random_forest.cc:1411): informative features must outrank noise."""
+    rng = np.random.RandomState(4)
+    n = 3000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    noise = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2, "noise": noise,
+         "label": np.where(2 * x1 - x2 + 0.3 * rng.randn(n) > 0, "a", "b")}
+    m = ydf.RandomForestLearner(
+        label="label", num_trees=20, max_depth=8,
+        compute_oob_variable_importances=True).train(d)
+    vi = m.variable_importances()["MEAN_DECREASE_IN_ACCURACY"]
+    ranked = [name for _, name in vi]
+    assert ranked[0] == "x1"
+    assert ranked[-1] == "noise"
+    scores = {name: s for s, name in vi}
+    assert scores["x1"] > 0.1
+    assert abs(scores["noise"]) < 0.05

@@ -426,6 +426,10 @@ class RandomForestLearner(GenericLearner):
                  num_candidate_attributes_ratio: float = -1.0,
                  winner_take_all: bool = True,
                  compute_oob_performances: bool = True,
+                 compute_oob_variable_importances: bool = False,
+                 honest: bool = False,
+                 honest_ratio_leaf_examples: float = 0.5,
+                 honest_fixed_separation: bool = False,
                  split_axis: str = "AXIS_ALIGNED",
                  sparse_oblique_num_projections_exponent: float = 2.0,
                  sparse_oblique_max_num_projections: int = 6000,
@@ -444,6 +448,11 @@ class RandomForestLearner(GenericLearner):
             num_candidate_attributes_ratio=num_candidate_attributes_ratio,
             winner_take_all=winner_take_all,
             compute_oob_performances=compute_oob_performances,
+            compute_oob_variable_importances=(
+                compute_oob_variable_importances),
+            honest=honest,
+            honest_ratio_leaf_examples=honest_ratio_leaf_examples,
+            honest_fixed_separation=honest_fixed_separation,
             split_axis=split_axis,
             sparse_oblique_num_projections_exponent=(
                 sparse_oblique_num_projections_exponent),
@@ -479,11 +488,6 @@ class RandomForestLearner(GenericLearner):
             data, device)
         if labels is None:
             raise ValueError(f"label column {self.label!r} missing")
-        obl = self._oblique_cfg(bins.shape[0], cat_flags)
-        raw_t = valid_raw_t = None
-        if obl:
-            raw_t = torch.from_numpy(
-                np.ascontiguousarray(ds.X)).to(device)
         classes = self._label_classes(ds) \
             if self._task == Task.CLASSIFICATION else None
         n_classes = len(classes) if classes else 2
@@ -506,6 +510,10 @@ class RandomForestLearner(GenericLearner):
             n_classes=n_classes, seed=self.random_seed,
             bootstrap=hp["bootstrap_training_dataset"],
             num_candidate_features=ncand,
+            honest=hp.get("honest", False),
+            honest_ratio=hp.get("honest_ratio_leaf_examples", 0.5),
+            honest_fixed_separation=hp.get("honest_fixed_separation",
+                                           False),
             **obl,
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
@@ -552,7 +560,75 @@ class RandomForestLearner(GenericLearner):
                 trees, [c.name for c in ds.dataspec.feature_columns]),
                 "winner_take_all": wta})
         model._self_evaluation = oob_eval
+        if hp.get("compute_oob_variable_importances") \
+                and hp["bootstrap_training_dataset"]:
+            vi = _oob_permutation_vi(model, ds, cfg, self._task, device)
+            if vi:
+                model.metadata["oob_permutation_importances"] = vi
         return model
+
+
+def _oob_permutation_vi(model, ds, cfg, task, device):
+    """Per-tree OOB permutation variable importances (reference
+    random_forest.cc:1411-1477 ComputeVariableImportancesFromAccumulated):
+    each tree is evaluated on ITS out-of-bag rows (bootstrap weights
+    regenerated from the per-tree seed) with each feature permuted in
+    turn; the metric drop is averaged over trees. Binary classification
+    -> MEAN_DECREASE_IN_ACCURACY (+ MEAN_DECREASE_IN_AUC omitted);
+    regression -> MEAN_INCREASE_IN_RMSE."""
+    from ydf_amd.learner import trainer as trainer_lib
+    from ydf_amd.model.generic_model import _DeviceForest
+
+    if task == Task.CLASSIFICATION and model.label_classes \
+            and len(model.label_classes) > 2:
+        return None  # multi-class per-tree votes: not supported yet
+    X = np.ascontiguousarray(ds.X)
+    y = ds.label_values.astype(np.float32)
+    F, N = X.shape
+    T = model.forest.n_trees
+    names = model.input_feature_names()
+    cpu = torch.device("cpu")
+    df = _DeviceForest(model.forest, cpu)
+    rng = np.random.RandomState(cfg.seed ^ 0x00bafeed)
+    acc_drop = np.zeros(F, dtype=np.float64)
+    n_used = 0
+
+    def tree_metric(Xsub_t, yb, t):
+        out = torch.empty(Xsub_t.shape[1], dtype=torch.float32)
+        from ydf_amd import ops as _ops
+
+        _ops.predict_forest(Xsub_t, df.feat, df.thr, df.left, df.roots,
+                            out, tree_start=t, tree_step=1, n_trees=1,
+                            cat_idx=df.cat_idx, masks=df.masks,
+                            obl_ranges=df.obl_ranges, obl_attr=df.obl_attr,
+                            obl_w=df.obl_w)
+        p = out.numpy()
+        if task == Task.REGRESSION:
+            return -float(np.sqrt(np.mean((p - yb) ** 2)))  # higher=better
+        return float(((p > 0.5) == (yb > 0.5)).mean())
+
+    for t in range(T):
+        w = trainer_lib.rf_bootstrap_weights(cfg.seed, t, N, device)
+        oob = (w == 0).cpu().numpy()
+        if oob.sum() < 10:
+            continue
+        n_used += 1
+        Xs = np.ascontiguousarray(X[:, oob])
+        yb = y[oob]
+        Xt = torch.from_numpy(Xs)
+        base = tree_metric(Xt, yb, t)
+        for f in range(F):
+            saved = Xs[f].copy()
+            Xs[f] = saved[rng.permutation(len(saved))]
+            acc_drop[f] += base - tree_metric(Xt, yb, t)
+            Xs[f] = saved
+    if n_used == 0:
+        return None
+    acc_drop /= n_used
+    key = "MEAN_INCREASE_IN_RMSE" if task == Task.REGRESSION \
+        else "MEAN_DECREASE_IN_ACCURACY"
+    order = np.argsort(-acc_drop, kind="stable")
+    return {key: [[float(acc_drop[i]), names[i]] for i in order]}
 
 
 class CartLearner(RandomForestLearner):

@@ -64,6 +64,11 @@ class TrainerConfig:
     # RF-specific
     bootstrap: bool = False
     num_candidate_features: int = 0  # 0 = all features
+    # honest trees (reference decision_tree.proto Honest message): tree
+    # structure from one random half, leaf values re-estimated on the other
+    honest: bool = False
+    honest_ratio: float = 0.5        # fraction reserved for leaf values
+    honest_fixed_separation: bool = False
     # early stopping (GBT; reference gradient_boosted_trees.proto:151-172)
     early_stopping: bool = False
     early_stopping_num_trees_look_ahead: int = 30
@@ -1054,6 +1059,32 @@ def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
     return float((s[0] / s[1]).item())
 
 
+def rf_bootstrap_weights(seed: int, tree_idx: int, N: int,
+                         dev) -> torch.Tensor:
+    """Poisson(1) bootstrap draw for tree `tree_idx`, clipped at 15.
+    Seeded per tree so the draw can be REGENERATED after training (OOB
+    permutation importances re-derive each tree's out-of-bag rows)."""
+    if dev.type == "cuda":
+        g = torch.Generator(device=dev)
+        g.manual_seed((seed * 31337 + tree_idx) % (1 << 31))
+        return torch.poisson(torch.ones(N, device=dev),
+                             generator=g).clamp_(max=15)
+    rs = np.random.RandomState((seed * 31337 + tree_idx) % (1 << 31))
+    w = np.minimum(rs.poisson(1.0, size=N), 15).astype(np.float32)
+    return torch.from_numpy(w).to(dev)
+
+
+def honest_split_mask(seed: int, tree_idx: int, N: int, ratio: float,
+                      fixed: bool, dev) -> torch.Tensor:
+    """True = row reserved for LEAF-VALUE estimation (reference Honest
+    message, decision_tree.proto; fixed_separation reuses one split for
+    every tree)."""
+    t = 0 if fixed else tree_idx
+    rs = np.random.RandomState((seed * 7919 + t * 104729 + 13) % (1 << 31))
+    m = rs.random_sample(N) < ratio
+    return torch.from_numpy(m).to(dev)
+
+
 def train_rf(trainer: ForestTrainer, log=None,
              compute_oob: bool = False):
     """Random-forest bagging loop (reference random_forest.cc:917).
@@ -1083,28 +1114,48 @@ def train_rf(trainer: ForestTrainer, log=None,
         if cfg.bootstrap:
             # Poisson(1) bootstrap, clipped at 15 (P < 1e-12) — the packed
             # u64 histogram path requires per-example h <= 16
-            if dev.type == "cuda":
-                g = torch.Generator(device=dev)
-                g.manual_seed((cfg.seed * 31337 + it) % (1 << 31))
-                weights = torch.poisson(
-                    torch.ones(N, device=dev), generator=g).clamp_(max=15)
-            else:
-                w = np.minimum(trainer.rng.poisson(1.0, size=N), 15).astype(
-                    np.float32)
-                weights = torch.from_numpy(w).to(dev)
+            weights = rf_bootstrap_weights(cfg.seed, it, N, dev)
         if trainer.weights is not None:
             # user example weights compose with the bootstrap draw counts
             weights = trainer.weights if weights is None \
                 else (weights * trainer.weights).clamp_(max=15)
+        hmask = None
+        if cfg.honest:
+            hmask = honest_split_mask(cfg.seed, it, N, cfg.honest_ratio,
+                                      cfg.honest_fixed_separation, dev)
         for c in range(C):
             if multi:
                 if onehot is None:
                     onehot = torch.empty(N, dtype=torch.float32, device=dev)
                 onehot.copy_((trainer.labels == c).float())
-                ops.weighted_target(onehot, weights, trainer.gh)
+                target = onehot
             else:
-                ops.weighted_target(trainer.labels, weights, trainer.gh)
+                target = trainer.labels
+            if hmask is None:
+                ops.weighted_target(target, weights, trainer.gh)
+            else:
+                # structure set only: leaf-estimation rows get weight 0
+                sw = torch.where(hmask, torch.zeros((), device=dev),
+                                 torch.ones((), device=dev))
+                if weights is not None:
+                    sw = sw * weights
+                ops.weighted_target(target, sw, trainer.gh)
             tree = trainer.grow_tree(it * C + c)
+            if hmask is not None:
+                # re-estimate leaf values from the held-out half
+                # (node_ids hold every row's final leaf)
+                ids = trainer.node_ids.long()
+                w_est = hmask.float() if weights is None \
+                    else hmask.float() * weights
+                num = torch.zeros_like(trainer.leaf_vals)
+                den = torch.zeros_like(trainer.leaf_vals)
+                num.scatter_add_(0, ids, target * w_est)
+                den.scatter_add_(0, ids, w_est)
+                est = num / den.clamp(min=1.0)
+                # leaves with no estimation rows keep the structure value
+                trainer.leaf_vals.copy_(
+                    torch.where(den > 0, est, trainer.leaf_vals))
+                tree = trainer.extract_host_tree()
             trees.append(tree)
             if oob_sum is not None:
                 # node_ids already hold every row's leaf (zero-weight rows

@@ -82,6 +82,11 @@ def structure_importances(model) -> Dict[str, List[Tuple[float, str]]]:
     if gains:
         out["SUM_SCORE"] = sorted(
             ((float(v), k) for k, v in gains.items() if v > 0), reverse=True)
+    oob_vi = model.metadata.get("oob_permutation_importances") \
+        if model.metadata else None
+    if oob_vi:
+        for metric, ranked_list in oob_vi.items():
+            out[metric] = [(float(s), n) for s, n in ranked_list]
     return out
 
 
