@@ -56,3 +56,46 @@ def test_embed_cpp_codegen(binary_data, tmp_path):
         p = lib.gen_predict_c(
             row.ctypes.data_as(ctypes.POINTER(ctypes.c_float)))
         assert abs(p - ref[i]) < 1e-5
+
+
+def test_model_comparison_and_cis():
+    """compare_models (reference metric/comparison.h) + closed-form and
+    bootstrap confidence intervals (metric.h:150-177)."""
+    import ydf_amd as ydf
+    from ydf_amd.metric.metric import bootstrap_confidence_intervals
+
+    rng = np.random.RandomState(0)
+    n = 4000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2,
+         "label": np.where(2 * x1 - x2 + 0.5 * rng.randn(n) > 0, "a", "b")}
+    weak = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=2, validation_ratio=0).train(d)
+    strong = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=50, validation_ratio=0).train(d)
+    cmp = ydf.compare_models(weak, strong, d)
+    assert cmp.p_value < 0.01  # strong significantly better
+    assert cmp.metrics_2["accuracy"] > cmp.metrics_1["accuracy"]
+    ev = strong.evaluate(d)
+    lo, hi = ev.accuracy_ci95
+    assert lo < ev.accuracy < hi
+    lo, hi = ev.auc_ci95
+    assert lo < ev.auc <= hi
+    y = (np.asarray(d["label"]) == strong.label_classes[1]).astype(
+        np.float32)
+    cis = bootstrap_confidence_intervals(y, strong.predict(d),
+                                         ydf.Task.CLASSIFICATION,
+                                         n_samples=30)
+    assert cis["accuracy"][0] <= ev.accuracy <= cis["accuracy"][1]
+    # regression comparison path
+    yr = (2 * x1 - x2).astype(np.float32)
+    dr = {"x1": x1, "x2": x2, "label": yr}
+    w = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=2,
+        validation_ratio=0).train(dr)
+    s = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=50,
+        validation_ratio=0).train(dr)
+    cmp = ydf.compare_models(w, s, dr)
+    assert cmp.p_value < 0.01

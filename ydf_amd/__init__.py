@@ -85,6 +85,8 @@ class Monotonic:
     DECREASING = -1
 
 # Metric
+from ydf_amd.metric.comparison import ModelComparison, compare_models
+from ydf_amd.model.sklearn_io import from_sklearn
 from ydf_amd.metric.metric import Evaluation, evaluate_predictions
 
 # Utilities

@@ -1,6 +1,7 @@
 """Evaluation metrics (capability analogue of the reference metric layer,
 yggdrasil_decision_forests/metric/metric.h:42-177: accuracy, AUC, PR-AUC,
-log-loss, RMSE/MAE, confusion matrix; bootstrap CIs to come)."""
+log-loss, RMSE/MAE, NDCG/MRR, confusion matrix, closed-form + bootstrap
+confidence intervals)."""
 from __future__ import annotations
 
 import dataclasses
@@ -101,6 +102,78 @@ def ndcg(labels: np.ndarray, scores: np.ndarray, groups: np.ndarray,
     return total / n_groups if n_groups else float("nan")
 
 
+def mrr(labels: np.ndarray, scores: np.ndarray, groups: np.ndarray,
+        truncation: int = 5) -> float:
+    """Mean reciprocal rank of the first relevant (label > 0) item
+    within the truncation (reference metric/ranking_mrr.h)."""
+    total, n_groups = 0.0, 0
+    for g in np.unique(groups):
+        m = groups == g
+        rel = labels[m]
+        if not (rel > 0).any():
+            continue
+        order = np.argsort(-scores[m], kind="mergesort")[:truncation]
+        hit = np.nonzero(rel[order] > 0)[0]
+        total += 1.0 / (hit[0] + 1) if len(hit) else 0.0
+        n_groups += 1
+    return total / n_groups if n_groups else float("nan")
+
+
+def accuracy_confidence_interval(acc: float, n: int,
+                                 level: float = 0.95):
+    """Wilson score interval (closed form; reference
+    metric.h ComputeXsLossConfidenceInterval family)."""
+    if n == 0:
+        return (float("nan"), float("nan"))
+    from scipy.stats import norm
+
+    z = norm.ppf(0.5 + level / 2.0)
+    denom = 1.0 + z * z / n
+    center = (acc + z * z / (2 * n)) / denom
+    half = z * np.sqrt(acc * (1 - acc) / n + z * z / (4 * n * n)) / denom
+    return (float(center - half), float(center + half))
+
+
+def auc_confidence_interval(auc: float, n_pos: int, n_neg: int,
+                            level: float = 0.95):
+    """Hanley-McNeil closed-form AUC standard error."""
+    if n_pos == 0 or n_neg == 0 or not np.isfinite(auc):
+        return (float("nan"), float("nan"))
+    from scipy.stats import norm
+
+    q1 = auc / (2 - auc)
+    q2 = 2 * auc * auc / (1 + auc)
+    se = np.sqrt((auc * (1 - auc) + (n_pos - 1) * (q1 - auc * auc)
+                  + (n_neg - 1) * (q2 - auc * auc)) / (n_pos * n_neg))
+    z = norm.ppf(0.5 + level / 2.0)
+    return (float(max(0.0, auc - z * se)), float(min(1.0, auc + z * se)))
+
+
+def bootstrap_confidence_intervals(labels: np.ndarray,
+                                   predictions: np.ndarray, task,
+                                   n_samples: int = 500,
+                                   level: float = 0.95,
+                                   seed: int = 1234):
+    """Percentile-bootstrap CIs for the task's main metrics (reference
+    bootstrap CIs, metric.h:150-177). Returns {metric: (lo, hi)}."""
+    from ydf_amd.dataset.dataspec import Task
+
+    rng = np.random.RandomState(seed)
+    n = len(labels)
+    stats = {}
+    for _ in range(n_samples):
+        idx = rng.randint(0, n, n)
+        ev = evaluate_predictions(predictions[idx], labels[idx], task)
+        for k in ("accuracy", "auc", "rmse", "mae", "loss"):
+            v = getattr(ev, k)
+            if v is not None and np.isfinite(v):
+                stats.setdefault(k, []).append(v)
+    lo_q, hi_q = 100 * (0.5 - level / 2), 100 * (0.5 + level / 2)
+    return {k: (float(np.percentile(v, lo_q)),
+                float(np.percentile(v, hi_q)))
+            for k, v in stats.items() if len(v) > 1}
+
+
 @dataclasses.dataclass
 class Evaluation:
     """Evaluation report (mirrors ydf.metric.Evaluation fields)."""
@@ -113,11 +186,16 @@ class Evaluation:
     rmse: Optional[float] = None
     mae: Optional[float] = None
     ndcg: Optional[float] = None
+    mrr: Optional[float] = None
     confusion: Optional[np.ndarray] = None
+    # closed-form 95% confidence intervals (lo, hi)
+    accuracy_ci95: Optional[tuple] = None
+    auc_ci95: Optional[tuple] = None
 
     def to_dict(self) -> Dict:
         d = {"num_examples": self.num_examples}
-        for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae", "ndcg"):
+        for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae",
+                  "ndcg", "mrr", "accuracy_ci95", "auc_ci95"):
             v = getattr(self, k)
             if v is not None:
                 d[k] = v
@@ -125,10 +203,15 @@ class Evaluation:
 
     def __str__(self) -> str:
         parts = [f"num examples: {self.num_examples}"]
-        for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae", "ndcg"):
+        for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae",
+                  "ndcg", "mrr"):
             v = getattr(self, k)
             if v is not None:
                 parts.append(f"{k}: {v:.6g}")
+        for k in ("accuracy_ci95", "auc_ci95"):
+            v = getattr(self, k)
+            if v is not None:
+                parts.append(f"{k}: [{v[0]:.6g}, {v[1]:.6g}]")
         return "\n".join(parts)
 
     def _repr_html_(self) -> str:
@@ -156,6 +239,12 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
         ev.loss = log_loss(labels, predictions)
         ev.confusion = confusion_matrix(labels.astype(np.int64), pred_cls,
                                         n_classes)
+        ev.accuracy_ci95 = accuracy_confidence_interval(
+            ev.accuracy, len(labels))
+        if ev.auc is not None and np.isfinite(ev.auc):
+            n_pos = int((labels > 0.5).sum())
+            ev.auc_ci95 = auc_confidence_interval(
+                ev.auc, n_pos, len(labels) - n_pos)
     elif task == Task.REGRESSION:
         ev.rmse = rmse(labels, predictions)
         ev.mae = mae(labels, predictions)
