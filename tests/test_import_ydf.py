@@ -156,3 +156,47 @@ def test_oblique_embed_cpp(tmp_path):
     got = np.array([float(v) for v in out.stdout.split()])
     want = m.predict(te, device="cpu")
     np.testing.assert_allclose(got, want, atol=2e-5)
+
+
+def test_export_roundtrip_rf(tmp_path, binary_data):
+    """RF export to the reference format (random_forest_header.pb +
+    classifier-distribution leaves), bit-exact round-trip, both
+    winner-take-all settings."""
+    for wta in (True, False):
+        m = ydf.RandomForestLearner(
+            label="label", num_trees=8, max_depth=8, winner_take_all=wta,
+            compute_oob_performances=False).train(binary_data)
+        p = str(tmp_path / f"rf{wta}")
+        ydf.export_ydf_model(m, p)
+        m2 = ydf.load_ydf_model(p)
+        np.testing.assert_array_equal(m.predict(binary_data, device="cpu"),
+                                      m2.predict(binary_data, device="cpu"))
+
+
+def test_export_roundtrip_rf_regression(tmp_path):
+    rng = np.random.RandomState(1)
+    x = rng.randn(2000).astype(np.float32)
+    d = {"x": x, "y": rng.randn(2000).astype(np.float32),
+         "label": (2 * x).astype(np.float32)}
+    m = ydf.RandomForestLearner(label="label", task=ydf.Task.REGRESSION,
+                                num_trees=8,
+                                compute_oob_performances=False).train(d)
+    p = str(tmp_path / "rfreg")
+    ydf.export_ydf_model(m, p)
+    np.testing.assert_array_equal(m.predict(d, device="cpu"),
+                                  ydf.load_ydf_model(p).predict(
+                                      d, device="cpu"))
+
+
+def test_export_roundtrip_oblique(tmp_path, binary_data):
+    """Oblique conditions survive the reference wire format."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=10, max_depth=3, validation_ratio=0,
+        split_axis="SPARSE_OBLIQUE").train(binary_data)
+    assert (m.forest.cat_idx <= -2).sum() > 0
+    p = str(tmp_path / "obl")
+    ydf.export_ydf_model(m, p)
+    m2 = ydf.load_ydf_model(p)
+    assert (m2.forest.cat_idx <= -2).sum() > 0
+    np.testing.assert_array_equal(m.predict(binary_data, device="cpu"),
+                                  m2.predict(binary_data, device="cpu"))

@@ -91,8 +91,18 @@ def encode_data_spec(dataspec) -> bytes:
 
 # --- nodes blob sequence ---------------------------------------------------
 def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
-                      is_bool: bool) -> bytes:
-    if is_cat:
+                      is_bool: bool, oblique=None, cover: float = 0.0
+                      ) -> bytes:
+    if oblique is not None:
+        # Oblique (decision_tree.proto:114-131): attributes=1 packed,
+        # weights=2 packed f32, threshold=3; semantics sum >= threshold
+        attrs, ws = oblique
+        t = float(np.nextafter(np.float32(thr), np.float32("inf")))
+        packed_attrs = b"".join(_varint(int(a)) for a in attrs)
+        packed_ws = struct.pack(f"<{len(ws)}f", *[float(w) for w in ws])
+        inner = f_msg(7, f_bytes(1, packed_attrs) + f_bytes(2, packed_ws)
+                      + f_float(3, t))
+    elif is_cat:
         bm = np.asarray(mask, dtype=np.uint64).view(np.uint8).tobytes()
         inner = f_msg(5, f_bytes(1, bm))           # ContainsBitmap
     elif is_bool:
@@ -103,36 +113,52 @@ def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
         t = float(np.nextafter(np.float32(thr), np.float32("inf")))
         inner = f_msg(2, f_float(1, t))             # Higher
     cond = f_varint(2, feat) + f_msg(3, inner)      # NodeCondition
+    cond += f_varint(4, max(0, int(cover)))         # n examples (unweighted)
+    cond += f_double(5, float(cover))               # n examples (weighted)
     return f_msg(3, cond)                           # Node.condition
 
 
-def encode_forest_nodes(model) -> List[bytes]:
-    """Pre-order Node records per tree (negative child first)."""
+def encode_forest_nodes(model, classifier_leaves: bool = False
+                        ) -> List[bytes]:
+    """Pre-order Node records per tree (negative child first). With
+    classifier_leaves (RF classification), leaves carry a class
+    distribution whose counts reproduce our leaf probability."""
     f = model.forest
-    task = model.task()
-    classification_leaf = False  # GBT leaves are regressor values
     bool_feats = set()
-    cat_feats = set()
     for i, c in enumerate(model.dataspec.feature_columns):
         if c.semantic == Semantic.BOOLEAN:
             bool_feats.add(i)
-        elif c.semantic == Semantic.CATEGORICAL:
-            cat_feats.add(i)
 
     records: List[bytes] = []
 
     def emit(n: int):
         body = b""
         if f.feat[n] < 0:
-            body += f_msg(2, f_float(1, float(f.thr[n])))  # regressor leaf
+            if classifier_leaves:
+                # counts over [OOV, class1, class2]: p(class2) = thr
+                p = float(np.clip(f.thr[n], 0.0, 1.0))
+                total = float(f.cover[n]) if f.cover[n] > 0 else 1.0
+                counts = struct.pack("<3d", 0.0, (1.0 - p) * total,
+                                     p * total)
+                dist = f_bytes(1, counts) + f_double(2, total)
+                cls = f_varint(1, 2 if p > 0.5 else 1) + f_msg(2, dist)
+                body += f_msg(1, cls)               # Node.classifier
+            else:
+                body += f_msg(2, f_float(1, float(f.thr[n])))
         else:
             body += f_msg(2, f_float(1, 0.0))
             fi = int(f.feat[n])
             ci = int(f.cat_idx[n])
+            obl = None
+            if ci <= -2:
+                oi = -(ci + 2)
+                s0, nn = int(f.obl_ranges[oi, 0]), int(f.obl_ranges[oi, 1])
+                obl = (f.obl_attr[s0:s0 + nn], f.obl_w[s0:s0 + nn])
             body += _encode_condition(
                 fi, float(f.thr[n]),
                 f.masks[ci] if ci >= 0 else None,
-                ci >= 0, fi in bool_feats and ci < 0)
+                ci >= 0, fi in bool_feats and ci == -1, oblique=obl,
+                cover=float(f.cover[n]))
         records.append(body)
         if f.feat[n] >= 0:
             left = int(f.left[n])
@@ -159,16 +185,26 @@ _LOSS = {"sigmoid": 1, "softmax": 3, "identity": 2}
 
 
 def export_ydf_model(model, path: str) -> None:
-    """Writes `model` as a reference-format model directory."""
-    from ydf_amd.model.specialized import GradientBoostedTreesModel
+    """Writes `model` as a reference-format model directory (GBT and
+    RF — binary classification and regression)."""
+    from ydf_amd.model.specialized import (GradientBoostedTreesModel,
+                                           RandomForestModel)
 
-    if not isinstance(model, GradientBoostedTreesModel):
+    is_gbt = isinstance(model, GradientBoostedTreesModel)
+    is_rf = isinstance(model, RandomForestModel) and not is_gbt
+    if not (is_gbt or is_rf):
         raise NotImplementedError(
-            "export to the reference format currently supports GBT models")
+            "export to the reference format supports GBT and RF models")
+    if is_rf and model._n_outputs() > 1:
+        raise NotImplementedError(
+            "RF export supports binary classification / regression "
+            "(our multi-class RF stores one tree per class, which the "
+            "reference single-tree-distribution format cannot express)")
     os.makedirs(path, exist_ok=True)
     label_idx = len(model.dataspec.columns) - 1
     # AbstractModel: name=1, task=2, label_col_idx=3, input_features=5
-    header = f_str(1, "GRADIENT_BOOSTED_TREES")
+    header = f_str(1, "GRADIENT_BOOSTED_TREES" if is_gbt
+                   else "RANDOM_FOREST")
     header += f_varint(2, _TASK.get(model.task(), 1))
     header += f_varint(3, label_idx)
     for i in range(label_idx):
@@ -177,18 +213,32 @@ def export_ydf_model(model, path: str) -> None:
         fp.write(header)
     with open(os.path.join(path, "data_spec.pb"), "wb") as fp:
         fp.write(encode_data_spec(model.dataspec))
-    # GBT header: num_trees=2, loss=3, initial_predictions=4,
-    # num_trees_per_iter=5, node_format=7
-    gh = f_varint(2, model.forest.n_trees)
-    gh += f_varint(3, _LOSS.get(model.activation, 2))
-    for v in model.init_predictions:
-        gh += f_float(4, v)
-    gh += f_varint(5, model.num_trees_per_iter)
-    gh += f_str(7, "BLOB_SEQUENCE")
-    with open(os.path.join(path, "gradient_boosted_trees_header.pb"),
-              "wb") as fp:
-        fp.write(gh)
-    write_blob_sequence(os.path.join(path, "nodes-00000-of-00001"),
-                        encode_forest_nodes(model))
+    if is_gbt:
+        # GBT header: num_trees=2, loss=3, initial_predictions=4,
+        # num_trees_per_iter=5, node_format=7
+        gh = f_varint(2, model.forest.n_trees)
+        gh += f_varint(3, _LOSS.get(model.activation, 2))
+        for v in model.init_predictions:
+            gh += f_float(4, v)
+        gh += f_varint(5, model.num_trees_per_iter)
+        gh += f_str(7, "BLOB_SEQUENCE")
+        with open(os.path.join(path, "gradient_boosted_trees_header.pb"),
+                  "wb") as fp:
+            fp.write(gh)
+        classifier_leaves = False
+    else:
+        # RF header (random_forest.proto): num_trees=2,
+        # winner_take_all_inference=3, node_format=6
+        wta = bool((model.metadata or {}).get("winner_take_all", False))
+        rh = f_varint(2, model.forest.n_trees)
+        rh += f_varint(3, 1 if wta else 0)
+        rh += f_str(6, "BLOB_SEQUENCE")
+        with open(os.path.join(path, "random_forest_header.pb"),
+                  "wb") as fp:
+            fp.write(rh)
+        classifier_leaves = model.task() == Task.CLASSIFICATION
+    write_blob_sequence(
+        os.path.join(path, "nodes-00000-of-00001"),
+        encode_forest_nodes(model, classifier_leaves=classifier_leaves))
     with open(os.path.join(path, "done"), "w") as fp:
         fp.write("")
