@@ -187,6 +187,8 @@ class Evaluation:
     mae: Optional[float] = None
     ndcg: Optional[float] = None
     mrr: Optional[float] = None
+    auuc: Optional[float] = None
+    qini: Optional[float] = None
     confusion: Optional[np.ndarray] = None
     # closed-form 95% confidence intervals (lo, hi)
     accuracy_ci95: Optional[tuple] = None
@@ -195,7 +197,8 @@ class Evaluation:
     def to_dict(self) -> Dict:
         d = {"num_examples": self.num_examples}
         for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae",
-                  "ndcg", "mrr", "accuracy_ci95", "auc_ci95"):
+                  "ndcg", "mrr", "auuc", "qini", "accuracy_ci95",
+                  "auc_ci95"):
             v = getattr(self, k)
             if v is not None:
                 d[k] = v
@@ -204,7 +207,7 @@ class Evaluation:
     def __str__(self) -> str:
         parts = [f"num examples: {self.num_examples}"]
         for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae",
-                  "ndcg", "mrr"):
+                  "ndcg", "mrr", "auuc", "qini"):
             v = getattr(self, k)
             if v is not None:
                 parts.append(f"{k}: {v:.6g}")

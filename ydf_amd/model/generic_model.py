@@ -220,6 +220,24 @@ class GenericModel:
             else:
                 labels = np.asarray(cols[lname], dtype=np.float32)
         n_classes = len(self.label_classes) if self.label_classes else 2
+        if self._task in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):
+            from ydf_amd.metric.uplift import auuc_qini
+
+            tcol = (self.metadata or {}).get("uplift_treatment")
+            if cols is None or tcol not in cols:
+                raise ValueError(
+                    f"uplift evaluation needs the treatment column "
+                    f"{tcol!r} in the dataset")
+            treat = np.asarray(cols[tcol])
+            if treat.dtype.kind in "UOS":
+                tvocab = (self.metadata or {}).get("treatment_vocab")
+                pos = tvocab[1] if tvocab and len(tvocab) > 1 else None
+                treat = (treat.astype(str) == pos).astype(np.float32) \
+                    if pos is not None else (treat != treat[0]).astype(
+                        np.float32)
+            ev = Evaluation(num_examples=len(labels))
+            ev.auuc, ev.qini = auuc_qini(labels, treat, preds)
+            return ev
         ev = evaluate_predictions(preds, labels, self._task, n_classes)
         if self._task == Task.RANKING:
             gcol = (self.metadata or {}).get("ranking_group")
