@@ -254,11 +254,12 @@ def test_oblique_predict_gpu_vs_cpu():
     X = rng.randn(F, N).astype(np.float32)
     from ydf_amd.model.forest import FlatForest
 
-    # tree: oblique root (0.5*x0 - x2 + 2*x4 > 0.1), numerical children
+    # tree: oblique root (0.5*x0 - x2 + 2*x4 > 0.1); children split on
+    # x1 (no branch) / x3 (yes branch); leaves 1..4
     forest = FlatForest(
-        feat=np.array([0, 1, -1, -1, 3, -1, -1], np.int32),
-        thr=np.array([0.1, 0.0, 1.0, 2.0, -0.3, 3.0, 4.0], np.float32),
-        left=np.array([1, 2, 0, 0, 5, 0, 0], np.int32),
+        feat=np.array([0, 1, 3, -1, -1, -1, -1], np.int32),
+        thr=np.array([0.1, 0.0, -0.3, 1.0, 2.0, 3.0, 4.0], np.float32),
+        left=np.array([1, 3, 5, 0, 0, 0, 0], np.int32),
         roots=np.array([0], np.int32),
         cat_idx=np.array([-2, -1, -1, -1, -1, -1, -1], np.int32),
         obl_ranges=np.array([[0, 3]], np.int32),
