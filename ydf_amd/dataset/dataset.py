@@ -33,7 +33,8 @@ def expand_sharded_paths(path: str):
     import glob as globlib
 
     fmt = "csv"
-    if ":" in path and path.split(":", 1)[0] in ("csv", "tfrecord", "avro"):
+    if ":" in path and path.split(":", 1)[0] in (
+            "csv", "tfrecord", "avro", "tfrecord+gzip", "tfrecordv2+tfe"):
         fmt, path = path.split(":", 1)
     out = []
     for part in path.split(","):
@@ -56,6 +57,14 @@ def expand_sharded_paths(path: str):
 def _to_column_dict(data: InputData) -> Dict[str, np.ndarray]:
     if isinstance(data, str):
         fmt, paths = expand_sharded_paths(data)
+        if fmt in ("tfrecord", "tfrecord+gzip", "tfrecordv2+tfe"):
+            from ydf_amd.dataset.tfrecord import read_tfrecord_columns
+
+            return read_tfrecord_columns(paths)
+        if fmt == "avro":
+            from ydf_amd.dataset.avro import read_avro_columns
+
+            return read_avro_columns(paths)
         if fmt != "csv":
             raise NotImplementedError(
                 f"dataset format {fmt!r} not supported yet (ROADMAP)")
