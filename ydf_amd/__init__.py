@@ -88,6 +88,8 @@ class Monotonic:
 from ydf_amd.metric.comparison import ModelComparison, compare_models
 from ydf_amd.dataset.synthetic import (SyntheticDatasetOptions,
                                         generate_synthetic_dataset)
+from ydf_amd.deep import (DeepModel, MultiLayerPerceptronLearner,
+                          TabularTransformerLearner)
 from ydf_amd.model.sklearn_io import from_sklearn
 from ydf_amd.metric.metric import Evaluation, evaluate_predictions
 

@@ -17,6 +17,10 @@ def load_model(path: str) -> GenericModel:
 
     with open(os.path.join(path, "header.json")) as f:
         header = json.load(f)
+    if header.get("model_type") in ("MLP", "TABULAR_TRANSFORMER"):
+        from ydf_amd.deep import DeepModel
+
+        return DeepModel.load(path)
     with open(os.path.join(path, "dataspec.json")) as f:
         dataspec = DataSpecification.from_json(json.load(f))
     z = np.load(os.path.join(path, "forest.npz"))
