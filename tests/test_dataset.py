@@ -113,3 +113,39 @@ def test_pav_calibration():
     # round-trip
     cal2 = type(cal).from_json(cal.to_json())
     np.testing.assert_allclose(cal2.apply(s), phat)
+
+
+def test_categorical_set_features(tmp_path):
+    """Multi-valued categorical columns (reference CategoricalSet): list
+    cells expand into per-token contains features; string cells are
+    space-tokenized when declared CATEGORICAL_SET."""
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(0)
+    n = 4000
+    vocabulary = ["buy", "cheap", "hello", "meeting", "urgent", "work"]
+    sets, y = [], []
+    for _ in range(n):
+        toks = list(rng.choice(vocabulary, rng.randint(1, 5),
+                               replace=False))
+        spam = ("buy" in toks or "cheap" in toks) ^ (rng.rand() < 0.05)
+        sets.append(toks)
+        y.append("spam" if spam else "ham")
+    d = {"words": np.array(sets, dtype=object),
+         "x": rng.randn(n).astype(np.float32),
+         "label": np.array(y)}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=20, validation_ratio=0).train(d)
+    assert m.evaluate(d).accuracy > 0.9
+    assert any(f.startswith("words.") for f in m.input_feature_names())
+    m.save(str(tmp_path / "s"))
+    m2 = ydf.load_model(str(tmp_path / "s"))
+    np.testing.assert_array_equal(m.predict(d), m2.predict(d))
+    # declared CATEGORICAL_SET on a space-delimited string column
+    d2 = {"words": np.array([" ".join(s) for s in sets]),
+          "x": d["x"], "label": d["label"]}
+    feats = [ydf.Column("words", ydf.Semantic.CATEGORICAL_SET), "x"]
+    m3 = ydf.GradientBoostedTreesLearner(
+        label="label", features=feats, num_trees=20,
+        validation_ratio=0).train(d2)
+    assert m3.evaluate(d2).accuracy > 0.9

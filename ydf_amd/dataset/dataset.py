@@ -126,8 +126,30 @@ def infer_dataspec(
                 # Small-cardinality integers stay numerical too (the binned
                 # store handles them fine); strings become categorical.
                 sem = Semantic.NUMERICAL
+            elif arr.dtype == object and len(arr) and isinstance(
+                    next((c for c in arr if c is not None), None),
+                    (list, tuple, set, frozenset, np.ndarray)):
+                sem = Semantic.CATEGORICAL_SET
             else:
                 sem = Semantic.CATEGORICAL
+        if sem == Semantic.CATEGORICAL_SET and not is_label:
+            # multi-valued categorical (reference CategoricalSet columns):
+            # expanded into per-token boolean "contains" virtual features;
+            # set conditions are approximated by contains conditions
+            from collections import Counter
+
+            counter = Counter()
+            for cell in arr:
+                counter.update(_set_tokens(cell))
+            top = [t for t, c in counter.most_common(64)
+                   if c >= min_vocab_frequency]
+            for tok in top:
+                columns.append(ColumnSpec(
+                    name=f"{name}.{tok}", semantic=Semantic.BOOLEAN,
+                    min_value=0.0, max_value=1.0,
+                    boundaries=np.asarray([0.5], dtype=np.float32),
+                    set_source=name, set_token=tok))
+            continue
         if sem == Semantic.CATEGORICAL:
             vocab = categorical_vocab(arr, max_vocab_count,
                                       min_vocab_frequency)
@@ -156,12 +178,26 @@ def infer_dataspec(
     return DataSpecification(columns=columns, label=label)
 
 
+def _set_tokens(cell) -> set:
+    if cell is None:
+        return set()
+    if isinstance(cell, (list, tuple, set, frozenset, np.ndarray)):
+        return {str(t) for t in cell}
+    return {t for t in str(cell).split(" ") if t}
+
+
 def encode_column(arr: np.ndarray, spec: ColumnSpec) -> np.ndarray:
     """Encodes one raw column to float32 according to its spec.
 
     CATEGORICAL -> vocabulary index (0 = OOV); NUMERICAL -> float32 with
     NaN imputed by the training-set mean (reference GLOBAL_IMPUTATION
-    missing-value policy, decision_tree.proto:85-103)."""
+    missing-value policy, decision_tree.proto:85-103); virtual
+    set-membership columns -> 1.0 iff set_token in the cell's tokens."""
+    if spec.set_source is not None:
+        tok = spec.set_token
+        return np.fromiter((1.0 if tok in _set_tokens(c) else 0.0
+                            for c in arr), dtype=np.float32,
+                           count=len(arr))
     if spec.semantic == Semantic.CATEGORICAL:
         lookup = {item: i for i, item in enumerate(spec.vocab)}
         out = np.fromiter((lookup.get(s, 0) for s in arr.astype(str)),
@@ -231,9 +267,10 @@ def create_vertical_dataset(
     n = len(next(iter(cols.values()))) if cols else 0
     X = np.empty((len(feature_specs), n), dtype=np.float32)
     for i, spec in enumerate(feature_specs):
-        if spec.name not in cols:
-            raise ValueError(f"missing feature column {spec.name!r}")
-        X[i] = encode_column(cols[spec.name], spec)
+        src = spec.set_source or spec.name
+        if src not in cols:
+            raise ValueError(f"missing feature column {src!r}")
+        X[i] = encode_column(cols[src], spec)
 
     label_values = None
     if dataspec.label is not None and dataspec.label in cols:

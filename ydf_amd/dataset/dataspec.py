@@ -67,6 +67,11 @@ class ColumnSpec:
     max_value: float = 0.0
     num_nas: int = 0
     boundaries: Optional[np.ndarray] = None
+    # CATEGORICAL_SET expansion: this column is the virtual boolean
+    # "set_token in data[set_source]" (reference categorical-set
+    # conditions, approximated by per-token contains conditions)
+    set_source: Optional[str] = None
+    set_token: Optional[str] = None
 
     @property
     def vocab_size(self) -> int:
@@ -85,6 +90,9 @@ class ColumnSpec:
             d["vocab"] = list(self.vocab)
         if self.boundaries is not None:
             d["boundaries"] = [float(v) for v in self.boundaries]
+        if self.set_source is not None:
+            d["set_source"] = self.set_source
+            d["set_token"] = self.set_token
         return d
 
     @classmethod
@@ -100,6 +108,8 @@ class ColumnSpec:
             boundaries=np.asarray(d["boundaries"], dtype=np.float32)
             if "boundaries" in d
             else None,
+            set_source=d.get("set_source"),
+            set_token=d.get("set_token"),
         )
 
 

@@ -135,9 +135,10 @@ class GenericModel:
         n = len(next(iter(cols.values()))) if cols else 0
         X = np.empty((len(specs), n), dtype=np.float32)
         for i, spec in enumerate(specs):
-            if spec.name not in cols:
-                raise ValueError(f"missing input feature {spec.name!r}")
-            X[i] = encode_column(cols[spec.name], spec)
+            src = spec.set_source or spec.name
+            if src not in cols:
+                raise ValueError(f"missing input feature {src!r}")
+            X[i] = encode_column(cols[src], spec)
         return X
 
     def _forest_on(self, device: torch.device) -> _DeviceForest:
