@@ -99,3 +99,29 @@ def test_model_comparison_and_cis():
         validation_ratio=0).train(dr)
     cmp = ydf.compare_models(w, s, dr)
     assert cmp.p_value < 0.01
+
+
+def test_to_java_codegen():
+    """Java embed codegen (reference serving/embed/java): structural
+    checks — javac is not available in this image, so the C++ twin
+    (compiled + compared elsewhere) anchors the shared emitter logic."""
+    import ydf_amd as ydf
+
+    d = ydf.generate_synthetic_dataset(num_examples=2000, num_numerical=4,
+                                       num_categorical=1, seed=3)
+    m = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=5, validation_ratio=0).train(d)
+    src = ydf.to_java(m, "M")
+    assert "public final class M" in src
+    assert "public static float predict(float[] f)" in src
+    assert src.count("private static float tree") == 5
+    assert "Math.exp" in src  # sigmoid link
+    assert src.count("{") == src.count("}")
+    # multi-class emits predictMulti
+    d3 = ydf.generate_synthetic_dataset(num_examples=1500, num_classes=3,
+                                        seed=4)
+    m3 = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=6, validation_ratio=0).train(d3)
+    src3 = ydf.to_java(m3)
+    assert "predictMulti" in src3
+    assert src3.count("{") == src3.count("}")
