@@ -149,3 +149,29 @@ def test_categorical_set_features(tmp_path):
         label="label", features=feats, num_trees=20,
         validation_ratio=0).train(d2)
     assert m3.evaluate(d2).accuracy > 0.9
+
+
+def test_allow_na_conditions():
+    """'x is missing' conditions (reference allow_na_conditions /
+    NaCondition): when the label depends on missingness itself and the
+    imputed value collides with a real value, only the NA condition can
+    separate them."""
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(0)
+    n = 6000
+    x = rng.choice([-1.0, 0.0, 1.0], n).astype(np.float32)  # mean ~ 0
+    miss = rng.rand(n) < 0.3
+    y = np.where(miss, "m", "p")
+    x2 = x.copy()
+    x2[miss] = np.nan
+    d = {"x": x2, "label": y}
+    kw = dict(label="label", num_trees=10, validation_ratio=0)
+    acc_without = ydf.GradientBoostedTreesLearner(**kw).train(d).evaluate(
+        d).accuracy
+    m = ydf.GradientBoostedTreesLearner(
+        allow_na_conditions=True, **kw).train(d)
+    acc_with = m.evaluate(d).accuracy
+    assert acc_with > 0.999
+    assert acc_with > acc_without + 0.05
+    assert "x.is_na" in m.input_feature_names()

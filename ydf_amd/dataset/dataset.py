@@ -94,6 +94,7 @@ def infer_dataspec(
     max_vocab_count: int = 2000,
     min_vocab_frequency: int = 1,
     max_bins: int = 256,
+    allow_na_conditions: bool = False,
 ) -> DataSpecification:
     """Single-pass dataspec inference (reference: data_spec_inference.h:55)."""
     from ydf_amd.dataset.dataspec import Task
@@ -175,6 +176,15 @@ def infer_dataspec(
                     num_nas=num_nas,
                     boundaries=numerical_boundaries(v, max_bins=max_bins),
                 ))
+            if allow_na_conditions and num_nas > 0 and not is_label:
+                # "x is missing" conditions (reference NaCondition,
+                # decision_tree.proto Condition.na_condition) as a
+                # virtual boolean feature
+                columns.append(ColumnSpec(
+                    name=f"{name}.is_na", semantic=Semantic.BOOLEAN,
+                    min_value=0.0, max_value=1.0,
+                    boundaries=np.asarray([0.5], dtype=np.float32),
+                    set_source=name, set_token=None))
     return DataSpecification(columns=columns, label=label)
 
 
@@ -195,6 +205,9 @@ def encode_column(arr: np.ndarray, spec: ColumnSpec) -> np.ndarray:
     set-membership columns -> 1.0 iff set_token in the cell's tokens."""
     if spec.set_source is not None:
         tok = spec.set_token
+        if tok is None:  # "is NA" virtual column (allow_na_conditions)
+            v = np.asarray(arr, dtype=np.float32)
+            return (~np.isfinite(v)).astype(np.float32)
         return np.fromiter((1.0 if tok in _set_tokens(c) else 0.0
                             for c in arr), dtype=np.float32,
                            count=len(arr))
@@ -254,6 +267,7 @@ def create_vertical_dataset(
     max_vocab_count: int = 2000,
     min_vocab_frequency: int = 1,
     max_bins: int = 256,
+    allow_na_conditions: bool = False,
 ) -> VerticalDataset:
     """Builds a VerticalDataset, inferring the dataspec unless provided."""
     cols = _to_column_dict(data)
@@ -262,7 +276,8 @@ def create_vertical_dataset(
                                   features=features,
                                   max_vocab_count=max_vocab_count,
                                   min_vocab_frequency=min_vocab_frequency,
-                                  max_bins=max_bins)
+                                  max_bins=max_bins,
+                                  allow_na_conditions=allow_na_conditions)
     feature_specs = dataspec.feature_columns
     n = len(next(iter(cols.values()))) if cols else 0
     X = np.empty((len(feature_specs), n), dtype=np.float32)

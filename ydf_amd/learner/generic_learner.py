@@ -21,8 +21,10 @@ class GenericLearner:
                  features: Optional[Sequence[Union[str, Column]]] = None,
                  weights: Optional[str] = None, tuner=None,
                  max_vocab_count: int = 2000, min_vocab_frequency: int = 1,
+                 allow_na_conditions: bool = False,
                  random_seed: int = 123456, device=None,
                  num_threads: Optional[int] = None):
+        self.allow_na_conditions = allow_na_conditions
         self.label = label
         self._task = task
         self.features = features
@@ -109,7 +111,8 @@ class GenericLearner:
             ds = create_vertical_dataset(
                 cols, label=self.label, task=self._task,
                 features=features, max_vocab_count=self.max_vocab_count,
-                min_vocab_frequency=self.min_vocab_frequency)
+                min_vocab_frequency=self.min_vocab_frequency,
+                allow_na_conditions=self.allow_na_conditions)
             if self.weights_col is not None:
                 if self.weights_col not in cols:
                     raise ValueError(
