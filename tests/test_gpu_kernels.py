@@ -303,3 +303,25 @@ def test_oblique_gbt_train_gpu():
     t2 = rng.randn(4000).astype(np.float32)
     te = {"x1": t1, "x2": t2, "label": np.where(t1 + t2 > 0, "p", "n")}
     assert m.evaluate(te, device="cuda").accuracy > 0.99
+
+
+def test_rf_deep_partitioned_i16_vs_cpu():
+    """Deep levels route through the feature-interleaved gathered
+    kernel (hist_build_gathered16); integer-valued RF gradients make the
+    GPU forest bit-comparable to the CPU sparse path."""
+    rng = np.random.RandomState(11)
+    n, F = 60000, 40
+    d = {f"x{i}": rng.randn(n).astype(np.float32) for i in range(F)}
+    d["label"] = np.where(
+        rng.randn(n) + d["x0"] * 2 - d["x1"] + d["x2"] * d["x3"] > 0,
+        "a", "b")
+    kw = dict(label="label", num_trees=3, max_depth=12, min_examples=2,
+              bootstrap_training_dataset=False,
+              num_candidate_attributes=-1,
+              compute_oob_performances=False)
+    m_cpu = ydf.RandomForestLearner(device="cpu", **kw).train(d)
+    m_gpu = ydf.RandomForestLearner(device="cuda", **kw).train(d)
+    np.testing.assert_array_equal(m_gpu.forest.feat, m_cpu.forest.feat)
+    np.testing.assert_array_equal(m_gpu.forest.left, m_cpu.forest.left)
+    np.testing.assert_allclose(m_gpu.forest.thr, m_cpu.forest.thr,
+                               rtol=1e-5, atol=1e-6)

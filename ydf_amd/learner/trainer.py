@@ -265,6 +265,7 @@ class ForestTrainer:
         if valid_bins is not None:
             self.valid_node_ids = torch.empty(valid_bins.shape[1],
                                               dtype=torch.int32, device=dev)
+        self._bins16 = None  # lazy interleaved copy for deep levels
         self.rng = np.random.RandomState(cfg.seed)
 
     # -- helpers ----------------------------------------------------------
@@ -478,6 +479,13 @@ class ForestTrainer:
                             // (ops.MAX_BINS * 16))
             use_partition = (self.device.type == "cuda"
                              and n_active > 2 * lds_group)
+            # feature-interleaved variant: 16 features per 16-byte load
+            # (see hist_build_gathered16_kernel); invalid under oblique
+            # (virtual feature rows change per level)
+            use_i16 = (use_partition and self.P == 0 and self.F >= 32
+                       and os.environ.get("YDFA_HIST_I16", "1") == "1")
+            if use_i16 and self._bins16 is None:
+                self._bins16 = ops.pack_bins16(self.bins)
             row_order = None
             if use_partition:
                 # build_map (not slot_map): derived (histogram-subtraction)
@@ -502,7 +510,20 @@ class ForestTrainer:
                 ns = min(self.max_slots, n_active - s0)
                 hist_view = self.hist[:ns]
                 hist_view.zero_()
-                if use_partition:
+                if use_i16:
+                    spg = 2
+                    g_slots = list(range(s0, s0 + ns, spg)) + [s0 + ns]
+                    garr = offs[g_slots]
+                    goffs = torch.from_numpy(
+                        np.ascontiguousarray(garr)).to(self.device)
+                    n_groups = len(g_slots) - 1
+                    max_rows = int(np.diff(garr).max()) if n_groups else 0
+                    ops.hist_build_gathered16(
+                        self._bins16, self.gh, self.node_ids, build_map,
+                        row_order, goffs, hist_view, self.N, self.F,
+                        level_base, level_size, s0, spg, n_groups,
+                        max_rows)
+                elif use_partition:
                     for g0 in range(s0, s0 + ns, lds_group):
                         g1 = min(g0 + lds_group, s0 + ns)
                         ops.hist_build_gathered(

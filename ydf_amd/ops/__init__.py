@@ -111,6 +111,35 @@ def hist_build(bins: torch.Tensor, gh: torch.Tensor, node_ids: torch.Tensor,
     return hist
 
 
+def hist_build_gathered16(bins16: torch.Tensor, gh: torch.Tensor,
+                          node_ids: torch.Tensor, slot_map: torch.Tensor,
+                          row_order: torch.Tensor,
+                          group_offs: torch.Tensor, hist: torch.Tensor,
+                          N: int, F: int, level_base: int,
+                          level_size: int, win0: int, spg: int,
+                          n_groups: int, max_group_rows: int):
+    """Feature-interleaved partitioned histograms (GPU only): bins16
+    [ceil(F/16), N, 16] u8; one uint4 load covers 16 features. Groups of
+    `spg` slots map to blockIdx.z with row ranges from group_offs."""
+    assert bins16.is_cuda
+    _C.gpu_hist_build_gathered16(
+        bins16.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
+        slot_map.data_ptr(), row_order.data_ptr(), group_offs.data_ptr(),
+        hist.data_ptr(), N, F, level_base, level_size, win0, spg,
+        n_groups, max_group_rows, _stream())
+    return hist
+
+
+def pack_bins16(bins: torch.Tensor) -> torch.Tensor:
+    """[F, N] u8 -> [ceil(F/16), N, 16] u8 interleaved copy."""
+    F, N = bins.shape
+    F16 = (F + 15) // 16
+    padded = torch.zeros((F16 * 16, N), dtype=torch.uint8,
+                         device=bins.device)
+    padded[:F] = bins
+    return padded.view(F16, 16, N).permute(0, 2, 1).contiguous()
+
+
 def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
                node_stats: torch.Tensor, best_gain_nf: torch.Tensor,
                best_bin_nf: torch.Tensor, best_feat: torch.Tensor,
