@@ -599,3 +599,36 @@ def test_rf_oob_permutation_importances():
     scores = {name: s for s, name in vi}
     assert scores["x1"] > 0.1
     assert abs(scores["noise"]) < 0.05
+
+
+def test_rf_bootstrap_ratio_and_max_duration():
+    import time as _time
+
+    d = ydf.generate_synthetic_dataset(num_examples=3000, seed=5)
+    m = ydf.RandomForestLearner(label="LABEL", num_trees=5,
+                                bootstrap_size_ratio=0.5,
+                                compute_oob_performances=False).train(d)
+    assert m.num_trees() == 5
+    t0 = _time.monotonic()
+    m2 = ydf.RandomForestLearner(
+        label="LABEL", num_trees=100000,
+        maximum_training_duration_seconds=1.0,
+        compute_oob_performances=False).train(d)
+    assert _time.monotonic() - t0 < 10
+    assert 0 < m2.num_trees() < 100000
+
+
+def test_analyze_prediction():
+    d = ydf.generate_synthetic_dataset(num_examples=2000, seed=6)
+    m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=10,
+                                        validation_ratio=0).train(d)
+    one = {k: v[:1] for k, v in d.items()}
+    ap = m.analyze_prediction(one)
+    assert "__BIAS__" in ap
+    assert set(m.input_feature_names()) <= set(ap)
+    # contributions + bias reconstruct the margin (SHAP completeness)
+    import torch
+    X = torch.from_numpy(m._encode_features(one))
+    margin = float(m.predict_margin(X)[0, 0])
+    total = sum(float(v) for v in ap.values())
+    np.testing.assert_allclose(total, margin, rtol=1e-3, atol=1e-3)

@@ -433,6 +433,7 @@ class RandomForestLearner(GenericLearner):
                  uplift_treatment: Optional[str] = None,
                  uplift_split_score: str = "KULLBACK_LEIBLER",
                  uplift_min_examples_in_treatment: int = 5,
+                 maximum_training_duration_seconds: float = -1.0,
                  split_axis: str = "AXIS_ALIGNED",
                  sparse_oblique_num_projections_exponent: float = 2.0,
                  sparse_oblique_max_num_projections: int = 6000,
@@ -460,6 +461,8 @@ class RandomForestLearner(GenericLearner):
             uplift_split_score=uplift_split_score,
             uplift_min_examples_in_treatment=(
                 uplift_min_examples_in_treatment),
+            maximum_training_duration_seconds=(
+                maximum_training_duration_seconds),
             split_axis=split_axis,
             sparse_oblique_num_projections_exponent=(
                 sparse_oblique_num_projections_exponent),
@@ -588,7 +591,10 @@ class RandomForestLearner(GenericLearner):
             min_examples=hp["min_examples"], min_hessian=0.0,
             n_classes=n_classes, seed=self.random_seed,
             bootstrap=hp["bootstrap_training_dataset"],
+            bootstrap_ratio=hp.get("bootstrap_size_ratio", 1.0),
             num_candidate_features=ncand,
+            max_duration_seconds=hp.get(
+                "maximum_training_duration_seconds", -1.0),
             honest=hp.get("honest", False),
             honest_ratio=hp.get("honest_ratio_leaf_examples", 0.5),
             honest_fixed_separation=hp.get("honest_fixed_separation",
@@ -687,7 +693,8 @@ def _oob_permutation_vi(model, ds, cfg, task, device):
         return float(((p > 0.5) == (yb > 0.5)).mean())
 
     for t in range(T):
-        w = trainer_lib.rf_bootstrap_weights(cfg.seed, t, N, device)
+        w = trainer_lib.rf_bootstrap_weights(cfg.seed, t, N, device,
+                                             cfg.bootstrap_ratio)
         oob = (w == 0).cpu().numpy()
         if oob.sum() < 10:
             continue

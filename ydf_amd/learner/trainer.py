@@ -63,7 +63,9 @@ class TrainerConfig:
     seed: int = 123456
     # RF-specific
     bootstrap: bool = False
+    bootstrap_ratio: float = 1.0     # Poisson rate (bootstrap_size_ratio)
     num_candidate_features: int = 0  # 0 = all features
+    max_duration_seconds: float = -1.0
     # honest trees (reference decision_tree.proto Honest message): tree
     # structure from one random half, leaf values re-estimated on the other
     honest: bool = False
@@ -1080,18 +1082,19 @@ def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
     return float((s[0] / s[1]).item())
 
 
-def rf_bootstrap_weights(seed: int, tree_idx: int, N: int,
-                         dev) -> torch.Tensor:
-    """Poisson(1) bootstrap draw for tree `tree_idx`, clipped at 15.
-    Seeded per tree so the draw can be REGENERATED after training (OOB
-    permutation importances re-derive each tree's out-of-bag rows)."""
+def rf_bootstrap_weights(seed: int, tree_idx: int, N: int, dev,
+                         rate: float = 1.0) -> torch.Tensor:
+    """Poisson(rate) bootstrap draw for tree `tree_idx`, clipped at 15
+    (rate = bootstrap_size_ratio). Seeded per tree so the draw can be
+    REGENERATED after training (OOB permutation importances re-derive
+    each tree's out-of-bag rows)."""
     if dev.type == "cuda":
         g = torch.Generator(device=dev)
         g.manual_seed((seed * 31337 + tree_idx) % (1 << 31))
-        return torch.poisson(torch.ones(N, device=dev),
+        return torch.poisson(torch.full((N,), float(rate), device=dev),
                              generator=g).clamp_(max=15)
     rs = np.random.RandomState((seed * 31337 + tree_idx) % (1 << 31))
-    w = np.minimum(rs.poisson(1.0, size=N), 15).astype(np.float32)
+    w = np.minimum(rs.poisson(rate, size=N), 15).astype(np.float32)
     return torch.from_numpy(w).to(dev)
 
 
@@ -1124,18 +1127,30 @@ def train_rf(trainer: ForestTrainer, log=None,
     N = trainer.N
     multi = cfg.n_classes > 2 and cfg.loss == LOSS_RF
     C = cfg.n_classes if multi else 1
+    import time as _time
+
     trees: List[HostTree] = []
     onehot = None
     oob_sum = oob_cnt = None
     if compute_oob and cfg.bootstrap:
         oob_sum = torch.zeros((C, N), dtype=torch.float32, device=dev)
         oob_cnt = torch.zeros(N, dtype=torch.float32, device=dev)
+    t_start = _time.monotonic()
     for it in range(cfg.num_trees):
+        if cfg.max_duration_seconds > 0 and it > 0 and \
+                _time.monotonic() - t_start > cfg.max_duration_seconds:
+            # reference AdaptativeWork (utils/adaptive_work.h) shrinks
+            # the per-tree sample to fit the budget; we stop adding
+            # trees instead (every grown tree is full-quality)
+            if log:
+                log(f"maximum_training_duration reached after {it} trees")
+            break
         weights = None
         if cfg.bootstrap:
-            # Poisson(1) bootstrap, clipped at 15 (P < 1e-12) — the packed
+            # Poisson bootstrap, clipped at 15 (P < 1e-12) — the packed
             # u64 histogram path requires per-example h <= 16
-            weights = rf_bootstrap_weights(cfg.seed, it, N, dev)
+            weights = rf_bootstrap_weights(cfg.seed, it, N, dev,
+                                           cfg.bootstrap_ratio)
         if trainer.weights is not None:
             # user example weights compose with the bootstrap draw counts
             weights = trainer.weights if weights is None \

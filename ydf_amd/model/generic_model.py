@@ -299,6 +299,15 @@ class GenericModel:
         out["__BIAS__"] = phi[:, F]
         return out
 
+    def analyze_prediction(self, single_example) -> Dict:
+        """Per-example prediction analysis (mirrors PYDF
+        model.analyze_prediction, generic_model.py:674 family): TreeSHAP
+        feature attributions for one (or a few) example(s). Returns
+        {feature: contribution} plus "__BIAS__"."""
+        shap = self.predict_shap(single_example)
+        return {k: (float(v[0]) if len(v) == 1 else v)
+                for k, v in shap.items()}
+
     def variable_importances(self) -> Dict:
         """Structure-based variable importances (reference
         AbstractModel::GetVariableImportance; SUM_SCORE requires a model
