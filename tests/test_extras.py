@@ -125,3 +125,40 @@ def test_to_java_codegen():
     src3 = ydf.to_java(m3)
     assert "predictMulti" in src3
     assert src3.count("{") == src3.count("}")
+
+
+def test_to_docker(tmp_path):
+    """to_docker (PYDF export_docker analogue): generated FastAPI app
+    serves the saved model (exercised in-process via TestClient)."""
+    import sys
+
+    import ydf_amd as ydf
+
+    pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    d = ydf.generate_synthetic_dataset(num_examples=1500, seed=10)
+    m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=5,
+                                        validation_ratio=0).train(d)
+    out = tmp_path / "serve"
+    ydf.to_docker(m, str(out))
+    for fn in ("Dockerfile", "main.py", "requirements.txt",
+               "model/header.json"):
+        assert (out / fn).exists()
+    sys.path.insert(0, str(out))
+    try:
+        import importlib
+
+        main = importlib.import_module("main")
+        importlib.reload(main)
+        client = TestClient(main.app)
+        assert client.get("/").json()["model"]
+        example = {k: v[0].item() if hasattr(v[0], "item") else str(v[0])
+                   for k, v in d.items() if k != "LABEL"}
+        r = client.post("/predict", json=example).json()
+        want = float(m.predict({k: v[:1] for k, v in d.items()},
+                               device="cpu")[0])
+        np.testing.assert_allclose(r["predictions"][0], want, rtol=1e-5)
+    finally:
+        sys.path.remove(str(out))
+        sys.modules.pop("main", None)

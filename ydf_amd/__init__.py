@@ -44,6 +44,7 @@ from_yggdrasil_model = load_ydf_model
 from ydf_amd.model import tree
 
 # Model export / serving extras
+from ydf_amd.serving.deploy import to_docker
 from ydf_amd.serving.embed import to_cpp, to_java
 from ydf_amd.learner.extras import (
     BackwardSelectionFeatureSelector,
