@@ -325,3 +325,26 @@ def test_rf_deep_partitioned_i16_vs_cpu():
     np.testing.assert_array_equal(m_gpu.forest.left, m_cpu.forest.left)
     np.testing.assert_allclose(m_gpu.forest.thr, m_cpu.forest.thr,
                                rtol=1e-5, atol=1e-6)
+
+
+def test_rf_deep_partitioned_i16_masked_vs_cpu():
+    """Masked interleaved build (per-slot feature-sampling bits) must
+    agree with the CPU sparse path exactly: integer-valued RF gradient
+    sums make histogram subtraction exact, so sub (CPU) vs direct build
+    (GPU masked) yield identical splits."""
+    rng = np.random.RandomState(12)
+    n, F = 60000, 40
+    d = {f"x{i}": rng.randn(n).astype(np.float32) for i in range(F)}
+    d["label"] = np.where(
+        rng.randn(n) + d["x0"] * 2 - d["x1"] + d["x2"] * d["x3"] > 0,
+        "a", "b")
+    kw = dict(label="label", num_trees=3, max_depth=12, min_examples=2,
+              bootstrap_training_dataset=False,
+              num_candidate_attributes=6,
+              compute_oob_performances=False)
+    m_cpu = ydf.RandomForestLearner(device="cpu", **kw).train(d)
+    m_gpu = ydf.RandomForestLearner(device="cuda", **kw).train(d)
+    np.testing.assert_array_equal(m_gpu.forest.feat, m_cpu.forest.feat)
+    np.testing.assert_array_equal(m_gpu.forest.left, m_cpu.forest.left)
+    np.testing.assert_allclose(m_gpu.forest.thr, m_cpu.forest.thr,
+                               rtol=1e-5, atol=1e-6)

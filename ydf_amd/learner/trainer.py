@@ -438,6 +438,17 @@ class ForestTrainer:
                                   device=self.device)
             slot_map[rel_t] = self.arange_buf[:n_active]
             feat_mask = self._feat_mask(n_active, tree_idx, level)
+            lds_group = max(1, (160 * 1024 - min(4 * level_size, 32768))
+                            // (ops.MAX_BINS * 16))
+            use_partition = (self.device.type == "cuda"
+                             and n_active > 2 * lds_group)
+            # feature-interleaved variant: 16 features per 16-byte load
+            # (see hist_build_gathered16_kernel); invalid under oblique
+            # (virtual feature rows change per level)
+            use_i16 = (use_partition and self.P == 0 and self.F >= 32
+                       and os.environ.get("YDFA_HIST_I16", "1") == "1")
+            if use_i16 and self._bins16 is None:
+                self._bins16 = ops.pack_bins16(self.bins)
 
             # Histogram subtraction (reference-free optimization; standard
             # GBT trick): build histograms only for the SMALLER child of
@@ -445,7 +456,8 @@ class ForestTrainer:
             # both this and the previous level fit un-chunked.
             use_sub = (self.use_hist_sub and prev_fit
                        and prev_slot_of is not None
-                       and n_active <= self.max_slots)
+                       and n_active <= self.max_slots
+                       and not (use_i16 and feat_mask is not None))
             derived = []  # (slot, parent_slot, sibling_slot)
             if use_sub:
                 active_set = {int(a): s for s, a in enumerate(active_abs)}
@@ -477,17 +489,6 @@ class ForestTrainer:
             # sort active rows by slot once so each group pass sweeps only
             # its own contiguous row range (vs rescanning the full table
             # per group). Group boundaries come from local per-slot counts.
-            lds_group = max(1, (160 * 1024 - min(4 * level_size, 32768))
-                            // (ops.MAX_BINS * 16))
-            use_partition = (self.device.type == "cuda"
-                             and n_active > 2 * lds_group)
-            # feature-interleaved variant: 16 features per 16-byte load
-            # (see hist_build_gathered16_kernel); invalid under oblique
-            # (virtual feature rows change per level)
-            use_i16 = (use_partition and self.P == 0 and self.F >= 32
-                       and os.environ.get("YDFA_HIST_I16", "1") == "1")
-            if use_i16 and self._bins16 is None:
-                self._bins16 = ops.pack_bins16(self.bins)
             row_order = None
             if use_partition:
                 # build_map (not slot_map): derived (histogram-subtraction)
@@ -520,11 +521,26 @@ class ForestTrainer:
                         np.ascontiguousarray(garr)).to(self.device)
                     n_groups = len(g_slots) - 1
                     max_rows = int(np.diff(garr).max()) if n_groups else 0
+                    maskbits = None
+                    if feat_mask is not None:
+                        # per-slot sampled-feature bits; feature 0 forced
+                        # on (its histogram provides the node totals)
+                        F16 = (self.F + 15) // 16
+                        fm = torch.zeros((ns, F16 * 16),
+                                         dtype=torch.int32,
+                                         device=self.device)
+                        fm[:, :self.F] = feat_mask[s0:s0 + ns].int()
+                        fm[:, 0] = 1
+                        weightsb = (1 << torch.arange(
+                            16, dtype=torch.int32, device=self.device))
+                        maskbits = (fm.view(ns, F16, 16)
+                                    * weightsb).sum(-1).to(torch.int16)
+                        maskbits = maskbits.contiguous()
                     ops.hist_build_gathered16(
                         self._bins16, self.gh, self.node_ids, build_map,
                         row_order, goffs, hist_view, self.N, self.F,
                         level_base, level_size, s0, spg, n_groups,
-                        max_rows)
+                        max_rows, maskbits=maskbits)
                 elif use_partition:
                     for g0 in range(s0, s0 + ns, lds_group):
                         g1 = min(g0 + lds_group, s0 + ns)

@@ -117,15 +117,19 @@ def hist_build_gathered16(bins16: torch.Tensor, gh: torch.Tensor,
                           group_offs: torch.Tensor, hist: torch.Tensor,
                           N: int, F: int, level_base: int,
                           level_size: int, win0: int, spg: int,
-                          n_groups: int, max_group_rows: int):
+                          n_groups: int, max_group_rows: int,
+                          maskbits=None):
     """Feature-interleaved partitioned histograms (GPU only): bins16
     [ceil(F/16), N, 16] u8; one uint4 load covers 16 features. Groups of
-    `spg` slots map to blockIdx.z with row ranges from group_offs."""
+    `spg` slots map to blockIdx.z with row ranges from group_offs.
+    maskbits [n_slots, ceil(F/16)] u16: per-slot feature-sampling bits
+    (skips loads/atomics for unsampled features)."""
     assert bins16.is_cuda
+    mb = maskbits.data_ptr() if maskbits is not None else 0
     _C.gpu_hist_build_gathered16(
         bins16.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
         slot_map.data_ptr(), row_order.data_ptr(), group_offs.data_ptr(),
-        hist.data_ptr(), N, F, level_base, level_size, win0, spg,
+        hist.data_ptr(), mb, N, F, level_base, level_size, win0, spg,
         n_groups, max_group_rows, _stream())
     return hist
 

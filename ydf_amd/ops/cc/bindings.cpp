@@ -26,8 +26,8 @@ void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
 void gpu_hist_build_gathered16(const uint8_t*, const float*,
                                const int32_t*, const int32_t*,
                                const int32_t*, const int64_t*, float*,
-                               int64_t, int, int, int, int, int, int,
-                               int64_t, void*);
+                               const uint16_t*, int64_t, int, int, int,
+                               int, int, int, int64_t, void*);
 void gpu_hist_build_gathered(const uint8_t*, const float*, const int32_t*,
                              const int32_t*, const int32_t*, float*, int64_t,
                              int, int, int, int, int, int, int64_t, int64_t,
@@ -164,13 +164,14 @@ PYBIND11_MODULE(_ydf_ops, m) {
   m.def("gpu_hist_build_gathered16",
         [](uintptr_t bins16, uintptr_t gh, uintptr_t node_ids,
            uintptr_t slot_map, uintptr_t row_order, uintptr_t group_offs,
-           uintptr_t hist, int64_t N, int F, int level_base,
-           int level_size, int win0, int spg, int n_groups,
+           uintptr_t hist, uintptr_t maskbits, int64_t N, int F,
+           int level_base, int level_size, int win0, int spg, int n_groups,
            int64_t max_group_rows, uintptr_t stream) {
           gpu_hist_build_gathered16(
               P<uint8_t>(bins16), P<float>(gh), P<int32_t>(node_ids),
               P<int32_t>(slot_map), P<int32_t>(row_order),
-              P<int64_t>(group_offs), P<float>(hist), N, F, level_base,
+              P<int64_t>(group_offs), P<float>(hist),
+              P<uint16_t>(maskbits), N, F, level_base,
               level_size, win0, spg, n_groups, max_group_rows,
               (void*)stream);
         },

@@ -500,6 +500,7 @@ __global__ void hist_build_gathered16_kernel(
     const int32_t* __restrict__ slot_map,
     const int32_t* __restrict__ row_order,
     const int64_t* __restrict__ group_offs, float* __restrict__ hist,
+    const uint16_t* __restrict__ maskbits,  // [n_slots][F16] or null
     int64_t N, int F, int n_bins, int level_base, int level_size,
     int win0, int spg, int n_chunks) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -522,20 +523,33 @@ __global__ void hist_build_gathered16_kernel(
   const uint4* fb =
       reinterpret_cast<const uint4*>(bins16 + (int64_t)fg * N * 16);
   const int slot_lo = z * spg;
+  const int F16 = (F + 15) / 16;
+  unsigned m_of_slot[8];  // spg <= 8
+  if (maskbits != nullptr) {
+    for (int sI = 0; sI < spg; ++sI)
+      m_of_slot[sI] =
+          maskbits[(int64_t)(slot_lo + sI) * F16 + fg];
+  }
   for (int64_t j = j0 + threadIdx.x; j < j1; j += blockDim.x) {
     const int row = row_order[j];
     const int rel = node_ids[row] - level_base;
     if (rel < 0 || rel >= level_size) continue;
     const int slot = slot_map[rel] - win0 - slot_lo;
     if (slot < 0 || slot >= spg) continue;
+    unsigned m = 0xFFFFu;
+    if (maskbits != nullptr) {
+      m = m_of_slot[slot];
+      if (m == 0u) continue;  // no sampled feature in this group
+    }
     const float2 v = gh[row];
     const unsigned long long hq =
         (unsigned long long)(v.y * kHScale + 0.5f) |
         ((unsigned long long)(v.y != 0.f) << 44);
     const uint4 b = fb[row];
     const unsigned words[4] = {b.x, b.y, b.z, b.w};
-#pragma unroll
-    for (int k = 0; k < 16; ++k) {
+    while (m) {
+      const int k = __ffs(m) - 1;
+      m &= m - 1;
       const int bin = (words[k >> 2] >> ((k & 3) * 8)) & 0xFF;
       const int cell = 2 * ((k * spg + slot) * n_bins + bin);
       atomicAdd(lg + cell, (double)v.x);
@@ -609,6 +623,17 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
   const int slot = blockIdx.x;
   const int f = blockIdx.y;
   const int b = threadIdx.x;  // blockDim.x == n_bins (power of two <= 256)
+  // masked-out (slot, feature) blocks exit before reading the histogram
+  // (f == 0 always scans: it computes the node totals) — with RF's
+  // sqrt(F) per-node sampling this skips ~95% of the scan traffic
+  if (f != 0 && feat_mask != nullptr &&
+      !feat_mask[(int64_t)(slot0 + slot) * F + f]) {
+    if (b == 0) {
+      best_gain_nf[(int64_t)slot * F + f] = -1e30f;
+      best_bin_nf[(int64_t)slot * F + f] = 0;
+    }
+    return;
+  }
   __shared__ float sg[kMaxBins], sh[kMaxBins], sc[kMaxBins];
   const float* hp = hist + ((int64_t)slot * F + f) * (n_bins * 3);
   float rg0 = hp[b * 3];
@@ -1211,10 +1236,10 @@ void gpu_hist_build_gathered16(const uint8_t* bins16, const float* gh,
                                const int32_t* slot_map,
                                const int32_t* row_order,
                                const int64_t* group_offs, float* hist,
-                               int64_t N, int F, int level_base,
-                               int level_size, int win0, int spg,
-                               int n_groups, int64_t max_group_rows,
-                               void* stream) {
+                               const uint16_t* maskbits, int64_t N, int F,
+                               int level_base, int level_size, int win0,
+                               int spg, int n_groups,
+                               int64_t max_group_rows, void* stream) {
   const int n_bins = kMaxBins;
   const int F16 = (F + 15) / 16;
   int chunks = 1;
@@ -1229,8 +1254,9 @@ void gpu_hist_build_gathered16(const uint8_t* bins16, const float* gh,
   hipLaunchKernelGGL(hist_build_gathered16_kernel,
                      dim3(F16, chunks, n_groups), dim3(kBlock), lds,
                      (hipStream_t)stream, bins16, (const float2*)gh,
-                     node_ids, slot_map, row_order, group_offs, hist, N, F,
-                     n_bins, level_base, level_size, win0, spg, chunks);
+                     node_ids, slot_map, row_order, group_offs, hist,
+                     maskbits, N, F, n_bins, level_base, level_size, win0,
+                     spg, chunks);
 }
 
 void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
