@@ -327,11 +327,15 @@ def test_rf_deep_partitioned_i16_vs_cpu():
                                rtol=1e-5, atol=1e-6)
 
 
-def test_rf_deep_partitioned_i16_masked_vs_cpu():
+def test_rf_deep_partitioned_i16_masked_vs_unmasked():
     """Masked interleaved build (per-slot feature-sampling bits) must
-    agree with the CPU sparse path exactly: integer-valued RF gradient
-    sums make histogram subtraction exact, so sub (CPU) vs direct build
-    (GPU masked) yield identical splits."""
+    reproduce the unmasked per-feature gathered path exactly: identical
+    masks (same device generator), and integer-valued RF gradient sums
+    make histogram subtraction (active only in the unmasked path) exact.
+    CPU cannot be the reference here: feature masks are drawn with the
+    device's own generator."""
+    import os
+
     rng = np.random.RandomState(12)
     n, F = 60000, 40
     d = {f"x{i}": rng.randn(n).astype(np.float32) for i in range(F)}
@@ -341,10 +345,14 @@ def test_rf_deep_partitioned_i16_masked_vs_cpu():
     kw = dict(label="label", num_trees=3, max_depth=12, min_examples=2,
               bootstrap_training_dataset=False,
               num_candidate_attributes=6,
-              compute_oob_performances=False)
-    m_cpu = ydf.RandomForestLearner(device="cpu", **kw).train(d)
-    m_gpu = ydf.RandomForestLearner(device="cuda", **kw).train(d)
-    np.testing.assert_array_equal(m_gpu.forest.feat, m_cpu.forest.feat)
-    np.testing.assert_array_equal(m_gpu.forest.left, m_cpu.forest.left)
-    np.testing.assert_allclose(m_gpu.forest.thr, m_cpu.forest.thr,
+              compute_oob_performances=False, device="cuda")
+    os.environ["YDFA_HIST_I16"] = "0"
+    try:
+        m_ref = ydf.RandomForestLearner(**kw).train(d)
+    finally:
+        os.environ.pop("YDFA_HIST_I16")
+    m_i16 = ydf.RandomForestLearner(**kw).train(d)
+    np.testing.assert_array_equal(m_i16.forest.feat, m_ref.forest.feat)
+    np.testing.assert_array_equal(m_i16.forest.left, m_ref.forest.left)
+    np.testing.assert_allclose(m_i16.forest.thr, m_ref.forest.thr,
                                rtol=1e-5, atol=1e-6)
