@@ -35,6 +35,7 @@ class Task(enum.Enum):
     CATEGORICAL_UPLIFT = 4
     NUMERICAL_UPLIFT = 5
     ANOMALY_DETECTION = 6
+    SURVIVAL_ANALYSIS = 7
 
 
 @dataclasses.dataclass

@@ -37,6 +37,8 @@ class GradientBoostedTreesLearner(GenericLearner):
     def __init__(self, label: str, task: Task = Task.CLASSIFICATION,
                  features: Optional[Sequence[Union[str, Column]]] = None,
                  ranking_group: Optional[str] = None,
+                 label_event_observed: Optional[str] = None,
+                 label_entry_age: Optional[str] = None,
                  ndcg_truncation: int = 5,
                  num_trees: int = 300, max_depth: int = 6,
                  shrinkage: float = 0.1, subsample: float = 1.0,
@@ -71,6 +73,8 @@ class GradientBoostedTreesLearner(GenericLearner):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
         self.ranking_group = ranking_group
+        self.label_event_observed = label_event_observed
+        self.label_entry_age = label_entry_age
         self.ndcg_truncation = ndcg_truncation
         self.hyperparameters = dict(
             num_trees=num_trees, max_depth=max_depth, shrinkage=shrinkage,
@@ -115,6 +119,24 @@ class GradientBoostedTreesLearner(GenericLearner):
         hp = self.hyperparameters
         device = self._resolve_device()
         group_ids = None
+        surv_events = surv_entry = None
+        if self._task == Task.SURVIVAL_ANALYSIS:
+            # Cox proportional hazards (reference loss_imp_cox.cc):
+            # label = time, label_event_observed = indicator column
+            if self.label_event_observed is None:
+                raise ValueError("task=SURVIVAL_ANALYSIS needs "
+                                 "label_event_observed=")
+            from ydf_amd.dataset.dataset import _to_column_dict
+
+            cols = _to_column_dict(data)
+            surv_events = np.asarray(
+                cols.pop(self.label_event_observed)).astype(bool)
+            if self.label_entry_age is not None:
+                surv_entry = np.asarray(
+                    cols.pop(self.label_entry_age), dtype=np.float64)
+            data = cols
+            if self.features is None:
+                self.features = [k for k in cols if k != self.label]
         if self._task == Task.RANKING:
             if self.ranking_group is None:
                 raise ValueError("task=RANKING needs ranking_group=")
@@ -169,6 +191,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             loss = (trainer_lib.LOSS_XE_NDCG
                     if named in ("XE_NDCG", "XE_NDCG_MART")
                     else trainer_lib.LOSS_LAMBDA_MART_NDCG)
+        elif self._task == Task.SURVIVAL_ANALYSIS:
+            loss = trainer_lib.LOSS_COX
         else:
             raise NotImplementedError(
                 f"GBT task {self._task} not yet supported")
@@ -208,6 +232,10 @@ class GradientBoostedTreesLearner(GenericLearner):
                                      device,
                                      truncation=self.ndcg_truncation)
         elif valid is not None:
+            if self._task == Task.SURVIVAL_ANALYSIS:
+                raise NotImplementedError(
+                    "survival: use validation_ratio (user valid= needs "
+                    "event columns threaded through; ROADMAP)")
             valid_bins, valid_labels, vds = self._prepare_valid(
                 valid, ds, device)
             if raw_t is not None:
@@ -230,6 +258,24 @@ class GradientBoostedTreesLearner(GenericLearner):
                 if raw_t is not None:
                     valid_raw_t = raw_t[:, vi].contiguous()
                     raw_t = raw_t[:, ti].contiguous()
+                if surv_events is not None:
+                    sv, st = perm[:n_valid], perm[n_valid:]
+                    surv_events_v = surv_events[sv]
+                    surv_events = surv_events[st]
+                    if surv_entry is not None:
+                        surv_entry_v = surv_entry[sv]
+                        surv_entry = surv_entry[st]
+
+        cox = valid_cox = None
+        if self._task == Task.SURVIVAL_ANALYSIS:
+            from ydf_amd.learner.survival import CoxData
+
+            cox = CoxData(labels.cpu().numpy(), surv_events, device,
+                          surv_entry)
+            if valid_bins is not None:
+                valid_cox = CoxData(
+                    valid_labels.cpu().numpy(), surv_events_v, device,
+                    surv_entry_v if surv_entry is not None else None)
 
         F = bins.shape[0]
         ncand = 0
@@ -288,6 +334,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                 num_trees_per_iter=C, activation=activation,
                 metadata={"feature_gains": gains,
                           "ranking_group": self.ranking_group,
+                          "label_event_observed":
+                              self.label_event_observed,
+                          "label_entry_age": self.label_entry_age,
                           "ndcg_truncation": self.ndcg_truncation})
 
         # checkpoint/resume (reference try_resume_training +
@@ -360,7 +409,7 @@ class GradientBoostedTreesLearner(GenericLearner):
             resume_margins=resume_margins,
             resume_valid_margins=resume_valid_margins,
             custom_loss=custom_loss, ranking=ranking,
-            valid_ranking=valid_ranking,
+            valid_ranking=valid_ranking, cox=cox, valid_cox=valid_cox,
             snapshot_cb=snapshot_cb,
             snapshot_interval_seconds=hp.get(
                 "resume_training_snapshot_interval_seconds", 1800.0),

@@ -35,6 +35,7 @@ LOSS_MAE = 8
 LOSS_LAMBDA_MART_NDCG = 9
 LOSS_FOCAL = 11            # binary focal loss (Lin et al. 2017)
 LOSS_XE_NDCG = 12          # cross-entropy NDCG (Bruch et al. 2020)
+LOSS_COX = 13              # Cox proportional hazards (survival)
 LOSS_RF = 100  # weighted-target mode (RF/CART): not a GBT loss
 
 
@@ -759,7 +760,8 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
               snapshot_cb=None,
               snapshot_interval_seconds: float = 1800.0,
               max_duration_seconds: float = -1.0,
-              custom_loss=None, ranking=None, valid_ranking=None):
+              custom_loss=None, ranking=None, valid_ranking=None,
+              cox=None, valid_cox=None):
     """The boosting loop (reference gradient_boosted_trees.cc:1460).
 
     Returns (trees, init_preds, training_logs). For multinomial loss,
@@ -798,6 +800,12 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             raise NotImplementedError(
                 "ranking is single-process for now (groups are not "
                 "row-shardable without group-aware sharding)")
+        init = 0.0
+        init_preds = [0.0]
+    elif cfg.loss == LOSS_COX:
+        if trainer.distributed:
+            raise NotImplementedError(
+                "survival risk sets are not row-shardable yet")
         init = 0.0
         init_preds = [0.0]
     elif cfg.loss in (LOSS_BINOMIAL, LOSS_FOCAL):
@@ -920,6 +928,9 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 fg, fh = _focal_grad_hess(pc, y, cfg.focal_gamma,
                                           cfg.focal_alpha)
                 trainer.gh.copy_(torch.stack([fg, fh], dim=1))
+            elif cfg.loss == LOSS_COX:
+                cg, chh = cox.grad_hess(pc)
+                trainer.gh.copy_(torch.stack([cg, chh], dim=1))
             elif multi:
                 ops.grad_hess_softmax(preds.view(-1), y, trainer.gh, C, c)
             else:
@@ -994,7 +1005,10 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             snapshot_cb(trees, it + 1, init_preds)
             t_last_snapshot = _time.monotonic()
         if has_valid:
-            if cfg.loss in (LOSS_LAMBDA_MART_NDCG, LOSS_XE_NDCG):
+            if cfg.loss == LOSS_COX:
+                vloss = valid_cox.loss(valid_preds[0]) \
+                    if valid_cox is not None else float("nan")
+            elif cfg.loss in (LOSS_LAMBDA_MART_NDCG, LOSS_XE_NDCG):
                 vloss = -valid_ranking.ndcg(valid_preds[0]) \
                     if valid_ranking is not None else float("nan")
             elif custom_loss is not None and custom_loss.loss is not None:

@@ -189,6 +189,7 @@ class Evaluation:
     mrr: Optional[float] = None
     auuc: Optional[float] = None
     qini: Optional[float] = None
+    cindex: Optional[float] = None
     confusion: Optional[np.ndarray] = None
     # closed-form 95% confidence intervals (lo, hi)
     accuracy_ci95: Optional[tuple] = None
@@ -197,8 +198,8 @@ class Evaluation:
     def to_dict(self) -> Dict:
         d = {"num_examples": self.num_examples}
         for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae",
-                  "ndcg", "mrr", "auuc", "qini", "accuracy_ci95",
-                  "auc_ci95"):
+                  "ndcg", "mrr", "auuc", "qini", "cindex",
+                  "accuracy_ci95", "auc_ci95"):
             v = getattr(self, k)
             if v is not None:
                 d[k] = v
@@ -207,7 +208,7 @@ class Evaluation:
     def __str__(self) -> str:
         parts = [f"num examples: {self.num_examples}"]
         for k in ("accuracy", "loss", "auc", "pr_auc", "rmse", "mae",
-                  "ndcg", "mrr", "auuc", "qini"):
+                  "ndcg", "mrr", "auuc", "qini", "cindex"):
             v = getattr(self, k)
             if v is not None:
                 parts.append(f"{k}: {v:.6g}")

@@ -221,6 +221,25 @@ class GenericModel:
             else:
                 labels = np.asarray(cols[lname], dtype=np.float32)
         n_classes = len(self.label_classes) if self.label_classes else 2
+        if self._task == Task.SURVIVAL_ANALYSIS:
+            from ydf_amd.learner.survival import CoxData
+            from ydf_amd.metric.survival import concordance_index
+
+            ecol = (self.metadata or {}).get("label_event_observed")
+            if cols is None or ecol not in cols:
+                raise ValueError(
+                    f"survival evaluation needs the event column "
+                    f"{ecol!r} in the dataset")
+            events = np.asarray(cols[ecol]).astype(bool)
+            acol = (self.metadata or {}).get("label_entry_age")
+            entry = np.asarray(cols[acol], np.float64) \
+                if acol and acol in cols else None
+            ev = Evaluation(num_examples=len(labels))
+            ev.cindex = concordance_index(labels, events, preds)
+            cd = CoxData(labels, events, torch.device("cpu"), entry)
+            ev.loss = cd.loss(torch.from_numpy(
+                np.asarray(preds, np.float32)))
+            return ev
         if self._task in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):
             from ydf_amd.metric.uplift import auuc_qini
 
