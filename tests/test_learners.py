@@ -632,3 +632,33 @@ def test_analyze_prediction():
     margin = float(m.predict_margin(X)[0, 0])
     total = sum(float(v) for v in ap.values())
     np.testing.assert_allclose(total, margin, rtol=1e-3, atol=1e-3)
+
+
+def test_selgb_sampling_ranking():
+    rng = np.random.RandomState(1)
+    q = np.repeat(np.arange(300), 10)
+    rel = (rng.rand(3000) < 0.15) * rng.randint(1, 5, 3000)
+    d = {"q": q, "rel": rel.astype(np.float32),
+         "f1": (rel + rng.randn(3000)).astype(np.float32),
+         "f2": rng.randn(3000).astype(np.float32)}
+    m = ydf.GradientBoostedTreesLearner(
+        label="rel", ranking_group="q", task=ydf.Task.RANKING,
+        num_trees=30, sampling_method="SELGB",
+        selective_gradient_boosting_ratio=0.2).train(d)
+    assert m.evaluate(d).ndcg > 0.9
+
+
+def test_hyperparameter_templates():
+    d = ydf.generate_synthetic_dataset(num_examples=2500, seed=12)
+    m = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=10, validation_ratio=0,
+        hyperparameter_template="benchmark_rank1").train(d)
+    assert (m.forest.cat_idx <= -2).sum() > 0  # template turns on oblique
+    m2 = ydf.RandomForestLearner(
+        label="LABEL", num_trees=5,
+        hyperparameter_template="better_default",
+        compute_oob_performances=False).train(d)
+    assert m2.num_trees() == 5
+    with pytest.raises(ValueError):
+        ydf.GradientBoostedTreesLearner(
+            label="LABEL", hyperparameter_template="nope")

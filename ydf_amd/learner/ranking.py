@@ -49,6 +49,10 @@ class RankingLambdas:
         self.sizes = self.valid.sum(dim=1)
         self.N = len(group_ids)
         self.device = device
+        # positive examples (relevance > 0) — used by SelGB sampling
+        pm = torch.zeros(self.N, dtype=torch.bool, device=device)
+        pm[self.safe_idx[self.valid & (self.rel > 0)]] = True
+        self.positive_mask = pm
 
     def _discounts(self, device):
         r = torch.arange(self.M, device=device, dtype=torch.float32)

@@ -24,6 +24,38 @@ from ydf_amd.model.specialized import (GradientBoostedTreesModel,
                                        RandomForestModel)
 from ydf_amd.utils.log import info
 
+# Predefined hyperparameter templates (reference
+# GetPredefinedHyperParameterTemplates: gradient_boosted_trees.cc /
+# random_forest.cc). growing_strategy=BEST_FIRST_GLOBAL entries are
+# dropped: trees grow level-wise here (documented deviation).
+HYPERPARAMETER_TEMPLATES = {
+    "GBT": {
+        "better_default@v1": {},
+        "benchmark_rank1@v1": {
+            "split_axis": "SPARSE_OBLIQUE",
+            "sparse_oblique_normalization": "MIN_MAX",
+            "sparse_oblique_num_projections_exponent": 1.0,
+        },
+    },
+    "RF": {
+        "better_default@v1": {"winner_take_all": True},
+    },
+}
+
+
+def _apply_template(hp: dict, family: str, name) -> None:
+    if not name:
+        return
+    table = HYPERPARAMETER_TEMPLATES[family]
+    if name not in table and f"{name}@v1" in table:
+        name = f"{name}@v1"
+    if name not in table:
+        raise ValueError(
+            f"unknown hyperparameter template {name!r}; available: "
+            f"{sorted(table)}")
+    hp.update(table[name])
+    info(f"applied hyperparameter template {name}")
+
 
 class GradientBoostedTreesLearner(GenericLearner):
     """GBT learner (reference learner/gradient_boosted_trees/; Python
@@ -44,6 +76,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  shrinkage: float = 0.1, subsample: float = 1.0,
                  sampling_method: str = "RANDOM",
                  goss_alpha: float = 0.2, goss_beta: float = 0.1,
+                 selective_gradient_boosting_ratio: float = 0.01,
                  min_examples: int = 5, l2_regularization: float = 0.0,
                  min_sum_hessian_in_leaf: float = 1e-3,
                  validation_ratio: float = 0.1,
@@ -69,6 +102,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  resume_training: bool = False,
                  resume_training_snapshot_interval_seconds: float = 1800.0,
                  maximum_training_duration_seconds: float = -1.0,
+                 hyperparameter_template: Optional[str] = None,
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
@@ -80,6 +114,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             num_trees=num_trees, max_depth=max_depth, shrinkage=shrinkage,
             subsample=subsample, sampling_method=sampling_method,
             goss_alpha=goss_alpha, goss_beta=goss_beta,
+            selective_gradient_boosting_ratio=(
+                selective_gradient_boosting_ratio),
             min_examples=min_examples,
             l2_regularization=l2_regularization,
             min_sum_hessian_in_leaf=min_sum_hessian_in_leaf,
@@ -111,6 +147,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             maximum_training_duration_seconds=(
                 maximum_training_duration_seconds),
         )
+        _apply_template(self.hyperparameters, "GBT",
+                        hyperparameter_template)
 
     def train(self, data, valid=None, verbose=None
               ) -> GradientBoostedTreesModel:
@@ -291,6 +329,7 @@ class GradientBoostedTreesLearner(GenericLearner):
             sampling_method=hp.get("sampling_method", "RANDOM"),
             goss_alpha=hp.get("goss_alpha", 0.2),
             goss_beta=hp.get("goss_beta", 0.1),
+            selgb_ratio=hp.get("selective_gradient_boosting_ratio", 0.01),
             n_classes=n_classes,
             seed=self.random_seed, num_candidate_features=ncand,
             early_stopping=(hp["early_stopping"] != "NONE"
@@ -489,9 +528,11 @@ class RandomForestLearner(GenericLearner):
                  sparse_oblique_projection_density_factor: float = 2.0,
                  sparse_oblique_normalization: str = "NONE",
                  sparse_oblique_weights: str = "BINARY",
+                 hyperparameter_template: Optional[str] = None,
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
+        self._hp_template = hyperparameter_template
         self.hyperparameters = dict(
             num_trees=num_trees, max_depth=max_depth,
             min_examples=min_examples,

@@ -53,6 +53,7 @@ class TrainerConfig:
     sampling_method: str = "RANDOM"   # RANDOM | GOSS
     goss_alpha: float = 0.2
     goss_beta: float = 0.1
+    selgb_ratio: float = 0.01   # SelGB share of kept negatives
     cat_smooth: float = 1.0      # l2_categorical_regularization
     n_classes: int = 2           # multinomial only
     focal_gamma: float = 2.0     # focal loss misprediction exponent
@@ -939,7 +940,21 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 # weighted loss: g,h scale linearly with the example weight
                 # (reference dataset/weight.h GetWeights path)
                 trainer.gh.mul_(trainer.weights.view(-1, 1))
-            if cfg.sampling_method == "GOSS":
+            if cfg.sampling_method == "SELGB" and ranking is not None:
+                # Selective Gradient Boosting (Lucchese et al. 2018;
+                # reference selective_gradient_boosting.h): keep every
+                # positive example, and the `selgb_ratio` share of
+                # negatives with the largest |gradient| per iteration
+                absg = trainer.gh[:, 0].abs()
+                pos = ranking.positive_mask
+                neg_scores = torch.where(pos, torch.full_like(absg, -1.0),
+                                         absg)
+                n_neg = int((~pos).sum().item())
+                k = max(1, int(cfg.selgb_ratio * n_neg))
+                thr_v = torch.kthvalue(
+                    neg_scores, max(1, N - k)).values
+                sample_mask = pos | (neg_scores >= thr_v)
+            elif cfg.sampling_method == "GOSS":
                 # Gradient-based one-side sampling (reference
                 # gradient_boosted_trees.cc:1488-1522 GOSS): keep the top
                 # alpha fraction by |g|, sample beta of the rest with
