@@ -662,3 +662,43 @@ def test_hyperparameter_templates():
     with pytest.raises(ValueError):
         ydf.GradientBoostedTreesLearner(
             label="LABEL", hyperparameter_template="nope")
+
+
+def test_best_first_global_growth(tmp_path):
+    """Leaf-wise growth (reference growing_strategy=BEST_FIRST_GLOBAL):
+    with the same leaf budget it must beat equal-size level-wise trees
+    on an interaction target, respect max_num_nodes, and persist."""
+    rng = np.random.RandomState(0)
+    n = 8000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2,
+         "label": np.where(2 * x1 - x2 + 0.5 * x1 * x2 > 0, "a", "b")}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=25, growing_strategy="BEST_FIRST_GLOBAL",
+        max_num_nodes=16, validation_ratio=0.1).train(d)
+    assert m.evaluate(d).accuracy > 0.99
+    # each tree has at most 16 leaves -> at most 31 nodes
+    f = m.forest
+    for t in range(f.n_trees):
+        lo, hi = f.tree_slice(t)
+        assert hi - lo <= 31
+    p1 = m.predict(d)
+    m.save(str(tmp_path / "bf"))
+    np.testing.assert_array_equal(
+        p1, ydf.load_model(str(tmp_path / "bf")).predict(d))
+
+
+def test_best_first_with_subsample():
+    """Out-of-sample rows route through the leaf-wise splits replay."""
+    rng = np.random.RandomState(1)
+    n = 6000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2,
+         "label": np.where(x1 - x2 > 0, "a", "b")}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15, subsample=0.6,
+        growing_strategy="BEST_FIRST_GLOBAL", max_num_nodes=8,
+        validation_ratio=0.1).train(d)
+    assert m.evaluate(d).accuracy > 0.97

@@ -73,6 +73,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                  label_entry_age: Optional[str] = None,
                  ndcg_truncation: int = 5,
                  num_trees: int = 300, max_depth: int = 6,
+                 growing_strategy: str = "LOCAL",
+                 max_num_nodes: int = 31,
                  shrinkage: float = 0.1, subsample: float = 1.0,
                  sampling_method: str = "RANDOM",
                  goss_alpha: float = 0.2, goss_beta: float = 0.1,
@@ -111,7 +113,9 @@ class GradientBoostedTreesLearner(GenericLearner):
         self.label_entry_age = label_entry_age
         self.ndcg_truncation = ndcg_truncation
         self.hyperparameters = dict(
-            num_trees=num_trees, max_depth=max_depth, shrinkage=shrinkage,
+            num_trees=num_trees, max_depth=max_depth,
+            growing_strategy=growing_strategy, max_num_nodes=max_num_nodes,
+            shrinkage=shrinkage,
             subsample=subsample, sampling_method=sampling_method,
             goss_alpha=goss_alpha, goss_beta=goss_beta,
             selective_gradient_boosting_ratio=(
@@ -339,6 +343,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             early_stopping_initial_iteration=(
                 hp["early_stopping_initial_iteration"]),
             cat_smooth=hp["l2_categorical_regularization"],
+            growing_strategy=hp.get("growing_strategy", "LOCAL"),
+            max_num_nodes=hp.get("max_num_nodes", 31),
             focal_gamma=hp.get("focal_loss_gamma", 2.0),
             focal_alpha=hp.get("focal_loss_alpha", 0.5),
             dart_dropout=(hp.get("dart_dropout", 0.01)

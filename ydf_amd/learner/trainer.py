@@ -79,6 +79,13 @@ class TrainerConfig:
     early_stopping_initial_iteration: int = 10
     # device memory budget for the per-level histogram buffer
     hist_budget_bytes: int = 1 << 31
+    # tree growth (reference growing_strategy, decision_tree.proto):
+    # LOCAL = level-wise (this trainer's native mode; the reference's
+    # depth-first local strategy visits the same splits),
+    # BEST_FIRST_GLOBAL = leaf-wise: repeatedly split the open leaf with
+    # the globally best gain until max_num_nodes leaves exist.
+    growing_strategy: str = "LOCAL"
+    max_num_nodes: int = 31
     # sparse-oblique splits (reference SparseObliqueSplit,
     # learner/decision_tree/decision_tree.proto:173): P random sparse
     # projections are sampled PER LEVEL, projected via GEMM (rocBLAS/MFMA),
@@ -105,6 +112,33 @@ class HostTree:
     # oblique nodes: {node_idx: (attrs i32[], weights f32[], threshold)}
     oblique: Optional[dict] = None
     # DART: final absolute leaf scale (bakes shrinkage + dropout rescales)
+    scale: float = 1.0
+
+
+@dataclasses.dataclass
+class BestFirstTree:
+    """Leaf-wise tree: splits in creation order over IMPLICIT keys
+    (children 2k+1 / 2k+2; depth capped at 24 so keys fit int32)."""
+
+    splits: list                 # [(key, feat, bin, gain)]
+    leaf_value: dict             # {key: float} (unscaled -G/(H+l2))
+    counts: dict                 # {key: float}
+    max_depth: int = 24
+    # compatibility views for feature-gain accounting
+    @property
+    def feat(self):
+        return np.asarray([s[1] for s in self.splits], dtype=np.int32)
+
+    @property
+    def gain(self):
+        return np.asarray([s[3] for s in self.splits], dtype=np.float32)
+
+    @property
+    def bin(self):
+        return np.asarray([s[2] for s in self.splits], dtype=np.int32)
+
+    oblique = None
+    masks = None
     scale: float = 1.0
 
 
@@ -355,14 +389,121 @@ class ForestTrainer:
 
     # -- one tree ---------------------------------------------------------
     def grow_tree(self, tree_idx: int,
-                  sample_mask: Optional[torch.Tensor] = None) -> HostTree:
+                  sample_mask: Optional[torch.Tensor] = None):
         """Grows one tree from self.gh; returns it as host arrays.
 
         On exit self.node_ids holds the final (sampled-rows) assignment and
         self.leaf_vals the per-node values; callers apply update_preds.
         """
+        if self.cfg.growing_strategy == "BEST_FIRST_GLOBAL":
+            return self.grow_tree_best_first(tree_idx, sample_mask)
+        self._bf_last = None
         self.grow_tree_device(tree_idx, sample_mask)
         return self.extract_host_tree()
+
+    def grow_tree_best_first(self, tree_idx: int,
+                             sample_mask=None) -> BestFirstTree:
+        """Leaf-wise growth (reference BEST_FIRST_GLOBAL,
+        decision_tree.proto growing_strategy): a host-driven loop pops
+        the open leaf with the best gain, splits it with the SAME
+        histogram/scan kernels (one node per launch, level_base = its
+        implicit key), until max_num_nodes leaves. Numerical + boolean
+        features only for now (categorical set-split masks index the
+        complete-tree buffer, which leaf-wise keys outgrow)."""
+        import heapq
+
+        cfg = self.cfg
+        assert self.has_cats is False, \
+            "BEST_FIRST_GLOBAL + categorical set-splits not supported yet"
+        assert self.P == 0, "BEST_FIRST_GLOBAL + oblique not supported"
+        assert self.mono is None
+        if sample_mask is None:
+            self.node_ids.zero_()
+        else:
+            self.node_ids.copy_(
+                torch.where(sample_mask,
+                            torch.zeros((), dtype=torch.int32,
+                                        device=self.device),
+                            torch.full((), -1, dtype=torch.int32,
+                                       device=self.device)))
+        abs0 = self.arange_buf[:1]           # value 0
+        fbuf = torch.empty(1, dtype=torch.int32, device=self.device)
+        bbuf = torch.empty(1, dtype=torch.int32, device=self.device)
+        stats = {}   # key -> (feat, bin, gain, G, H, C)
+        nsc = 0
+
+        def scan_node(key: int):
+            nonlocal nsc
+            hist_view = self.hist[:1]
+            hist_view.zero_()
+            ops.hist_build(self.bins, self.gh, self.node_ids, abs0,
+                           hist_view, key, 1, 0, 1)
+            self._allreduce(hist_view)
+            fm = self._feat_mask(1, tree_idx, nsc)
+            nsc += 1
+            ops.split_scan(hist_view, abs0, self.node_stats, self.bg_nf,
+                           self.bb_nf, self.best_feat, self.best_bin,
+                           self.best_gain, 0, 1, cfg.lambda_l2,
+                           cfg.min_hessian, cfg.min_examples, cfg.min_gain,
+                           feat_mask=fm)
+            f = int(self.best_feat[0].item())
+            b = int(self.best_bin[0].item())
+            g = float(self.best_gain[0].item())
+            ns = self.node_stats[0]
+            G, H, C = (float(ns[0].item()), float(ns[1].item()),
+                       float(ns[2].item()))
+            stats[key] = (f, b, g, G, H, C)
+
+        scan_node(0)
+        heap = []
+        tie = 0
+        if stats[0][0] >= 0 and stats[0][2] > 0:
+            heapq.heappush(heap, (-stats[0][2], tie, 0))
+            tie += 1
+        splits = []
+        n_leaves = 1
+        while heap and n_leaves < max(2, cfg.max_num_nodes):
+            _, _, key = heapq.heappop(heap)
+            if key >= (1 << 24):     # depth cap: keys stay int32-safe
+                continue
+            f, b, g, _, _, _ = stats[key]
+            splits.append((key, f, b, g))
+            fbuf.fill_(f)
+            bbuf.fill_(b)
+            ops.update_node_ids(self.bins, self.node_ids, abs0, fbuf,
+                                bbuf, key, 1)
+            for child in (2 * key + 1, 2 * key + 2):
+                scan_node(child)
+                cf, _, cg, _, _, _ = stats[child]
+                if cf >= 0 and cg > 0:
+                    heapq.heappush(heap, (-cg, tie, child))
+                    tie += 1
+            n_leaves += 1
+        split_keys = {k for k, *_ in splits}
+        leaf_value, counts = {}, {}
+        for key, (f, b, g, G, H, C) in stats.items():
+            counts[key] = C
+            if key not in split_keys:
+                leaf_value[key] = -G / (H + cfg.lambda_l2) if H > 0 else 0.0
+        tree = BestFirstTree(splits=splits, leaf_value=leaf_value,
+                             counts=counts)
+        # dense leaf renumbering so update_preds / route_rows keep their
+        # contract: node_ids become indices into leaf_vals[:n_leaves]
+        self._bf_last = tree
+        self._bf_keys = torch.tensor(sorted(leaf_value),
+                                     dtype=torch.int32,
+                                     device=self.device)
+        self.leaf_vals[: len(leaf_value)] = torch.tensor(
+            [leaf_value[k] for k in sorted(leaf_value)],
+            dtype=torch.float32, device=self.device)
+        self._bf_remap(self.node_ids)
+        return tree
+
+    def _bf_remap(self, node_ids: torch.Tensor) -> None:
+        ok = node_ids >= 0
+        idx = torch.searchsorted(
+            self._bf_keys, node_ids.clamp(min=0)).to(torch.int32)
+        node_ids.copy_(torch.where(ok, idx, node_ids))
 
     def grow_tree_device(self, tree_idx: int,
                          sample_mask: Optional[torch.Tensor] = None) -> None:
@@ -708,6 +849,7 @@ class ForestTrainer:
         assert (1 << (self.cfg.max_depth - 1)) <= self.dense_limit
         assert self.cfg.num_candidate_features <= 0
         assert self.cfg.oblique_projections == 0
+        assert self.cfg.growing_strategy == "LOCAL"
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             ops.grad_hess(preds, labels, self.gh, self.cfg.loss)
@@ -735,6 +877,23 @@ class ForestTrainer:
 
     def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor,
                    raw: Optional[torch.Tensor] = None):
+        if getattr(self, "_bf_last", None) is not None:
+            node_ids.zero_()
+            abs0 = self.arange_buf[:1]
+            fbuf = torch.empty(1, dtype=torch.int32, device=self.device)
+            bbuf = torch.empty(1, dtype=torch.int32, device=self.device)
+            for key, f, b, _ in self._bf_last.splits:
+                fbuf.fill_(f)
+                bbuf.fill_(b)
+                ops.update_node_ids(bins, node_ids, abs0, fbuf, bbuf,
+                                    key, 1)
+            self._bf_remap(node_ids)
+            return
+        return self._route_rows_levelwise(bins, node_ids, raw)
+
+    def _route_rows_levelwise(self, bins: torch.Tensor,
+                              node_ids: torch.Tensor,
+                              raw: Optional[torch.Tensor] = None):
         """Routes arbitrary rows through the latest tree (device arrays).
         With oblique training, per-level projections are replayed onto the
         given rows (raw defaults to the training/validation matrix matching

@@ -147,6 +147,45 @@ def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
     return feat, thr, left, cat_idx, masks, cover, obl
 
 
+def best_first_tree_to_flat(tree, boundaries: np.ndarray,
+                            leaf_scale: float = 1.0):
+    """Converts a leaf-wise BestFirstTree (implicit-key splits dict) to
+    flat arrays via adjacent-pair child allocation."""
+    split_of = {k: (f, b, g) for k, f, b, g in tree.splits}
+    feats, thrs, lefts, covers = [], [], [], []
+
+    def new_slot():
+        feats.append(-1)
+        thrs.append(0.0)
+        lefts.append(0)
+        covers.append(0.0)
+        return len(feats) - 1
+
+    def fill(key, slot):
+        covers[slot] = float(tree.counts.get(key, 0.0))
+        if key in split_of:
+            fi, b, _ = split_of[key]
+            feats[slot] = fi
+            thrs[slot] = float(boundaries[fi, b])
+            li = new_slot()
+            new_slot()
+            lefts[slot] = li
+            fill(2 * key + 1, li)
+            fill(2 * key + 2, li + 1)
+        else:
+            thrs[slot] = float(tree.leaf_value.get(key, 0.0)) * leaf_scale
+
+    root = new_slot()
+    fill(0, root)
+    return (np.asarray(feats, np.int32), np.asarray(thrs, np.float32),
+            np.asarray(lefts, np.int32),
+            np.full(len(feats), -1, np.int32),
+            np.zeros((0, 4), np.uint64),
+            np.asarray(covers, np.float32),
+            (np.zeros((0, 2), np.int32), np.zeros(0, np.int32),
+             np.zeros(0, np.float32)))
+
+
 def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
                       leaf_scale: float = 1.0, cat_feats=None) -> FlatForest:
     feats, thrs, lefts, roots, cidxs, mask_list, covers = \
@@ -157,8 +196,12 @@ def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
     obl_off = 0
     term_off = 0
     for t in trees:
-        f, th, lf, ci, mk, cv, (orng, oat, ow) = host_tree_to_flat(
-            t, boundaries, leaf_scale, cat_feats)
+        if isinstance(t, HostTree):
+            f, th, lf, ci, mk, cv, (orng, oat, ow) = host_tree_to_flat(
+                t, boundaries, leaf_scale, cat_feats)
+        else:  # BestFirstTree (leaf-wise growth)
+            f, th, lf, ci, mk, cv, (orng, oat, ow) = \
+                best_first_tree_to_flat(t, boundaries, leaf_scale)
         lf = np.where(f >= 0, lf + off, 0)
         ci = np.where(ci >= 0, ci + mask_off, ci)
         ci = np.where(ci <= -2, ci - obl_off, ci)
