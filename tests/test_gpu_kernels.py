@@ -356,3 +356,35 @@ def test_rf_deep_partitioned_i16_masked_vs_unmasked():
     np.testing.assert_array_equal(m_i16.forest.left, m_ref.forest.left)
     np.testing.assert_allclose(m_i16.forest.thr, m_ref.forest.thr,
                                rtol=1e-5, atol=1e-6)
+
+
+def test_best_first_gpu():
+    """Leaf-wise growth on the GPU (single-node hist/scan launches)."""
+    rng = np.random.RandomState(13)
+    n = 50000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    d = {"x1": x1, "x2": x2,
+         "label": np.where(2 * x1 - x2 + 0.5 * x1 * x2 > 0, "a", "b")}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15, growing_strategy="BEST_FIRST_GLOBAL",
+        max_num_nodes=16, validation_ratio=0.1, device="cuda").train(d)
+    assert m.evaluate(d, device="cuda").accuracy > 0.98
+
+
+def test_cox_gpu():
+    rng = np.random.RandomState(14)
+    n = 30000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    hazard = np.exp(x1 - 0.5 * x2)
+    T = rng.exponential(1.0 / hazard)
+    C = rng.exponential(2.0, n)
+    d = {"x1": x1, "x2": x2,
+         "time": np.minimum(T, C).astype(np.float32),
+         "event": T <= C}
+    m = ydf.GradientBoostedTreesLearner(
+        label="time", label_event_observed="event",
+        task=ydf.Task.SURVIVAL_ANALYSIS, num_trees=30,
+        validation_ratio=0.1, device="cuda").train(d)
+    assert m.evaluate(d).cindex > 0.7

@@ -107,6 +107,24 @@ class GenericModel:
     def input_feature_names(self) -> List[str]:
         return [c.name for c in self.dataspec.feature_columns]
 
+    def input_features(self):
+        """[(name, semantic)] like PYDF model.input_features()."""
+        return [(c.name, c.semantic) for c in self.dataspec.feature_columns]
+
+    def data_spec(self):
+        return self.dataspec
+
+    def to_tensorflow_saved_model(self, *a, **k):
+        raise ImportError(
+            "TensorFlow is not available in this environment; use "
+            "to_cpp()/to_java()/to_docker() for deployment, or "
+            "export_ydf_model() for the reference on-disk format")
+
+    def to_jax_function(self, *a, **k):
+        raise ImportError(
+            "JAX is not available in this environment; use to_cpp()/"
+            "to_java()/to_docker() for deployment")
+
     def num_trees(self) -> int:
         return self.forest.n_trees
 
