@@ -583,8 +583,10 @@ class ForestTrainer:
             feat_mask = self._feat_mask(n_active, tree_idx, level)
             lds_group = max(1, (160 * 1024 - min(4 * level_size, 32768))
                             // (ops.MAX_BINS * 16))
+            part_min = int(os.environ.get("YDFA_PART_MIN",
+                                          str(2 * lds_group)))
             use_partition = (self.device.type == "cuda"
-                             and n_active > 2 * lds_group)
+                             and n_active > part_min)
             # feature-interleaved variant: 16 features per 16-byte load
             # (see hist_build_gathered16_kernel); invalid under oblique
             # (virtual feature rows change per level)
