@@ -266,6 +266,9 @@ class ForestTrainer:
                 1 << int(math.floor(math.log2(self.max_slots))), 64)
         else:
             self.dense_limit = 1
+        self._i16_ok = (self.device.type == "cuda" and self.P == 0
+                        and self.F >= 32
+                        and os.environ.get("YDFA_HIST_I16", "1") == "1")
         self.build_map_buf = torch.empty(self.dense_limit, dtype=torch.int32,
                                          device=dev)
         self.derived_buf = torch.empty(self.dense_limit, dtype=torch.uint8,
@@ -588,9 +591,7 @@ class ForestTrainer:
             # (virtual feature rows change per level). When available it
             # pays from n_active > 2 (measured +34% on the RF bench vs
             # partitioning only past the LDS-group size).
-            i16_ok = (self.device.type == "cuda" and self.P == 0
-                      and self.F >= 32
-                      and os.environ.get("YDFA_HIST_I16", "1") == "1")
+            i16_ok = self._i16_ok
             part_min = int(os.environ.get(
                 "YDFA_PART_MIN", "2" if i16_ok else str(2 * lds_group)))
             use_partition = (self.device.type == "cuda"
@@ -639,6 +640,7 @@ class ForestTrainer:
             # its own contiguous row range (vs rescanning the full table
             # per group). Group boundaries come from local per-slot counts.
             row_order = None
+            offs_dev = None
             if use_partition:
                 # build_map (not slot_map): derived (histogram-subtraction)
                 # slots must not be built, so their rows sort to the end
@@ -652,11 +654,23 @@ class ForestTrainer:
                                    torch.full((), n_active,
                                               dtype=torch.int32,
                                               device=self.device))
-                row_order = torch.argsort(keys, stable=True).to(torch.int32)
                 slot_counts = torch.bincount(
-                    keys.long(), minlength=n_active + 1)[:n_active]
-                offs = np.zeros(n_active + 1, dtype=np.int64)
-                np.cumsum(slot_counts.cpu().numpy(), out=offs[1:])
+                    keys.long(), minlength=n_active + 1)
+                offs_dev = torch.zeros(n_active + 2, dtype=torch.int64,
+                                       device=self.device)
+                torch.cumsum(slot_counts, 0, out=offs_dev[1:])
+                if use_i16:
+                    # counting-sort scatter (one kernel) instead of the
+                    # generic merge argsort
+                    cursor = offs_dev[:-1].to(torch.int32).contiguous()
+                    row_order = torch.empty(self.N, dtype=torch.int32,
+                                            device=self.device)
+                    ops.row_scatter(keys, cursor, row_order)
+                else:
+                    row_order = torch.argsort(
+                        keys, stable=True).to(torch.int32)
+                    offs = np.zeros(n_active + 1, dtype=np.int64)
+                    offs[1:] = offs_dev[1:n_active + 1].cpu().numpy()
 
             for s0 in range(0, n_active, self.max_slots):
                 ns = min(self.max_slots, n_active - s0)
@@ -664,12 +678,14 @@ class ForestTrainer:
                 hist_view.zero_()
                 if use_i16:
                     spg = 2
-                    g_slots = list(range(s0, s0 + ns, spg)) + [s0 + ns]
-                    garr = offs[g_slots]
-                    goffs = torch.from_numpy(
-                        np.ascontiguousarray(garr)).to(self.device)
-                    n_groups = len(g_slots) - 1
-                    max_rows = int(np.diff(garr).max()) if n_groups else 0
+                    gidx = torch.cat([
+                        torch.arange(s0, s0 + ns, spg, dtype=torch.int64,
+                                     device=self.device),
+                        torch.tensor([s0 + ns], dtype=torch.int64,
+                                     device=self.device)])
+                    goffs = offs_dev[gidx].contiguous()
+                    n_groups = int(gidx.numel()) - 1
+                    max_rows = int(self.N)
                     maskbits = None
                     if feat_mask is not None:
                         # per-slot sampled-feature bits; feature 0 forced
@@ -814,12 +830,69 @@ class ForestTrainer:
             build_map = self.build_map_buf[:level_size]
             derived = self.derived_buf[:level_size]
         feat_mask = self._feat_mask(level_size, tree_idx, level)
+        use_i16d = self._i16_ok and level_size >= 4
+        # interleaved masked build disables subtraction (masks differ
+        # across levels); unmasked i16 keeps it
+        if use_i16d and feat_mask is not None and use_sub:
+            use_sub = False
+            if level > 0:
+                ops.plan_level(self.node_stats, self.best_feat, level_base,
+                               level_size, need, 0, self.build_map_buf,
+                               self.derived_buf)
+                build_map = self.build_map_buf[:level_size]
+                derived = self.derived_buf[:level_size]
         hist_view = self.hist[:level_size]
         hist_view.zero_()
-        ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
-                       hist_view, level_base, level_size, 0, level_size,
-                       filtered_hint=use_sub and os.environ.get(
-                           "YDFA_HIST_FILTER_SUB", "0") == "1")
+        if use_i16d:
+            if self._bins16 is None:
+                self._bins16 = ops.pack_bins16(self.bins)
+            rel_all = self.node_ids - level_base
+            keys = torch.where(
+                (rel_all >= 0) & (rel_all < level_size),
+                build_map[rel_all.clamp(0, level_size - 1)],
+                torch.full((), -1, dtype=torch.int32,
+                           device=self.device))
+            keys = torch.where(keys >= 0, keys,
+                               torch.full((), level_size,
+                                          dtype=torch.int32,
+                                          device=self.device))
+            slot_counts = torch.bincount(keys.long(),
+                                         minlength=level_size + 1)
+            offs_dev = torch.zeros(level_size + 2, dtype=torch.int64,
+                                   device=self.device)
+            torch.cumsum(slot_counts, 0, out=offs_dev[1:])
+            cursor = offs_dev[:-1].to(torch.int32).contiguous()
+            row_order = torch.empty(self.N, dtype=torch.int32,
+                                    device=self.device)
+            ops.row_scatter(keys, cursor, row_order)
+            maskbits = None
+            if feat_mask is not None:
+                F16 = (self.F + 15) // 16
+                fm = torch.zeros((level_size, F16 * 16),
+                                 dtype=torch.int32, device=self.device)
+                fm[:, :self.F] = feat_mask.int()
+                fm[:, 0] = 1
+                wb = (1 << torch.arange(16, dtype=torch.int32,
+                                        device=self.device))
+                maskbits = (fm.view(level_size, F16, 16)
+                            * wb).sum(-1).to(torch.int16).contiguous()
+            gidx = torch.cat([
+                torch.arange(0, level_size, 2, dtype=torch.int64,
+                             device=self.device),
+                torch.tensor([level_size], dtype=torch.int64,
+                             device=self.device)])
+            goffs = offs_dev[gidx].contiguous()
+            ops.hist_build_gathered16(
+                self._bins16, self.gh, self.node_ids, build_map,
+                row_order, goffs, hist_view, self.N, self.F, level_base,
+                level_size, 0, 2, int(gidx.numel()) - 1, int(self.N),
+                maskbits=maskbits)
+        else:
+            ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
+                           hist_view, level_base, level_size, 0,
+                           level_size,
+                           filtered_hint=use_sub and os.environ.get(
+                               "YDFA_HIST_FILTER_SUB", "0") == "1")
         self._allreduce(hist_view)
         if use_sub and derived is not None:
             ops.subtract_hist(hist_view, self.hist_prev, derived, level_size)

@@ -134,6 +134,17 @@ def hist_build_gathered16(bins16: torch.Tensor, gh: torch.Tensor,
     return hist
 
 
+def row_scatter(keys: torch.Tensor, cursor: torch.Tensor,
+                row_order: torch.Tensor):
+    """Counting-sort scatter: row_order[cursor[keys[r]]++] = r.
+    cursor length = number of distinct keys (block-local LDS counting
+    when it fits)."""
+    _C.gpu_row_scatter(keys.data_ptr(), cursor.data_ptr(),
+                       row_order.data_ptr(), keys.numel(),
+                       int(cursor.numel()), _stream())
+    return row_order
+
+
 def pack_bins16(bins: torch.Tensor) -> torch.Tensor:
     """[F, N] u8 -> [ceil(F/16), N, 16] u8 interleaved copy."""
     F, N = bins.shape

@@ -23,6 +23,8 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
                     const int32_t*, float*, int64_t, int, int, int, int, int,
                     int, int, uint8_t*, void*);
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
+void gpu_row_scatter(const int32_t*, int32_t*, int32_t*, int64_t, int,
+                     void*);
 void gpu_hist_build_gathered16(const uint8_t*, const float*,
                                const int32_t*, const int32_t*,
                                const int32_t*, const int64_t*, float*,
@@ -159,6 +161,14 @@ PYBIND11_MODULE(_ydf_ops, m) {
               P<int32_t>(slot_map), P<int32_t>(row_order), P<float>(hist),
               N, F, n_bins, level_base, level_size, slot0, n_slots, row_lo,
               row_hi, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_row_scatter",
+        [](uintptr_t keys, uintptr_t cursor, uintptr_t row_order,
+           int64_t N, int n_keys, uintptr_t stream) {
+          gpu_row_scatter(P<int32_t>(keys), P<int32_t>(cursor),
+                          P<int32_t>(row_order), N, n_keys,
+                          (void*)stream);
         },
         nogil);
   m.def("gpu_hist_build_gathered16",

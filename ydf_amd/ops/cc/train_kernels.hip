@@ -485,6 +485,52 @@ __global__ void hist_build_gathered_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Row partition scatter: counting-sort pass replacing the generic
+// device merge sort (keys are small slot ids). cursor[k] starts at the
+// slot's exclusive-prefix offset; arrival order within a slot is
+// arbitrary (histogram accumulation is order-insensitive up to ulps).
+// ---------------------------------------------------------------------------
+__global__ void row_scatter_kernel(const int32_t* __restrict__ keys,
+                                   int32_t* __restrict__ cursor,
+                                   int32_t* __restrict__ row_order,
+                                   int64_t N) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t r = i; r < N; r += stride) {
+    const int pos = atomicAdd(cursor + keys[r], 1);
+    row_order[pos] = (int32_t)r;
+  }
+}
+
+// Few-key variant: global atomics on a handful of counters serialize
+// (measured 2.3 ms/call at shallow levels), so each block counts its
+// contiguous row chunk in LDS, reserves one global range per key, then
+// places rows through fast LDS cursors.
+__global__ void row_scatter_block_kernel(const int32_t* __restrict__ keys,
+                                         int32_t* __restrict__ cursor,
+                                         int32_t* __restrict__ row_order,
+                                         int64_t N, int n_keys) {
+  extern __shared__ int lcnt[];
+  for (int k = threadIdx.x; k < n_keys; k += blockDim.x) lcnt[k] = 0;
+  __syncthreads();
+  const int64_t per = (N + gridDim.x - 1) / gridDim.x;
+  const int64_t c0 = (int64_t)blockIdx.x * per;
+  const int64_t c1 = min(c0 + per, N);
+  for (int64_t r = c0 + threadIdx.x; r < c1; r += blockDim.x)
+    atomicAdd(lcnt + keys[r], 1);
+  __syncthreads();
+  for (int k = threadIdx.x; k < n_keys; k += blockDim.x) {
+    const int c = lcnt[k];
+    lcnt[k] = c ? atomicAdd(cursor + k, c) : 0;
+  }
+  __syncthreads();
+  for (int64_t r = c0 + threadIdx.x; r < c1; r += blockDim.x) {
+    const int pos = atomicAdd(lcnt + keys[r], 1);
+    row_order[pos] = (int32_t)r;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Feature-interleaved gathered histograms: bins16 stores the binned
 // matrix as [ceil(F/16)][N][16] u8, so ONE 16-byte load fetches a row's
 // bins for 16 features (vs 16 scattered single-byte gathers from the
@@ -1229,6 +1275,26 @@ void gpu_hist_build_gathered(const uint8_t* bins, const float* gh,
                      (const float2*)gh, node_ids, slot_map, row_order, hist,
                      N, F, n_bins, level_base, level_size, slot0, n_slots,
                      lds_map, row_lo, row_hi, rpb);
+}
+
+void gpu_row_scatter(const int32_t* keys, int32_t* cursor,
+                     int32_t* row_order, int64_t N, int n_keys,
+                     void* stream) {
+  if (n_keys > 0 && n_keys <= 8192) {
+    int grid = 512;
+    if (N < (int64_t)grid * kBlock)
+      grid = (int)((N + kBlock - 1) / kBlock);
+    if (grid < 1) grid = 1;
+    hipLaunchKernelGGL(row_scatter_block_kernel, dim3(grid), dim3(kBlock),
+                       (size_t)n_keys * sizeof(int), (hipStream_t)stream,
+                       keys, cursor, row_order, N, n_keys);
+    return;
+  }
+  int grid = (int)((N + kBlock - 1) / kBlock);
+  if (grid > 8192) grid = 8192;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(row_scatter_kernel, dim3(grid), dim3(kBlock), 0,
+                     (hipStream_t)stream, keys, cursor, row_order, N);
 }
 
 void gpu_hist_build_gathered16(const uint8_t* bins16, const float* gh,
