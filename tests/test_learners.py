@@ -702,3 +702,54 @@ def test_best_first_with_subsample():
         growing_strategy="BEST_FIRST_GLOBAL", max_num_nodes=8,
         validation_ratio=0.1).train(d)
     assert m.evaluate(d).accuracy > 0.97
+
+
+def test_interrupt_returns_partial_model(binary_data, monkeypatch):
+    """Fault-injection analogue (reference simulate_worker_failure /
+    stop_training_trigger_): an interrupt mid-boosting returns a usable
+    partial model instead of crashing."""
+    from ydf_amd.learner import trainer as trainer_lib
+
+    orig = trainer_lib.ForestTrainer.grow_tree
+    calls = {"n": 0}
+
+    def failing(self, tree_idx, sample_mask=None):
+        calls["n"] += 1
+        if calls["n"] == 6:
+            raise KeyboardInterrupt
+        return orig(self, tree_idx, sample_mask)
+
+    monkeypatch.setattr(trainer_lib.ForestTrainer, "grow_tree", failing)
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=50, validation_ratio=0).train(binary_data)
+    assert 0 < m.num_trees() <= 5
+    assert m.evaluate(binary_data).accuracy > 0.7
+
+
+def test_crash_recovery_via_snapshot(tmp_path, binary_data, monkeypatch):
+    """Crash mid-training (exception after some snapshots), then a new
+    learner with resume_training=True continues from the snapshot and
+    finishes with the full tree count."""
+    from ydf_amd.learner import trainer as trainer_lib
+
+    wd = str(tmp_path / "work")
+    kw = dict(label="label", num_trees=30, validation_ratio=0,
+              working_dir=wd, resume_training=True,
+              resume_training_snapshot_interval_seconds=0.0)
+
+    orig = trainer_lib.ForestTrainer.grow_tree
+    calls = {"n": 0}
+
+    def crashing(self, tree_idx, sample_mask=None):
+        calls["n"] += 1
+        if calls["n"] == 12:
+            raise RuntimeError("simulated worker failure")
+        return orig(self, tree_idx, sample_mask)
+
+    monkeypatch.setattr(trainer_lib.ForestTrainer, "grow_tree", crashing)
+    with pytest.raises(RuntimeError):
+        ydf.GradientBoostedTreesLearner(**kw).train(binary_data)
+    monkeypatch.setattr(trainer_lib.ForestTrainer, "grow_tree", orig)
+    m = ydf.GradientBoostedTreesLearner(**kw).train(binary_data)
+    assert m.num_trees() == 30
+    assert m.evaluate(binary_data).accuracy > 0.85
