@@ -388,3 +388,28 @@ def test_cox_gpu():
         task=ydf.Task.SURVIVAL_ANALYSIS, num_trees=30,
         validation_ratio=0.1, device="cuda").train(d)
     assert m.evaluate(d).cindex > 0.7
+
+
+def test_quickscorer_matches_flat_kernel():
+    """HIP QuickScorer engine (reference quick_scorer_extended) must
+    reproduce the flat-node kernel exactly on a numerical GBT."""
+    from ydf_amd.model.forest import build_quickscorer
+
+    d = ydf.generate_synthetic_dataset(num_examples=20000,
+                                       num_numerical=8,
+                                       num_categorical=0, num_boolean=0,
+                                       seed=9)
+    m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=50,
+                                        validation_ratio=0,
+                                        device="cuda").train(d)
+    conds, offs, lv = build_quickscorer(m.forest)
+    X = torch.from_numpy(np.ascontiguousarray(
+        m._encode_features(d))).cuda()
+    want = m.predict_margin(X)[0]
+    out = torch.empty(X.shape[1], dtype=torch.float32, device="cuda")
+    ops.predict_forest_qs(
+        X, torch.from_numpy(conds).cuda(),
+        torch.from_numpy(offs).cuda(), torch.from_numpy(lv).cuda(), out,
+        init=float(m.init_predictions[0]), scale=m._leaf_scale())
+    np.testing.assert_allclose(out.cpu().numpy(), want.cpu().numpy(),
+                               rtol=1e-5, atol=1e-5)

@@ -283,3 +283,54 @@ def concat_forests(a: FlatForest, b: FlatForest) -> FlatForest:
         obl_attr=np.concatenate([a.obl_attr, b.obl_attr]).astype(np.int32),
         obl_w=np.concatenate([a.obl_w, b.obl_w]).astype(np.float32),
     )
+
+
+def build_quickscorer(forest: FlatForest):
+    """Precomputes QuickScorer tables (reference
+    quick_scorer_extended.h): per tree <= 64 in-order leaves, each
+    internal node's left-subtree leaf mask. Numerical conditions only.
+    Returns (conds i32 [C,4] = {feat, thr-bits, mask_lo, mask_hi},
+    offs i32 [T+1], leaf_vals f32 [T,64])."""
+    import struct as _struct
+
+    if len(forest.masks) or len(forest.obl_ranges):
+        raise ValueError("QuickScorer supports numerical conditions only")
+    conds = []
+    offs = [0]
+    leaf_vals = np.zeros((forest.n_trees, 64), dtype=np.float32)
+    for t in range(forest.n_trees):
+        leaves = []
+
+        def rec(n):
+            if forest.feat[n] < 0:
+                leaves.append(float(forest.thr[n]))
+                i = len(leaves) - 1
+                return i, i
+            li = int(forest.left[n])
+            l0, l1 = rec(li)
+            r0, r1 = rec(li + 1)
+            mask = 0
+            for b in range(l0, l1 + 1):
+                mask |= 1 << b
+            conds.append((int(forest.feat[n]), float(forest.thr[n]),
+                          mask))
+            return l0, r1
+
+        rec(int(forest.roots[t]))
+        if len(leaves) > 64:
+            raise ValueError(
+                f"tree {t} has {len(leaves)} leaves (> 64); QuickScorer "
+                "needs depth <= 6")
+        leaf_vals[t, :len(leaves)] = leaves
+        offs.append(len(conds))
+    packed = np.zeros((max(len(conds), 1), 4), dtype=np.int32)
+    for i, (feat, thr, mask) in enumerate(conds):
+        packed[i, 0] = feat
+        packed[i, 1] = np.frombuffer(
+            _struct.pack("<f", thr), dtype=np.int32)[0]
+        packed[i, 2] = np.frombuffer(
+            _struct.pack("<I", mask & 0xFFFFFFFF), dtype=np.int32)[0]
+        packed[i, 3] = np.frombuffer(
+            _struct.pack("<I", (mask >> 32) & 0xFFFFFFFF),
+            dtype=np.int32)[0]
+    return (packed, np.asarray(offs, dtype=np.int32), leaf_vals)

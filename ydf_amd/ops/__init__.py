@@ -328,6 +328,22 @@ def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
     return out
 
 
+def predict_forest_qs(X: torch.Tensor, conds: torch.Tensor,
+                      cond_offs: torch.Tensor, leaf_vals: torch.Tensor,
+                      out: torch.Tensor, init: float = 0.0,
+                      scale: float = 1.0):
+    """QuickScorer batch inference (GPU; numerical conditions,
+    <= 64 leaves/tree). conds [C,4] i32, cond_offs [T+1] i32,
+    leaf_vals [T,64] f32 (build_quickscorer output)."""
+    assert X.is_cuda
+    F, N = X.shape
+    T = cond_offs.numel() - 1
+    _C.gpu_predict_forest_qs(X.data_ptr(), N, F, conds.data_ptr(),
+                             cond_offs.data_ptr(), leaf_vals.data_ptr(),
+                             T, out.data_ptr(), init, scale, _stream())
+    return out
+
+
 def sigmoid(x: torch.Tensor, out: torch.Tensor):
     if x.is_cuda:
         _C.gpu_sigmoid(x.data_ptr(), out.data_ptr(), x.numel(), _stream())

@@ -56,6 +56,9 @@ void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const int32_t*, const int32_t*, const float*, int,
                         int, int, int, float*, float, float, void*);
 void gpu_sigmoid(const float*, float*, int64_t, void*);
+void gpu_predict_forest_qs(const float*, int64_t, int, const int32_t*,
+                           const int32_t*, const float*, int, float*,
+                           float, float, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int);
 void cpu_grad_hess(const float*, const float*, float*, int64_t, int);
@@ -279,6 +282,16 @@ PYBIND11_MODULE(_ydf_ops, m) {
                              P<float>(obl_w), has_cats,
                              tree_start, tree_step, n_trees, P<float>(out),
                              init, scale, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_predict_forest_qs",
+        [](uintptr_t X, int64_t N, int F, uintptr_t conds,
+           uintptr_t cond_offs, uintptr_t leaf_vals, int n_trees,
+           uintptr_t out, float init, float scale, uintptr_t stream) {
+          gpu_predict_forest_qs(P<float>(X), N, F, P<int32_t>(conds),
+                                P<int32_t>(cond_offs), P<float>(leaf_vals),
+                                n_trees, P<float>(out), init, scale,
+                                (void*)stream);
         },
         nogil);
   m.def("gpu_sigmoid",

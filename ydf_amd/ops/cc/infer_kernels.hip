@@ -163,6 +163,51 @@ __global__ void predict_forest_global_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// QuickScorer engine (reference serving/decision_forest/
+// quick_scorer_extended.h:24-61): each tree <= 64 leaves; every internal
+// node carries the 64-bit mask of its LEFT subtree's leaves. For each
+// condition where the example goes RIGHT (x > thr), those leaves are
+// removed; the exit leaf is the lowest surviving bit. Trades the flat
+// walk's 6 dependent node fetches for ~63 streamed, branch-free
+// condition evaluations per tree — conditions are block-uniform reads
+// (L2 broadcast), examples stay feature-major in LDS.
+// ---------------------------------------------------------------------------
+struct QSCond {
+  int32_t feat;
+  float thr;
+  unsigned long long mask;
+};
+
+__global__ void qs_predict_kernel(
+    const float* __restrict__ X, int64_t N, int F,
+    const QSCond* __restrict__ conds,
+    const int32_t* __restrict__ cond_offs,   // [T+1]
+    const float* __restrict__ leaf_vals,     // [T*64]
+    int n_trees, float* __restrict__ out, float init, float scale) {
+  extern __shared__ float xs[];  // [F][kTile]
+  const int tid = threadIdx.x;
+  const int64_t base = (int64_t)blockIdx.x * kTile;
+  const int64_t n_here = min((int64_t)kTile, N - base);
+  if (n_here <= 0) return;
+  for (int f = 0; f < F; ++f) {
+    if (tid < n_here) xs[f * kTile + tid] = X[(int64_t)f * N + base + tid];
+  }
+  __syncthreads();
+  if (tid >= n_here) return;
+  float acc = 0.f;
+  for (int t = 0; t < n_trees; ++t) {
+    unsigned long long live = ~0ull;
+    const int c1 = cond_offs[t + 1];
+    for (int c = cond_offs[t]; c < c1; ++c) {
+      const QSCond q = conds[c];
+      if (xs[q.feat * kTile + tid] > q.thr) live &= ~q.mask;
+    }
+    acc += leaf_vals[(int64_t)t * 64 + (__ffsll((long long)live) - 1)];
+  }
+  out[base + tid] = init + acc * scale;
+}
+
 __global__ void sigmoid_kernel(const float* __restrict__ in,
                                float* __restrict__ out, int64_t N) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -172,6 +217,18 @@ __global__ void sigmoid_kernel(const float* __restrict__ in,
 }
 
 extern "C" {
+
+void gpu_predict_forest_qs(const float* X, int64_t N, int F,
+                           const int32_t* conds, const int32_t* cond_offs,
+                           const float* leaf_vals, int n_trees, float* out,
+                           float init, float scale, void* stream) {
+  const size_t lds = (size_t)F * kTile * sizeof(float);
+  const int grid = (int)((N + kTile - 1) / kTile);
+  hipLaunchKernelGGL(qs_predict_kernel, dim3(grid), dim3(kTile), lds,
+                     (hipStream_t)stream, X, N, F,
+                     reinterpret_cast<const QSCond*>(conds), cond_offs,
+                     leaf_vals, n_trees, out, init, scale);
+}
 
 void gpu_predict_forest(const float* X, int64_t N, int F,
                         const int32_t* packed_nodes, const int32_t* roots,
