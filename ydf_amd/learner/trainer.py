@@ -666,12 +666,6 @@ class ForestTrainer:
                     row_order = torch.empty(self.N, dtype=torch.int32,
                                             device=self.device)
                     ops.row_scatter(keys, cursor, row_order)
-                    # permute gh/bins into partition order: the build
-                    # kernel then streams everything (slot implied by
-                    # row position)
-                    ro64 = row_order.long()
-                    gh_part = self.gh.index_select(0, ro64)
-                    bins16p = self._bins16.index_select(1, ro64)
                 else:
                     row_order = torch.argsort(
                         keys, stable=True).to(torch.int32)
@@ -684,7 +678,14 @@ class ForestTrainer:
                 hist_view.zero_()
                 if use_i16:
                     spg = 2
-                    slot_offs = offs_dev[s0:s0 + ns + 1].contiguous()
+                    gidx = torch.cat([
+                        torch.arange(s0, s0 + ns, spg, dtype=torch.int64,
+                                     device=self.device),
+                        torch.tensor([s0 + ns], dtype=torch.int64,
+                                     device=self.device)])
+                    goffs = offs_dev[gidx].contiguous()
+                    n_groups = int(gidx.numel()) - 1
+                    max_rows = int(self.N)
                     maskbits = None
                     if feat_mask is not None:
                         # per-slot sampled-feature bits; feature 0 forced
@@ -700,10 +701,11 @@ class ForestTrainer:
                         maskbits = (fm.view(ns, F16, 16)
                                     * weightsb).sum(-1).to(torch.int16)
                         maskbits = maskbits.contiguous()
-                    ops.hist_build_gathered16p(
-                        bins16p, gh_part, slot_offs, hist_view, self.N,
-                        self.F, ns, spg, int(self.N),
-                        maskbits=maskbits)
+                    ops.hist_build_gathered16(
+                        self._bins16, self.gh, self.node_ids, build_map,
+                        row_order, goffs, hist_view, self.N, self.F,
+                        level_base, level_size, s0, spg, n_groups,
+                        max_rows, maskbits=maskbits)
                 elif use_partition:
                     for g0 in range(s0, s0 + ns, lds_group):
                         g1 = min(g0 + lds_group, s0 + ns)
@@ -863,9 +865,6 @@ class ForestTrainer:
             row_order = torch.empty(self.N, dtype=torch.int32,
                                     device=self.device)
             ops.row_scatter(keys, cursor, row_order)
-            ro64 = row_order.long()
-            gh_part = self.gh.index_select(0, ro64)
-            bins16p = self._bins16.index_select(1, ro64)
             maskbits = None
             if feat_mask is not None:
                 F16 = (self.F + 15) // 16
@@ -877,10 +876,17 @@ class ForestTrainer:
                                         device=self.device))
                 maskbits = (fm.view(level_size, F16, 16)
                             * wb).sum(-1).to(torch.int16).contiguous()
-            slot_offs = offs_dev[:level_size + 1].contiguous()
-            ops.hist_build_gathered16p(
-                bins16p, gh_part, slot_offs, hist_view, self.N, self.F,
-                level_size, 2, int(self.N), maskbits=maskbits)
+            gidx = torch.cat([
+                torch.arange(0, level_size, 2, dtype=torch.int64,
+                             device=self.device),
+                torch.tensor([level_size], dtype=torch.int64,
+                             device=self.device)])
+            goffs = offs_dev[gidx].contiguous()
+            ops.hist_build_gathered16(
+                self._bins16, self.gh, self.node_ids, build_map,
+                row_order, goffs, hist_view, self.N, self.F, level_base,
+                level_size, 0, 2, int(gidx.numel()) - 1, int(self.N),
+                maskbits=maskbits)
         else:
             ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
                            hist_view, level_base, level_size, 0,

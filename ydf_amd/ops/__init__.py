@@ -145,19 +145,6 @@ def row_scatter(keys: torch.Tensor, cursor: torch.Tensor,
     return row_order
 
 
-def hist_build_gathered16p(bins16p: torch.Tensor, gh_part: torch.Tensor,
-                           slot_offs: torch.Tensor, hist: torch.Tensor,
-                           N: int, F: int, ns: int, spg: int, rows: int,
-                           maskbits=None):
-    """Streaming interleaved histograms over PERMUTED (partition-order)
-    bins16/gh: the slot is implied by row position."""
-    mb = maskbits.data_ptr() if maskbits is not None else 0
-    _C.gpu_hist_build_gathered16p(
-        bins16p.data_ptr(), gh_part.data_ptr(), slot_offs.data_ptr(),
-        hist.data_ptr(), mb, N, F, ns, spg, rows, _stream())
-    return hist
-
-
 def pack_bins16(bins: torch.Tensor) -> torch.Tensor:
     """[F, N] u8 -> [ceil(F/16), N, 16] u8 interleaved copy."""
     F, N = bins.shape

@@ -25,9 +25,6 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
 void gpu_row_scatter(const int32_t*, int32_t*, int32_t*, int64_t, int,
                      void*);
-void gpu_hist_build_gathered16p(const uint8_t*, const float*,
-                                const int64_t*, float*, const uint16_t*,
-                                int64_t, int, int, int, int64_t, void*);
 void gpu_hist_build_gathered16(const uint8_t*, const float*,
                                const int32_t*, const int32_t*,
                                const int32_t*, const int64_t*, float*,
@@ -175,16 +172,6 @@ PYBIND11_MODULE(_ydf_ops, m) {
           gpu_row_scatter(P<int32_t>(keys), P<int32_t>(cursor),
                           P<int32_t>(row_order), N, n_keys,
                           (void*)stream);
-        },
-        nogil);
-  m.def("gpu_hist_build_gathered16p",
-        [](uintptr_t bins16p, uintptr_t gh, uintptr_t slot_offs,
-           uintptr_t hist, uintptr_t maskbits, int64_t N, int F, int ns,
-           int spg, int64_t rows, uintptr_t stream) {
-          gpu_hist_build_gathered16p(
-              P<uint8_t>(bins16p), P<float>(gh), P<int64_t>(slot_offs),
-              P<float>(hist), P<uint16_t>(maskbits), N, F, ns, spg, rows,
-              (void*)stream);
         },
         nogil);
   m.def("gpu_hist_build_gathered16",

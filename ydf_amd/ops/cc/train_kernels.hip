@@ -622,90 +622,6 @@ __global__ void hist_build_gathered16_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Fully-streaming variant: gh and the interleaved bins are PERMUTED
-// into partition order once per level (torch index_select), so the slot
-// is implied by the row position (j >= slot boundary) and every access
-// in the hot loop is a coalesced stream — no node_ids/slot_map/row_order
-// gathers at all.
-// ---------------------------------------------------------------------------
-__global__ void hist_build_gathered16p_kernel(
-    const uint8_t* __restrict__ bins16p,   // [F16][N][16], partition order
-    const float2* __restrict__ gh_part,    // [N], partition order
-    const int64_t* __restrict__ slot_offs, // [ns+1] absolute row offsets
-    float* __restrict__ hist,
-    const uint16_t* __restrict__ maskbits, // [ns][F16] or null
-    int64_t N, int F, int n_bins, int ns, int spg, int n_chunks) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  double* lg = reinterpret_cast<double*>(smem);
-  unsigned long long* lp =
-      reinterpret_cast<unsigned long long*>(smem) + 1;
-  const int fg = blockIdx.x;
-  const int slot_lo = blockIdx.z * spg;
-  const int slot_hi = min(slot_lo + spg, ns);
-  const int tot = 16 * spg * n_bins;
-  {
-    unsigned long long* zz = reinterpret_cast<unsigned long long*>(smem);
-    for (int i = threadIdx.x; i < tot * 2; i += blockDim.x) zz[i] = 0ull;
-  }
-  __syncthreads();
-  const int64_t r0 = slot_offs[slot_lo];
-  const int64_t r1 = slot_offs[slot_hi];
-  const int64_t mid = (slot_hi - slot_lo) > 1 ? slot_offs[slot_lo + 1]
-                                              : r1;
-  const int64_t per = (r1 - r0 + n_chunks - 1) / n_chunks;
-  const int64_t j0 = r0 + (int64_t)blockIdx.y * per;
-  const int64_t j1 = min(j0 + per, r1);
-  const uint4* fb =
-      reinterpret_cast<const uint4*>(bins16p + (int64_t)fg * N * 16);
-  const int F16 = (F + 15) / 16;
-  unsigned m_of_slot[8];
-  if (maskbits != nullptr) {
-    for (int sI = 0; sI < spg && slot_lo + sI < ns; ++sI)
-      m_of_slot[sI] = maskbits[(int64_t)(slot_lo + sI) * F16 + fg];
-  }
-  for (int64_t j = j0 + threadIdx.x; j < j1; j += blockDim.x) {
-    const int slot = j >= mid ? 1 : 0;
-    unsigned m = 0xFFFFu;
-    if (maskbits != nullptr) {
-      m = m_of_slot[slot];
-      if (m == 0u) continue;
-    }
-    const float2 v = gh_part[j];
-    const unsigned long long hq =
-        (unsigned long long)(v.y * kHScale + 0.5f) |
-        ((unsigned long long)(v.y != 0.f) << 44);
-    const uint4 b = fb[j];
-    const unsigned words[4] = {b.x, b.y, b.z, b.w};
-    while (m) {
-      const int k = __ffs(m) - 1;
-      m &= m - 1;
-      const int bin = (words[k >> 2] >> ((k & 3) * 8)) & 0xFF;
-      const int cell = 2 * ((k * spg + slot) * n_bins + bin);
-      atomicAdd(lg + cell, (double)v.x);
-      atomicAdd(lp + cell, hq);
-    }
-  }
-  __syncthreads();
-  for (int idx = threadIdx.x; idx < tot; idx += blockDim.x) {
-    const double g = lg[2 * idx];
-    const unsigned long long pk = lp[2 * idx];
-    if (pk == 0ull && g == 0.0) continue;
-    const int k = idx / (spg * n_bins);
-    const int f = fg * 16 + k;
-    if (f >= F) continue;
-    const int rem = idx - k * spg * n_bins;
-    const int slot = rem / n_bins;
-    if (slot_lo + slot >= ns) continue;
-    const int bin = rem - slot * n_bins;
-    float* p = hist + ((int64_t)(slot_lo + slot) * F + f) * (n_bins * 3)
-               + bin * 3;
-    atomicAdd(p, (float)g);
-    atomicAdd(p + 1, (float)((double)(pk & kHMask) * (double)kHInvScale));
-    atomicAdd(p + 2, (float)(pk >> 44));
-  }
-}
-
-// ---------------------------------------------------------------------------
 // Split scan, stage A: one block per (slot, feature). Each thread owns one
 // bin; inclusive prefix sums of {g,h,c} over bins via LDS Hillis-Steele
 // (deterministic), then per-boundary gain and a deterministic block argmax.
@@ -1379,30 +1295,6 @@ void gpu_row_scatter(const int32_t* keys, int32_t* cursor,
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(row_scatter_kernel, dim3(grid), dim3(kBlock), 0,
                      (hipStream_t)stream, keys, cursor, row_order, N);
-}
-
-void gpu_hist_build_gathered16p(const uint8_t* bins16p, const float* gh,
-                                const int64_t* slot_offs, float* hist,
-                                const uint16_t* maskbits, int64_t N,
-                                int F, int ns, int spg, int64_t rows,
-                                void* stream) {
-  const int n_bins = kMaxBins;
-  const int F16 = (F + 15) / 16;
-  const int n_groups = (ns + spg - 1) / spg;
-  int chunks = 1;
-  const int target_blocks = 4096;
-  if ((int64_t)F16 * n_groups < target_blocks) {
-    const int want = target_blocks / (F16 * (n_groups > 0 ? n_groups : 1));
-    const int64_t cap = (rows / (n_groups > 0 ? n_groups : 1) + 511) / 512;
-    chunks = (int)std::min<int64_t>(std::max(1, want),
-                                    std::max<int64_t>(1, cap));
-  }
-  const size_t lds = (size_t)16 * spg * n_bins * 16;
-  hipLaunchKernelGGL(hist_build_gathered16p_kernel,
-                     dim3(F16, chunks, n_groups), dim3(kBlock), lds,
-                     (hipStream_t)stream, bins16p, (const float2*)gh,
-                     slot_offs, hist, maskbits, N, F, n_bins, ns, spg,
-                     chunks);
 }
 
 void gpu_hist_build_gathered16(const uint8_t* bins16, const float* gh,
