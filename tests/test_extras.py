@@ -162,3 +162,32 @@ def test_to_docker(tmp_path):
     finally:
         sys.path.remove(str(out))
         sys.modules.pop("main", None)
+
+
+def test_registry_usage_hooks_folds():
+    """Small reference-parity utils: name->factory registry
+    (registration.h), usage telemetry hooks (usage.h), fold generator
+    (fold_generator.h)."""
+    import ydf_amd as ydf
+
+    events = []
+    ydf.usage.register_on_inference(lambda **k: events.append(k))
+    try:
+        d = ydf.generate_synthetic_dataset(num_examples=400, seed=1)
+        lrn = ydf.get_learner("GRADIENT_BOOSTED_TREES")(
+            label="LABEL", num_trees=3, validation_ratio=0)
+        m = lrn.train(d)
+        m.predict(d)
+        assert events and events[0]["num_examples"] == 400
+    finally:
+        ydf.usage.clear()
+    with pytest.raises(KeyError):
+        ydf.get_learner("NOPE")
+    folds = ydf.generate_folds(103, 5, seed=2)
+    assert sum(len(f) for f in folds) == 103
+    assert len(np.unique(np.concatenate(folds))) == 103
+    groups = np.repeat(np.arange(20), 5)
+    gf = ydf.generate_folds(100, 4, groups=groups)
+    for f in gf:  # whole groups stay together
+        assert set(groups[f]) & set(
+            groups[np.setdiff1d(np.arange(100), f)]) == set()

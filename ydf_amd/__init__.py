@@ -45,6 +45,9 @@ from ydf_amd.model import tree
 
 # Model export / serving extras
 from ydf_amd.serving.deploy import to_docker
+from ydf_amd.utils import usage
+from ydf_amd.utils.folds import fold_splits, generate_folds
+from ydf_amd.utils.registry import get_learner
 from ydf_amd.serving.embed import to_cpp, to_java
 from ydf_amd.learner.extras import (
     BackwardSelectionFeatureSelector,

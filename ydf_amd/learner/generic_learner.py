@@ -93,6 +93,24 @@ class GenericLearner:
         del X
         return bins
 
+    def _notify_usage_start(self, ds) -> float:
+        import time as _time
+
+        from ydf_amd.utils import usage
+
+        usage.on_training_start(type(self).__name__, ds.n_examples)
+        return _time.monotonic()
+
+    def _notify_usage_end(self, ds, model, t0: float) -> None:
+        import time as _time
+
+        from ydf_amd.utils import usage
+
+        usage.on_training_end(
+            type(self).__name__, ds.n_examples,
+            model.num_trees() if hasattr(model, "num_trees") else None,
+            _time.monotonic() - t0)
+
     def _prepare(self, data, device: torch.device):
         """Dataset -> (VerticalDataset, binned u8 [F,N] on device,
         labels f32 [N] on device, padded boundary matrix np [F,n_cuts],

@@ -200,7 +200,10 @@ class GenericModel:
         multi-class -> [N, C]; regression/anomaly -> [N]. (Mirrors
         ydf GenericModel.predict semantics.)"""
         dev = torch.device(device) if device is not None else default_device()
+        from ydf_amd.utils import usage
+
         X_np = self._encode_features(data)
+        usage.on_inference(X_np.shape[1] if X_np.ndim == 2 else len(X_np))
         X = torch.from_numpy(np.ascontiguousarray(X_np)).to(dev)
         m = self.predict_margin(X)
         out = self._apply_activation(m)
@@ -395,7 +398,10 @@ class GenericModel:
         """Inference throughput benchmark (mirrors ydf model.benchmark;
         reference cli/benchmark_inference.cc)."""
         dev = torch.device(device) if device is not None else default_device()
+        from ydf_amd.utils import usage
+
         X_np = self._encode_features(data)
+        usage.on_inference(X_np.shape[1] if X_np.ndim == 2 else len(X_np))
         X = torch.from_numpy(np.ascontiguousarray(X_np)).to(dev)
         n = X.shape[1]
 
