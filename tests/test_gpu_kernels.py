@@ -413,3 +413,33 @@ def test_quickscorer_matches_flat_kernel():
         init=float(m.init_predictions[0]), scale=m._leaf_scale())
     np.testing.assert_allclose(out.cpu().numpy(), want.cpu().numpy(),
                                rtol=1e-5, atol=1e-5)
+
+
+def test_8bit_binned_engine_matches_flat():
+    """8-bit serving engine (pre-binned u8 features, bin-index
+    thresholds) must match the float flat-node kernel on our own
+    models (thresholds sit exactly on training cuts)."""
+    from ydf_amd.model.forest import (pack_binned_nodes,
+                                      padded_boundaries)
+
+    d = ydf.generate_synthetic_dataset(num_examples=30000,
+                                       num_numerical=10,
+                                       num_categorical=0, num_boolean=0,
+                                       seed=21)
+    m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=40,
+                                        validation_ratio=0,
+                                        device="cuda").train(d)
+    bnd = padded_boundaries(m.dataspec.feature_columns)
+    X = torch.from_numpy(np.ascontiguousarray(
+        m._encode_features(d))).cuda()
+    want = m.predict_margin(X)[0]
+    bins = torch.empty(X.shape, dtype=torch.uint8, device="cuda")
+    ops.bin_data(X, torch.from_numpy(bnd).cuda(), bins)
+    packed = torch.from_numpy(pack_binned_nodes(m.forest, bnd)).cuda()
+    roots = torch.from_numpy(m.forest.roots).cuda()
+    out = torch.empty(X.shape[1], dtype=torch.float32, device="cuda")
+    ops.predict_forest_binned(bins, packed, roots, out,
+                              init=float(m.init_predictions[0]),
+                              scale=m._leaf_scale())
+    np.testing.assert_allclose(out.cpu().numpy(), want.cpu().numpy(),
+                               rtol=1e-5, atol=1e-5)

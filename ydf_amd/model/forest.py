@@ -334,3 +334,31 @@ def build_quickscorer(forest: FlatForest):
             _struct.pack("<I", (mask >> 32) & 0xFFFFFFFF),
             dtype=np.int32)[0]
     return (packed, np.asarray(offs, dtype=np.int32), leaf_vals)
+
+
+def pack_binned_nodes(forest: FlatForest, boundaries: np.ndarray
+                      ) -> np.ndarray:
+    """Packs nodes for the 8-bit engine (reference
+    8bits_numerical_features.h): each numerical threshold becomes its
+    BIN INDEX in the training cut table ("bin > b" <=> "x > cut[b]"),
+    stored as an int in the thr slot. Numerical conditions only."""
+    if len(forest.masks) or len(forest.obl_ranges):
+        raise ValueError("8-bit engine supports numerical conditions only")
+    packed = np.zeros((forest.n_nodes, 4), dtype=np.int32)
+    packed[:, 0] = forest.feat
+    packed[:, 2] = forest.left
+    internal = forest.feat >= 0
+    thr_bits = forest.thr.view(np.int32).copy()
+    for n in np.nonzero(internal)[0]:
+        fi = int(forest.feat[n])
+        cuts = boundaries[fi]
+        b = int(np.searchsorted(cuts, forest.thr[n]))
+        if b >= len(cuts) or cuts[b] != forest.thr[n]:
+            # threshold not on a cut (e.g. imported model): fall back to
+            # the number of cuts strictly below it minus matching epsilon
+            b = int(np.searchsorted(cuts, forest.thr[n], side="right")) - 1
+            b = max(b, 0)
+        thr_bits[n] = b
+    packed[:, 1] = thr_bits
+    packed[:, 3] = -1
+    return packed

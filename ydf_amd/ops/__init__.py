@@ -328,6 +328,25 @@ def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
     return out
 
 
+def predict_forest_binned(B: torch.Tensor, packed: torch.Tensor,
+                          roots: torch.Tensor, out: torch.Tensor,
+                          tree_start: int = 0, tree_step: int = 1,
+                          n_trees: int = -1, init: float = 0.0,
+                          scale: float = 1.0):
+    """8-bit engine (GPU): B [F,N] u8 pre-binned features; `packed`
+    holds the split BIN index in the thr slot
+    (see model.forest binned packing)."""
+    assert B.is_cuda
+    F, N = B.shape
+    if n_trees < 0:
+        n_trees = roots.numel()
+    _C.gpu_predict_forest_binned(B.data_ptr(), N, F, packed.data_ptr(),
+                                 roots.data_ptr(), tree_start, tree_step,
+                                 n_trees, out.data_ptr(), init, scale,
+                                 _stream())
+    return out
+
+
 def predict_forest_qs(X: torch.Tensor, conds: torch.Tensor,
                       cond_offs: torch.Tensor, leaf_vals: torch.Tensor,
                       out: torch.Tensor, init: float = 0.0,

@@ -56,6 +56,9 @@ void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const int32_t*, const int32_t*, const float*, int,
                         int, int, int, float*, float, float, void*);
 void gpu_sigmoid(const float*, float*, int64_t, void*);
+void gpu_predict_forest_binned(const uint8_t*, int64_t, int,
+                               const int32_t*, const int32_t*, int, int,
+                               int, float*, float, float, void*);
 void gpu_predict_forest_qs(const float*, int64_t, int, const int32_t*,
                            const int32_t*, const float*, int, float*,
                            float, float, void*);
@@ -282,6 +285,17 @@ PYBIND11_MODULE(_ydf_ops, m) {
                              P<float>(obl_w), has_cats,
                              tree_start, tree_step, n_trees, P<float>(out),
                              init, scale, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_predict_forest_binned",
+        [](uintptr_t B, int64_t N, int F, uintptr_t packed_nodes,
+           uintptr_t roots, int tree_start, int tree_step, int n_trees,
+           uintptr_t out, float init, float scale, uintptr_t stream) {
+          gpu_predict_forest_binned(P<uint8_t>(B), N, F,
+                                    P<int32_t>(packed_nodes),
+                                    P<int32_t>(roots), tree_start,
+                                    tree_step, n_trees, P<float>(out),
+                                    init, scale, (void*)stream);
         },
         nogil);
   m.def("gpu_predict_forest_qs",

@@ -208,6 +208,42 @@ __global__ void qs_predict_kernel(
   out[base + tid] = init + acc * scale;
 }
 
+// ---------------------------------------------------------------------------
+// 8-bit engine (reference serving/decision_forest/
+// 8bits_numerical_features.h:82): features arrive PRE-BINNED as u8 (the
+// training-side quantile bins), thresholds are bin indices, the example
+// tile costs 1 byte per feature in LDS. For repeated serving of the
+// same rows this skips float encode + threshold lookups entirely.
+// ---------------------------------------------------------------------------
+__global__ void predict_forest_binned_kernel(
+    const uint8_t* __restrict__ B, int64_t N, int F,
+    const PackedNode* __restrict__ nodes,
+    const int32_t* __restrict__ roots, int tree_start, int tree_step,
+    int n_trees, float* __restrict__ out, float init, float scale) {
+  extern __shared__ uint8_t bs[];  // [F][kTile]
+  const int tid = threadIdx.x;
+  const int64_t base = (int64_t)blockIdx.x * kTile;
+  const int64_t n_here = min((int64_t)kTile, N - base);
+  if (n_here <= 0) return;
+  for (int f = 0; f < F; ++f) {
+    if (tid < n_here) bs[f * kTile + tid] = B[(int64_t)f * N + base + tid];
+  }
+  __syncthreads();
+  if (tid >= n_here) return;
+  float acc = init;
+  for (int tt = 0; tt < n_trees; ++tt) {
+    PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
+    while (nd.feat >= 0) {
+      // thr field holds the split BIN as an int bit-pattern
+      const int right =
+          (int)bs[nd.feat * kTile + tid] > __float_as_int(nd.thr) ? 1 : 0;
+      nd = nodes[nd.left + right];
+    }
+    acc += nd.thr;
+  }
+  out[base + tid] = init + (acc - init) * scale;
+}
+
 __global__ void sigmoid_kernel(const float* __restrict__ in,
                                float* __restrict__ out, int64_t N) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -217,6 +253,20 @@ __global__ void sigmoid_kernel(const float* __restrict__ in,
 }
 
 extern "C" {
+
+void gpu_predict_forest_binned(const uint8_t* B, int64_t N, int F,
+                               const int32_t* packed_nodes,
+                               const int32_t* roots, int tree_start,
+                               int tree_step, int n_trees, float* out,
+                               float init, float scale, void* stream) {
+  const size_t lds = (size_t)F * kTile;
+  const int grid = (int)((N + kTile - 1) / kTile);
+  hipLaunchKernelGGL(predict_forest_binned_kernel, dim3(grid), dim3(kTile),
+                     lds, (hipStream_t)stream, B, N, F,
+                     reinterpret_cast<const PackedNode*>(packed_nodes),
+                     roots, tree_start, tree_step, n_trees, out, init,
+                     scale);
+}
 
 void gpu_predict_forest_qs(const float* X, int64_t N, int F,
                            const int32_t* conds, const int32_t* cond_offs,
