@@ -191,3 +191,28 @@ def test_registry_usage_hooks_folds():
     for f in gf:  # whole groups stay together
         assert set(groups[f]) & set(
             groups[np.setdiff1d(np.arange(100), f)]) == set()
+
+
+def test_weighted_evaluation():
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(0)
+    n = 3000
+    x = rng.randn(n).astype(np.float32)
+    d = {"x": x, "label": np.where(x > 0, "a", "b"),
+         "w": np.where(x > 0, 5.0, 1.0).astype(np.float32)}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=5, features=["x"],
+        validation_ratio=0).train(d)
+    ev_u = m.evaluate(d)
+    ev_w = m.evaluate(d, weights="w")
+    assert ev_u.accuracy != ev_w.accuracy or ev_u.loss != ev_w.loss
+    # direct array form + regression path
+    yr = (2 * x).astype(np.float32)
+    dr = {"x": x, "label": yr}
+    mr = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=5,
+        validation_ratio=0).train(dr)
+    w = np.abs(x) + 0.1
+    ev = mr.evaluate(dr, weights=w)
+    assert ev.rmse is not None

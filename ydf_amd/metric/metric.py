@@ -227,11 +227,16 @@ class Evaluation:
 
 
 def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
-                         task, n_classes: int = 2) -> Evaluation:
-    """Evaluates raw predictions (mirrors ydf.evaluate_predictions)."""
+                         task, n_classes: int = 2,
+                         weights: Optional[np.ndarray] = None
+                         ) -> Evaluation:
+    """Evaluates raw predictions (mirrors ydf.evaluate_predictions).
+    `weights` makes accuracy/loss/rmse/mae example-weighted (reference
+    weighted evaluation, metric.cc)."""
     from ydf_amd.dataset.dataspec import Task
 
     ev = Evaluation(num_examples=len(labels))
+    w = None if weights is None else np.asarray(weights, np.float64)
     if task == Task.CLASSIFICATION:
         if predictions.ndim == 1:
             pred_cls = (predictions >= 0.5).astype(np.int64)
@@ -239,8 +244,23 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
             ev.pr_auc = pr_auc(labels > 0.5, predictions)
         else:
             pred_cls = predictions.argmax(axis=1)
-        ev.accuracy = accuracy(labels.astype(np.int64), pred_cls)
-        ev.loss = log_loss(labels, predictions)
+        correct = (labels.astype(np.int64) == pred_cls)
+        if w is None:
+            ev.accuracy = accuracy(labels.astype(np.int64), pred_cls)
+            ev.loss = log_loss(labels, predictions)
+        else:
+            ev.accuracy = float((correct * w).sum() / w.sum())
+            if predictions.ndim == 1:
+                p = np.clip(predictions.astype(np.float64), 1e-12,
+                            1 - 1e-12)
+                y = labels.astype(np.float64)
+                per = -(y * np.log(p) + (1 - y) * np.log(1 - p))
+            else:
+                per = -np.log(np.clip(
+                    predictions[np.arange(len(labels)),
+                                labels.astype(int)].astype(np.float64),
+                    1e-12, 1.0))
+            ev.loss = float((per * w).sum() / w.sum())
         ev.confusion = confusion_matrix(labels.astype(np.int64), pred_cls,
                                         n_classes)
         ev.accuracy_ci95 = accuracy_confidence_interval(
@@ -250,7 +270,12 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
             ev.auc_ci95 = auc_confidence_interval(
                 ev.auc, n_pos, len(labels) - n_pos)
     elif task == Task.REGRESSION:
-        ev.rmse = rmse(labels, predictions)
-        ev.mae = mae(labels, predictions)
+        if w is None:
+            ev.rmse = rmse(labels, predictions)
+            ev.mae = mae(labels, predictions)
+        else:
+            err = (labels - predictions).astype(np.float64)
+            ev.rmse = float(np.sqrt((err ** 2 * w).sum() / w.sum()))
+            ev.mae = float((np.abs(err) * w).sum() / w.sum())
         ev.loss = ev.rmse ** 2
     return ev

@@ -223,7 +223,10 @@ class GenericModel:
         return m.T.contiguous()
 
     # ------------------------------------------------------------------
-    def evaluate(self, data, device=None) -> Evaluation:
+    def evaluate(self, data, device=None, weights=None) -> Evaluation:
+        """`weights` names a column in `data` (or passes an array) for
+        example-weighted metrics; defaults to the training weights
+        column if the model recorded one."""
         cols = _to_column_dict(data) if not isinstance(data, VerticalDataset) \
             else None
         preds = self.predict(data, device=device)
@@ -241,6 +244,14 @@ class GenericModel:
                     dtype=np.float32, count=len(cols[lname]))
             else:
                 labels = np.asarray(cols[lname], dtype=np.float32)
+        w = None
+        if weights is None:
+            weights = (self.metadata or {}).get("weights_column")
+        if weights is not None and cols is not None:
+            w = np.asarray(cols[weights], np.float64) \
+                if isinstance(weights, str) and weights in cols \
+                else (np.asarray(weights, np.float64)
+                      if not isinstance(weights, str) else None)
         n_classes = len(self.label_classes) if self.label_classes else 2
         if self._task == Task.SURVIVAL_ANALYSIS:
             from ydf_amd.learner.survival import CoxData
@@ -279,7 +290,8 @@ class GenericModel:
             ev = Evaluation(num_examples=len(labels))
             ev.auuc, ev.qini = auuc_qini(labels, treat, preds)
             return ev
-        ev = evaluate_predictions(preds, labels, self._task, n_classes)
+        ev = evaluate_predictions(preds, labels, self._task, n_classes,
+                                  weights=w)
         if self._task == Task.RANKING:
             gcol = (self.metadata or {}).get("ranking_group")
             if gcol and cols is not None and gcol in cols:
