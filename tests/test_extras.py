@@ -216,3 +216,24 @@ def test_weighted_evaluation():
     w = np.abs(x) + 0.1
     ev = mr.evaluate(dr, weights=w)
     assert ev.rmse is not None
+
+
+def test_engine_api_cpu():
+    import ydf_amd as ydf
+
+    d = ydf.generate_synthetic_dataset(num_examples=1500,
+                                       num_numerical=5,
+                                       num_categorical=0, num_boolean=0,
+                                       seed=2)
+    m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=8,
+                                        validation_ratio=0).train(d)
+    assert set(m.list_compatible_engines()) == {"flat", "8bit", "qs"}
+    m.force_engine("qs")   # CPU predict falls back to the flat twin
+    assert m.predict(d, device="cpu").shape == (1500,)
+    with pytest.raises(ValueError):
+        m.force_engine("bogus")
+    d2 = ydf.generate_synthetic_dataset(num_examples=800,
+                                        num_categorical=2, seed=3)
+    m2 = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=4,
+                                         validation_ratio=0).train(d2)
+    assert m2.list_compatible_engines() == ["flat"]

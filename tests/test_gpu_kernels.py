@@ -443,3 +443,22 @@ def test_8bit_binned_engine_matches_flat():
                               scale=m._leaf_scale())
     np.testing.assert_allclose(out.cpu().numpy(), want.cpu().numpy(),
                                rtol=1e-5, atol=1e-5)
+
+
+def test_force_engine_equality_gpu():
+    """model.force_engine('qs'/'8bit'): engine-routed predictions match
+    the default flat kernel (reference engine cross-check discipline)."""
+    d = ydf.generate_synthetic_dataset(num_examples=20000,
+                                       num_numerical=6,
+                                       num_categorical=0, num_boolean=0,
+                                       seed=22)
+    m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=30,
+                                        validation_ratio=0,
+                                        device="cuda").train(d)
+    want = m.predict(d, device="cuda")
+    for eng in m.list_compatible_engines():
+        m.force_engine(eng)
+        got = m.predict(d, device="cuda")
+        np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-5,
+                                   err_msg=eng)
+    m.force_engine(None)

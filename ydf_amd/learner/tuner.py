@@ -34,10 +34,14 @@ class RandomSearchTuner:
         self.seed = seed
         self._choices: Dict[str, list] = {}
         if automatic_search_space:
+            # predefined space (reference PredefinedHyperParameterSpace,
+            # gradient_boosted_trees.cc hyperparameter templates)
             self.choice("shrinkage", [0.02, 0.05, 0.1, 0.15])
             self.choice("max_depth", [3, 4, 6, 8])
             self.choice("subsample", [0.6, 0.8, 1.0])
             self.choice("l2_regularization", [0.0, 0.1, 1.0])
+            self.choice("num_candidate_attributes_ratio", [0.5, 0.9, 1.0])
+            self.choice("split_axis", ["AXIS_ALIGNED", "SPARSE_OBLIQUE"])
 
     def choice(self, name: str, values: list, merge: bool = False):
         if merge and name in self._choices:

@@ -84,6 +84,7 @@ class GenericModel:
         self.activation = activation
         self.metadata = metadata or {}
         self._dev_forest: Dict[str, _DeviceForest] = {}
+        self._engine = None  # None = automatic (flat kernel)
         self.training_logs = None
         self.tuner_logs = None
         self._self_evaluation = None
@@ -159,6 +160,37 @@ class GenericModel:
             X[i] = encode_column(cols[src], spec)
         return X
 
+    def list_compatible_engines(self):
+        """Names of serving engines usable by this model (reference
+        ListCompatibleFastEngines; mirrors PYDF
+        model.list_compatible_fast_engines)."""
+        out = ["flat"]
+        pure_numerical = (len(self.forest.masks) == 0
+                          and len(self.forest.obl_ranges) == 0)
+        if pure_numerical:
+            out.append("8bit")
+            # QuickScorer needs <= 64 leaves per tree
+            f = self.forest
+            ok = True
+            for t in range(f.n_trees):
+                lo, hi = f.tree_slice(t)
+                if int((f.feat[lo:hi] < 0).sum()) > 64:
+                    ok = False
+                    break
+            if ok:
+                out.append("qs")
+        return out
+
+    def force_engine(self, name) -> None:
+        """Pins the serving engine used by predict() on GPU
+        (mirrors PYDF model.force_engine; None = automatic = flat)."""
+        if name is not None and name not in self.list_compatible_engines():
+            raise ValueError(
+                f"engine {name!r} not compatible; options: "
+                f"{self.list_compatible_engines()}")
+        self._engine = name
+        self._dev_forest.clear()
+
     def _forest_on(self, device: torch.device) -> _DeviceForest:
         key = str(device)
         if key not in self._dev_forest:
@@ -176,6 +208,10 @@ class GenericModel:
 
     def predict_margin(self, X: torch.Tensor) -> torch.Tensor:
         """Raw per-output forest sums/means. X [F,N] f32 on any device."""
+        eng = getattr(self, "_engine", None)
+        if eng in ("qs", "8bit") and X.is_cuda \
+                and self._n_outputs() == 1:
+            return self._predict_margin_engine(X, eng)
         df = self._forest_on(X.device)
         C = self._n_outputs()
         N = X.shape[1]
@@ -193,6 +229,43 @@ class GenericModel:
                                cat_idx=df.cat_idx, masks=df.masks,
                                packed=df.packed, obl_ranges=df.obl_ranges,
                                obl_attr=df.obl_attr, obl_w=df.obl_w)
+        return out
+
+    def _predict_margin_engine(self, X: torch.Tensor,
+                               eng: str) -> torch.Tensor:
+        from ydf_amd.model.forest import (build_quickscorer,
+                                          pack_binned_nodes,
+                                          padded_boundaries)
+
+        dev = X.device
+        key = f"{eng}:{dev}"
+        cache = self._dev_forest
+        N = X.shape[1]
+        out = torch.empty((1, N), dtype=torch.float32, device=dev)
+        init = float(self.init_predictions[0])
+        scale = self._leaf_scale()
+        if eng == "qs":
+            if key not in cache:
+                conds, offs, lv = build_quickscorer(self.forest)
+                cache[key] = (torch.from_numpy(conds).to(dev),
+                              torch.from_numpy(offs).to(dev),
+                              torch.from_numpy(lv).to(dev))
+            c, o, lv = cache[key]
+            ops.predict_forest_qs(X, c, o, lv, out[0], init=init,
+                                  scale=scale)
+            return out
+        if key not in cache:
+            bnd = padded_boundaries(self.dataspec.feature_columns)
+            cache[key] = (
+                torch.from_numpy(
+                    pack_binned_nodes(self.forest, bnd)).to(dev),
+                torch.from_numpy(self.forest.roots).to(dev),
+                torch.from_numpy(bnd).to(dev))
+        packed, roots, bnd_t = cache[key]
+        bins = torch.empty(X.shape, dtype=torch.uint8, device=dev)
+        ops.bin_data(X, bnd_t, bins)
+        ops.predict_forest_binned(bins, packed, roots, out[0], init=init,
+                                  scale=scale)
         return out
 
     def predict(self, data, device=None) -> np.ndarray:
