@@ -28,7 +28,11 @@ def init_from_env(backend: str = None) -> int:
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         if backend == "nccl":
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+            # modulo: lets N ranks share fewer GPUs (single-GPU rehearsal
+            # of the multi-rank launch path)
+            torch.cuda.set_device(
+                int(os.environ.get("LOCAL_RANK", "0"))
+                % max(torch.cuda.device_count(), 1))
         dist.init_process_group(backend=backend,
                                 timeout=datetime.timedelta(seconds=300))
     return dist.get_rank()
