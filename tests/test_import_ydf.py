@@ -200,3 +200,54 @@ def test_export_roundtrip_oblique(tmp_path, binary_data):
     assert (m2.forest.cat_idx <= -2).sum() > 0
     np.testing.assert_array_equal(m.predict(binary_data, device="cpu"),
                                   m2.predict(binary_data, device="cpu"))
+
+
+def test_na_value_routing_golden_ranking():
+    """The reference's na_value condition bits route MISSING inputs; the
+    ranking model's golden predictions only reproduce when NaNs follow
+    them (global imputation was off by up to 1.7 before)."""
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/synthetic_ranking_gbdt")
+    assert m.forest.has_na_routing
+    te = pd.read_csv(f"{BASE}/dataset/synthetic_ranking_test.csv")
+    p = m.predict(te, device="cpu")
+    g = pd.read_csv(
+        f"{BASE}/prediction/synthetic_ranking_gbdt_test.csv")["LABEL"]
+    assert np.abs(p - g.values).max() < 1e-4
+
+
+def test_golden_isolation_forest():
+    """Imported IF model vs the reference's sklearn-scored golden file
+    (c(n) conventions differ slightly between implementations)."""
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/gaussians_anomaly_if")
+    te = pd.read_csv(f"{BASE}/dataset/gaussians_test.csv")
+    p = m.predict(te, device="cpu")
+    g = np.loadtxt(f"{BASE}/prediction/gaussians_anomaly_if_skl.csv")
+    assert np.abs(p - g).max() < 0.01
+    assert np.corrcoef(p, g)[0, 1] > 0.999
+
+
+def test_import_uplift_rf():
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/sim_pte_categorical_uplift_rf")
+    assert m.task() == ydf.Task.CATEGORICAL_UPLIFT
+    assert m.metadata.get("uplift_treatment") == "treat"
+    te = pd.read_csv(f"{BASE}/dataset/sim_pte_test.csv")
+    p = m.predict(te, device="cpu")
+    assert -1.0 <= p.min() and p.max() <= 1.0
+    assert p.std() > 0.01
+
+
+def test_na_routing_trained_model_unaffected(binary_data, tmp_path):
+    """Our own trained models keep imputation-era behavior (all na bits
+    zero) and still round-trip."""
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=5,
+                                        validation_ratio=0).train(
+                                            binary_data)
+    assert not m.forest.has_na_routing
+    p1 = m.predict(binary_data, device="cpu")
+    m.save(str(tmp_path / "m"))
+    np.testing.assert_array_equal(
+        p1, ydf.load_model(str(tmp_path / "m")).predict(binary_data,
+                                                        device="cpu"))

@@ -196,7 +196,8 @@ def _set_tokens(cell) -> set:
     return {t for t in str(cell).split(" ") if t}
 
 
-def encode_column(arr: np.ndarray, spec: ColumnSpec) -> np.ndarray:
+def encode_column(arr: np.ndarray, spec: ColumnSpec,
+                  keep_na: bool = False) -> np.ndarray:
     """Encodes one raw column to float32 according to its spec.
 
     CATEGORICAL -> vocabulary index (0 = OOV); NUMERICAL -> float32 with
@@ -213,10 +214,25 @@ def encode_column(arr: np.ndarray, spec: ColumnSpec) -> np.ndarray:
                            count=len(arr))
     if spec.semantic == Semantic.CATEGORICAL:
         lookup = {item: i for i, item in enumerate(spec.vocab)}
-        out = np.fromiter((lookup.get(s, 0) for s in arr.astype(str)),
-                          dtype=np.float32, count=len(arr))
-        return out
+        miss = -1.0 if keep_na else 0.0
+
+        def code(cell):
+            if cell is None:
+                return miss
+            if isinstance(cell, float) and cell != cell:  # NaN
+                return miss
+            s = str(cell)
+            if keep_na and s in ("", "nan", "NA"):
+                return miss
+            return lookup.get(s, 0)
+
+        return np.fromiter((code(c) for c in arr), dtype=np.float32,
+                           count=len(arr))
     v = np.asarray(arr, dtype=np.float32).copy()
+    if keep_na:
+        # NA routing models (reference na_value): NaN passes through and
+        # the serving kernels follow the stored direction
+        return v
     bad = ~np.isfinite(v)
     if bad.any():
         v[bad] = spec.mean

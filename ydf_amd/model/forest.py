@@ -37,6 +37,11 @@ class FlatForest:
     obl_ranges: np.ndarray = None  # i32 [n_obl, 2] (start, count)
     obl_attr: np.ndarray = None    # i32 [total_terms]
     obl_w: np.ndarray = None       # f32 [total_terms]
+    # na_value routing (reference NodeCondition.na_value): when the
+    # node's input is MISSING (NaN numerical / -1 categorical code) the
+    # example goes right iff na_right[node]. All-zeros = imputation-era
+    # models (missing values were imputed at encode time).
+    na_right: np.ndarray = None    # u8 [total]
 
     def __post_init__(self):
         if self.cat_idx is None:
@@ -51,6 +56,12 @@ class FlatForest:
             self.obl_attr = np.zeros(0, dtype=np.int32)
         if self.obl_w is None:
             self.obl_w = np.zeros(0, dtype=np.float32)
+        if self.na_right is None:
+            self.na_right = np.zeros(len(self.feat), dtype=np.uint8)
+
+    @property
+    def has_na_routing(self) -> bool:
+        return bool(self.na_right.any())
 
     @property
     def has_cats(self) -> bool:

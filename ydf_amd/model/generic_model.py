@@ -57,6 +57,8 @@ class _DeviceForest:
             self.obl_w = torch.from_numpy(forest.obl_w).to(device)
         else:
             self.obl_ranges = self.obl_attr = self.obl_w = None
+        self.na_right = torch.from_numpy(forest.na_right).to(device) \
+            if forest.has_na_routing else None
         self.packed = None
         if device.type == "cuda":
             ci = self.cat_idx if self.cat_idx is not None else torch.full(
@@ -153,11 +155,12 @@ class GenericModel:
         specs = self.dataspec.feature_columns
         n = len(next(iter(cols.values()))) if cols else 0
         X = np.empty((len(specs), n), dtype=np.float32)
+        keep_na = self.forest.has_na_routing
         for i, spec in enumerate(specs):
             src = spec.set_source or spec.name
             if src not in cols:
                 raise ValueError(f"missing input feature {src!r}")
-            X[i] = encode_column(cols[src], spec)
+            X[i] = encode_column(cols[src], spec, keep_na=keep_na)
         return X
 
     def list_compatible_engines(self):
@@ -166,7 +169,8 @@ class GenericModel:
         model.list_compatible_fast_engines)."""
         out = ["flat"]
         pure_numerical = (len(self.forest.masks) == 0
-                          and len(self.forest.obl_ranges) == 0)
+                          and len(self.forest.obl_ranges) == 0
+                          and not self.forest.has_na_routing)
         if pure_numerical:
             out.append("8bit")
             # QuickScorer needs <= 64 leaves per tree
@@ -228,7 +232,8 @@ class GenericModel:
                                scale=self._leaf_scale(),
                                cat_idx=df.cat_idx, masks=df.masks,
                                packed=df.packed, obl_ranges=df.obl_ranges,
-                               obl_attr=df.obl_attr, obl_w=df.obl_w)
+                               obl_attr=df.obl_attr, obl_w=df.obl_w,
+                               na_right=df.na_right)
         return out
 
     def _predict_margin_engine(self, X: torch.Tensor,
@@ -407,12 +412,14 @@ class GenericModel:
         obl_ranges = np.ascontiguousarray(f.obl_ranges)
         obl_attr = np.ascontiguousarray(f.obl_attr)
         obl_w = np.ascontiguousarray(f.obl_w)
+        na_arr = np.ascontiguousarray(f.na_right)
         cpu_tree_shap(X.ctypes.data, N, F, feat.ctypes.data, thr.ctypes.data,
                       left.ctypes.data,
                       cat_idx.ctypes.data if f.has_cats else 0,
                       masks.ctypes.data if f.has_cats else 0,
                       obl_ranges.ctypes.data, obl_attr.ctypes.data,
                       obl_w.ctypes.data,
+                      na_arr.ctypes.data if f.has_na_routing else 0,
                       cover.ctypes.data, roots.ctypes.data, 0, 1,
                       f.n_trees, scale, 0.0, phi.ctypes.data)
         ev = cpu_forest_expected_value(
@@ -616,7 +623,8 @@ class GenericModel:
                  roots=self.forest.roots, cat_idx=self.forest.cat_idx,
                  masks=self.forest.masks, cover=self.forest.cover,
                  obl_ranges=self.forest.obl_ranges,
-                 obl_attr=self.forest.obl_attr, obl_w=self.forest.obl_w)
+                 obl_attr=self.forest.obl_attr, obl_w=self.forest.obl_w,
+                 na_right=self.forest.na_right)
         with open(os.path.join(path, "done"), "w") as f:
             f.write("")
 

@@ -224,6 +224,21 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
     elif 2 in node:  # regressor output
         reg = _msg(node[2][0])
         r.value = _f32(reg.get(1, [0.0])[0]) if 1 in reg else 0.0
+    elif 5 in node:  # uplift leaf (NodeUpliftOutput)
+        up = _msg(node[5][0])
+        te = _packed_floats(up[4][0]) if up.get(4) else []
+        if te:
+            r.value = float(te[0])
+        else:
+            swpt = _packed_doubles(up[2][0]) if up.get(2) else []
+            swpto = _packed_doubles(up[3][0]) if up.get(3) else []
+            rt = swpto[1] / swpt[1] if len(swpt) > 1 and swpt[1] else 0.0
+            rc = swpto[0] / swpt[0] if swpt and swpt[0] else 0.0
+            r.value = rt - rc
+    elif 6 in node:  # anomaly-detection leaf: example count; the
+        # importer converts to depth + c(n) while walking the tree
+        an = _msg(node[6][0])
+        r.value = ("anomaly", int(an.get(1, [0])[0]))
     if not r.is_leaf:
         cond = _msg(node[3][0])
         r.na_value = bool(cond.get(1, [0])[0])
@@ -288,6 +303,7 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
     feats, thrs, lefts, roots, cidx, masks, covers = \
         [], [], [], [], [], [], []
     obl_ranges, obl_attr, obl_w = [], [], []
+    na_right = []
     pos = 0
 
     def new_slot():
@@ -296,9 +312,10 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         lefts.append(0)
         cidx.append(-1)
         covers.append(0.0)
+        na_right.append(0)
         return len(feats) - 1
 
-    def fill_node(idx):
+    def fill_node(idx, depth=0):
         # consumes the next record into slot idx; children are allocated as
         # an ADJACENT pair (our flat layout needs right == left + 1, which
         # the on-disk DFS pre-order does not give for free)
@@ -307,7 +324,15 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         pos += 1
         covers[idx] = rec.cover
         if rec.is_leaf:
-            if wta and not isinstance(rec.value, list):
+            if isinstance(rec.value, tuple) and rec.value[0] == "anomaly":
+                # isolation-forest leaf: score contribution is the path
+                # length depth + c(n_leaf) (isolation_forest.h:51)
+                from ydf_amd.model.specialized import IsolationForestModel
+
+                thrs[idx] = depth + \
+                    IsolationForestModel.expected_path_length(
+                        float(rec.value[1]))
+            elif wta and not isinstance(rec.value, list):
                 # winner-take-all: the tree votes its majority class
                 # (binary: leaf P(class2) > 0.5 -> vote 1)
                 thrs[idx] = 1.0 if rec.value > 0.5 else 0.0
@@ -316,6 +341,7 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
                     if not isinstance(rec.value, list) else 0.0
             return
         feats[idx] = rec.attr
+        na_right[idx] = 1 if rec.na_value else 0
         if rec.mask is not None:
             cidx[idx] = len(masks)
             masks.append(rec.mask)
@@ -332,8 +358,8 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         li = new_slot()
         new_slot()
         lefts[idx] = li
-        fill_node(li)        # negative child first on disk
-        fill_node(li + 1)
+        fill_node(li, depth + 1)    # negative child first on disk
+        fill_node(li + 1, depth + 1)
 
     while pos < len(records):
         root = new_slot()
@@ -348,7 +374,8 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         cover=np.asarray(covers, np.float32),
         obl_ranges=np.asarray(obl_ranges, np.int32).reshape(-1, 2),
         obl_attr=np.asarray(obl_attr, np.int32),
-        obl_w=np.asarray(obl_w, np.float32))
+        obl_w=np.asarray(obl_w, np.float32),
+        na_right=np.asarray(na_right, np.uint8))
 
 
 def load_ydf_model(path: str, file_prefix: str = ""):
@@ -358,26 +385,30 @@ def load_ydf_model(path: str, file_prefix: str = ""):
     with open(os.path.join(path, file_prefix + "data_spec.pb"), "rb") as f:
         columns, disc_bounds = parse_data_spec(f.read())
     task = Task(header.get(2, [1])[0]) if header.get(2, [1])[0] in (
-        1, 2, 3, 6) else Task.CLASSIFICATION
+        1, 2, 3, 4, 5, 6) else Task.CLASSIFICATION
     label_idx = header.get(3, [len(columns) - 1])[0]
+    if label_idx >= (1 << 62):  # negative varint (-1): no label column
+        label_idx = -1
     input_features = []
     for v in header.get(5, []):
         if isinstance(v, bytes):
             input_features.extend(_packed_varints(v))
         else:
             input_features.append(v)
-    label_name = columns[label_idx].name
+    has_label = 0 <= label_idx < len(columns)
+    label_name = columns[label_idx].name if has_label else None
     # order dataspec features like the model's input_features
     feat_cols = [columns[i] for i in input_features]
-    dataspec = DataSpecification(columns=feat_cols + [columns[label_idx]],
-                                 label=label_name)
+    dataspec = DataSpecification(
+        columns=feat_cols + ([columns[label_idx]] if has_label else []),
+        label=label_name)
     # remap attribute indices (original column idx -> dense feature idx)
     remap = {ci: i for i, ci in enumerate(input_features)}
 
     gbt_hdr_path = os.path.join(
         path, file_prefix + "gradient_boosted_trees_header.pb")
     rf_hdr_path = os.path.join(path, file_prefix + "random_forest_header.pb")
-    label_vocab = columns[label_idx].vocab
+    label_vocab = columns[label_idx].vocab if has_label else None
     classes = list(label_vocab[1:]) if label_vocab else None
     n_classes = len(classes) if classes else 2
 
@@ -418,14 +449,37 @@ def load_ydf_model(path: str, file_prefix: str = ""):
         wta = bool(rh.get(3, [1])[0])
         forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
                              wta=wta and task == Task.CLASSIFICATION)
+        meta = {"imported_from": "yggdrasil-decision-forests",
+                "winner_take_all": wta}
+        if task in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):
+            # AbstractModel.uplift_treatment_col_idx = 9
+            tcol = header.get(9, [-1])[0]
+            if 0 <= tcol < len(columns):
+                meta["uplift_treatment"] = columns[tcol].name
+                tv = columns[tcol].vocab
+                meta["treatment_vocab"] = list(tv[1:]) if tv else None
         model = RandomForestModel(
             forest=_remap_forest(forest, remap), dataspec=dataspec,
             task=task, label_classes=classes,
             init_predictions=[0.0],
             num_trees_per_iter=1, activation="identity",
-            metadata={"imported_from": "yggdrasil-decision-forests",
-                      "winner_take_all": wta})
+            metadata=meta)
         return model
+    if_hdr_path = os.path.join(path,
+                               file_prefix + "isolation_forest_header.pb")
+    if os.path.exists(if_hdr_path):
+        # isolation_forest.proto: num_trees=2, num_examples_per_trees=4
+        from ydf_amd.model.specialized import IsolationForestModel
+
+        with open(if_hdr_path, "rb") as f:
+            ih = _msg(f.read())
+        forest = _read_trees(path, file_prefix, disc_bounds, n_classes)
+        return IsolationForestModel(
+            forest=_remap_forest(forest, remap), dataspec=dataspec,
+            task=Task.ANOMALY_DETECTION, init_predictions=[0.0],
+            num_trees_per_iter=1, activation="identity",
+            num_examples_per_tree=int(ih.get(4, [256])[0]),
+            metadata={"imported_from": "yggdrasil-decision-forests"})
     raise ValueError(f"unsupported or missing model header in {path}")
 
 

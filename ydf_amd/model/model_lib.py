@@ -32,7 +32,9 @@ def load_model(path: str) -> GenericModel:
                         obl_ranges=z["obl_ranges"] if "obl_ranges" in z
                         else None,
                         obl_attr=z["obl_attr"] if "obl_attr" in z else None,
-                        obl_w=z["obl_w"] if "obl_w" in z else None)
+                        obl_w=z["obl_w"] if "obl_w" in z else None,
+                        na_right=z["na_right"] if "na_right" in z
+                        else None)
     cls = MODEL_CLASSES.get(header["model_type"], GenericModel)
     model = cls(
         forest=forest,
@@ -63,7 +65,8 @@ def serialize_model(model: GenericModel) -> bytes:
                  cat_idx=model.forest.cat_idx, masks=model.forest.masks,
                  cover=model.forest.cover,
                  obl_ranges=model.forest.obl_ranges,
-                 obl_attr=model.forest.obl_attr, obl_w=model.forest.obl_w)
+                 obl_attr=model.forest.obl_attr, obl_w=model.forest.obl_w,
+                 na_right=model.forest.na_right)
         zf.writestr("forest.npz", fbuf.getvalue())
     return buf.getvalue()
 

@@ -295,7 +295,8 @@ def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
                    tree_start: int = 0, tree_step: int = 1,
                    n_trees: int = -1, init: float = 0.0, scale: float = 1.0,
                    cat_idx=None, masks=None, packed=None,
-                   obl_ranges=None, obl_attr=None, obl_w=None):
+                   obl_ranges=None, obl_attr=None, obl_w=None,
+                   na_right=None):
     """Flat-forest batch inference. X [F,N] f32, out [N] f32. On GPU,
     pass `packed` (pack_forest_nodes output) for the fast path. Oblique
     nodes (cat_idx <= -2) read obl_ranges [n,2] i32 / obl_attr i32 /
@@ -307,14 +308,16 @@ def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
     orr = obl_ranges.data_ptr() if obl_ranges is not None else 0
     oa = obl_attr.data_ptr() if obl_attr is not None else 0
     ow = obl_w.data_ptr() if obl_w is not None else 0
-    special = cat_idx is not None or obl_ranges is not None
+    nr = na_right.data_ptr() if na_right is not None else 0
+    special = cat_idx is not None or obl_ranges is not None \
+        or na_right is not None
     if X.is_cuda:
         if packed is None:
             ci = cat_idx if cat_idx is not None else torch.full(
                 (feat.numel(),), -1, dtype=torch.int32, device=X.device)
             packed = pack_forest_nodes(feat, thr, left, ci)
         _C.gpu_predict_forest(X.data_ptr(), N, F, packed.data_ptr(),
-                              roots.data_ptr(), mk, orr, oa, ow,
+                              roots.data_ptr(), mk, orr, oa, ow, nr,
                               1 if special else 0, tree_start,
                               tree_step, n_trees, out.data_ptr(), init,
                               scale, _stream())
@@ -322,7 +325,7 @@ def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
         ci = cat_idx.data_ptr() if cat_idx is not None else 0
         _C.cpu_predict_forest(X.data_ptr(), N, F, feat.data_ptr(),
                               thr.data_ptr(), left.data_ptr(),
-                              roots.data_ptr(), ci, mk, orr, oa, ow,
+                              roots.data_ptr(), ci, mk, orr, oa, ow, nr,
                               tree_start, tree_step,
                               n_trees, out.data_ptr(), init, scale)
     return out

@@ -53,7 +53,8 @@ void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
 // infer_kernels.hip
 void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const int32_t*, const unsigned long long*,
-                        const int32_t*, const int32_t*, const float*, int,
+                        const int32_t*, const int32_t*, const float*,
+                        const uint8_t*, int,
                         int, int, int, float*, float, float, void*);
 void gpu_sigmoid(const float*, float*, int64_t, void*);
 void gpu_predict_forest_binned(const uint8_t*, int64_t, int,
@@ -87,11 +88,13 @@ void cpu_binary_logloss(const float*, const float*, float*, int64_t);
 void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const float*, const int32_t*, const int32_t*,
                         const int32_t*, const unsigned long long*,
-                        const int32_t*, const int32_t*, const float*, int,
+                        const int32_t*, const int32_t*, const float*,
+                        const uint8_t*, int,
                         int, int, float*, float, float);
 void cpu_tree_shap(const float*, int64_t, int, const int32_t*, const float*,
                    const int32_t*, const int32_t*, const unsigned long long*,
                    const int32_t*, const int32_t*, const float*,
+                   const uint8_t*,
                    const float*, const int32_t*, int, int, int, float, float,
                    float*);
 void cpu_forest_expected_value(const int32_t*, const float*, const int32_t*,
@@ -275,14 +278,16 @@ PYBIND11_MODULE(_ydf_ops, m) {
   m.def("gpu_predict_forest",
         [](uintptr_t X, int64_t N, int F, uintptr_t packed_nodes,
            uintptr_t roots, uintptr_t masks, uintptr_t obl_ranges,
-           uintptr_t obl_attr, uintptr_t obl_w, int has_cats, int tree_start,
+           uintptr_t obl_attr, uintptr_t obl_w, uintptr_t na_right,
+           int has_cats, int tree_start,
            int tree_step, int n_trees, uintptr_t out, float init,
            float scale, uintptr_t stream) {
           gpu_predict_forest(P<float>(X), N, F, P<int32_t>(packed_nodes),
                              P<int32_t>(roots),
                              P<unsigned long long>(masks),
                              P<int32_t>(obl_ranges), P<int32_t>(obl_attr),
-                             P<float>(obl_w), has_cats,
+                             P<float>(obl_w), P<uint8_t>(na_right),
+                             has_cats,
                              tree_start, tree_step, n_trees, P<float>(out),
                              init, scale, (void*)stream);
         },
@@ -421,12 +426,14 @@ PYBIND11_MODULE(_ydf_ops, m) {
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
            uintptr_t left, uintptr_t cat_idx, uintptr_t masks,
            uintptr_t obl_ranges, uintptr_t obl_attr, uintptr_t obl_w,
+           uintptr_t na_right,
            uintptr_t cover, uintptr_t roots, int tree_start, int tree_step,
            int n_trees, float scale, float init, uintptr_t phi_out) {
           cpu_tree_shap(P<float>(X), N, F, P<int32_t>(feat), P<float>(thr),
                         P<int32_t>(left), P<int32_t>(cat_idx),
                         P<unsigned long long>(masks), P<int32_t>(obl_ranges),
                         P<int32_t>(obl_attr), P<float>(obl_w),
+                        P<uint8_t>(na_right),
                         P<float>(cover),
                         P<int32_t>(roots), tree_start, tree_step, n_trees,
                         scale, init, P<float>(phi_out));
@@ -447,14 +454,16 @@ PYBIND11_MODULE(_ydf_ops, m) {
         [](uintptr_t X, int64_t N, int F, uintptr_t feat, uintptr_t thr,
            uintptr_t left, uintptr_t roots, uintptr_t cat_idx,
            uintptr_t masks, uintptr_t obl_ranges, uintptr_t obl_attr,
-           uintptr_t obl_w, int tree_start, int tree_step, int n_trees,
+           uintptr_t obl_w, uintptr_t na_right, int tree_start,
+           int tree_step, int n_trees,
            uintptr_t out, float init, float scale) {
           cpu_predict_forest(P<float>(X), N, F, P<int32_t>(feat),
                              P<float>(thr), P<int32_t>(left),
                              P<int32_t>(roots), P<int32_t>(cat_idx),
                              P<unsigned long long>(masks),
                              P<int32_t>(obl_ranges), P<int32_t>(obl_attr),
-                             P<float>(obl_w), tree_start,
+                             P<float>(obl_w), P<uint8_t>(na_right),
+                             tree_start,
                              tree_step, n_trees, P<float>(out), init, scale);
         },
         nogil);

@@ -459,7 +459,8 @@ void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
                         const int32_t* roots, const int32_t* cat_idx,
                         const unsigned long long* masks,
                         const int32_t* obl_ranges, const int32_t* obl_attr,
-                        const float* obl_w, int tree_start,
+                        const float* obl_w, const uint8_t* na_right,
+                        int tree_start,
                         int tree_step, int n_trees, float* out, float init,
                         float scale) {
   (void)F;
@@ -476,8 +477,12 @@ void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
         while (f >= 0) {
           int right;
           const int ci = cat_idx ? cat_idx[n] : -1;
-          if (ci >= 0) {
-            int cbin = (int)X[(int64_t)f * N + k];
+          const float xv = X[(int64_t)f * N + k];
+          if (na_right != nullptr &&
+              (std::isnan(xv) || (xv < 0.f && ci >= 0))) {
+            right = na_right[n];  // missing input: stored na_value side
+          } else if (ci >= 0) {
+            int cbin = (int)xv;
             cbin = cbin < 0 ? 0 : (cbin > 255 ? 255 : cbin);
             right = (int)((masks[(int64_t)ci * 4 + (cbin >> 6)]
                            >> (cbin & 63)) & 1ull);
@@ -489,7 +494,7 @@ void cpu_predict_forest(const float* X, int64_t N, int F, const int32_t* feat,
               dot += obl_w[s0 + q] * X[(int64_t)obl_attr[s0 + q] * N + k];
             right = dot > thr[n] ? 1 : 0;
           } else {
-            right = X[(int64_t)f * N + k] > thr[n] ? 1 : 0;
+            right = xv > thr[n] ? 1 : 0;
           }
           n = left[n] + right;
           f = feat[n];
@@ -575,6 +580,7 @@ struct ShapCtx {
   const int32_t* obl_ranges;
   const int32_t* obl_attr;
   const float* obl_w;
+  const uint8_t* na_right;
   const float* cover;
   double* phi;  // [F+1]
   int64_t row;
@@ -583,6 +589,10 @@ struct ShapCtx {
 
 int ShapGoesRight(const ShapCtx& c, int node) {
   const int ci = c.cat_idx ? c.cat_idx[node] : -1;
+  if (c.na_right != nullptr) {
+    const float xv = c.X[(int64_t)c.feat[node] * c.N + c.row];
+    if (std::isnan(xv) || (xv < 0.f && ci >= 0)) return c.na_right[node];
+  }
   if (ci >= 0) {
     int cb = (int)c.X[(int64_t)c.feat[node] * c.N + c.row];
     cb = cb < 0 ? 0 : (cb > 255 ? 255 : cb);
@@ -643,6 +653,7 @@ extern "C" void cpu_tree_shap(const float* X, int64_t N, int F,
                               const unsigned long long* masks,
                               const int32_t* obl_ranges,
                               const int32_t* obl_attr, const float* obl_w,
+                              const uint8_t* na_right,
                               const float* cover, const int32_t* roots,
                               int tree_start, int tree_step, int n_trees,
                               float scale, float init, float* phi_out) {
@@ -661,7 +672,7 @@ extern "C" void cpu_tree_shap(const float* X, int64_t N, int F,
             const int root = roots[tree_start + (int64_t)tt * tree_step];
             ShapCtx c{X,          N,        feat,  thr,   left,
                       cat_idx,    masks,    obl_ranges, obl_attr, obl_w,
-                      cover,      phi.data(), row,  scale};
+                      na_right,   cover,    phi.data(), row,  scale};
             PathElem dummy[1];
             ShapRecurse(c, root, dummy, 0, 1.f, 1.f, -1);
           }

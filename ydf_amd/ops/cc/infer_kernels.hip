@@ -42,7 +42,8 @@ __global__ void predict_forest_lds_kernel(
     const unsigned long long* __restrict__ masks,
     const int32_t* __restrict__ obl_ranges,
     const int32_t* __restrict__ obl_attr,
-    const float* __restrict__ obl_w, int has_cats,
+    const float* __restrict__ obl_w,
+    const uint8_t* __restrict__ na_right, int has_cats,
     int tree_start, int tree_step, int n_trees, float* __restrict__ out,
     float init, float scale) {
   extern __shared__ float xs[];  // [F][kTile]
@@ -57,7 +58,7 @@ __global__ void predict_forest_lds_kernel(
   __syncthreads();
   if (tid >= n_here) return;
   float acc = init;
-  if (!has_cats) {
+  if (!has_cats && na_right == nullptr) {
     // pure-numerical fast path: 4 trees walk in parallel per thread (a
     // single walk is a chain of DEPENDENT L2 gathers), one 16-B packed
     // node load per step.
@@ -91,11 +92,17 @@ __global__ void predict_forest_lds_kernel(
     }
   } else {
     for (int tt = 0; tt < n_trees; ++tt) {
-      PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
+      int ni = roots[tree_start + (int64_t)tt * tree_step];
+      PackedNode nd = nodes[ni];
       while (nd.feat >= 0) {
         int right;
-        if (nd.cat_idx >= 0) {
-          const float xv = xs[nd.feat * kTile + tid];
+        const float xv = xs[nd.feat * kTile + tid];
+        if (na_right != nullptr && (isnan(xv) || xv < 0.f) &&
+            (isnan(xv) || nd.cat_idx >= 0)) {
+          // missing input (NaN numerical / negative categorical code):
+          // follow the stored na_value direction
+          right = na_right[ni];
+        } else if (nd.cat_idx >= 0) {
           int c = (int)xv;
           c = c < 0 ? 0 : (c > 255 ? 255 : c);
           right = (int)((masks[(int64_t)nd.cat_idx * 4 + (c >> 6)]
@@ -109,9 +116,10 @@ __global__ void predict_forest_lds_kernel(
             dot += obl_w[s0 + k] * xs[obl_attr[s0 + k] * kTile + tid];
           right = dot > nd.thr ? 1 : 0;
         } else {
-          right = xs[nd.feat * kTile + tid] > nd.thr ? 1 : 0;
+          right = xv > nd.thr ? 1 : 0;
         }
-        nd = nodes[nd.left + right];
+        ni = nd.left + right;
+        nd = nodes[ni];
       }
       acc += nd.thr;
     }
@@ -127,7 +135,8 @@ __global__ void predict_forest_global_kernel(
     const unsigned long long* __restrict__ masks,
     const int32_t* __restrict__ obl_ranges,
     const int32_t* __restrict__ obl_attr,
-    const float* __restrict__ obl_w, int has_cats,
+    const float* __restrict__ obl_w,
+    const uint8_t* __restrict__ na_right, int has_cats,
     int tree_start, int tree_step, int n_trees, float* __restrict__ out,
     float init, float scale) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -135,11 +144,15 @@ __global__ void predict_forest_global_kernel(
   for (int64_t k = i; k < N; k += stride) {
     float acc = init;
     for (int tt = 0; tt < n_trees; ++tt) {
-      PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
+      int ni = roots[tree_start + (int64_t)tt * tree_step];
+      PackedNode nd = nodes[ni];
       while (nd.feat >= 0) {
         int right;
-        if (has_cats && nd.cat_idx >= 0) {
-          const float xv = X[(int64_t)nd.feat * N + k];
+        const float xv = X[(int64_t)nd.feat * N + k];
+        if (na_right != nullptr && (isnan(xv) || xv < 0.f) &&
+            (isnan(xv) || nd.cat_idx >= 0)) {
+          right = na_right[ni];
+        } else if (has_cats && nd.cat_idx >= 0) {
           int c = (int)xv;
           c = c < 0 ? 0 : (c > 255 ? 255 : c);
           right = (int)((masks[(int64_t)nd.cat_idx * 4 + (c >> 6)]
@@ -153,9 +166,10 @@ __global__ void predict_forest_global_kernel(
             dot += obl_w[s0 + kk] * X[(int64_t)obl_attr[s0 + kk] * N + k];
           right = dot > nd.thr ? 1 : 0;
         } else {
-          right = X[(int64_t)nd.feat * N + k] > nd.thr ? 1 : 0;
+          right = xv > nd.thr ? 1 : 0;
         }
-        nd = nodes[nd.left + right];
+        ni = nd.left + right;
+        nd = nodes[ni];
       }
       acc += nd.thr;
     }
@@ -284,7 +298,8 @@ void gpu_predict_forest(const float* X, int64_t N, int F,
                         const int32_t* packed_nodes, const int32_t* roots,
                         const unsigned long long* masks,
                         const int32_t* obl_ranges, const int32_t* obl_attr,
-                        const float* obl_w, int has_cats, int tree_start,
+                        const float* obl_w, const uint8_t* na_right,
+                        int has_cats, int tree_start,
                         int tree_step, int n_trees, float* out, float init,
                         float scale, void* stream) {
   const PackedNode* nodes =
@@ -294,16 +309,16 @@ void gpu_predict_forest(const float* X, int64_t N, int F,
     const int grid = (int)((N + kTile - 1) / kTile);
     hipLaunchKernelGGL(predict_forest_lds_kernel, dim3(grid), dim3(kTile), lds,
                        (hipStream_t)stream, X, N, F, nodes, roots, masks,
-                       obl_ranges, obl_attr, obl_w, has_cats, tree_start,
-                       tree_step, n_trees, out, init, scale);
+                       obl_ranges, obl_attr, obl_w, na_right, has_cats,
+                       tree_start, tree_step, n_trees, out, init, scale);
   } else {
     int grid = (int)((N + kTile - 1) / kTile);
     if (grid > 4096) grid = 4096;
     if (grid < 1) grid = 1;
     hipLaunchKernelGGL(predict_forest_global_kernel, dim3(grid), dim3(kTile),
                        0, (hipStream_t)stream, X, N, F, nodes, roots, masks,
-                       obl_ranges, obl_attr, obl_w, has_cats, tree_start,
-                       tree_step, n_trees, out, init, scale);
+                       obl_ranges, obl_attr, obl_w, na_right, has_cats,
+                       tree_start, tree_step, n_trees, out, init, scale);
   }
 }
 
