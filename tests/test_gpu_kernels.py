@@ -462,3 +462,40 @@ def test_force_engine_equality_gpu():
         np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-5,
                                    err_msg=eng)
     m.force_engine(None)
+
+
+def test_na_routing_gpu_vs_cpu():
+    """na_value routing (imported reference models): NaN inputs follow
+    the stored per-node direction identically on CPU and GPU."""
+    from ydf_amd.model.forest import FlatForest
+    from ydf_amd.model.generic_model import _DeviceForest
+
+    rng = np.random.RandomState(6)
+    F, N = 4, 40000
+    X = rng.randn(F, N).astype(np.float32)
+    X[0, rng.rand(N) < 0.3] = np.nan
+    X[1, rng.rand(N) < 0.3] = np.nan
+    forest = FlatForest(
+        feat=np.array([0, 1, 1, -1, -1, -1, -1], np.int32),
+        thr=np.array([0.2, -0.1, 0.4, 1.0, 2.0, 3.0, 4.0], np.float32),
+        left=np.array([1, 3, 5, 0, 0, 0, 0], np.int32),
+        roots=np.array([0], np.int32),
+        na_right=np.array([1, 0, 1, 0, 0, 0, 0], np.uint8))
+    outs = {}
+    for dev in ("cpu", "cuda"):
+        d = torch.device(dev)
+        df = _DeviceForest(forest, d)
+        Xd = torch.from_numpy(X).to(d)
+        o = torch.empty(N, dtype=torch.float32, device=d)
+        ops.predict_forest(Xd, df.feat, df.thr, df.left, df.roots, o,
+                           cat_idx=df.cat_idx, masks=df.masks,
+                           packed=df.packed, na_right=df.na_right)
+        outs[dev] = o.cpu().numpy()
+    # reference semantics in numpy
+    r0 = np.where(np.isnan(X[0]), 1, (X[0] > 0.2).astype(int))
+    rl = np.where(np.isnan(X[1]), 0, (X[1] > -0.1).astype(int))
+    rr = np.where(np.isnan(X[1]), 1, (X[1] > 0.4).astype(int))
+    want = np.where(r0 == 1, np.where(rr == 1, 4.0, 3.0),
+                    np.where(rl == 1, 2.0, 1.0)).astype(np.float32)
+    np.testing.assert_allclose(outs["cpu"], want, atol=1e-6)
+    np.testing.assert_allclose(outs["cuda"], want, atol=1e-6)
