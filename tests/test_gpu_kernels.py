@@ -499,3 +499,14 @@ def test_na_routing_gpu_vs_cpu():
                     np.where(rl == 1, 2.0, 1.0)).astype(np.float32)
     np.testing.assert_allclose(outs["cpu"], want, atol=1e-6)
     np.testing.assert_allclose(outs["cuda"], want, atol=1e-6)
+
+
+def test_deep_mlp_gpu():
+    """Deep learners train on the GPU through torch (rocBLAS/MIOpen)."""
+    d = ydf.generate_synthetic_dataset(num_examples=4000,
+                                       num_numerical=5,
+                                       num_categorical=1, seed=30)
+    m = ydf.MultiLayerPerceptronLearner(
+        label="LABEL", num_epochs=15, num_layers=2, layer_size=48,
+        device="cuda").train(d)
+    assert m.evaluate(d, device="cuda").auc > 0.75
