@@ -251,3 +251,16 @@ def test_na_routing_trained_model_unaffected(binary_data, tmp_path):
     np.testing.assert_array_equal(
         p1, ydf.load_model(str(tmp_path / "m")).predict(binary_data,
                                                         device="cpu"))
+
+
+def test_import_categorical_set_model_sst():
+    """sst_binary_class_gbdt: bag-of-words CATEGORICAL_SET conditions
+    (vocab 2001 > 256-bit masks); host set-walk serving reproduces the
+    model's published quality and persists."""
+    pd = pytest.importorskip("pandas")
+    m = ydf.load_ydf_model(f"{BASE}/model/sst_binary_class_gbdt")
+    assert m.forest.has_set_conditions
+    te = pd.read_csv(f"{BASE}/dataset/sst_binary_test.csv")
+    ev = m.evaluate(te)
+    assert ev.accuracy > 0.78
+    assert ev.auc > 0.86

@@ -37,6 +37,13 @@ class FlatForest:
     obl_ranges: np.ndarray = None  # i32 [n_obl, 2] (start, count)
     obl_attr: np.ndarray = None    # i32 [total_terms]
     obl_w: np.ndarray = None       # f32 [total_terms]
+    # categorical-SET conditions (reference ContainsVector/Bitmap over
+    # CATEGORICAL_SET columns, vocab can exceed 256): node n with
+    # set_idx[n] = s >= 0 goes right iff the cell's token codes
+    # intersect set_items[set_offs[s]:set_offs[s+1]]
+    set_idx: np.ndarray = None     # i32 [total], -1 = not a set node
+    set_offs: np.ndarray = None    # i64 [n_set+1]
+    set_items: np.ndarray = None   # i32 [...]
     # na_value routing (reference NodeCondition.na_value): when the
     # node's input is MISSING (NaN numerical / -1 categorical code) the
     # example goes right iff na_right[node]. All-zeros = imputation-era
@@ -58,6 +65,16 @@ class FlatForest:
             self.obl_w = np.zeros(0, dtype=np.float32)
         if self.na_right is None:
             self.na_right = np.zeros(len(self.feat), dtype=np.uint8)
+        if self.set_idx is None:
+            self.set_idx = np.full(len(self.feat), -1, dtype=np.int32)
+        if self.set_offs is None:
+            self.set_offs = np.zeros(1, dtype=np.int64)
+        if self.set_items is None:
+            self.set_items = np.zeros(0, dtype=np.int32)
+
+    @property
+    def has_set_conditions(self) -> bool:
+        return bool((self.set_idx >= 0).any())
 
     @property
     def has_na_routing(self) -> bool:

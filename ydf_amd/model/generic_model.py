@@ -280,12 +280,97 @@ class GenericModel:
         dev = torch.device(device) if device is not None else default_device()
         from ydf_amd.utils import usage
 
+        if self.forest.has_set_conditions:
+            # categorical-set conditions: ragged token inputs — walks
+            # run on host (reference CategoricalSetContains engines)
+            m = self._margin_set_model(data)
+            usage.on_inference(m.shape[1])
+            return self._apply_activation(m).cpu().numpy()
         X_np = self._encode_features(data)
         usage.on_inference(X_np.shape[1] if X_np.ndim == 2 else len(X_np))
         X = torch.from_numpy(np.ascontiguousarray(X_np)).to(dev)
         m = self.predict_margin(X)
         out = self._apply_activation(m)
         return out.cpu().numpy()
+
+    def _margin_set_model(self, data) -> torch.Tensor:
+        """Host tree walk for models with categorical-SET conditions
+        (reference ContainsVector over CATEGORICAL_SET columns): cells
+        are token sets; a set node goes right iff the cell intersects
+        its item list. Vectorized per node over the rows reaching it."""
+        cols = _to_column_dict(data)
+        specs = self.dataspec.feature_columns
+        f = self.forest
+        n = len(next(iter(cols.values()))) if cols else 0
+        dense = {}
+        setcol = {}
+        for i, spec in enumerate(specs):
+            if spec.semantic == Semantic.CATEGORICAL_SET:
+                lut = {v: j for j, v in enumerate(spec.vocab or [])}
+
+                def to_set(cell, lut=lut):
+                    if isinstance(cell, (list, tuple, set, frozenset,
+                                         np.ndarray)):
+                        toks = [str(t) for t in cell]
+                    else:
+                        toks = str(cell).split(" ")
+                    return frozenset(lut.get(t, 0) for t in toks if t)
+
+                setcol[i] = [to_set(c) for c in cols[spec.name]]
+            else:
+                dense[i] = encode_column(
+                    cols[spec.name], spec,
+                    keep_na=self.forest.has_na_routing)
+        C = self._n_outputs()
+        out = np.zeros((C, n), dtype=np.float64)
+        offs, items = f.set_offs, f.set_items
+        scale = self._leaf_scale()
+
+        def rec(node, idx, acc):
+            fi = int(f.feat[node])
+            if fi < 0:
+                acc[idx] += float(f.thr[node])
+                return
+            si = int(f.set_idx[node])
+            ci = int(f.cat_idx[node])
+            if si >= 0:
+                cond = frozenset(
+                    int(v) for v in items[offs[si]:offs[si + 1]])
+                col = setcol[fi]
+                right = np.fromiter((not cond.isdisjoint(col[i])
+                                     for i in idx), bool, count=len(idx))
+            elif ci >= 0:
+                cb = dense[fi][idx].astype(np.int64).clip(0, 255)
+                right = (f.masks[ci][cb >> 6]
+                         >> (cb & 63).astype(np.uint64)) & np.uint64(1)
+                right = right.astype(bool)
+            else:
+                xv = dense[fi][idx]
+                right = xv > f.thr[node]
+                if f.has_na_routing:
+                    nanm = np.isnan(xv)
+                    if nanm.any():
+                        right = np.where(nanm, bool(f.na_right[node]),
+                                         right)
+            left = int(f.left[node])
+            rec(left, idx[~right], acc)
+            rec(left + 1, idx[right], acc)
+
+        T = f.n_trees
+        per = T // C if C > 1 else T
+        all_idx = np.arange(n)
+        for c in range(C):
+            acc = out[c]
+            acc += float(self.init_predictions[c]
+                         if c < len(self.init_predictions)
+                         else self.init_predictions[0])
+            for t in range(c, T, C) if C > 1 else range(T):
+                rec(int(f.roots[t]), all_idx, acc)
+            init = float(self.init_predictions[c]
+                         if c < len(self.init_predictions)
+                         else self.init_predictions[0])
+            out[c] = init + (acc - init) * scale
+        return torch.from_numpy(out.astype(np.float32))
 
     def _apply_activation(self, m: torch.Tensor) -> torch.Tensor:
         if self.activation == "sigmoid":
@@ -624,7 +709,10 @@ class GenericModel:
                  masks=self.forest.masks, cover=self.forest.cover,
                  obl_ranges=self.forest.obl_ranges,
                  obl_attr=self.forest.obl_attr, obl_w=self.forest.obl_w,
-                 na_right=self.forest.na_right)
+                 na_right=self.forest.na_right,
+                 set_idx=self.forest.set_idx,
+                 set_offs=self.forest.set_offs,
+                 set_items=self.forest.set_items)
         with open(os.path.join(path, "done"), "w") as f:
             f.write("")
 

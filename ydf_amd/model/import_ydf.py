@@ -188,7 +188,7 @@ def parse_data_spec(raw: bytes):
 # ---------------------------------------------------------------------------
 class _NodeRec:
     __slots__ = ("is_leaf", "attr", "thr", "mask", "value", "cover",
-                 "na_value", "obl")
+                 "na_value", "obl", "elements", "bitmap")
 
 
 def parse_node(raw: bytes, disc_bounds, n_classes: int,
@@ -203,6 +203,8 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
     r.cover = 0.0
     r.na_value = False
     r.obl = None
+    r.elements = None
+    r.bitmap = None
     if 5 in node:
         pass
     if 1 in node:  # classifier output
@@ -264,8 +266,9 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
             r.thr = 0.5
         elif 4 in inner:  # ContainsVector
             els = _msg(inner[4][0])
+            r.elements = _packed_varints(els[1][0]) if els.get(1) else []
             m = np.zeros(4, dtype=np.uint64)
-            for e in (_packed_varints(els[1][0]) if els.get(1) else []):
+            for e in r.elements:
                 if e < 256:
                     m[e >> 6] |= np.uint64(1 << (e & 63))
             r.mask = m
@@ -281,6 +284,7 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
             r.thr = float(np.nextafter(np.float32(t), np.float32("-inf")))
         elif 5 in inner:  # ContainsBitmap
             bm = _msg(inner[5][0]).get(1, [b""])[0]
+            r.bitmap = bm
             m = np.zeros(32, dtype=np.uint8)
             m[:min(len(bm), 32)] = np.frombuffer(bm[:32], dtype=np.uint8)
             r.mask = m.view(np.uint64)
@@ -294,7 +298,8 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
 # Model assembly
 # ---------------------------------------------------------------------------
 def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
-                value_scale: float = 1.0, wta: bool = False):
+                value_scale: float = 1.0, wta: bool = False,
+                set_feats=frozenset()):
     shards = sorted(p for p in os.listdir(model_dir)
                     if p.startswith(prefix + "nodes-"))
     records: List[bytes] = []
@@ -304,6 +309,7 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         [], [], [], [], [], [], []
     obl_ranges, obl_attr, obl_w = [], [], []
     na_right = []
+    set_idx_l, set_items, set_offs = [], [], [0]
     pos = 0
 
     def new_slot():
@@ -313,6 +319,7 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         cidx.append(-1)
         covers.append(0.0)
         na_right.append(0)
+        set_idx_l.append(-1)
         return len(feats) - 1
 
     def fill_node(idx, depth=0):
@@ -342,7 +349,20 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
             return
         feats[idx] = rec.attr
         na_right[idx] = 1 if rec.na_value else 0
-        if rec.mask is not None:
+        if rec.mask is not None and rec.attr in set_feats:
+            # categorical-SET condition: keep full element list (vocab
+            # may exceed the 256-bit mask)
+            if rec.elements is not None:
+                items = list(rec.elements)
+            else:
+                bits = np.unpackbits(
+                    np.frombuffer(rec.bitmap, dtype=np.uint8),
+                    bitorder="little")
+                items = list(np.nonzero(bits)[0])
+            set_idx_l[idx] = len(set_offs) - 1
+            set_items.extend(int(v) for v in items)
+            set_offs.append(len(set_items))
+        elif rec.mask is not None:
             cidx[idx] = len(masks)
             masks.append(rec.mask)
         elif rec.obl is not None:
@@ -375,7 +395,10 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
         obl_ranges=np.asarray(obl_ranges, np.int32).reshape(-1, 2),
         obl_attr=np.asarray(obl_attr, np.int32),
         obl_w=np.asarray(obl_w, np.float32),
-        na_right=np.asarray(na_right, np.uint8))
+        na_right=np.asarray(na_right, np.uint8),
+        set_idx=np.asarray(set_idx_l, np.int32),
+        set_offs=np.asarray(set_offs, np.int64),
+        set_items=np.asarray(set_items, np.int32))
 
 
 def load_ydf_model(path: str, file_prefix: str = ""):
@@ -405,6 +428,9 @@ def load_ydf_model(path: str, file_prefix: str = ""):
     # remap attribute indices (original column idx -> dense feature idx)
     remap = {ci: i for i, ci in enumerate(input_features)}
 
+    set_feats = frozenset(
+        i for i, c in enumerate(columns)
+        if c.semantic == Semantic.CATEGORICAL_SET)
     gbt_hdr_path = os.path.join(
         path, file_prefix + "gradient_boosted_trees_header.pb")
     rf_hdr_path = os.path.join(path, file_prefix + "random_forest_header.pb")
@@ -429,7 +455,8 @@ def load_ydf_model(path: str, file_prefix: str = ""):
         if not init_preds:
             init_preds = [0.0]
         loss = gh.get(3, [0])[0]
-        forest = _read_trees(path, file_prefix, disc_bounds, n_classes)
+        forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
+                             set_feats=set_feats)
         ntpi = gh.get(5, [1])[0]
         activation = "identity"
         if task == Task.CLASSIFICATION:
@@ -448,7 +475,8 @@ def load_ydf_model(path: str, file_prefix: str = ""):
             rh = _msg(f.read())
         wta = bool(rh.get(3, [1])[0])
         forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
-                             wta=wta and task == Task.CLASSIFICATION)
+                             wta=wta and task == Task.CLASSIFICATION,
+                             set_feats=set_feats)
         meta = {"imported_from": "yggdrasil-decision-forests",
                 "winner_take_all": wta}
         if task in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):

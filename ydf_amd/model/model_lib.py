@@ -34,6 +34,11 @@ def load_model(path: str) -> GenericModel:
                         obl_attr=z["obl_attr"] if "obl_attr" in z else None,
                         obl_w=z["obl_w"] if "obl_w" in z else None,
                         na_right=z["na_right"] if "na_right" in z
+                        else None,
+                        set_idx=z["set_idx"] if "set_idx" in z else None,
+                        set_offs=z["set_offs"] if "set_offs" in z
+                        else None,
+                        set_items=z["set_items"] if "set_items" in z
                         else None)
     cls = MODEL_CLASSES.get(header["model_type"], GenericModel)
     model = cls(
@@ -66,7 +71,10 @@ def serialize_model(model: GenericModel) -> bytes:
                  cover=model.forest.cover,
                  obl_ranges=model.forest.obl_ranges,
                  obl_attr=model.forest.obl_attr, obl_w=model.forest.obl_w,
-                 na_right=model.forest.na_right)
+                 na_right=model.forest.na_right,
+                 set_idx=model.forest.set_idx,
+                 set_offs=model.forest.set_offs,
+                 set_items=model.forest.set_items)
         zf.writestr("forest.npz", fbuf.getvalue())
     return buf.getvalue()
 
