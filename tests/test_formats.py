@@ -112,3 +112,28 @@ def test_synthetic_missing_values():
     m = ydf.GradientBoostedTreesLearner(
         label="LABEL", num_trees=10, validation_ratio=0).train(d)
     assert m.evaluate(d).accuracy > 0.6
+
+
+def test_read_reference_recordio_tfexamples():
+    """Our TFRecord reader parses the reference's own recordio tf.Example
+    shards (plain and gzip) and the imported adult GBT scores them
+    identically to the CSV."""
+    base = "/root/reference/yggdrasil_decision_forests/test_data"
+    if not os.path.exists(f"{base}/dataset/adult_test.recordio"):
+        pytest.skip("reference test_data not available")
+    pd = pytest.importorskip("pandas")
+    from ydf_amd.dataset.tfrecord import read_tfrecord_columns
+
+    cols = read_tfrecord_columns([f"{base}/dataset/adult_test.recordio"])
+    csv = pd.read_csv(f"{base}/dataset/adult_test.csv")
+    assert len(cols["age"]) == len(csv)
+    np.testing.assert_allclose(cols["age"], csv["age"].values)
+    assert list(cols["income"][:3]) == list(csv["income"][:3])
+    # gzip variant via the typed dataset path
+    gz = read_tfrecord_columns(
+        [f"{base}/dataset/adult_train.recordio.gz"])
+    assert len(gz["age"]) > 20000
+    m = ydf.load_ydf_model(f"{base}/model/adult_binary_class_gbdt")
+    p_rec = m.predict(cols, device="cpu")
+    p_csv = m.predict(csv, device="cpu")
+    np.testing.assert_allclose(p_rec, p_csv, atol=1e-6)
