@@ -167,20 +167,32 @@ def read_tfrecord_columns(paths: List[str]) -> Dict[str, np.ndarray]:
     """Reads tf.Example shards into a column dict (scalar features; the
     first value of each list is taken, missing -> NaN/empty)."""
     rows: List[Dict[str, object]] = []
-    names: Dict[str, str] = {}  # name -> kind
+    names: Dict[str, str] = {}  # name -> kind: s(tring) | i(nt) | f(loat)
     for path in paths:
         for rec in read_tfrecords(path):
             row = parse_example(rec)
             rows.append(row)
             for k, v in row.items():
-                if k not in names:
-                    names[k] = ("s" if len(v) and isinstance(v[0], str)
+                if k not in names and len(v):
+                    names[k] = ("s" if isinstance(v[0], str)
+                                else "i" if isinstance(v[0], int)
                                 else "f")
     cols: Dict[str, np.ndarray] = {}
     for name, kind in names.items():
         if kind == "s":
             cols[name] = np.array(
                 [(row.get(name) or [""])[0] for row in rows], dtype=object)
+        elif kind == "i":
+            # integers keep their dtype: categorical-integer vocabularies
+            # match on the "13" (not "13.0") string form
+            vals = [row.get(name) for row in rows]
+            if any(v is None or not v for v in vals):
+                cols[name] = np.array(
+                    [float(v[0]) if v else np.nan for v in vals],
+                    dtype=np.float32)
+            else:
+                cols[name] = np.array([int(v[0]) for v in vals],
+                                      dtype=np.int64)
         else:
             cols[name] = np.array(
                 [float((row.get(name) or [np.nan])[0]) for row in rows],
