@@ -264,3 +264,22 @@ def test_import_categorical_set_model_sst():
     ev = m.evaluate(te)
     assert ev.accuracy > 0.78
     assert ev.auc > 0.86
+
+
+def test_import_model_variants():
+    """Every importable reference adult GBT variant loads and keeps
+    published-range quality (32-category, integerized, numerical-only,
+    v2 format, tuned)."""
+    pd = pytest.importorskip("pandas")
+    te = pd.read_csv(f"{BASE}/dataset/adult_test.csv")
+    floors = {"adult_binary_class_gbdt_32cat": 0.86,
+              "adult_binary_class_gbdt_integerized": 0.82,
+              "adult_binary_class_gbdt_only_num": 0.82,
+              "adult_binary_class_gbdt_v2": 0.86,
+              "adult_binary_class_gbdt_tuned": 0.86}
+    for name, floor in floors.items():
+        m = ydf.load_ydf_model(f"{BASE}/model/{name}")
+        assert m.evaluate(te).accuracy > floor, name
+    for name in ("iris_multi_class_gbdt_v2", "abalone_regression_gbdt_v2",
+                 "8bits_numerical_binary_class_gbdt"):
+        assert ydf.load_ydf_model(f"{BASE}/model/{name}").num_trees() > 0
