@@ -675,7 +675,8 @@ class ForestTrainer:
             for s0 in range(0, n_active, self.max_slots):
                 ns = min(self.max_slots, n_active - s0)
                 hist_view = self.hist[:ns]
-                hist_view.zero_()
+                if not (use_i16 and feat_mask is not None):
+                    hist_view.zero_()
                 if use_i16:
                     # spg=1: 64 KiB LDS -> 2 workgroups/CU; the masked build
                     # is latency-bound, so occupancy beats slot batching
@@ -704,6 +705,9 @@ class ForestTrainer:
                         maskbits = (fm.view(ns, F16, 16)
                                     * weightsb).sum(-1).to(torch.int16)
                         maskbits = maskbits.contiguous()
+                        # masked zero: only live cells are read/written
+                        ops.zero_hist_masked(hist_view, maskbits,
+                                             self.F, ns)
                     ops.hist_build_gathered16(
                         self._bins16, self.gh, self.node_ids, build_map,
                         row_order, goffs, hist_view, self.N, self.F,
@@ -848,7 +852,6 @@ class ForestTrainer:
                 build_map = self.build_map_buf[:level_size]
                 derived = self.derived_buf[:level_size]
         hist_view = self.hist[:level_size]
-        hist_view.zero_()
         if use_i16d:
             if self._bins16 is None:
                 self._bins16 = ops.pack_bins16(self.bins)
@@ -882,6 +885,11 @@ class ForestTrainer:
                                         device=self.device))
                 maskbits = (fm.view(level_size, F16, 16)
                             * wb).sum(-1).to(torch.int16).contiguous()
+                # masked zero: only live (slot, sampled-feature) cells
+                ops.zero_hist_masked(hist_view, maskbits, self.F,
+                                     level_size)
+            else:
+                hist_view.zero_()
             spg_d = int(os.environ.get("YDFA_I16_SPG", "1"))
             gidx = torch.cat([
                 torch.arange(0, level_size, spg_d, dtype=torch.int64,
@@ -895,6 +903,7 @@ class ForestTrainer:
                 level_size, 0, spg_d, int(gidx.numel()) - 1, int(self.N),
                 maskbits=maskbits)
         else:
+            hist_view.zero_()
             ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
                            hist_view, level_base, level_size, 0,
                            level_size,

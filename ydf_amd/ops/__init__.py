@@ -134,6 +134,14 @@ def hist_build_gathered16(bins16: torch.Tensor, gh: torch.Tensor,
     return hist
 
 
+def zero_hist_masked(hist: torch.Tensor, maskbits: torch.Tensor,
+                     F: int, ns: int):
+    """Zeroes only the (slot, sampled-feature) histogram cells."""
+    _C.gpu_zero_hist_masked(hist.data_ptr(), maskbits.data_ptr(), F, ns,
+                            _stream())
+    return hist
+
+
 def row_scatter(keys: torch.Tensor, cursor: torch.Tensor,
                 row_order: torch.Tensor):
     """Counting-sort scatter: row_order[cursor[keys[r]]++] = r.

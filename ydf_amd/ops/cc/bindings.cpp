@@ -23,6 +23,7 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
                     const int32_t*, float*, int64_t, int, int, int, int, int,
                     int, int, uint8_t*, void*);
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
+void gpu_zero_hist_masked(float*, const uint16_t*, int, int, void*);
 void gpu_row_scatter(const int32_t*, int32_t*, int32_t*, int64_t, int,
                      void*);
 void gpu_hist_build_gathered16(const uint8_t*, const float*,
@@ -170,6 +171,13 @@ PYBIND11_MODULE(_ydf_ops, m) {
               P<int32_t>(slot_map), P<int32_t>(row_order), P<float>(hist),
               N, F, n_bins, level_base, level_size, slot0, n_slots, row_lo,
               row_hi, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_zero_hist_masked",
+        [](uintptr_t hist, uintptr_t maskbits, int F, int ns,
+           uintptr_t stream) {
+          gpu_zero_hist_masked(P<float>(hist), P<uint16_t>(maskbits), F,
+                               ns, (void*)stream);
         },
         nogil);
   m.def("gpu_row_scatter",

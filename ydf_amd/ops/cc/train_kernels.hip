@@ -531,6 +531,30 @@ __global__ void row_scatter_block_kernel(const int32_t* __restrict__ keys,
 }
 
 // ---------------------------------------------------------------------------
+// Masked histogram zeroing: with per-slot feature sampling only the
+// sampled (slot, feature) cells (plus feature 0, the totals source) are
+// ever written or read, so zeroing the whole [slots, F, 256, 3] buffer
+// (measured ~10% of the RF step) is wasted — clear just the live cells.
+// ---------------------------------------------------------------------------
+__global__ void zero_hist_masked_kernel(float* __restrict__ hist,
+                                        const uint16_t* __restrict__
+                                            maskbits,
+                                        int F, int n_bins, int ns) {
+  const int fg = blockIdx.x;
+  const int slot = blockIdx.z;
+  const int F16 = (F + 15) / 16;
+  unsigned m = maskbits[(int64_t)slot * F16 + fg];
+  while (m) {
+    const int k = __ffs(m) - 1;
+    m &= m - 1;
+    const int f = fg * 16 + k;
+    if (f >= F) continue;
+    float* p = hist + ((int64_t)slot * F + f) * (n_bins * 3);
+    for (int i = threadIdx.x; i < n_bins * 3; i += blockDim.x) p[i] = 0.f;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Feature-interleaved gathered histograms: bins16 stores the binned
 // matrix as [ceil(F/16)][N][16] u8, so ONE 16-byte load fetches a row's
 // bins for 16 features (vs 16 scattered single-byte gathers from the
@@ -1275,6 +1299,14 @@ void gpu_hist_build_gathered(const uint8_t* bins, const float* gh,
                      (const float2*)gh, node_ids, slot_map, row_order, hist,
                      N, F, n_bins, level_base, level_size, slot0, n_slots,
                      lds_map, row_lo, row_hi, rpb);
+}
+
+void gpu_zero_hist_masked(float* hist, const uint16_t* maskbits, int F,
+                          int ns, void* stream) {
+  const int F16 = (F + 15) / 16;
+  hipLaunchKernelGGL(zero_hist_masked_kernel, dim3(F16, 1, ns), dim3(192),
+                     0, (hipStream_t)stream, hist, maskbits, F, kMaxBins,
+                     ns);
 }
 
 void gpu_row_scatter(const int32_t* keys, int32_t* cursor,
