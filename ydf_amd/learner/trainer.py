@@ -252,8 +252,16 @@ class ForestTrainer:
                                 dtype=torch.float32, device=dev)
         # Histogram-subtraction trick (sibling = parent - smaller child):
         # previous level's histograms, indexed by its slot order.
+        i16_masked = (self.device.type == "cuda" and self.P == 0
+                      and self.F >= 32
+                      and os.environ.get("YDFA_HIST_I16", "1") == "1"
+                      and 0 < cfg.num_candidate_features < self.F)
+        # per-node feature sampling + interleaved builds: subtraction is
+        # never valid (masks differ across levels), so skip the
+        # hist_prev buffer (slots * F * 256 * 3 f32) and its per-level
+        # copies entirely
         self.use_hist_sub = os.environ.get("YDFA_NO_HIST_SUB", "0") != "1" \
-            and self.P == 0
+            and self.P == 0 and not i16_masked
         self.hist_prev = torch.empty_like(self.hist) if self.use_hist_sub \
             else None
         # dense mode: whole levels stay device-resident while the level fits
