@@ -315,6 +315,7 @@ class ForestTrainer:
             self.valid_node_ids = torch.empty(valid_bins.shape[1],
                                               dtype=torch.int32, device=dev)
         self._bins16 = None  # lazy interleaved copy for deep levels
+        self._bins32 = None
         self.rng = np.random.RandomState(cfg.seed)
 
     # -- helpers ----------------------------------------------------------
@@ -685,6 +686,10 @@ class ForestTrainer:
                 hist_view = self.hist[:ns]
                 if not (use_i16 and feat_mask is not None):
                     hist_view.zero_()
+                use_fg32 = use_i16 and os.environ.get(
+                    "YDFA_I16_FG", "16") == "32"
+                if use_fg32 and self._bins32 is None:
+                    self._bins32 = ops.pack_bins32(self.bins)
                 if use_i16:
                     # spg=1: 64 KiB LDS -> 2 workgroups/CU; the masked build
                     # is latency-bound, so occupancy beats slot batching
@@ -716,11 +721,36 @@ class ForestTrainer:
                         # masked zero: only live cells are read/written
                         ops.zero_hist_masked(hist_view, maskbits,
                                              self.F, ns)
-                    ops.hist_build_gathered16(
-                        self._bins16, self.gh, self.node_ids, build_map,
-                        row_order, goffs, hist_view, self.N, self.F,
-                        level_base, level_size, s0, spg, n_groups,
-                        max_rows, maskbits=maskbits)
+                    if use_fg32:
+                        mb32 = None
+                        if feat_mask is not None:
+                            F32 = (self.F + 31) // 32
+                            fm32 = torch.zeros((ns, F32 * 32),
+                                               dtype=torch.int64,
+                                               device=self.device)
+                            fm32[:, :self.F] = feat_mask[s0:s0 + ns].long()
+                            fm32[:, 0] = 1
+                            wb32 = (1 << torch.arange(
+                                32, dtype=torch.int64,
+                                device=self.device))
+                            mb32 = (fm32.view(ns, F32, 32)
+                                    * wb32).sum(-1).to(
+                                        torch.int32).contiguous()
+                        g1 = torch.arange(s0, s0 + ns + 1,
+                                          dtype=torch.int64,
+                                          device=self.device)
+                        ops.hist_build_gathered32(
+                            self._bins32, self.gh, self.node_ids,
+                            build_map, row_order,
+                            offs_dev[g1].contiguous(), hist_view, self.N,
+                            self.F, level_base, level_size, s0, ns,
+                            int(self.N), maskbits=mb32)
+                    else:
+                        ops.hist_build_gathered16(
+                            self._bins16, self.gh, self.node_ids,
+                            build_map, row_order, goffs, hist_view,
+                            self.N, self.F, level_base, level_size, s0,
+                            spg, n_groups, max_rows, maskbits=maskbits)
                 elif use_partition:
                     for g0 in range(s0, s0 + ns, lds_group):
                         g1 = min(g0 + lds_group, s0 + ns)

@@ -134,6 +134,34 @@ def hist_build_gathered16(bins16: torch.Tensor, gh: torch.Tensor,
     return hist
 
 
+def hist_build_gathered32(bins32: torch.Tensor, gh: torch.Tensor,
+                          node_ids: torch.Tensor, slot_map: torch.Tensor,
+                          row_order: torch.Tensor,
+                          group_offs: torch.Tensor, hist: torch.Tensor,
+                          N: int, F: int, level_base: int,
+                          level_size: int, win0: int, n_groups: int,
+                          max_group_rows: int, maskbits=None):
+    """32-feature interleaved partitioned histograms (one slot/block);
+    maskbits [ns, ceil(F/32)] u32."""
+    mb = maskbits.data_ptr() if maskbits is not None else 0
+    _C.gpu_hist_build_gathered32(
+        bins32.data_ptr(), gh.data_ptr(), node_ids.data_ptr(),
+        slot_map.data_ptr(), row_order.data_ptr(), group_offs.data_ptr(),
+        hist.data_ptr(), mb, N, F, level_base, level_size, win0,
+        n_groups, max_group_rows, _stream())
+    return hist
+
+
+def pack_bins32(bins: torch.Tensor) -> torch.Tensor:
+    """[F, N] u8 -> [ceil(F/32), N, 32] u8 interleaved copy."""
+    F, N = bins.shape
+    F32 = (F + 31) // 32
+    padded = torch.zeros((F32 * 32, N), dtype=torch.uint8,
+                         device=bins.device)
+    padded[:F] = bins
+    return padded.view(F32, 32, N).permute(0, 2, 1).contiguous()
+
+
 def zero_hist_masked(hist: torch.Tensor, maskbits: torch.Tensor,
                      F: int, ns: int):
     """Zeroes only the (slot, sampled-feature) histogram cells."""

@@ -23,6 +23,11 @@ void gpu_hist_build(const uint8_t*, const float*, const int32_t*,
                     const int32_t*, float*, int64_t, int, int, int, int, int,
                     int, int, uint8_t*, void*);
 void gpu_weighted_target(const float*, const float*, float*, int64_t, void*);
+void gpu_hist_build_gathered32(const uint8_t*, const float*,
+                               const int32_t*, const int32_t*,
+                               const int32_t*, const int64_t*, float*,
+                               const uint32_t*, int64_t, int, int, int,
+                               int, int, int64_t, void*);
 void gpu_zero_hist_masked(float*, const uint16_t*, int, int, void*);
 void gpu_row_scatter(const int32_t*, int32_t*, int32_t*, int64_t, int,
                      void*);
@@ -171,6 +176,20 @@ PYBIND11_MODULE(_ydf_ops, m) {
               P<int32_t>(slot_map), P<int32_t>(row_order), P<float>(hist),
               N, F, n_bins, level_base, level_size, slot0, n_slots, row_lo,
               row_hi, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_hist_build_gathered32",
+        [](uintptr_t bins32, uintptr_t gh, uintptr_t node_ids,
+           uintptr_t slot_map, uintptr_t row_order, uintptr_t group_offs,
+           uintptr_t hist, uintptr_t maskbits, int64_t N, int F,
+           int level_base, int level_size, int win0, int n_groups,
+           int64_t max_group_rows, uintptr_t stream) {
+          gpu_hist_build_gathered32(
+              P<uint8_t>(bins32), P<float>(gh), P<int32_t>(node_ids),
+              P<int32_t>(slot_map), P<int32_t>(row_order),
+              P<int64_t>(group_offs), P<float>(hist),
+              P<uint32_t>(maskbits), N, F, level_base, level_size, win0,
+              n_groups, max_group_rows, (void*)stream);
         },
         nogil);
   m.def("gpu_zero_hist_masked",

@@ -531,6 +531,85 @@ __global__ void row_scatter_block_kernel(const int32_t* __restrict__ keys,
 }
 
 // ---------------------------------------------------------------------------
+// 32-feature interleave variant: bins32 is [ceil(F/32)][N][32] u8, one
+// row costs two uint4 loads from the SAME 32-byte line (vs two separate
+// scattered lines with the 16-wide layout). One slot per block;
+// LDS = 32 * 256 * 16 B = 128 KiB.
+// ---------------------------------------------------------------------------
+__global__ void hist_build_gathered32_kernel(
+    const uint8_t* __restrict__ bins32, const float2* __restrict__ gh,
+    const int32_t* __restrict__ node_ids,
+    const int32_t* __restrict__ slot_map,
+    const int32_t* __restrict__ row_order,
+    const int64_t* __restrict__ group_offs, float* __restrict__ hist,
+    const uint32_t* __restrict__ maskbits,  // [ns][F32] or null
+    int64_t N, int F, int n_bins, int level_base, int level_size,
+    int win0, int n_chunks) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double* lg = reinterpret_cast<double*>(smem);
+  unsigned long long* lp =
+      reinterpret_cast<unsigned long long*>(smem) + 1;
+  const int fg = blockIdx.x;
+  const int slot_abs = blockIdx.z;   // one slot per block
+  const int tot = 32 * n_bins;
+  {
+    unsigned long long* zz = reinterpret_cast<unsigned long long*>(smem);
+    for (int i = threadIdx.x; i < tot * 2; i += blockDim.x) zz[i] = 0ull;
+  }
+  __syncthreads();
+  const int64_t r0 = group_offs[slot_abs];
+  const int64_t r1 = group_offs[slot_abs + 1];
+  const int64_t per = (r1 - r0 + n_chunks - 1) / n_chunks;
+  const int64_t j0 = r0 + (int64_t)blockIdx.y * per;
+  const int64_t j1 = min(j0 + per, r1);
+  const uint4* fb =
+      reinterpret_cast<const uint4*>(bins32 + (int64_t)fg * N * 32);
+  const int F32 = (F + 31) / 32;
+  unsigned m = 0xFFFFFFFFu;
+  if (maskbits != nullptr)
+    m = maskbits[(int64_t)slot_abs * F32 + fg];
+  if (m != 0u) {
+    for (int64_t j = j0 + threadIdx.x; j < j1; j += blockDim.x) {
+      const int row = row_order[j];
+      const int rel = node_ids[row] - level_base;
+      if (rel < 0 || rel >= level_size) continue;
+      if (slot_map[rel] - win0 != slot_abs) continue;
+      const float2 v = gh[row];
+      const unsigned long long hq =
+          (unsigned long long)(v.y * kHScale + 0.5f) |
+          ((unsigned long long)(v.y != 0.f) << 44);
+      const uint4 a = fb[(int64_t)row * 2];
+      const uint4 b = fb[(int64_t)row * 2 + 1];
+      const unsigned words[8] = {a.x, a.y, a.z, a.w, b.x, b.y, b.z, b.w};
+      unsigned mm = m;
+      while (mm) {
+        const int k = __ffs(mm) - 1;
+        mm &= mm - 1;
+        const int bin = (words[k >> 2] >> ((k & 3) * 8)) & 0xFF;
+        const int cell = 2 * (k * n_bins + bin);
+        atomicAdd(lg + cell, (double)v.x);
+        atomicAdd(lp + cell, hq);
+      }
+    }
+  }
+  __syncthreads();
+  for (int idx = threadIdx.x; idx < tot; idx += blockDim.x) {
+    const double g = lg[2 * idx];
+    const unsigned long long pk = lp[2 * idx];
+    if (pk == 0ull && g == 0.0) continue;
+    const int k = idx / n_bins;
+    const int f = fg * 32 + k;
+    if (f >= F) continue;
+    const int bin = idx - k * n_bins;
+    float* p = hist + ((int64_t)slot_abs * F + f) * (n_bins * 3)
+               + bin * 3;
+    atomicAdd(p, (float)g);
+    atomicAdd(p + 1, (float)((double)(pk & kHMask) * (double)kHInvScale));
+    atomicAdd(p + 2, (float)(pk >> 44));
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Masked histogram zeroing: with per-slot feature sampling only the
 // sampled (slot, feature) cells (plus feature 0, the totals source) are
 // ever written or read, so zeroing the whole [slots, F, 256, 3] buffer
@@ -1299,6 +1378,36 @@ void gpu_hist_build_gathered(const uint8_t* bins, const float* gh,
                      (const float2*)gh, node_ids, slot_map, row_order, hist,
                      N, F, n_bins, level_base, level_size, slot0, n_slots,
                      lds_map, row_lo, row_hi, rpb);
+}
+
+void gpu_hist_build_gathered32(const uint8_t* bins32, const float* gh,
+                               const int32_t* node_ids,
+                               const int32_t* slot_map,
+                               const int32_t* row_order,
+                               const int64_t* group_offs, float* hist,
+                               const uint32_t* maskbits, int64_t N, int F,
+                               int level_base, int level_size, int win0,
+                               int n_groups, int64_t max_group_rows,
+                               void* stream) {
+  const int n_bins = kMaxBins;
+  const int F32 = (F + 31) / 32;
+  int chunks = 1;
+  const int target_blocks = 4096;
+  if ((int64_t)F32 * n_groups < target_blocks) {
+    const int want = target_blocks / (F32 * (n_groups > 0 ? n_groups : 1));
+    const int64_t cap = (max_group_rows + 511) / 512;
+    chunks = (int)std::min<int64_t>(std::max(1, want),
+                                    std::max<int64_t>(1, cap));
+  }
+  const size_t lds = (size_t)32 * n_bins * 16;
+  int block = 1024;
+  if (const char* e = std::getenv("YDFA_I32_BLOCK")) block = atoi(e);
+  hipLaunchKernelGGL(hist_build_gathered32_kernel,
+                     dim3(F32, chunks, n_groups), dim3(block), lds,
+                     (hipStream_t)stream, bins32, (const float2*)gh,
+                     node_ids, slot_map, row_order, group_offs, hist,
+                     maskbits, N, F, n_bins, level_base, level_size, win0,
+                     chunks);
 }
 
 void gpu_zero_hist_masked(float* hist, const uint16_t* maskbits, int F,
