@@ -687,7 +687,7 @@ class ForestTrainer:
                 if not (use_i16 and feat_mask is not None):
                     hist_view.zero_()
                 use_fg32 = use_i16 and os.environ.get(
-                    "YDFA_I16_FG", "16") == "32"
+                    "YDFA_I16_FG", "32") == "32"
                 if use_fg32 and self._bins32 is None:
                     self._bins32 = ops.pack_bins32(self.bins)
                 if use_i16:
@@ -928,18 +928,37 @@ class ForestTrainer:
                                      level_size)
             else:
                 hist_view.zero_()
-            spg_d = int(os.environ.get("YDFA_I16_SPG", "1"))
-            gidx = torch.cat([
-                torch.arange(0, level_size, spg_d, dtype=torch.int64,
-                             device=self.device),
-                torch.tensor([level_size], dtype=torch.int64,
-                             device=self.device)])
-            goffs = offs_dev[gidx].contiguous()
-            ops.hist_build_gathered16(
-                self._bins16, self.gh, self.node_ids, build_map,
-                row_order, goffs, hist_view, self.N, self.F, level_base,
-                level_size, 0, spg_d, int(gidx.numel()) - 1, int(self.N),
-                maskbits=maskbits)
+            if os.environ.get("YDFA_I16_FG", "32") == "32":
+                if self._bins32 is None:
+                    self._bins32 = ops.pack_bins32(self.bins)
+                F32 = (self.F + 31) // 32
+                fm32 = torch.zeros((level_size, F32 * 32),
+                                   dtype=torch.int64, device=self.device)
+                fm32[:, :self.F] = feat_mask.long()
+                fm32[:, 0] = 1
+                wb32 = (1 << torch.arange(32, dtype=torch.int64,
+                                          device=self.device))
+                mb32 = (fm32.view(level_size, F32, 32)
+                        * wb32).sum(-1).to(torch.int32).contiguous()
+                ops.hist_build_gathered32(
+                    self._bins32, self.gh, self.node_ids, build_map,
+                    row_order, offs_dev[:level_size + 1].contiguous(),
+                    hist_view, self.N, self.F, level_base, level_size, 0,
+                    level_size, int(self.N), maskbits=mb32)
+            else:
+                spg_d = int(os.environ.get("YDFA_I16_SPG", "1"))
+                gidx = torch.cat([
+                    torch.arange(0, level_size, spg_d, dtype=torch.int64,
+                                 device=self.device),
+                    torch.tensor([level_size], dtype=torch.int64,
+                                 device=self.device)])
+                goffs = offs_dev[gidx].contiguous()
+                ops.hist_build_gathered16(
+                    self._bins16, self.gh, self.node_ids, build_map,
+                    row_order, goffs, hist_view, self.N, self.F,
+                    level_base, level_size, 0, spg_d,
+                    int(gidx.numel()) - 1, int(self.N),
+                    maskbits=maskbits)
         else:
             hist_view.zero_()
             ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
