@@ -1172,6 +1172,10 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             raise NotImplementedError(
                 "DART + checkpoint/resume is not supported (tree scales "
                 "change retroactively)")
+        if cfg.growing_strategy != "LOCAL":
+            raise NotImplementedError(
+                "DART replays dropped trees from the complete-tree "
+                "buffers; BEST_FIRST_GLOBAL trees live on implicit keys")
         dart_rec = []   # per tree: (feat_dev, bin_dev, masks_dev, leaf_dev)
         dart_scale = []  # per tree: current absolute leaf scale
         dart_rng = np.random.RandomState(cfg.seed ^ 0x5bd1e995)
