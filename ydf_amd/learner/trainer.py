@@ -876,9 +876,11 @@ class ForestTrainer:
             derived = self.derived_buf[:level_size]
         feat_mask = self._feat_mask(level_size, tree_idx, level)
         # dense-mode interleaved build pays only with feature sampling
-        # (measured ~4% regression for unmasked wide-F GBT)
+        # (measured ~4% regression for unmasked wide-F GBT);
+        # YDFA_I16_DENSE=1 forces it for unmasked models too
         use_i16d = self._i16_ok and level_size >= 4 \
-            and feat_mask is not None
+            and (feat_mask is not None
+                 or os.environ.get("YDFA_I16_DENSE", "0") == "1")
         # interleaved masked build disables subtraction (masks differ
         # across levels); unmasked i16 keeps it
         if use_i16d and feat_mask is not None and use_sub:
@@ -931,15 +933,18 @@ class ForestTrainer:
             if os.environ.get("YDFA_I16_FG", "32") == "32":
                 if self._bins32 is None:
                     self._bins32 = ops.pack_bins32(self.bins)
-                F32 = (self.F + 31) // 32
-                fm32 = torch.zeros((level_size, F32 * 32),
-                                   dtype=torch.int64, device=self.device)
-                fm32[:, :self.F] = feat_mask.long()
-                fm32[:, 0] = 1
-                wb32 = (1 << torch.arange(32, dtype=torch.int64,
-                                          device=self.device))
-                mb32 = (fm32.view(level_size, F32, 32)
-                        * wb32).sum(-1).to(torch.int32).contiguous()
+                mb32 = None
+                if feat_mask is not None:
+                    F32 = (self.F + 31) // 32
+                    fm32 = torch.zeros((level_size, F32 * 32),
+                                       dtype=torch.int64,
+                                       device=self.device)
+                    fm32[:, :self.F] = feat_mask.long()
+                    fm32[:, 0] = 1
+                    wb32 = (1 << torch.arange(32, dtype=torch.int64,
+                                              device=self.device))
+                    mb32 = (fm32.view(level_size, F32, 32)
+                            * wb32).sum(-1).to(torch.int32).contiguous()
                 ops.hist_build_gathered32(
                     self._bins32, self.gh, self.node_ids, build_map,
                     row_order, offs_dev[:level_size + 1].contiguous(),
