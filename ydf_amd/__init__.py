@@ -95,6 +95,10 @@ from ydf_amd.dataset.synthetic import (SyntheticDatasetOptions,
 from ydf_amd.deep import (DeepModel, MultiLayerPerceptronLearner,
                           TabularTransformerLearner)
 from ydf_amd.model.sklearn_io import from_sklearn
+from ydf_amd.model.tree import (Leaf, NonLeaf, Tree,
+                               build_forest_from_trees,
+                               build_model_from_trees,
+                               extract_tree, format_tree)
 from ydf_amd.metric.metric import Evaluation, evaluate_predictions
 
 # Utilities

@@ -103,3 +103,87 @@ def format_tree(tree: Tree, dataspec=None, max_depth: int = 6) -> str:
 
     fmt(tree.root, 0, "")
     return "\n".join(out)
+
+
+def build_forest_from_trees(trees, n_features: int):
+    """Inverse of extract_tree (capability analogue of the reference
+    TreeBuilder, model/decision_tree/builder.h): assembles a FlatForest
+    from Python Leaf/NonLeaf trees (numerical + categorical-mask +
+    oblique conditions)."""
+    import numpy as np
+
+    feats, thrs, lefts, covers, cidx = [], [], [], [], []
+    masks = []
+    obl_ranges, obl_attr, obl_w = [], [], []
+    roots = []
+
+    def new_slot():
+        feats.append(-1)
+        thrs.append(0.0)
+        lefts.append(0)
+        cidx.append(-1)
+        covers.append(0.0)
+        return len(feats) - 1
+
+    def fill(node, slot):
+        if isinstance(node, Leaf):
+            thrs[slot] = float(node.value)
+            return
+        if node.feature < 0 or node.feature >= n_features:
+            raise ValueError(f"feature index {node.feature} out of range")
+        feats[slot] = int(node.feature)
+        if node.oblique is not None:
+            attrs, ws = node.oblique
+            cidx[slot] = -(2 + len(obl_ranges))
+            obl_ranges.append((len(obl_attr), len(attrs)))
+            obl_attr.extend(int(a) for a in attrs)
+            obl_w.extend(float(w) for w in ws)
+            thrs[slot] = float(node.threshold)
+        elif node.mask is not None:
+            cidx[slot] = len(masks)
+            masks.append(np.asarray(node.mask, dtype=np.uint64))
+            thrs[slot] = 0.0
+        else:
+            thrs[slot] = float(node.threshold)
+        li = new_slot()
+        new_slot()
+        lefts[slot] = li
+        fill(node.neg_child, li)
+        fill(node.pos_child, li + 1)
+
+    for t in trees:
+        root = t.root if isinstance(t, Tree) else t
+        slot = new_slot()
+        roots.append(slot)
+        fill(root, slot)
+    import numpy as np
+
+    return FlatForest(
+        feat=np.asarray(feats, np.int32),
+        thr=np.asarray(thrs, np.float32),
+        left=np.asarray(lefts, np.int32),
+        roots=np.asarray(roots, np.int32),
+        cat_idx=np.asarray(cidx, np.int32),
+        masks=np.stack(masks).astype(np.uint64) if masks
+        else np.zeros((0, 4), np.uint64),
+        obl_ranges=np.asarray(obl_ranges, np.int32).reshape(-1, 2),
+        obl_attr=np.asarray(obl_attr, np.int32),
+        obl_w=np.asarray(obl_w, np.float32))
+
+
+def build_model_from_trees(trees, dataspec, task=None,
+                           init_predictions=None, activation="identity",
+                           model_type="GRADIENT_BOOSTED_TREES"):
+    """Assembles a servable model from hand-built trees (mirrors
+    constructing a model with the reference TreeBuilder + headers)."""
+    from ydf_amd.dataset.dataspec import Task
+    from ydf_amd.model.specialized import MODEL_CLASSES
+
+    forest = build_forest_from_trees(
+        trees, n_features=len(dataspec.feature_columns))
+    cls = MODEL_CLASSES[model_type]
+    return cls(forest=forest, dataspec=dataspec,
+               task=task or Task.REGRESSION,
+               init_predictions=init_predictions or [0.0],
+               num_trees_per_iter=1, activation=activation,
+               metadata={"hand_built": True})
