@@ -753,3 +753,47 @@ def test_crash_recovery_via_snapshot(tmp_path, binary_data, monkeypatch):
     m = ydf.GradientBoostedTreesLearner(**kw).train(binary_data)
     assert m.num_trees() == 30
     assert m.evaluate(binary_data).accuracy > 0.85
+
+
+def test_appendix_a_hyperparameters():
+    """Round-out of the reference's public hyperparameter surface
+    (SURVEY Appendix A): l1_regularization shrinks leaves, oblique
+    weight families + per-projection feature cap, lambda_loss,
+    validation_interval_in_trees, total_max_num_nodes, RF sampling
+    without replacement, pure_serving_model, and explicit
+    NotImplementedError for LOCAL imputation."""
+    d = ydf.generate_synthetic_dataset(num_examples=3000, seed=20)
+    m0 = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=15, validation_ratio=0).train(d)
+    m1 = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=15, l1_regularization=5.0,
+        validation_ratio=0).train(d)
+    assert np.abs(m1.forest.thr[m1.forest.feat < 0]).mean() < \
+        np.abs(m0.forest.thr[m0.forest.feat < 0]).mean()
+    for w in ("POWER_OF_TWO", "INTEGER"):
+        mo = ydf.GradientBoostedTreesLearner(
+            label="LABEL", num_trees=5, split_axis="SPARSE_OBLIQUE",
+            sparse_oblique_weights=w, sparse_oblique_max_num_features=2,
+            validation_ratio=0).train(d)
+        if len(mo.forest.obl_ranges):
+            assert mo.forest.obl_ranges[:, 1].max() <= 2
+    mc = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=500, total_max_num_nodes=300,
+        validation_ratio=0).train(d)
+    assert mc.num_trees() < 500
+    mi = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=20, validation_interval_in_trees=5,
+        validation_ratio=0.2).train(d)
+    assert len(mi.training_logs) == 4
+    mr = ydf.RandomForestLearner(
+        label="LABEL", num_trees=6, sampling_with_replacement=False,
+        bootstrap_size_ratio=0.7,
+        compute_oob_performances=False).train(d)
+    assert mr.num_trees() == 6
+    mp = ydf.GradientBoostedTreesLearner(
+        label="LABEL", num_trees=3, pure_serving_model=True,
+        validation_ratio=0).train(d)
+    assert mp.training_logs is None
+    with pytest.raises(NotImplementedError):
+        ydf.GradientBoostedTreesLearner(
+            label="LABEL", missing_value_policy="LOCAL_IMPUTATION")

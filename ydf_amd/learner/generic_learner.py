@@ -22,9 +22,24 @@ class GenericLearner:
                  weights: Optional[str] = None, tuner=None,
                  max_vocab_count: int = 2000, min_vocab_frequency: int = 1,
                  allow_na_conditions: bool = False,
+                 pure_serving_model: bool = False,
+                 missing_value_policy: str = "GLOBAL_IMPUTATION",
+                 categorical_algorithm: str = "CART",
+                 maximum_model_size_in_memory_in_bytes: float = -1.0,
                  random_seed: int = 123456, device=None,
                  num_threads: Optional[int] = None):
         self.allow_na_conditions = allow_na_conditions
+        self.pure_serving_model = pure_serving_model
+        if missing_value_policy != "GLOBAL_IMPUTATION":
+            raise NotImplementedError(
+                "missing_value_policy: only GLOBAL_IMPUTATION is "
+                "implemented (LOCAL/RANDOM_LOCAL: ROADMAP; learned "
+                "'is missing' splits via allow_na_conditions=True)")
+        if categorical_algorithm != "CART":
+            raise NotImplementedError(
+                "categorical_algorithm: only CART (sorted set-splits) "
+                "is implemented")
+        self.max_model_bytes = maximum_model_size_in_memory_in_bytes
         self.label = label
         self._task = task
         self.features = features
@@ -54,6 +69,30 @@ class GenericLearner:
             [c.semantic == Semantic.CATEGORICAL
              for c in ds.dataspec.feature_columns], dtype=bool)
 
+    def _finalize_model(self, model) -> None:
+        """pure_serving_model strips training-only payloads (reference
+        MakePureServing, abstract_model.h:433); the in-memory size cap
+        (maximum_model_size_in_memory_in_bytes) is checked post-train."""
+        if getattr(self, "pure_serving_model", False):
+            model.training_logs = None
+            model.tuner_logs = None
+            model._self_evaluation = None
+            if model.metadata:
+                model.metadata.pop("feature_gains", None)
+            model.forest.cover = np.zeros_like(model.forest.cover)
+        cap = getattr(self, "max_model_bytes", -1.0)
+        if cap and cap > 0:
+            f = model.forest
+            size = sum(a.nbytes for a in (f.feat, f.thr, f.left, f.roots,
+                                          f.cat_idx, f.masks, f.cover,
+                                          f.obl_attr, f.obl_w,
+                                          f.na_right))
+            if size > cap:
+                from ydf_amd.utils.log import info as _info
+
+                _info(f"model size {size}B exceeds "
+                      f"maximum_model_size_in_memory_in_bytes={cap:g}")
+
     def _oblique_cfg(self, F: int, cat_flags) -> Dict:
         """TrainerConfig kwargs for sparse-oblique splits (reference
         SparseObliqueSplit defaults, decision_tree.proto:173-296):
@@ -76,6 +115,8 @@ class GenericLearner:
                 "sparse_oblique_projection_density_factor", 2.0),
             oblique_weights=hp.get("sparse_oblique_weights", "BINARY"),
             oblique_norm=hp.get("sparse_oblique_normalization", "NONE"),
+            oblique_max_features=hp.get(
+                "sparse_oblique_max_num_features", -1),
         )
 
     def _bin_matrix(self, ds_X: np.ndarray, cat_feats: np.ndarray,

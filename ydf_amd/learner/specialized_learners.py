@@ -80,8 +80,12 @@ class GradientBoostedTreesLearner(GenericLearner):
                  goss_alpha: float = 0.2, goss_beta: float = 0.1,
                  selective_gradient_boosting_ratio: float = 0.01,
                  min_examples: int = 5, l2_regularization: float = 0.0,
+                 l1_regularization: float = 0.0,
                  min_sum_hessian_in_leaf: float = 1e-3,
                  validation_ratio: float = 0.1,
+                 validation_interval_in_trees: int = 1,
+                 lambda_loss: float = 1.0,
+                 total_max_num_nodes: int = -1,
                  early_stopping: str = "LOSS_INCREASE",
                  early_stopping_num_trees_look_ahead: int = 30,
                  early_stopping_initial_iteration: int = 10,
@@ -99,6 +103,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  sparse_oblique_projection_density_factor: float = 2.0,
                  sparse_oblique_normalization: str = "NONE",
                  sparse_oblique_weights: str = "BINARY",
+                 sparse_oblique_max_num_features: int = -1,
                  loss: str = "DEFAULT",
                  working_dir: Optional[str] = None,
                  resume_training: bool = False,
@@ -122,8 +127,13 @@ class GradientBoostedTreesLearner(GenericLearner):
                 selective_gradient_boosting_ratio),
             min_examples=min_examples,
             l2_regularization=l2_regularization,
+            l1_regularization=l1_regularization,
             min_sum_hessian_in_leaf=min_sum_hessian_in_leaf,
-            validation_ratio=validation_ratio, early_stopping=early_stopping,
+            validation_ratio=validation_ratio,
+            validation_interval_in_trees=validation_interval_in_trees,
+            lambda_loss=lambda_loss,
+            total_max_num_nodes=total_max_num_nodes,
+            early_stopping=early_stopping,
             early_stopping_num_trees_look_ahead=(
                 early_stopping_num_trees_look_ahead),
             early_stopping_initial_iteration=early_stopping_initial_iteration,
@@ -144,6 +154,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                 sparse_oblique_projection_density_factor),
             sparse_oblique_normalization=sparse_oblique_normalization,
             sparse_oblique_weights=sparse_oblique_weights,
+            sparse_oblique_max_num_features=(
+                sparse_oblique_max_num_features),
             loss=loss,
             working_dir=working_dir, resume_training=resume_training,
             resume_training_snapshot_interval_seconds=(
@@ -272,7 +284,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                 group_ids = group_ids[~vmask]
             ranking = RankingLambdas(group_ids, labels.cpu().numpy(),
                                      device,
-                                     truncation=self.ndcg_truncation)
+                                     truncation=self.ndcg_truncation,
+                                     sigma=hp.get("lambda_loss", 1.0))
         elif valid is not None:
             if self._task == Task.SURVIVAL_ANALYSIS:
                 raise NotImplementedError(
@@ -327,6 +340,7 @@ class GradientBoostedTreesLearner(GenericLearner):
         cfg = trainer_lib.TrainerConfig(
             loss=loss, num_trees=hp["num_trees"], max_depth=hp["max_depth"],
             shrinkage=hp["shrinkage"], lambda_l2=hp["l2_regularization"],
+            lambda_l1=hp.get("l1_regularization", 0.0),
             min_examples=hp["min_examples"],
             min_hessian=hp["min_sum_hessian_in_leaf"],
             subsample=hp["subsample"],
@@ -343,6 +357,9 @@ class GradientBoostedTreesLearner(GenericLearner):
             early_stopping_initial_iteration=(
                 hp["early_stopping_initial_iteration"]),
             cat_smooth=hp["l2_categorical_regularization"],
+            lambda_loss=hp.get("lambda_loss", 1.0),
+            validation_interval=hp.get("validation_interval_in_trees", 1),
+            total_max_num_nodes=hp.get("total_max_num_nodes", -1),
             growing_strategy=hp.get("growing_strategy", "LOCAL"),
             max_num_nodes=hp.get("max_num_nodes", 31),
             focal_gamma=hp.get("focal_loss_gamma", 2.0),
@@ -479,6 +496,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                 gains[k] = gains.get(k, 0.0) + v
         model = make_model(flat, init_preds, gains)
         model.training_logs = logs
+        self._finalize_model(model)
         if snapshot_cb is not None:
             # final state also becomes the snapshot (enables continuing
             # with a larger num_trees later)
@@ -518,9 +536,11 @@ class RandomForestLearner(GenericLearner):
                  bootstrap_size_ratio: float = 1.0,
                  num_candidate_attributes: int = 0,
                  num_candidate_attributes_ratio: float = -1.0,
+                 sampling_with_replacement: bool = True,
                  winner_take_all: bool = True,
                  compute_oob_performances: bool = True,
                  compute_oob_variable_importances: bool = False,
+                 num_oob_variable_importances_permutations: int = 1,
                  honest: bool = False,
                  honest_ratio_leaf_examples: float = 0.5,
                  honest_fixed_separation: bool = False,
@@ -546,7 +566,10 @@ class RandomForestLearner(GenericLearner):
             bootstrap_size_ratio=bootstrap_size_ratio,
             num_candidate_attributes=num_candidate_attributes,
             num_candidate_attributes_ratio=num_candidate_attributes_ratio,
+            sampling_with_replacement=sampling_with_replacement,
             winner_take_all=winner_take_all,
+            num_oob_variable_importances_permutations=(
+                num_oob_variable_importances_permutations),
             compute_oob_performances=compute_oob_performances,
             compute_oob_variable_importances=(
                 compute_oob_variable_importances),
@@ -688,6 +711,9 @@ class RandomForestLearner(GenericLearner):
             n_classes=n_classes, seed=self.random_seed,
             bootstrap=hp["bootstrap_training_dataset"],
             bootstrap_ratio=hp.get("bootstrap_size_ratio", 1.0),
+            with_replacement=hp.get("sampling_with_replacement", True),
+            oob_vi_permutations=hp.get(
+                "num_oob_variable_importances_permutations", 1),
             num_candidate_features=ncand,
             max_duration_seconds=hp.get(
                 "maximum_training_duration_seconds", -1.0),
@@ -741,6 +767,7 @@ class RandomForestLearner(GenericLearner):
                 trees, [c.name for c in ds.dataspec.feature_columns]),
                 "winner_take_all": wta})
         model._self_evaluation = oob_eval
+        self._finalize_model(model)
         if hp.get("compute_oob_variable_importances") \
                 and hp["bootstrap_training_dataset"]:
             vi = _oob_permutation_vi(model, ds, cfg, self._task, device)
@@ -788,9 +815,11 @@ def _oob_permutation_vi(model, ds, cfg, task, device):
             return -float(np.sqrt(np.mean((p - yb) ** 2)))  # higher=better
         return float(((p > 0.5) == (yb > 0.5)).mean())
 
+    n_perm = max(1, int(getattr(cfg, "oob_vi_permutations", 1)))
     for t in range(T):
         w = trainer_lib.rf_bootstrap_weights(cfg.seed, t, N, device,
-                                             cfg.bootstrap_ratio)
+                                             cfg.bootstrap_ratio,
+                                             cfg.with_replacement)
         oob = (w == 0).cpu().numpy()
         if oob.sum() < 10:
             continue
@@ -801,8 +830,11 @@ def _oob_permutation_vi(model, ds, cfg, task, device):
         base = tree_metric(Xt, yb, t)
         for f in range(F):
             saved = Xs[f].copy()
-            Xs[f] = saved[rng.permutation(len(saved))]
-            acc_drop[f] += base - tree_metric(Xt, yb, t)
+            drop = 0.0
+            for _ in range(n_perm):
+                Xs[f] = saved[rng.permutation(len(saved))]
+                drop += base - tree_metric(Xt, yb, t)
+            acc_drop[f] += drop / n_perm
             Xs[f] = saved
     if n_used == 0:
         return None

@@ -46,6 +46,7 @@ class TrainerConfig:
     max_depth: int = 6
     shrinkage: float = 0.1
     lambda_l2: float = 0.0
+    lambda_l1: float = 0.0
     min_examples: int = 5
     min_hessian: float = 1e-3
     min_gain: float = 0.0
@@ -66,13 +67,21 @@ class TrainerConfig:
     # RF-specific
     bootstrap: bool = False
     bootstrap_ratio: float = 1.0     # Poisson rate (bootstrap_size_ratio)
+    with_replacement: bool = True    # False: Bernoulli row subsampling
     num_candidate_features: int = 0  # 0 = all features
     max_duration_seconds: float = -1.0
     # honest trees (reference decision_tree.proto Honest message): tree
     # structure from one random half, leaf values re-estimated on the other
+    oob_vi_permutations: int = 1
     honest: bool = False
     honest_ratio: float = 0.5        # fraction reserved for leaf values
     honest_fixed_separation: bool = False
+    # ranking sigma (reference lambda_loss)
+    lambda_loss: float = 1.0
+    # evaluate validation every k trees (validation_interval_in_trees)
+    validation_interval: int = 1
+    # stop boosting once the forest holds this many nodes (<=0: off)
+    total_max_num_nodes: int = -1
     # early stopping (GBT; reference gradient_boosted_trees.proto:151-172)
     early_stopping: bool = False
     early_stopping_num_trees_look_ahead: int = 30
@@ -94,7 +103,8 @@ class TrainerConfig:
     # oblique node. 0 = axis-aligned only.
     oblique_projections: int = 0
     oblique_density: float = 2.0      # expected nonzeros per projection
-    oblique_weights: str = "BINARY"   # BINARY | CONTINUOUS
+    oblique_weights: str = "BINARY"   # BINARY|CONTINUOUS|POWER_OF_TWO|INTEGER
+    oblique_max_features: int = -1    # cap nonzeros per projection
     oblique_norm: str = "NONE"        # NONE | STANDARD_DEVIATION | MIN_MAX
 
 
@@ -369,9 +379,30 @@ class ForestTrainer:
             sel[np.nonzero(empty)[0], rs.randint(0, nf, empty.sum())] = True
         if cfg.oblique_weights == "CONTINUOUS":
             w = rs.uniform(-1.0, 1.0, size=(P, nf)).astype(np.float32)
+        elif cfg.oblique_weights == "POWER_OF_TWO":
+            # s * 2^i, i ~ U{-3..3}, s ~ U{-1,1} (decision_tree.proto:260)
+            i = rs.randint(-3, 4, size=(P, nf))
+            sgn = rs.randint(0, 2, size=(P, nf)) * 2 - 1
+            w = (sgn * np.exp2(i)).astype(np.float32)
+        elif cfg.oblique_weights == "INTEGER":
+            # uniform integers in [-5, 5] (decision_tree.proto:266)
+            w = rs.randint(-5, 6, size=(P, nf)).astype(np.float32)
         else:  # BINARY (reference default)
             w = (rs.randint(0, 2, size=(P, nf)) * 2 - 1).astype(np.float32)
         w *= sel
+        if cfg.oblique_max_features > 0:
+            # cap nonzeros per projection (sparse_oblique_max_num_features)
+            for p in range(P):
+                nz = np.nonzero(w[p])[0]
+                if len(nz) > cfg.oblique_max_features:
+                    drop = rs.choice(nz,
+                                     len(nz) - cfg.oblique_max_features,
+                                     replace=False)
+                    w[p, drop] = 0.0
+            empty2 = ~(w != 0).any(axis=1)
+            if empty2.any():
+                w[np.nonzero(empty2)[0],
+                  rs.randint(0, nf, empty2.sum())] = 1.0
         w *= self.proj_scale[self.num_feat_idx][None, :]
         W = np.zeros((P, bF), dtype=np.float32)
         W[:, self.num_feat_idx] = w
@@ -457,7 +488,7 @@ class ForestTrainer:
                            self.bb_nf, self.best_feat, self.best_bin,
                            self.best_gain, 0, 1, cfg.lambda_l2,
                            cfg.min_hessian, cfg.min_examples, cfg.min_gain,
-                           feat_mask=fm)
+                           feat_mask=fm, lambda_l1=cfg.lambda_l1)
             f = int(self.best_feat[0].item())
             b = int(self.best_bin[0].item())
             g = float(self.best_gain[0].item())
@@ -785,7 +816,8 @@ class ForestTrainer:
                                feat_mask=feat_mask, cat_flags=self.cat_flags,
                                masks=self.tree_masks,
                                cat_smooth=cfg.cat_smooth, mono=self.mono,
-                               node_bounds=self.node_bounds)
+                               node_bounds=self.node_bounds,
+                               lambda_l1=cfg.lambda_l1)
 
             prev_fit = n_active <= self.max_slots
             if self.use_hist_sub and prev_fit and level + 1 < cfg.max_depth:
@@ -822,7 +854,8 @@ class ForestTrainer:
                             for a, c in zip(children, ccounts)}
 
         ops.leaf_values(self.node_stats, self.leaf_vals, cfg.lambda_l2,
-                        node_bounds=self.node_bounds)
+                        node_bounds=self.node_bounds,
+                        lambda_l1=cfg.lambda_l1)
 
     def extract_host_tree(self) -> HostTree:
         cfg = self.cfg
@@ -980,7 +1013,8 @@ class ForestTrainer:
                        cfg.min_hessian, cfg.min_examples, cfg.min_gain,
                        feat_mask=feat_mask, cat_flags=self.cat_flags,
                        masks=self.tree_masks, cat_smooth=cfg.cat_smooth,
-                       mono=self.mono, node_bounds=self.node_bounds)
+                       mono=self.mono, node_bounds=self.node_bounds,
+                       lambda_l1=cfg.lambda_l1)
         fits = True
         if self.use_hist_sub and level + 1 < cfg.max_depth:
             self.hist_prev[:level_size].copy_(hist_view)
@@ -1339,7 +1373,15 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 snapshot_interval_seconds:
             snapshot_cb(trees, it + 1, init_preds)
             t_last_snapshot = _time.monotonic()
-        if has_valid:
+        if cfg.total_max_num_nodes > 0:
+            n_nodes = sum(int((t_.feat >= 0).sum()) * 2 + 1
+                          for t_ in trees)
+            if n_nodes >= cfg.total_max_num_nodes:
+                if log:
+                    log(f"total_max_num_nodes reached at iteration "
+                        f"{it + 1}")
+                break
+        if has_valid and (it + 1) % max(cfg.validation_interval, 1) == 0:
             if cfg.loss == LOSS_COX:
                 vloss = valid_cox.loss(valid_preds[0]) \
                     if valid_cox is not None else float("nan")
@@ -1448,18 +1490,27 @@ def _eval_loss(trainer, preds, labels, cfg, loss_buf) -> float:
 
 
 def rf_bootstrap_weights(seed: int, tree_idx: int, N: int, dev,
-                         rate: float = 1.0) -> torch.Tensor:
-    """Poisson(rate) bootstrap draw for tree `tree_idx`, clipped at 15
-    (rate = bootstrap_size_ratio). Seeded per tree so the draw can be
-    REGENERATED after training (OOB permutation importances re-derive
-    each tree's out-of-bag rows)."""
+                         rate: float = 1.0,
+                         with_replacement: bool = True) -> torch.Tensor:
+    """Bootstrap draw for tree `tree_idx`: Poisson(rate) clipped at 15
+    (sampling with replacement; rate = bootstrap_size_ratio) or
+    Bernoulli(rate) 0/1 weights (without replacement,
+    reference sampling_with_replacement=false). Seeded per tree so the
+    draw can be REGENERATED after training (OOB permutation importances
+    re-derive each tree's out-of-bag rows)."""
     if dev.type == "cuda":
         g = torch.Generator(device=dev)
         g.manual_seed((seed * 31337 + tree_idx) % (1 << 31))
+        if not with_replacement:
+            return (torch.rand(N, device=dev, generator=g)
+                    < rate).float()
         return torch.poisson(torch.full((N,), float(rate), device=dev),
                              generator=g).clamp_(max=15)
     rs = np.random.RandomState((seed * 31337 + tree_idx) % (1 << 31))
-    w = np.minimum(rs.poisson(rate, size=N), 15).astype(np.float32)
+    if not with_replacement:
+        w = (rs.random_sample(N) < rate).astype(np.float32)
+    else:
+        w = np.minimum(rs.poisson(rate, size=N), 15).astype(np.float32)
     return torch.from_numpy(w).to(dev)
 
 
@@ -1515,7 +1566,8 @@ def train_rf(trainer: ForestTrainer, log=None,
             # Poisson bootstrap, clipped at 15 (P < 1e-12) — the packed
             # u64 histogram path requires per-example h <= 16
             weights = rf_bootstrap_weights(cfg.seed, it, N, dev,
-                                           cfg.bootstrap_ratio)
+                                           cfg.bootstrap_ratio,
+                                           cfg.with_replacement)
         if trainer.weights is not None:
             # user example weights compose with the bootstrap draw counts
             weights = trainer.weights if weights is None \

@@ -198,7 +198,7 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
                n_slots: int, lambda_l2: float, min_hessian: float,
                min_examples: int, min_gain: float, feat_mask=None,
                cat_flags=None, masks=None, cat_smooth: float = 1.0,
-               mono=None, node_bounds=None):
+               mono=None, node_bounds=None, lambda_l1: float = 0.0):
     F = hist.shape[1]
     n_bins = hist.shape[2]
     mp = feat_mask.data_ptr() if feat_mask is not None else 0
@@ -210,7 +210,7 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
             best_gain_nf.data_ptr(), best_bin_nf.data_ptr(),
             best_feat.data_ptr(), best_bin.data_ptr(), best_gain.data_ptr(),
             mp, cf, mk, mn, nb, F, n_bins, slot0, n_slots, lambda_l2,
-            min_hessian, min_examples, min_gain, cat_smooth)
+            min_hessian, min_examples, min_gain, cat_smooth, lambda_l1)
     if hist.is_cuda:
         _C.gpu_split_scan(*args, _stream())
     else:
@@ -276,15 +276,16 @@ def update_node_ids(bins: torch.Tensor, node_ids: torch.Tensor,
 
 
 def leaf_values(node_stats: torch.Tensor, out: torch.Tensor,
-                lambda_l2: float, node_bounds=None):
+                lambda_l2: float, node_bounds=None,
+                lambda_l1: float = 0.0):
     total = out.numel()
     nb = node_bounds.data_ptr() if node_bounds is not None else 0
     if node_stats.is_cuda:
         _C.gpu_leaf_values(node_stats.data_ptr(), nb, out.data_ptr(), total,
-                           lambda_l2, _stream())
+                           lambda_l2, lambda_l1, _stream())
     else:
         _C.cpu_leaf_values(node_stats.data_ptr(), nb, out.data_ptr(), total,
-                           lambda_l2)
+                           lambda_l2, lambda_l1)
     return out
 
 

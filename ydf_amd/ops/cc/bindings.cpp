@@ -52,7 +52,8 @@ void gpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, const uint8_t*,
                          const unsigned long long*, int64_t, int, int,
                          void*);
-void gpu_leaf_values(const float*, const float*, float*, int, float, void*);
+void gpu_leaf_values(const float*, const float*, float*, int, float,
+                     float, void*);
 void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
                       void*);
 void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
@@ -88,7 +89,8 @@ void cpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int);
 void cpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, const uint8_t*,
                          const unsigned long long*, int64_t, int, int);
-void cpu_leaf_values(const float*, const float*, float*, int, float);
+void cpu_leaf_values(const float*, const float*, float*, int, float,
+                     float);
 void cpu_update_preds(float*, const int32_t*, const float*, int64_t, float);
 void cpu_binary_logloss(const float*, const float*, float*, int64_t);
 void cpu_predict_forest(const float*, int64_t, int, const int32_t*,
@@ -114,9 +116,10 @@ T* P(uintptr_t p) {
   return reinterpret_cast<T*>(p);
 }
 SplitParams MakeSP(float lambda_l2, float min_hessian, int min_examples,
-                   float min_gain, float cat_smooth) {
+                   float min_gain, float cat_smooth, float lambda_l1) {
   SplitParams sp;
   sp.lambda_l2 = lambda_l2;
+  sp.lambda_l1 = lambda_l1;
   sp.min_hessian = min_hessian;
   sp.min_examples = min_examples;
   sp.min_gain = min_gain;
@@ -236,7 +239,8 @@ PYBIND11_MODULE(_ydf_ops, m) {
            uintptr_t cat_flags, uintptr_t masks, uintptr_t mono,
            uintptr_t node_bounds, int F, int n_bins, int slot0,
            int n_slots, float lambda_l2, float min_hessian, int min_examples,
-           float min_gain, float cat_smooth, uintptr_t stream) {
+           float min_gain, float cat_smooth, float lambda_l1,
+           uintptr_t stream) {
           gpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
                          P<float>(node_stats), P<float>(best_gain_nf),
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
@@ -245,7 +249,7 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          P<unsigned long long>(masks), P<int8_t>(mono),
                          P<float>(node_bounds), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
-                                min_gain, cat_smooth),
+                                min_gain, cat_smooth, lambda_l1),
                          (void*)stream);
         },
         nogil);
@@ -281,9 +285,11 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("gpu_leaf_values",
         [](uintptr_t node_stats, uintptr_t node_bounds, uintptr_t leaf_values,
-           int total_nodes, float lambda_l2, uintptr_t stream) {
+           int total_nodes, float lambda_l2, float lambda_l1,
+           uintptr_t stream) {
           gpu_leaf_values(P<float>(node_stats), P<float>(node_bounds),
                           P<float>(leaf_values), total_nodes, lambda_l2,
+                          lambda_l1,
                           (void*)stream);
         },
         nogil);
@@ -390,7 +396,7 @@ PYBIND11_MODULE(_ydf_ops, m) {
            uintptr_t cat_flags, uintptr_t masks, uintptr_t mono,
            uintptr_t node_bounds, int F, int n_bins, int slot0,
            int n_slots, float lambda_l2, float min_hessian, int min_examples,
-           float min_gain, float cat_smooth) {
+           float min_gain, float cat_smooth, float lambda_l1) {
           cpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
                          P<float>(node_stats), P<float>(best_gain_nf),
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
@@ -399,7 +405,7 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          P<unsigned long long>(masks), P<int8_t>(mono),
                          P<float>(node_bounds), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
-                                min_gain, cat_smooth));
+                                min_gain, cat_smooth, lambda_l1));
         },
         nogil);
   m.def("cpu_plan_level",
@@ -431,9 +437,10 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("cpu_leaf_values",
         [](uintptr_t node_stats, uintptr_t node_bounds, uintptr_t leaf_values,
-           int total_nodes, float lambda_l2) {
+           int total_nodes, float lambda_l2, float lambda_l1) {
           cpu_leaf_values(P<float>(node_stats), P<float>(node_bounds),
-                          P<float>(leaf_values), total_nodes, lambda_l2);
+                          P<float>(leaf_values), total_nodes, lambda_l2,
+                          lambda_l1);
         },
         nogil);
   m.def("cpu_update_preds",

@@ -34,11 +34,19 @@ enum LossKind : int {
 // Split-scan hyper-parameters (subset of the reference decision-tree proto).
 struct SplitParams {
   float lambda_l2;        // l2_regularization
+  float lambda_l1;        // l1_regularization (soft-thresholds gradients,
+                          // XGBoost eq. 2 formulation)
   float min_hessian;      // min_sum_hessian_in_leaf
   int min_examples;       // min_examples (reference default 5)
   float min_gain;         // splits with gain <= min_gain become leaves
   float cat_smooth;       // l2_categorical_regularization (category order
                           // statistic smoothing; reference default 1.0)
 };
+
+// l1 soft threshold: T(G) = sign(G) * max(|G| - l1, 0)
+__host__ __device__ inline float l1_thresh(float g, float l1) {
+  const float a = fabsf(g) - l1;
+  return a > 0.f ? (g > 0.f ? a : -a) : 0.f;
+}
 
 }  // namespace ydfa

@@ -268,7 +268,8 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
       for (int b = 0; b < n_bins; ++b) {
         Gf += hp[b * 3]; Hf += hp[b * 3 + 1]; Cf += hp[b * 3 + 2];
       }
-      const float pterm = Gf * Gf / (Hf + sp.lambda_l2);
+      const float tpf = ydfa::l1_thresh(Gf, sp.lambda_l1);
+      const float pterm = tpf * tpf / (Hf + sp.lambda_l2);
       float GL = 0.f, HL = 0.f, CL = 0.f;
       float fbest = -1e30f;
       int fbin = 0;
@@ -280,12 +281,16 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
             HL >= sp.min_hessian && HR >= sp.min_hessian) {
           bool okm = true;
           if (mono != nullptr && mono[f] != 0) {
-            const float wl = -GL / (HL + sp.lambda_l2);
-            const float wr = -GR / (HR + sp.lambda_l2);
+            const float wl = -ydfa::l1_thresh(GL, sp.lambda_l1)
+                             / (HL + sp.lambda_l2);
+            const float wr = -ydfa::l1_thresh(GR, sp.lambda_l1)
+                             / (HR + sp.lambda_l2);
             okm = (mono[f] > 0) ? (wl <= wr) : (wl >= wr);
           }
-          const float gain = GL * GL / (HL + sp.lambda_l2) +
-                             GR * GR / (HR + sp.lambda_l2) - pterm;
+          const float tl = ydfa::l1_thresh(GL, sp.lambda_l1);
+          const float tr = ydfa::l1_thresh(GR, sp.lambda_l1);
+          const float gain = tl * tl / (HL + sp.lambda_l2) +
+                             tr * tr / (HR + sp.lambda_l2) - pterm;
           if (okm && gain > fbest) { fbest = gain; fbin = b; }
         }
       }
@@ -338,9 +343,11 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
       float llo = lo, lhi = hi, rlo = lo, rhi = hi;
       if (!is_cat && mono != nullptr && mono[node_best_f] != 0) {
         const float wl =
-            std::min(std::max(-GL / (HL + sp.lambda_l2), lo), hi);
+            std::min(std::max(-ydfa::l1_thresh(GL, sp.lambda_l1)
+                              / (HL + sp.lambda_l2), lo), hi);
         const float wr = std::min(
-            std::max(-(G - GL) / (H - HL + sp.lambda_l2), lo), hi);
+            std::max(-ydfa::l1_thresh(G - GL, sp.lambda_l1)
+                     / (H - HL + sp.lambda_l2), lo), hi);
         const float mid = 0.5f * (wl + wr);
         if (mono[node_best_f] > 0) { lhi = mid; rlo = mid; }
         else { llo = mid; rhi = mid; }
@@ -416,10 +423,14 @@ void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
 }
 
 void cpu_leaf_values(const float* node_stats, const float* node_bounds,
-                     float* leaf_values, int total_nodes, float lambda_l2) {
+                     float* leaf_values, int total_nodes, float lambda_l2,
+                     float lambda_l1) {
   for (int i = 0; i < total_nodes; ++i) {
     const float* ns = node_stats + (int64_t)i * 3;
-    float v = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+    float v = (ns[1] != 0.f)
+                  ? (-ydfa::l1_thresh(ns[0], lambda_l1)
+                     / (ns[1] + lambda_l2))
+                  : 0.f;
     if (node_bounds != nullptr)
       v = std::min(std::max(v, node_bounds[2 * i]), node_bounds[2 * i + 1]);
     leaf_values[i] = v;

@@ -834,13 +834,17 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
         // monotonic constraint (reference monotonic_constraints;
         // XGBoost-style): reject splits whose child values violate the
         // declared direction
-        const float wl = -GL / (HL + sp.lambda_l2);
-        const float wr = -GR / (HR + sp.lambda_l2);
+        const float wl = -l1_thresh(GL, sp.lambda_l1) / (HL + sp.lambda_l2);
+        const float wr = -l1_thresh(GR, sp.lambda_l1) / (HR + sp.lambda_l2);
         ok = (mono[f] > 0) ? (wl <= wr) : (wl >= wr);
       }
-      if (ok)
-        gain = GL * GL / (HL + sp.lambda_l2) + GR * GR / (HR + sp.lambda_l2) -
-               G * G / (H + sp.lambda_l2);
+      if (ok) {
+        const float tl = l1_thresh(GL, sp.lambda_l1);
+        const float tr = l1_thresh(GR, sp.lambda_l1);
+        const float tp = l1_thresh(G, sp.lambda_l1);
+        gain = tl * tl / (HL + sp.lambda_l2) + tr * tr / (HR + sp.lambda_l2) -
+               tp * tp / (H + sp.lambda_l2);
+      }
     }
   }
   // Deterministic argmax reduce: higher gain wins; ties -> smaller bin.
@@ -947,9 +951,11 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
       float llo = lo, lhi = hi, rlo = lo, rhi = hi;
       if (mono != nullptr && mono[f] != 0) {
         const float wl =
-            fminf(fmaxf(-GL / (HL + sp.lambda_l2), lo), hi);
+            fminf(fmaxf(-l1_thresh(GL, sp.lambda_l1)
+                        / (HL + sp.lambda_l2), lo), hi);
         const float wr = fminf(
-            fmaxf(-(ns[0] - GL) / (ns[1] - HL + sp.lambda_l2), lo), hi);
+            fmaxf(-l1_thresh(ns[0] - GL, sp.lambda_l1)
+                  / (ns[1] - HL + sp.lambda_l2), lo), hi);
         const float mid = 0.5f * (wl + wr);
         if (mono[f] > 0) { lhi = mid; rlo = mid; }
         else { llo = mid; rhi = mid; }
@@ -1136,11 +1142,14 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
 __global__ void leaf_values_kernel(const float* __restrict__ node_stats,
                                    const float* __restrict__ node_bounds,
                                    float* __restrict__ leaf_values,
-                                   int total_nodes, float lambda_l2) {
+                                   int total_nodes, float lambda_l2,
+                                   float lambda_l1) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= total_nodes) return;
   const float* ns = node_stats + (int64_t)i * 3;
-  float v = (ns[1] != 0.f) ? (-ns[0] / (ns[1] + lambda_l2)) : 0.f;
+  float v = (ns[1] != 0.f)
+                ? (-l1_thresh(ns[0], lambda_l1) / (ns[1] + lambda_l2))
+                : 0.f;
   if (node_bounds != nullptr)
     v = fminf(fmaxf(v, node_bounds[2 * i]), node_bounds[2 * i + 1]);
   leaf_values[i] = v;
@@ -1525,11 +1534,12 @@ void gpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
 
 void gpu_leaf_values(const float* node_stats, const float* node_bounds,
                      float* leaf_values, int total_nodes, float lambda_l2,
+                     float lambda_l1,
                      void* stream) {
   const int grid = (total_nodes + kBlock - 1) / kBlock;
   hipLaunchKernelGGL(leaf_values_kernel, dim3(grid), dim3(kBlock), 0,
                      (hipStream_t)stream, node_stats, node_bounds,
-                     leaf_values, total_nodes, lambda_l2);
+                     leaf_values, total_nodes, lambda_l2, lambda_l1);
 }
 
 void gpu_update_preds(float* preds, const int32_t* node_ids,
