@@ -797,3 +797,27 @@ def test_appendix_a_hyperparameters():
     with pytest.raises(NotImplementedError):
         ydf.GradientBoostedTreesLearner(
             label="LABEL", missing_value_policy="LOCAL_IMPUTATION")
+
+
+def test_edge_cases_robustness():
+    """Degenerate inputs train or fail with clear errors (constant
+    features, single class, tiny N, all-NaN columns, no features)."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="l", num_trees=3, validation_ratio=0).train(
+        {"x": np.zeros(100, np.float32),
+         "y": np.arange(100, dtype=np.float32),
+         "l": np.array(["a", "b"] * 50)})
+    assert m.evaluate({"x": np.zeros(4, np.float32),
+                       "y": np.arange(4, dtype=np.float32),
+                       "l": np.array(["a", "b", "a", "b"])}).accuracy >= 0.5
+    ydf.GradientBoostedTreesLearner(
+        label="l", num_trees=2, validation_ratio=0).train(
+        {"x": np.full(60, np.nan, np.float32),
+         "z": np.arange(60, dtype=np.float32),
+         "l": np.array(["a", "b"] * 30)})
+    with pytest.raises(ValueError):
+        ydf.GradientBoostedTreesLearner(
+            label="l", features=[], num_trees=2,
+            validation_ratio=0).train(
+            {"x": np.arange(50, dtype=np.float32),
+             "l": np.array(["a", "b"] * 25)})

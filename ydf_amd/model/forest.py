@@ -272,6 +272,8 @@ def padded_boundaries(specs, max_bins: int = 256) -> np.ndarray:
     """Stacks per-column ragged cut lists into a dense [F, n_cuts] matrix,
     padding with +inf (padding adds no cut below any finite value, so bin
     assignment is unchanged)."""
+    if not specs:
+        raise ValueError("the dataset has no input feature columns")
     n_cuts = max(1, max((len(s.boundaries) if s.boundaries is not None else 0)
                         for s in specs))
     n_cuts = min(n_cuts, max_bins - 1)
