@@ -875,3 +875,21 @@ def test_end_to_end_user_journey(tmp_path):
     ydf.to_docker(model, str(tmp_path / "serve"))
     assert (tmp_path / "serve" / "Dockerfile").exists()
     assert model.describe("html").startswith("<h2>")
+
+
+def test_isolation_forest_evaluation_auc():
+    """ANOMALY_DETECTION evaluation reports AUC of the anomaly score
+    against binary labels (gaussians: reference IF test dataset)."""
+    import os
+
+    base = ("/root/reference/yggdrasil_decision_forests/test_data/"
+            "dataset")
+    if not os.path.exists(base):
+        pytest.skip("reference test_data not available")
+    pd = pytest.importorskip("pandas")
+    tr = pd.read_csv(f"{base}/gaussians_train.csv")
+    te = pd.read_csv(f"{base}/gaussians_test.csv")
+    m = ydf.IsolationForestLearner(
+        label="label",
+        features=[c for c in tr.columns if c != "label"]).train(tr)
+    assert m.evaluate(te).auc > 0.98

@@ -269,6 +269,10 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
             n_pos = int((labels > 0.5).sum())
             ev.auc_ci95 = auc_confidence_interval(
                 ev.auc, n_pos, len(labels) - n_pos)
+    elif task == Task.ANOMALY_DETECTION:
+        # labels: 1 = anomaly; predictions: anomaly score in [0, 1]
+        ev.auc = roc_auc(labels > 0.5, predictions)
+        ev.pr_auc = pr_auc(labels > 0.5, predictions)
     elif task == Task.REGRESSION:
         if w is None:
             ev.rmse = rmse(labels, predictions)
