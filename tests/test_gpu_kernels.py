@@ -510,3 +510,28 @@ def test_deep_mlp_gpu():
         label="LABEL", num_epochs=15, num_layers=2, layer_size=48,
         device="cuda").train(d)
     assert m.evaluate(d, device="cuda").auc > 0.75
+
+
+def test_local_imputation_gpu_vs_cpu():
+    """LOCAL_IMPUTATION forests are bit-comparable across devices (RF
+    integer gradients; same na-bin merge on both paths)."""
+    rng = np.random.RandomState(31)
+    n = 40000
+    x = rng.randn(n).astype(np.float32)
+    miss = rng.rand(n) < 0.3
+    xna = x.copy()
+    xna[miss] = np.nan
+    d = {"x": xna, "z": rng.randn(n).astype(np.float32),
+         "label": np.where(np.where(miss, 2.0, x) > 0.4, "a", "b")}
+    kw = dict(label="label", num_trees=3, max_depth=8,
+              bootstrap_training_dataset=False,
+              num_candidate_attributes=-1,
+              missing_value_policy="LOCAL_IMPUTATION",
+              compute_oob_performances=False)
+    m_cpu = ydf.RandomForestLearner(device="cpu", **kw).train(d)
+    m_gpu = ydf.RandomForestLearner(device="cuda", **kw).train(d)
+    np.testing.assert_array_equal(m_gpu.forest.feat, m_cpu.forest.feat)
+    np.testing.assert_array_equal(m_gpu.forest.na_right,
+                                  m_cpu.forest.na_right)
+    np.testing.assert_allclose(m_gpu.forest.thr, m_cpu.forest.thr,
+                               rtol=1e-5, atol=1e-6)

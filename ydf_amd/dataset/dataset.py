@@ -284,8 +284,11 @@ def create_vertical_dataset(
     min_vocab_frequency: int = 1,
     max_bins: int = 256,
     allow_na_conditions: bool = False,
+    keep_na: bool = False,
 ) -> VerticalDataset:
-    """Builds a VerticalDataset, inferring the dataspec unless provided."""
+    """Builds a VerticalDataset, inferring the dataspec unless provided.
+    keep_na: numerical NaN passes through (LOCAL_IMPUTATION training);
+    categorical missing encodes as -1."""
     cols = _to_column_dict(data)
     if dataspec is None:
         dataspec = infer_dataspec(cols, label=label, task=task,
@@ -301,7 +304,7 @@ def create_vertical_dataset(
         src = spec.set_source or spec.name
         if src not in cols:
             raise ValueError(f"missing feature column {src!r}")
-        X[i] = encode_column(cols[src], spec)
+        X[i] = encode_column(cols[src], spec, keep_na=keep_na)
 
     label_values = None
     if dataspec.label is not None and dataspec.label in cols:

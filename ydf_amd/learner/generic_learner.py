@@ -30,11 +30,13 @@ class GenericLearner:
                  num_threads: Optional[int] = None):
         self.allow_na_conditions = allow_na_conditions
         self.pure_serving_model = pure_serving_model
-        if missing_value_policy != "GLOBAL_IMPUTATION":
+        if missing_value_policy not in ("GLOBAL_IMPUTATION",
+                                        "LOCAL_IMPUTATION"):
             raise NotImplementedError(
-                "missing_value_policy: only GLOBAL_IMPUTATION is "
-                "implemented (LOCAL/RANDOM_LOCAL: ROADMAP; learned "
-                "'is missing' splits via allow_na_conditions=True)")
+                "missing_value_policy: GLOBAL_IMPUTATION and "
+                "LOCAL_IMPUTATION are implemented (RANDOM_LOCAL: "
+                "ROADMAP)")
+        self.missing_value_policy = missing_value_policy
         if categorical_algorithm != "CART":
             raise NotImplementedError(
                 "categorical_algorithm: only CART (sorted set-splits) "
@@ -122,11 +124,14 @@ class GenericLearner:
     def _bin_matrix(self, ds_X: np.ndarray, cat_feats: np.ndarray,
                     bnd: np.ndarray, device: torch.device) -> torch.Tensor:
         """Bins numericals by quantile cuts; categorical codes pass through
-        as their own bin index (clamped to 255)."""
+        as their own bin index (clamped to 255). Under LOCAL_IMPUTATION
+        NaN rows land in reserved bin 255."""
+        na_mode = getattr(self, "missing_value_policy",
+                          "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"
         X = torch.from_numpy(np.ascontiguousarray(ds_X)).to(device)
         bnd_t = torch.from_numpy(bnd).to(device)
         bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
-        ops.bin_data(X, bnd_t, bins)
+        ops.bin_data(X, bnd_t, bins, na_to_255=na_mode)
         ci = np.nonzero(cat_feats)[0]
         if ci.size:
             idx = torch.from_numpy(ci).to(device)
@@ -167,11 +172,15 @@ class GenericLearner:
             if self.weights_col is not None and features is None:
                 features = [c for c in cols
                             if c not in (self.label, self.weights_col)]
+            local_na = getattr(self, "missing_value_policy",
+                               "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"
             ds = create_vertical_dataset(
                 cols, label=self.label, task=self._task,
                 features=features, max_vocab_count=self.max_vocab_count,
                 min_vocab_frequency=self.min_vocab_frequency,
-                allow_na_conditions=self.allow_na_conditions)
+                allow_na_conditions=self.allow_na_conditions,
+                keep_na=local_na,
+                max_bins=255 if local_na else 256)
             if self.weights_col is not None:
                 if self.weights_col not in cols:
                     raise ValueError(

@@ -364,6 +364,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             max_num_nodes=hp.get("max_num_nodes", 31),
             focal_gamma=hp.get("focal_loss_gamma", 2.0),
             focal_alpha=hp.get("focal_loss_alpha", 0.5),
+            na_mode=(getattr(self, "missing_value_policy",
+                             "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"),
             dart_dropout=(hp.get("dart_dropout", 0.01)
                           if hp.get("forest_extraction") == "DART" else 0.0),
             **obl,
@@ -395,6 +397,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                 label_classes=classes, init_predictions=init_preds,
                 num_trees_per_iter=C, activation=activation,
                 metadata={"feature_gains": gains,
+                          "missing_value_policy": getattr(
+                              self, "missing_value_policy",
+                              "GLOBAL_IMPUTATION"),
                           "ranking_group": self.ranking_group,
                           "label_event_observed":
                               self.label_event_observed,
@@ -718,6 +723,8 @@ class RandomForestLearner(GenericLearner):
             max_duration_seconds=hp.get(
                 "maximum_training_duration_seconds", -1.0),
             honest=hp.get("honest", False),
+            na_mode=(getattr(self, "missing_value_policy",
+                             "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"),
             honest_ratio=hp.get("honest_ratio_leaf_examples", 0.5),
             honest_fixed_separation=hp.get("honest_fixed_separation",
                                            False),
@@ -765,6 +772,8 @@ class RandomForestLearner(GenericLearner):
             num_trees_per_iter=C, activation="identity",
             metadata={"feature_gains": self._feature_gains(
                 trees, [c.name for c in ds.dataspec.feature_columns]),
+                "missing_value_policy": getattr(
+                    self, "missing_value_policy", "GLOBAL_IMPUTATION"),
                 "winner_take_all": wta})
         model._self_evaluation = oob_eval
         self._finalize_model(model)

@@ -57,6 +57,10 @@ class TrainerConfig:
     selgb_ratio: float = 0.01   # SelGB share of kept negatives
     cat_smooth: float = 1.0      # l2_categorical_regularization
     n_classes: int = 2           # multinomial only
+    # LOCAL_IMPUTATION (reference missing_value_policy): NaN rows live
+    # in reserved bin 255 and are folded into the node-local mean bin at
+    # scan time; the winner records a per-node na direction
+    na_mode: bool = False
     focal_gamma: float = 2.0     # focal loss misprediction exponent
     focal_alpha: float = 0.5     # focal loss positive-class weight
     # DART (Rashmi & Gilad-Bachrach 2015; reference forest_extraction=DART,
@@ -123,6 +127,8 @@ class HostTree:
     oblique: Optional[dict] = None
     # DART: final absolute leaf scale (bakes shrinkage + dropout rescales)
     scale: float = 1.0
+    # LOCAL_IMPUTATION: per-node "missing goes right" bits
+    na: "np.ndarray" = None
 
 
 @dataclasses.dataclass
@@ -191,6 +197,10 @@ class ForestTrainer:
         self.base_F = self.F
         self.raw = raw
         self.valid_raw = valid_raw
+        if cfg.na_mode and cfg.oblique_projections > 0:
+            raise NotImplementedError(
+                "LOCAL_IMPUTATION + SPARSE_OBLIQUE is not supported "
+                "(projections would propagate NaN)")
         if self.P > 0:
             assert raw is not None, \
                 "oblique training needs the raw feature matrix (raw=)"
@@ -312,6 +322,11 @@ class ForestTrainer:
         self.tree_masks = torch.zeros((self.total_nodes, 4),
                                       dtype=torch.int64, device=dev) \
             if self.has_cats else None
+        self.na_mb_nf = torch.zeros((self.max_slots, self.F),
+                                    dtype=torch.int32, device=dev) \
+            if cfg.na_mode else None
+        self.tree_na = torch.zeros(self.total_nodes, dtype=torch.uint8,
+                                   device=dev) if cfg.na_mode else None
         self.bg_nf = torch.empty((self.max_slots, self.F),
                                  dtype=torch.float32, device=dev)
         self.bb_nf = torch.empty((self.max_slots, self.F), dtype=torch.int32,
@@ -489,6 +504,8 @@ class ForestTrainer:
                            self.best_gain, 0, 1, cfg.lambda_l2,
                            cfg.min_hessian, cfg.min_examples, cfg.min_gain,
                            feat_mask=fm, lambda_l1=cfg.lambda_l1)
+            assert not cfg.na_mode, \
+                "BEST_FIRST_GLOBAL + LOCAL_IMPUTATION not supported"
             f = int(self.best_feat[0].item())
             b = int(self.best_bin[0].item())
             g = float(self.best_gain[0].item())
@@ -559,6 +576,8 @@ class ForestTrainer:
         self.node_stats.zero_()
         if self.tree_masks is not None:
             self.tree_masks.zero_()
+        if self.tree_na is not None:
+            self.tree_na.zero_()
         if self.node_bounds is not None:
             self.node_bounds[:, 0] = float("-inf")
             self.node_bounds[:, 1] = float("inf")
@@ -817,7 +836,9 @@ class ForestTrainer:
                                masks=self.tree_masks,
                                cat_smooth=cfg.cat_smooth, mono=self.mono,
                                node_bounds=self.node_bounds,
-                               lambda_l1=cfg.lambda_l1)
+                               lambda_l1=cfg.lambda_l1,
+                               na_meanb_nf=self.na_mb_nf,
+                               tree_na=self.tree_na)
 
             prev_fit = n_active <= self.max_slots
             if self.use_hist_sub and prev_fit and level + 1 < cfg.max_depth:
@@ -832,7 +853,8 @@ class ForestTrainer:
             ops.update_node_ids(self.bins, self.node_ids, slot_map,
                                 self.best_feat, self.best_bin, level_base,
                                 level_size, cat_flags=self.cat_flags,
-                                masks=self.tree_masks)
+                                masks=self.tree_masks,
+                                tree_na=self.tree_na)
 
             if level + 1 < cfg.max_depth:
                 # choose next level's open nodes (host sync; deterministic
@@ -887,6 +909,8 @@ class ForestTrainer:
             if self.tree_masks is not None else None,
             gain=self.tree_gain.cpu().numpy().copy(),
             oblique=oblique,
+            na=self.tree_na.cpu().numpy().copy()
+            if self.tree_na is not None else None,
         )
 
     def _dense_level(self, tree_idx: int, level: int, need: int,
@@ -1014,7 +1038,8 @@ class ForestTrainer:
                        feat_mask=feat_mask, cat_flags=self.cat_flags,
                        masks=self.tree_masks, cat_smooth=cfg.cat_smooth,
                        mono=self.mono, node_bounds=self.node_bounds,
-                       lambda_l1=cfg.lambda_l1)
+                       lambda_l1=cfg.lambda_l1,
+                       na_meanb_nf=self.na_mb_nf, tree_na=self.tree_na)
         fits = True
         if self.use_hist_sub and level + 1 < cfg.max_depth:
             self.hist_prev[:level_size].copy_(hist_view)
@@ -1027,7 +1052,7 @@ class ForestTrainer:
         ops.update_node_ids(self.bins, self.node_ids, identity,
                             self.best_feat, self.best_bin, level_base,
                             level_size, cat_flags=self.cat_flags,
-                            masks=self.tree_masks)
+                            masks=self.tree_masks, tree_na=self.tree_na)
         return fits
 
     def capture_step_graph(self, preds: torch.Tensor, labels: torch.Tensor,
@@ -1103,7 +1128,7 @@ class ForestTrainer:
                 self.tree_feat[level_base:level_base + level_size],
                 self.tree_bin[level_base:level_base + level_size],
                 level_base, level_size, cat_flags=self.cat_flags,
-                masks=self.tree_masks)
+                masks=self.tree_masks, tree_na=self.tree_na)
 
 
 def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
@@ -1215,6 +1240,10 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             raise NotImplementedError(
                 "DART replays dropped trees from the complete-tree "
                 "buffers; BEST_FIRST_GLOBAL trees live on implicit keys")
+        if cfg.na_mode:
+            raise NotImplementedError(
+                "DART + LOCAL_IMPUTATION: dropped-tree replay does not "
+                "carry the per-tree na bits yet")
         dart_rec = []   # per tree: (feat_dev, bin_dev, masks_dev, leaf_dev)
         dart_scale = []  # per tree: current absolute leaf scale
         dart_rng = np.random.RandomState(cfg.seed ^ 0x5bd1e995)

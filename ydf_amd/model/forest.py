@@ -170,9 +170,12 @@ def host_tree_to_flat(tree: HostTree, boundaries: np.ndarray,
     li = np.nonzero(~n_int)[0]
     thr[li] = tree.leaf_value[nodes[li]] * leaf_scale
     cover = tree.counts[nodes].astype(np.float32)
+    na = None
+    if getattr(tree, "na", None) is not None:
+        na = np.where(n_int, tree.na[nodes], 0).astype(np.uint8)
     obl = (np.asarray(obl_ranges, np.int32).reshape(-1, 2),
            np.asarray(obl_attr, np.int32), np.asarray(obl_w, np.float32))
-    return feat, thr, left, cat_idx, masks, cover, obl
+    return feat, thr, left, cat_idx, masks, cover, obl, na
 
 
 def best_first_tree_to_flat(tree, boundaries: np.ndarray,
@@ -223,13 +226,17 @@ def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
     mask_off = 0
     obl_off = 0
     term_off = 0
+    na_list = []
     for t in trees:
         if isinstance(t, HostTree):
-            f, th, lf, ci, mk, cv, (orng, oat, ow) = host_tree_to_flat(
-                t, boundaries, leaf_scale, cat_feats)
+            f, th, lf, ci, mk, cv, (orng, oat, ow), na = \
+                host_tree_to_flat(t, boundaries, leaf_scale, cat_feats)
         else:  # BestFirstTree (leaf-wise growth)
             f, th, lf, ci, mk, cv, (orng, oat, ow) = \
                 best_first_tree_to_flat(t, boundaries, leaf_scale)
+            na = None
+        na_list.append(na if na is not None
+                       else np.zeros(len(f), np.uint8))
         lf = np.where(f >= 0, lf + off, 0)
         ci = np.where(ci >= 0, ci + mask_off, ci)
         ci = np.where(ci <= -2, ci - obl_off, ci)
@@ -265,6 +272,8 @@ def build_flat_forest(trees: List[HostTree], boundaries: np.ndarray,
         obl_attr=np.concatenate(attr_list).astype(np.int32)
         if attr_list else None,
         obl_w=np.concatenate(w_list).astype(np.float32) if w_list else None,
+        na_right=np.concatenate(na_list).astype(np.uint8)
+        if na_list else None,
     )
 
 

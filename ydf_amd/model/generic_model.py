@@ -155,7 +155,9 @@ class GenericModel:
         specs = self.dataspec.feature_columns
         n = len(next(iter(cols.values()))) if cols else 0
         X = np.empty((len(specs), n), dtype=np.float32)
-        keep_na = self.forest.has_na_routing
+        keep_na = self.forest.has_na_routing or (
+            (self.metadata or {}).get("missing_value_policy")
+            == "LOCAL_IMPUTATION")
         for i, spec in enumerate(specs):
             src = spec.set_source or spec.name
             if src not in cols:

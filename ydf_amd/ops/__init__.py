@@ -31,8 +31,10 @@ def _chk(t: torch.Tensor, dtype, name: str) -> None:
 
 
 def bin_data(x: torch.Tensor, boundaries: torch.Tensor,
-             out: torch.Tensor) -> torch.Tensor:
-    """x [F,N] f32, boundaries [F,n_cuts] f32 (ascending) -> out [F,N] u8."""
+             out: torch.Tensor, na_to_255: bool = False) -> torch.Tensor:
+    """x [F,N] f32, boundaries [F,n_cuts] f32 (ascending) -> out [F,N]
+    u8. na_to_255: NaN lands in the reserved bin 255
+    (LOCAL_IMPUTATION)."""
     F, N = x.shape
     n_cuts = boundaries.shape[1]
     _chk(x, torch.float32, "x")
@@ -40,10 +42,10 @@ def bin_data(x: torch.Tensor, boundaries: torch.Tensor,
     _chk(out, torch.uint8, "out")
     if x.is_cuda:
         _C.gpu_bin_data(x.data_ptr(), boundaries.data_ptr(), out.data_ptr(),
-                        N, F, n_cuts, _stream())
+                        N, F, n_cuts, 1 if na_to_255 else 0, _stream())
     else:
         _C.cpu_bin_data(x.data_ptr(), boundaries.data_ptr(), out.data_ptr(),
-                        N, F, n_cuts)
+                        N, F, n_cuts, 1 if na_to_255 else 0)
     return out
 
 
@@ -198,7 +200,8 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
                n_slots: int, lambda_l2: float, min_hessian: float,
                min_examples: int, min_gain: float, feat_mask=None,
                cat_flags=None, masks=None, cat_smooth: float = 1.0,
-               mono=None, node_bounds=None, lambda_l1: float = 0.0):
+               mono=None, node_bounds=None, lambda_l1: float = 0.0,
+               na_meanb_nf=None, tree_na=None):
     F = hist.shape[1]
     n_bins = hist.shape[2]
     mp = feat_mask.data_ptr() if feat_mask is not None else 0
@@ -210,7 +213,10 @@ def split_scan(hist: torch.Tensor, abs_of_slot: torch.Tensor,
             best_gain_nf.data_ptr(), best_bin_nf.data_ptr(),
             best_feat.data_ptr(), best_bin.data_ptr(), best_gain.data_ptr(),
             mp, cf, mk, mn, nb, F, n_bins, slot0, n_slots, lambda_l2,
-            min_hessian, min_examples, min_gain, cat_smooth, lambda_l1)
+            min_hessian, min_examples, min_gain, cat_smooth, lambda_l1,
+            na_meanb_nf.data_ptr() if na_meanb_nf is not None else 0,
+            tree_na.data_ptr() if tree_na is not None else 0,
+            1 if na_meanb_nf is not None else 0)
     if hist.is_cuda:
         _C.gpu_split_scan(*args, _stream())
     else:
@@ -259,20 +265,21 @@ def subtract_hist(hist: torch.Tensor, hist_prev: torch.Tensor,
 def update_node_ids(bins: torch.Tensor, node_ids: torch.Tensor,
                     slot_map: torch.Tensor, best_feat: torch.Tensor,
                     best_bin: torch.Tensor, level_base: int, level_size: int,
-                    cat_flags=None, masks=None):
+                    cat_flags=None, masks=None, tree_na=None):
     F, N = bins.shape
     cf = cat_flags.data_ptr() if cat_flags is not None else 0
     mk = masks.data_ptr() if masks is not None else 0
+    na = tree_na.data_ptr() if tree_na is not None else 0
     if bins.is_cuda:
         _C.gpu_update_node_ids(bins.data_ptr(), node_ids.data_ptr(),
                                slot_map.data_ptr(), best_feat.data_ptr(),
-                               best_bin.data_ptr(), cf, mk, N, level_base,
-                               level_size, _stream())
+                               best_bin.data_ptr(), cf, mk, na, N,
+                               level_base, level_size, _stream())
     else:
         _C.cpu_update_node_ids(bins.data_ptr(), node_ids.data_ptr(),
                                slot_map.data_ptr(), best_feat.data_ptr(),
-                               best_bin.data_ptr(), cf, mk, N, level_base,
-                               level_size)
+                               best_bin.data_ptr(), cf, mk, na, N,
+                               level_base, level_size)
 
 
 def leaf_values(node_stats: torch.Tensor, out: torch.Tensor,

@@ -15,7 +15,7 @@ using ydfa::SplitParams;
 extern "C" {
 // train_kernels.hip
 void gpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
-                  void*);
+                  int, void*);
 void gpu_grad_hess(const float*, const float*, float*, int64_t, int, void*);
 void gpu_grad_hess_softmax(const float*, const float*, float*, int64_t, int,
                            int, void*);
@@ -43,15 +43,16 @@ void gpu_hist_build_gathered(const uint8_t*, const float*, const int32_t*,
 void gpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*,
                     const uint8_t*, unsigned long long*, const int8_t*,
-                    float*, int, int, int, int, SplitParams, void*);
+                    float*, int32_t*, uint8_t*, int, int, int, int,
+                    SplitParams, void*);
 void gpu_plan_level(const float*, const int32_t*, int, int, int, int,
                     int32_t*, uint8_t*, void*);
 void gpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int,
                        void*);
 void gpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, const uint8_t*,
-                         const unsigned long long*, int64_t, int, int,
-                         void*);
+                         const unsigned long long*, const uint8_t*,
+                         int64_t, int, int, void*);
 void gpu_leaf_values(const float*, const float*, float*, int, float,
                      float, void*);
 void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
@@ -71,7 +72,8 @@ void gpu_predict_forest_qs(const float*, int64_t, int, const int32_t*,
                            const int32_t*, const float*, int, float*,
                            float, float, void*);
 // cpu_ops.cpp
-void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int);
+void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
+                  int);
 void cpu_grad_hess(const float*, const float*, float*, int64_t, int);
 void cpu_grad_hess_softmax(const float*, const float*, float*, int64_t, int,
                            int);
@@ -82,13 +84,15 @@ void cpu_weighted_target(const float*, const float*, float*, int64_t);
 void cpu_split_scan(const float*, const int32_t*, float*, float*, int32_t*,
                     int32_t*, int32_t*, float*, const uint8_t*,
                     const uint8_t*, unsigned long long*, const int8_t*,
-                    float*, int, int, int, int, SplitParams);
+                    float*, int32_t*, uint8_t*, int, int, int, int,
+                    SplitParams);
 void cpu_plan_level(const float*, const int32_t*, int, int, int, int,
                     int32_t*, uint8_t*);
 void cpu_subtract_hist(float*, const float*, const uint8_t*, int, int, int);
 void cpu_update_node_ids(const uint8_t*, int32_t*, const int32_t*,
                          const int32_t*, const int32_t*, const uint8_t*,
-                         const unsigned long long*, int64_t, int, int);
+                         const unsigned long long*, const uint8_t*,
+                         int64_t, int, int);
 void cpu_leaf_values(const float*, const float*, float*, int, float,
                      float);
 void cpu_update_preds(float*, const int32_t*, const float*, int64_t, float);
@@ -116,7 +120,8 @@ T* P(uintptr_t p) {
   return reinterpret_cast<T*>(p);
 }
 SplitParams MakeSP(float lambda_l2, float min_hessian, int min_examples,
-                   float min_gain, float cat_smooth, float lambda_l1) {
+                   float min_gain, float cat_smooth, float lambda_l1,
+                   int na_mode) {
   SplitParams sp;
   sp.lambda_l2 = lambda_l2;
   sp.lambda_l1 = lambda_l1;
@@ -124,6 +129,7 @@ SplitParams MakeSP(float lambda_l2, float min_hessian, int min_examples,
   sp.min_examples = min_examples;
   sp.min_gain = min_gain;
   sp.cat_smooth = cat_smooth;
+  sp.na_mode = na_mode;
   return sp;
 }
 }  // namespace
@@ -136,9 +142,9 @@ PYBIND11_MODULE(_ydf_ops, m) {
   // --- GPU ---
   m.def("gpu_bin_data",
         [](uintptr_t x, uintptr_t bnd, uintptr_t out, int64_t N, int F,
-           int n_cuts, uintptr_t stream) {
+           int n_cuts, int na_to_255, uintptr_t stream) {
           gpu_bin_data(P<float>(x), P<float>(bnd), P<uint8_t>(out), N, F,
-                       n_cuts, (void*)stream);
+                       n_cuts, na_to_255, (void*)stream);
         },
         nogil);
   m.def("gpu_grad_hess",
@@ -240,6 +246,7 @@ PYBIND11_MODULE(_ydf_ops, m) {
            uintptr_t node_bounds, int F, int n_bins, int slot0,
            int n_slots, float lambda_l2, float min_hessian, int min_examples,
            float min_gain, float cat_smooth, float lambda_l1,
+           uintptr_t na_meanb_nf, uintptr_t tree_na, int na_mode,
            uintptr_t stream) {
           gpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
                          P<float>(node_stats), P<float>(best_gain_nf),
@@ -247,9 +254,10 @@ PYBIND11_MODULE(_ydf_ops, m) {
                          P<int32_t>(best_bin), P<float>(best_gain),
                          P<uint8_t>(feat_mask), P<uint8_t>(cat_flags),
                          P<unsigned long long>(masks), P<int8_t>(mono),
-                         P<float>(node_bounds), F, n_bins, slot0, n_slots,
+                         P<float>(node_bounds), P<int32_t>(na_meanb_nf),
+                         P<uint8_t>(tree_na), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
-                                min_gain, cat_smooth, lambda_l1),
+                                min_gain, cat_smooth, lambda_l1, na_mode),
                          (void*)stream);
         },
         nogil);
@@ -274,12 +282,13 @@ PYBIND11_MODULE(_ydf_ops, m) {
   m.def("gpu_update_node_ids",
         [](uintptr_t bins, uintptr_t node_ids, uintptr_t slot_map,
            uintptr_t best_feat, uintptr_t best_bin, uintptr_t cat_flags,
-           uintptr_t masks, int64_t N, int level_base, int level_size,
-           uintptr_t stream) {
+           uintptr_t masks, uintptr_t tree_na, int64_t N, int level_base,
+           int level_size, uintptr_t stream) {
           gpu_update_node_ids(P<uint8_t>(bins), P<int32_t>(node_ids),
                               P<int32_t>(slot_map), P<int32_t>(best_feat),
                               P<int32_t>(best_bin), P<uint8_t>(cat_flags),
-                              P<unsigned long long>(masks), N, level_base,
+                              P<unsigned long long>(masks),
+                              P<uint8_t>(tree_na), N, level_base,
                               level_size, (void*)stream);
         },
         nogil);
@@ -355,9 +364,9 @@ PYBIND11_MODULE(_ydf_ops, m) {
   // --- CPU ---
   m.def("cpu_bin_data",
         [](uintptr_t x, uintptr_t bnd, uintptr_t out, int64_t N, int F,
-           int n_cuts) {
+           int n_cuts, int na_to_255) {
           cpu_bin_data(P<float>(x), P<float>(bnd), P<uint8_t>(out), N, F,
-                       n_cuts);
+                       n_cuts, na_to_255);
         },
         nogil);
   m.def("cpu_grad_hess",
@@ -396,16 +405,18 @@ PYBIND11_MODULE(_ydf_ops, m) {
            uintptr_t cat_flags, uintptr_t masks, uintptr_t mono,
            uintptr_t node_bounds, int F, int n_bins, int slot0,
            int n_slots, float lambda_l2, float min_hessian, int min_examples,
-           float min_gain, float cat_smooth, float lambda_l1) {
+           float min_gain, float cat_smooth, float lambda_l1,
+           uintptr_t na_meanb_nf, uintptr_t tree_na, int na_mode) {
           cpu_split_scan(P<float>(hist), P<int32_t>(abs_of_slot),
                          P<float>(node_stats), P<float>(best_gain_nf),
                          P<int32_t>(best_bin_nf), P<int32_t>(best_feat),
                          P<int32_t>(best_bin), P<float>(best_gain),
                          P<uint8_t>(feat_mask), P<uint8_t>(cat_flags),
                          P<unsigned long long>(masks), P<int8_t>(mono),
-                         P<float>(node_bounds), F, n_bins, slot0, n_slots,
+                         P<float>(node_bounds), P<int32_t>(na_meanb_nf),
+                         P<uint8_t>(tree_na), F, n_bins, slot0, n_slots,
                          MakeSP(lambda_l2, min_hessian, min_examples,
-                                min_gain, cat_smooth, lambda_l1));
+                                min_gain, cat_smooth, lambda_l1, na_mode));
         },
         nogil);
   m.def("cpu_plan_level",
@@ -427,11 +438,13 @@ PYBIND11_MODULE(_ydf_ops, m) {
   m.def("cpu_update_node_ids",
         [](uintptr_t bins, uintptr_t node_ids, uintptr_t slot_map,
            uintptr_t best_feat, uintptr_t best_bin, uintptr_t cat_flags,
-           uintptr_t masks, int64_t N, int level_base, int level_size) {
+           uintptr_t masks, uintptr_t tree_na, int64_t N, int level_base,
+           int level_size) {
           cpu_update_node_ids(P<uint8_t>(bins), P<int32_t>(node_ids),
                               P<int32_t>(slot_map), P<int32_t>(best_feat),
                               P<int32_t>(best_bin), P<uint8_t>(cat_flags),
-                              P<unsigned long long>(masks), N, level_base,
+                              P<unsigned long long>(masks),
+                              P<uint8_t>(tree_na), N, level_base,
                               level_size);
         },
         nogil);

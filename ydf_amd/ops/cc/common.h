@@ -41,6 +41,9 @@ struct SplitParams {
   float min_gain;         // splits with gain <= min_gain become leaves
   float cat_smooth;       // l2_categorical_regularization (category order
                           // statistic smoothing; reference default 1.0)
+  int na_mode;            // LOCAL_IMPUTATION: bin 255 holds NaN rows;
+                          // the scan merges it into the node-local mean
+                          // bin and records the na direction
 };
 
 // l1 soft threshold: T(G) = sign(G) * max(|G| - l1, 0)

@@ -110,13 +110,17 @@ class ThreadPool {
 extern "C" {
 
 void cpu_bin_data(const float* x, const float* boundaries, uint8_t* out,
-                  int64_t N, int F, int n_cuts) {
+                  int64_t N, int F, int n_cuts, int na_to_255) {
   ThreadPool::Get().ParallelFor(F, [&](int f) {
     const float* bnd = boundaries + (int64_t)f * n_cuts;
     const float* xf = x + (int64_t)f * N;
     uint8_t* of = out + (int64_t)f * N;
     for (int64_t i = 0; i < N; ++i) {
       const float v = xf[i];
+      if (na_to_255 && std::isnan(v)) {
+        of[i] = 255;
+        continue;
+      }
       int lo = 0, hi = n_cuts;
       while (lo < hi) {
         const int mid = (lo + hi) >> 1;
@@ -234,7 +238,8 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                     int32_t* best_bin, float* best_gain,
                     const uint8_t* feat_mask, const uint8_t* cat_flags,
                     unsigned long long* masks, const int8_t* mono,
-                    float* node_bounds, int F, int n_bins, int slot0,
+                    float* node_bounds, int32_t* na_meanb_nf,
+                    uint8_t* tree_na, int F, int n_bins, int slot0,
                     int n_slots, SplitParams sp) {
   ThreadPool::Get().ParallelFor(n_slots, [&](int slot) {
     std::vector<int> order;
@@ -262,11 +267,33 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
       const bool is_cat = cat_flags != nullptr && cat_flags[f];
       if (is_cat) sorted_bin_order(hp, n_bins, sp.cat_smooth, order);
       auto bin_at = [&](int b) { return is_cat ? order[b] : b; };
+      // LOCAL_IMPUTATION: fold the NA bin (255) into the node-local
+      // mean bin before scanning (mirrors the GPU kernel)
+      int meanb = -1;
+      if (sp.na_mode && !is_cat) {
+        float wsum = 0.f, csum = 0.f;
+        for (int b = 0; b < n_bins - 1; ++b) {
+          wsum += (float)b * hp[b * 3 + 2];
+          csum += hp[b * 3 + 2];
+        }
+        meanb = csum > 0.f ? (int)(wsum / csum + 0.5f) : 0;
+        if (meanb > n_bins - 2) meanb = n_bins - 2;
+        if (na_meanb_nf != nullptr)
+          na_meanb_nf[(int64_t)slot * F + f] = meanb;
+      }
+      auto hv = [&](int b, int j) -> float {
+        float v = hp[b * 3 + j];
+        if (meanb >= 0) {
+          if (b == n_bins - 1) return 0.f;
+          if (b == meanb) v += hp[(n_bins - 1) * 3 + j];
+        }
+        return v;
+      };
       // Per-feature totals (matches the GPU kernel exactly; identical to
       // the feature-0 totals for any real histogram).
       float Gf = 0.f, Hf = 0.f, Cf = 0.f;
       for (int b = 0; b < n_bins; ++b) {
-        Gf += hp[b * 3]; Hf += hp[b * 3 + 1]; Cf += hp[b * 3 + 2];
+        Gf += hv(b, 0); Hf += hv(b, 1); Cf += hv(b, 2);
       }
       const float tpf = ydfa::l1_thresh(Gf, sp.lambda_l1);
       const float pterm = tpf * tpf / (Hf + sp.lambda_l2);
@@ -275,7 +302,7 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
       int fbin = 0;
       for (int b = 0; b < n_bins - 1; ++b) {
         const int bb = bin_at(b);
-        GL += hp[bb * 3]; HL += hp[bb * 3 + 1]; CL += hp[bb * 3 + 2];
+        GL += hv(bb, 0); HL += hv(bb, 1); CL += hv(bb, 2);
         const float GR = Gf - GL, HR = Hf - HL, CR = Cf - CL;
         if (CL >= sp.min_examples && CR >= sp.min_examples &&
             HL >= sp.min_hessian && HR >= sp.min_hessian) {
@@ -331,6 +358,16 @@ void cpu_split_scan(const float* hist, const int32_t* abs_of_slot,
     } else {
       for (int b = 0; b <= node_best_b; ++b) {
         GL += hp[b * 3]; HL += hp[b * 3 + 1]; CL += hp[b * 3 + 2];
+      }
+      if (sp.na_mode && na_meanb_nf != nullptr) {
+        const int mb = na_meanb_nf[(int64_t)slot * F + node_best_f];
+        if (tree_na != nullptr)
+          tree_na[abs_node] = mb > node_best_b ? (uint8_t)1 : (uint8_t)0;
+        if (mb <= node_best_b) {
+          GL += hp[(n_bins - 1) * 3];
+          HL += hp[(n_bins - 1) * 3 + 1];
+          CL += hp[(n_bins - 1) * 3 + 2];
+        }
       }
     }
     float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
@@ -395,7 +432,8 @@ void cpu_subtract_hist(float* hist, const float* hist_prev,
 void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
                          const int32_t* slot_map, const int32_t* best_feat,
                          const int32_t* best_bin, const uint8_t* cat_flags,
-                         const unsigned long long* masks, int64_t N,
+                         const unsigned long long* masks,
+                         const uint8_t* tree_na, int64_t N,
                          int level_base, int level_size) {
   const int nb = std::max(1, std::min<int>(ThreadPool::Get().size(),
                                            (int)(N / 16384) + 1));
@@ -415,6 +453,8 @@ void cpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
       if (cat_flags != nullptr && cat_flags[f])
         right = (int)((masks[(int64_t)nid * (kMaxBins / 64) + (b >> 6)]
                        >> (b & 63)) & 1ull);
+      else if (tree_na != nullptr && b == kMaxBins - 1)
+        right = tree_na[nid];
       else
         right = b > best_bin[slot] ? 1 : 0;
       node_ids[k] = 2 * nid + 1 + right;

@@ -45,7 +45,8 @@ constexpr int kBlock = 256;
 __global__ void bin_data_kernel(const float* __restrict__ x,
                                 const float* __restrict__ boundaries,
                                 uint8_t* __restrict__ out, int64_t N, int F,
-                                int n_cuts, int64_t rows_per_block) {
+                                int n_cuts, int64_t rows_per_block,
+                                int na_to_255) {
   __shared__ float bnd[kMaxBins - 1];
   const int f = blockIdx.x;
   for (int i = threadIdx.x; i < n_cuts; i += blockDim.x)
@@ -57,6 +58,10 @@ __global__ void bin_data_kernel(const float* __restrict__ x,
   uint8_t* of = out + (int64_t)f * N;
   for (int64_t i = row0 + threadIdx.x; i < row1; i += blockDim.x) {
     const float v = xf[i];
+    if (na_to_255 && isnan(v)) {
+      of[i] = 255;  // reserved NA bin (LOCAL_IMPUTATION)
+      continue;
+    }
     int lo = 0, hi = n_cuts;
     while (lo < hi) {
       const int mid = (lo + hi) >> 1;
@@ -767,6 +772,7 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
                                        const uint8_t* __restrict__ feat_mask,
                                        const uint8_t* __restrict__ cat_flags,
                                        const int8_t* __restrict__ mono,
+                                       int32_t* __restrict__ na_meanb_nf,
                                        int F, int n_bins, int slot0,
                                        SplitParams sp) {
   const int slot = blockIdx.x;
@@ -805,6 +811,36 @@ __global__ void split_scan_feat_kernel(const float* __restrict__ hist,
   sh[b] = rh0;
   sc[b] = rc0;
   __syncthreads();
+  if (sp.na_mode && (cat_flags == nullptr || !cat_flags[f])) {
+    // LOCAL_IMPUTATION (decision_tree.proto:85-103): bin 255 holds the
+    // node's MISSING rows; fold them into the node-local mean bin and
+    // remember it so the winning split can store the na direction
+    __shared__ int s_meanb;
+    if (b == 0) {
+      float wsum = 0.f, csum = 0.f;
+      for (int i = 0; i < n_bins - 1; ++i) {
+        wsum += (float)i * sc[i];
+        csum += sc[i];
+      }
+      s_meanb = csum > 0.f ? (int)(wsum / csum + 0.5f) : 0;
+      if (s_meanb > n_bins - 2) s_meanb = n_bins - 2;
+      if (na_meanb_nf != nullptr)
+        na_meanb_nf[(int64_t)slot * F + f] = s_meanb;
+    }
+    __syncthreads();
+    if (b == s_meanb) {
+      sg[b] += sg[n_bins - 1];
+      sh[b] += sh[n_bins - 1];
+      sc[b] += sc[n_bins - 1];
+    }
+    __syncthreads();
+    if (b == n_bins - 1) {
+      sg[b] = 0.f;
+      sh[b] = 0.f;
+      sc[b] = 0.f;
+    }
+    __syncthreads();
+  }
   for (int off = 1; off < n_bins; off <<= 1) {
     float tg = 0.f, th = 0.f, tc = 0.f;
     if (b >= off) { tg = sg[b - off]; th = sh[b - off]; tc = sc[b - off]; }
@@ -883,6 +919,8 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
                                     unsigned long long* __restrict__ masks,
                                     const int8_t* __restrict__ mono,
                                     float* __restrict__ node_bounds,
+                                    const int32_t* __restrict__ na_meanb_nf,
+                                    uint8_t* __restrict__ tree_na,
                                     int F, int n_bins, int slot0,
                                     SplitParams sp) {
   const int slot = blockIdx.x;
@@ -939,6 +977,18 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
     float GL = 0.f, HL = 0.f, CL = 0.f;
     for (int bb = 0; bb <= bin; ++bb) {
       GL += hp[bb * 3]; HL += hp[bb * 3 + 1]; CL += hp[bb * 3 + 2];
+    }
+    if (sp.na_mode && na_meanb_nf != nullptr) {
+      // the scan folded the NA bin (255) into the node-local mean bin;
+      // mirror that here and record the na direction for routing
+      const int meanb = na_meanb_nf[(int64_t)slot * F + f];
+      if (tree_na != nullptr)
+        tree_na[abs_node] = meanb > bin ? (uint8_t)1 : (uint8_t)0;
+      if (meanb <= bin) {
+        GL += hp[(n_bins - 1) * 3];
+        HL += hp[(n_bins - 1) * 3 + 1];
+        CL += hp[(n_bins - 1) * 3 + 2];
+      }
     }
     const float* ns = node_stats + (int64_t)abs_node * 3;
     float* nl = node_stats + (int64_t)(2 * abs_node + 1) * 3;
@@ -1081,6 +1131,7 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
                                        const uint8_t* __restrict__ cat_flags,
                                        const unsigned long long* __restrict__
                                            masks,
+                                       const uint8_t* __restrict__ tree_na,
                                        int64_t N, int level_base,
                                        int level_size) {
   // 4-row unroll: four independent (node -> split -> bin) load chains in
@@ -1113,6 +1164,8 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
       if (cat_flags != nullptr && cat_flags[f[u]])
         right = (int)((masks[(int64_t)nid[u] * (kMaxBins / 64) + (b >> 6)]
                        >> (b & 63)) & 1ull);
+      else if (tree_na != nullptr && b == kMaxBins - 1)
+        right = tree_na[nid[u]];  // missing row: stored na direction
       else
         right = b > sbin[u];
       node_ids[k + u * stride] = 2 * nid[u] + 1 + right;
@@ -1131,6 +1184,8 @@ __global__ void update_node_ids_kernel(const uint8_t* __restrict__ bins,
     if (cat_flags != nullptr && cat_flags[f])
       right = (int)((masks[(int64_t)nid * (kMaxBins / 64) + (b >> 6)]
                      >> (b & 63)) & 1ull);
+    else if (tree_na != nullptr && b == kMaxBins - 1)
+      right = tree_na[nid];
     else
       right = b > best_bin[slot];
     node_ids[k] = 2 * nid + 1 + right;
@@ -1231,12 +1286,13 @@ static inline int hist_block_threads() {
 extern "C" {
 
 void gpu_bin_data(const float* x, const float* boundaries, uint8_t* out,
-                  int64_t N, int F, int n_cuts, void* stream) {
+                  int64_t N, int F, int n_cuts, int na_to_255,
+                  void* stream) {
   const int chunks = row_chunks(N, F);
   const int64_t rpb = (N + chunks - 1) / chunks;
   hipLaunchKernelGGL(bin_data_kernel, dim3(F, chunks), dim3(kBlock), 0,
                      (hipStream_t)stream, x, boundaries, out, N, F, n_cuts,
-                     rpb);
+                     rpb, na_to_255);
 }
 
 static int elem_grid(int64_t N, int cap = 2048) {
@@ -1488,17 +1544,18 @@ void gpu_split_scan(const float* hist, const int32_t* abs_of_slot,
                     int32_t* best_bin, float* best_gain,
                     const uint8_t* feat_mask, const uint8_t* cat_flags,
                     unsigned long long* masks, const int8_t* mono,
-                    float* node_bounds, int F, int n_bins, int slot0,
+                    float* node_bounds, int32_t* na_meanb_nf,
+                    uint8_t* tree_na, int F, int n_bins, int slot0,
                     int n_slots, SplitParams sp, void* stream) {
   hipLaunchKernelGGL(split_scan_feat_kernel, dim3(n_slots, F), dim3(n_bins),
                      0, (hipStream_t)stream, hist, abs_of_slot, node_stats,
                      best_gain_nf, best_bin_nf, feat_mask, cat_flags, mono,
-                     F, n_bins, slot0, sp);
+                     na_meanb_nf, F, n_bins, slot0, sp);
   hipLaunchKernelGGL(split_select_kernel, dim3(n_slots), dim3(kBlock), 0,
                      (hipStream_t)stream, hist, abs_of_slot, best_gain_nf,
                      best_bin_nf, node_stats, best_feat, best_bin, best_gain,
-                     cat_flags, masks, mono, node_bounds, F, n_bins, slot0,
-                     sp);
+                     cat_flags, masks, mono, node_bounds, na_meanb_nf,
+                     tree_na, F, n_bins, slot0, sp);
 }
 
 void gpu_plan_level(const float* node_stats, const int32_t* prev_best_feat,
@@ -1524,11 +1581,13 @@ void gpu_subtract_hist(float* hist, const float* hist_prev,
 void gpu_update_node_ids(const uint8_t* bins, int32_t* node_ids,
                          const int32_t* slot_map, const int32_t* best_feat,
                          const int32_t* best_bin, const uint8_t* cat_flags,
-                         const unsigned long long* masks, int64_t N,
+                         const unsigned long long* masks,
+                         const uint8_t* tree_na, int64_t N,
                          int level_base, int level_size, void* stream) {
   hipLaunchKernelGGL(update_node_ids_kernel, dim3(elem_grid(N, 4096)),
                      dim3(kBlock), 0, (hipStream_t)stream, bins, node_ids,
-                     slot_map, best_feat, best_bin, cat_flags, masks, N,
+                     slot_map, best_feat, best_bin, cat_flags, masks,
+                     tree_na, N,
                      level_base, level_size);
 }
 
