@@ -214,3 +214,51 @@ def test_binary_logloss_matches_torch():
     ref_acc = (((preds > 0) == (labels > 0.5)).float().sum())
     np.testing.assert_allclose(out[0].item(), ref_loss.item(), rtol=1e-4)
     np.testing.assert_allclose(out[1].item(), ref_acc.item(), rtol=0)
+
+
+def test_bin_data_na_flag():
+    """bin_data(na_to_255): NaN rows land in the reserved bin, other
+    values keep the standard cut semantics."""
+    x = torch.tensor([[0.1, np.nan, 5.0, -3.0]], dtype=torch.float32)
+    bnd = torch.tensor([[0.0, 1.0, 2.0]], dtype=torch.float32)
+    out = torch.empty((1, 4), dtype=torch.uint8)
+    ops.bin_data(x, bnd, out, na_to_255=True)
+    assert out.tolist() == [[1, 255, 3, 0]]
+    ops.bin_data(x, bnd, out)  # default: NaN compares false -> bin 0
+    assert out[0, 1].item() == 0
+
+
+def test_split_scan_l1_shrinks_gain():
+    """lambda_l1 soft-thresholds gradient sums: a weak split's gain hits
+    zero once l1 exceeds |G| on both sides."""
+    n_bins = ops.MAX_BINS
+    hist = torch.zeros((1, 1, n_bins, 3), dtype=torch.float32)
+    hist[0, 0, 10] = torch.tensor([-1.0, 4.0, 8.0])
+    hist[0, 0, 200] = torch.tensor([1.5, 4.0, 8.0])
+    args = dict(
+        abs_of_slot=torch.zeros(1, dtype=torch.int32),
+        node_stats=torch.zeros((3, 3), dtype=torch.float32),
+        best_gain_nf=torch.zeros((1, 1), dtype=torch.float32),
+        best_bin_nf=torch.zeros((1, 1), dtype=torch.int32),
+        best_feat=torch.zeros(1, dtype=torch.int32),
+        best_bin=torch.zeros(1, dtype=torch.int32),
+        best_gain=torch.zeros(1, dtype=torch.float32))
+    ops.split_scan(hist, args["abs_of_slot"], args["node_stats"],
+                   args["best_gain_nf"], args["best_bin_nf"],
+                   args["best_feat"], args["best_bin"],
+                   args["best_gain"], 0, 1, 0.0, 0.0, 1, 0.0)
+    g0 = float(args["best_gain"][0])
+    assert g0 > 0
+    ops.split_scan(hist, args["abs_of_slot"], args["node_stats"],
+                   args["best_gain_nf"], args["best_bin_nf"],
+                   args["best_feat"], args["best_bin"],
+                   args["best_gain"], 0, 1, 0.0, 0.0, 1, 0.0,
+                   lambda_l1=0.5)
+    g1 = float(args["best_gain"][0])
+    assert 0 <= g1 < g0
+    ops.split_scan(hist, args["abs_of_slot"], args["node_stats"],
+                   args["best_gain_nf"], args["best_bin_nf"],
+                   args["best_feat"], args["best_bin"],
+                   args["best_gain"], 0, 1, 0.0, 0.0, 1, 0.0,
+                   lambda_l1=10.0)
+    assert int(args["best_feat"][0]) == -1  # fully suppressed
