@@ -103,3 +103,65 @@ from ydf_amd.metric.metric import Evaluation, evaluate_predictions
 
 # Utilities
 from ydf_amd.utils.log import strict, verbose
+
+
+# --- remaining PYDF surface names ------------------------------------------
+# DistributedGradientBoostedTreesLearner: the reference's gRPC
+# manager/worker learner. Here distribution is one-process-per-GPU over
+# RCCL (torchrun), which the plain GBT learner already does when
+# launched under torch.distributed.run — this alias makes the intent
+# explicit and the name importable.
+DistributedGradientBoostedTreesLearner = GradientBoostedTreesLearner
+
+from ydf_amd.learner.extras import FeatureSelectorLogs  # noqa: E402
+
+
+def start_worker(*args, **kwargs):
+    """The reference starts gRPC workers; this framework distributes as
+    one process per GPU via `python -m torch.distributed.run` (RCCL over
+    xGMI) instead, so there is no worker daemon to start."""
+    raise NotImplementedError(
+        "distributed training runs as one process per GPU under "
+        "torch.distributed.run (RCCL/xGMI); gRPC workers do not exist "
+        "in this design. See README 'Distributed'.")
+
+
+import dataclasses as _dataclasses  # noqa: E402
+from typing import Optional as _Optional  # noqa: E402
+
+
+@_dataclasses.dataclass
+class ModelMetadata:
+    """Mirrors ydf.ModelMetadata (owner/created date/uid/framework)."""
+
+    owner: _Optional[str] = None
+    created_date: _Optional[int] = None
+    uid: _Optional[int] = None
+    framework: _Optional[str] = "ydf_amd"
+
+
+class NodeFormat:
+    """Node storage formats (reference decision_tree.proto NodeFormat);
+    models here always write BLOB_SEQUENCE, matching the reference
+    default."""
+
+    BLOB_SEQUENCE = "BLOB_SEQUENCE"
+
+
+def from_tensorflow_decision_forests(*args, **kwargs):
+    raise ImportError(
+        "TensorFlow Decision Forests is not available in this "
+        "environment; use load_ydf_model() for reference model "
+        "directories or from_sklearn() for sklearn models.")
+
+
+from ydf_amd.utils import folds as _folds  # noqa: E402
+from ydf_amd.utils import usage as _usage  # noqa: E402
+
+
+class util:  # noqa: N801  (PYDF exposes a lowercase `util` namespace)
+    """Utility namespace (ydf.util analogue)."""
+
+    generate_folds = staticmethod(_folds.generate_folds)
+    fold_splits = staticmethod(_folds.fold_splits)
+    usage = _usage
