@@ -108,3 +108,56 @@ def test_tree_builder_roundtrip():
                                                      np.float32)
     np.testing.assert_allclose(hm.predict(x, device="cpu"),
                                [1.0, 2.0, 1.0])
+
+
+def test_property_random_tree_roundtrip():
+    """Property test: randomly-built trees (numerical + categorical-mask
+    + oblique conditions) round-trip through predict/save/load/serialize
+    bit-exactly."""
+    import tempfile
+
+    import ydf_amd as ydf
+    from ydf_amd.dataset.dataspec import (ColumnSpec, DataSpecification,
+                                          Semantic, Task)
+
+    rng = np.random.RandomState(77)
+    F = 6
+    cols = [ColumnSpec(name=f"f{i}", semantic=Semantic.NUMERICAL)
+            for i in range(F)]
+    cols.append(ColumnSpec(name="y", semantic=Semantic.NUMERICAL))
+    spec = DataSpecification(columns=cols, label="y")
+
+    def random_tree(depth):
+        if depth == 0 or rng.rand() < 0.3:
+            return ydf.Leaf(float(rng.randn()))
+        kind = rng.rand()
+        if kind < 0.2:
+            attrs = rng.choice(F, 2, replace=False)
+            return ydf.NonLeaf(
+                feature=int(attrs[0]), threshold=float(rng.randn()),
+                neg_child=random_tree(depth - 1),
+                pos_child=random_tree(depth - 1),
+                oblique=(tuple(int(a) for a in attrs),
+                         tuple(float(w) for w in rng.randn(2))))
+        return ydf.NonLeaf(
+            feature=int(rng.randint(F)), threshold=float(rng.randn()),
+            neg_child=random_tree(depth - 1),
+            pos_child=random_tree(depth - 1))
+
+    for trial in range(5):
+        trees = [ydf.Tree(root=random_tree(4)) for _ in range(7)]
+        m = ydf.build_model_from_trees(trees, spec, task=ydf.Task.REGRESSION)
+        d = {f"f{i}": rng.randn(500).astype(np.float32) for i in range(F)}
+        p1 = m.predict(d, device="cpu")
+        assert np.isfinite(p1).all()
+        with tempfile.TemporaryDirectory() as td:
+            m.save(td)
+            np.testing.assert_array_equal(
+                p1, ydf.load_model(td).predict(d, device="cpu"))
+        m3 = ydf.deserialize_model(ydf.serialize_model(m))
+        np.testing.assert_array_equal(p1, m3.predict(d, device="cpu"))
+        # extraction round-trips the random structure too
+        back = [ydf.extract_tree(m.forest, t) for t in range(7)]
+        m4 = ydf.build_model_from_trees(back, spec,
+                                        task=ydf.Task.REGRESSION)
+        np.testing.assert_array_equal(p1, m4.predict(d, device="cpu"))
