@@ -120,6 +120,24 @@ def test_rf_distributed_bit_exact(tmp_path):
     np.testing.assert_allclose(got["thr"], m.forest.thr, rtol=1e-6)
 
 
+def test_rf_distributed_dense_compacted_allreduce(tmp_path):
+    """Dense levels under histogram subtraction all-reduce only the
+    BUILT slots (derived siblings reduce locally): the 2-rank forest
+    must still equal single-process training bit for bit."""
+    out = str(tmp_path / "rfd.pkl")
+    _spawn(_worker_rf_dense, out)
+    with open(out, "rb") as f:
+        got = pickle.load(f)
+    os.environ["YDFA_DENSE_LIMIT"] = "8"
+    try:
+        m = ydf.RandomForestLearner(**_rf_args()).train(_make_data())
+    finally:
+        os.environ.pop("YDFA_DENSE_LIMIT", None)
+    np.testing.assert_array_equal(got["feat"], m.forest.feat)
+    np.testing.assert_array_equal(got["left"], m.forest.left)
+    np.testing.assert_allclose(got["thr"], m.forest.thr, rtol=1e-6)
+
+
 def test_gbt_distributed_quality(tmp_path):
     out = str(tmp_path / "gbt.pkl")
     _spawn(_worker_gbt, out)
@@ -127,6 +145,30 @@ def test_gbt_distributed_quality(tmp_path):
         got = pickle.load(f)
     assert got["auc"] > 0.97, got
     assert got["n_trees"] >= 10
+
+
+def _worker_rf_dense(rank, world, port, out_path):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["YDFA_DENSE_LIMIT"] = "8"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from ydf_amd.parallel.dist import shard_rows
+
+        data = _make_data()
+        ds = ydf.create_vertical_dataset(data, label="label",
+                                         task=ydf.Task.CLASSIFICATION)
+        lo, hi = shard_rows(ds.n_examples, rank, world)
+        m = ydf.RandomForestLearner(**_rf_args()).train(ds.shard(lo, hi))
+        if rank == 0:
+            with open(out_path, "wb") as f:
+                pickle.dump({"feat": m.forest.feat, "thr": m.forest.thr,
+                             "left": m.forest.left}, f)
+    finally:
+        os.environ.pop("YDFA_DENSE_LIMIT", None)
+        dist.destroy_process_group()
 
 
 def _worker_oblique(rank, world, port, out_path):

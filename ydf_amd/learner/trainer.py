@@ -294,6 +294,9 @@ class ForestTrainer:
                 1 << int(math.floor(math.log2(self.max_slots))), 64)
         else:
             self.dense_limit = 1
+        if os.environ.get("YDFA_DENSE_LIMIT"):
+            self.dense_limit = min(int(os.environ["YDFA_DENSE_LIMIT"]),
+                                   self.max_slots)
         self._i16_ok = (self.device.type == "cuda" and self.P == 0
                         and self.F >= 32
                         and os.environ.get("YDFA_HIST_I16", "1") == "1")
@@ -1028,7 +1031,18 @@ class ForestTrainer:
                            level_size,
                            filtered_hint=use_sub and os.environ.get(
                                "YDFA_HIST_FILTER_SUB", "0") == "1")
-        self._allreduce(hist_view)
+        if self.distributed and use_sub and derived is not None \
+                and level > 0:
+            # derived slots are still all-zero here: all-reduce only the
+            # BUILT slots (halves the xGMI payload at depth >= 1), then
+            # derive siblings locally from the reduced histograms
+            built = torch.nonzero(build_map >= 0).view(-1)
+            if built.numel() > 0:
+                compact = hist_view.index_select(0, built).contiguous()
+                self._allreduce(compact)
+                hist_view.index_copy_(0, built, compact)
+        else:
+            self._allreduce(hist_view)
         if use_sub and derived is not None:
             ops.subtract_hist(hist_view, self.hist_prev, derived, level_size)
         ops.split_scan(hist_view, abs_t, self.node_stats, self.bg_nf,
