@@ -397,6 +397,10 @@ class GradientBoostedTreesLearner(GenericLearner):
                 label_classes=classes, init_predictions=init_preds,
                 num_trees_per_iter=C, activation=activation,
                 metadata={"feature_gains": gains,
+                          # custom python losses have no reference Loss
+                          # enum value; export refuses to mislabel them
+                          "loss": (int(loss) if custom_loss is None
+                                   else "custom"),
                           "missing_value_policy": getattr(
                               self, "missing_value_policy",
                               "GLOBAL_IMPUTATION"),

@@ -1337,6 +1337,11 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 # weighted loss: g,h scale linearly with the example weight
                 # (reference dataset/weight.h GetWeights path)
                 trainer.gh.mul_(trainer.weights.view(-1, 1))
+                # packed-u64 kernel invariant: per-example h <= 16. The
+                # pre-weight clamp bounds h at 16 and weights at 8, so
+                # h*w can reach 128 (e.g. weighted Poisson) and would
+                # corrupt the 44-bit fixed-point h field — re-clamp.
+                trainer.gh[:, 1].clamp_(max=16.0)
             if cfg.sampling_method == "SELGB" and ranking is not None:
                 # Selective Gradient Boosting (Lucchese et al. 2018;
                 # reference selective_gradient_boosting.h): keep every

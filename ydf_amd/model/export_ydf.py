@@ -82,8 +82,15 @@ def encode_data_spec(dataspec) -> bytes:
                 items += f_msg(7, entry)
             cat = f_varint(2, len(c.vocab)) + items
             body += f_msg(6, cat)
+        elif c.semantic == Semantic.BOOLEAN:
+            # BooleanSpec (data_spec.proto): count_true=1, count_false=2.
+            # We only track the mean; preserve the imputation decision
+            # (most-frequent value) with indicator counts.
+            t = 1 if float(c.mean) >= 0.5 else 0
+            body += f_msg(9, f_varint(1, t) + f_varint(2, 1 - t))
         else:
-            num = f_float(1, c.mean)
+            # NumericalSpec: mean=1 is a DOUBLE in the reference
+            num = f_double(1, c.mean)
             body += f_msg(5, num)
         cols += f_msg(1, body)
     return cols
@@ -91,8 +98,8 @@ def encode_data_spec(dataspec) -> bytes:
 
 # --- nodes blob sequence ---------------------------------------------------
 def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
-                      is_bool: bool, oblique=None, cover: float = 0.0
-                      ) -> bytes:
+                      is_bool: bool, oblique=None, cover: float = 0.0,
+                      na_right: bool = False) -> bytes:
     if oblique is not None:
         # Oblique (decision_tree.proto:114-131): attributes=1 packed,
         # weights=2 packed f32, threshold=3; semantics sum >= threshold
@@ -112,7 +119,11 @@ def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
         # x > thr  <=>  x >= nextafter(thr, +inf)
         t = float(np.nextafter(np.float32(thr), np.float32("inf")))
         inner = f_msg(2, f_float(1, t))             # Higher
-    cond = f_varint(2, feat) + f_msg(3, inner)      # NodeCondition
+    # NodeCondition (decision_tree.proto:179-199): na_value=1 (NA routes
+    # to the positive child iff set — our na_right bit), attribute=2,
+    # condition=3, counts=4/5.
+    cond = f_varint(1, 1) if na_right else b""
+    cond += f_varint(2, feat) + f_msg(3, inner)
     cond += f_varint(4, max(0, int(cover)))         # n examples (unweighted)
     cond += f_double(5, float(cover))               # n examples (weighted)
     return f_msg(3, cond)                           # Node.condition
@@ -158,7 +169,8 @@ def encode_forest_nodes(model, classifier_leaves: bool = False
                 fi, float(f.thr[n]),
                 f.masks[ci] if ci >= 0 else None,
                 ci >= 0, fi in bool_feats and ci == -1, oblique=obl,
-                cover=float(f.cover[n]))
+                cover=float(f.cover[n]),
+                na_right=bool(f.na_right[n]))
         records.append(body)
         if f.feat[n] >= 0:
             left = int(f.left[n])
@@ -181,7 +193,33 @@ def write_blob_sequence(path: str, records: List[bytes]) -> None:
 # --- model export ----------------------------------------------------------
 _TASK = {Task.CLASSIFICATION: 1, Task.REGRESSION: 2, Task.RANKING: 3,
          Task.ANOMALY_DETECTION: 6}
-_LOSS = {"sigmoid": 1, "softmax": 3, "identity": 2}
+# Internal trainer loss id -> reference proto::Loss enum value
+# (model/gradient_boosted_trees/gradient_boosted_trees.proto:54-82).
+# Internal ids 1,2,3,7,8,9 already match the reference; 11/12/13 are
+# focal/xe-ndcg/cox which the reference numbers 6/5/10.
+_LOSS_INTERNAL_TO_REF = {1: 1, 2: 2, 3: 3, 7: 7, 8: 8, 9: 9,
+                         11: 6, 12: 5, 13: 10}
+# Fallback when the model carries no loss id (e.g. hand-built trees):
+# infer from the activation/link function.
+_LOSS_FROM_ACTIVATION = {"sigmoid": 1, "softmax": 3, "identity": 2,
+                         "exp": 7}
+
+
+def _ref_loss_enum(model) -> int:
+    loss_id = (model.metadata or {}).get("loss")
+    if loss_id is not None:
+        ref = _LOSS_INTERNAL_TO_REF.get(loss_id) \
+            if isinstance(loss_id, int) else None
+        if ref is None:
+            raise NotImplementedError(
+                f"GBT loss id {loss_id} has no reference Loss enum value; "
+                "refusing to export with a mislabeled loss")
+        return ref
+    act = model.activation
+    if act not in _LOSS_FROM_ACTIVATION:
+        raise NotImplementedError(
+            f"cannot infer reference Loss enum for activation {act!r}")
+    return _LOSS_FROM_ACTIVATION[act]
 
 
 def export_ydf_model(model, path: str) -> None:
@@ -214,10 +252,14 @@ def export_ydf_model(model, path: str) -> None:
     with open(os.path.join(path, "data_spec.pb"), "wb") as fp:
         fp.write(encode_data_spec(model.dataspec))
     if is_gbt:
-        # GBT header: num_trees=2, loss=3, initial_predictions=4,
-        # num_trees_per_iter=5, node_format=7
-        gh = f_varint(2, model.forest.n_trees)
-        gh += f_varint(3, _LOSS.get(model.activation, 2))
+        # GBT header (gradient_boosted_trees.proto:28-42):
+        # num_node_shards=1, num_trees=2, loss=3, initial_predictions=4,
+        # num_trees_per_iter=5, node_format=7. num_node_shards is
+        # REQUIRED by the reference reader (LoadTreesFromDisk uses it to
+        # build the sharded nodes file spec).
+        gh = f_varint(1, 1)
+        gh += f_varint(2, model.forest.n_trees)
+        gh += f_varint(3, _ref_loss_enum(model))
         for v in model.init_predictions:
             gh += f_float(4, v)
         gh += f_varint(5, model.num_trees_per_iter)
@@ -227,12 +269,15 @@ def export_ydf_model(model, path: str) -> None:
             fp.write(gh)
         classifier_leaves = False
     else:
-        # RF header (random_forest.proto): num_trees=2,
-        # winner_take_all_inference=3, node_format=6
+        # RF header (random_forest.proto:28-41): num_node_shards=1,
+        # num_trees=2, winner_take_all_inference=3, node_format=7
+        # (field 6 is mean_increase_in_rmse — a repeated message, NOT
+        # node_format).
         wta = bool((model.metadata or {}).get("winner_take_all", False))
-        rh = f_varint(2, model.forest.n_trees)
+        rh = f_varint(1, 1)
+        rh += f_varint(2, model.forest.n_trees)
         rh += f_varint(3, 1 if wta else 0)
-        rh += f_str(6, "BLOB_SEQUENCE")
+        rh += f_str(7, "BLOB_SEQUENCE")
         with open(os.path.join(path, "random_forest_header.pb"),
                   "wb") as fp:
             fp.write(rh)

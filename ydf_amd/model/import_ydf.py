@@ -454,13 +454,23 @@ def load_ydf_model(path: str, file_prefix: str = ""):
                 init_preds.append(_f32(v))
         if not init_preds:
             init_preds = [0.0]
-        loss = gh.get(3, [0])[0]
+        loss_ref = gh.get(3, [0])[0]
+        # Reference proto::Loss enum -> our internal trainer loss ids
+        # (they agree on 1,2,3,7,8,9; focal=6->11, xe-ndcg=5->12,
+        # cox=10->13, deprecated ndcg5=4->9).
+        loss = {1: 1, 2: 2, 3: 3, 4: 9, 5: 12, 6: 11, 7: 7, 8: 8,
+                9: 9, 10: 13}.get(int(loss_ref), int(loss_ref))
         forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
                              set_feats=set_feats)
         ntpi = gh.get(5, [1])[0]
         activation = "identity"
         if task == Task.CLASSIFICATION:
             activation = "softmax" if ntpi > 1 else "sigmoid"
+        elif loss == 7:  # POISSON: reference applies exp at predict
+            activation = "exp"
+        if loss == 0:  # DEFAULT: infer like the reference does
+            loss = (3 if ntpi > 1 else 1) \
+                if task == Task.CLASSIFICATION else 2
         model = GradientBoostedTreesModel(
             forest=_remap_forest(forest, remap), dataspec=dataspec,
             task=task, label_classes=classes, init_predictions=init_preds,
