@@ -143,7 +143,7 @@ def test_to_docker(tmp_path):
     out = tmp_path / "serve"
     ydf.to_docker(m, str(out))
     for fn in ("Dockerfile", "main.py", "requirements.txt",
-               "model/header.json"):
+               "model/header.pb"):
         assert (out / fn).exists()
     sys.path.insert(0, str(out))
     try:

@@ -12,8 +12,14 @@ def main():
     ap.add_argument("--dataspec", help="dataspec.json path")
     args = ap.parse_args()
     path = args.dataspec or os.path.join(args.model, "dataspec.json")
-    with open(path) as f:
-        spec = DataSpecification.from_json(json.load(f))
+    if os.path.exists(path):
+        with open(path) as f:
+            spec = DataSpecification.from_json(json.load(f))
+    else:
+        # reference-format model dir (the default save() layout)
+        from ydf_amd.model.model_lib import load_model
+
+        spec = load_model(args.model).dataspec
     print(f"{len(spec.columns)} column(s); label: {spec.label!r}")
     for c in spec.columns:
         extra = ""

@@ -446,7 +446,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                 tmp = snap_dir + ".tmp"
                 if _os.path.exists(tmp):
                     shutil.rmtree(tmp)
-                ms.save(tmp)
+                ms.save(tmp, format="npz")  # snapshots: fast container
                 if _os.path.exists(snap_dir):
                     shutil.rmtree(snap_dir)
                 _os.replace(tmp, snap_dir)

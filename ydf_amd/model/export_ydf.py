@@ -99,8 +99,14 @@ def encode_data_spec(dataspec) -> bytes:
 # --- nodes blob sequence ---------------------------------------------------
 def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
                       is_bool: bool, oblique=None, cover: float = 0.0,
-                      na_right: bool = False) -> bytes:
-    if oblique is not None:
+                      na_right: bool = False, set_items=None) -> bytes:
+    if set_items is not None:
+        # ContainsVector (decision_tree.proto Condition.contains=4):
+        # elements=1 packed varints — categorical-SET conditions whose
+        # vocab can exceed the 256-bit bitmap
+        packed = b"".join(_varint(int(e)) for e in set_items)
+        inner = f_msg(4, f_bytes(1, packed))
+    elif oblique is not None:
         # Oblique (decision_tree.proto:114-131): attributes=1 packed,
         # weights=2 packed f32, threshold=3; semantics sum >= threshold
         attrs, ws = oblique
@@ -129,23 +135,53 @@ def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
     return f_msg(3, cond)                           # Node.condition
 
 
-def encode_forest_nodes(model, classifier_leaves: bool = False
-                        ) -> List[bytes]:
+def _invert_path_length(x: float, n_max: int) -> int:
+    """Recovers the leaf example count n from c(n) = x (the importer
+    stores isolation leaves as depth + c(count); see
+    specialized.IsolationForestModel.expected_path_length)."""
+    from ydf_amd.model.specialized import IsolationForestModel
+    c = IsolationForestModel.expected_path_length
+    best, best_err = 1, abs(c(1) - x)
+    lo, hi = 1, max(4, n_max)
+    while lo <= hi:             # c is monotonic: binary search
+        mid = (lo + hi) // 2
+        err = c(mid) - x
+        if abs(err) < best_err:
+            best, best_err = mid, abs(err)
+        if err < 0:
+            lo = mid + 1
+        else:
+            hi = mid - 1
+    return best
+
+
+def encode_forest_nodes(model, classifier_leaves: bool = False,
+                        anomaly_leaves: bool = False,
+                        n_per_tree: int = 256,
+                        feat_to_col=None) -> List[bytes]:
     """Pre-order Node records per tree (negative child first). With
     classifier_leaves (RF classification), leaves carry a class
-    distribution whose counts reproduce our leaf probability."""
+    distribution whose counts reproduce our leaf probability; with
+    anomaly_leaves (isolation forest), leaves carry the example count
+    recovered from the stored path-length contribution."""
     f = model.forest
     bool_feats = set()
     for i, c in enumerate(model.dataspec.feature_columns):
         if c.semantic == Semantic.BOOLEAN:
             bool_feats.add(i)
+    if feat_to_col is None:
+        feat_to_col = list(range(len(model.dataspec.columns)))
 
     records: List[bytes] = []
 
-    def emit(n: int):
+    def emit(n: int, depth: int = 0):
         body = b""
         if f.feat[n] < 0:
-            if classifier_leaves:
+            if anomaly_leaves:
+                cnt = _invert_path_length(float(f.thr[n]) - depth,
+                                          n_per_tree)
+                body += f_msg(6, f_varint(1, cnt))  # Node.anomaly_detection
+            elif classifier_leaves:
                 # counts over [OOV, class1, class2]: p(class2) = thr
                 p = float(np.clip(f.thr[n], 0.0, 1.0))
                 total = float(f.cover[n]) if f.cover[n] > 0 else 1.0
@@ -160,22 +196,28 @@ def encode_forest_nodes(model, classifier_leaves: bool = False
             body += f_msg(2, f_float(1, 0.0))
             fi = int(f.feat[n])
             ci = int(f.cat_idx[n])
+            si = int(f.set_idx[n]) if f.set_idx is not None else -1
             obl = None
-            if ci <= -2:
+            set_items = None
+            if si >= 0:
+                s0, s1 = int(f.set_offs[si]), int(f.set_offs[si + 1])
+                set_items = f.set_items[s0:s1]
+            elif ci <= -2:
                 oi = -(ci + 2)
                 s0, nn = int(f.obl_ranges[oi, 0]), int(f.obl_ranges[oi, 1])
-                obl = (f.obl_attr[s0:s0 + nn], f.obl_w[s0:s0 + nn])
+                obl = ([feat_to_col[a] for a in f.obl_attr[s0:s0 + nn]],
+                       f.obl_w[s0:s0 + nn])
             body += _encode_condition(
-                fi, float(f.thr[n]),
+                feat_to_col[fi], float(f.thr[n]),
                 f.masks[ci] if ci >= 0 else None,
-                ci >= 0, fi in bool_feats and ci == -1, oblique=obl,
-                cover=float(f.cover[n]),
-                na_right=bool(f.na_right[n]))
+                ci >= 0, fi in bool_feats and ci == -1 and si < 0,
+                oblique=obl, cover=float(f.cover[n]),
+                na_right=bool(f.na_right[n]), set_items=set_items)
         records.append(body)
         if f.feat[n] >= 0:
             left = int(f.left[n])
-            emit(left)       # negative child first
-            emit(left + 1)
+            emit(left, depth + 1)       # negative child first
+            emit(left + 1, depth + 1)
 
     for t in range(f.n_trees):
         emit(int(f.roots[t]))
@@ -226,27 +268,57 @@ def export_ydf_model(model, path: str) -> None:
     """Writes `model` as a reference-format model directory (GBT and
     RF — binary classification and regression)."""
     from ydf_amd.model.specialized import (GradientBoostedTreesModel,
+                                           IsolationForestModel,
                                            RandomForestModel)
 
     is_gbt = isinstance(model, GradientBoostedTreesModel)
-    is_rf = isinstance(model, RandomForestModel) and not is_gbt
-    if not (is_gbt or is_rf):
+    is_if = isinstance(model, IsolationForestModel)
+    is_rf = isinstance(model, RandomForestModel) and not (is_gbt or is_if)
+    if not (is_gbt or is_rf or is_if):
         raise NotImplementedError(
-            "export to the reference format supports GBT and RF models")
+            "export to the reference format supports GBT, RF and "
+            "IsolationForest models")
     if is_rf and model._n_outputs() > 1:
         raise NotImplementedError(
             "RF export supports binary classification / regression "
             "(our multi-class RF stores one tree per class, which the "
             "reference single-tree-distribution format cannot express)")
+    if model.task() not in _TASK:
+        raise NotImplementedError(
+            f"task {model.task()} has no reference-format export")
+    if any(getattr(c, "set_source", None)
+           for c in model.dataspec.columns):
+        raise NotImplementedError(
+            "models trained on expanded categorical-set token features "
+            "use a virtual-column representation the reference data "
+            "spec cannot express")
     os.makedirs(path, exist_ok=True)
-    label_idx = len(model.dataspec.columns) - 1
-    # AbstractModel: name=1, task=2, label_col_idx=3, input_features=5
-    header = f_str(1, "GRADIENT_BOOSTED_TREES" if is_gbt
-                   else "RANDOM_FOREST")
-    header += f_varint(2, _TASK.get(model.task(), 1))
+    # Column indexing: node conditions reference COLUMN indices in the
+    # data spec, not dense feature indices (the label can sit anywhere —
+    # ranking models put it first). input_features lists the feature
+    # columns in the model's dense feature order; anomaly-detection
+    # models have no label column (label_col_idx=-1).
+    cols = model.dataspec.columns
+    label_name = model.dataspec.label
+    has_label = label_name is not None
+    label_idx = next((i for i, c in enumerate(cols)
+                      if c.name == label_name), -1) if has_label else -1
+    feat_to_col = [i for i, c in enumerate(cols) if c.name != label_name]
+    # AbstractModel: name=1, task=2, label_col_idx=3, input_features=5,
+    # ranking_group_col_idx=6
+    name = ("GRADIENT_BOOSTED_TREES" if is_gbt
+            else "ISOLATION_FOREST" if is_if else "RANDOM_FOREST")
+    header = f_str(1, name)
+    header += f_varint(2, _TASK[model.task()])
     header += f_varint(3, label_idx)
-    for i in range(label_idx):
+    for i in feat_to_col:
         header += f_varint(5, i)
+    rg = (model.metadata or {}).get("ranking_group")
+    if rg:
+        for i, c in enumerate(model.dataspec.columns):
+            if c.name == rg:
+                header += f_varint(6, i)
+                break
     with open(os.path.join(path, "header.pb"), "wb") as fp:
         fp.write(header)
     with open(os.path.join(path, "data_spec.pb"), "wb") as fp:
@@ -268,7 +340,7 @@ def export_ydf_model(model, path: str) -> None:
                   "wb") as fp:
             fp.write(gh)
         classifier_leaves = False
-    else:
+    elif is_rf:
         # RF header (random_forest.proto:28-41): num_node_shards=1,
         # num_trees=2, winner_take_all_inference=3, node_format=7
         # (field 6 is mean_increase_in_rmse — a repeated message, NOT
@@ -282,8 +354,25 @@ def export_ydf_model(model, path: str) -> None:
                   "wb") as fp:
             fp.write(rh)
         classifier_leaves = model.task() == Task.CLASSIFICATION
+    else:
+        classifier_leaves = False
+    if is_if:
+        # IF header (isolation_forest.proto): num_node_shards=1,
+        # num_trees=2, node_format=3 (NOTE: 3 here, not 7),
+        # num_examples_per_trees=4
+        ih = f_varint(1, 1)
+        ih += f_varint(2, model.forest.n_trees)
+        ih += f_str(3, "BLOB_SEQUENCE")
+        ih += f_varint(4, model.num_examples_per_tree)
+        with open(os.path.join(path, "isolation_forest_header.pb"),
+                  "wb") as fp:
+            fp.write(ih)
     write_blob_sequence(
         os.path.join(path, "nodes-00000-of-00001"),
-        encode_forest_nodes(model, classifier_leaves=classifier_leaves))
+        encode_forest_nodes(
+            model, classifier_leaves=classifier_leaves,
+            anomaly_leaves=is_if,
+            n_per_tree=getattr(model, "num_examples_per_tree", 256),
+            feat_to_col=feat_to_col))
     with open(os.path.join(path, "done"), "w") as fp:
         fp.write("")

@@ -690,16 +690,51 @@ class GenericModel:
             "num_trees_per_iter": self.num_trees_per_iter,
             "activation": self.activation,
             "metadata": self.metadata,
+            "training_logs": self.training_logs,
         }
 
     def _load_extra(self, header: dict) -> None:
         pass
 
-    def save(self, path: str, advanced_options: ModelIOOptions = None) -> None:
-        """Model directory: header.json + dataspec.json + forest.npz + done
-        (structural analogue of the reference model dir,
-        model/model_library.cc:92-107: header.pb / data_spec.pb / nodes-* /
-        done)."""
+    def save(self, path: str, advanced_options: ModelIOOptions = None,
+             format: str = "ydf") -> None:
+        """Saves the model directory.
+
+        Default format="ydf": the REFERENCE on-disk layout
+        (model/model_library.cc:92-107: header.pb / data_spec.pb /
+        <family>_header.pb / nodes-00000-of-00001 / done) readable by
+        the reference C++/Go/JS/TF-DF loaders, plus `extra.json`
+        carrying state the reference format has no field for (training
+        logs, framework metadata, exact activation) which the reference
+        reader ignores. Models the layout cannot express (multi-class
+        RF, uplift, survival, custom losses) fall back to the npz
+        container with a log line. format="npz" forces the fast npz
+        container (used for training snapshots)."""
+        if format == "ydf":
+            import shutil
+            import tempfile
+
+            try:
+                from ydf_amd.model.export_ydf import export_ydf_model
+
+                os.makedirs(path, exist_ok=True)
+                tmp = tempfile.mkdtemp(
+                    dir=os.path.dirname(os.path.abspath(path)) or ".")
+                try:
+                    export_ydf_model(self, tmp)
+                    with open(os.path.join(tmp, "extra.json"), "w") as f:
+                        json.dump(self._header(), f, indent=1)
+                    for name in os.listdir(tmp):
+                        os.replace(os.path.join(tmp, name),
+                                   os.path.join(path, name))
+                finally:
+                    shutil.rmtree(tmp, ignore_errors=True)
+                return
+            except NotImplementedError as e:
+                from ydf_amd.utils.log import info
+
+                info(f"reference-format save not available for this "
+                     f"model ({e}); writing npz container")
         os.makedirs(path, exist_ok=True)
         with open(os.path.join(path, "header.json"), "w") as f:
             json.dump(self._header(), f, indent=1)

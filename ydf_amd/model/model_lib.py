@@ -15,6 +15,30 @@ from ydf_amd.model.generic_model import GenericModel
 def load_model(path: str) -> GenericModel:
     from ydf_amd.model.specialized import MODEL_CLASSES
 
+    if not os.path.exists(os.path.join(path, "header.json")):
+        if os.path.exists(os.path.join(path, "header.pb")):
+            # reference-format directory (the default save() layout):
+            # forest/dataspec from the wire files, framework-only state
+            # from extra.json (absent when the dir came from the
+            # reference itself)
+            from ydf_amd.model.import_ydf import load_ydf_model
+
+            model = load_ydf_model(path)
+            ej = os.path.join(path, "extra.json")
+            if os.path.exists(ej):
+                with open(ej) as f:
+                    extra = json.load(f)
+                model.activation = extra.get("activation",
+                                             model.activation)
+                meta = dict(extra.get("metadata") or {})
+                meta.setdefault("loss", (model.metadata or {}).get("loss"))
+                model.metadata = meta
+                model.training_logs = extra.get("training_logs")
+                model._load_extra(extra)
+            return model
+        raise FileNotFoundError(
+            f"{path}: neither header.json (npz container) nor header.pb "
+            f"(reference model directory) found")
     with open(os.path.join(path, "header.json")) as f:
         header = json.load(f)
     if header.get("model_type") in ("MLP", "TABULAR_TRANSFORMER"):
@@ -51,6 +75,7 @@ def load_model(path: str) -> GenericModel:
         activation=header.get("activation", "identity"),
         metadata=header.get("metadata"),
     )
+    model.training_logs = header.get("training_logs")
     model._load_extra(header)
     return model
 
