@@ -60,8 +60,8 @@ def device_boundaries(X: torch.Tensor, n_cuts: int = 255):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=40)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=300)
+    ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--rows", type=int, default=N_ROWS,
                     help="total rows (testing only; default = Higgs 11M)")
     ap.add_argument("--device", default=None)
@@ -110,8 +110,12 @@ def main():
     # sequence is device-resident, so one tree == one graph replay plus the
     # host copy of the finished tree). Falls back to eager launches.
     graph = None
-    if device.type == "cuda" and world == 1 \
+    import torch.distributed as _td
+    if device.type == "cuda" and not _td.is_initialized() \
             and os.environ.get("YDFA_BENCH_GRAPH", "1") == "1":
+        # graph replay only without a process group: collectives inside
+        # a hipGraph capture are not exercised here (a world-size-1
+        # torchrun rehearsal must run the REAL eager RCCL path)
         try:
             ops.grad_hess(preds, labels, tr.gh, cfg.loss)  # warm allocs
             tr.grow_tree(0)
@@ -166,7 +170,7 @@ def main():
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
             "scaling": "strong",
-            "vs_baseline": None,
+            "vs_baseline": value / 281.03,  # round-1 driver-measured value (BENCH_r01.json)
             "dtype": "fp32",
             "data": "synthetic",
             "config": {

@@ -211,3 +211,152 @@ def test_oblique_distributed_rank_consistent(tmp_path):
     np.testing.assert_array_equal(a["obl_attr"], b["obl_attr"])
     np.testing.assert_array_equal(a["obl_w"], b["obl_w"])
     assert len(a["obl_attr"]) > 0
+
+
+# ---------------------------------------------------------------------------
+# Fault injection + recovery (reference simulate_worker_failure,
+# distributed_gradient_boosted_trees_test.cc:148-181 + worker.h:121
+# MaybeSimulateFailure: a worker dies mid-training, the job restarts and
+# must resume from the last checkpoint and converge)
+# ---------------------------------------------------------------------------
+def _worker_gbt_snap(rank, world, port, out_path, workdir, fault_iter):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    if fault_iter >= 0:
+        os.environ["YDFA_FAULT_ITER"] = str(fault_iter)
+        os.environ["YDFA_FAULT_RANK"] = "1"
+    else:
+        os.environ.pop("YDFA_FAULT_ITER", None)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from ydf_amd.parallel.dist import shard_rows
+
+        data = _make_data()
+        ds = ydf.create_vertical_dataset(data, label="label",
+                                         task=ydf.Task.CLASSIFICATION)
+        lo, hi = shard_rows(ds.n_examples, rank, world)
+        m = ydf.GradientBoostedTreesLearner(
+            label="label", num_trees=24, validation_ratio=0.0,
+            early_stopping="NONE",
+            working_dir=workdir, resume_training=True,
+            resume_training_snapshot_interval_seconds=0.0,
+            device="cpu").train(ds.shard(lo, hi))
+        if rank == 0:
+            with open(out_path, "wb") as f:
+                pickle.dump({"preds": m.predict(data, device="cpu"),
+                             "n_trees": m.num_trees()}, f)
+    finally:
+        dist.destroy_process_group()
+
+
+def _spawn_may_fail(target, args, world=2):
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=target, args=(r, world, port) + args)
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    return [p.exitcode for p in procs]
+
+
+def test_distributed_fault_recovery(tmp_path):
+    """Rank 1 dies at iteration 8 (snapshots every iteration); the
+    restarted job must resume from the snapshot and produce the same
+    forest as an uninterrupted run."""
+    wd_f = str(tmp_path / "wd_fault")
+    out = str(tmp_path / "snap.pkl")
+
+    codes = _spawn_may_fail(_worker_gbt_snap,
+                            (out, wd_f, 8))
+    assert any(c != 0 for c in codes), "injected fault did not fire"
+    snap = os.path.join(wd_f, "snapshot")
+    assert os.path.exists(os.path.join(snap, "done")), \
+        "no snapshot written before the fault"
+
+    # restart (no fault): resumes from snapshot
+    codes = _spawn_may_fail(_worker_gbt_snap, (out, wd_f, -1))
+    assert codes == [0, 0], codes
+    with open(out, "rb") as f:
+        resumed = pickle.load(f)
+    assert resumed["n_trees"] == 24
+
+    # uninterrupted reference run in a fresh working dir
+    wd_c = str(tmp_path / "wd_clean")
+    out2 = str(tmp_path / "clean.pkl")
+    codes = _spawn_may_fail(_worker_gbt_snap, (out2, wd_c, -1))
+    assert codes == [0, 0], codes
+    with open(out2, "rb") as f:
+        clean = pickle.load(f)
+    assert clean["n_trees"] == 24
+    # margins are re-summed on resume -> ulp differences may flip
+    # near-tie splits for a handful of examples (same tolerance as the
+    # single-process exact-resume test)
+    p2, p3 = resumed["preds"], clean["preds"]
+    assert np.mean(np.abs(p2 - p3) < 1e-4) > 0.99
+    assert np.abs(p2 - p3).max() < 0.2
+
+
+# ---------------------------------------------------------------------------
+# torchrun rehearsals: the exact launch path the driver uses for the
+# 8-GPU scaling bench, exercised end-to-end (world-size-1 initializes a
+# REAL process group; on a GPU box that is a real RCCL communicator +
+# ncclAllReduce calls)
+# ---------------------------------------------------------------------------
+def _run_torchrun_bench(nproc, extra_env=None):
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update(extra_env or {})
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--standalone",
+           "--local-addr", "127.0.0.1", f"--nproc-per-node={nproc}",
+           os.path.join(repo, "bench.py"), "--gpus", str(nproc),
+           "--rows", "200000", "--steps", "4", "--warmup", "1"]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=repo, env=env)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    import json as _json
+
+    line = [ln for ln in r.stdout.splitlines()
+            if ln.startswith("{")][-1]
+    return _json.loads(line)
+
+
+def test_torchrun_world1_cpu():
+    out = _run_torchrun_bench(1, {"YDFA_DIST_BACKEND": "gloo"})
+    assert out["n_gpus"] == 1
+    assert out["value"] > 0
+
+
+@pytest.mark.gpu
+def test_torchrun_world1_rccl():
+    """Single-rank torchrun on a real GPU: full RCCL init + world-size-1
+    ncclAllReduce path — the rehearsal that de-risks the driver's 8-GPU
+    launch (VERDICT round-1 item #1)."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    out = _run_torchrun_bench(1)
+    assert out["n_gpus"] == 1
+    assert out["value"] > 0
+
+
+@pytest.mark.gpu
+def test_torchrun_oversubscribed_gloo_fallback():
+    """2 ranks on 1 GPU: RCCL would refuse (Duplicate GPU); the gloo
+    fallback must keep the rehearsal runnable end-to-end."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    if torch.cuda.device_count() >= 2:
+        pytest.skip("only meaningful when ranks exceed GPUs")
+    out = _run_torchrun_bench(2)
+    assert out["n_gpus"] == 2
+    assert out["value"] > 0

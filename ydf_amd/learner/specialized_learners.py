@@ -434,6 +434,13 @@ class GradientBoostedTreesLearner(GenericLearner):
                 import shutil
                 import tempfile
 
+                from ydf_amd.parallel.dist import is_main
+
+                if not is_main():
+                    # data-parallel ranks grow identical trees (seeded
+                    # masks + summed histograms): rank 0's snapshot is
+                    # the job's snapshot; others must not race on the dir
+                    return
                 flat_s = build_flat_forest(trees_so_far, bnd,
                                            leaf_scale=hp["shrinkage"],
                                            cat_feats=cat_feats)

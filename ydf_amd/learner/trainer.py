@@ -159,9 +159,12 @@ class BestFirstTree:
 
 
 def _dist_ok() -> bool:
+    # any initialized process group counts — a world-size-1 group (e.g.
+    # torchrun --nproc-per-node=1) runs REAL collectives, which is the
+    # rehearsal that catches RCCL init/dtype/shape bugs before the first
+    # 8-GPU launch
     return torch.distributed.is_available() and \
-        torch.distributed.is_initialized() and \
-        torch.distributed.get_world_size() > 1
+        torch.distributed.is_initialized()
 
 
 class ForestTrainer:
@@ -349,7 +352,17 @@ class ForestTrainer:
     # -- helpers ----------------------------------------------------------
     def _allreduce(self, t: torch.Tensor):
         if self.distributed:
-            torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.SUM)
+            if t.is_cuda and \
+                    torch.distributed.get_backend() == "gloo":
+                # oversubscribed rehearsal (more ranks than GPUs): gloo
+                # cannot reduce CUDA tensors — round-trip through host
+                h = t.cpu()
+                torch.distributed.all_reduce(
+                    h, op=torch.distributed.ReduceOp.SUM)
+                t.copy_(h)
+            else:
+                torch.distributed.all_reduce(
+                    t, op=torch.distributed.ReduceOp.SUM)
 
     def _feat_mask(self, n_active: int, tree_idx: int,
                    level: int) -> Optional[torch.Tensor]:
@@ -1272,7 +1285,19 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
     t_start = _time.monotonic()
     t_last_snapshot = t_start
     interrupted = False
+    fault_iter = int(os.environ.get("YDFA_FAULT_ITER", "-1"))
+    fault_rank = int(os.environ.get("YDFA_FAULT_RANK", "0"))
     for it in range(start_iteration, n_iters):
+        if it == fault_iter:
+            # fault injection (reference MaybeSimulateFailure,
+            # learner/distributed_gradient_boosted_trees/worker.h:121):
+            # this rank dies mid-training; recovery = restart the job,
+            # which resumes from the last snapshot
+            rk = torch.distributed.get_rank() \
+                if _dist_ok() else 0
+            if rk == fault_rank:
+                raise RuntimeError(
+                    f"injected fault at iteration {it} (YDFA_FAULT_ITER)")
         if max_duration_seconds > 0 and \
                 _time.monotonic() - t_start > max_duration_seconds:
             if log:

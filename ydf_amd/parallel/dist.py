@@ -18,21 +18,32 @@ import torch.distributed as dist
 def init_from_env(backend: str = None) -> int:
     """Initializes torch.distributed from torchrun env vars; returns rank.
 
-    No-op (returns 0) when WORLD_SIZE is absent or 1."""
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    if world <= 1:
+    No-op (returns 0) when not launched by torchrun (RANK absent). A
+    world-size-1 torchrun launch DOES initialize a process group, so the
+    full RCCL init + collective path runs in single-GPU rehearsals."""
+    if "RANK" not in os.environ or "MASTER_ADDR" not in os.environ:
         return 0
+    world = int(os.environ.get("WORLD_SIZE", "1"))
     if not dist.is_initialized():
         if backend is None:
             backend = os.environ.get("YDFA_DIST_BACKEND")
         if backend is None:
             backend = "nccl" if torch.cuda.is_available() else "gloo"
         if backend == "nccl":
-            # modulo: lets N ranks share fewer GPUs (single-GPU rehearsal
-            # of the multi-rank launch path)
-            torch.cuda.set_device(
-                int(os.environ.get("LOCAL_RANK", "0"))
-                % max(torch.cuda.device_count(), 1))
+            n_dev = max(torch.cuda.device_count(), 1)
+            if world > n_dev:
+                # RCCL refuses two ranks on one device ("Duplicate GPU
+                # detected"); fall back to gloo so over-subscribed
+                # rehearsal launches still run end-to-end
+                import sys
+
+                print(f"# ydf_amd: {world} ranks > {n_dev} visible "
+                      f"GPU(s); using gloo backend for this rehearsal",
+                      file=sys.stderr)
+                backend = "gloo"
+            else:
+                torch.cuda.set_device(
+                    int(os.environ.get("LOCAL_RANK", "0")) % n_dev)
         dist.init_process_group(backend=backend,
                                 timeout=datetime.timedelta(seconds=300))
     return dist.get_rank()
