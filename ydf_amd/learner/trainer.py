@@ -1045,10 +1045,12 @@ class ForestTrainer:
                            filtered_hint=use_sub and os.environ.get(
                                "YDFA_HIST_FILTER_SUB", "0") == "1")
         if self.distributed and use_sub and derived is not None \
-                and level > 0:
+                and level > 0 and not getattr(self, "capturing", False):
             # derived slots are still all-zero here: all-reduce only the
             # BUILT slots (halves the xGMI payload at depth >= 1), then
-            # derive siblings locally from the reduced histograms
+            # derive siblings locally from the reduced histograms.
+            # (Skipped under graph capture: torch.nonzero needs a host
+            # sync; the captured graph all-reduces the full view.)
             built = torch.nonzero(build_map >= 0).view(-1)
             if built.numel() > 0:
                 compact = hist_view.index_select(0, built).contiguous()
@@ -1086,18 +1088,31 @@ class ForestTrainer:
                            shrinkage: float):
         """Captures one boosting step (gradients -> dense tree growth ->
         prediction update) as a hipGraph; replay + extract_host_tree()
-        per tree. Requires: CUDA device, single process, all levels dense,
-        no feature sampling / subsampling."""
-        assert self.device.type == "cuda" and not self.distributed
+        per tree. Requires: CUDA device, all levels dense, no feature
+        sampling / subsampling. In data-parallel mode the per-level RCCL
+        all-reduces are captured INTO the graph (the xGMI collective
+        launch overhead — ~40us per eager collective — disappears from
+        the replayed step); capture then uses the full-histogram
+        all-reduce (the compacted variant needs a host sync that cannot
+        be captured)."""
+        assert self.device.type == "cuda"
+        if self.distributed:
+            assert torch.distributed.get_backend() == "nccl", \
+                "graph capture of collectives requires the nccl backend"
         assert (1 << (self.cfg.max_depth - 1)) <= self.dense_limit
         assert self.cfg.num_candidate_features <= 0
         assert self.cfg.oblique_projections == 0
         assert self.cfg.growing_strategy == "LOCAL"
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
-            ops.grad_hess(preds, labels, self.gh, self.cfg.loss)
-            self.grow_tree_device(0, None)
-            ops.update_preds(preds, self.node_ids, self.leaf_vals, shrinkage)
+        self.capturing = True
+        try:
+            with torch.cuda.graph(g):
+                ops.grad_hess(preds, labels, self.gh, self.cfg.loss)
+                self.grow_tree_device(0, None)
+                ops.update_preds(preds, self.node_ids, self.leaf_vals,
+                                 shrinkage)
+        finally:
+            self.capturing = False
         return g
 
     def route_tree(self, bins: torch.Tensor, node_ids: torch.Tensor,
