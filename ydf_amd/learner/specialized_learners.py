@@ -105,6 +105,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  sparse_oblique_weights: str = "BINARY",
                  sparse_oblique_max_num_features: int = -1,
                  loss: str = "DEFAULT",
+                 discretize_numerical_columns: bool = True,
                  working_dir: Optional[str] = None,
                  resume_training: bool = False,
                  resume_training_snapshot_interval_seconds: float = 1800.0,
@@ -117,6 +118,10 @@ class GradientBoostedTreesLearner(GenericLearner):
         self.label_event_observed = label_event_observed
         self.label_entry_age = label_entry_age
         self.ndcg_truncation = ndcg_truncation
+        # reference DEFAULT trains exact numerical splits; the MI355X
+        # hot path always bins (discretize=True, documented deviation).
+        # False selects the CPU exact-split oracle (learner/exact.py).
+        self.discretize_numerical_columns = discretize_numerical_columns
         self.hyperparameters = dict(
             num_trees=num_trees, max_depth=max_depth,
             growing_strategy=growing_strategy, max_num_nodes=max_num_nodes,
@@ -166,10 +171,78 @@ class GradientBoostedTreesLearner(GenericLearner):
         _apply_template(self.hyperparameters, "GBT",
                         hyperparameter_template)
 
+    def _train_exact(self, data) -> GradientBoostedTreesModel:
+        """Exact (non-binned) numerical split training on CPU — the
+        reference's DEFAULT split semantics (splitter_scanner.h:1290,
+        preprocessing.h:106). Serves as the correctness oracle that
+        bounds the quality effect of the 256-bin hot path."""
+        from ydf_amd.learner.exact import (exact_trees_to_forest,
+                                           train_gbt_exact)
+
+        hp = self.hyperparameters
+        if self._task not in (Task.CLASSIFICATION, Task.REGRESSION):
+            raise NotImplementedError(
+                "discretize_numerical_columns=False supports "
+                "classification and regression")
+        if hp.get("split_axis", "AXIS_ALIGNED") != "AXIS_ALIGNED" \
+                or hp.get("forest_extraction") == "DART" \
+                or hp.get("sampling_method") not in (None, "RANDOM") \
+                or hp.get("subsample", 1.0) < 1.0:
+            raise NotImplementedError(
+                "exact-split mode supports axis-aligned MART without "
+                "subsampling")
+        device = torch.device("cpu")
+        ds, bins, labels, bnd, cat_flags, weights, mono = self._prepare(
+            data, device)
+        if labels is None:
+            raise ValueError(f"label column {self.label!r} missing")
+        if weights is not None or mono is not None:
+            raise NotImplementedError(
+                "exact-split mode: weights/monotonic not supported")
+        classes = self._label_classes(ds)
+        if classes and len(classes) != 2:
+            raise NotImplementedError(
+                "exact-split mode supports binary classification")
+        loss_named = hp.get("loss")
+        if isinstance(loss_named, str) and loss_named not in (
+                "DEFAULT", "SQUARED_ERROR", "BINOMIAL_LOG_LIKELIHOOD"):
+            raise NotImplementedError(
+                f"exact-split mode does not support loss {loss_named}")
+        X = np.ascontiguousarray(ds.X)
+        if np.isnan(X).any():
+            raise NotImplementedError(
+                "exact-split mode: impute missing values first "
+                "(GLOBAL_IMPUTATION runs at encode time by default)")
+        y = labels.cpu().numpy().astype(np.float64)
+        loss = 1 if self._task == Task.CLASSIFICATION else 2
+        cf = cat_flags.cpu().numpy().astype(bool) \
+            if cat_flags is not None else None
+        info(f"exact-split training ({hp['num_trees']} trees, no "
+             f"validation/early-stopping in this mode)")
+        trees, init = train_gbt_exact(
+            X, y, cf, loss, hp["num_trees"], hp["shrinkage"],
+            hp["max_depth"], hp["min_examples"],
+            hp["min_sum_hessian_in_leaf"],
+            hp.get("l1_regularization", 0.0), hp["l2_regularization"],
+            cat_smooth=hp["l2_categorical_regularization"])
+        forest = exact_trees_to_forest(trees, hp["shrinkage"])
+        activation = "identity"
+        if self._task == Task.CLASSIFICATION and \
+                hp["apply_link_function"]:
+            activation = "sigmoid"
+        return GradientBoostedTreesModel(
+            forest=forest, dataspec=ds.dataspec, task=self._task,
+            label_classes=classes, init_predictions=[init],
+            num_trees_per_iter=1, activation=activation,
+            metadata={"loss": loss, "exact_splits": True,
+                      "missing_value_policy": "GLOBAL_IMPUTATION"})
+
     def train(self, data, valid=None, verbose=None
               ) -> GradientBoostedTreesModel:
         if self.tuner is not None:
             return self._train_with_tuner(data, valid=valid)
+        if not self.discretize_numerical_columns:
+            return self._train_exact(data)
         hp = self.hyperparameters
         device = self._resolve_device()
         group_ids = None
