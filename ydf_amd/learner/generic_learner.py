@@ -121,11 +121,58 @@ class GenericLearner:
                 "sparse_oblique_max_num_features", -1),
         )
 
+    def _finalize_forest(self, flat):
+        """Post-build forest fixups: expands group-space masks on
+        large-vocab categorical features into full-dictionary set
+        conditions."""
+        bigcat = getattr(self, "_bigcat_group_of_code", None)
+        if bigcat:
+            from ydf_amd.model.forest import expand_bigcat_masks
+
+            flat = expand_bigcat_masks(flat, bigcat)
+        return flat
+
+    def _compute_bigcat_maps(self, ds, cat_feats: np.ndarray):
+        """Large-vocab categorical support (vocab > 256): orders the
+        full dictionary by the global mean-label statistic (CART-style
+        ordering, reference splitter_scanner.h:859 applies it per node)
+        and packs contiguous runs into <=256 mass-balanced groups. The
+        kernels train on group indices; chosen group masks expand back
+        to full-dictionary set conditions at model build."""
+        na_mode = getattr(self, "missing_value_policy",
+                          "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"
+        n_groups = 255 if na_mode else 256
+        specs = ds.dataspec.feature_columns
+        y = ds.label_values
+        maps = {}
+        for fi in np.nonzero(cat_feats)[0]:
+            spec = specs[fi]
+            V = spec.vocab_size
+            if V <= n_groups or y is None:
+                continue
+            codes = ds.X[fi].astype(np.int64)
+            valid = codes >= 0
+            cnts = np.bincount(codes[valid], minlength=V).astype(
+                np.float64)
+            sums = np.bincount(codes[valid], weights=y[valid],
+                               minlength=V)
+            stat = sums / np.maximum(cnts, 1.0)
+            order = np.argsort(stat, kind="stable")
+            csum = np.cumsum(np.maximum(cnts[order], 1.0))
+            g_of_rank = np.minimum(
+                ((csum - 1.0) * n_groups / csum[-1]).astype(np.int64),
+                n_groups - 1)
+            group_of_code = np.empty(V, dtype=np.int32)
+            group_of_code[order] = g_of_rank
+            maps[int(fi)] = group_of_code
+        return maps or None
+
     def _bin_matrix(self, ds_X: np.ndarray, cat_feats: np.ndarray,
                     bnd: np.ndarray, device: torch.device) -> torch.Tensor:
         """Bins numericals by quantile cuts; categorical codes pass through
-        as their own bin index (clamped to 255). Under LOCAL_IMPUTATION
-        NaN rows land in reserved bin 255."""
+        as their own bin index (clamped to 255; large vocabularies remap
+        through the CART-ordered group map). Under LOCAL_IMPUTATION NaN
+        rows land in reserved bin 255."""
         na_mode = getattr(self, "missing_value_policy",
                           "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"
         X = torch.from_numpy(np.ascontiguousarray(ds_X)).to(device)
@@ -133,9 +180,22 @@ class GenericLearner:
         bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
         ops.bin_data(X, bnd_t, bins, na_to_255=na_mode)
         ci = np.nonzero(cat_feats)[0]
+        bigcat = getattr(self, "_bigcat_group_of_code", None) or {}
         if ci.size:
-            idx = torch.from_numpy(ci).to(device)
-            bins[idx] = X[idx].clamp_(0, 255).to(torch.uint8)
+            idx_np = np.asarray([i for i in ci if i not in bigcat])
+            if idx_np.size:
+                idx = torch.from_numpy(idx_np).to(device)
+                bins[idx] = X[idx].clamp_(0, 255).to(torch.uint8)
+            for fi, g_of_c in bigcat.items():
+                gt = torch.from_numpy(
+                    g_of_c.astype(np.int64)).to(device)
+                codes = X[fi].long()
+                na = codes < 0
+                grp = gt[codes.clamp(0, len(g_of_c) - 1)]
+                if na_mode:
+                    grp = torch.where(na, torch.full(
+                        (), 255, dtype=torch.int64, device=device), grp)
+                bins[fi] = grp.to(torch.uint8)
         del X
         return bins
 
@@ -189,6 +249,8 @@ class GenericLearner:
                                         dtype=np.float32)
         bnd = padded_boundaries(ds.dataspec.feature_columns)
         cat_feats = self._cat_feature_flags(ds)
+        self._bigcat_group_of_code = self._compute_bigcat_maps(
+            ds, cat_feats)
         bins = self._bin_matrix(ds.X, cat_feats, bnd, device)
         labels = None
         if ds.label_values is not None:

@@ -514,9 +514,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                     # masks + summed histograms): rank 0's snapshot is
                     # the job's snapshot; others must not race on the dir
                     return
-                flat_s = build_flat_forest(trees_so_far, bnd,
-                                           leaf_scale=hp["shrinkage"],
-                                           cat_feats=cat_feats)
+                flat_s = self._finalize_forest(build_flat_forest(
+                    trees_so_far, bnd, leaf_scale=hp["shrinkage"],
+                    cat_feats=cat_feats))
                 if partial is not None:
                     from ydf_amd.model.forest import concat_forests
 
@@ -570,11 +570,12 @@ class GradientBoostedTreesLearner(GenericLearner):
             # HostTree.scale holds each tree's final absolute leaf scale
             for tr in trees:
                 tr.leaf_value = tr.leaf_value * tr.scale
-            flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
-                                     cat_feats=cat_feats)
+            flat = self._finalize_forest(build_flat_forest(
+                trees, bnd, leaf_scale=1.0, cat_feats=cat_feats))
         else:
-            flat = build_flat_forest(trees, bnd, leaf_scale=hp["shrinkage"],
-                                     cat_feats=cat_feats)
+            flat = self._finalize_forest(build_flat_forest(
+                trees, bnd, leaf_scale=hp["shrinkage"],
+                cat_feats=cat_feats))
         gains = self._feature_gains(trees, names)
         if partial is not None:
             from ydf_amd.model.forest import concat_forests
@@ -735,8 +736,9 @@ class RandomForestLearner(GenericLearner):
             num_candidate_features=self._num_candidate(F),
             bootstrap=hp["bootstrap_training_dataset"],
             seed=self.random_seed, weights=weights, log=info)
-        flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
-                                 cat_feats=self._cat_feature_flags(ds))
+        flat = self._finalize_forest(build_flat_forest(
+            trees, bnd, leaf_scale=1.0,
+            cat_feats=self._cat_feature_flags(ds)))
         classes = self._label_classes(ds) \
             if self._task == Task.CATEGORICAL_UPLIFT else None
         model = RandomForestModel(
@@ -847,8 +849,9 @@ class RandomForestLearner(GenericLearner):
         if wta:
             for tr in trees:
                 tr.leaf_value = (tr.leaf_value > 0.5).astype(np.float32)
-        flat = build_flat_forest(trees, bnd, leaf_scale=1.0,
-                                 cat_feats=self._cat_feature_flags(ds))
+        flat = self._finalize_forest(build_flat_forest(
+            trees, bnd, leaf_scale=1.0,
+            cat_feats=self._cat_feature_flags(ds)))
         C = n_classes if (classes and n_classes > 2) else 1
         model = RandomForestModel(
             forest=flat, dataspec=ds.dataspec, task=self._task,
@@ -987,8 +990,8 @@ class CartLearner(RandomForestLearner):
         pruned = prune_tree(tree, vds.X, vds.label_values, bnd, cat_feats,
                             self._task)
         info(f"CART pruning removed {pruned} nodes")
-        model.forest = build_flat_forest([tree], bnd, leaf_scale=1.0,
-                                         cat_feats=cat_feats)
+        model.forest = self._finalize_forest(build_flat_forest(
+            [tree], bnd, leaf_scale=1.0, cat_feats=cat_feats))
         model._dev_forest = {}
         return model
 

@@ -401,3 +401,40 @@ def pack_binned_nodes(forest: FlatForest, boundaries: np.ndarray
     packed[:, 1] = thr_bits
     packed[:, 3] = -1
     return packed
+
+
+def expand_bigcat_masks(forest: FlatForest, bigcat) -> FlatForest:
+    """Converts group-space categorical masks on large-vocab features
+    (trained through the <=256 CART-ordered group map) into
+    full-dictionary set conditions (reference ContainsVector), so the
+    model predicts on raw vocabulary codes.
+
+    bigcat: {feature_idx: group_of_code i32 [vocab]}."""
+    if not bigcat:
+        return forest
+    n = forest.n_nodes
+    set_idx = forest.set_idx.copy() if forest.set_idx is not None \
+        else np.full(n, -1, dtype=np.int32)
+    offs = list(forest.set_offs) if forest.set_offs is not None else [0]
+    items = list(forest.set_items) if forest.set_items is not None else []
+    cat_idx = forest.cat_idx.copy()
+    for node in range(n):
+        fi = int(forest.feat[node])
+        ci = int(cat_idx[node])
+        if ci < 0 or fi not in bigcat:
+            continue
+        g_of_c = bigcat[fi]
+        mask = forest.masks[ci]  # u64[4] over GROUP indices
+        grp = g_of_c.astype(np.int64)
+        bit = (mask[grp >> 6] >> (grp & 63).astype(np.uint64)) \
+            & np.uint64(1)
+        codes = np.nonzero(bit.astype(bool))[0]
+        set_idx[node] = len(offs) - 1
+        items.extend(int(c) for c in codes)
+        offs.append(len(items))
+        cat_idx[node] = -1
+    forest.set_idx = set_idx
+    forest.set_offs = np.asarray(offs, dtype=np.int64)
+    forest.set_items = np.asarray(items, dtype=np.int32)
+    forest.cat_idx = cat_idx
+    return forest

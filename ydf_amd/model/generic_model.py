@@ -336,11 +336,19 @@ class GenericModel:
             si = int(f.set_idx[node])
             ci = int(f.cat_idx[node])
             if si >= 0:
-                cond = frozenset(
-                    int(v) for v in items[offs[si]:offs[si + 1]])
-                col = setcol[fi]
-                right = np.fromiter((not cond.isdisjoint(col[i])
-                                     for i in idx), bool, count=len(idx))
+                cond_items = items[offs[si]:offs[si + 1]]
+                if fi in setcol:
+                    cond = frozenset(int(v) for v in cond_items)
+                    col = setcol[fi]
+                    right = np.fromiter(
+                        (not cond.isdisjoint(col[i]) for i in idx),
+                        bool, count=len(idx))
+                else:
+                    # plain CATEGORICAL feature with a full-dictionary
+                    # set condition (large-vocab training): code in set
+                    codes = dense[fi][idx].astype(np.int64)
+                    right = np.isin(codes, np.asarray(cond_items,
+                                                      dtype=np.int64))
             elif ci >= 0:
                 cb = dense[fi][idx].astype(np.int64).clip(0, 255)
                 right = (f.masks[ci][cb >> 6]

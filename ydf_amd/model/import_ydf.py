@@ -349,9 +349,13 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
             return
         feats[idx] = rec.attr
         na_right[idx] = 1 if rec.na_value else 0
-        if rec.mask is not None and rec.attr in set_feats:
-            # categorical-SET condition: keep full element list (vocab
-            # may exceed the 256-bit mask)
+        if rec.mask is not None and (
+                rec.attr in set_feats
+                or (rec.elements is not None and len(rec.elements)
+                    and max(rec.elements) >= 256)):
+            # categorical-SET condition, or a plain-categorical
+            # ContainsVector whose dictionary exceeds the 256-bit mask
+            # (large-vocab training): keep the full element list
             if rec.elements is not None:
                 items = list(rec.elements)
             else:
