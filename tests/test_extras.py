@@ -237,3 +237,60 @@ def test_engine_api_cpu():
     m2 = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=4,
                                          validation_ratio=0).train(d2)
     assert m2.list_compatible_engines() == ["flat"]
+
+
+def test_embed_cpp_routing_codegen(binary_data, tmp_path):
+    """Table-driven (ROUTING) codegen — the default algorithm
+    (reference embed.proto:38) — on a 1000-tree forest: compiles in
+    O(nodes) data and matches model.predict."""
+    m = ydf.RandomForestLearner(label="label", num_trees=1000,
+                                max_depth=6,
+                                winner_take_all=False).train(binary_data)
+    src = ydf.to_cpp(m, "big")  # ROUTING is the default
+    assert "big_roots" in src and "while ((fi" in src
+    cpp = tmp_path / "big.cpp"
+    cpp.write_text(src + '\nextern "C" float big_predict_c(const float* f)'
+                   '{return big_predict(f);}\n')
+    so = str(tmp_path / "big.so")
+    subprocess.run(["g++", "-O1", "-shared", "-fPIC", str(cpp), "-o", so],
+                   check=True, timeout=600)
+    lib = ctypes.CDLL(so)
+    lib.big_predict_c.restype = ctypes.c_float
+    X = m._encode_features(binary_data)
+    ref = m.predict(binary_data, device="cpu")
+    for i in range(0, X.shape[1], 709):
+        row = np.ascontiguousarray(X[:, i])
+        p = lib.big_predict_c(
+            row.ctypes.data_as(ctypes.POINTER(ctypes.c_float)))
+        assert abs(p - ref[i]) < 1e-5, (i, p, ref[i])
+
+
+def test_embed_routing_categorical_and_na(tmp_path):
+    """ROUTING handles categorical masks + NA routing."""
+    rng = np.random.RandomState(4)
+    n = 6000
+    cat = rng.choice(["a", "b", "c", "d", "e"], n)
+    x = rng.randn(n).astype(np.float32)
+    x[rng.rand(n) < 0.2] = np.nan
+    y = np.where((np.isin(cat, ["a", "c"])) ^ (np.nan_to_num(x) > 0.5),
+                 "p", "q")
+    data = {"cat": cat, "x": x, "label": y}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=30, validation_ratio=0,
+        missing_value_policy="LOCAL_IMPUTATION").train(data)
+    src = ydf.to_cpp(m, "cn")
+    cpp = tmp_path / "cn.cpp"
+    cpp.write_text(src + '\nextern "C" float cn_predict_c(const float* f)'
+                   '{return cn_predict(f);}\n')
+    so = str(tmp_path / "cn.so")
+    subprocess.run(["g++", "-O2", "-shared", "-fPIC", str(cpp), "-o", so],
+                   check=True, timeout=300)
+    lib = ctypes.CDLL(so)
+    lib.cn_predict_c.restype = ctypes.c_float
+    X = m._encode_features(data)
+    ref = m.predict(data, device="cpu")
+    for i in range(0, X.shape[1], 499):
+        row = np.ascontiguousarray(X[:, i])
+        p = lib.cn_predict_c(
+            row.ctypes.data_as(ctypes.POINTER(ctypes.c_float)))
+        assert abs(p - ref[i]) < 1e-5, (i, p, ref[i])

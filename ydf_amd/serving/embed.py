@@ -61,9 +61,174 @@ def _emit_node(f, forest, n: int, indent: str, lines: List[str]):
     lines.append(f"{indent}}}" + ("}" if ci >= 0 else ""))
 
 
-def to_cpp(model, function_name: str = "ydf_model") -> str:
-    """Generates a standalone C++ source string for `model`."""
+def _check_embeddable(forest):
+    if forest.has_set_conditions:
+        raise NotImplementedError(
+            "embed codegen does not support categorical-set / "
+            "large-vocab set conditions yet")
+
+
+def _routing_tables(model, prefix: str) -> str:
+    """Emits the model as node TABLES + a fixed traversal loop
+    (reference serving/embed Algorithm=ROUTING, the default:
+    cpp_target_lowering.cc). O(nodes) data instead of O(nodes) code, so
+    1000-tree forests compile fast and small."""
+    f = model.forest
+    n = f.n_nodes
+    feat = ", ".join(str(int(v)) for v in f.feat)
+    left = ", ".join(str(int(v)) for v in f.left)
+    thr = ", ".join(f"{float(v)!r}f" for v in f.thr)
+    cat = ", ".join(str(int(v)) for v in f.cat_idx)
+    roots = ", ".join(str(int(v)) for v in f.roots)
+    out = [
+        f"static const int32_t {prefix}_feat[{n}] = {{{feat}}};",
+        f"static const float {prefix}_thr[{n}] = {{{thr}}};",
+        f"static const uint32_t {prefix}_left[{n}] = {{{left}}};",
+        f"static const int32_t {prefix}_cat[{n}] = {{{cat}}};",
+        f"static const uint32_t {prefix}_roots[{f.n_trees}] = {{{roots}}};",
+    ]
+    if len(f.masks):
+        words = ", ".join(
+            f"0x{int(w):016x}ull"
+            for w in np.asarray(f.masks, dtype=np.uint64).reshape(-1))
+        out.append(f"static const uint64_t {prefix}_masks"
+                   f"[{len(f.masks) * 4}] = {{{words}}};")
+    if f.has_na_routing:
+        bits = np.zeros((n + 63) // 64, dtype=np.uint64)
+        for i in range(n):
+            if f.na_right[i]:
+                bits[i >> 6] |= np.uint64(1 << (i & 63))
+        words = ", ".join(f"0x{int(w):016x}ull" for w in bits)
+        out.append(f"static const uint64_t {prefix}_na[{len(bits)}] = "
+                   f"{{{words}}};")
+    if len(f.obl_ranges):
+        rng = ", ".join(f"{int(a)}, {int(b)}" for a, b in f.obl_ranges)
+        attr = ", ".join(str(int(v)) for v in f.obl_attr)
+        w = ", ".join(f"{float(v)!r}f" for v in f.obl_w)
+        out += [
+            f"static const int32_t {prefix}_orng"
+            f"[{2 * len(f.obl_ranges)}] = {{{rng}}};",
+            f"static const int32_t {prefix}_oattr"
+            f"[{max(1, len(f.obl_attr))}] = {{{attr or '0'}}};",
+            f"static const float {prefix}_ow"
+            f"[{max(1, len(f.obl_w))}] = {{{w or '0.f'}}};",
+        ]
+    return "\n".join(out)
+
+
+def _routing_walk(model, prefix: str, acc_expr: str) -> str:
+    """The traversal loop over the node tables."""
+    f = model.forest
+    scale = model._leaf_scale()
+    body = [
+        f"  for (uint32_t t = 0; t < {f.n_trees}u; ++t) {{",
+        f"    uint32_t n = {prefix}_roots[t];",
+        "    int32_t fi;",
+        f"    while ((fi = {prefix}_feat[n]) >= 0) {{",
+        f"      const int32_t c = {prefix}_cat[n];",
+        "      bool right;",
+    ]
+    if len(f.masks):
+        body += [
+            "      if (c >= 0) {",
+            "        const float x = f[fi];",
+            "        const int v = x < 0 ? 0 : (x > 255 ? 255 : (int)x);",
+            f"        right = ({prefix}_masks[4 * c + (v >> 6)] >> "
+            "(v & 63)) & 1ull;",
+            "      } else",
+        ]
+    if len(f.obl_ranges):
+        body += [
+            "      if (c <= -2) {",
+            "        const int32_t r = -(c + 2);",
+            f"        const int32_t s0 = {prefix}_orng[2 * r];",
+            f"        const int32_t nn = {prefix}_orng[2 * r + 1];",
+            "        float dot = 0.f;",
+            "        for (int32_t q = 0; q < nn; ++q)",
+            f"          dot += {prefix}_ow[s0 + q] * "
+            f"f[{prefix}_oattr[s0 + q]];",
+            f"        right = dot > {prefix}_thr[n];",
+            "      } else",
+        ]
+    if f.has_na_routing:
+        body += [
+            "      if (std::isnan(f[fi])) {",
+            f"        right = ({prefix}_na[n >> 6] >> (n & 63)) & 1ull;",
+            "      } else",
+        ]
+    body += [
+        f"      right = f[fi] > {prefix}_thr[n];",
+        f"      n = {prefix}_left[n] + (right ? 1u : 0u);",
+        "    }",
+        f"    {acc_expr} += {prefix}_thr[n] * {scale!r}f;",
+        "  }",
+    ]
+    return "\n".join(body)
+
+
+def to_cpp_routing(model, function_name: str = "ydf_model") -> str:
+    """Table-driven (ROUTING) C++ codegen — the default algorithm."""
     forest = model.forest
+    _check_embeddable(forest)
+    feats = model.dataspec.feature_columns
+    C = model._n_outputs()
+    lines = [
+        "// Generated by ydf_amd (MI355X-native decision forests).",
+        "// Algorithm: ROUTING (node tables + fixed traversal loop).",
+        "// Feature order: " + ", ".join(
+            f"{i}:{c.name}" for i, c in enumerate(feats)),
+        "#include <cstdint>",
+        "#include <cmath>",
+        "",
+        _routing_tables(model, function_name),
+        "",
+    ]
+    if C == 1:
+        lines.append(f"float {function_name}_predict(const float* f) {{")
+        lines.append(
+            f"  float acc = {float(model.init_predictions[0])!r}f;")
+        lines.append(_routing_walk(model, function_name, "acc"))
+        if model.activation == "sigmoid":
+            lines.append("  return 1.0f / (1.0f + std::exp(-acc));")
+        elif model.activation == "exp":
+            lines.append("  return std::exp(acc);")
+        else:
+            lines.append("  return acc;")
+        lines.append("}")
+    else:
+        lines.append(f"void {function_name}_predict_multi("
+                     "const float* f, float* out) {")
+        for c in range(C):
+            init = float(model.init_predictions[c]
+                         if c < len(model.init_predictions) else 0.0)
+            lines.append(f"  out[{c}] = {init!r}f;")
+        lines.append(_routing_walk(model, function_name,
+                                   f"out[t % {C}u]"))
+        if model.activation == "softmax":
+            lines.append(
+                "  float m = out[0];\n"
+                f"  for (int c = 1; c < {C}; ++c) m = out[c] > m ? out[c] "
+                ": m;\n"
+                "  float s = 0.f;\n"
+                f"  for (int c = 0; c < {C}; ++c) {{ out[c] = "
+                "std::exp(out[c] - m); s += out[c]; }\n"
+                f"  for (int c = 0; c < {C}; ++c) out[c] /= s;")
+        lines.append("}")
+    return "\n".join(lines) + "\n"
+
+
+def to_cpp(model, function_name: str = "ydf_model",
+           algorithm: str = "ROUTING") -> str:
+    """Generates a standalone C++ source string for `model`.
+
+    algorithm="ROUTING" (default, like the reference embed.proto:38):
+    node tables + fixed walk loop, O(nodes) data — scales to
+    1000-tree forests. "IF_ELSE": per-node branch code (small models,
+    fully branch-predictable)."""
+    if algorithm == "ROUTING":
+        return to_cpp_routing(model, function_name)
+    forest = model.forest
+    _check_embeddable(forest)
     feats = model.dataspec.feature_columns
     C = model._n_outputs()
     lines = [
@@ -184,6 +349,7 @@ def to_java(model, class_name: str = "YdfModel") -> str:
     serving/embed/java/java_embed.cc). Binary classification returns the
     positive-class probability; multi-class emits predictMulti."""
     forest = model.forest
+    _check_embeddable(forest)
     feats = model.dataspec.feature_columns
     C = model._n_outputs()
     scale = model._leaf_scale()
