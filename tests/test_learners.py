@@ -248,6 +248,33 @@ def test_random_search_tuner(binary_data):
         t.score for t in m.tuner_logs.trials) - 1e-9
 
 
+def test_parallel_tuner_matches_sequential(binary_data):
+    """parallel_trials=4 (one trial per process slot; one per GPU when
+    present — reference distributed HPO analogue) must sample the same
+    trials as sequential search and pick an equally good winner, and
+    the tuning logs must show up in describe()."""
+    def make_tuner(par):
+        t = ydf.RandomSearchTuner(num_trials=4, seed=7,
+                                  parallel_trials=par)
+        t.choice("shrinkage", [0.05, 0.2])
+        t.choice("max_depth", [3, 5])
+        return t
+
+    kw = dict(label="label", num_trees=10, device="cpu")
+    m_par = ydf.GradientBoostedTreesLearner(
+        tuner=make_tuner(4), **kw).train(binary_data)
+    m_seq = ydf.GradientBoostedTreesLearner(
+        tuner=make_tuner(1), **kw).train(binary_data)
+    assert len(m_par.tuner_logs.trials) == 4
+    hp_par = [t.hyperparameters for t in m_par.tuner_logs.trials]
+    hp_seq = [t.hyperparameters for t in m_seq.tuner_logs.trials]
+    assert hp_par == hp_seq  # same seed -> same sampled trials
+    np.testing.assert_allclose(m_par.tuner_logs.best_trial.score,
+                               m_seq.tuner_logs.best_trial.score,
+                               rtol=1e-5)
+    assert "tuning: 4 trials" in m_par.describe()
+
+
 def test_cross_validation(binary_data):
     ev = ydf.GradientBoostedTreesLearner(
         label="label", num_trees=15).cross_validation(binary_data, folds=3)

@@ -95,6 +95,8 @@ def infer_dataspec(
     min_vocab_frequency: int = 1,
     max_bins: int = 256,
     allow_na_conditions: bool = False,
+    vecseq_num_anchors: int = 20,
+    vecseq_seed: int = 1234,
 ) -> DataSpecification:
     """Single-pass dataspec inference (reference: data_spec_inference.h:55)."""
     from ydf_amd.dataset.dataspec import Task
@@ -130,9 +132,27 @@ def infer_dataspec(
             elif arr.dtype == object and len(arr) and isinstance(
                     next((c for c in arr if c is not None), None),
                     (list, tuple, set, frozenset, np.ndarray)):
-                sem = Semantic.CATEGORICAL_SET
+                from ydf_amd.dataset.vecseq import is_vecseq_cell
+
+                first = next((c for c in arr if c is not None), None)
+                sem = Semantic.NUMERICAL_VECTOR_SEQUENCE \
+                    if is_vecseq_cell(first) else Semantic.CATEGORICAL_SET
             else:
                 sem = Semantic.CATEGORICAL
+        if sem == Semantic.NUMERICAL_VECTOR_SEQUENCE and not is_label:
+            # ragged vector-sequence column -> per-anchor virtual
+            # projection columns (ydf_amd/dataset/vecseq.py; reference
+            # gpu.cu.cc conditions CloserThan/ProjectedMoreThan)
+            from ydf_amd.dataset.vecseq import (extract_ragged,
+                                                virtual_specs)
+
+            values, offs, dim = extract_ragged(arr)
+            columns.extend(virtual_specs(
+                name, values, offs, dim, n_anchors=vecseq_num_anchors,
+                seed=vecseq_seed,
+                numerical_boundaries_fn=lambda v: numerical_boundaries(
+                    v, max_bins=max_bins)))
+            continue
         if sem == Semantic.CATEGORICAL_SET and not is_label:
             # multi-valued categorical (reference CategoricalSet columns):
             # expanded into per-token boolean "contains" virtual features;
@@ -300,11 +320,19 @@ def create_vertical_dataset(
     feature_specs = dataspec.feature_columns
     n = len(next(iter(cols.values()))) if cols else 0
     X = np.empty((len(feature_specs), n), dtype=np.float32)
+    has_vecseq = False
     for i, spec in enumerate(feature_specs):
+        if spec.vecseq_source is not None:
+            has_vecseq = True
+            continue
         src = spec.set_source or spec.name
         if src not in cols:
             raise ValueError(f"missing feature column {src!r}")
         X[i] = encode_column(cols[src], spec, keep_na=keep_na)
+    if has_vecseq:
+        from ydf_amd.dataset.vecseq import fill_vecseq_columns
+
+        fill_vecseq_columns(X, feature_specs, cols)
 
     label_values = None
     if dataspec.label is not None and dataspec.label in cols:

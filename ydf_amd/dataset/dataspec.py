@@ -24,6 +24,7 @@ class Semantic(enum.Enum):
     HASH = 4
     CATEGORICAL_SET = 5
     DISCRETIZED_NUMERICAL = 6
+    NUMERICAL_VECTOR_SEQUENCE = 7
 
 
 class Task(enum.Enum):
@@ -73,6 +74,17 @@ class ColumnSpec:
     # conditions, approximated by per-token contains conditions)
     set_source: Optional[str] = None
     set_token: Optional[str] = None
+    # NUMERICAL_VECTOR_SEQUENCE (reference data_spec.proto:73-81 + the
+    # only GPU code in the reference, learner/decision_tree/gpu.cu.cc):
+    # the parent column stores vecseq_dim; virtual projection columns
+    # store (vecseq_source, vecseq_kind, vecseq_anchor) and hold
+    #   kind "dot":  max_k <vec_k, anchor>    (ProjectedMoreThan)
+    #   kind "dist": -min_k |vec_k - anchor|^2 (CloserThan, negated so
+    #                 every condition keeps the ">= threshold" shape)
+    vecseq_dim: int = 0
+    vecseq_source: Optional[str] = None
+    vecseq_kind: Optional[str] = None
+    vecseq_anchor: Optional[np.ndarray] = None
 
     @property
     def vocab_size(self) -> int:
@@ -94,6 +106,12 @@ class ColumnSpec:
         if self.set_source is not None:
             d["set_source"] = self.set_source
             d["set_token"] = self.set_token
+        if self.vecseq_dim:
+            d["vecseq_dim"] = int(self.vecseq_dim)
+        if self.vecseq_source is not None:
+            d["vecseq_source"] = self.vecseq_source
+            d["vecseq_kind"] = self.vecseq_kind
+            d["vecseq_anchor"] = [float(v) for v in self.vecseq_anchor]
         return d
 
     @classmethod
@@ -111,6 +129,11 @@ class ColumnSpec:
             else None,
             set_source=d.get("set_source"),
             set_token=d.get("set_token"),
+            vecseq_dim=d.get("vecseq_dim", 0),
+            vecseq_source=d.get("vecseq_source"),
+            vecseq_kind=d.get("vecseq_kind"),
+            vecseq_anchor=np.asarray(d["vecseq_anchor"], dtype=np.float32)
+            if "vecseq_anchor" in d else None,
         )
 
 

@@ -348,8 +348,14 @@ class GenericLearner:
     def _train_with_tuner(self, data, valid=None):
         """Random-search trials; keeps the best model by validation loss
         (reference hyperparameters_optimizer.cc random trials)."""
-        from ydf_amd.learner.tuner import OptimizerLogs, TrialLog
+        from ydf_amd.learner.tuner import (OptimizerLogs, TrialLog,
+                                           run_parallel_trials)
 
+        if getattr(self.tuner, "parallel_trials", 1) > 1:
+            # one trial per process slot (one per GPU when available):
+            # the MI355X mapping of the reference's distributed HPO
+            # (hyperparameters_optimizer.h:46 + generic_worker/)
+            return run_parallel_trials(self, self.tuner, data)
         rng = np.random.RandomState(self.tuner.seed)
         tuner = self.tuner
         best = None

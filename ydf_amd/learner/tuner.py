@@ -26,12 +26,22 @@ class OptimizerLogs:
 
 
 class RandomSearchTuner:
-    """Random search over declared hyper-parameter choices."""
+    """Random search over declared hyper-parameter choices.
+
+    parallel_trials > 1 evaluates trials concurrently, one process per
+    trial slot (the MI355X mapping of the reference's distributed HPO:
+    hyperparameters_optimizer.h:46 + learner/generic_worker/ train one
+    model per remote worker). On a multi-GPU box each slot pins a GPU
+    round-robin; on CPU the slots are plain processes. Trial sampling
+    happens up front from the seed, so results are independent of the
+    execution order."""
 
     def __init__(self, num_trials: int = 50, automatic_search_space:
-                 bool = False, seed: int = 1234):
+                 bool = False, seed: int = 1234,
+                 parallel_trials: int = 1):
         self.num_trials = num_trials
         self.seed = seed
+        self.parallel_trials = parallel_trials
         self._choices: Dict[str, list] = {}
         if automatic_search_space:
             # predefined space (reference PredefinedHyperParameterSpace,
@@ -58,3 +68,55 @@ class RandomSearchTuner:
 class VizierTuner(RandomSearchTuner):
     """Placeholder keeping API parity: falls back to random search (the
     reference's Vizier backend is Google-internal)."""
+
+
+def _trial_worker(payload):
+    """One tuning trial in a worker process (spawn-safe top-level).
+
+    payload: (pickled learner copy, hp_override, data, slot_idx).
+    Returns (score, serialized model bytes)."""
+    import torch
+
+    from ydf_amd.model.model_lib import serialize_model
+
+    learner, hp, data, slot = payload
+    if torch.cuda.is_available():
+        # one trial per GPU, round-robin over slots
+        learner.device = f"cuda:{slot % torch.cuda.device_count()}"
+    learner.hyperparameters = dict(learner.hyperparameters)
+    learner.hyperparameters.update(
+        {k: v for k, v in hp.items() if k in learner.hyperparameters})
+    if learner.hyperparameters.get("validation_ratio", 0) == 0:
+        learner.hyperparameters["validation_ratio"] = 0.1
+    model = learner.train(data)
+    vloss = None
+    if model.training_logs:
+        vloss = model.training_logs[-1].get("valid_loss")
+    score = -(vloss if vloss is not None else float("inf"))
+    return score, serialize_model(model)
+
+
+def run_parallel_trials(learner, tuner, data):
+    """Evaluates all trials in a process pool (parallel_trials slots)."""
+    import multiprocessing as mp
+
+    from ydf_amd.dataset.dataset import _to_column_dict
+    from ydf_amd.model.model_lib import deserialize_model
+
+    import copy
+
+    rng = np.random.RandomState(tuner.seed)
+    samples = [tuner.sample(rng) for _ in range(tuner.num_trials)]
+    cols = _to_column_dict(data)
+    ln = copy.copy(learner)
+    ln.tuner = None  # the worker runs a plain (non-tuning) train
+    payloads = [(ln, hp, cols, i) for i, hp in enumerate(samples)]
+    ctx = mp.get_context("spawn")
+    with ctx.Pool(processes=tuner.parallel_trials) as pool:
+        results = pool.map(_trial_worker, payloads)
+    logs = [TrialLog(hyperparameters=hp, score=score)
+            for hp, (score, _) in zip(samples, results)]
+    best_i = int(np.argmax([r[0] for r in results]))
+    model = deserialize_model(results[best_i][1])
+    model.tuner_logs = OptimizerLogs(trials=logs)
+    return model

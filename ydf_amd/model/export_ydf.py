@@ -292,6 +292,11 @@ def export_ydf_model(model, path: str) -> None:
             "models trained on expanded categorical-set token features "
             "use a virtual-column representation the reference data "
             "spec cannot express")
+    if any(getattr(c, "vecseq_source", None)
+           for c in model.dataspec.columns):
+        raise NotImplementedError(
+            "vector-sequence models: reference-format export of "
+            "NumericalVectorSequence conditions is not implemented yet")
     os.makedirs(path, exist_ok=True)
     # Column indexing: node conditions reference COLUMN indices in the
     # data spec, not dense feature indices (the label can sit anywhere —

@@ -158,11 +158,19 @@ class GenericModel:
         keep_na = self.forest.has_na_routing or (
             (self.metadata or {}).get("missing_value_policy")
             == "LOCAL_IMPUTATION")
+        has_vecseq = False
         for i, spec in enumerate(specs):
+            if spec.vecseq_source is not None:
+                has_vecseq = True
+                continue
             src = spec.set_source or spec.name
             if src not in cols:
                 raise ValueError(f"missing input feature {src!r}")
             X[i] = encode_column(cols[src], spec, keep_na=keep_na)
+        if has_vecseq:
+            from ydf_amd.dataset.vecseq import fill_vecseq_columns
+
+            fill_vecseq_columns(X, specs, cols)
         return X
 
     def list_compatible_engines(self):
@@ -654,6 +662,13 @@ class GenericModel:
         if ev is not None and getattr(ev, "accuracy", None) is not None:
             lines.append(f"self evaluation (OOB/validation): "
                          f"accuracy={ev.accuracy:.4f}")
+        if self.tuner_logs is not None and self.tuner_logs.trials:
+            bt = self.tuner_logs.best_trial
+            lines.append(
+                f"tuning: {len(self.tuner_logs.trials)} trials, best "
+                f"score {bt.score:.6g}, best hyperparameters "
+                + ", ".join(f"{k}={v}"
+                            for k, v in bt.hyperparameters.items()))
         if output_format == "html":
             rows = "".join(f"<tr><td>{ln.split(':', 1)[0]}</td>"
                            f"<td>{ln.split(':', 1)[1] if ':' in ln else ''}"

@@ -416,3 +416,28 @@ def sigmoid(x: torch.Tensor, out: torch.Tensor):
     else:
         torch.sigmoid(x, out=out)
     return out
+
+
+def vecseq_project(values: torch.Tensor, offs: torch.Tensor,
+                   anchors: torch.Tensor):
+    """NUMERICAL_VECTOR_SEQUENCE projections (reference gpu.cu.cc:46-136
+    redesigned for wave64 + LDS anchors): returns (maxdot [A,N],
+    negminsq [A,N]) f32. CPU falls back to the vectorized numpy twin."""
+    N = offs.numel() - 1
+    A = anchors.shape[0]
+    dim = anchors.shape[1] if A else 1
+    if values.is_cuda:
+        maxdot = torch.empty((A, N), dtype=torch.float32,
+                             device=values.device)
+        negminsq = torch.empty((A, N), dtype=torch.float32,
+                               device=values.device)
+        maxdot.fill_(-3.0e38)
+        negminsq.fill_(-3.0e38)
+        _C.gpu_vecseq_project(values.data_ptr(), offs.data_ptr(),
+                              anchors.data_ptr(), maxdot.data_ptr(),
+                              negminsq.data_ptr(), N, dim, A, _stream())
+        return maxdot, negminsq
+    from ydf_amd.dataset.vecseq import project_numpy
+
+    md, ns = project_numpy(values.numpy(), offs.numpy(), anchors.numpy())
+    return torch.from_numpy(md), torch.from_numpy(ns)

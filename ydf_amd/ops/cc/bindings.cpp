@@ -71,6 +71,8 @@ void gpu_predict_forest_binned(const uint8_t*, int64_t, int,
 void gpu_predict_forest_qs(const float*, int64_t, int, const int32_t*,
                            const int32_t*, const float*, int, float*,
                            float, float, void*);
+void gpu_vecseq_project(const float*, const int64_t*, const float*,
+                        float*, float*, int64_t, int, int, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
                   int);
@@ -353,6 +355,16 @@ PYBIND11_MODULE(_ydf_ops, m) {
                                 P<int32_t>(cond_offs), P<float>(leaf_vals),
                                 n_trees, P<float>(out), init, scale,
                                 (void*)stream);
+        },
+        nogil);
+  m.def("gpu_vecseq_project",
+        [](uintptr_t values, uintptr_t offs, uintptr_t anchors,
+           uintptr_t maxdot, uintptr_t negminsq, int64_t N, int dim,
+           int A, uintptr_t stream) {
+          gpu_vecseq_project(P<float>(values), P<int64_t>(offs),
+                             P<float>(anchors), P<float>(maxdot),
+                             P<float>(negminsq), N, dim, A,
+                             (void*)stream);
         },
         nogil);
   m.def("gpu_sigmoid",

@@ -266,7 +266,60 @@ __global__ void sigmoid_kernel(const float* __restrict__ in,
     out[k] = 1.0f / (1.0f + __expf(-in[k]));
 }
 
+// NUMERICAL_VECTOR_SEQUENCE projections — the one feature type the
+// reference itself accelerated on GPU (learner/decision_tree/
+// gpu.cu.cc:46-136), redesigned for wave64: one block covers a
+// 256-example tile for one anchor; the anchor (+|a|^2) is staged in
+// LDS; each lane scans its example's vector run computing BOTH
+// condition statistics in one pass:
+//   maxdot[a][i]   = max_k <vec_ik, anchor_a>       (ProjectedMoreThan)
+//   negminsq[a][i] = -min_k |vec_ik - anchor_a|^2   (CloserThan)
+// Empty runs stay at -3e38 (every "exists" condition false).
+__global__ void vecseq_project_kernel(
+    const float* __restrict__ values,   // [K][dim]
+    const int64_t* __restrict__ offs,   // [N+1]
+    const float* __restrict__ anchors,  // [A][dim]
+    float* __restrict__ maxdot,         // [A][N]
+    float* __restrict__ negminsq,       // [A][N]
+    int64_t N, int dim) {
+  extern __shared__ float a_s[];  // [dim]
+  const int aidx = blockIdx.y;
+  for (int d = threadIdx.x; d < dim; d += blockDim.x)
+    a_s[d] = anchors[(int64_t)aidx * dim + d];
+  __syncthreads();
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= N) return;
+  float best_dot = -3.0e38f, best_neg = -3.0e38f;
+  const int64_t k0 = offs[i], k1 = offs[i + 1];
+  for (int64_t k = k0; k < k1; ++k) {
+    const float* v = values + k * dim;
+    float dot = 0.f, sq = 0.f;
+    for (int d = 0; d < dim; ++d) {
+      const float x = v[d];
+      dot = fmaf(x, a_s[d], dot);
+      const float diff = x - a_s[d];
+      sq = fmaf(diff, diff, sq);
+    }
+    best_dot = fmaxf(best_dot, dot);
+    best_neg = fmaxf(best_neg, -sq);
+  }
+  maxdot[(int64_t)aidx * N + i] = best_dot;
+  negminsq[(int64_t)aidx * N + i] = best_neg;
+}
+
 extern "C" {
+
+void gpu_vecseq_project(const float* values, const int64_t* offs,
+                        const float* anchors, float* maxdot,
+                        float* negminsq, int64_t N, int dim, int A,
+                        void* stream) {
+  if (A <= 0 || N <= 0) return;
+  const int grid_x = (int)((N + 255) / 256);
+  const size_t lds = (size_t)dim * sizeof(float);
+  hipLaunchKernelGGL(vecseq_project_kernel, dim3(grid_x, A), dim3(256),
+                     lds, (hipStream_t)stream, values, offs, anchors,
+                     maxdot, negminsq, N, dim);
+}
 
 void gpu_predict_forest_binned(const uint8_t* B, int64_t N, int F,
                                const int32_t* packed_nodes,
