@@ -14,7 +14,19 @@
 //     count} are the collective unit: data-parallel ranks AllReduce this
 //     tensor (RCCL over xGMI) and then select splits redundantly.
 #pragma once
+#include <cmath>
 #include <cstdint>
+
+// Host-only builds (the C++ user API links cpu_ops.cpp with a plain
+// C++ compiler): the HIP function-space qualifiers become no-ops.
+#ifndef __HIPCC__
+#ifndef __host__
+#define __host__
+#endif
+#ifndef __device__
+#define __device__
+#endif
+#endif
 
 namespace ydfa {
 
