@@ -71,6 +71,7 @@ from ydf_amd.learner.tuner import (
 )
 
 # Dataset
+from ydf_amd.dataset.cache import DatasetCache, create_dataset_cache
 from ydf_amd.dataset.dataset import VerticalDataset, create_vertical_dataset
 from ydf_amd.dataset.dataspec import (
     Column,

@@ -31,11 +31,10 @@ class GenericLearner:
         self.allow_na_conditions = allow_na_conditions
         self.pure_serving_model = pure_serving_model
         if missing_value_policy not in ("GLOBAL_IMPUTATION",
-                                        "LOCAL_IMPUTATION"):
-            raise NotImplementedError(
-                "missing_value_policy: GLOBAL_IMPUTATION and "
-                "LOCAL_IMPUTATION are implemented (RANDOM_LOCAL: "
-                "ROADMAP)")
+                                        "LOCAL_IMPUTATION",
+                                        "RANDOM_LOCAL_IMPUTATION"):
+            raise ValueError(
+                f"unknown missing_value_policy {missing_value_policy!r}")
         self.missing_value_policy = missing_value_policy
         if categorical_algorithm != "CART":
             raise NotImplementedError(
@@ -173,12 +172,38 @@ class GenericLearner:
         as their own bin index (clamped to 255; large vocabularies remap
         through the CART-ordered group map). Under LOCAL_IMPUTATION NaN
         rows land in reserved bin 255."""
-        na_mode = getattr(self, "missing_value_policy",
-                          "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"
+        policy = getattr(self, "missing_value_policy",
+                         "GLOBAL_IMPUTATION")
+        na_mode = policy == "LOCAL_IMPUTATION"
+        random_na = policy == "RANDOM_LOCAL_IMPUTATION"
         X = torch.from_numpy(np.ascontiguousarray(ds_X)).to(device)
         bnd_t = torch.from_numpy(bnd).to(device)
         bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
-        ops.bin_data(X, bnd_t, bins, na_to_255=na_mode)
+        ops.bin_data(X, bnd_t, bins, na_to_255=na_mode or random_na)
+        if random_na:
+            # RANDOM_LOCAL_IMPUTATION (reference decision_tree.proto:
+            # 99-103, Random Survival Forests): missing values imputed
+            # by randomly sampled observed values. Deviation
+            # (documented): sampled once per forest from the feature's
+            # GLOBAL observed bin distribution (the reference redraws
+            # per node) — preserves the feature distribution, which is
+            # the policy's point, while keeping the binned store
+            # immutable across trees. Serving imputes the global mean
+            # (deterministic predictions).
+            g = torch.Generator(device="cpu")
+            g.manual_seed(int(self.random_seed) * 7919 + 13)
+            for f in range(bins.shape[0]):
+                na = bins[f] == 255
+                n_na = int(na.sum().item())
+                if n_na == 0:
+                    continue
+                obs = bins[f][~na]
+                if obs.numel() == 0:
+                    bins[f][na] = 0
+                    continue
+                pick = torch.randint(0, obs.numel(), (n_na,),
+                                     generator=g).to(device)
+                bins[f][na] = obs[pick]
         ci = np.nonzero(cat_feats)[0]
         bigcat = getattr(self, "_bigcat_group_of_code", None) or {}
         if ci.size:
@@ -233,7 +258,8 @@ class GenericLearner:
                 features = [c for c in cols
                             if c not in (self.label, self.weights_col)]
             local_na = getattr(self, "missing_value_policy",
-                               "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"
+                               "GLOBAL_IMPUTATION") in (
+                "LOCAL_IMPUTATION", "RANDOM_LOCAL_IMPUTATION")
             ds = create_vertical_dataset(
                 cols, label=self.label, task=self._task,
                 features=features, max_vocab_count=self.max_vocab_count,

@@ -237,8 +237,73 @@ class GradientBoostedTreesLearner(GenericLearner):
             metadata={"loss": loss, "exact_splits": True,
                       "missing_value_policy": "GLOBAL_IMPUTATION"})
 
+    def _train_streaming(self, cache) -> GradientBoostedTreesModel:
+        """Out-of-core GBT over an on-disk binned DatasetCache
+        (learner/streaming.py; reference ShardedSamplingTrain,
+        gradient_boosted_trees.cc:655). Chunks stream through the
+        device; the dataset can exceed HBM."""
+        from ydf_amd.dataset.dataspec import Semantic
+        from ydf_amd.learner.streaming import train_gbt_streaming
+        from ydf_amd.model.forest import padded_boundaries
+
+        hp = self.hyperparameters
+        if self._task not in (Task.CLASSIFICATION, Task.REGRESSION):
+            raise NotImplementedError(
+                "streaming training supports classification/regression")
+        if hp.get("subsample", 1.0) < 1.0 or \
+                hp.get("split_axis", "AXIS_ALIGNED") != "AXIS_ALIGNED" \
+                or hp.get("forest_extraction") == "DART":
+            raise NotImplementedError(
+                "streaming training: axis-aligned MART without "
+                "subsampling")
+        device = self._resolve_device()
+        lspec = cache.dataspec.label_column
+        classes = None
+        if self._task == Task.CLASSIFICATION:
+            if lspec.semantic != Semantic.CATEGORICAL:
+                raise ValueError("classification label must be "
+                                 "categorical")
+            classes = list(lspec.vocab[1:])
+            if len(classes) != 2:
+                raise NotImplementedError(
+                    "streaming training supports binary classification")
+        loss = trainer_lib.LOSS_BINOMIAL \
+            if self._task == Task.CLASSIFICATION \
+            else trainer_lib.LOSS_SQUARED_ERROR
+        cfg = trainer_lib.TrainerConfig(
+            loss=loss, num_trees=hp["num_trees"],
+            max_depth=hp["max_depth"], shrinkage=hp["shrinkage"],
+            lambda_l2=hp["l2_regularization"],
+            lambda_l1=hp.get("l1_regularization", 0.0),
+            min_examples=hp["min_examples"],
+            min_hessian=hp["min_sum_hessian_in_leaf"],
+            cat_smooth=hp["l2_categorical_regularization"],
+            seed=self.random_seed)
+        trees, init = train_gbt_streaming(cache, cfg, device, log=info)
+        bnd = padded_boundaries(cache.dataspec.feature_columns)
+        cat_feats = np.asarray(
+            [c.semantic == Semantic.CATEGORICAL
+             for c in cache.dataspec.feature_columns], dtype=bool)
+        flat = build_flat_forest(trees, bnd,
+                                 leaf_scale=hp["shrinkage"],
+                                 cat_feats=cat_feats)
+        activation = "identity"
+        if self._task == Task.CLASSIFICATION and \
+                hp["apply_link_function"]:
+            activation = "sigmoid"
+        return GradientBoostedTreesModel(
+            forest=flat, dataspec=cache.dataspec, task=self._task,
+            label_classes=classes, init_predictions=[init],
+            num_trees_per_iter=1, activation=activation,
+            metadata={"loss": int(loss), "streaming_cache": True,
+                      "missing_value_policy": "GLOBAL_IMPUTATION"})
+
     def train(self, data, valid=None, verbose=None
               ) -> GradientBoostedTreesModel:
+        from ydf_amd.dataset.cache import DatasetCache
+
+        if isinstance(data, DatasetCache):
+            return self._train_streaming(data)
         if self.tuner is not None:
             return self._train_with_tuner(data, valid=valid)
         if not self.discretize_numerical_columns:
