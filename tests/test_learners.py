@@ -995,3 +995,36 @@ def test_random_local_imputation():
     p1 = m_rand.predict(d)
     p2 = m_rand.predict(d)
     np.testing.assert_array_equal(p1, p2)
+
+
+def test_dart_multiclass_and_na():
+    """DART (reference forest_extraction=DART,
+    gradient_boosted_trees.h:338) now covers multi-class losses and
+    LOCAL_IMPUTATION NA routing: dropout drops whole iterations (all
+    class trees together) and dropped-tree replay carries the per-node
+    NA direction bits."""
+    rng = np.random.RandomState(13)
+    n = 6000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where(x1 > 0.5, "a", np.where(x2 > 0, "b", "c"))
+    d = {"x1": x1, "x2": x2, "label": y}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=30, forest_extraction="DART",
+        dart_dropout=0.1, validation_ratio=0.0).train(d)
+    ev = m.evaluate(d)
+    assert ev.accuracy > 0.95
+    p = m.predict(d)
+    assert p.shape == (n, 3)
+    np.testing.assert_allclose(p.sum(axis=1), 1.0, atol=1e-4)
+
+    # NA routing + DART
+    x1na = x1.copy()
+    x1na[rng.rand(n) < 0.3] = np.nan
+    d2 = {"x1": x1na, "x2": x2,
+          "label": np.where(np.nan_to_num(x1na) + x2 > 0, "p", "q")}
+    m2 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=20, forest_extraction="DART",
+        dart_dropout=0.1, validation_ratio=0.0,
+        missing_value_policy="LOCAL_IMPUTATION").train(d2)
+    assert m2.evaluate(d2).accuracy > 0.8

@@ -1117,7 +1117,7 @@ class ForestTrainer:
 
     def route_tree(self, bins: torch.Tensor, node_ids: torch.Tensor,
                    feat_t: torch.Tensor, bin_t: torch.Tensor,
-                   masks_t: Optional[torch.Tensor]):
+                   masks_t: Optional[torch.Tensor], na_t=None):
         """Routes rows through a PREVIOUSLY extracted tree given its
         device arrays (DART dropout re-evaluation). Oblique trees are not
         supported here (projections are per-level ephemeral)."""
@@ -1130,7 +1130,7 @@ class ForestTrainer:
                 feat_t[level_base:level_base + level_size],
                 bin_t[level_base:level_base + level_size],
                 level_base, level_size, cat_flags=self.cat_flags,
-                masks=masks_t)
+                masks=masks_t, tree_na=na_t)
         return node_ids
 
     def route_rows(self, bins: torch.Tensor, node_ids: torch.Tensor,
@@ -1269,8 +1269,6 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
     trees: List[HostTree] = []
     dart = cfg.dart_dropout > 0.0
     if dart:
-        if C > 1:
-            raise NotImplementedError("DART supports single-output losses")
         if cfg.oblique_projections > 0:
             raise NotImplementedError("DART + oblique is not supported "
                                       "(per-level projections)")
@@ -1282,12 +1280,11 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             raise NotImplementedError(
                 "DART replays dropped trees from the complete-tree "
                 "buffers; BEST_FIRST_GLOBAL trees live on implicit keys")
-        if cfg.na_mode:
-            raise NotImplementedError(
-                "DART + LOCAL_IMPUTATION: dropped-tree replay does not "
-                "carry the per-tree na bits yet")
-        dart_rec = []   # per tree: (feat_dev, bin_dev, masks_dev, leaf_dev)
-        dart_scale = []  # per tree: current absolute leaf scale
+        # per ITERATION: a list of C per-class records
+        # (feat_dev, bin_dev, masks_dev, na_dev, leaf_dev); dropout
+        # drops whole iterations (all classes together)
+        dart_rec = []
+        dart_scale = []  # per iteration: current absolute leaf scale
         dart_rng = np.random.RandomState(cfg.seed ^ 0x5bd1e995)
         dart_ids = torch.empty_like(trainer.node_ids)
         dart_valid_ids = torch.empty(
@@ -1338,14 +1335,17 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                 dmask[dart_rng.randint(len(dart_rec))] = True
             dropped = list(np.nonzero(dmask)[0])
             for ti in dropped:
-                ft, bt, mt, lv = dart_rec[ti]
-                trainer.route_tree(trainer.bins, dart_ids, ft, bt, mt)
-                preds[0].sub_(lv[dart_ids.long()], alpha=dart_scale[ti])
-                if has_valid:
-                    trainer.route_tree(trainer.valid_bins, dart_valid_ids,
-                                       ft, bt, mt)
-                    valid_preds[0].sub_(lv[dart_valid_ids.long()],
-                                        alpha=dart_scale[ti])
+                for c2, (ft, bt, mt, nt, lv) in enumerate(dart_rec[ti]):
+                    trainer.route_tree(trainer.bins, dart_ids, ft, bt,
+                                       mt, na_t=nt)
+                    preds[c2].sub_(lv[dart_ids.long()],
+                                   alpha=dart_scale[ti])
+                    if has_valid:
+                        trainer.route_tree(trainer.valid_bins,
+                                           dart_valid_ids, ft, bt, mt,
+                                           na_t=nt)
+                        valid_preds[c2].sub_(lv[dart_valid_ids.long()],
+                                             alpha=dart_scale[ti])
         try:
           for c in range(C):
             pc = preds[c]
@@ -1424,12 +1424,16 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             if dart:
                 k = len(dropped)
                 step_scale = cfg.shrinkage / (k + 1)
-                dart_rec.append((
+                if c == 0:
+                    dart_rec.append([])
+                    dart_scale.append(step_scale)
+                dart_rec[-1].append((
                     trainer.tree_feat.clone(), trainer.tree_bin.clone(),
                     trainer.tree_masks.clone()
                     if trainer.tree_masks is not None else None,
+                    trainer.tree_na.clone()
+                    if trainer.tree_na is not None else None,
                     trainer.leaf_vals.clone()))
-                dart_scale.append(step_scale)
             ops.update_preds(pc, trainer.node_ids, trainer.leaf_vals,
                              step_scale)
             if has_valid:
@@ -1437,19 +1441,22 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                                    trainer.valid_node_ids)
                 ops.update_preds(valid_preds[c], trainer.valid_node_ids,
                                  trainer.leaf_vals, step_scale)
-            if dart and dropped:
-                k = len(dropped)
-                for ti in dropped:
-                    ft, bt, mt, lv = dart_rec[ti]
-                    new_scale = dart_scale[ti] * k / (k + 1)
-                    dart_scale[ti] = new_scale
-                    trainer.route_tree(trainer.bins, dart_ids, ft, bt, mt)
-                    preds[0].add_(lv[dart_ids.long()], alpha=new_scale)
+          if dart and dropped:
+            k = len(dropped)
+            for ti in dropped:
+                new_scale = dart_scale[ti] * k / (k + 1)
+                dart_scale[ti] = new_scale
+                for c2, (ft, bt, mt, nt, lv) in enumerate(dart_rec[ti]):
+                    trainer.route_tree(trainer.bins, dart_ids, ft, bt,
+                                       mt, na_t=nt)
+                    preds[c2].add_(lv[dart_ids.long()],
+                                   alpha=new_scale)
                     if has_valid:
                         trainer.route_tree(trainer.valid_bins,
-                                           dart_valid_ids, ft, bt, mt)
-                        valid_preds[0].add_(lv[dart_valid_ids.long()],
-                                            alpha=new_scale)
+                                           dart_valid_ids, ft, bt, mt,
+                                           na_t=nt)
+                        valid_preds[c2].add_(lv[dart_valid_ids.long()],
+                                             alpha=new_scale)
         except KeyboardInterrupt:
             if log:
                 log(f"interrupted at iteration {it}; returning partial model")
@@ -1506,8 +1513,10 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
         trees = trees[:keep]
     if dart:
         # bake each tree's final absolute scale into HostTree.scale
-        # (the learner builds the flat forest with leaf_scale=1.0)
-        for t, sc in zip(trees, dart_scale):
+        # (the learner builds the flat forest with leaf_scale=1.0);
+        # dart_scale is per iteration -> repeat per class
+        per_tree = [s for s in dart_scale for _ in range(C)]
+        for t, sc in zip(trees, per_tree):
             t.scale = float(sc)
     return trees, init_preds, logs
 
