@@ -294,3 +294,64 @@ def test_embed_routing_categorical_and_na(tmp_path):
         p = lib.cn_predict_c(
             row.ctypes.data_as(ctypes.POINTER(ctypes.c_float)))
         assert abs(p - ref[i]) < 1e-5, (i, p, ref[i])
+
+
+def test_to_js_codegen(binary_data, tmp_path):
+    """JavaScript codegen (capability analogue of the reference JS port,
+    port/javascript WASM inference): executed under node with
+    prediction parity against model.predict."""
+    import shutil
+
+    if shutil.which("node") is None:
+        pytest.skip("node not available")
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=20,
+                                        validation_ratio=0).train(
+                                            binary_data)
+    src = ydf.to_js(m, "gen")
+    js = tmp_path / "model.js"
+    X = m._encode_features(binary_data)
+    idx = list(range(0, X.shape[1], 401))
+    rows = [[float(v) for v in X[:, i]] for i in idx]
+    driver = (src + "\nconst m = module.exports;\n"
+              + f"const rows = {rows!r};\n"
+              + "for (const r of rows) console.log(m.predict(r));\n")
+    js.write_text(driver)
+    r = subprocess.run(["node", str(js)], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    got = np.array([float(s) for s in r.stdout.split()])
+    want = m.predict(binary_data, device="cpu")[idx]
+    np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)
+
+
+def test_to_js_multiclass_and_categorical(tmp_path):
+    import shutil
+
+    if shutil.which("node") is None:
+        pytest.skip("node not available")
+    rng = np.random.RandomState(5)
+    n = 4000
+    cat = rng.choice(["a", "b", "c", "d", "e"], n)
+    x = rng.randn(n).astype(np.float32)
+    y = np.where(np.isin(cat, ["a", "c"]), "u",
+                 np.where(x > 0, "v", "w"))
+    d = {"cat": cat, "x": x, "label": y}
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=12,
+                                        validation_ratio=0).train(d)
+    src = ydf.to_js(m, "mc")
+    X = m._encode_features(d)
+    idx = list(range(0, n, 307))
+    rows = [[float(v) for v in X[:, i]] for i in idx]
+    driver = (src + "\nconst m = module.exports;\n"
+              + f"const rows = {rows!r};\n"
+              + "for (const r of rows) "
+              "console.log(m.predict(r).join(','));\n")
+    js = tmp_path / "mc.js"
+    js.write_text(driver)
+    r = subprocess.run(["node", str(js)], capture_output=True,
+                       text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    got = np.array([[float(v) for v in line.split(",")]
+                    for line in r.stdout.strip().splitlines()])
+    want = m.predict(d, device="cpu")[idx]
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)

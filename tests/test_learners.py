@@ -1028,3 +1028,37 @@ def test_dart_multiclass_and_na():
         dart_dropout=0.1, validation_ratio=0.0,
         missing_value_policy="LOCAL_IMPUTATION").train(d2)
     assert m2.evaluate(d2).accuracy > 0.8
+
+
+def test_hyperparameter_specification_matches_reference_defaults():
+    """Spec generated FROM the signatures (reference
+    GetGenericHyperParameterSpecification analogue) pins the Appendix-A
+    names/defaults — one source of truth, no drift."""
+    from ydf_amd.learner.generic_learner import (
+        hyperparameter_specification)
+
+    spec = hyperparameter_specification(ydf.GradientBoostedTreesLearner)
+    # reference proto defaults (SURVEY.md Appendix A)
+    pins = {
+        "num_trees": 300, "max_depth": 6, "shrinkage": 0.1,
+        "subsample": 1.0, "min_examples": 5, "l2_regularization": 0.0,
+        "l1_regularization": 0.0, "validation_ratio": 0.1,
+        "early_stopping_num_trees_look_ahead": 30,
+        "l2_categorical_regularization": 1.0,
+        "sparse_oblique_num_projections_exponent": 2.0,
+        "sparse_oblique_max_num_projections": 6000,
+        "goss_alpha": 0.2, "goss_beta": 0.1,
+        "dart_dropout": 0.01, "random_seed": 123456,
+        "num_candidate_attributes_ratio": -1.0,
+    }
+    for k, v in pins.items():
+        assert k in spec, f"missing hyperparameter {k}"
+        assert spec[k]["default"] == v, (k, spec[k]["default"], v)
+
+    rf = hyperparameter_specification(ydf.RandomForestLearner)
+    for k, v in {"num_trees": 300, "bootstrap_size_ratio": 1.0,
+                 "winner_take_all": True, "max_depth": 16}.items():
+        assert rf[k]["default"] == v, (k, rf[k])
+
+    if_spec = hyperparameter_specification(ydf.IsolationForestLearner)
+    assert if_spec["subsample_count"]["default"] == 256

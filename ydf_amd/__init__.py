@@ -48,7 +48,7 @@ from ydf_amd.serving.deploy import to_docker
 from ydf_amd.utils import usage
 from ydf_amd.utils.folds import fold_splits, generate_folds
 from ydf_amd.utils.registry import get_learner
-from ydf_amd.serving.embed import to_cpp, to_java
+from ydf_amd.serving.embed import to_cpp, to_java, to_js
 from ydf_amd.learner.extras import (
     BackwardSelectionFeatureSelector,
     MultitaskerLearner,

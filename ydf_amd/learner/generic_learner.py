@@ -414,3 +414,39 @@ class GenericLearner:
 
     def validate_hyperparameters(self) -> None:
         pass
+
+
+def hyperparameter_specification(learner_cls) -> dict:
+    """Machine-readable hyper-parameter spec generated from the learner
+    signature (reference GetGenericHyperParameterSpecification,
+    abstract_learner.h:126 — the spec the reference uses to generate
+    PYDF signatures and docs; here the signature IS the source and the
+    spec is derived from it, eliminating signature/spec drift).
+
+    Returns {name: {"type": "int|float|str|bool|...", "default": v}}.
+    """
+    import inspect
+
+    out = {}
+    for cls in reversed(learner_cls.__mro__):
+        if cls is object:
+            continue
+        try:
+            sig = inspect.signature(cls.__init__)
+        except (TypeError, ValueError):
+            continue
+        for name, p in sig.parameters.items():
+            if name in ("self", "args", "kwargs", "label", "task",
+                        "features", "tuner", "device"):
+                continue
+            if p.kind in (inspect.Parameter.VAR_POSITIONAL,
+                          inspect.Parameter.VAR_KEYWORD):
+                continue
+            default = None if p.default is inspect.Parameter.empty \
+                else p.default
+            out[name] = {
+                "type": type(default).__name__
+                if default is not None else "optional",
+                "default": default,
+            }
+    return out

@@ -1337,6 +1337,16 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   const size_t budget = 160 * 1024 - (lds_map ? map_bytes_full : 0);
   int max_lds_slots = (int)(budget / ((size_t)n_bins * 16));
   if (max_lds_slots < 1) max_lds_slots = 1;
+  // Optional cap: smaller slot groups shrink per-block LDS (e.g. a
+  // 32-slot level at 16 B/bin needs 131 KB -> 1 WG/CU; capping at 16
+  // gives 64 KB -> 2 WG/CU at the cost of a second filtered pass).
+  static int max_slots_env = -1;
+  if (max_slots_env < 0) {
+    const char* e = getenv("YDFA_HIST_MAX_SLOTS");
+    max_slots_env = e ? atoi(e) : 0;
+  }
+  if (max_slots_env > 0 && max_lds_slots > max_slots_env)
+    max_lds_slots = max_slots_env;
   const int group = n_slots < max_lds_slots ? n_slots : max_lds_slots;
   // multi-feature blocks: when the whole level fits with room to spare,
   // each block histograms fpb features from ONE pass over the rows

@@ -407,3 +407,150 @@ def to_java(model, class_name: str = "YdfModel") -> str:
         lines.append("  }")
     lines.append("}")
     return "\n".join(lines) + "\n"
+
+
+def to_js(model, function_name: str = "ydfModel") -> str:
+    """Generates a standalone JavaScript module for `model` — the
+    capability analogue of the reference's JavaScript port
+    (port/javascript: WASM inference over the same model format; here
+    the model is compiled to dependency-free JS with routing tables).
+
+    Exports (CommonJS + browser global):
+      <name>_predict(features)        binary/regression, features =
+                                      array in data-spec feature order
+                                      (categorical = vocab index)
+      <name>_predictMulti(features)   multi-class -> array of probs
+      <name>_features                 feature name/type metadata
+    """
+    f = model.forest
+    _check_embeddable(f)
+    feats = model.dataspec.feature_columns
+    C = model._n_outputs()
+    scale = model._leaf_scale()
+    n = f.n_nodes
+
+    def arr(name, vals, typ):
+        body = ",".join(str(v) for v in vals)
+        return f"const {name} = new {typ}([{body}]);"
+
+    lines = [
+        "// Generated by ydf_amd (MI355X-native decision forests).",
+        "// Feature order: " + ", ".join(
+            f"{i}:{c.name}" for i, c in enumerate(feats)),
+        "(function (root, factory) {",
+        "  if (typeof module === 'object' && module.exports) "
+        "{ module.exports = factory(); }",
+        f"  else {{ root.{function_name} = factory(); }}",
+        "}(typeof self !== 'undefined' ? self : this, function () {",
+        arr("FEAT", (int(v) for v in f.feat), "Int32Array"),
+        arr("LEFT", (int(v) for v in f.left), "Uint32Array"),
+        arr("THR", (float(v) for v in f.thr), "Float32Array"),
+        arr("CAT", (int(v) for v in f.cat_idx), "Int32Array"),
+        arr("ROOTS", (int(v) for v in f.roots), "Uint32Array"),
+    ]
+    if len(f.masks):
+        # JS lacks u64 literals in arrays pre-BigUint64; use 8 x u32
+        words = []
+        for m in np.asarray(f.masks, dtype=np.uint64).reshape(-1):
+            words.append(int(m) & 0xFFFFFFFF)
+            words.append(int(m) >> 32)
+        lines.append(arr("MASKS", words, "Uint32Array"))
+    if f.has_na_routing:
+        lines.append(arr("NA", (int(v) for v in f.na_right),
+                         "Uint8Array"))
+    if len(f.obl_ranges):
+        lines.append(arr("ORNG",
+                         (int(v) for v in
+                          np.asarray(f.obl_ranges).reshape(-1)),
+                         "Int32Array"))
+        lines.append(arr("OATTR", (int(v) for v in f.obl_attr),
+                         "Int32Array"))
+        lines.append(arr("OW", (float(v) for v in f.obl_w),
+                         "Float32Array"))
+    walk = [
+        "  function margin(fv, cls) {",
+        "    let acc = 0.0;",
+        f"    for (let t = cls; t < {f.n_trees}; t += "
+        f"{C if C > 1 else 1}) {{",
+        "      let node = ROOTS[t];",
+        "      for (;;) {",
+        "        const fi = FEAT[node];",
+        "        if (fi < 0) break;",
+        "        const c = CAT[node];",
+        "        let right;",
+    ]
+    if len(f.masks):
+        walk += [
+            "        if (c >= 0) {",
+            "          const v = fv[fi] < 0 ? 0 : (fv[fi] > 255 ? 255 "
+            ": Math.floor(fv[fi]));",
+            "          const w = MASKS[8 * c + ((v >> 5) | 0)];",
+            "          right = ((w >>> (v & 31)) & 1) !== 0;",
+            "        } else",
+        ]
+    if len(f.obl_ranges):
+        walk += [
+            "        if (c <= -2) {",
+            "          const r = -(c + 2);",
+            "          let dot = 0.0;",
+            "          for (let q = 0; q < ORNG[2*r+1]; ++q)",
+            "            dot += OW[ORNG[2*r] + q] * "
+            "fv[OATTR[ORNG[2*r] + q]];",
+            "          right = dot > THR[node];",
+            "        } else",
+        ]
+    if f.has_na_routing:
+        walk += [
+            "        if (Number.isNaN(fv[fi])) {",
+            "          right = NA[node] !== 0;",
+            "        } else",
+        ]
+    walk += [
+        "        right = fv[fi] > THR[node];",
+        "        node = LEFT[node] + (right ? 1 : 0);",
+        "      }",
+        f"      acc += THR[node] * {scale!r};",
+        "    }",
+        "    return acc;",
+        "  }",
+    ]
+    lines += walk
+    init0 = float(model.init_predictions[0])
+    if C == 1:
+        lines.append("  function predict(fv) {")
+        lines.append(f"    const m = {init0!r} + margin(fv, 0);")
+        if model.activation == "sigmoid":
+            lines.append("    return 1.0 / (1.0 + Math.exp(-m));")
+        elif model.activation == "exp":
+            lines.append("    return Math.exp(m);")
+        else:
+            lines.append("    return m;")
+        lines.append("  }")
+    else:
+        inits = ", ".join(
+            repr(float(model.init_predictions[c]
+                       if c < len(model.init_predictions) else 0.0))
+            for c in range(C))
+        lines += [
+            "  function predict(fv) {",
+            f"    const out = [{inits}];",
+            f"    for (let c = 0; c < {C}; ++c) out[c] += margin(fv, c);",
+        ]
+        if model.activation == "softmax":
+            lines += [
+                "    const mx = Math.max.apply(null, out);",
+                "    let s = 0.0;",
+                f"    for (let c = 0; c < {C}; ++c) "
+                "{ out[c] = Math.exp(out[c] - mx); s += out[c]; }",
+                f"    for (let c = 0; c < {C}; ++c) out[c] /= s;",
+            ]
+        lines.append("    return out;")
+        lines.append("  }")
+    meta = ", ".join(
+        f"{{name: {c.name!r}, type: {c.semantic.name!r}}}"
+        for c in feats)
+    lines += [
+        f"  return {{ predict: predict, features: [{meta}] }};",
+        "}));",
+    ]
+    return "\n".join(lines) + "\n"
