@@ -175,3 +175,47 @@ def test_allow_na_conditions():
     assert acc_with > 0.999
     assert acc_with > acc_without + 0.05
     assert "x.is_na" in m.input_feature_names()
+
+
+def test_pluggable_filesystem(tmp_path):
+    """Filesystem registry (reference utils/filesystem.h pluggable FS):
+    a registered scheme:// backend serves dataset reads end-to-end."""
+    import io
+
+    import ydf_amd as ydf
+    from ydf_amd.utils import fs
+
+    csv = "x,label\n" + "\n".join(
+        f"{i * 0.1},{'a' if i % 2 else 'b'}" for i in range(200))
+
+    class MemFS:
+        files = {"bucket/data.csv": csv.encode()}
+
+        def open(self, path, mode="rb"):
+            data = self.files[path]
+            return io.BytesIO(data) if "b" in mode else io.StringIO(
+                data.decode())
+
+        def glob(self, pattern):
+            import fnmatch
+
+            return sorted(p for p in self.files
+                          if fnmatch.fnmatch(p, pattern))
+
+        def exists(self, path):
+            return path in self.files
+
+    fs.register_filesystem("mem", MemFS())
+    try:
+        assert fs.exists("mem://bucket/data.csv")
+        assert fs.glob_files("mem://bucket/*.csv") == \
+            ["mem://bucket/data.csv"]
+        ds = ydf.create_vertical_dataset("csv:mem://bucket/data.csv",
+                                         label="label")
+        assert ds.n_examples == 200
+        m = ydf.GradientBoostedTreesLearner(
+            label="label", num_trees=3, validation_ratio=0,
+            device="cpu").train("csv:mem://bucket/data.csv")
+        assert m.num_trees() == 3
+    finally:
+        fs._REGISTRY.pop("mem", None)

@@ -1062,3 +1062,28 @@ def test_hyperparameter_specification_matches_reference_defaults():
 
     if_spec = hyperparameter_specification(ydf.IsolationForestLearner)
     assert if_spec["subsample_count"]["default"] == 256
+
+
+def test_adaptive_work_keeps_tree_count(binary_data):
+    """adapt_*_for_maximum_training_duration (reference AdaptativeWork,
+    utils/adaptive_work.h:32): the sample shrinks so the forest keeps
+    its FULL tree count inside the budget, instead of truncating."""
+    import time
+
+    t0 = time.time()
+    m = ydf.RandomForestLearner(
+        label="label", num_trees=60, max_depth=10,
+        adapt_bootstrap_size_ratio_for_maximum_training_duration=True,
+        maximum_training_duration_seconds=2.0,
+        compute_oob_performances=False, device="cpu").train(binary_data)
+    took = time.time() - t0
+    assert m.num_trees() == 60, m.num_trees()
+    assert took < 20
+    assert m.evaluate(binary_data).accuracy > 0.85
+
+    m2 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=60, validation_ratio=0.0,
+        adapt_subsample_for_maximum_training_duration=True,
+        maximum_training_duration_seconds=2.0, device="cpu").train(
+        binary_data)
+    assert m2.num_trees() == 60

@@ -29,8 +29,10 @@ InputData = Union[dict, "pandas.DataFrame", str]  # noqa: F821
 def expand_sharded_paths(path: str):
     """Typed/sharded dataset paths (reference dataset/formats.proto:23-31 +
     utils/sharded_io.h ExpandInputShards): supports "csv:" prefixes,
-    "path@N" shard counts, comma lists and globs."""
-    import glob as globlib
+    "path@N" shard counts, comma lists and globs. Globs resolve through
+    the pluggable filesystem registry (utils/fs.py, reference
+    utils/filesystem.h), so scheme:// paths work end-to-end."""
+    from ydf_amd.utils.fs import glob_files
 
     fmt = "csv"
     if ":" in path and path.split(":", 1)[0] in (
@@ -48,7 +50,7 @@ def expand_sharded_paths(path: str):
             else:
                 out.extend(f"{base}-{i:05d}-of-{n:05d}" for i in range(n))
         elif any(ch in part for ch in "*?["):
-            out.extend(sorted(globlib.glob(part)))
+            out.extend(glob_files(part))
         else:
             out.append(part)
     return fmt, out
@@ -70,7 +72,9 @@ def _to_column_dict(data: InputData) -> Dict[str, np.ndarray]:
                 f"dataset format {fmt!r} not supported yet (ROADMAP)")
         import pandas as pd
 
-        frames = [pd.read_csv(p) for p in paths]
+        from ydf_amd.utils.fs import open_file
+
+        frames = [pd.read_csv(open_file(p, "rb")) for p in paths]
         df = frames[0] if len(frames) == 1 else pd.concat(
             frames, ignore_index=True)
         return _to_column_dict(df)

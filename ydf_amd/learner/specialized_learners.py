@@ -97,6 +97,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                  focal_loss_gamma: float = 2.0,
                  forest_extraction: str = "MART",
                  dart_dropout: float = 0.01,
+                 adapt_subsample_for_maximum_training_duration:
+                 bool = False,
                  split_axis: str = "AXIS_ALIGNED",
                  sparse_oblique_num_projections_exponent: float = 2.0,
                  sparse_oblique_max_num_projections: int = 6000,
@@ -150,6 +152,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             focal_loss_gamma=focal_loss_gamma,
             forest_extraction=forest_extraction,
             dart_dropout=dart_dropout,
+            adapt_subsample_for_maximum_training_duration=(
+                adapt_subsample_for_maximum_training_duration),
             split_axis=split_axis,
             sparse_oblique_num_projections_exponent=(
                 sparse_oblique_num_projections_exponent),
@@ -506,6 +510,8 @@ class GradientBoostedTreesLearner(GenericLearner):
                              "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"),
             dart_dropout=(hp.get("dart_dropout", 0.01)
                           if hp.get("forest_extraction") == "DART" else 0.0),
+            adapt_sample_for_duration=hp.get(
+                "adapt_subsample_for_maximum_training_duration", False),
             **obl,
         )
         t = trainer_lib.ForestTrainer(bins, labels, cfg,
@@ -702,6 +708,8 @@ class RandomForestLearner(GenericLearner):
                  uplift_treatment: Optional[str] = None,
                  uplift_split_score: str = "KULLBACK_LEIBLER",
                  uplift_min_examples_in_treatment: int = 5,
+                 adapt_bootstrap_size_ratio_for_maximum_training_duration:
+                 bool = False,
                  maximum_training_duration_seconds: float = -1.0,
                  split_axis: str = "AXIS_ALIGNED",
                  sparse_oblique_num_projections_exponent: float = 2.0,
@@ -737,6 +745,8 @@ class RandomForestLearner(GenericLearner):
                 uplift_min_examples_in_treatment),
             maximum_training_duration_seconds=(
                 maximum_training_duration_seconds),
+            adapt_bootstrap_size_ratio_for_maximum_training_duration=(
+                adapt_bootstrap_size_ratio_for_maximum_training_duration),
             split_axis=split_axis,
             sparse_oblique_num_projections_exponent=(
                 sparse_oblique_num_projections_exponent),
@@ -873,6 +883,9 @@ class RandomForestLearner(GenericLearner):
             num_candidate_features=ncand,
             max_duration_seconds=hp.get(
                 "maximum_training_duration_seconds", -1.0),
+            adapt_sample_for_duration=hp.get(
+                "adapt_bootstrap_size_ratio_for_maximum_training_duration",
+                False),
             honest=hp.get("honest", False),
             na_mode=(getattr(self, "missing_value_policy",
                              "GLOBAL_IMPUTATION") == "LOCAL_IMPUTATION"),

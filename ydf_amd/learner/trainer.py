@@ -74,6 +74,10 @@ class TrainerConfig:
     with_replacement: bool = True    # False: Bernoulli row subsampling
     num_candidate_features: int = 0  # 0 = all features
     max_duration_seconds: float = -1.0
+    # reference AdaptativeWork (utils/adaptive_work.h:32): shrink the
+    # per-tree sample so ALL num_trees trees fit the time budget,
+    # instead of truncating the forest
+    adapt_sample_for_duration: bool = False
     # honest trees (reference decision_tree.proto Honest message): tree
     # structure from one random half, leaf values re-estimated on the other
     oob_vi_permutations: int = 1
@@ -1310,17 +1314,31 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             if rk == fault_rank:
                 raise RuntimeError(
                     f"injected fault at iteration {it} (YDFA_FAULT_ITER)")
+        elapsed_it = _time.monotonic() - t_start
+        adapt_ratio = 1.0
+        if max_duration_seconds > 0 and it > start_iteration and \
+                cfg.adapt_sample_for_duration:
+            # reference adapt_subsample_for_maximum_training_duration
+            # (utils/adaptive_work.h:32): shrink this iteration's
+            # subsample so the remaining trees fit the budget
+            done = it - start_iteration
+            per_full = elapsed_it / max(done, 1)
+            remaining_t = max_duration_seconds - elapsed_it
+            remaining_n = n_iters - it
+            adapt_ratio = min(1.0, max(
+                0.02, remaining_t / max(remaining_n * per_full, 1e-9)))
         if max_duration_seconds > 0 and \
-                _time.monotonic() - t_start > max_duration_seconds:
+                elapsed_it > max_duration_seconds:
             if log:
                 log(f"maximum_training_duration reached at iteration {it}")
             break
         sample_mask = None
-        if cfg.sampling_method != "GOSS" and cfg.subsample < 1.0:
+        eff_sub = cfg.subsample * adapt_ratio
+        if cfg.sampling_method != "GOSS" and eff_sub < 1.0:
             sample_mask = (
                 torch.from_numpy(
                     trainer.rng.random_sample(N).astype(np.float32))
-                .to(dev) < cfg.subsample)
+                .to(dev) < eff_sub)
         custom_gh = None
         if custom_loss is not None:
             y_np = y.cpu().numpy()
@@ -1649,21 +1667,37 @@ def train_rf(trainer: ForestTrainer, log=None,
         oob_sum = torch.zeros((C, N), dtype=torch.float32, device=dev)
         oob_cnt = torch.zeros(N, dtype=torch.float32, device=dev)
     t_start = _time.monotonic()
+    adapt_ratio = 1.0
+    adapt_spent = 0.0  # sum of ratios already trained
     for it in range(cfg.num_trees):
+        elapsed = _time.monotonic() - t_start
         if cfg.max_duration_seconds > 0 and it > 0 and \
-                _time.monotonic() - t_start > cfg.max_duration_seconds:
-            # reference AdaptativeWork (utils/adaptive_work.h) shrinks
-            # the per-tree sample to fit the budget; we stop adding
-            # trees instead (every grown tree is full-quality)
+                cfg.adapt_sample_for_duration:
+            # reference AdaptativeWork (utils/adaptive_work.h:32):
+            # estimate the full-sample per-tree cost from the work done
+            # so far and shrink this tree's bootstrap so the REMAINING
+            # trees fit the remaining budget (the forest keeps its full
+            # tree count; late trees see smaller samples)
+            per_full = elapsed / max(adapt_spent, 1e-9)
+            remaining_t = cfg.max_duration_seconds - elapsed
+            remaining_n = cfg.num_trees - it
+            adapt_ratio = min(1.0, max(
+                0.02, remaining_t / max(remaining_n * per_full, 1e-9)))
+        if cfg.max_duration_seconds > 0 and it > 0 and \
+                elapsed > cfg.max_duration_seconds:
+            # without adaptation (or if even 2% samples overrun): stop
+            # adding trees (every grown tree is full-quality)
             if log:
                 log(f"maximum_training_duration reached after {it} trees")
             break
+        adapt_spent += adapt_ratio
         weights = None
         if cfg.bootstrap:
             # Poisson bootstrap, clipped at 15 (P < 1e-12) — the packed
             # u64 histogram path requires per-example h <= 16
             weights = rf_bootstrap_weights(cfg.seed, it, N, dev,
-                                           cfg.bootstrap_ratio,
+                                           cfg.bootstrap_ratio
+                                           * adapt_ratio,
                                            cfg.with_replacement)
         if trainer.weights is not None:
             # user example weights compose with the bootstrap draw counts
