@@ -140,3 +140,36 @@ def test_golden_margin_abalone_regression():
     # training-set RMSE of a fresh 100-tree model must at least match
     # the golden model's (which saw a train/test split)
     assert rmse < rmse_golden * 1.10, (rmse, rmse_golden)
+
+
+def test_mhld_oblique_beats_axis_aligned_on_rotated_data():
+    """MHLD oblique (reference oblique.h:33, greedy LDA subsets): on a
+    rotated decision boundary the LDA projection recovers the
+    separating direction that axis-aligned stumps cannot express."""
+    rng = np.random.RandomState(21)
+    n = 6000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    # boundary along x1 + x2 (45 degrees)
+    y = np.where(x1 + x2 + 0.15 * rng.randn(n) > 0, "p", "q")
+    d = {"x1": x1, "x2": x2, "label": y}
+    kw = dict(label="label", num_trees=4, max_depth=2,
+              validation_ratio=0.0, device="cpu")
+    m_axis = ydf.GradientBoostedTreesLearner(**kw).train(d)
+    m_mhld = ydf.GradientBoostedTreesLearner(
+        split_axis="MHLD_OBLIQUE", **kw).train(d)
+    acc_a = m_axis.evaluate(d).accuracy
+    acc_m = m_mhld.evaluate(d).accuracy
+    assert acc_m > 0.95, acc_m
+    assert acc_m > acc_a + 0.03, (acc_m, acc_a)
+    # oblique conditions present + model round-trips
+    assert len(m_mhld.forest.obl_ranges) > 0
+    p1 = m_mhld.predict(d)
+    import tempfile
+
+    td = tempfile.mkdtemp()
+    m_mhld.save(td)
+    import ydf_amd
+
+    m2 = ydf_amd.load_model(td)
+    np.testing.assert_allclose(p1, m2.predict(d), rtol=1e-5, atol=1e-6)

@@ -380,3 +380,201 @@ def train_gbt_exact(X: np.ndarray, y: np.ndarray,
         leaf_of_row = sp._last_node_of_row
         preds += shrinkage * t.thr[leaf_of_row].astype(np.float64)
     return trees, init
+
+
+class MhldSplitter(ExactSplitter):
+    """MHLD oblique splits (reference oblique.h:33-38 + oblique.cc
+    FindBestConditionMHLDObliqueTemplate): greedily grow a feature
+    subset (<= max_attributes); for each candidate subset the
+    projection coefficients come from Linear Discriminant Analysis on
+    the node's examples (w = Sw^-1 (mu1 - mu0) for the binary case),
+    and the threshold from the regular numerical scan over the
+    projected values. Classification only, like the reference
+    (oblique.cc:690)."""
+
+    def __init__(self, X, cat_flags, y01, max_attributes: int = 4,
+                 **kw):
+        super().__init__(X, cat_flags, **kw)
+        self.y01 = np.asarray(y01, dtype=bool)
+        self.max_attributes = max_attributes
+        self.num_feats = [f for f in range(self.F)
+                          if not self.cat_flags[f]]
+
+    def _lda_direction(self, rows, feats):
+        Xs = self.X[np.ix_(feats, rows)].astype(np.float64)  # [k, n]
+        y = self.y01[rows]
+        if y.all() or (~y).any() == 0 or not y.any():
+            return None
+        mu0 = Xs[:, ~y].mean(axis=1)
+        mu1 = Xs[:, y].mean(axis=1)
+        d0 = Xs[:, ~y] - mu0[:, None]
+        d1 = Xs[:, y] - mu1[:, None]
+        sw = d0 @ d0.T + d1 @ d1.T
+        sw[np.diag_indices_from(sw)] += 1e-6 * max(1.0, np.trace(sw))
+        try:
+            w = np.linalg.solve(sw, mu1 - mu0)
+        except np.linalg.LinAlgError:
+            return None
+        nrm = np.linalg.norm(w)
+        if not np.isfinite(nrm) or nrm == 0:
+            return None
+        return w / nrm
+
+    def _best_threshold(self, proj, g, h):
+        """Best (gain, threshold) of a 1-D exact scan (sorted values)."""
+        o = np.argsort(proj, kind="stable")
+        sv = proj[o]
+        cg = np.cumsum(g[o])
+        ch = np.cumsum(h[o])
+        G, H = cg[-1], ch[-1]
+        n = len(o)
+        pos = np.arange(1, n)
+        valid = sv[:-1] != sv[1:]
+        CL = pos
+        CR = n - pos
+        GL, HL = cg[:-1], ch[:-1]
+        GR, HR = G - GL, H - HL
+        ok = valid & (CL >= self.min_examples) & \
+            (CR >= self.min_examples) & (HL >= self.min_hessian) & \
+            (HR >= self.min_hessian)
+        if not ok.any():
+            return -np.inf, 0.0
+        with np.errstate(divide="ignore", invalid="ignore"):
+            gain = GL * GL / (HL + self.lambda_l2) \
+                + GR * GR / (HR + self.lambda_l2) \
+                - G * G / (H + self.lambda_l2)
+        gain = np.where(ok & np.isfinite(gain), gain, -np.inf)
+        i = int(np.argmax(gain))
+        return float(gain[i]), float(_mid_threshold(
+            np.float32(sv[i]), np.float32(sv[i + 1])))
+
+    def grow_tree(self, g, h) -> ExactTree:
+        """Greedy MHLD growth per node (recursive; node sizes shrink
+        geometrically so the per-node LDA solves stay cheap)."""
+        feat: list = [-1]
+        thr: list = [0.0]
+        left: list = [0]
+        cover: list = [float(self.N)]
+        gains: list = [0.0]
+        oblique: dict = {}
+        node_of_row = np.zeros(self.N, dtype=np.int64)
+
+        def split_node(node, rows, depth):
+            if depth >= self.max_depth or \
+                    rows.size < 2 * self.min_examples:
+                return
+            # greedy subset growth
+            subset: list = []
+            best = (-np.inf, None, None)  # gain, w, thr
+            improved = True
+            while improved and len(subset) < self.max_attributes:
+                improved = False
+                for f in self.num_feats:
+                    if f in subset:
+                        continue
+                    cand = subset + [f]
+                    w = self._lda_direction(rows, cand)
+                    if w is None:
+                        continue
+                    proj = (w[None, :] @ self.X[np.ix_(
+                        cand, rows)].astype(np.float64))[0]
+                    gain, t = self._best_threshold(
+                        proj.astype(np.float32), g[rows], h[rows])
+                    if gain > best[0]:
+                        best = (gain, (list(cand), w.copy()), t)
+                        chosen = f
+                        improved = True
+                if improved:
+                    subset.append(chosen)
+            if best[1] is None or best[0] <= self.min_gain:
+                return
+            attrs, w = best[1]
+            t = best[2]
+            proj = (w[None, :] @ self.X[np.ix_(
+                attrs, rows)].astype(np.float64))[0]
+            go_right = proj > t
+            if go_right.all() or not go_right.any():
+                return
+            li = len(feat)
+            feat[node] = attrs[0]
+            left[node] = li
+            thr[node] = t
+            gains[node] = float(best[0])
+            oblique[node] = (np.asarray(attrs, dtype=np.int32),
+                             w.astype(np.float32), float(t))
+            for _ in range(2):
+                feat.append(-1)
+                thr.append(0.0)
+                left.append(0)
+                gains.append(0.0)
+                cover.append(0.0)
+            lrows = rows[~go_right]
+            rrows = rows[go_right]
+            node_of_row[lrows] = li
+            node_of_row[rrows] = li + 1
+            cover[li] = float(lrows.size)
+            cover[li + 1] = float(rrows.size)
+            split_node(li, lrows, depth + 1)
+            split_node(li + 1, rrows, depth + 1)
+
+        split_node(0, np.arange(self.N), 0)
+        # leaf values
+        for n in range(len(feat)):
+            if feat[n] < 0:
+                rows = np.nonzero(node_of_row == n)[0] if n else None
+                if n == 0 and len(feat) == 1:
+                    rows = np.arange(self.N)
+                if rows is None or rows.size == 0:
+                    continue
+                G = g[rows].sum()
+                H = h[rows].sum()
+                if H != 0.0:
+                    thr[n] = float(-G / (H + self.lambda_l2))
+        self._last_node_of_row = node_of_row
+        t = ExactTree(feat=np.asarray(feat, dtype=np.int32),
+                      thr=np.asarray(thr, dtype=np.float32),
+                      left=np.asarray(left, dtype=np.int32),
+                      cover=np.asarray(cover, dtype=np.float32),
+                      cat_mask={}, gain=np.asarray(gains,
+                                                   dtype=np.float32))
+        t.oblique = oblique
+        return t
+
+
+def mhld_trees_to_forest(trees, shrinkage: float):
+    """ExactTrees with .oblique records -> FlatForest with oblique
+    conditions (cat_idx <= -2 indexing obl_* arrays)."""
+    from ydf_amd.model.forest import FlatForest
+
+    feat, thr, left, roots, cover, cat_idx = [], [], [], [], [], []
+    obl_ranges, obl_attr, obl_w = [], [], []
+    base = 0
+    for t in trees:
+        roots.append(base)
+        n = len(t.feat)
+        tt = t.thr.copy()
+        leaf = t.feat < 0
+        tt[leaf] *= shrinkage
+        ci = np.full(n, -1, dtype=np.int32)
+        for node, (attrs, w, thr_v) in t.oblique.items():
+            ci[node] = -2 - len(obl_ranges)
+            obl_ranges.append((len(obl_attr), len(attrs)))
+            obl_attr.extend(int(a) for a in attrs)
+            obl_w.extend(float(v) for v in w)
+        feat.append(t.feat)
+        thr.append(tt)
+        left.append(np.where(t.feat >= 0, t.left + base, 0))
+        cover.append(t.cover)
+        cat_idx.append(ci)
+        base += n
+    return FlatForest(
+        feat=np.concatenate(feat), thr=np.concatenate(thr),
+        left=np.concatenate(left),
+        roots=np.asarray(roots, dtype=np.int32),
+        cat_idx=np.concatenate(cat_idx),
+        cover=np.concatenate(cover),
+        obl_ranges=np.asarray(obl_ranges, dtype=np.int32).reshape(-1, 2)
+        if obl_ranges else None,
+        obl_attr=np.asarray(obl_attr, dtype=np.int32)
+        if obl_attr else None,
+        obl_w=np.asarray(obl_w, dtype=np.float32) if obl_w else None)

@@ -106,6 +106,7 @@ class GradientBoostedTreesLearner(GenericLearner):
                  sparse_oblique_normalization: str = "NONE",
                  sparse_oblique_weights: str = "BINARY",
                  sparse_oblique_max_num_features: int = -1,
+                 mhld_oblique_max_num_attributes: int = 4,
                  loss: str = "DEFAULT",
                  discretize_numerical_columns: bool = True,
                  working_dir: Optional[str] = None,
@@ -165,6 +166,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             sparse_oblique_weights=sparse_oblique_weights,
             sparse_oblique_max_num_features=(
                 sparse_oblique_max_num_features),
+            mhld_oblique_max_num_attributes=(
+                mhld_oblique_max_num_attributes),
             loss=loss,
             working_dir=working_dir, resume_training=resume_training,
             resume_training_snapshot_interval_seconds=(
@@ -302,6 +305,59 @@ class GradientBoostedTreesLearner(GenericLearner):
             metadata={"loss": int(loss), "streaming_cache": True,
                       "missing_value_policy": "GLOBAL_IMPUTATION"})
 
+    def _train_mhld(self, data) -> GradientBoostedTreesModel:
+        """MHLD oblique splits (reference oblique.h:33 + oblique.cc
+        FindBestConditionMHLDObliqueTemplate): greedy LDA-projection
+        subsets per node. Runs on the CPU exact path (per-node LDA
+        solves; classification only, like the reference)."""
+        from ydf_amd.learner.exact import (MhldSplitter,
+                                           mhld_trees_to_forest)
+
+        hp = self.hyperparameters
+        if self._task != Task.CLASSIFICATION:
+            raise NotImplementedError(
+                "MHLD_OBLIQUE is classification-only "
+                "(reference oblique.cc:690); use SPARSE_OBLIQUE")
+        ds, bins, labels, bnd, cat_flags, weights, mono = self._prepare(
+            data, torch.device("cpu"))
+        classes = self._label_classes(ds)
+        if classes is None or len(classes) != 2:
+            raise NotImplementedError(
+                "MHLD_OBLIQUE supports binary classification")
+        if weights is not None or mono is not None:
+            raise NotImplementedError("MHLD_OBLIQUE: weights/monotonic "
+                                      "not supported")
+        X = np.ascontiguousarray(ds.X)
+        y = labels.cpu().numpy().astype(np.float64)
+        cf = cat_flags.cpu().numpy().astype(bool) \
+            if cat_flags is not None else None
+        p = np.clip(y.mean(), 1e-6, 1 - 1e-6)
+        init = float(np.log(p / (1 - p)))
+        preds = np.full(len(y), init)
+        sp = MhldSplitter(
+            X, cf, y01=y > 0.5,
+            max_attributes=hp.get("mhld_oblique_max_num_attributes", 4),
+            max_depth=hp["max_depth"], min_examples=hp["min_examples"],
+            min_hessian=hp["min_sum_hessian_in_leaf"],
+            lambda_l2=hp["l2_regularization"])
+        trees = []
+        for _ in range(hp["num_trees"]):
+            pr = 1.0 / (1.0 + np.exp(-preds))
+            g = pr - y
+            h = np.maximum(pr * (1.0 - pr), 1e-16)
+            t = sp.grow_tree(g, h)
+            trees.append(t)
+            preds += hp["shrinkage"] * t.thr[sp._last_node_of_row]
+        flat = mhld_trees_to_forest(trees, hp["shrinkage"])
+        return GradientBoostedTreesModel(
+            forest=flat, dataspec=ds.dataspec, task=self._task,
+            label_classes=classes, init_predictions=[init],
+            num_trees_per_iter=1,
+            activation="sigmoid" if hp["apply_link_function"]
+            else "identity",
+            metadata={"loss": 1, "mhld_oblique": True,
+                      "missing_value_policy": "GLOBAL_IMPUTATION"})
+
     def train(self, data, valid=None, verbose=None
               ) -> GradientBoostedTreesModel:
         from ydf_amd.dataset.cache import DatasetCache
@@ -310,6 +366,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             return self._train_streaming(data)
         if self.tuner is not None:
             return self._train_with_tuner(data, valid=valid)
+        if self.hyperparameters.get("split_axis") == "MHLD_OBLIQUE":
+            return self._train_mhld(data)
         if not self.discretize_numerical_columns:
             return self._train_exact(data)
         hp = self.hyperparameters
