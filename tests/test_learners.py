@@ -1087,3 +1087,17 @@ def test_adaptive_work_keeps_tree_count(binary_data):
         maximum_training_duration_seconds=2.0, device="cpu").train(
         binary_data)
     assert m2.num_trees() == 60
+
+
+def test_analysis_html_report(binary_data):
+    """Analysis HTML report with SVG charts (reference
+    model_analysis.h CreateHtmlReport + utils/plot.*)."""
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=15,
+                                        validation_ratio=0).train(
+        binary_data)
+    an = m.analyze(binary_data)
+    html = an._repr_html_()
+    assert "<svg" in html and "polyline" in html  # PDP line plots
+    assert "rect" in html                         # VI bar chart
+    assert "Partial dependence" in html
+    assert html.count("<svg") >= 4  # >=1 VI chart + 3 PDP panels

@@ -49,13 +49,101 @@ class Analysis:
         return self.to_text()
 
     def _repr_html_(self) -> str:
+        """Full HTML report: VI bar charts + per-feature PDP curves
+        (reference utils/model_analysis.h:36-89 CreateHtmlReport /
+        utils/plot.* — rendered as dependency-free inline SVG instead
+        of plotly)."""
         rows = []
         for name, vi in self.variable_importances.items():
             body = "".join(
                 f"<tr><td>{f}</td><td>{s:.6g}</td></tr>" for s, f in vi)
-            rows.append(f"<h3>{name}</h3><table><tr><th>feature</th>"
-                        f"<th>score</th></tr>{body}</table>")
+            rows.append(f"<h3>{name}</h3>{_svg_bar_chart(vi)}"
+                        f"<details><summary>table</summary>"
+                        f"<table><tr><th>feature</th>"
+                        f"<th>score</th></tr>{body}</table></details>")
+        if self.partial_dependences:
+            rows.append("<h3>Partial dependence</h3>"
+                        "<div style='display:flex;flex-wrap:wrap'>")
+            for pd in self.partial_dependences:
+                rows.append(
+                    f"<div style='margin:4px'>"
+                    f"<b style='font-size:12px'>{pd.feature}</b><br/>"
+                    + _svg_pdp(pd) + "</div>")
+            rows.append("</div>")
         return "".join(rows)
+
+
+def _svg_bar_chart(vi: List[Tuple[float, str]], width: int = 420,
+                   bar_h: int = 14, top: int = 12) -> str:
+    """Horizontal bar chart of the top variable importances."""
+    items = vi[:top]
+    if not items:
+        return ""
+    mx = max(abs(s) for s, _ in items) or 1.0
+    h = len(items) * (bar_h + 3) + 4
+    parts = [f'<svg width="{width}" height="{h}" '
+             'xmlns="http://www.w3.org/2000/svg" '
+             'style="font:10px sans-serif">']
+    for i, (s, f) in enumerate(items):
+        y = 2 + i * (bar_h + 3)
+        w = max(1.0, abs(s) / mx * (width - 190))
+        parts.append(
+            f'<rect x="150" y="{y}" width="{w:.1f}" height="{bar_h}" '
+            'fill="#4a7abc"/>')
+        parts.append(f'<text x="146" y="{y + bar_h - 3}" '
+                     f'text-anchor="end">{f[:24]}</text>')
+        parts.append(f'<text x="{152 + w:.1f}" y="{y + bar_h - 3}">'
+                     f'{s:.4g}</text>')
+    parts.append("</svg>")
+    return "".join(parts)
+
+
+def _svg_pdp(pd: PartialDependence, width: int = 220,
+             height: int = 120) -> str:
+    """One PDP panel: line plot (numerical) or bars (categorical)."""
+    ys = np.asarray(pd.mean_prediction, dtype=np.float64)
+    if ys.size == 0:
+        return ""
+    lo, hi = float(ys.min()), float(ys.max())
+    span = (hi - lo) or 1.0
+    pad = 14
+    ph = height - 2 * pad
+    pw = width - 2 * pad
+
+    def sy(v):
+        return pad + ph - (v - lo) / span * ph
+
+    parts = [f'<svg width="{width}" height="{height}" '
+             'xmlns="http://www.w3.org/2000/svg" '
+             'style="font:9px sans-serif;background:#fafafa">']
+    parts.append(f'<text x="2" y="10">{hi:.3g}</text>')
+    parts.append(f'<text x="2" y="{height - 2}">{lo:.3g}</text>')
+    n = ys.size
+    if pd.is_categorical:
+        bw = pw / max(n, 1)
+        for i, v in enumerate(ys):
+            x = pad + i * bw
+            parts.append(
+                f'<rect x="{x:.1f}" y="{sy(v):.1f}" '
+                f'width="{max(bw - 2, 1):.1f}" '
+                f'height="{pad + ph - sy(v):.1f}" fill="#4a7abc"/>')
+    else:
+        pts = " ".join(
+            f"{pad + i / max(n - 1, 1) * pw:.1f},{sy(v):.1f}"
+            for i, v in enumerate(ys))
+        parts.append(f'<polyline points="{pts}" fill="none" '
+                     'stroke="#c0392b" stroke-width="1.5"/>')
+        gx = np.asarray(pd.grid, dtype=np.float64)
+        if gx.size:
+            parts.append(f'<text x="{pad}" y="{height - 2}" '
+                         f'text-anchor="start"></text>')
+            parts.append(
+                f'<text x="{width - 2}" y="{height - 2}" '
+                f'text-anchor="end">{gx[-1]:.3g}</text>')
+            parts.append(
+                f'<text x="{pad}" y="{height - 2}">{gx[0]:.3g}</text>')
+    parts.append("</svg>")
+    return "".join(parts)
 
 
 def structure_importances(model) -> Dict[str, List[Tuple[float, str]]]:
