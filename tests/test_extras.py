@@ -227,7 +227,7 @@ def test_engine_api_cpu():
                                        seed=2)
     m = ydf.GradientBoostedTreesLearner(label="LABEL", num_trees=8,
                                         validation_ratio=0).train(d)
-    assert set(m.list_compatible_engines()) == {"flat", "8bit", "qs"}
+    assert set(m.list_compatible_engines()) == {"flat", "8bit", "binned8", "qs"}
     m.force_engine("qs")   # CPU predict falls back to the flat twin
     assert m.predict(d, device="cpu").shape == (1500,)
     with pytest.raises(ValueError):

@@ -53,7 +53,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--device", default=None)
     ap.add_argument("--engine", default="flat",
-                    choices=["flat", "qs", "both"])
+                    choices=["flat", "qs", "both", "binned8", "all"])
     args = ap.parse_args()
     device = torch.device(args.device) if args.device else (
         torch.device("cuda") if torch.cuda.is_available()
@@ -72,8 +72,26 @@ def main():
     out = torch.empty(args.rows, dtype=torch.float32, device=device)
     act = torch.empty_like(out)
 
+    b8 = None
+    if args.engine in ("binned8", "all") and device.type == "cuda":
+        from ydf_amd.model.forest import (FlatForest, pack_binned8_nodes,
+                                          padded_boundaries)
+
+        # quantile cut table from a sample; node thresholds map to bins
+        sample, _ = torch.sort(X[:, :1 << 18], dim=1)
+        m = sample.shape[1]
+        qi = torch.linspace(0, m - 1, 257, device=device)[1:-1].long()
+        bnd = sample[:, qi].contiguous()
+        bins = torch.empty(X.shape, dtype=torch.uint8, device=device)
+        from ydf_amd import ops as _ops
+
+        ff8 = FlatForest(feat=feat, thr=thr, left=left, roots=roots)
+        packed8 = torch.from_numpy(pack_binned8_nodes(
+            ff8, bnd.cpu().numpy())).to(device)
+        b8 = (bins, bnd, packed8)
+
     qs = None
-    if args.engine in ("qs", "both"):
+    if args.engine in ("qs", "both", "all"):
         from ydf_amd.model.forest import FlatForest, build_quickscorer
 
         ff = FlatForest(feat=feat, thr=thr, left=left, roots=roots)
@@ -90,9 +108,20 @@ def main():
         ops.predict_forest_qs(X, qs[0], qs[1], qs[2], out)
         ops.sigmoid(out, act)
 
-    engines = {"flat": run_flat} if args.engine == "flat" else (
-        {"qs": run_qs} if args.engine == "qs"
-        else {"flat": run_flat, "qs": run_qs})
+    def run_b8():
+        # binning included in the timed region (reads raw X per batch)
+        ops.bin_data(X, b8[1], b8[0])
+        ops.predict_forest_binned8(b8[0], b8[2], roots_d8, out)
+        ops.sigmoid(out, act)
+
+    roots_d8 = rootsd
+    engines = {}
+    if args.engine in ("flat", "both", "all"):
+        engines["flat"] = run_flat
+    if args.engine in ("qs", "both", "all"):
+        engines["qs"] = run_qs
+    if args.engine in ("binned8", "all") and b8 is not None:
+        engines["binned8"] = run_b8
     results = {}
     for name, run in engines.items():
         for _ in range(args.warmup):

@@ -441,3 +441,18 @@ def vecseq_project(values: torch.Tensor, offs: torch.Tensor,
 
     md, ns = project_numpy(values.numpy(), offs.numpy(), anchors.numpy())
     return torch.from_numpy(md), torch.from_numpy(ns)
+
+
+def predict_forest_binned8(B: torch.Tensor, packed8: torch.Tensor,
+                           roots: torch.Tensor, out: torch.Tensor,
+                           init: float = 0.0):
+    """Compact-node 8-bit engine (GPU): B [F,N] u8 pre-binned features;
+    packed8 [n_nodes, 2] u32 (pack_binned8_nodes output, leaf values
+    pre-scaled). Half the node bytes per visit of the 16-B engines."""
+    assert B.is_cuda
+    F, N = B.shape
+    _C.gpu_predict_forest_binned8(B.data_ptr(), N, F,
+                                  packed8.data_ptr(), roots.data_ptr(),
+                                  roots.numel(), out.data_ptr(), init,
+                                  1.0, _stream())
+    return out
