@@ -1101,3 +1101,31 @@ def test_analysis_html_report(binary_data):
     assert "rect" in html                         # VI bar chart
     assert "Partial dependence" in html
     assert html.count("<svg") >= 4  # >=1 VI chart + 3 PDP panels
+
+
+def test_best_first_categorical():
+    """BEST_FIRST_GLOBAL + categorical set-splits (round-1 exclusion
+    lifted): leaf-wise growth now scans categorical features host-side
+    from the pulled pair histogram."""
+    rng = np.random.RandomState(17)
+    n = 12000
+    cats = rng.randint(0, 10, n)
+    probs = np.array([0.9, 0.1, 0.85, 0.15, 0.9, 0.1, 0.8, 0.2,
+                      0.88, 0.12])
+    y = rng.rand(n) < probs[cats]
+    d = {"c": np.array([f"k{v}" for v in cats]),
+         "x": rng.randn(n).astype(np.float32),
+         "label": np.where(y, "p", "n")}
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=20, growing_strategy="BEST_FIRST_GLOBAL",
+        max_num_nodes=15, validation_ratio=0.0, device="cpu").train(d)
+    assert (m.forest.cat_idx >= 0).any(), "no categorical splits chosen"
+    assert m.evaluate(d).accuracy > 0.8
+    # round-trip
+    import tempfile
+
+    td = tempfile.mkdtemp()
+    m.save(td)
+    m2 = ydf.load_model(td)
+    np.testing.assert_allclose(m.predict(d), m2.predict(d), rtol=1e-5,
+                               atol=1e-6)

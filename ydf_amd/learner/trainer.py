@@ -140,7 +140,7 @@ class BestFirstTree:
     """Leaf-wise tree: splits in creation order over IMPLICIT keys
     (children 2k+1 / 2k+2; depth capped at 24 so keys fit int32)."""
 
-    splits: list                 # [(key, feat, bin, gain)]
+    splits: list    # [(key, feat, bin, gain, mask_u64x4_or_None)]
     leaf_value: dict             # {key: float} (unscaled -G/(H+l2))
     counts: dict                 # {key: float}
     max_depth: int = 24
@@ -479,22 +479,82 @@ class ForestTrainer:
         self.grow_tree_device(tree_idx, sample_mask)
         return self.extract_host_tree()
 
+    @staticmethod
+    def _host_cat_scan(hv: np.ndarray, cat_idx_list, fm_row, sp_l2,
+                       sp_smooth, min_ex, min_h, l1):
+        """Host CART categorical scan over a pulled histogram slot
+        (mirrors cpu_split_scan's sorted_bin_order + scan; used by the
+        best-first path where device masks cannot be key-indexed).
+        Returns (gain, feat, rank, mask u64[4]) or None."""
+        best = None
+        for f in cat_idx_list:
+            if fm_row is not None and not fm_row[f]:
+                continue
+            h = hv[f]  # [256, 3]
+            cnt = h[:, 2]
+            G, H = float(h[:, 0].sum()), float(h[:, 1].sum())
+            key = np.where(cnt > 0, h[:, 0] / (h[:, 1] + sp_smooth),
+                           1e30)
+            order = np.argsort(key, kind="stable")
+            og, oh, oc = (h[order, 0], h[order, 1], h[order, 2])
+            GL = np.cumsum(og)[:-1]
+            HL = np.cumsum(oh)[:-1]
+            CL = np.cumsum(oc)[:-1]
+            GR, HR, CR = G - GL, H - HL, cnt.sum() - CL
+            ok = (CL >= min_ex) & (CR >= min_ex) & (HL >= min_h) & \
+                (HR >= min_h)
+            if not ok.any():
+                continue
+            tl = GL if l1 <= 0 else np.sign(GL) * np.maximum(
+                np.abs(GL) - l1, 0)
+            tr = GR if l1 <= 0 else np.sign(GR) * np.maximum(
+                np.abs(GR) - l1, 0)
+            tp = G if l1 <= 0 else np.sign(G) * max(abs(G) - l1, 0)
+            with np.errstate(divide="ignore", invalid="ignore"):
+                gain = tl * tl / (HL + sp_l2) + tr * tr / (HR + sp_l2) \
+                    - tp * tp / (H + sp_l2)
+            gain = np.where(ok & np.isfinite(gain), gain, -np.inf)
+            r = int(np.argmax(gain))
+            g = float(gain[r])
+            if g <= 0 or (best is not None and g <= best[0]):
+                continue
+            mask = np.zeros(4, dtype=np.uint64)
+            for rank in range(r + 1, 256):
+                b = int(order[rank])
+                if cnt[b] > 0:
+                    mask[b >> 6] |= np.uint64(1 << (b & 63))
+            best = (g, int(f), r, mask)
+        return best
+
+    def _bf_route_cat(self, bins, node_ids, key, f, mask4):
+        """Routes rows at implicit key through a categorical set-split
+        (torch-level; device mask tables are abs-indexed and leaf-wise
+        keys outgrow them)."""
+        at = node_ids == key
+        b = bins[f].long()
+        mask_t = torch.from_numpy(mask4.view(np.int64)).to(bins.device)
+        bit = (mask_t[b >> 6] >> (b & 63)) & 1
+        node_ids.copy_(torch.where(
+            at, 2 * key + 1 + bit.to(torch.int32), node_ids))
+
     def grow_tree_best_first(self, tree_idx: int,
                              sample_mask=None) -> BestFirstTree:
         """Leaf-wise growth (reference BEST_FIRST_GLOBAL,
         decision_tree.proto growing_strategy): a host-driven loop pops
-        the open leaf with the best gain, splits it with the SAME
-        histogram/scan kernels (one node per launch, level_base = its
-        implicit key), until max_num_nodes leaves. Numerical + boolean
-        features only for now (categorical set-split masks index the
-        complete-tree buffer, which leaf-wise keys outgrow)."""
+        the open leaf with the best gain and splits it. After each
+        split BOTH children are histogrammed and scanned in ONE kernel
+        launch (level_base = left child's implicit key, 2 slots) — one
+        full-row pass and one host sync per split instead of two.
+        Numerical + boolean + categorical set-splits (categorical
+        scans run on the pulled histogram host-side: device mask
+        tables are abs-node-indexed, which leaf-wise keys outgrow)."""
         import heapq
 
         cfg = self.cfg
-        assert self.has_cats is False, \
-            "BEST_FIRST_GLOBAL + categorical set-splits not supported yet"
         assert self.P == 0, "BEST_FIRST_GLOBAL + oblique not supported"
         assert self.mono is None
+        assert not cfg.na_mode, \
+            "BEST_FIRST_GLOBAL + LOCAL_IMPUTATION not supported"
         if sample_mask is None:
             self.node_ids.zero_()
         else:
@@ -504,37 +564,66 @@ class ForestTrainer:
                                         device=self.device),
                             torch.full((), -1, dtype=torch.int32,
                                        device=self.device)))
-        abs0 = self.arange_buf[:1]           # value 0
+        cat_list = []
+        num_mask = None
+        if self.has_cats:
+            cf = self.cat_flags.cpu().numpy().astype(bool)
+            cat_list = list(np.nonzero(cf)[0])
+            num_mask = torch.from_numpy(
+                (~cf).astype(np.uint8)).to(self.device)
+        arange2 = self.arange_buf[:2]
         fbuf = torch.empty(1, dtype=torch.int32, device=self.device)
         bbuf = torch.empty(1, dtype=torch.int32, device=self.device)
-        stats = {}   # key -> (feat, bin, gain, G, H, C)
+        stats = {}   # key -> (feat, bin, gain, G, H, C, mask_or_None)
         nsc = 0
 
-        def scan_node(key: int):
+        def scan_keys(keys):
+            """Builds + scans 1 or 2 ADJACENT implicit keys in one
+            launch each."""
             nonlocal nsc
-            hist_view = self.hist[:1]
+            k = len(keys)
+            hist_view = self.hist[:k]
             hist_view.zero_()
-            ops.hist_build(self.bins, self.gh, self.node_ids, abs0,
-                           hist_view, key, 1, 0, 1)
+            ops.hist_build(self.bins, self.gh, self.node_ids,
+                           arange2[:k], hist_view, keys[0], k, 0, k)
             self._allreduce(hist_view)
-            fm = self._feat_mask(1, tree_idx, nsc)
+            fm = self._feat_mask(k, tree_idx, nsc)
             nsc += 1
-            ops.split_scan(hist_view, abs0, self.node_stats, self.bg_nf,
-                           self.bb_nf, self.best_feat, self.best_bin,
-                           self.best_gain, 0, 1, cfg.lambda_l2,
-                           cfg.min_hessian, cfg.min_examples, cfg.min_gain,
-                           feat_mask=fm, lambda_l1=cfg.lambda_l1)
-            assert not cfg.na_mode, \
-                "BEST_FIRST_GLOBAL + LOCAL_IMPUTATION not supported"
-            f = int(self.best_feat[0].item())
-            b = int(self.best_bin[0].item())
-            g = float(self.best_gain[0].item())
-            ns = self.node_stats[0]
-            G, H, C = (float(ns[0].item()), float(ns[1].item()),
-                       float(ns[2].item()))
-            stats[key] = (f, b, g, G, H, C)
+            gfm = fm
+            if num_mask is not None:
+                gfm = (fm if fm is not None else
+                       torch.ones((k, self.F), dtype=torch.uint8,
+                                  device=self.device)) * num_mask
+            ops.split_scan(hist_view, arange2[:k], self.node_stats,
+                           self.bg_nf, self.bb_nf, self.best_feat,
+                           self.best_bin, self.best_gain, 0, k,
+                           cfg.lambda_l2, cfg.min_hessian,
+                           cfg.min_examples, cfg.min_gain,
+                           feat_mask=gfm, lambda_l1=cfg.lambda_l1)
+            bf = self.best_feat[:k].cpu().numpy()
+            bb = self.best_bin[:k].cpu().numpy()
+            bg = self.best_gain[:k].cpu().numpy()
+            ns = self.node_stats[:k].cpu().numpy().reshape(k, 3)
+            hv = hist_view.cpu().numpy().reshape(
+                k, self.F, -1, 3) if cat_list else None
+            fm_np = fm.cpu().numpy() if fm is not None else None
+            for s, key in enumerate(keys):
+                f, b, g = int(bf[s]), int(bb[s]), float(bg[s])
+                mask = None
+                if cat_list:
+                    cb = self._host_cat_scan(
+                        hv[s], cat_list,
+                        fm_np[s] if fm_np is not None else None,
+                        cfg.lambda_l2, cfg.cat_smooth,
+                        cfg.min_examples, cfg.min_hessian,
+                        cfg.lambda_l1)
+                    if cb is not None and cb[0] > max(g, 0):
+                        g, f, b = cb[0], cb[1], cb[2]
+                        mask = cb[3]
+                stats[key] = (f, b, g, float(ns[s, 0]),
+                              float(ns[s, 1]), float(ns[s, 2]), mask)
 
-        scan_node(0)
+        scan_keys((0,))
         heap = []
         tie = 0
         if stats[0][0] >= 0 and stats[0][2] > 0:
@@ -546,22 +635,26 @@ class ForestTrainer:
             _, _, key = heapq.heappop(heap)
             if key >= (1 << 24):     # depth cap: keys stay int32-safe
                 continue
-            f, b, g, _, _, _ = stats[key]
-            splits.append((key, f, b, g))
-            fbuf.fill_(f)
-            bbuf.fill_(b)
-            ops.update_node_ids(self.bins, self.node_ids, abs0, fbuf,
-                                bbuf, key, 1)
+            f, b, g, _, _, _, mask = stats[key]
+            splits.append((key, f, b, g, mask))
+            if mask is not None:
+                self._bf_route_cat(self.bins, self.node_ids, key, f,
+                                   mask)
+            else:
+                fbuf.fill_(f)
+                bbuf.fill_(b)
+                ops.update_node_ids(self.bins, self.node_ids,
+                                    arange2[:1], fbuf, bbuf, key, 1)
+            scan_keys((2 * key + 1, 2 * key + 2))
             for child in (2 * key + 1, 2 * key + 2):
-                scan_node(child)
-                cf, _, cg, _, _, _ = stats[child]
-                if cf >= 0 and cg > 0:
+                cf_, _, cg, _, _, _, _ = stats[child]
+                if cf_ >= 0 and cg > 0:
                     heapq.heappush(heap, (-cg, tie, child))
                     tie += 1
             n_leaves += 1
         split_keys = {k for k, *_ in splits}
         leaf_value, counts = {}, {}
-        for key, (f, b, g, G, H, C) in stats.items():
+        for key, (f, b, g, G, H, C, _m) in stats.items():
             counts[key] = C
             if key not in split_keys:
                 leaf_value[key] = -G / (H + cfg.lambda_l2) if H > 0 else 0.0
@@ -1144,11 +1237,14 @@ class ForestTrainer:
             abs0 = self.arange_buf[:1]
             fbuf = torch.empty(1, dtype=torch.int32, device=self.device)
             bbuf = torch.empty(1, dtype=torch.int32, device=self.device)
-            for key, f, b, _ in self._bf_last.splits:
-                fbuf.fill_(f)
-                bbuf.fill_(b)
-                ops.update_node_ids(bins, node_ids, abs0, fbuf, bbuf,
-                                    key, 1)
+            for key, f, b, _, mask in self._bf_last.splits:
+                if mask is not None:
+                    self._bf_route_cat(bins, node_ids, key, f, mask)
+                else:
+                    fbuf.fill_(f)
+                    bbuf.fill_(b)
+                    ops.update_node_ids(bins, node_ids, abs0, fbuf,
+                                        bbuf, key, 1)
             self._bf_remap(node_ids)
             return
         return self._route_rows_levelwise(bins, node_ids, raw)

@@ -182,22 +182,31 @@ def best_first_tree_to_flat(tree, boundaries: np.ndarray,
                             leaf_scale: float = 1.0):
     """Converts a leaf-wise BestFirstTree (implicit-key splits dict) to
     flat arrays via adjacent-pair child allocation."""
-    split_of = {k: (f, b, g) for k, f, b, g in tree.splits}
-    feats, thrs, lefts, covers = [], [], [], []
+    split_of = {s[0]: s[1:] for s in tree.splits}
+    feats, thrs, lefts, covers, cidx = [], [], [], [], []
+    masks = []
 
     def new_slot():
         feats.append(-1)
         thrs.append(0.0)
         lefts.append(0)
         covers.append(0.0)
+        cidx.append(-1)
         return len(feats) - 1
 
     def fill(key, slot):
         covers[slot] = float(tree.counts.get(key, 0.0))
         if key in split_of:
-            fi, b, _ = split_of[key]
+            rec = split_of[key]
+            fi, b = rec[0], rec[1]
+            mask = rec[3] if len(rec) > 3 else None
             feats[slot] = fi
-            thrs[slot] = float(boundaries[fi, b])
+            if mask is not None:
+                cidx[slot] = len(masks)
+                masks.append(np.asarray(mask, dtype=np.uint64))
+                thrs[slot] = 0.0
+            else:
+                thrs[slot] = float(boundaries[fi, b])
             li = new_slot()
             new_slot()
             lefts[slot] = li
@@ -210,8 +219,9 @@ def best_first_tree_to_flat(tree, boundaries: np.ndarray,
     fill(0, root)
     return (np.asarray(feats, np.int32), np.asarray(thrs, np.float32),
             np.asarray(lefts, np.int32),
-            np.full(len(feats), -1, np.int32),
-            np.zeros((0, 4), np.uint64),
+            np.asarray(cidx, np.int32),
+            (np.stack(masks) if masks
+             else np.zeros((0, 4), np.uint64)),
             np.asarray(covers, np.float32),
             (np.zeros((0, 2), np.int32), np.zeros(0, np.int32),
              np.zeros(0, np.float32)))
