@@ -566,3 +566,26 @@ def test_binned8_engine_matches_flat(binary_data):
                                init=float(m.init_predictions[0]))
     got = torch.sigmoid(out)
     assert (got - want).abs().max().item() < 1e-5
+
+
+@pytest.mark.gpu
+def test_binned8_engine_multiclass():
+    """Compact-node engine with class-tree striding == default path."""
+    import torch
+
+    import ydf_amd as ydf
+
+    assert torch.cuda.is_available()
+    rng = np.random.RandomState(6)
+    n = 70000  # above the auto-select batch threshold
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = np.where(x1 > 0.5, "a", np.where(x2 > 0, "b", "c"))
+    d = {"x1": x1, "x2": x2, "label": y}
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=20,
+                                        validation_ratio=0).train(d)
+    p_auto = m.predict(d)           # auto-selects binned8 (large batch)
+    m.force_engine("flat")
+    p_flat = m.predict(d)
+    assert p_auto.shape == (n, 3)
+    np.testing.assert_allclose(p_auto, p_flat, rtol=1e-5, atol=1e-6)

@@ -212,11 +212,17 @@ class GenericModel:
             self._thr_on_cuts = False
             return False
         f = self.forest
+        internal = f.feat >= 0
+        feats = f.feat[internal]
+        thrs = f.thr[internal]
         ok = True
-        for n in np.nonzero(f.feat >= 0)[0]:
-            cuts = bnd[int(f.feat[n])]
-            i = int(np.searchsorted(cuts, f.thr[n]))
-            if i >= len(cuts) or cuts[i] != f.thr[n]:
+        for fi in np.unique(feats):
+            cuts = bnd[int(fi)]
+            t = thrs[feats == fi]
+            idx = np.searchsorted(cuts, t)
+            valid = idx < len(cuts)
+            if not valid.all() or not np.array_equal(
+                    cuts[idx[valid]], t[valid]):
                 ok = False
                 break
         self._thr_on_cuts = bool(ok)
@@ -250,7 +256,7 @@ class GenericModel:
     def predict_margin(self, X: torch.Tensor) -> torch.Tensor:
         """Raw per-output forest sums/means. X [F,N] f32 on any device."""
         eng = getattr(self, "_engine", None)
-        if eng is None and X.is_cuda and self._n_outputs() == 1 \
+        if eng is None and X.is_cuda \
                 and len(self.forest.masks) == 0 \
                 and len(self.forest.obl_ranges) == 0 \
                 and not self.forest.has_na_routing \
@@ -260,7 +266,9 @@ class GenericModel:
             # profiles/serving_engines_r02.md) and bit-equivalent when
             # thresholds sit on training cuts
             eng = "binned8"
-        if eng in ("qs", "8bit", "binned8") and X.is_cuda \
+        if eng == "binned8" and X.is_cuda:
+            return self._predict_margin_binned8(X)
+        if eng in ("qs", "8bit") and X.is_cuda \
                 and self._n_outputs() == 1:
             return self._predict_margin_engine(X, eng)
         df = self._forest_on(X.device)
@@ -306,23 +314,6 @@ class GenericModel:
             ops.predict_forest_qs(X, c, o, lv, out[0], init=init,
                                   scale=scale)
             return out
-        if eng == "binned8":
-            from ydf_amd.model.forest import pack_binned8_nodes
-
-            if key not in cache:
-                bnd = padded_boundaries(self.dataspec.feature_columns)
-                cache[key] = (
-                    torch.from_numpy(pack_binned8_nodes(
-                        self.forest, bnd,
-                        leaf_scale=scale)).to(dev),
-                    torch.from_numpy(self.forest.roots).to(dev),
-                    torch.from_numpy(bnd).to(dev))
-            packed8, roots, bnd_t = cache[key]
-            bins = torch.empty(X.shape, dtype=torch.uint8, device=dev)
-            ops.bin_data(X, bnd_t, bins)
-            ops.predict_forest_binned8(bins, packed8, roots, out[0],
-                                       init=init)
-            return out
         if key not in cache:
             bnd = padded_boundaries(self.dataspec.feature_columns)
             cache[key] = (
@@ -335,6 +326,40 @@ class GenericModel:
         ops.bin_data(X, bnd_t, bins)
         ops.predict_forest_binned(bins, packed, roots, out[0], init=init,
                                   scale=scale)
+        return out
+
+    def _predict_margin_binned8(self, X: torch.Tensor) -> torch.Tensor:
+        """Compact-node binned engine (multi-output capable: class
+        trees stride through the shared node table)."""
+        from ydf_amd.model.forest import (pack_binned8_nodes,
+                                          padded_boundaries)
+
+        dev = X.device
+        key = f"binned8:{dev}"
+        cache = self._dev_forest
+        if key not in cache:
+            bnd = padded_boundaries(self.dataspec.feature_columns)
+            cache[key] = (
+                torch.from_numpy(pack_binned8_nodes(
+                    self.forest, bnd,
+                    leaf_scale=self._leaf_scale())).to(dev),
+                torch.from_numpy(self.forest.roots).to(dev),
+                torch.from_numpy(bnd).to(dev))
+        packed8, roots, bnd_t = cache[key]
+        C = self._n_outputs()
+        N = X.shape[1]
+        out = torch.empty((C, N), dtype=torch.float32, device=dev)
+        bins = torch.empty(X.shape, dtype=torch.uint8, device=dev)
+        ops.bin_data(X, bnd_t, bins)
+        T = self.forest.n_trees
+        for c in range(C):
+            ops.predict_forest_binned8(
+                bins, packed8, roots, out[c],
+                init=float(self.init_predictions[c]
+                           if c < len(self.init_predictions)
+                           else self.init_predictions[0]),
+                tree_start=c, tree_step=C if C > 1 else 1,
+                n_trees=T // C if C > 1 else T)
         return out
 
     def predict(self, data, device=None) -> np.ndarray:

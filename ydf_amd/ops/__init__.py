@@ -445,14 +445,18 @@ def vecseq_project(values: torch.Tensor, offs: torch.Tensor,
 
 def predict_forest_binned8(B: torch.Tensor, packed8: torch.Tensor,
                            roots: torch.Tensor, out: torch.Tensor,
-                           init: float = 0.0):
+                           init: float = 0.0, tree_start: int = 0,
+                           tree_step: int = 1, n_trees: int = -1):
     """Compact-node 8-bit engine (GPU): B [F,N] u8 pre-binned features;
     packed8 [n_nodes, 2] u32 (pack_binned8_nodes output, leaf values
-    pre-scaled). Half the node bytes per visit of the 16-B engines."""
+    pre-scaled). Half the node bytes per visit of the 16-B engines.
+    tree_start/tree_step stride class trees for multi-output models."""
     assert B.is_cuda
     F, N = B.shape
+    if n_trees < 0:
+        n_trees = roots.numel()
     _C.gpu_predict_forest_binned8(B.data_ptr(), N, F,
                                   packed8.data_ptr(), roots.data_ptr(),
-                                  roots.numel(), out.data_ptr(), init,
-                                  1.0, _stream())
+                                  tree_start, tree_step, n_trees,
+                                  out.data_ptr(), init, 1.0, _stream())
     return out

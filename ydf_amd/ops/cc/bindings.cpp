@@ -75,7 +75,7 @@ void gpu_vecseq_project(const float*, const int64_t*, const float*,
                         float*, float*, int64_t, int, int, void*);
 void gpu_predict_forest_binned8(const uint8_t*, int64_t, int,
                                 const uint32_t*, const int32_t*, int,
-                                float*, float, float, void*);
+                                int, int, float*, float, float, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
                   int);
@@ -362,11 +362,12 @@ PYBIND11_MODULE(_ydf_ops, m) {
         nogil);
   m.def("gpu_predict_forest_binned8",
         [](uintptr_t B, int64_t N, int F, uintptr_t nodes8,
-           uintptr_t roots, int n_trees, uintptr_t out, float init,
-           float scale, uintptr_t stream) {
+           uintptr_t roots, int tree_start, int tree_step, int n_trees,
+           uintptr_t out, float init, float scale, uintptr_t stream) {
           gpu_predict_forest_binned8(P<uint8_t>(B), N, F,
                                      P<uint32_t>(nodes8),
-                                     P<int32_t>(roots), n_trees,
+                                     P<int32_t>(roots), tree_start,
+                                     tree_step, n_trees,
                                      P<float>(out), init, scale,
                                      (void*)stream);
         },

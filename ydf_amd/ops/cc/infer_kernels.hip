@@ -271,7 +271,8 @@ struct Node8 {
 __global__ void predict_forest_binned8_kernel(
     const uint8_t* __restrict__ B, int64_t N, int F,
     const Node8* __restrict__ nodes, const int32_t* __restrict__ roots,
-    int n_trees, float* __restrict__ out, float init, float scale) {
+    int tree_start, int tree_step, int n_trees, float* __restrict__ out,
+    float init, float scale) {
   extern __shared__ uint8_t bs[];  // [F][kTile]
   const int tid = threadIdx.x;
   const int64_t base = (int64_t)blockIdx.x * kTile;
@@ -291,7 +292,8 @@ __global__ void predict_forest_binned8_kernel(
   for (; t + 4 <= n_trees; t += 4) {
     Node8 nd[4];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) nd[u] = nodes[roots[t + u]];
+    for (int u = 0; u < 4; ++u)
+      nd[u] = nodes[roots[tree_start + (int64_t)(t + u) * tree_step]];
     bool done = false;
     while (!done) {
       done = true;
@@ -309,7 +311,7 @@ __global__ void predict_forest_binned8_kernel(
     for (int u = 0; u < 4; ++u) acc += __uint_as_float(nd[u].left_or_val);
   }
   for (; t < n_trees; ++t) {
-    Node8 nd = nodes[roots[t]];
+    Node8 nd = nodes[roots[tree_start + (int64_t)t * tree_step]];
     while (nd.feat != 0xFFFFu) {
       const int right = (int)bs[nd.feat * kTile + tid] > (int)nd.bin;
       nd = nodes[nd.left_or_val + right];
@@ -398,7 +400,8 @@ void gpu_predict_forest_binned(const uint8_t* B, int64_t N, int F,
 
 void gpu_predict_forest_binned8(const uint8_t* B, int64_t N, int F,
                                 const uint32_t* nodes8,
-                                const int32_t* roots, int n_trees,
+                                const int32_t* roots, int tree_start,
+                                int tree_step, int n_trees,
                                 float* out, float init, float scale,
                                 void* stream) {
   const size_t lds = (size_t)F * kTile;
@@ -406,7 +409,7 @@ void gpu_predict_forest_binned8(const uint8_t* B, int64_t N, int F,
   hipLaunchKernelGGL(predict_forest_binned8_kernel, dim3(grid),
                      dim3(kTile), lds, (hipStream_t)stream, B, N, F,
                      reinterpret_cast<const Node8*>(nodes8), roots,
-                     n_trees, out, init, scale);
+                     tree_start, tree_step, n_trees, out, init, scale);
 }
 
 void gpu_predict_forest_qs(const float* X, int64_t N, int F,
