@@ -87,3 +87,32 @@ def test_synthesize_and_convert_dataset(tmp_path):
 
     m = ydf.load_model(str(tmp_path / "model"))
     assert m.num_trees() == 5
+
+
+def test_distribute_run(tmp_path):
+    """Batch command runner (reference utils/distribute_cli analogue):
+    N commands over K worker processes; exit code reflects failures."""
+    import subprocess
+    import sys
+
+    cmds = tmp_path / "cmds.txt"
+    out = tmp_path / "out"
+    out.mkdir()
+    cmds.write_text("\n".join(
+        f"{sys.executable} -c \"open(r'{out}/f{i}','w').write('x')\""
+        for i in range(6)))
+    r = subprocess.run(
+        [sys.executable, "-m", "ydf_amd.cli.distribute_run",
+         "--workers", "3", "--commands", str(cmds)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert sorted(p.name for p in out.iterdir()) == \
+        [f"f{i}" for i in range(6)]
+    # a failing command fails the batch
+    r = subprocess.run(
+        [sys.executable, "-m", "ydf_amd.cli.distribute_run",
+         "--workers", "2", "true", "false"],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 1

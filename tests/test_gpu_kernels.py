@@ -589,3 +589,34 @@ def test_binned8_engine_multiclass():
     p_flat = m.predict(d)
     assert p_auto.shape == (n, 3)
     np.testing.assert_allclose(p_auto, p_flat, rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_weighted_poisson_histogram_integrity():
+    """Advisor follow-up: weighted Poisson can reach per-example h*w
+    near the packed-u64 limit; the post-weight clamp must keep the
+    44-bit fixed-point h field sane. GPU and CPU models must agree in
+    quality (corruption would destroy one of them)."""
+    import torch
+
+    import ydf_amd as ydf
+
+    assert torch.cuda.is_available()
+    rng = np.random.RandomState(8)
+    n = 30000
+    x = rng.randn(n).astype(np.float32)
+    lam = np.exp(1.2 * x)
+    w = rng.uniform(0.5, 50.0, n).astype(np.float32)  # rescaled to max 8
+    d = {"x": x, "z": rng.randn(n).astype(np.float32),
+         "w": w, "label": rng.poisson(lam).astype(np.float32)}
+    kw = dict(label="label", task=ydf.Task.REGRESSION, loss="POISSON",
+              weights="w", num_trees=25, validation_ratio=0.0)
+    mg = ydf.GradientBoostedTreesLearner(**kw).train(d)
+    mc = ydf.GradientBoostedTreesLearner(device="cpu", **kw).train(d)
+    pg = mg.predict(d)
+    pc = mc.predict(d)
+    assert np.isfinite(pg).all() and (pg > 0).all()
+    # both runs clamp identically; quality must match closely
+    err_g = float(np.mean((pg - lam) ** 2))
+    err_c = float(np.mean((pc - lam) ** 2))
+    assert err_g < err_c * 1.3 + 1e-6, (err_g, err_c)
