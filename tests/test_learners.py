@@ -1129,3 +1129,21 @@ def test_best_first_categorical():
     m2 = ydf.load_model(td)
     np.testing.assert_allclose(m.predict(d), m2.predict(d), rtol=1e-5,
                                atol=1e-6)
+
+
+def test_custom_metrics_in_training_logs(binary_data):
+    """User-provided secondary metrics (PYDF custom_metric.py): values
+    recorded per validation interval in the training logs."""
+    def brier(labels, margins, weights):
+        p = 1.0 / (1.0 + np.exp(-margins))
+        return float(np.average((p - labels) ** 2, weights=weights))
+
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15, validation_ratio=0.2,
+        custom_metrics=[ydf.BinaryClassificationMetric(
+            "brier", brier)]).train(binary_data)
+    assert m.training_logs
+    vals = [d["brier"] for d in m.training_logs if "brier" in d]
+    assert len(vals) >= 10
+    assert vals[-1] < vals[0]  # boosting improves it
+    assert 0.0 < vals[-1] < 0.25

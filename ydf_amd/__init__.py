@@ -56,6 +56,13 @@ from ydf_amd.learner.extras import (
 )
 
 # Custom losses
+from ydf_amd.learner.custom_metric import (
+    AbstractCustomMetric,
+    BinaryClassificationMetric,
+    MultiClassificationMetric,
+    RegressionMetric,
+)
+from ydf_amd import experimental, help
 from ydf_amd.learner.custom_loss import (
     Activation,
     BinaryClassificationLoss,
@@ -166,3 +173,6 @@ class util:  # noqa: N801  (PYDF exposes a lowercase `util` namespace)
     generate_folds = staticmethod(_folds.generate_folds)
     fold_splits = staticmethod(_folds.fold_splits)
     usage = _usage
+
+version = "2.0.0+mi355x"
+__version__ = version

@@ -1,0 +1,44 @@
+"""ydf.help.* analogue: quick interactive documentation."""
+from __future__ import annotations
+
+
+def loading_data() -> str:
+    """How to feed data to learners (PYDF ydf.help.loading_data)."""
+    text = """Supported dataset inputs (everywhere a `data` argument is taken):
+  - dict of numpy arrays: {"feature": np.array(...), "label": ...}
+  - pandas DataFrame (or any object with .columns / __getitem__)
+  - typed paths: "csv:/path/file.csv", "tfrecord:...", "avro:...",
+    sharded "path@10", globs, comma lists; scheme:// paths resolve
+    through ydf_amd.utils.fs.register_filesystem backends
+  - ydf.create_vertical_dataset(...) for reuse across learners
+  - ydf.create_dataset_cache(...) for out-of-core (>HBM) training
+Cells may be: scalars (numerical/categorical/boolean), token sets
+(categorical-set), or [n_k, dim] vector sequences."""
+    print(text)
+    return text
+
+
+def learners() -> str:
+    """Lists the available learners."""
+    from ydf_amd.utils.registry import LEARNERS
+
+    text = "\n".join(sorted(LEARNERS)) if isinstance(LEARNERS, dict) \
+        else "GRADIENT_BOOSTED_TREES, RANDOM_FOREST, CART, " \
+             "ISOLATION_FOREST"
+    print(text)
+    return text
+
+
+def hyperparameters(learner_cls) -> str:
+    """Prints the hyper-parameter spec (names, types, defaults) of a
+    learner class — generated from its signature
+    (reference GetGenericHyperParameterSpecification analogue)."""
+    from ydf_amd.learner.generic_learner import (
+        hyperparameter_specification)
+
+    spec = hyperparameter_specification(learner_cls)
+    lines = [f"{name:45s} {d['type']:10s} default={d['default']!r}"
+             for name, d in sorted(spec.items())]
+    text = "\n".join(lines)
+    print(text)
+    return text

@@ -114,9 +114,11 @@ class GradientBoostedTreesLearner(GenericLearner):
                  resume_training_snapshot_interval_seconds: float = 1800.0,
                  maximum_training_duration_seconds: float = -1.0,
                  hyperparameter_template: Optional[str] = None,
+                 custom_metrics=None,
                  random_seed: int = 123456, **kwargs):
         super().__init__(label=label, task=task, features=features,
                          random_seed=random_seed, **kwargs)
+        self.custom_metrics = custom_metrics
         self.ranking_group = ranking_group
         self.label_event_observed = label_event_observed
         self.label_entry_age = label_entry_age
@@ -694,7 +696,8 @@ class GradientBoostedTreesLearner(GenericLearner):
             snapshot_interval_seconds=hp.get(
                 "resume_training_snapshot_interval_seconds", 1800.0),
             max_duration_seconds=hp.get(
-                "maximum_training_duration_seconds", -1.0))
+                "maximum_training_duration_seconds", -1.0),
+            custom_metrics=self.custom_metrics)
         if hp.get("forest_extraction") == "DART":
             # HostTree.scale holds each tree's final absolute leaf scale
             for tr in trees:

@@ -1279,7 +1279,7 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
               snapshot_interval_seconds: float = 1800.0,
               max_duration_seconds: float = -1.0,
               custom_loss=None, ranking=None, valid_ranking=None,
-              cox=None, valid_cox=None):
+              cox=None, valid_cox=None, custom_metrics=None):
     """The boosting loop (reference gradient_boosted_trees.cc:1460).
 
     Returns (trees, init_preds, training_logs). For multinomial loss,
@@ -1608,7 +1608,24 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
             else:
                 vloss = _eval_loss(trainer, valid_preds,
                                    trainer.valid_labels, cfg, loss_buf)
-            logs.append({"iteration": it + 1, "valid_loss": vloss})
+            entry = {"iteration": it + 1, "valid_loss": vloss}
+            if custom_metrics:
+                # user-provided secondary metrics on the validation set
+                # (PYDF custom_metric.py: evaluation_func(labels,
+                # predictions-without-activation, weights))
+                vy_np = trainer.valid_labels.cpu().numpy()
+                vp_np = (valid_preds[0] if C == 1
+                         else valid_preds).cpu().numpy()
+                wv = np.ones_like(vy_np, dtype=np.float32)
+                for cm in custom_metrics:
+                    try:
+                        entry[cm.name] = float(
+                            cm.evaluation_func(vy_np, vp_np, wv))
+                    except Exception as e:  # noqa: BLE001
+                        entry[cm.name] = float("nan")
+                        if log:
+                            log(f"custom metric {cm.name!r} failed: {e}")
+            logs.append(entry)
             if vloss < best_loss:
                 best_loss = vloss
                 best_num_trees = (it + 1) * C
