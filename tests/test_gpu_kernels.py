@@ -620,3 +620,34 @@ def test_weighted_poisson_histogram_integrity():
     err_g = float(np.mean((pg - lam) ** 2))
     err_c = float(np.mean((pc - lam) ** 2))
     assert err_g < err_c * 1.3 + 1e-6, (err_g, err_c)
+
+
+@pytest.mark.gpu
+def test_binned4_engine_matches_flat(binary_data):
+    """4-byte-node engine == flat engine on a trained model."""
+    import torch
+
+    import ydf_amd as ydf
+    from ydf_amd.model.forest import (pack_binned4_nodes,
+                                      padded_boundaries)
+
+    assert torch.cuda.is_available()
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=40, validation_ratio=0).train(
+        binary_data)
+    dev = torch.device("cuda:0")
+    X = torch.from_numpy(m._encode_features(binary_data)).to(dev)
+    want = torch.from_numpy(m.predict(binary_data)).to(dev)
+    bnd = padded_boundaries(m.dataspec.feature_columns)
+    n4, lv4 = pack_binned4_nodes(m.forest, bnd,
+                                 leaf_scale=m._leaf_scale())
+    bins = torch.empty(X.shape, dtype=torch.uint8, device=dev)
+    ops.bin_data(X, torch.from_numpy(bnd).to(dev), bins)
+    out = torch.empty(X.shape[1], dtype=torch.float32, device=dev)
+    ops.predict_forest_binned4(
+        bins, torch.from_numpy(n4).to(dev),
+        torch.from_numpy(lv4).to(dev),
+        torch.from_numpy(m.forest.roots).to(dev), out,
+        init=float(m.init_predictions[0]))
+    got = torch.sigmoid(out)
+    assert (got - want).abs().max().item() < 1e-5

@@ -53,7 +53,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--device", default=None)
     ap.add_argument("--engine", default="flat",
-                    choices=["flat", "qs", "both", "binned8", "all"])
+                    choices=["flat", "qs", "both", "binned8", "binned4",
+                             "all"])
     args = ap.parse_args()
     device = torch.device(args.device) if args.device else (
         torch.device("cuda") if torch.cuda.is_available()
@@ -73,7 +74,8 @@ def main():
     act = torch.empty_like(out)
 
     b8 = None
-    if args.engine in ("binned8", "all") and device.type == "cuda":
+    if args.engine in ("binned8", "binned4", "all") \
+            and device.type == "cuda":
         from ydf_amd.model.forest import (FlatForest, pack_binned8_nodes,
                                           padded_boundaries)
 
@@ -88,7 +90,12 @@ def main():
         ff8 = FlatForest(feat=feat, thr=thr, left=left, roots=roots)
         packed8 = torch.from_numpy(pack_binned8_nodes(
             ff8, bnd.cpu().numpy())).to(device)
-        b8 = (bins, bnd, packed8)
+        from ydf_amd.model.forest import pack_binned4_nodes
+
+        n4, lv4 = pack_binned4_nodes(ff8, bnd.cpu().numpy())
+        b8 = (bins, bnd, packed8,
+              torch.from_numpy(n4).to(device),
+              torch.from_numpy(lv4).to(device))
 
     qs = None
     if args.engine in ("qs", "both", "all"):
@@ -122,6 +129,13 @@ def main():
         engines["qs"] = run_qs
     if args.engine in ("binned8", "all") and b8 is not None:
         engines["binned8"] = run_b8
+    if args.engine in ("binned4", "all") and b8 is not None:
+        def run_b4():
+            ops.bin_data(X, b8[1], b8[0])
+            ops.predict_forest_binned4(b8[0], b8[3], b8[4], roots_d8,
+                                       out)
+            ops.sigmoid(out, act)
+        engines["binned4"] = run_b4
     results = {}
     for name, run in engines.items():
         for _ in range(args.warmup):

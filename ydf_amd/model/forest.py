@@ -483,3 +483,40 @@ def pack_binned8_nodes(forest: FlatForest, boundaries: np.ndarray,
                             forest.left.astype(np.uint32),
                             leaf_vals.view(np.uint32))
     return packed
+
+
+def pack_binned4_nodes(forest: FlatForest, boundaries: np.ndarray,
+                       leaf_scale: float = 1.0):
+    """4-byte nodes for the binned4 engine: feat (6 b, 63 = leaf) |
+    bin (8 b) | left-child-or-leaf-index (18 b); leaf values (scaled)
+    in a dense side table. Returns (nodes u32 [n], leaf_vals f32 [L]).
+    Limits: numerical-only, F <= 63, nodes and leaves <= 2^18."""
+    if len(forest.masks) or len(forest.obl_ranges) \
+            or forest.has_na_routing or forest.has_set_conditions:
+        raise ValueError("binned4 engine supports numerical conditions "
+                         "only")
+    n = forest.n_nodes
+    fmax = int(forest.feat.max(initial=-1))
+    if fmax >= 63 or n >= (1 << 18):
+        raise ValueError("binned4 limits exceeded (F < 63, nodes < 2^18)")
+    internal = forest.feat >= 0
+    leaves = np.nonzero(~internal)[0]
+    if len(leaves) >= (1 << 18):
+        raise ValueError("binned4: too many leaves")
+    leaf_idx = np.zeros(n, dtype=np.uint32)
+    leaf_idx[leaves] = np.arange(len(leaves), dtype=np.uint32)
+    leaf_vals = (forest.thr[leaves]
+                 * np.float32(leaf_scale)).astype(np.float32)
+    bins = np.zeros(n, dtype=np.uint32)
+    for node in np.nonzero(internal)[0]:
+        fi = int(forest.feat[node])
+        cuts = boundaries[fi]
+        b = int(np.searchsorted(cuts, forest.thr[node]))
+        if b >= len(cuts) or cuts[b] != forest.thr[node]:
+            b = max(int(np.searchsorted(cuts, forest.thr[node],
+                                        side="right")) - 1, 0)
+        bins[node] = b
+    feat_u = np.where(internal, forest.feat.astype(np.uint32), 63)
+    link = np.where(internal, forest.left.astype(np.uint32), leaf_idx)
+    nodes = feat_u | (bins << 6) | (link << 14)
+    return nodes.astype(np.uint32), leaf_vals

@@ -334,18 +334,29 @@ class GenericModel:
         from ydf_amd.model.forest import (pack_binned8_nodes,
                                           padded_boundaries)
 
+        from ydf_amd.model.forest import pack_binned4_nodes
+
         dev = X.device
         key = f"binned8:{dev}"
         cache = self._dev_forest
         if key not in cache:
             bnd = padded_boundaries(self.dataspec.feature_columns)
-            cache[key] = (
-                torch.from_numpy(pack_binned8_nodes(
+            try:
+                # 4-byte nodes when the forest fits the packing limits
+                # (F < 63, < 2^18 nodes/leaves) — measured slightly
+                # faster than the 8-byte form
+                n4, lv4 = pack_binned4_nodes(
+                    self.forest, bnd, leaf_scale=self._leaf_scale())
+                packed = (torch.from_numpy(n4).to(dev),
+                          torch.from_numpy(lv4).to(dev))
+            except ValueError:
+                packed = (torch.from_numpy(pack_binned8_nodes(
                     self.forest, bnd,
-                    leaf_scale=self._leaf_scale())).to(dev),
-                torch.from_numpy(self.forest.roots).to(dev),
-                torch.from_numpy(bnd).to(dev))
-        packed8, roots, bnd_t = cache[key]
+                    leaf_scale=self._leaf_scale())).to(dev),)
+            cache[key] = (packed,
+                          torch.from_numpy(self.forest.roots).to(dev),
+                          torch.from_numpy(bnd).to(dev))
+        packed, roots, bnd_t = cache[key]
         C = self._n_outputs()
         N = X.shape[1]
         out = torch.empty((C, N), dtype=torch.float32, device=dev)
@@ -353,13 +364,20 @@ class GenericModel:
         ops.bin_data(X, bnd_t, bins)
         T = self.forest.n_trees
         for c in range(C):
-            ops.predict_forest_binned8(
-                bins, packed8, roots, out[c],
-                init=float(self.init_predictions[c]
+            init_c = float(self.init_predictions[c]
                            if c < len(self.init_predictions)
-                           else self.init_predictions[0]),
-                tree_start=c, tree_step=C if C > 1 else 1,
-                n_trees=T // C if C > 1 else T)
+                           else self.init_predictions[0])
+            if len(packed) == 2:
+                ops.predict_forest_binned4(
+                    bins, packed[0], packed[1], roots, out[c],
+                    init=init_c, tree_start=c,
+                    tree_step=C if C > 1 else 1,
+                    n_trees=T // C if C > 1 else T)
+            else:
+                ops.predict_forest_binned8(
+                    bins, packed[0], roots, out[c], init=init_c,
+                    tree_start=c, tree_step=C if C > 1 else 1,
+                    n_trees=T // C if C > 1 else T)
         return out
 
     def predict(self, data, device=None) -> np.ndarray:

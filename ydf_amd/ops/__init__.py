@@ -460,3 +460,20 @@ def predict_forest_binned8(B: torch.Tensor, packed8: torch.Tensor,
                                   tree_start, tree_step, n_trees,
                                   out.data_ptr(), init, 1.0, _stream())
     return out
+
+
+def predict_forest_binned4(B: torch.Tensor, nodes4: torch.Tensor,
+                           leaf_vals: torch.Tensor, roots: torch.Tensor,
+                           out: torch.Tensor, init: float = 0.0,
+                           tree_start: int = 0, tree_step: int = 1,
+                           n_trees: int = -1):
+    """4-byte-node binned engine (GPU): see pack_binned4_nodes."""
+    assert B.is_cuda
+    F, N = B.shape
+    if n_trees < 0:
+        n_trees = roots.numel()
+    _C.gpu_predict_forest_binned4(B.data_ptr(), N, F, nodes4.data_ptr(),
+                                  leaf_vals.data_ptr(), roots.data_ptr(),
+                                  tree_start, tree_step, n_trees,
+                                  out.data_ptr(), init, 1.0, _stream())
+    return out

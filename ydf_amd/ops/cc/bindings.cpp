@@ -76,6 +76,10 @@ void gpu_vecseq_project(const float*, const int64_t*, const float*,
 void gpu_predict_forest_binned8(const uint8_t*, int64_t, int,
                                 const uint32_t*, const int32_t*, int,
                                 int, int, float*, float, float, void*);
+void gpu_predict_forest_binned4(const uint8_t*, int64_t, int,
+                                const uint32_t*, const float*,
+                                const int32_t*, int, int, int, float*,
+                                float, float, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
                   int);
@@ -358,6 +362,19 @@ PYBIND11_MODULE(_ydf_ops, m) {
                                 P<int32_t>(cond_offs), P<float>(leaf_vals),
                                 n_trees, P<float>(out), init, scale,
                                 (void*)stream);
+        },
+        nogil);
+  m.def("gpu_predict_forest_binned4",
+        [](uintptr_t B, int64_t N, int F, uintptr_t nodes4,
+           uintptr_t leaf_vals, uintptr_t roots, int tree_start,
+           int tree_step, int n_trees, uintptr_t out, float init,
+           float scale, uintptr_t stream) {
+          gpu_predict_forest_binned4(P<uint8_t>(B), N, F,
+                                     P<uint32_t>(nodes4),
+                                     P<float>(leaf_vals),
+                                     P<int32_t>(roots), tree_start,
+                                     tree_step, n_trees, P<float>(out),
+                                     init, scale, (void*)stream);
         },
         nogil);
   m.def("gpu_predict_forest_binned8",

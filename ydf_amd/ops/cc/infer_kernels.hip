@@ -321,6 +321,65 @@ __global__ void predict_forest_binned8_kernel(
   out[base + tid] = init + (acc - init) * scale;
 }
 
+// 4-byte node: feat (6 b, 63 = leaf) | bin (8 b) | left-child-or-
+// leaf-value-index (18 b). Leaf values live in a dense side table, so
+// a depth-6 visit costs 4 B + one 4-B leaf fetch per tree — ~1.7x
+// less L2 traffic than the 8-B nodes. Limits: F <= 63, <= 2^18 nodes
+// and leaves per forest (a 1000-tree depth-6 GBT has ~127k nodes).
+__global__ void predict_forest_binned4_kernel(
+    const uint8_t* __restrict__ B, int64_t N, int F,
+    const uint32_t* __restrict__ nodes,
+    const float* __restrict__ leaf_vals,
+    const int32_t* __restrict__ roots, int tree_start, int tree_step,
+    int n_trees, float* __restrict__ out, float init, float scale) {
+  extern __shared__ uint8_t bs[];  // [F][kTile]
+  const int tid = threadIdx.x;
+  const int64_t base = (int64_t)blockIdx.x * kTile;
+  const int64_t n_here = min((int64_t)kTile, N - base);
+  if (n_here <= 0) return;
+  for (int idx = tid; idx < F * kTile; idx += blockDim.x) {
+    const int f = idx >> 8;
+    const int i = idx & 255;
+    bs[idx] = (i < n_here) ? B[(int64_t)f * N + base + i] : 0;
+  }
+  __syncthreads();
+  if (tid >= n_here) return;
+  float acc = init;
+  int t = 0;
+  for (; t + 4 <= n_trees; t += 4) {
+    uint32_t nd[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      nd[u] = nodes[roots[tree_start + (int64_t)(t + u) * tree_step]];
+    bool done = false;
+    while (!done) {
+      done = true;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int f = nd[u] & 63;
+        if (f != 63) {
+          const int right =
+              (int)bs[f * kTile + tid] > (int)((nd[u] >> 6) & 255);
+          nd[u] = nodes[(nd[u] >> 14) + right];
+          done &= (nd[u] & 63) == 63;
+        }
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc += leaf_vals[nd[u] >> 14];
+  }
+  for (; t < n_trees; ++t) {
+    uint32_t nd = nodes[roots[tree_start + (int64_t)t * tree_step]];
+    while ((nd & 63) != 63) {
+      const int right =
+          (int)bs[(nd & 63) * kTile + tid] > (int)((nd >> 6) & 255);
+      nd = nodes[(nd >> 14) + right];
+    }
+    acc += leaf_vals[nd >> 14];
+  }
+  out[base + tid] = init + (acc - init) * scale;
+}
+
 __global__ void sigmoid_kernel(const float* __restrict__ in,
                                float* __restrict__ out, int64_t N) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -410,6 +469,20 @@ void gpu_predict_forest_binned8(const uint8_t* B, int64_t N, int F,
                      dim3(kTile), lds, (hipStream_t)stream, B, N, F,
                      reinterpret_cast<const Node8*>(nodes8), roots,
                      tree_start, tree_step, n_trees, out, init, scale);
+}
+
+void gpu_predict_forest_binned4(const uint8_t* B, int64_t N, int F,
+                                const uint32_t* nodes4,
+                                const float* leaf_vals,
+                                const int32_t* roots, int tree_start,
+                                int tree_step, int n_trees, float* out,
+                                float init, float scale, void* stream) {
+  const size_t lds = (size_t)F * kTile;
+  const int grid = (int)((N + kTile - 1) / kTile);
+  hipLaunchKernelGGL(predict_forest_binned4_kernel, dim3(grid),
+                     dim3(kTile), lds, (hipStream_t)stream, B, N, F,
+                     nodes4, leaf_vals, roots, tree_start, tree_step,
+                     n_trees, out, init, scale);
 }
 
 void gpu_predict_forest_qs(const float* X, int64_t N, int F,
