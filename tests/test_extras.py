@@ -355,3 +355,14 @@ def test_to_js_multiclass_and_categorical(tmp_path):
                     for line in r.stdout.strip().splitlines()])
     want = m.predict(d, device="cpu")[idx]
     np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_help_and_version():
+    import ydf_amd as ydf
+
+    assert "GRADIENT_BOOSTED_TREES" in ydf.help.learners()
+    assert "num_trees" in ydf.help.hyperparameters(
+        ydf.GradientBoostedTreesLearner)
+    assert "csv:" in ydf.help.loading_data()
+    assert ydf.version == ydf.__version__
+    assert ydf.experimental.MultiLayerPerceptronLearner is not None

@@ -1147,3 +1147,21 @@ def test_custom_metrics_in_training_logs(binary_data):
     assert len(vals) >= 10
     assert vals[-1] < vals[0]  # boosting improves it
     assert 0.0 < vals[-1] < 0.25
+
+
+def test_conditional_expectations(binary_data):
+    """CEP plots (reference model_analysis CEP companion to PDP):
+    means over actual examples bucketed by feature value."""
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=20,
+                                        validation_ratio=0).train(
+        binary_data)
+    an = m.analyze(binary_data)
+    assert an.conditional_expectations
+    ce = next(c for c in an.conditional_expectations
+              if c.feature == "x1")
+    assert ce.counts.sum() > 0
+    # P(positive class) is monotone in x1 in the fixture (direction
+    # depends on the frequency-ordered class vocabulary)
+    mp = ce.mean_prediction[ce.counts > 50]
+    assert abs(float(mp[-1]) - float(mp[0])) > 0.3
+    assert "Conditional expectation" in an._repr_html_()

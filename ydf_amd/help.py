@@ -19,12 +19,11 @@ Cells may be: scalars (numerical/categorical/boolean), token sets
 
 
 def learners() -> str:
-    """Lists the available learners."""
-    from ydf_amd.utils.registry import LEARNERS
+    """Lists the registered learners (reference GetLearner registry)."""
+    from ydf_amd.utils import registry
 
-    text = "\n".join(sorted(LEARNERS)) if isinstance(LEARNERS, dict) \
-        else "GRADIENT_BOOSTED_TREES, RANDOM_FOREST, CART, " \
-             "ISOLATION_FOREST"
+    registry.get_learner("GRADIENT_BOOSTED_TREES")  # bootstrap
+    text = "\n".join(sorted(registry.learner_registry._items))
     print(text)
     return text
 

@@ -1132,6 +1132,7 @@ class CartLearner(RandomForestLearner):
         model.forest = self._finalize_forest(build_flat_forest(
             [tree], bnd, leaf_scale=1.0, cat_feats=cat_feats))
         model._dev_forest = {}
+        model._thr_on_cuts = None  # forest changed: re-check eligibility
         return model
 
 

@@ -26,11 +26,27 @@ class PartialDependence:
 
 
 @dataclasses.dataclass
+class ConditionalExpectation:
+    """Mean prediction and mean label over ACTUAL examples grouped by
+    feature value (reference CEP plots, utils/model_analysis.cc)."""
+
+    feature: str
+    grid: np.ndarray
+    mean_prediction: np.ndarray
+    mean_label: np.ndarray
+    counts: np.ndarray
+    is_categorical: bool = False
+    categories: Optional[List[str]] = None
+
+
+@dataclasses.dataclass
 class Analysis:
     """Analysis report (mirrors ydf model.analyze() output)."""
 
     variable_importances: Dict[str, List[Tuple[float, str]]]
     partial_dependences: List[PartialDependence]
+    conditional_expectations: List[ConditionalExpectation] = \
+        dataclasses.field(default_factory=list)
 
     def to_text(self) -> str:
         out = []
@@ -61,6 +77,20 @@ class Analysis:
                         f"<details><summary>table</summary>"
                         f"<table><tr><th>feature</th>"
                         f"<th>score</th></tr>{body}</table></details>")
+        if self.conditional_expectations:
+            rows.append("<h3>Conditional expectation (on data)</h3>"
+                        "<div style='display:flex;flex-wrap:wrap'>")
+            for ce in self.conditional_expectations:
+                pd_like = PartialDependence(
+                    feature=ce.feature, grid=ce.grid,
+                    mean_prediction=ce.mean_prediction,
+                    is_categorical=ce.is_categorical,
+                    categories=ce.categories)
+                rows.append(
+                    f"<div style='margin:4px'>"
+                    f"<b style='font-size:12px'>{ce.feature}</b><br/>"
+                    + _svg_pdp(pd_like) + "</div>")
+            rows.append("</div>")
         if self.partial_dependences:
             rows.append("<h3>Partial dependence</h3>"
                         "<div style='display:flex;flex-wrap:wrap'>")
@@ -275,6 +305,75 @@ def partial_dependences(model, data, features: Optional[List[str]] = None,
     return out
 
 
+def conditional_expectations(model, data, labels=None,
+                             num_bins: int = 16,
+                             max_examples: int = 20000,
+                             device=None):
+    """Mean prediction (and mean label when given) over actual
+    examples bucketed by each feature's value — the reference's CEP
+    companion to PDP (utils/model_analysis.cc)."""
+    import torch
+
+    X = model._encode_features(data)
+    if labels is not None:
+        labels = np.asarray(labels, dtype=np.float64)
+    if X.shape[1] > max_examples:
+        idx = np.random.RandomState(0).choice(X.shape[1], max_examples,
+                                              replace=False)
+        X = X[:, idx]
+        if labels is not None:
+            labels = labels[idx]
+    dev = (torch.device(device) if device is not None
+           else (torch.device("cuda") if torch.cuda.is_available()
+                 else torch.device("cpu")))
+    Xt = torch.from_numpy(np.ascontiguousarray(X)).to(dev)
+    p = model._apply_activation(model.predict_margin(Xt)).cpu().numpy()
+    if p.ndim == 2:
+        p = p[:, min(1, p.shape[1] - 1)]
+    p = p.astype(np.float64)
+    out = []
+    for fi, spec in enumerate(model.dataspec.feature_columns):
+        vals = X[fi]
+        if spec.semantic == Semantic.CATEGORICAL:
+            K = min(spec.vocab_size, 16)
+            codes = np.clip(vals.astype(np.int64), 0, K - 1)
+            cnt = np.bincount(codes, minlength=K)
+            mp = np.bincount(codes, weights=p, minlength=K) \
+                / np.maximum(cnt, 1)
+            ml = (np.bincount(codes, weights=labels, minlength=K)
+                  / np.maximum(cnt, 1)) if labels is not None \
+                else np.full(K, np.nan)
+            out.append(ConditionalExpectation(
+                feature=spec.name,
+                grid=np.arange(K, dtype=np.float32),
+                mean_prediction=mp.astype(np.float32),
+                mean_label=ml.astype(np.float32),
+                counts=cnt.astype(np.int64), is_categorical=True,
+                categories=[spec.vocab[c] for c in range(K)]))
+        else:
+            edges = np.quantile(vals, np.linspace(0, 1, num_bins + 1))
+            edges = np.unique(edges)
+            if len(edges) < 2:
+                continue
+            codes = np.clip(np.searchsorted(edges, vals, side="right")
+                            - 1, 0, len(edges) - 2)
+            K = len(edges) - 1
+            cnt = np.bincount(codes, minlength=K)
+            mp = np.bincount(codes, weights=p, minlength=K) \
+                / np.maximum(cnt, 1)
+            ml = (np.bincount(codes, weights=labels, minlength=K)
+                  / np.maximum(cnt, 1)) if labels is not None \
+                else np.full(K, np.nan)
+            centers = (edges[:-1] + edges[1:]) / 2
+            out.append(ConditionalExpectation(
+                feature=spec.name,
+                grid=centers.astype(np.float32),
+                mean_prediction=mp.astype(np.float32),
+                mean_label=ml.astype(np.float32),
+                counts=cnt.astype(np.int64)))
+    return out
+
+
 def analyze(model, data, labels: Optional[np.ndarray] = None,
             permutation_variable_importance: bool = True,
             partial_dependence: bool = True,
@@ -287,8 +386,12 @@ def analyze(model, data, labels: Optional[np.ndarray] = None,
                else "MEAN_INCREASE_IN_RMSE")
         vi[key] = permutation_importances(model, data, labels, device=device)
     pdps = []
+    ceps = []
     if partial_dependence:
         pdps = partial_dependences(model, data, features=features,
                                    num_grid_points=num_grid_points,
                                    device=device)
-    return Analysis(variable_importances=vi, partial_dependences=pdps)
+        ceps = conditional_expectations(model, data, labels=labels,
+                                        device=device)
+    return Analysis(variable_importances=vi, partial_dependences=pdps,
+                    conditional_expectations=ceps)
