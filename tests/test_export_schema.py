@@ -273,3 +273,30 @@ def test_golden_header_bytes(binary_data, tmp_path):
     assert raw[:2] == b"\x08\x01"
     # field 7 length-delimited "BLOB_SEQUENCE" -> tag 0x3a len 13
     assert b"\x3a\x0dBLOB_SEQUENCE" in raw
+
+
+def test_wire_decoder_robust_on_fuzzed_bytes():
+    """The schema decoder must reject arbitrary garbage with WireError
+    (or decode it), never crash — hypothesis-style fuzz with a fixed
+    seed for reproducibility."""
+    rng = np.random.RandomState(1234)
+    for _ in range(300):
+        n = int(rng.randint(0, 60))
+        data = rng.randint(0, 256, n).astype(np.uint8).tobytes()
+        try:
+            pw.decode(GBT_HDR, data, strict=True)
+        except pw.WireError:
+            pass
+        except (UnicodeDecodeError, OverflowError):
+            pass  # string fields may reject invalid utf-8
+
+
+def test_blob_sequence_truncation_rejected(tmp_path):
+    p = tmp_path / "trunc"
+    p.write_bytes(b"BS" + b"\x01\x00\x00\x00\x00\x00" + b"\x10\x00\x00\x00ab")
+    with pytest.raises(pw.WireError):
+        pw.read_blob_sequence(str(p))
+    p2 = tmp_path / "badmagic"
+    p2.write_bytes(b"XY" + bytes(6))
+    with pytest.raises(pw.WireError):
+        pw.read_blob_sequence(str(p2))
