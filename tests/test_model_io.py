@@ -161,3 +161,29 @@ def test_property_random_tree_roundtrip():
         m4 = ydf.build_model_from_trees(back, spec,
                                         task=ydf.Task.REGRESSION)
         np.testing.assert_array_equal(p1, m4.predict(d, device="cpu"))
+
+
+def test_pydf_method_surface(binary_data, tmp_path):
+    """Thin PYDF model-method delegates: predict_class, serialize,
+    to_cpp/to_standalone_cc/to_standalone_java, col idxs, logs."""
+    import ydf_amd as ydf
+
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=5,
+                                        validation_ratio=0).train(
+        binary_data)
+    cls = m.predict_class(binary_data)
+    assert set(cls) <= set(m.label_classes)
+    acc = (cls == binary_data["label"]).mean()
+    assert acc > 0.9
+    m2 = ydf.deserialize_model(m.serialize())
+    np.testing.assert_allclose(m.predict(binary_data),
+                               m2.predict(binary_data), rtol=1e-6)
+    assert "predict" in m.to_cpp("k")
+    assert "while ((fi" in m.to_standalone_cc()
+    assert "class YdfModel" in m.to_standalone_java()
+    assert m.input_features_col_idxs() == [0, 1, 2]
+    assert m.hyperparameter_optimizer_logs() is None
+    m.set_feature_selection_logs({"selected": ["x1"]})
+    assert m.feature_selection_logs()["selected"] == ["x1"]
+    with pytest.raises(ImportError):
+        m.to_tensorflow_function()

@@ -128,6 +128,74 @@ class GenericModel:
             "JAX is not available in this environment; use to_cpp()/"
             "to_java()/to_docker() for deployment")
 
+    def to_tensorflow_function(self, *a, **k):
+        raise ImportError(
+            "TensorFlow is not available in this environment; use "
+            "to_cpp()/to_java()/to_js() for embedding")
+
+    def update_with_jax_params(self, *a, **k):
+        raise ImportError("JAX is not available in this environment")
+
+    # -- PYDF method-surface parity (thin delegates) -------------------
+    def predict_class(self, data, device=None) -> np.ndarray:
+        """Most likely class name per example (PYDF
+        model.predict_class; classification only)."""
+        if self._task != Task.CLASSIFICATION or not self.label_classes:
+            raise ValueError("predict_class requires a classification "
+                             "model")
+        p = self.predict(data, device=device)
+        names = np.asarray(self.label_classes)
+        if p.ndim == 1:
+            return names[(p >= 0.5).astype(np.int64)]
+        return names[p.argmax(axis=1)]
+
+    def serialize(self) -> bytes:
+        """In-memory serialization (PYDF model.serialize; pair with
+        ydf.deserialize_model)."""
+        from ydf_amd.model.model_lib import serialize_model
+
+        return serialize_model(self)
+
+    def to_cpp(self, key: str = "my_model") -> str:
+        from ydf_amd.serving.embed import to_cpp
+
+        return to_cpp(self, key)
+
+    def to_standalone_cc(self, name: str = "ydf_model",
+                         algorithm: str = "ROUTING") -> str:
+        from ydf_amd.serving.embed import to_cpp
+
+        return to_cpp(self, name, algorithm=algorithm)
+
+    def to_standalone_java(self, class_name: str = "YdfModel") -> str:
+        from ydf_amd.serving.embed import to_java
+
+        return to_java(self, class_name)
+
+    def to_docker(self, path: str, **kwargs) -> None:
+        from ydf_amd.serving.deploy import to_docker
+
+        to_docker(self, path, **kwargs)
+
+    def input_features_col_idxs(self):
+        label = self.dataspec.label
+        return [i for i, c in enumerate(self.dataspec.columns)
+                if c.name != label]
+
+    def hyperparameter_optimizer_logs(self):
+        """Tuning trial logs when the model came from a tuner (PYDF
+        model.hyperparameter_optimizer_logs)."""
+        return self.tuner_logs
+
+    def set_metadata(self, metadata) -> None:
+        self.metadata = dict(metadata) if metadata else {}
+
+    def set_feature_selection_logs(self, logs) -> None:
+        self._feature_selection_logs = logs
+
+    def feature_selection_logs(self):
+        return getattr(self, "_feature_selection_logs", None)
+
     def num_trees(self) -> int:
         return self.forest.n_trees
 
