@@ -366,3 +366,17 @@ def test_help_and_version():
     assert "csv:" in ydf.help.loading_data()
     assert ydf.version == ydf.__version__
     assert ydf.experimental.MultiLayerPerceptronLearner is not None
+
+
+def test_learner_feature_selector_integration(binary_data):
+    """learner(feature_selector=...) trains the final model on the
+    selected subset and attaches the logs (PYDF integration)."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15,
+        feature_selector=ydf.BackwardSelectionFeatureSelector(
+            objective_metric="accuracy")).train(binary_data)
+    logs = m.feature_selection_logs()
+    assert logs is not None
+    assert "x1" in logs.selected_features
+    assert set(m.input_feature_names()) == set(logs.selected_features)
+    assert m.evaluate(binary_data).accuracy > 0.9

@@ -27,7 +27,9 @@ class GenericLearner:
                  categorical_algorithm: str = "CART",
                  maximum_model_size_in_memory_in_bytes: float = -1.0,
                  random_seed: int = 123456, device=None,
+                 feature_selector=None,
                  num_threads: Optional[int] = None):
+        self.feature_selector = feature_selector
         self.allow_na_conditions = allow_na_conditions
         self.pure_serving_model = pure_serving_model
         if missing_value_policy not in ("GLOBAL_IMPUTATION",
@@ -119,6 +121,23 @@ class GenericLearner:
             oblique_max_features=hp.get(
                 "sparse_oblique_max_num_features", -1),
         )
+
+    def _train_with_feature_selection(self, data, valid):
+        """PYDF learner(feature_selector=...) integration: run the
+        selector, train the final model on the selected features and
+        attach the selection logs (PYDF set_feature_selection_logs)."""
+        import copy as _copy
+
+        fs = self.feature_selector
+        base = _copy.copy(self)
+        base.feature_selector = None
+        logs = fs.run(base, data, valid if valid is not None else data)
+        final = _copy.copy(self)
+        final.feature_selector = None
+        final.features = list(logs.selected_features)
+        model = final.train(data)
+        model.set_feature_selection_logs(logs)
+        return model
 
     def _finalize_forest(self, flat):
         """Post-build forest fixups: expands group-space masks on

@@ -366,6 +366,8 @@ class GradientBoostedTreesLearner(GenericLearner):
 
         if isinstance(data, DatasetCache):
             return self._train_streaming(data)
+        if getattr(self, "feature_selector", None) is not None:
+            return self._train_with_feature_selection(data, valid)
         if self.tuner is not None:
             return self._train_with_tuner(data, valid=valid)
         if self.hyperparameters.get("split_axis") == "MHLD_OBLIQUE":
@@ -906,6 +908,8 @@ class RandomForestLearner(GenericLearner):
         return max(1, F // 3)
 
     def train(self, data, valid=None, verbose=None) -> RandomForestModel:
+        if getattr(self, "feature_selector", None) is not None:
+            return self._train_with_feature_selection(data, valid)
         if self.tuner is not None:
             return self._train_with_tuner(data, valid=valid)
         hp = self.hyperparameters
