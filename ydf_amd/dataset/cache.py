@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import json
 import os
-from typing import Iterator, Optional, Tuple
+from typing import Iterator, Tuple
 
 import numpy as np
 

@@ -9,8 +9,6 @@ from __future__ import annotations
 import dataclasses
 from typing import Callable
 
-import numpy as np
-
 
 @dataclasses.dataclass(frozen=True)
 class AbstractCustomMetric:

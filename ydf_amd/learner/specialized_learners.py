@@ -638,7 +638,6 @@ class GradientBoostedTreesLearner(GenericLearner):
 
             def _snapshot(trees_so_far, iteration, init_preds_s):
                 import shutil
-                import tempfile
 
                 from ydf_amd.parallel.dist import is_main
 

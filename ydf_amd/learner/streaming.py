@@ -17,7 +17,7 @@ to floating-point accumulation order across chunks.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 import torch

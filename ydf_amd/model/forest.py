@@ -11,7 +11,7 @@ FlatNodeModel :201), laid out SoA for the MI355X inference kernels:
 from __future__ import annotations
 
 import dataclasses
-from typing import List, Optional
+from typing import List
 
 import numpy as np
 

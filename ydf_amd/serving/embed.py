@@ -17,7 +17,6 @@ from typing import List
 
 import numpy as np
 
-from ydf_amd.dataset.dataspec import Semantic
 
 
 def _emit_node(f, forest, n: int, indent: str, lines: List[str]):
