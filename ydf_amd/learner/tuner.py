@@ -97,13 +97,30 @@ def _trial_worker(payload):
 
 
 def run_parallel_trials(learner, tuner, data):
-    """Evaluates all trials in a process pool (parallel_trials slots)."""
+    """Evaluates all trials in a process pool (parallel_trials slots).
+
+    Falls back to sequential trials when worker processes cannot be
+    spawned (multiprocessing "spawn" must re-import __main__, which an
+    interactive session / heredoc script does not have)."""
     import multiprocessing as mp
+    import sys
 
     from ydf_amd.dataset.dataset import _to_column_dict
     from ydf_amd.model.model_lib import deserialize_model
 
     import copy
+
+    main_file = getattr(sys.modules.get("__main__"), "__file__", None)
+    if not main_file or not __import__("os").path.exists(main_file):
+        from ydf_amd.utils.log import info
+
+        info("parallel_trials: __main__ is not an importable file "
+             "(interactive session); running trials sequentially")
+        seq = copy.copy(tuner)
+        seq.parallel_trials = 1
+        seq_learner = copy.copy(learner)
+        seq_learner.tuner = seq
+        return seq_learner.train(data)
 
     rng = np.random.RandomState(tuner.seed)
     samples = [tuner.sample(rng) for _ in range(tuner.num_trials)]
