@@ -138,6 +138,16 @@ def main():
 
     for i in range(args.warmup):
         step(i)
+    if graph is not None and not bool(torch.isfinite(preds).all()):
+        # graph replay produced garbage (e.g. multi-rank collective
+        # capture silently wrong): fall back to the eager path with a
+        # clean slate rather than timing a broken configuration
+        print("# graph replay sanity check FAILED; falling back to "
+              "eager", file=sys.stderr)
+        graph = None
+        preds.fill_(init)
+        for i in range(args.warmup):
+            step(i)
     dist_lib.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
