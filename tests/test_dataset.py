@@ -219,3 +219,43 @@ def test_pluggable_filesystem(tmp_path):
         assert m.num_trees() == 3
     finally:
         fs._REGISTRY.pop("mem", None)
+
+
+def test_dataframe_ducktype_polars_like():
+    """Any object with .columns and __getitem__ returning array-likes
+    works (this is how polars DataFrames flow in without a polars
+    dependency — PYDF ships a dedicated polars_io backend; here the
+    generic DataFrame path covers it)."""
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(0)
+    n = 2000
+
+    class FakeSeries:
+        def __init__(self, v):
+            self._v = v
+
+        def __array__(self, dtype=None):
+            return np.asarray(self._v, dtype=dtype)
+
+        def __len__(self):
+            return len(self._v)
+
+    class FakeFrame:
+        def __init__(self, cols):
+            self._cols = cols
+
+        @property
+        def columns(self):
+            return list(self._cols)
+
+        def __getitem__(self, name):
+            return FakeSeries(self._cols[name])
+
+    x = rng.randn(n).astype(np.float32)
+    df = FakeFrame({"x": x,
+                    "label": np.where(x > 0, "p", "n")})
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=5,
+                                        validation_ratio=0,
+                                        device="cpu").train(df)
+    assert m.evaluate(df).accuracy > 0.95
