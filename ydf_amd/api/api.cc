@@ -539,23 +539,41 @@ float Model::PredictRow(const float* f) const {
 std::vector<float> Model::Predict(const Dataset& ds) const {
   const size_t N = ds.num_rows();
   const size_t F = features_.size();
+  // resolve column pointers + vocab lookup tables ONCE (not per row)
+  struct Col {
+    const std::vector<float>* num = nullptr;
+    const std::vector<std::string>* cat = nullptr;
+    std::map<std::string, int> lut;
+    float mean = 0.f;
+  };
+  std::vector<Col> cols(F);
+  for (size_t f = 0; f < F; ++f) {
+    const auto& spec = features_[f];
+    if (spec.type == ColumnType::kCategorical) {
+      const auto it = ds.categoricals().find(spec.name);
+      if (it == ds.categoricals().end())
+        throw std::runtime_error("missing feature " + spec.name);
+      cols[f].cat = &it->second;
+      for (size_t i = 0; i < spec.vocab.size(); ++i)
+        cols[f].lut[spec.vocab[i]] = (int)i;
+    } else {
+      const auto it = ds.numericals().find(spec.name);
+      if (it == ds.numericals().end())
+        throw std::runtime_error("missing feature " + spec.name);
+      cols[f].num = &it->second;
+      cols[f].mean = (float)spec.mean;
+    }
+  }
   std::vector<float> row(F);
   std::vector<float> out(N);
-  // per-feature accessors
   for (size_t i = 0; i < N; ++i) {
     for (size_t f = 0; f < F; ++f) {
-      const auto& spec = features_[f];
-      if (spec.type == ColumnType::kCategorical) {
-        const auto it = ds.categoricals().find(spec.name);
-        if (it == ds.categoricals().end())
-          throw std::runtime_error("missing feature " + spec.name);
-        row[f] = (float)vocab_index(spec, it->second[i]);
+      if (cols[f].cat != nullptr) {
+        const auto vit = cols[f].lut.find((*cols[f].cat)[i]);
+        row[f] = (float)(vit == cols[f].lut.end() ? 0 : vit->second);
       } else {
-        const auto it = ds.numericals().find(spec.name);
-        if (it == ds.numericals().end())
-          throw std::runtime_error("missing feature " + spec.name);
-        const float v = it->second[i];
-        row[f] = std::isfinite(v) ? v : (float)spec.mean;
+        const float v = (*cols[f].num)[i];
+        row[f] = std::isfinite(v) ? v : cols[f].mean;
       }
     }
     out[i] = PredictRow(row.data());
