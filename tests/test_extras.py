@@ -380,3 +380,23 @@ def test_learner_feature_selector_integration(binary_data):
     assert "x1" in logs.selected_features
     assert set(m.input_feature_names()) == set(logs.selected_features)
     assert m.evaluate(binary_data).accuracy > 0.9
+
+
+def test_evaluation_binary_statistics(binary_data):
+    """precision/recall/F1/FPR + per-threshold ROC characteristics
+    (PYDF evaluation surface; reference metric Roc curves)."""
+    m = ydf.GradientBoostedTreesLearner(label="label", num_trees=30,
+                                        validation_ratio=0).train(
+        binary_data)
+    ev = m.evaluate(binary_data)
+    assert 0.9 < ev.precision <= 1.0
+    assert 0.9 < ev.recall <= 1.0
+    assert 0.9 < ev.f1 <= 1.0
+    assert 0.0 <= ev.false_positive_rate < 0.1
+    ch = ev.characteristics[0]
+    assert len(ch["fpr"]) == len(ch["tpr"]) == len(ch["thresholds"])
+    assert (np.diff(ch["fpr"]) >= 0).all()  # monotone sweep
+    assert (np.diff(ch["tpr"]) >= 0).all()
+    # trapezoid integral of the curve must approximate the rank AUC
+    auc_trap = float(np.trapz(ch["tpr"], ch["fpr"]))
+    assert abs(auc_trap - ev.auc) < 0.01

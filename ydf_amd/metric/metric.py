@@ -21,6 +21,22 @@ def confusion_matrix(labels: np.ndarray, pred_classes: np.ndarray,
     return cm
 
 
+def roc_curve(labels: np.ndarray, scores: np.ndarray):
+    """(fpr, tpr, thresholds) swept over unique score cutoffs
+    (descending), the reference's Roc building blocks."""
+    labels = np.asarray(labels, dtype=bool)
+    order = np.argsort(-scores, kind="mergesort")
+    s = scores[order]
+    y = labels[order]
+    distinct = np.nonzero(np.diff(s))[0]
+    idx = np.r_[distinct, len(s) - 1]
+    tps = np.cumsum(y)[idx].astype(np.float64)
+    fps = (idx + 1) - tps
+    P = max(float(labels.sum()), 1.0)
+    Nn = max(float(len(labels) - labels.sum()), 1.0)
+    return fps / Nn, tps / P, s[idx]
+
+
 def roc_auc(labels: np.ndarray, scores: np.ndarray) -> float:
     """Rank-based AUC (equivalent to the trapezoidal ROC integral used by
     the reference, metric.h:150)."""
@@ -191,9 +207,55 @@ class Evaluation:
     qini: Optional[float] = None
     cindex: Optional[float] = None
     confusion: Optional[np.ndarray] = None
+    # per-threshold ROC points for binary classification (PYDF
+    # evaluation.characteristics; reference metric Roc curves):
+    # list of dicts {"name", "fpr", "tpr", "thresholds"}
+    characteristics: Optional[list] = None
     # closed-form 95% confidence intervals (lo, hi)
     accuracy_ci95: Optional[tuple] = None
     auc_ci95: Optional[tuple] = None
+
+    # -- derived binary-classification statistics (from the confusion
+    # matrix; convention: rows = truth, cols = prediction, class 1 =
+    # positive) ------------------------------------------------------
+    def _binary_counts(self):
+        cm = self.confusion
+        if cm is None or cm.shape != (2, 2):
+            return None
+        tn, fp = float(cm[0, 0]), float(cm[0, 1])
+        fn, tp = float(cm[1, 0]), float(cm[1, 1])
+        return tp, fp, tn, fn
+
+    @property
+    def precision(self) -> Optional[float]:
+        c = self._binary_counts()
+        if c is None:
+            return None
+        tp, fp, _, _ = c
+        return tp / (tp + fp) if tp + fp > 0 else float("nan")
+
+    @property
+    def recall(self) -> Optional[float]:
+        c = self._binary_counts()
+        if c is None:
+            return None
+        tp, _, _, fn = c
+        return tp / (tp + fn) if tp + fn > 0 else float("nan")
+
+    @property
+    def f1(self) -> Optional[float]:
+        p, r = self.precision, self.recall
+        if p is None or r is None or p + r == 0:
+            return None
+        return 2 * p * r / (p + r)
+
+    @property
+    def false_positive_rate(self) -> Optional[float]:
+        c = self._binary_counts()
+        if c is None:
+            return None
+        _, fp, tn, _ = c
+        return fp / (fp + tn) if fp + tn > 0 else float("nan")
 
     def to_dict(self) -> Dict:
         d = {"num_examples": self.num_examples}
@@ -241,6 +303,9 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
         if predictions.ndim == 1:
             pred_cls = (predictions >= 0.5).astype(np.int64)
             ev.auc = roc_auc(labels > 0.5, predictions)
+            fpr, tpr, thr = roc_curve(labels > 0.5, predictions)
+            ev.characteristics = [{"name": "default", "fpr": fpr,
+                                   "tpr": tpr, "thresholds": thr}]
             ev.pr_auc = pr_auc(labels > 0.5, predictions)
         else:
             pred_cls = predictions.argmax(axis=1)
@@ -272,6 +337,9 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
     elif task == Task.ANOMALY_DETECTION:
         # labels: 1 = anomaly; predictions: anomaly score in [0, 1]
         ev.auc = roc_auc(labels > 0.5, predictions)
+        fpr, tpr, thr = roc_curve(labels > 0.5, predictions)
+        ev.characteristics = [{"name": "default", "fpr": fpr,
+                               "tpr": tpr, "thresholds": thr}]
         ev.pr_auc = pr_auc(labels > 0.5, predictions)
     elif task == Task.REGRESSION:
         if w is None:
