@@ -285,7 +285,27 @@ class Evaluation:
             f"<tr><td>{k}</td><td>{v:.6g}</td></tr>"
             if isinstance(v, float) else f"<tr><td>{k}</td><td>{v}</td></tr>"
             for k, v in self.to_dict().items())
-        return f"<table>{rows}</table>"
+        html = f"<table>{rows}</table>"
+        if self.characteristics:
+            # ROC curve panel (PYDF evaluation HTML shows the plotly
+            # ROC; here dependency-free inline SVG)
+            ch = self.characteristics[0]
+            w = h = 180
+            pad = 16
+            pts = " ".join(
+                f"{pad + x * (w - 2 * pad):.1f},"
+                f"{h - pad - y * (h - 2 * pad):.1f}"
+                for x, y in zip(ch["fpr"], ch["tpr"]))
+            html += (
+                f'<div><b style="font:11px sans-serif">ROC</b><br/>'
+                f'<svg width="{w}" height="{h}" '
+                'xmlns="http://www.w3.org/2000/svg" '
+                'style="background:#fafafa">'
+                f'<line x1="{pad}" y1="{h - pad}" x2="{w - pad}" '
+                f'y2="{pad}" stroke="#ccc" stroke-dasharray="3"/>'
+                f'<polyline points="{pts}" fill="none" stroke="#2a7" '
+                'stroke-width="1.5"/></svg></div>')
+        return html
 
 
 def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
