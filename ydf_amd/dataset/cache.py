@@ -113,7 +113,8 @@ def create_dataset_cache(data, cache_dir: str, label: str,
     with open(os.path.join(cache_dir, "dataspec.json"), "w") as f:
         json.dump(ds.dataspec.to_json(), f)
     with open(os.path.join(cache_dir, "cache_meta.json"), "w") as f:
-        json.dump({"n_rows": int(N), "n_features": int(F),
+        json.dump({"version": 1,
+                   "n_rows": int(N), "n_features": int(F),
                    "chunk_rows": int(chunk_rows),
                    "n_chunks": int(n_chunks),
                    "cat_flags": [bool(v) for v in cat_flags]}, f)

@@ -33,6 +33,15 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
     Supports: binary classification (loss 1) / regression (loss 2),
     numerical + categorical (<=256 vocab) features, depth-wise growth.
     Returns (trees, init_prediction)."""
+    import torch.distributed as _td
+
+    if _td.is_available() and _td.is_initialized() \
+            and _td.get_world_size() > 1:
+        raise NotImplementedError(
+            "streaming (out-of-core) training is single-process for "
+            "now; chunk-sharded data-parallel streaming is a natural "
+            "extension (each rank owns a chunk subset + the same "
+            "per-level histogram all-reduce) but is not wired yet")
     F = cache.n_features
     N = cache.n_rows
     n_bins = 256
