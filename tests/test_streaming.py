@@ -87,3 +87,72 @@ def test_streaming_gbt_gpu(tmp_path):
         label="label", num_trees=20, max_depth=4,
         validation_ratio=0.0, device="cuda:0").train(cache)
     assert m.evaluate(data).accuracy > 0.9
+
+
+def _worker_streaming(rank, world, port, cache_dir, out_path):
+    import os
+
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import pickle
+
+        import ydf_amd as ydf
+        from ydf_amd.dataset.cache import DatasetCache
+
+        cache = DatasetCache(cache_dir)
+        m = ydf.GradientBoostedTreesLearner(
+            label="label", num_trees=10, max_depth=4,
+            validation_ratio=0.0, device="cpu").train(cache)
+        with open(f"{out_path}.{rank}", "wb") as f:
+            pickle.dump({"feat": m.forest.feat, "thr": m.forest.thr},
+                        f)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_streaming_distributed_chunk_sharded(tmp_path):
+    """Chunk-sharded data-parallel streaming: 2 ranks each own half
+    the chunks; per-level histograms all-reduce, so both ranks must
+    build the same forest, and it must match the single-process
+    streamed forest (same reduced histograms up to fp order)."""
+    import multiprocessing as mp
+    import pickle
+    import socket
+
+    data = _data(n=24000, seed=7)
+    cache = ydf.create_dataset_cache(data, str(tmp_path / "cache"),
+                                     label="label", chunk_rows=4000)
+    assert cache.n_chunks == 6
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    out = str(tmp_path / "m")
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_worker_streaming,
+                         args=(r, 2, port, cache.path, out))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=300)
+    for p in procs:
+        assert p.exitcode == 0, p.exitcode
+    with open(out + ".0", "rb") as f:
+        a = pickle.load(f)
+    with open(out + ".1", "rb") as f:
+        b = pickle.load(f)
+    np.testing.assert_array_equal(a["feat"], b["feat"])
+    np.testing.assert_allclose(a["thr"], b["thr"], rtol=1e-6)
+    # single-process reference
+    m1 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=10, max_depth=4, validation_ratio=0.0,
+        device="cpu").train(cache)
+    np.testing.assert_array_equal(a["feat"], m1.forest.feat)
+    np.testing.assert_allclose(a["thr"], m1.forest.thr, rtol=5e-3,
+                               atol=5e-3)

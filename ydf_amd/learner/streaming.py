@@ -5,6 +5,9 @@ Reference analogue: ShardedSamplingTrain
 distributed dataset-cache training loop (dataset_cache.h:15-58).
 
 MI355X design: one fused pass over the row chunks per tree LEVEL —
+(data-parallel: each rank owns the chunk subset c % world == rank and
+the per-level histograms all-reduce over RCCL, so one on-disk cache
+feeds all 8 GPUs of a node) —
 each chunk is uploaded (H2D on GPU, memory-mapped on CPU), the
 previous level's routing is applied to that chunk's node ids,
 gradients are recomputed from the chunk's resident predictions, and
@@ -35,13 +38,21 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
     Returns (trees, init_prediction)."""
     import torch.distributed as _td
 
-    if _td.is_available() and _td.is_initialized() \
-            and _td.get_world_size() > 1:
-        raise NotImplementedError(
-            "streaming (out-of-core) training is single-process for "
-            "now; chunk-sharded data-parallel streaming is a natural "
-            "extension (each rank owns a chunk subset + the same "
-            "per-level histogram all-reduce) but is not wired yet")
+    distributed = _td.is_available() and _td.is_initialized()
+    rank = _td.get_rank() if distributed else 0
+    world = _td.get_world_size() if distributed else 1
+
+    def allreduce(t):
+        if not distributed:
+            return
+        if t.is_cuda and _td.get_backend() == "gloo":
+            h = t.cpu()
+            _td.all_reduce(h, op=_td.ReduceOp.SUM)
+            t.copy_(h)
+        else:
+            _td.all_reduce(t, op=_td.ReduceOp.SUM)
+
+    my_chunks = [c for c in range(cache.n_chunks) if c % world == rank]
     F = cache.n_features
     N = cache.n_rows
     n_bins = 256
@@ -53,25 +64,28 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
         cat_flags_t = torch.from_numpy(
             cache.cat_flags.astype(np.uint8)).to(device)
 
-    # --- init prediction from a streaming pass over labels ----------
-    s = 0.0
-    for c in range(cache.n_chunks):
+    # --- init prediction from a streaming pass over OWNED labels ----
+    s = torch.zeros(1, dtype=torch.float64, device=device)
+    for c in my_chunks:
         _, labels = cache.chunk(c)
         s += float(np.asarray(labels, dtype=np.float64).sum())
+    allreduce(s)
+    s = float(s.item())
     if cfg.loss == 1:
         p = min(max(s / N, 1e-6), 1 - 1e-6)
         init = float(np.log(p / (1 - p)))
     else:
         init = float(s / N)
 
-    # --- per-chunk persistent host state ----------------------------
-    chunk_meta = []
-    preds_h: List[np.ndarray] = []
-    for c in range(cache.n_chunks):
+    # --- per-OWNED-chunk persistent host state ----------------------
+    chunk_meta = {}
+    preds_h = {}
+    node_ids_h = {}
+    for c in my_chunks:
         rows = min(cache.chunk_rows, N - c * cache.chunk_rows)
-        chunk_meta.append(rows)
-        preds_h.append(np.full(rows, init, dtype=np.float32))
-    node_ids_h = [np.zeros(r, dtype=np.int32) for r in chunk_meta]
+        chunk_meta[c] = rows
+        preds_h[c] = np.full(rows, init, dtype=np.float32)
+        node_ids_h[c] = np.zeros(rows, dtype=np.int32)
 
     # --- device/host buffers ----------------------------------------
     widest = 1 << (D - 1)
@@ -103,7 +117,7 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
             tree_masks.zero_()
         tree_feat_all.fill(-1)
         tree_bin_all.fill(0)
-        for c in range(cache.n_chunks):
+        for c in my_chunks:
             node_ids_h[c].fill(0)
         prev_bf = prev_bb = None
         for level in range(D):
@@ -112,7 +126,7 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
             hist_view = hist[:level_size]
             hist_view.zero_()
             slot_map = arange[:level_size]
-            for c in range(cache.n_chunks):
+            for c in my_chunks:
                 bins_np, labels_np = cache.chunk(c)
                 rows = chunk_meta[c]
                 bins_c = torch.from_numpy(
@@ -136,6 +150,7 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
                                hist_view, level_base, level_size, 0,
                                level_size)
                 del bins_c, nid_c, preds_c, labels_c, gh_c
+            allreduce(hist_view)
             split_scan(hist_view, slot_map + level_base, node_stats,
                        bg_nf[:level_size], bb_nf[:level_size],
                        best_feat, best_bin, best_gain, 0, level_size,
@@ -154,7 +169,7 @@ def train_gbt_streaming(cache, cfg, device: torch.device,
                         lambda_l1=cfg.lambda_l1)
         last_base = (1 << (D - 1)) - 1
         last_size = 1 << (D - 1)
-        for c in range(cache.n_chunks):
+        for c in my_chunks:
             bins_np, _ = cache.chunk(c)
             bins_c = torch.from_numpy(
                 np.ascontiguousarray(bins_np)).to(device)
