@@ -1257,12 +1257,20 @@ __global__ void binary_logloss_kernel(const float* __restrict__ preds,
 // Host launchers (raw pointers + explicit stream; exported via pybind).
 // ---------------------------------------------------------------------------
 static inline int row_chunks(int64_t N, int F, int max_blocks = 8192) {
-  // Enough blocks to fill 256 CUs x several waves, but bounded (every block
-  // merges its LDS histogram, so block count is also merge traffic).
+  // Enough blocks to fill 256 CUs x several waves, but bounded: every
+  // block merges (and first zeroes) its LDS histogram, so block count is
+  // also merge traffic. Measured on MI355X (profiles/
+  // kernel_stats_r02_final.md sweep): at small shards the kernel is
+  // merge-bound, and ~64k rows/chunk is the sweet spot at every shard
+  // size of the 8-GPU strong-scaling ladder (1.375M: 73->21 chunks =
+  // +16% trees/s; 11M unchanged because the max_blocks cap binds first).
   int per_f = (int)((max_blocks + F - 1) / F);
-  int64_t min_rows = 1024;
-  int64_t max_chunks = (N + min_rows - 1) / min_rows;
-  int chunks = (int)(max_chunks < per_f ? max_chunks : per_f);
+  int64_t target = (N + 65535) >> 16;           // >=64k rows per chunk
+  const int64_t fill = (512 + F - 1) / F;       // ...but keep >=512 WGs
+  if (target < fill) target = fill;
+  const int64_t max_chunks = (N + 1023) >> 10;  // never <1k rows/chunk
+  int chunks = (int)(target < per_f ? target : per_f);
+  if (chunks > max_chunks) chunks = (int)max_chunks;
   if (chunks < 1) chunks = 1;
   // count/h-field packing requires rows_per_block <= 2^19
   const int min_chunks = (int)((N + (1 << 19) - 1) >> 19);
@@ -1365,7 +1373,14 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
   }
   const int n_fgroups = (F + fpb - 1) / fpb;
   const int threads = hist_block_threads();
-  int chunks = row_chunks(N, n_fgroups, 8192 * 256 / threads);
+  static int max_blocks_env = -1;
+  if (max_blocks_env < 0) {
+    const char* e = getenv("YDFA_HIST_MAX_BLOCKS");
+    max_blocks_env = e ? atoi(e) : 0;
+  }
+  int chunks = row_chunks(
+      N, n_fgroups,
+      max_blocks_env > 0 ? max_blocks_env : 8192 * 256 / threads);
   {
     static int64_t max_rpb = -1;
     if (max_rpb < 0) {
