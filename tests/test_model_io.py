@@ -187,3 +187,64 @@ def test_pydf_method_surface(binary_data, tmp_path):
     assert m.feature_selection_logs()["selected"] == ["x1"]
     with pytest.raises(ImportError):
         m.to_tensorflow_function()
+
+
+def test_tree_editing(trained, binary_data):
+    """set_tree/add_tree/remove_tree (reference PYDF
+    decision_forest_model.py:148-173): edits round-trip through the
+    flat forest and are visible in predictions."""
+    n0 = trained.num_trees()
+    p0 = trained.predict(binary_data)
+
+    # replace tree 0 with a single constant leaf -> predictions move
+    t0 = trained.get_tree(0)
+    trained.set_tree(0, tree_lib.Tree(root=tree_lib.Leaf(value=0.0)))
+    assert trained.num_trees() == n0
+    p1 = trained.predict(binary_data)
+    assert np.abs(p1 - p0).max() > 1e-6
+    # restore the original tree -> predictions restored exactly
+    trained.set_tree(0, t0)
+    np.testing.assert_allclose(trained.predict(binary_data), p0,
+                               rtol=1e-6, atol=1e-7)
+
+    # add_tree / remove_tree keep the rest intact
+    trained.add_tree(tree_lib.Tree(root=tree_lib.Leaf(value=0.25)))
+    assert trained.num_trees() == n0 + 1
+    p2 = trained.predict(binary_data)
+    assert np.abs(p2 - p0).max() > 1e-3
+    trained.remove_tree(n0)
+    assert trained.num_trees() == n0
+    np.testing.assert_allclose(trained.predict(binary_data), p0,
+                               rtol=1e-6, atol=1e-7)
+
+
+def test_tree_edit_preserves_categorical_and_na(binary_data):
+    """Edit round-trip on a model with categorical masks and NA
+    routing: untouched trees must serve identically after a rebuild."""
+    data = dict(binary_data)
+    rng = np.random.RandomState(5)
+    x = data["x1"].astype(np.float32).copy()
+    x[rng.rand(len(x)) < 0.2] = np.nan
+    data["x1"] = x
+    data["cat"] = rng.choice(["a", "b", "c", "d", "e"], len(x))
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=12, validation_ratio=0.0,
+        allow_na_conditions=True).train(data)
+    p0 = m.predict(data)
+    t_last = m.get_tree(m.num_trees() - 1)
+    m.set_tree(m.num_trees() - 1, t_last)  # identity edit
+    np.testing.assert_allclose(m.predict(data), p0, rtol=1e-6,
+                               atol=1e-7)
+
+
+def test_plot_tree(trained, tmp_path):
+    plot = trained.plot_tree(0, max_depth=8)
+    html = plot.html()
+    assert html.startswith("<svg") and html.endswith("</svg>")
+    assert "value=" in html
+    # depth-pruned rendering still works
+    assert "<svg" in trained.plot_tree(0, max_depth=2).html()
+    assert plot._repr_html_() == html
+    f = str(tmp_path / "tree.svg")
+    plot.to_file(f)
+    assert "<svg" in open(f).read()

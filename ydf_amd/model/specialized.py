@@ -34,6 +34,55 @@ class DecisionForestModel(GenericModel):
         return tree_lib.format_tree(
             self.get_tree(idx), self.dataspec, max_depth=max_depth)
 
+    def plot_tree(self, idx: int = 0, max_depth: int = 6):
+        """SVG rendering of one tree (mirrors PYDF model.plot_tree,
+        port/python/ydf/model/decision_forest_model/
+        decision_forest_model.py:102)."""
+        from ydf_amd.model import tree as tree_lib
+
+        return tree_lib.plot_tree(self.get_tree(idx), self.dataspec,
+                                  max_depth=max_depth,
+                                  label_classes=self.label_classes)
+
+    def _rebuild_from_trees(self, trees) -> None:
+        from ydf_amd.model import tree as tree_lib
+
+        self.forest = tree_lib.build_forest_from_trees(
+            trees, n_features=len(self.dataspec.feature_columns))
+        # drop every derived/device representation of the old forest
+        self._dev_forest.clear()
+        if hasattr(self, "_thr_on_cuts"):
+            del self._thr_on_cuts
+
+    def set_tree(self, idx: int, tree) -> None:
+        """Replaces tree `idx` (mirrors PYDF model.set_tree,
+        decision_forest_model.py:148). The edited model serves the new
+        structure everywhere (all engines are rebuilt lazily)."""
+        trees = self.get_all_trees()
+        if not 0 <= idx < len(trees):
+            raise ValueError(f"tree index {idx} out of range")
+        trees[idx] = tree
+        self._rebuild_from_trees(trees)
+
+    def add_tree(self, tree) -> None:
+        """Appends a tree (mirrors PYDF model.add_tree,
+        decision_forest_model.py:158). For multi-output GBT models the
+        caller is responsible for keeping class striding consistent
+        (append num_trees_per_iter trees per round), as in the
+        reference."""
+        trees = self.get_all_trees()
+        trees.append(tree)
+        self._rebuild_from_trees(trees)
+
+    def remove_tree(self, idx: int) -> None:
+        """Removes tree `idx` (mirrors PYDF model.remove_tree,
+        decision_forest_model.py:167)."""
+        trees = self.get_all_trees()
+        if not 0 <= idx < len(trees):
+            raise ValueError(f"tree index {idx} out of range")
+        del trees[idx]
+        self._rebuild_from_trees(trees)
+
 
 class GradientBoostedTreesModel(DecisionForestModel):
     _model_type = "GRADIENT_BOOSTED_TREES"
