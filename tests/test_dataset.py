@@ -259,3 +259,28 @@ def test_dataframe_ducktype_polars_like():
                                         validation_ratio=0,
                                         device="cpu").train(df)
     assert m.evaluate(df).accuracy > 0.95
+
+
+def test_structured_array_ingestion():
+    """numpy structured arrays work as a dataset (reference dataset IO
+    accepts several column-store shapes; port/python ydf/dataset/io)."""
+    import ydf_amd as ydf
+
+    n = 2000
+    rng = np.random.RandomState(0)
+    arr = np.zeros(n, dtype=[("x1", "f4"), ("x2", "f4"), ("label", "U3")])
+    arr["x1"] = rng.randn(n)
+    arr["x2"] = rng.randn(n)
+    arr["label"] = np.where(arr["x1"] + arr["x2"] > 0, "yes", "no")
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=10, validation_ratio=0.0).train(arr)
+    assert m.evaluate(arr).accuracy > 0.9
+
+
+def test_unsupported_dataset_type_message():
+    import pytest
+
+    import ydf_amd as ydf
+
+    with pytest.raises(ValueError, match="supported"):
+        ydf.GradientBoostedTreesLearner(label="y").train(12345)

@@ -80,10 +80,21 @@ def _to_column_dict(data: InputData) -> Dict[str, np.ndarray]:
         return _to_column_dict(df)
     if isinstance(data, dict):
         return {k: np.asarray(v) for k, v in data.items()}
-    # pandas DataFrame (duck-typed to avoid a hard dependency)
+    # numpy structured array (one named field per column)
+    if isinstance(data, np.ndarray) and data.dtype.names:
+        return {str(n): np.ascontiguousarray(data[n])
+                for n in data.dtype.names}
+    # xarray.Dataset (duck-typed; reference port/python ydf/dataset/io)
+    if hasattr(data, "data_vars") and hasattr(data, "to_dataframe"):
+        return _to_column_dict(
+            data.to_dataframe().reset_index(drop=True))
+    # pandas/polars DataFrame (duck-typed to avoid a hard dependency)
     if hasattr(data, "columns") and hasattr(data, "__getitem__"):
         return {str(c): np.asarray(data[c]) for c in data.columns}
-    raise ValueError(f"unsupported dataset type: {type(data)}")
+    raise ValueError(
+        f"unsupported dataset type: {type(data)} (supported: dict of "
+        "columns, pandas/polars DataFrame, numpy structured array, "
+        "xarray.Dataset, 'csv:'/'tfrecord:'/'avro:' paths)")
 
 
 def _is_numerical(arr: np.ndarray) -> bool:
