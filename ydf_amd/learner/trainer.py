@@ -1185,11 +1185,14 @@ class ForestTrainer:
                     maskbits=maskbits)
         else:
             hist_view.zero_()
+            # the hint marks subtraction levels (many rows belong to a
+            # DERIVED sibling and are skipped); the launcher picks the
+            # slot8 1-B-per-row filter when the level fits one launch
             ops.hist_build(self.bins, self.gh, self.node_ids, build_map,
                            hist_view, level_base, level_size, 0,
                            level_size,
-                           filtered_hint=use_sub and os.environ.get(
-                               "YDFA_HIST_FILTER_SUB", "0") == "1")
+                           filtered_hint=bool(use_sub and level > 0),
+                           grp_scratch=self.grp_buf)
         if self.distributed and use_sub and derived is not None \
                 and level > 0 and not getattr(self, "capturing", False):
             # derived slots are still all-zero here: all-reduce only the

@@ -226,7 +226,49 @@ __global__ void hist_build_lds_kernel(const uint8_t* __restrict__ bins,
   int64_t i = row0 + threadIdx.x;
   const int64_t bulk_end = row1 - (kHistUnroll - 1) * stride;
 
-  if (filtered && row_grp != nullptr) {
+  if (filtered == 2 && row_grp != nullptr) {
+    // slot8 mode: row_grp holds the per-row GLOBAL slot id precomputed
+    // once per level (255 = row not built this level, e.g. the derived
+    // sibling under histogram subtraction, or subsampled out). Streams
+    // 1 B/row for the filter instead of node_ids (4 B) + slot-map
+    // lookup; gh/bins load only under the match predicate.
+    for (; i < bulk_end; i += kHistUnroll * stride) {
+      uint8_t gb[kHistUnroll];
+#pragma unroll
+      for (int u = 0; u < kHistUnroll; ++u) gb[u] = row_grp[i + u * stride];
+#pragma unroll
+      for (int u = 0; u < kHistUnroll; ++u) {
+        const int slot = (int)gb[u] - slot0;
+        if (slot < 0 || slot >= n_slots) continue;
+        const float2 v = gh[i + u * stride];
+        const unsigned long long hq =
+            (unsigned long long)(v.y * kHScale + 0.5f);
+        const unsigned long long pk =
+            hq | ((unsigned long long)(v.y != 0.f) << 44);
+        for (int j = 0; j < nf; ++j) {
+          const int b = bins[(int64_t)(f0 + j) * N + i + u * stride];
+          const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+          atomicAdd(lg + cell, (double)v.x);
+          atomicAdd(lp + cell, pk);
+        }
+      }
+    }
+    for (; i < row1; i += stride) {
+      const int slot = (int)row_grp[i] - slot0;
+      if (slot < 0 || slot >= n_slots) continue;
+      const float2 v = gh[i];
+      const unsigned long long hq =
+          (unsigned long long)(v.y * kHScale + 0.5f);
+      const unsigned long long pk =
+          hq | ((unsigned long long)(v.y != 0.f) << 44);
+      for (int j = 0; j < nf; ++j) {
+        const int b = bins[(int64_t)(f0 + j) * N + i];
+        const int cell = 2 * (((j * n_slots) + slot) * n_bins + b);
+        atomicAdd(lg + cell, (double)v.x);
+        atomicAdd(lp + cell, pk);
+      }
+    }
+  } else if (filtered && row_grp != nullptr) {
     // Cheapest multi-group pass: 1-byte group id per row; only matching
     // rows touch node_ids/gh/bins.
     const uint8_t my_grp = (uint8_t)(slot0 / n_slots_group);
@@ -1415,14 +1457,46 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
                        slot_map, grp_scratch, N, level_base, level_size,
                        group);
   }
+  // slot8 mode (single-launch levels where the caller hints that many
+  // rows are not built, i.e. histogram-subtraction levels): precompute a
+  // per-row u8 slot id so every feature-group block streams 1 B/row for
+  // the filter instead of re-reading node_ids (4 B) + the slot map.
+  // MEASURED SLOWER at every shard size (11M: 185 vs 286 trees/s;
+  // 1.375M: 999 vs 1248 — gpurun_out/sweep_slot8.log): the predicated
+  // gh/bins loads break the eager unrolled pipeline, same failure mode
+  // as YDFA_HIST_FILTER_SUB. Opt-in only.
+  static int slot8_enabled = -1;
+  if (slot8_enabled < 0) {
+    const char* e = getenv("YDFA_HIST_SLOT8");
+    slot8_enabled = e ? atoi(e) : 0;
+  }
+  // node-id-first filtered pass for single-group hinted levels: opt-in
+  // (measured slower than the eager unfiltered pass; slot8 supersedes it)
+  static int filter_sub_env = -1;
+  if (filter_sub_env < 0) {
+    const char* e = getenv("YDFA_HIST_FILTER_SUB");
+    filter_sub_env = e ? atoi(e) : 0;
+  }
+  const int use_slot8 = (slot8_enabled && filtered_hint && !use_grp &&
+                         grp_scratch != nullptr && group == n_slots &&
+                         slot0 + n_slots <= 255) ? 1 : 0;
+  if (use_slot8) {
+    hipLaunchKernelGGL(row_group_kernel, dim3(elem_grid(N, 4096)),
+                       dim3(kBlock), 0, (hipStream_t)stream, node_ids,
+                       slot_map, grp_scratch, N, level_base, level_size,
+                       1);
+  }
   for (int s0 = 0; s0 < n_slots; s0 += group) {
     const int ng = (n_slots - s0) < group ? (n_slots - s0) : group;
     const int this_fpb = (ng == n_slots) ? fpb : 1;
     const int this_nfg = (F + this_fpb - 1) / this_fpb;
     const size_t lds = (size_t)this_fpb * ng * n_bins * 16 +
                        (lds_map ? map_bytes_full : 0);
-    const int filtered = (n_slots > group || filtered_hint) ? 1 : 0;
-    const uint8_t* rg = use_grp ? grp_scratch : nullptr;
+    const int filtered =
+        use_slot8 ? 2
+                  : ((n_slots > group || (filtered_hint && filter_sub_env))
+                         ? 1 : 0);
+    const uint8_t* rg = (use_grp || use_slot8) ? grp_scratch : nullptr;
     if (swizzle) {
       const int grid_flat = ((chunks + 7) / 8) * 8 * this_nfg;
       hipLaunchKernelGGL(hist_build_lds_kernel, dim3(grid_flat),
