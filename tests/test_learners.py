@@ -1168,3 +1168,30 @@ def test_conditional_expectations(binary_data):
     mp = ce.mean_prediction[ce.counts > 50]
     assert abs(float(mp[-1]) - float(mp[0])) > 0.3
     assert "Conditional expectation" in an._repr_html_()
+
+
+def test_weight_rescale_semantics_pinned():
+    """The automatic heavy-weight rescale (w -> w * 8/max when max > 8;
+    packed-u64 histogram field needs per-example h bounded,
+    ops/cc/train_kernels.hip) is EXACTLY 'train on the scaled weights':
+    pre-scaling the weights by the same factor must give the identical
+    model, including with lambda_l2 > 0 where G/(H+lambda) is not
+    weight-scale-invariant (documented deviation, ROADMAP.md)."""
+    rng = np.random.RandomState(11)
+    n = 6000
+    x1 = rng.randn(n).astype(np.float32)
+    x2 = rng.randn(n).astype(np.float32)
+    y = (x1 + 0.5 * x2 + 0.2 * rng.randn(n)) > 0
+    w = rng.uniform(0.5, 50.0, n).astype(np.float32)  # max > 8
+    base = {"x1": x1, "x2": x2, "label": np.where(y, "a", "b")}
+    kw = dict(label="label", weights="w", num_trees=15, max_depth=4,
+              l2_regularization=2.0, validation_ratio=0.0, device="cpu")
+    m_auto = ydf.GradientBoostedTreesLearner(**kw).train(
+        {**base, "w": w})
+    m_pre = ydf.GradientBoostedTreesLearner(**kw).train(
+        {**base, "w": w * (8.0 / float(w.max()))})
+    np.testing.assert_array_equal(m_auto.forest.feat, m_pre.forest.feat)
+    np.testing.assert_allclose(m_auto.forest.thr, m_pre.forest.thr,
+                               rtol=1e-6, atol=1e-7)
+    np.testing.assert_allclose(m_auto.predict(base),
+                               m_pre.predict(base), rtol=1e-6, atol=1e-7)
