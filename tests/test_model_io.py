@@ -248,3 +248,28 @@ def test_plot_tree(trained, tmp_path):
     f = str(tmp_path / "tree.svg")
     plot.to_file(f)
     assert "<svg" in open(f).read()
+
+
+def test_distance_proximity(binary_data):
+    """model.distance (reference PYDF decision_forest_model.py:196):
+    tree-ensemble proximity — 0 on the diagonal, in [0,1], symmetric
+    for one dataset, and near-duplicate examples are closer than
+    random pairs."""
+    m = ydf.RandomForestLearner(label="label", num_trees=30,
+                                max_depth=10).train(binary_data)
+    sub = {k: v[:200] for k, v in binary_data.items()}
+    d = m.distance(sub)
+    assert d.shape == (200, 200)
+    np.testing.assert_allclose(np.diag(d), 0.0, atol=1e-7)
+    assert d.min() >= 0.0 and d.max() <= 1.0
+    np.testing.assert_allclose(d, d.T, atol=1e-7)
+    # a tiny perturbation of an example stays closer than average
+    pert = {k: (v[:50].astype(np.float32) + 1e-4
+                if v.dtype.kind == "f" else v[:50])
+            for k, v in binary_data.items()}
+    dp = m.distance({k: v[:50] for k, v in binary_data.items()}, pert)
+    assert np.diag(dp).mean() < d.mean() * 0.2
+
+    # leaf_indices shape/type
+    li = m.leaf_indices(sub)
+    assert li.shape == (200, m.num_trees()) and li.dtype == np.int32

@@ -83,6 +83,74 @@ class DecisionForestModel(GenericModel):
         del trees[idx]
         self._rebuild_from_trees(trees)
 
+    def leaf_indices(self, data) -> np.ndarray:
+        """[n, num_trees] i32: the leaf NODE index each example reaches
+        in each tree (vectorized recursive partition walk; handles
+        numerical, categorical-mask, oblique and NA-routing
+        conditions)."""
+        f = self.forest
+        if f.has_set_conditions:
+            raise NotImplementedError(
+                "leaf_indices/distance on categorical-SET models is not "
+                "supported")
+        X = self._encode_features(data)
+        n = X.shape[1]
+        out = np.empty((n, f.n_trees), dtype=np.int32)
+        has_na = f.has_na_routing
+
+        def rec(node, idx, col):
+            fi = int(f.feat[node])
+            if fi < 0:
+                col[idx] = node
+                return
+            ci = int(f.cat_idx[node])
+            if ci >= 0:
+                cb = X[fi, idx].astype(np.int64).clip(0, 255)
+                right = ((f.masks[ci][cb >> 6]
+                          >> (cb & 63).astype(np.uint64))
+                         & np.uint64(1)).astype(bool)
+            elif ci <= -2:
+                oi = -(ci + 2)
+                s0, nn = int(f.obl_ranges[oi, 0]), int(f.obl_ranges[oi, 1])
+                acc = (f.obl_w[s0:s0 + nn, None]
+                       * X[f.obl_attr[s0:s0 + nn]][:, idx]).sum(axis=0)
+                right = acc > f.thr[node]
+            else:
+                xv = X[fi, idx]
+                right = xv > f.thr[node]
+                if has_na:
+                    nanm = np.isnan(xv)
+                    if nanm.any():
+                        right = np.where(nanm, bool(f.na_right[node]),
+                                         right)
+            left = int(f.left[node])
+            rec(left, idx[~right], col)
+            rec(left + 1, idx[right], col)
+
+        all_idx = np.arange(n)
+        for t in range(f.n_trees):
+            rec(int(f.roots[t]), all_idx, out[:, t])
+        return out
+
+    def distance(self, data1, data2=None) -> np.ndarray:
+        """Pairwise tree-ensemble distance between examples (mirrors
+        PYDF model.distance, port/python/ydf/model/
+        decision_forest_model/decision_forest_model.py:196): 1 minus
+        the fraction of trees in which the two examples reach the same
+        leaf. distance[i, j] is between example i of data1 and example
+        j of data2 (data2 defaults to data1). In [0, 1]; not a metric
+        (no triangle inequality)."""
+        l1 = self.leaf_indices(data1)
+        l2 = l1 if data2 is None else self.leaf_indices(data2)
+        n1, T = l1.shape
+        n2 = l2.shape[0]
+        d = np.empty((n1, n2), dtype=np.float32)
+        step = max(1, 20_000_000 // max(T * max(n2, 1), 1))
+        for i0 in range(0, n1, step):
+            eq = l1[i0:i0 + step, None, :] == l2[None, :, :]
+            d[i0:i0 + step] = 1.0 - eq.mean(axis=-1, dtype=np.float32)
+        return d
+
 
 class GradientBoostedTreesModel(DecisionForestModel):
     _model_type = "GRADIENT_BOOSTED_TREES"
