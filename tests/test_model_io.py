@@ -273,3 +273,26 @@ def test_distance_proximity(binary_data):
     # leaf_indices shape/type
     li = m.leaf_indices(sub)
     assert li.shape == (200, m.num_trees()) and li.dtype == np.int32
+
+
+def test_reference_api_compat_surface(trained, binary_data):
+    """Reference-API compatibility: label_classes/metadata/
+    training_logs answer both attribute and method call forms; models
+    pickle; predict_leaves/iter_trees/set_data_spec/set_node_format
+    exist (PYDF generic_model.py / decision_forest_model.py)."""
+    import pickle
+
+    assert trained.label_classes == trained.label_classes()
+    assert trained.metadata == trained.metadata()
+    blob = pickle.dumps(trained)
+    m2 = pickle.loads(blob)
+    np.testing.assert_allclose(trained.predict(binary_data),
+                               m2.predict(binary_data), rtol=1e-6)
+    leaves = trained.predict_leaves(
+        {k: v[:64] for k, v in binary_data.items()})
+    assert leaves.shape == (64, trained.num_trees())
+    assert sum(1 for _ in trained.iter_trees()) == trained.num_trees()
+    trained.set_node_format("BLOB_SEQUENCE")
+    with pytest.raises(ValueError):
+        trained.set_node_format("TFE_RECORDIO")
+    trained.set_data_spec(trained.data_spec())

@@ -67,6 +67,20 @@ class _DeviceForest:
                                                 self.left, ci)
 
 
+class _CallableList(list):
+    """list that also answers `x()` — reference-API compatibility
+    (PYDF exposes label_classes/training_logs as METHODS; this
+    framework historically as attributes; both forms work here)."""
+
+    def __call__(self):
+        return self
+
+
+class _CallableDict(dict):
+    def __call__(self):
+        return self
+
+
 class GenericModel:
     """Base decision-forest model."""
 
@@ -90,6 +104,62 @@ class GenericModel:
         self.training_logs = None
         self.tuner_logs = None
         self._self_evaluation = None
+
+    # reference-API compat: these read as attributes AND call as methods
+    @property
+    def label_classes(self):
+        return self._label_classes
+
+    @label_classes.setter
+    def label_classes(self, v):
+        self._label_classes = _CallableList(v) if isinstance(
+            v, (list, tuple)) else v
+
+    @property
+    def metadata(self):
+        return self._metadata
+
+    @metadata.setter
+    def metadata(self, v):
+        self._metadata = _CallableDict(v) if isinstance(v, dict) else v
+
+    @property
+    def training_logs(self):
+        return self._training_logs
+
+    @training_logs.setter
+    def training_logs(self, v):
+        self._training_logs = _CallableList(v) if isinstance(
+            v, (list, tuple)) else v
+
+    def __getstate__(self):
+        # models pickle (reference PYDF __getstate__/__setstate__ via
+        # serialized blob); device/engine caches are rebuilt lazily
+        d = dict(self.__dict__)
+        d["_dev_forest"] = {}
+        return d
+
+    def __setstate__(self, d):
+        self.__dict__.update(d)
+        self._dev_forest = {}
+
+    def set_data_spec(self, data_spec) -> None:
+        """Replaces the dataspec (PYDF model.set_data_spec); engine
+        caches are invalidated because thresholds/vocabularies may no
+        longer match the packed tables."""
+        self.dataspec = data_spec
+        self._dev_forest.clear()
+        if hasattr(self, "_thr_on_cuts"):
+            del self._thr_on_cuts
+
+    def set_node_format(self, node_format: str) -> None:
+        """On-disk node container format (PYDF model.set_node_format).
+        Only BLOB_SEQUENCE (the reference default) is written."""
+        if str(node_format).upper() not in ("BLOB_SEQUENCE",):
+            raise ValueError(
+                f"unsupported node format {node_format!r}; the exporter "
+                "writes BLOB_SEQUENCE (reference blob_sequence.h)")
+        self.metadata["node_format"] = str(node_format).upper()
 
     # ------------------------------------------------------------------
     def task(self) -> Task:

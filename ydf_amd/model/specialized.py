@@ -83,6 +83,16 @@ class DecisionForestModel(GenericModel):
         del trees[idx]
         self._rebuild_from_trees(trees)
 
+    def iter_trees(self):
+        """Iterator over the trees (PYDF model.iter_trees)."""
+        for i in range(self.forest.n_trees):
+            yield self.get_tree(i)
+
+    def predict_leaves(self, data) -> np.ndarray:
+        """Index of the active leaf per tree, [n, num_trees] (PYDF
+        model.predict_leaves). Alias of leaf_indices."""
+        return self.leaf_indices(data)
+
     def leaf_indices(self, data) -> np.ndarray:
         """[n, num_trees] i32: the leaf NODE index each example reaches
         in each tree (vectorized recursive partition walk; handles
