@@ -400,3 +400,16 @@ def test_evaluation_binary_statistics(binary_data):
     # trapezoid integral of the curve must approximate the rank AUC
     auc_trap = float(np.trapz(ch["tpr"], ch["fpr"]))
     assert abs(auc_trap - ev.auc) < 0.01
+
+
+def test_learner_name_and_input_feature_names(binary_data):
+    """PYDF learner.learner_name + extract_input_feature_names."""
+    ln = ydf.GradientBoostedTreesLearner(label="label")
+    assert ln.learner_name == "GRADIENT_BOOSTED_TREES"
+    assert ydf.RandomForestLearner(label="label").learner_name == \
+        "RANDOM_FOREST"
+    names = ln.extract_input_feature_names(binary_data)
+    assert sorted(names) == ["x1", "x2", "x3"]
+    ln2 = ydf.GradientBoostedTreesLearner(label="label",
+                                          features=["x2", "nope"])
+    assert ln2.extract_input_feature_names(binary_data) == ["x2"]

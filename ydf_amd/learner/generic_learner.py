@@ -58,6 +58,41 @@ class GenericLearner:
     def task(self) -> Task:
         return self._task
 
+    @property
+    def learner_name(self) -> str:
+        """Registry name of this learner, e.g. "RANDOM_FOREST" (PYDF
+        learner.learner_name)."""
+        from ydf_amd.utils import registry
+
+        registry._bootstrap()
+        for name, cls in registry.learner_registry._items.items():
+            if cls is type(self):
+                return name
+        # fall back to CamelCase -> SNAKE of the class name
+        import re
+
+        return re.sub(r"(?<!^)(?=[A-Z])", "_",
+                      type(self).__name__.replace("Learner", "")).upper()
+
+    def extract_input_feature_names(self, ds) -> list:
+        """Input-feature column names of `ds` for this learner: every
+        column except the label/weights/group special columns, or the
+        explicit `features` list filtered to what the data provides
+        (PYDF learner.extract_input_feature_names)."""
+        from ydf_amd.dataset.dataset import _to_column_dict
+
+        cols = _to_column_dict(ds)
+        non_input = {self.label, self.weights_col,
+                     getattr(self, "ranking_group", None),
+                     getattr(self, "uplift_treatment", None),
+                     getattr(self, "label_event_observed", None),
+                     getattr(self, "label_entry_age", None)}
+        if self.features is not None:
+            want = [f if isinstance(f, str) else f.name
+                    for f in self.features]
+            return [f for f in want if f in cols]
+        return [c for c in cols if c not in non_input]
+
     # ------------------------------------------------------------------
     def _resolve_device(self) -> torch.device:
         if self.device is not None:
