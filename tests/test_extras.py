@@ -413,3 +413,42 @@ def test_learner_name_and_input_feature_names(binary_data):
     ln2 = ydf.GradientBoostedTreesLearner(label="label",
                                           features=["x2", "nope"])
     assert ln2.extract_input_feature_names(binary_data) == ["x2"]
+
+
+def test_log_book(tmp_path):
+    """ydf.util.LogBook (reference util/log_book.py): SQLite-backed
+    experiment tracking with default keys and superset filtering."""
+    lb = ydf.util.LogBook(str(tmp_path / "lb"),
+                          print_num_experiments=False,
+                          default_keys={"project": "demo"})
+    key = {"param1": 1, "param2": "abc"}
+    assert not lb.exist(key)
+    lb.add(key, {"accuracy": 0.9, "obs": [1, 2, 3]})
+    assert lb.exist(key)
+    assert lb.count_key(key) == 1
+    assert lb.num_experiments() == 1
+    with pytest.raises(ValueError):
+        lb.add(key, {"accuracy": 0.91})  # duplicate key
+    with pytest.raises(ValueError):
+        lb.add({"id": 1}, {})  # reserved key
+    lb.add({"param1": 2, "param2": "x"}, {"accuracy": 0.8})
+    df = lb.to_dataframe()
+    assert len(df) == 2 and "accuracy" in df.columns
+    assert (df["project"] == "demo").all()
+    assert len(lb.to_dataframe({"param1": 1})) == 1
+    # reopen from disk
+    lb2 = ydf.util.LogBook(str(tmp_path / "lb"),
+                           print_num_experiments=False,
+                           default_keys={"project": "demo"})
+    assert lb2.num_experiments() == 2
+
+
+def test_util_tf_record_roundtrip(tmp_path):
+    """ydf.util.read_tf_record / write_tf_record."""
+    cols = {"x": np.arange(5, dtype=np.float32),
+            "name": np.array(["a", "b", "c", "d", "e"])}
+    p = str(tmp_path / "data.tfrecord")
+    ydf.util.write_tf_record(cols, p)
+    back = ydf.util.read_tf_record(p)
+    np.testing.assert_allclose(back["x"], cols["x"])
+    assert list(back["name"].astype(str)) == list(cols["name"])

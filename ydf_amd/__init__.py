@@ -167,12 +167,52 @@ from ydf_amd.utils import folds as _folds  # noqa: E402
 from ydf_amd.utils import usage as _usage  # noqa: E402
 
 
+def _util_read_tf_record(path, *, compressed=None, process=None,
+                         verbose=False, threads=None):
+    """ydf.util.read_tf_record analogue: TFRecord shards -> dict of
+    numpy columns (compression auto-detected; `process` maps raw
+    decoded row dicts before column assembly)."""
+    from ydf_amd.dataset.dataset import expand_sharded_paths
+    from ydf_amd.dataset.tfrecord import read_tfrecord_columns
+
+    p = str(path)
+    if not p.startswith(("tfrecord:", "tfrecord+gzip:")):
+        p = "tfrecord:" + p
+    _, paths = expand_sharded_paths(p)
+    cols = read_tfrecord_columns(paths)
+    if process is not None:
+        import numpy as _np
+
+        n = len(next(iter(cols.values()))) if cols else 0
+        rows = [{k: v[i] for k, v in cols.items()} for i in range(n)]
+        rows = [r for r in (process(dict(r)) for r in rows)
+                if r is not None]
+        cols = {k: _np.asarray([r[k] for r in rows])
+                for k in (rows[0] if rows else {})}
+    return cols
+
+
+def _util_write_tf_record(data, path, *, compressed=None):
+    """ydf.util.write_tf_record analogue: dict of columns -> TFRecord."""
+    from ydf_amd.dataset.tfrecord import write_tfrecord_columns
+
+    write_tfrecord_columns(str(path), data,
+                           compress=bool(compressed))
+
+
 class util:  # noqa: N801  (PYDF exposes a lowercase `util` namespace)
     """Utility namespace (ydf.util analogue)."""
 
     generate_folds = staticmethod(_folds.generate_folds)
     fold_splits = staticmethod(_folds.fold_splits)
     usage = _usage
+    read_tf_record = staticmethod(_util_read_tf_record)
+    write_tf_record = staticmethod(_util_write_tf_record)
+
+
+from ydf_amd.utils.log_book import LogBook as _LogBook  # noqa: E402
+
+util.LogBook = _LogBook
 
 version = "2.0.0+mi355x"
 __version__ = version
