@@ -452,3 +452,47 @@ def test_util_tf_record_roundtrip(tmp_path):
     back = ydf.util.read_tf_record(p)
     np.testing.assert_allclose(back["x"], cols["x"])
     assert list(back["name"].astype(str)) == list(cols["name"])
+
+
+def test_evaluation_reference_surface(binary_data, regression_data):
+    """Reference Evaluation surface: confusion_matrix with class names,
+    num_examples_weighted, Characteristic precision/recall accessors +
+    precision_at_recall, html(), regression bootstrap RMSE CI."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=30, validation_ratio=0.0).train(
+        binary_data)
+    ev = m.evaluate(binary_data)
+    cm = ev.confusion_matrix
+    assert cm.matrix.shape == (2, 2)
+    assert set(cm.classes) == set(m.label_classes)
+    assert cm.value(cm.classes[0], cm.classes[0]) == cm.matrix[0, 0]
+    assert ev.num_examples_weighted == ev.num_examples
+    ch = ev.characteristics[0]
+    assert len(ch.precisions) == len(ch.recalls) == len(ch.thresholds)
+    assert 0.0 <= ch.precision_at_recall(0.5) <= 1.0
+    assert ch.precision_at_recall(0.0) == 1.0
+    assert ch.roc_auc == ev.auc
+    assert "<table>" in ev.html()
+
+    mr = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=30,
+        validation_ratio=0.0).train(regression_data)
+    evr = mr.evaluate(regression_data)
+    lo, hi = evr.rmse_ci95_bootstrap
+    assert lo <= evr.rmse <= hi
+
+
+def test_ranking_map_mrr():
+    rng = np.random.RandomState(3)
+    n = 4000
+    g = np.repeat(np.arange(n // 8), 8)
+    x = rng.randn(n).astype(np.float32)
+    rel = np.clip((x + rng.randn(n) * 0.3) * 2, 0, 4).astype(np.float32)
+    d = {"x": x, "rel": rel, "g": g}
+    m = ydf.GradientBoostedTreesLearner(
+        label="rel", task=ydf.Task.RANKING, ranking_group="g",
+        num_trees=30, validation_ratio=0.0).train(d)
+    ev = m.evaluate(d)
+    assert ev.ndcg > 0.75
+    assert 0.0 < ev.map <= 1.0
+    assert 0.0 < ev.mrr <= 1.0

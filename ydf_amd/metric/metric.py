@@ -135,6 +135,25 @@ def mrr(labels: np.ndarray, scores: np.ndarray, groups: np.ndarray,
     return total / n_groups if n_groups else float("nan")
 
 
+def mean_average_precision(labels: np.ndarray, scores: np.ndarray,
+                           groups: np.ndarray) -> float:
+    """Ranking MAP: mean over groups of average precision with binary
+    relevance label > 0 (reference ranking metric "map")."""
+    total, n_groups = 0.0, 0
+    for g in np.unique(groups):
+        m = groups == g
+        rel = labels[m] > 0
+        if not rel.any():
+            continue
+        order = np.argsort(-scores[m], kind="mergesort")
+        r = rel[order]
+        hits = np.cumsum(r)
+        prec_at_hit = hits[r] / (np.nonzero(r)[0] + 1)
+        total += float(prec_at_hit.mean())
+        n_groups += 1
+    return total / n_groups if n_groups else float("nan")
+
+
 def accuracy_confidence_interval(acc: float, n: int,
                                  level: float = 0.95):
     """Wilson score interval (closed form; reference
@@ -191,6 +210,72 @@ def bootstrap_confidence_intervals(labels: np.ndarray,
 
 
 @dataclasses.dataclass
+class Characteristic:
+    """Per-threshold binary-classification curve (mirrors PYDF
+    metric.Characteristic): ROC points plus derived precision/recall
+    arrays from the stored class counts. Also answers dict-style
+    access (ch["fpr"]) for backward compatibility."""
+
+    name: str
+    fpr: np.ndarray
+    tpr: np.ndarray
+    thresholds: np.ndarray
+    n_pos: int = 0
+    n_neg: int = 0
+    roc_auc: Optional[float] = None
+    pr_auc: Optional[float] = None
+
+    def __getitem__(self, k):
+        return getattr(self, k)
+
+    @property
+    def recalls(self) -> np.ndarray:
+        return self.tpr
+
+    @property
+    def false_positive_rates(self) -> np.ndarray:
+        return self.fpr
+
+    @property
+    def precisions(self) -> np.ndarray:
+        tp = self.tpr * self.n_pos
+        fp = self.fpr * self.n_neg
+        with np.errstate(invalid="ignore", divide="ignore"):
+            return np.where(tp + fp > 0, tp / (tp + fp), 1.0)
+
+    @property
+    def accuracies(self) -> np.ndarray:
+        tp = self.tpr * self.n_pos
+        tn = self.n_neg - self.fpr * self.n_neg
+        return (tp + tn) / max(self.n_pos + self.n_neg, 1)
+
+    def precision_at_recall(self, recall: float) -> float:
+        if recall <= 0.0:
+            return 1.0
+        m = self.recalls >= recall
+        return float(self.precisions[m].max()) if m.any() else 0.0
+
+
+@dataclasses.dataclass
+class ConfusionMatrix:
+    """Confusion matrix with class names (mirrors PYDF
+    metric.ConfusionMatrix): rows = truth, cols = prediction."""
+
+    classes: tuple
+    matrix: np.ndarray
+
+    def value(self, label, prediction) -> float:
+        return float(self.matrix[self.classes.index(label),
+                                 self.classes.index(prediction)])
+
+    def __str__(self) -> str:
+        head = "truth\\pred " + " ".join(str(c) for c in self.classes)
+        rows = [f"{c} " + " ".join(str(int(v)) for v in self.matrix[i])
+                for i, c in enumerate(self.classes)]
+        return "\n".join([head] + rows)
+
+
+@dataclasses.dataclass
 class Evaluation:
     """Evaluation report (mirrors ydf.metric.Evaluation fields)."""
 
@@ -207,6 +292,15 @@ class Evaluation:
     qini: Optional[float] = None
     cindex: Optional[float] = None
     confusion: Optional[np.ndarray] = None
+    # label class names (classification; feeds confusion_matrix)
+    classes: Optional[tuple] = None
+    num_examples_weighted: Optional[float] = None
+    custom_metrics: Optional[Dict] = None
+    # ranking mean average precision (reference ranking metric "map")
+    map: Optional[float] = None
+    # regression bootstrap 95% CI on RMSE (reference
+    # bootstrap_rmse_*_bounds_95p)
+    rmse_ci95_bootstrap: Optional[tuple] = None
     # per-threshold ROC points for binary classification (PYDF
     # evaluation.characteristics; reference metric Roc curves):
     # list of dicts {"name", "fpr", "tpr", "thresholds"}
@@ -280,6 +374,19 @@ class Evaluation:
                 parts.append(f"{k}: [{v[0]:.6g}, {v[1]:.6g}]")
         return "\n".join(parts)
 
+    @property
+    def confusion_matrix(self):
+        """ConfusionMatrix with class names (PYDF
+        evaluation.confusion_matrix); None for non-classification."""
+        if self.confusion is None:
+            return None
+        cls = self.classes or tuple(range(self.confusion.shape[0]))
+        return ConfusionMatrix(classes=tuple(cls), matrix=self.confusion)
+
+    def html(self) -> str:
+        """HTML report (PYDF evaluation.html())."""
+        return self._repr_html_()
+
     def _repr_html_(self) -> str:
         rows = "".join(
             f"<tr><td>{k}</td><td>{v:.6g}</td></tr>"
@@ -319,14 +426,19 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
 
     ev = Evaluation(num_examples=len(labels))
     w = None if weights is None else np.asarray(weights, np.float64)
+    ev.num_examples_weighted = float(w.sum()) if w is not None \
+        else float(len(labels))
     if task == Task.CLASSIFICATION:
         if predictions.ndim == 1:
             pred_cls = (predictions >= 0.5).astype(np.int64)
             ev.auc = roc_auc(labels > 0.5, predictions)
             fpr, tpr, thr = roc_curve(labels > 0.5, predictions)
-            ev.characteristics = [{"name": "default", "fpr": fpr,
-                                   "tpr": tpr, "thresholds": thr}]
             ev.pr_auc = pr_auc(labels > 0.5, predictions)
+            np_ = int((labels > 0.5).sum())
+            ev.characteristics = [Characteristic(
+                name="default", fpr=fpr, tpr=tpr, thresholds=thr,
+                n_pos=np_, n_neg=len(labels) - np_, roc_auc=ev.auc,
+                pr_auc=ev.pr_auc)]
         else:
             pred_cls = predictions.argmax(axis=1)
         correct = (labels.astype(np.int64) == pred_cls)
@@ -358,9 +470,12 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
         # labels: 1 = anomaly; predictions: anomaly score in [0, 1]
         ev.auc = roc_auc(labels > 0.5, predictions)
         fpr, tpr, thr = roc_curve(labels > 0.5, predictions)
-        ev.characteristics = [{"name": "default", "fpr": fpr,
-                               "tpr": tpr, "thresholds": thr}]
         ev.pr_auc = pr_auc(labels > 0.5, predictions)
+        np_ = int((labels > 0.5).sum())
+        ev.characteristics = [Characteristic(
+            name="default", fpr=fpr, tpr=tpr, thresholds=thr,
+            n_pos=np_, n_neg=len(labels) - np_, roc_auc=ev.auc,
+            pr_auc=ev.pr_auc)]
     elif task == Task.REGRESSION:
         if w is None:
             ev.rmse = rmse(labels, predictions)
@@ -370,4 +485,14 @@ def evaluate_predictions(predictions: np.ndarray, labels: np.ndarray,
             ev.rmse = float(np.sqrt((err ** 2 * w).sum() / w.sum()))
             ev.mae = float((np.abs(err) * w).sum() / w.sum())
         ev.loss = ev.rmse ** 2
+        # bootstrap 95% CI on RMSE (reference
+        # bootstrap_rmse_lower/upper_bounds_95p, 199 resamples)
+        se = ((labels - predictions).astype(np.float64)) ** 2
+        if len(se) > 1:
+            rng = np.random.RandomState(1234)
+            idx = rng.randint(0, len(se), size=(199, len(se)))
+            boots = np.sqrt(se[idx].mean(axis=1))
+            ev.rmse_ci95_bootstrap = (
+                float(np.percentile(boots, 2.5)),
+                float(np.percentile(boots, 97.5)))
     return ev

@@ -708,15 +708,22 @@ class GenericModel:
             return ev
         ev = evaluate_predictions(preds, labels, self._task, n_classes,
                                   weights=w)
+        if self._task == Task.CLASSIFICATION and self.label_classes:
+            ev.classes = tuple(self.label_classes)
         if self._task == Task.RANKING:
             gcol = (self.metadata or {}).get("ranking_group")
             if gcol and cols is not None and gcol in cols:
+                from ydf_amd.metric.metric import mean_average_precision
+                from ydf_amd.metric.metric import mrr as mrr_fn
                 from ydf_amd.metric.metric import ndcg as ndcg_fn
 
+                g = np.asarray(cols[gcol])
                 ev.ndcg = ndcg_fn(
-                    labels, preds, np.asarray(cols[gcol]),
+                    labels, preds, g,
                     truncation=(self.metadata or {}).get(
                         "ndcg_truncation", 5))
+                ev.mrr = mrr_fn(labels, preds, g)
+                ev.map = mean_average_precision(labels, preds, g)
                 ev.loss = -ev.ndcg
         return ev
 
