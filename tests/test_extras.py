@@ -496,3 +496,21 @@ def test_ranking_map_mrr():
     assert ev.ndcg > 0.75
     assert 0.0 < ev.map <= 1.0
     assert 0.0 < ev.mrr <= 1.0
+
+
+def test_tuner_optimize_metric(binary_data):
+    """RandomSearchTuner(optimize_metric=...) (PYDF OptimizeMetric):
+    trial selection by evaluated metric instead of validation loss."""
+    from ydf_amd.learner.tuner import OptimizeMetric
+
+    t = ydf.RandomSearchTuner(num_trials=3, seed=7,
+                              optimize_metric=OptimizeMetric.ACCURACY)
+    t.choice("max_depth", [2, 4])
+    t.choice("shrinkage", [0.05, 0.15])
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15, tuner=t).train(binary_data)
+    assert m.tuner_logs is not None
+    assert len(m.tuner_logs.trials) == 3
+    best = m.tuner_logs.best_trial
+    # accuracy objective: scores are accuracies in [0, 1]
+    assert 0.5 < best.score <= 1.0

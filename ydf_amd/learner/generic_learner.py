@@ -429,13 +429,15 @@ class GenericLearner:
         """Random-search trials; keeps the best model by validation loss
         (reference hyperparameters_optimizer.cc random trials)."""
         from ydf_amd.learner.tuner import (OptimizerLogs, TrialLog,
-                                           run_parallel_trials)
+                                           run_parallel_trials,
+                                           trial_score)
 
         if getattr(self.tuner, "parallel_trials", 1) > 1:
             # one trial per process slot (one per GPU when available):
             # the MI355X mapping of the reference's distributed HPO
             # (hyperparameters_optimizer.h:46 + generic_worker/)
-            return run_parallel_trials(self, self.tuner, data)
+            return run_parallel_trials(self, self.tuner, data,
+                                       valid=valid)
         rng = np.random.RandomState(self.tuner.seed)
         tuner = self.tuner
         best = None
@@ -452,10 +454,9 @@ class GenericLearner:
                 if self.hyperparameters.get("validation_ratio", 0) == 0:
                     self.hyperparameters["validation_ratio"] = 0.1
                 model = self.train(data, valid=valid)
-                vloss = None
-                if model.training_logs:
-                    vloss = model.training_logs[-1].get("valid_loss")
-                score = -(vloss if vloss is not None else float("inf"))
+                score = trial_score(
+                    model, valid if valid is not None else data,
+                    getattr(tuner, "optimize_metric", None))
                 logs.append(TrialLog(hyperparameters=hp, score=score))
                 if best is None or score > best[0]:
                     best = (score, model)
