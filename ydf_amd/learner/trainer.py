@@ -1000,32 +1000,27 @@ class ForestTrainer:
         ~50-75 us/tree that dominates small per-rank shards (the
         8-GPU strong-scaling regime)."""
         T = self.tree_feat.numel()
-        srcs = [(self.tree_feat, np.int32), (self.tree_bin, np.int32),
-                (self.leaf_vals[:T], np.float32),
-                (self.node_stats[:, 2].contiguous(), np.float32),
-                (self.tree_gain, np.float32)]
-        if self.tree_masks is not None:
-            srcs.append((self.tree_masks.view(-1), np.int64))
-        if self.tree_na is not None:
-            srcs.append((self.tree_na, np.uint8))
-        nbytes = [t.numel() * t.element_size() for t, _ in srcs]
-        total = sum(nbytes)
+        W = self.tree_masks.numel() if self.tree_masks is not None else 0
+        moff = (20 * T + 7) & ~7  # masks start 8-byte aligned
+        total = moff + 8 * W + (T if self.tree_na is not None else 0)
         stage = getattr(self, "_extract_stage", None)
         if stage is None or stage.numel() < total:
             stage = torch.empty(total, dtype=torch.uint8,
                                 device=self.device)
             self._extract_stage = stage
-        off = 0
-        for (t, _), nb in zip(srcs, nbytes):
-            stage[off:off + nb].copy_(
-                t.contiguous().view(torch.uint8).view(-1))
-            off += nb
+        ops.pack_extract(self.tree_feat, self.tree_bin, self.leaf_vals,
+                         self.node_stats, self.tree_gain,
+                         self.tree_masks, self.tree_na, stage, T, W)
         host = stage[:total].cpu().numpy()
-        out = []
-        off = 0
-        for (t, dt), nb in zip(srcs, nbytes):
-            out.append(host[off:off + nb].view(dt).copy())
-            off += nb
+        out = [host[0:4 * T].view(np.int32).copy(),
+               host[4 * T:8 * T].view(np.int32).copy(),
+               host[8 * T:12 * T].view(np.float32).copy(),
+               host[12 * T:16 * T].view(np.float32).copy(),
+               host[16 * T:20 * T].view(np.float32).copy()]
+        if self.tree_masks is not None:
+            out.append(host[moff:moff + 8 * W].view(np.int64).copy())
+        if self.tree_na is not None:
+            out.append(host[moff + 8 * W:moff + 8 * W + T].copy())
         return out
 
     def extract_host_tree(self) -> HostTree:

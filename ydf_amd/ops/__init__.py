@@ -88,6 +88,21 @@ def weighted_target(labels: torch.Tensor, weights, gh: torch.Tensor):
     return gh
 
 
+def pack_extract(feat: torch.Tensor, binv: torch.Tensor,
+                 leaf: torch.Tensor, node_stats: torch.Tensor,
+                 gain: torch.Tensor, tmasks, na, out: torch.Tensor,
+                 T: int, n_mask_words: int):
+    """Packs the per-tree extraction arrays into `out` (u8 staging
+    buffer) in ONE kernel launch; layout documented at
+    pack_extract_kernel (train_kernels.hip). GPU only."""
+    _C.gpu_pack_extract(
+        feat.data_ptr(), binv.data_ptr(), leaf.data_ptr(),
+        node_stats.data_ptr(), gain.data_ptr(),
+        tmasks.data_ptr() if tmasks is not None else 0,
+        na.data_ptr() if na is not None else 0,
+        out.data_ptr(), T, n_mask_words, _stream())
+
+
 def hist_build(bins: torch.Tensor, gh: torch.Tensor, node_ids: torch.Tensor,
                slot_map: torch.Tensor, hist: torch.Tensor, level_base: int,
                level_size: int, slot0: int, n_slots: int,

@@ -58,6 +58,9 @@ void gpu_leaf_values(const float*, const float*, float*, int, float,
 void gpu_update_preds(float*, const int32_t*, const float*, int64_t, float,
                       void*);
 void gpu_binary_logloss(const float*, const float*, float*, int64_t, void*);
+void gpu_pack_extract(const int32_t*, const int32_t*, const float*,
+                      const float*, const float*, const unsigned long long*,
+                      const uint8_t*, uint8_t*, int, int64_t, void*);
 // infer_kernels.hip
 void gpu_predict_forest(const float*, int64_t, int, const int32_t*,
                         const int32_t*, const unsigned long long*,
@@ -309,6 +312,19 @@ PYBIND11_MODULE(_ydf_ops, m) {
                           P<float>(leaf_values), total_nodes, lambda_l2,
                           lambda_l1,
                           (void*)stream);
+        },
+        nogil);
+  m.def("gpu_pack_extract",
+        [](uintptr_t feat, uintptr_t binv, uintptr_t leaf,
+           uintptr_t node_stats, uintptr_t gain, uintptr_t tmasks,
+           uintptr_t na, uintptr_t out, int T, int64_t n_mask_words,
+           uintptr_t stream) {
+          gpu_pack_extract(P<int32_t>(feat), P<int32_t>(binv),
+                           P<float>(leaf), P<float>(node_stats),
+                           P<float>(gain),
+                           P<unsigned long long>(tmasks), P<uint8_t>(na),
+                           P<uint8_t>(out), T, n_mask_words,
+                           (void*)stream);
         },
         nogil);
   m.def("gpu_update_preds",
