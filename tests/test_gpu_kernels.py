@@ -651,3 +651,26 @@ def test_binned4_engine_matches_flat(binary_data):
         init=float(m.init_predictions[0]))
     got = torch.sigmoid(out)
     assert (got - want).abs().max().item() < 1e-5
+
+
+@pytest.mark.gpu
+def test_deep_rf_extraction_integrity():
+    """Regression: pack_extract_kernel must cover node buffers larger
+    than 64k slots (RF depth-16 trees) — a capped grid without a
+    stride loop left the tail of the staging buffer stale, producing
+    garbage feature indices (caught by tools/bench_rf.py)."""
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(0)
+    n = 300000
+    d = {f"x{i}": rng.randn(n).astype(np.float32) for i in range(10)}
+    d["label"] = np.where(
+        d["x0"] + d["x1"] * d["x2"] + 0.5 * rng.randn(n) > 0, "a", "b")
+    m = ydf.RandomForestLearner(
+        label="label", num_trees=3, max_depth=16,
+        compute_oob_performances=False, device="cuda:0").train(d)
+    f = m.forest
+    valid = f.feat[f.feat >= 0]
+    assert valid.size > 1000
+    assert valid.max() < 10, int(valid.max())
+    assert m.evaluate(d).accuracy > 0.85

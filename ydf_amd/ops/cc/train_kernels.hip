@@ -439,7 +439,6 @@ __global__ void pack_extract_kernel(const int32_t* __restrict__ feat,
                                     const uint8_t* __restrict__ na,
                                     uint8_t* __restrict__ out, int T,
                                     int64_t n_mask_words) {
-  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int32_t* o_feat = reinterpret_cast<int32_t*>(out);
   int32_t* o_bin = o_feat + T;
   float* o_leaf = reinterpret_cast<float*>(o_bin + T);
@@ -450,15 +449,20 @@ __global__ void pack_extract_kernel(const int32_t* __restrict__ feat,
   unsigned long long* o_mask =
       reinterpret_cast<unsigned long long*>(out + moff);
   uint8_t* o_na = out + moff + 8 * n_mask_words;
-  if (i < T) {
-    o_feat[i] = feat[i];
-    o_bin[i] = binv[i];
-    o_leaf[i] = leaf[i];
-    o_cnt[i] = node_stats[i * 3 + 2];
-    o_gain[i] = gain[i];
-    if (na != nullptr) o_na[i] = na[i];
+  const int64_t n = T > n_mask_words ? T : n_mask_words;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < n; i += stride) {
+    if (i < T) {
+      o_feat[i] = feat[i];
+      o_bin[i] = binv[i];
+      o_leaf[i] = leaf[i];
+      o_cnt[i] = node_stats[i * 3 + 2];
+      o_gain[i] = gain[i];
+      if (na != nullptr) o_na[i] = na[i];
+    }
+    if (tmasks != nullptr && i < n_mask_words) o_mask[i] = tmasks[i];
   }
-  if (tmasks != nullptr && i < n_mask_words) o_mask[i] = tmasks[i];
 }
 
 // Precomputes the per-row slot-GROUP id (u8) for multi-group levels so each
