@@ -116,3 +116,26 @@ def test_distribute_run(tmp_path):
         capture_output=True, text=True, timeout=120,
         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
     assert r.returncode == 1
+
+
+def test_cli_edit_model(tmp_path, binary_data):
+    """cli/edit_model analogue: label rename + pure_serving strip."""
+    import ydf_amd as ydf
+
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=8, validation_ratio=0.0).train(
+        binary_data)
+    src = str(tmp_path / "m1")
+    dst = str(tmp_path / "m2")
+    m.save(src)
+    r = subprocess.run(
+        [sys.executable, "-m", "ydf_amd.cli.edit_model",
+         "--input", src, "--output", dst,
+         "--new_label_name", "income", "--pure_serving", "true"],
+        capture_output=True, text=True, cwd=REPO)
+    assert r.returncode == 0, r.stderr
+    m2 = ydf.load_model(dst)
+    assert m2.label() == "income"
+    assert not m2.training_logs
+    np.testing.assert_allclose(m.predict(binary_data),
+                               m2.predict(binary_data), rtol=1e-6)
