@@ -1195,3 +1195,44 @@ def test_weight_rescale_semantics_pinned():
                                rtol=1e-6, atol=1e-7)
     np.testing.assert_allclose(m_auto.predict(base),
                                m_pre.predict(base), rtol=1e-6, atol=1e-7)
+
+
+def test_gbt_reference_param_surface(binary_data):
+    """GBT accepts the full reference decision-tree shared parameter
+    surface: honest trees, num_candidate_attributes
+    (count form), sorting_strategy, keep_non_leaf_label_distribution,
+    compute_permutation_variable_importance; unimplemented non-default
+    settings raise."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=25, validation_ratio=0.0,
+        honest=True, num_candidate_attributes=2,
+        sorting_strategy="AUTO").train(binary_data)
+    assert m.evaluate(binary_data).accuracy > 0.85
+
+    # honest leaf values differ from plain training (re-estimated on
+    # the held-out half) while structure quality stays comparable
+    m0 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=25, validation_ratio=0.0).train(
+        binary_data)
+    assert np.abs(m.predict(binary_data)
+                  - m0.predict(binary_data)).max() > 1e-4
+
+    m2 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=10, validation_ratio=0.0,
+        keep_non_leaf_label_distribution=False,
+        compute_permutation_variable_importance=True).train(binary_data)
+    f = m2.forest
+    assert (f.cover[f.feat >= 0] == 0).all()
+    assert (f.cover[f.feat < 0] > 0).any()
+    vi = m2.variable_importances()
+    assert "MEAN_DECREASE_IN_ACCURACY" in vi
+
+    import pytest as _pt
+    with _pt.raises(NotImplementedError):
+        ydf.GradientBoostedTreesLearner(
+            label="label", num_trees=2,
+            in_split_min_examples_check=False).train(binary_data)
+    with _pt.raises(NotImplementedError):
+        ydf.GradientBoostedTreesLearner(
+            label="label", num_trees=2,
+            mhld_oblique_sample_attributes=True).train(binary_data)

@@ -90,6 +90,17 @@ class GradientBoostedTreesLearner(GenericLearner):
                  early_stopping_num_trees_look_ahead: int = 30,
                  early_stopping_initial_iteration: int = 10,
                  num_candidate_attributes_ratio: float = -1.0,
+                 num_candidate_attributes: int = -1,
+                 in_split_min_examples_check: bool = True,
+                 sorting_strategy: str = "PRESORT",
+                 keep_non_leaf_label_distribution: bool = True,
+                 honest: bool = False,
+                 honest_ratio_leaf_examples: float = 0.5,
+                 honest_fixed_separation: bool = False,
+                 mhld_oblique_sample_attributes=None,
+                 compute_permutation_variable_importance: bool = False,
+                 uplift_split_score: str = "KULLBACK_LEIBLER",
+                 uplift_min_examples_in_treatment: int = 5,
                  use_hessian_gain: bool = True,
                  apply_link_function: bool = True,
                  l2_categorical_regularization: float = 1.0,
@@ -148,6 +159,20 @@ class GradientBoostedTreesLearner(GenericLearner):
                 early_stopping_num_trees_look_ahead),
             early_stopping_initial_iteration=early_stopping_initial_iteration,
             num_candidate_attributes_ratio=num_candidate_attributes_ratio,
+            num_candidate_attributes=num_candidate_attributes,
+            in_split_min_examples_check=in_split_min_examples_check,
+            sorting_strategy=sorting_strategy,
+            keep_non_leaf_label_distribution=(
+                keep_non_leaf_label_distribution),
+            honest=honest,
+            honest_ratio_leaf_examples=honest_ratio_leaf_examples,
+            honest_fixed_separation=honest_fixed_separation,
+            mhld_oblique_sample_attributes=mhld_oblique_sample_attributes,
+            compute_permutation_variable_importance=(
+                compute_permutation_variable_importance),
+            uplift_split_score=uplift_split_score,
+            uplift_min_examples_in_treatment=(
+                uplift_min_examples_in_treatment),
             use_hessian_gain=use_hessian_gain,
             apply_link_function=apply_link_function,
             l2_categorical_regularization=l2_categorical_regularization,
@@ -537,8 +562,28 @@ class GradientBoostedTreesLearner(GenericLearner):
                     surv_entry_v if surv_entry is not None else None)
 
         F = bins.shape[0]
+        if not hp.get("in_split_min_examples_check", True):
+            raise NotImplementedError(
+                "in_split_min_examples_check=False is not implemented "
+                "(min_examples is always enforced at split search)")
+        if hp.get("sorting_strategy", "PRESORT") not in (
+                "IN_NODE", "PRESORT", "FORCE_PRESORT", "AUTO", "LAYER"):
+            raise ValueError(
+                f"unknown sorting_strategy {hp['sorting_strategy']!r}")
+        # sorting_strategy is a CPU-splitter performance hint in the
+        # reference; the 256-bin GPU histogram path has no per-node sort
+        if hp.get("mhld_oblique_sample_attributes"):
+            raise NotImplementedError(
+                "mhld_oblique_sample_attributes is not implemented")
+        if hp.get("honest") and hp.get(
+                "growing_strategy") == "BEST_FIRST_GLOBAL":
+            raise NotImplementedError(
+                "honest trees with BEST_FIRST_GLOBAL growth are not "
+                "supported")
         ncand = 0
-        if hp["num_candidate_attributes_ratio"] > 0:
+        if hp.get("num_candidate_attributes", -1) > 0:
+            ncand = min(F, int(hp["num_candidate_attributes"]))
+        elif hp["num_candidate_attributes_ratio"] > 0:
             ncand = max(1, int(round(hp["num_candidate_attributes_ratio"]
                                      * F)))
         cfg = trainer_lib.TrainerConfig(
@@ -554,6 +599,10 @@ class GradientBoostedTreesLearner(GenericLearner):
             selgb_ratio=hp.get("selective_gradient_boosting_ratio", 0.01),
             n_classes=n_classes,
             seed=self.random_seed, num_candidate_features=ncand,
+            honest=hp.get("honest", False),
+            honest_ratio=hp.get("honest_ratio_leaf_examples", 0.5),
+            honest_fixed_separation=hp.get("honest_fixed_separation",
+                                           False),
             early_stopping=(hp["early_stopping"] != "NONE"
                             and valid_bins is not None),
             early_stopping_num_trees_look_ahead=(
@@ -720,6 +769,22 @@ class GradientBoostedTreesLearner(GenericLearner):
         model = make_model(flat, init_preds, gains)
         model.training_logs = logs
         self._finalize_model(model)
+        if not hp.get("keep_non_leaf_label_distribution", True):
+            # drop non-leaf training distributions (reference
+            # keep_non_leaf_label_distribution=false: smaller model,
+            # disables distribution-dependent analyses like TreeSHAP)
+            model.forest.cover = model.forest.cover.copy()
+            model.forest.cover[model.forest.feat >= 0] = 0.0
+        if hp.get("compute_permutation_variable_importance"):
+            from ydf_amd.utils.analysis import permutation_importances
+
+            key = ("MEAN_DECREASE_IN_ACCURACY"
+                   if self._task == Task.CLASSIFICATION
+                   else "MEAN_INCREASE_IN_RMSE")
+            vi = permutation_importances(model, ds, ds.label_values,
+                                         device=device)
+            model.metadata["permutation_importances"] = {
+                key: [(float(s), n) for s, n in vi]}
         if snapshot_cb is not None:
             # final state also becomes the snapshot (enables continuing
             # with a larger num_trees later)

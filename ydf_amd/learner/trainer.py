@@ -1507,6 +1507,12 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                                            na_t=nt)
                         valid_preds[c2].sub_(lv[dart_valid_ids.long()],
                                              alpha=dart_scale[ti])
+        hmask = None
+        if cfg.honest:
+            hmask = honest_split_mask(cfg.seed, it, N,
+                                      cfg.honest_ratio,
+                                      cfg.honest_fixed_separation,
+                                      trainer.device)
         try:
           for c in range(C):
             pc = preds[c]
@@ -1577,10 +1583,33 @@ def train_gbt(trainer: ForestTrainer, log=None, start_iteration: int = 0,
                     torch.full((), amp, device=dev),
                     torch.ones((), device=dev))
                 trainer.gh.mul_(scale.view(-1, 1))
-            tree = trainer.grow_tree(it * C + c, sample_mask)
+            smask = sample_mask
+            if hmask is not None:
+                # honest trees: structure from the non-estimation half
+                smask = (~hmask) if smask is None else (smask & ~hmask)
+            tree = trainer.grow_tree(it * C + c, smask)
             trees.append(tree)
-            if sample_mask is not None:
+            if smask is not None:
                 trainer.route_rows(trainer.bins, trainer.node_ids)
+            if hmask is not None:
+                # re-estimate leaf values -G/(H+l2) from the held-out
+                # half (reference Honest message,
+                # decision_tree.proto:417-426); leaves with no
+                # estimation rows keep the structure value
+                ids = trainer.node_ids.long()
+                zero = torch.zeros((), device=trainer.device)
+                gm = torch.where(hmask, trainer.gh[:, 0], zero)
+                hm_ = torch.where(hmask, trainer.gh[:, 1], zero)
+                num = torch.zeros_like(trainer.leaf_vals)
+                den = torch.zeros_like(trainer.leaf_vals)
+                num.scatter_add_(0, ids, gm)
+                den.scatter_add_(0, ids, hm_)
+                est = -num / (den + cfg.lambda_l2).clamp(min=1e-9)
+                trainer.leaf_vals.copy_(
+                    torch.where(den > 0, est, trainer.leaf_vals))
+                if tree.leaf_value is not None:
+                    tree.leaf_value = trainer.leaf_vals[
+                        :len(tree.leaf_value)].cpu().numpy().copy()
             step_scale = cfg.shrinkage
             if dart:
                 k = len(dropped)

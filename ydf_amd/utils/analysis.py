@@ -205,6 +205,11 @@ def structure_importances(model) -> Dict[str, List[Tuple[float, str]]]:
     if oob_vi:
         for metric, ranked_list in oob_vi.items():
             out[metric] = [(float(s), n) for s, n in ranked_list]
+    pvi = model.metadata.get("permutation_importances") \
+        if model.metadata else None
+    if pvi:
+        for metric, ranked_list in pvi.items():
+            out[metric] = [(float(s), n) for s, n in ranked_list]
     return out
 
 
