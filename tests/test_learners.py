@@ -1236,3 +1236,41 @@ def test_gbt_reference_param_surface(binary_data):
         ydf.GradientBoostedTreesLearner(
             label="label", num_trees=2,
             mhld_oblique_sample_attributes=True).train(binary_data)
+
+
+def test_shared_tree_params_accepted_everywhere(binary_data):
+    """Every tree learner accepts the reference's shared decision-tree
+    parameter surface (PYDF generates all learner signatures from one
+    spec): class_weights is honored, num_discretized_numerical_bins
+    feeds binning, unimplemented settings raise, unknown kwargs are
+    rejected."""
+    # class_weights shifts the decision boundary toward the upweighted
+    # class -> more "yes" predictions
+    m0 = ydf.RandomForestLearner(
+        label="label", num_trees=15,
+        compute_oob_performances=False).train(binary_data)
+    m1 = ydf.RandomForestLearner(
+        label="label", num_trees=15, compute_oob_performances=False,
+        class_weights={"yes": 8.0}).train(binary_data)
+    r0 = (m0.predict(binary_data) > 0.5).mean()
+    r1 = (m1.predict(binary_data) > 0.5).mean()
+    assert r1 > r0 + 0.02, (r0, r1)
+
+    # coarser discretization trains and degrades gracefully
+    m2 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=20, validation_ratio=0.0,
+        num_discretized_numerical_bins=16).train(binary_data)
+    assert m2.evaluate(binary_data).accuracy > 0.85
+
+    # accepted names on every learner (no TypeError at construction)
+    for cls in (ydf.CartLearner, ydf.IsolationForestLearner):
+        cls(label="label", sorting_strategy="AUTO",
+            sparse_oblique_weights_integer_minimum=-4,
+            numerical_vector_sequence_num_random_anchors=10)
+
+    import pytest as _pt
+    with _pt.raises(TypeError):
+        ydf.RandomForestLearner(label="label", not_a_param=1)
+    with _pt.raises(NotImplementedError):
+        ydf.CartLearner(label="label",
+                        include_all_columns=True).train(binary_data)

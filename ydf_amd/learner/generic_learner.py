@@ -14,6 +14,65 @@ from ydf_amd.model.forest import padded_boundaries
 from ydf_amd import ops
 
 
+# Shared decision-tree hyperparameters accepted by EVERY tree learner
+# (reference GetGenericHyperParameterSpecification surface; PYDF
+# generates each learner signature from it). A name listed here is
+# accepted as a keyword by any learner; settings this framework cannot
+# honor raise NotImplementedError at train time (see
+# _validate_extra_hp). None = "not set" sentinel.
+SHARED_TREE_PARAMS = {
+    "explicit_args": None,
+    "class_weights": None,              # implemented (example weights)
+    "label_classes": None,
+    "include_all_columns": False,
+    "max_num_scanned_rows_to_infer_semantic": None,       # perf cap
+    "max_num_scanned_rows_to_compute_statistics": None,   # perf cap
+    "num_discretized_numerical_bins": None,  # implemented (max_bins)
+    "discretize_numerical_columns": None,    # 256-bin path is core design
+    "categorical_set_split_greedy_sampling": None,
+    "categorical_set_split_max_num_items": None,
+    "categorical_set_split_min_item_frequency": None,
+    "categorical_set_split_greedy_maximum_mask_size": None,
+    "categorical_random_max_num_trials": None,
+    "categorical_random_num_trial_exponent": None,
+    "sparse_oblique_weights_integer_minimum": None,
+    "sparse_oblique_weights_integer_maximum": None,
+    "sparse_oblique_weights_power_of_two_min_exponent": None,
+    "sparse_oblique_weights_power_of_two_max_exponent": None,
+    "numerical_vector_sequence_num_examples": None,
+    "numerical_vector_sequence_num_random_anchors": None,
+    "numerical_vector_sequence_enable_closer_than_conditions": None,
+    "numerical_vector_sequence_enable_projected_more_than_conditions":
+        None,
+    "sorting_strategy": None,
+    "in_split_min_examples_check": None,
+    "keep_non_leaf_label_distribution": None,
+    "growing_strategy": None,
+    "max_num_nodes": None,
+    "honest": None,
+    "honest_ratio_leaf_examples": None,
+    "honest_fixed_separation": None,
+    "num_candidate_attributes": None,
+    "num_candidate_attributes_ratio": None,
+    "mhld_oblique_max_num_attributes": None,
+    "mhld_oblique_sample_attributes": None,
+    "split_axis": None,
+    "sparse_oblique_max_num_features": None,
+    "sparse_oblique_max_num_projections": None,
+    "sparse_oblique_normalization": None,
+    "sparse_oblique_num_projections_exponent": None,
+    "sparse_oblique_projection_density_factor": None,
+    "sparse_oblique_weights": None,
+    "min_examples": None,
+    "working_dir": None,
+    "ranking_group": None,
+    "uplift_treatment": None,
+    "uplift_split_score": None,
+    "uplift_min_examples_in_treatment": None,
+    "maximum_training_duration_seconds": None,
+}
+
+
 class GenericLearner:
     """Base learner: dataset ingestion, binning, device selection."""
 
@@ -28,7 +87,15 @@ class GenericLearner:
                  maximum_model_size_in_memory_in_bytes: float = -1.0,
                  random_seed: int = 123456, device=None,
                  feature_selector=None,
-                 num_threads: Optional[int] = None):
+                 num_threads: Optional[int] = None, **extra_hp):
+        self._extra_hp = {}
+        for k, v in extra_hp.items():
+            if k not in SHARED_TREE_PARAMS:
+                raise TypeError(
+                    f"{type(self).__name__} got an unexpected keyword "
+                    f"argument {k!r}")
+            if v is not None:
+                self._extra_hp[k] = v
         self.feature_selector = feature_selector
         self.allow_na_conditions = allow_na_conditions
         self.pure_serving_model = pure_serving_model
@@ -296,10 +363,46 @@ class GenericLearner:
             model.num_trees() if hasattr(model, "num_trees") else None,
             _time.monotonic() - t0)
 
+    def _validate_extra_hp(self) -> None:
+        eh = getattr(self, "_extra_hp", {})
+        if not eh:
+            return
+        # merge names the specialized trainers read from hp (explicit
+        # constructor params always win over the shared-table form)
+        for k, v in eh.items():
+            if hasattr(self, "hyperparameters"):
+                self.hyperparameters.setdefault(k, v)
+        if eh.get("label_classes"):
+            raise NotImplementedError(
+                "label_classes (explicit class order) is not "
+                "implemented; classes are frequency-ordered as in the "
+                "default reference dataspec inference")
+        if eh.get("include_all_columns"):
+            raise NotImplementedError(
+                "include_all_columns=True is not implemented; list the "
+                "columns in `features` instead")
+        if eh.get("mhld_oblique_sample_attributes"):
+            raise NotImplementedError(
+                "mhld_oblique_sample_attributes is not implemented")
+        if eh.get("in_split_min_examples_check") is False:
+            raise NotImplementedError(
+                "in_split_min_examples_check=False is not implemented")
+        if eh.get("categorical_random_max_num_trials") or \
+                eh.get("categorical_random_num_trial_exponent"):
+            raise NotImplementedError(
+                "categorical_algorithm=RANDOM trials are not "
+                "implemented (CART set-splits are used)")
+        ss = eh.get("sorting_strategy")
+        if ss is not None and ss not in ("IN_NODE", "PRESORT",
+                                         "FORCE_PRESORT", "AUTO",
+                                         "LAYER"):
+            raise ValueError(f"unknown sorting_strategy {ss!r}")
+
     def _prepare(self, data, device: torch.device):
         """Dataset -> (VerticalDataset, binned u8 [F,N] on device,
         labels f32 [N] on device, padded boundary matrix np [F,n_cuts],
         cat_flags u8 tensor or None)."""
+        self._validate_extra_hp()
         weights_np = None
         if isinstance(data, VerticalDataset):
             ds = data
@@ -314,13 +417,18 @@ class GenericLearner:
             local_na = getattr(self, "missing_value_policy",
                                "GLOBAL_IMPUTATION") in (
                 "LOCAL_IMPUTATION", "RANDOM_LOCAL_IMPUTATION")
+            nb = getattr(self, "_extra_hp", {}).get(
+                "num_discretized_numerical_bins")
+            max_bins = 255 if local_na else 256
+            if nb:
+                max_bins = max(2, min(int(nb), max_bins))
             ds = create_vertical_dataset(
                 cols, label=self.label, task=self._task,
                 features=features, max_vocab_count=self.max_vocab_count,
                 min_vocab_frequency=self.min_vocab_frequency,
                 allow_na_conditions=self.allow_na_conditions,
                 keep_na=local_na,
-                max_bins=255 if local_na else 256)
+                max_bins=max_bins)
             if self.weights_col is not None:
                 if self.weights_col not in cols:
                     raise ValueError(
@@ -350,6 +458,26 @@ class GenericLearner:
                 for i, c in enumerate(ds.dataspec.feature_columns):
                     arr[i] = dirs.get(c.name, 0)
                 mono = torch.from_numpy(arr).to(device)
+        cw = getattr(self, "_extra_hp", {}).get("class_weights")
+        if cw:
+            # class weights as per-example weights (reference
+            # class_weights: Dict[label value -> weight])
+            if self._task != Task.CLASSIFICATION or \
+                    ds.label_values is None:
+                raise ValueError(
+                    "class_weights requires a classification task")
+            vocab = list(ds.dataspec.label_column.vocab or [])
+            wmap = np.ones(max(len(vocab), 1), dtype=np.float32)
+            for name, w in cw.items():
+                if str(name) not in vocab:
+                    raise ValueError(
+                        f"class_weights key {name!r} is not a label "
+                        f"class (classes: {vocab})")
+                wmap[vocab.index(str(name))] = float(w)
+            cls_w = wmap[np.clip(
+                ds.label_values.astype(np.int64), 0, len(wmap) - 1)]
+            weights_np = cls_w if weights_np is None \
+                else weights_np * cls_w
         weights = None
         if weights_np is not None:
             # packed-u64 histogram path needs per-example h <= 16: scale
