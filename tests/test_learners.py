@@ -1274,3 +1274,27 @@ def test_shared_tree_params_accepted_everywhere(binary_data):
     with _pt.raises(NotImplementedError):
         ydf.CartLearner(label="label",
                         include_all_columns=True).train(binary_data)
+
+
+def test_data_spec_override_and_dgbt(binary_data):
+    """Learner(data_spec=...) trains against a pre-built dataspec
+    (reference generic data_spec arg), and the distributed GBT class
+    exposes the reference worker-pool surface with RCCL guidance."""
+    m0 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=10, validation_ratio=0.0).train(
+        binary_data)
+    m1 = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=10, validation_ratio=0.0,
+        data_spec=m0.data_spec()).train(binary_data)
+    np.testing.assert_array_equal(m0.forest.feat, m1.forest.feat)
+    np.testing.assert_allclose(m0.predict(binary_data),
+                               m1.predict(binary_data), rtol=1e-6)
+
+    m2 = ydf.DistributedGradientBoostedTreesLearner(
+        label="label", num_trees=10, validation_ratio=0.0,
+        worker_logs=False, force_numerical_discretization=True).train(
+        binary_data)
+    assert m2.evaluate(binary_data).accuracy > 0.9
+    with pytest.raises(NotImplementedError, match="torch.distributed"):
+        ydf.DistributedGradientBoostedTreesLearner(
+            label="label", workers=["host:2001"])

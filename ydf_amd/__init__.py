@@ -114,12 +114,31 @@ from ydf_amd.utils.log import strict, verbose
 
 
 # --- remaining PYDF surface names ------------------------------------------
-# DistributedGradientBoostedTreesLearner: the reference's gRPC
-# manager/worker learner. Here distribution is one-process-per-GPU over
-# RCCL (torchrun), which the plain GBT learner already does when
-# launched under torch.distributed.run — this alias makes the intent
-# explicit and the name importable.
-DistributedGradientBoostedTreesLearner = GradientBoostedTreesLearner
+class DistributedGradientBoostedTreesLearner(GradientBoostedTreesLearner):
+    """Distributed GBT (reference learner/distributed_gradient_boosted_
+    trees). MI355X mapping: distribution is one process per GPU over
+    RCCL/xGMI — launch any GBT training under `torch.distributed.run`
+    (one rank per GPU; rows are sharded, per-level histograms
+    all-reduce). The reference's gRPC worker-pool parameters are
+    accepted for API parity; `workers` (remote machine addresses) has
+    no RCCL analogue here and raises with guidance."""
+
+    def __init__(self, *args, workers=None, worker_logs=True,
+                 force_numerical_discretization=False,
+                 max_unique_values_for_discretized_numerical=16000,
+                 **kwargs):
+        if workers:
+            raise NotImplementedError(
+                "gRPC worker pools are not used on MI355X: launch this "
+                "training under `python -m torch.distributed.run "
+                "--nproc-per-node <gpus>` instead (one RCCL rank per "
+                "GPU); see docs/DESIGN.md")
+        # numerical features are ALWAYS 256-bin discretized on this
+        # framework (the GPU histogram path), so
+        # force_numerical_discretization is effectively always true
+        # and the unique-value cap is the bin count
+        self._worker_logs = worker_logs
+        super().__init__(*args, **kwargs)
 
 from ydf_amd.learner.extras import FeatureSelectorLogs  # noqa: E402
 

@@ -87,7 +87,8 @@ class GenericLearner:
                  maximum_model_size_in_memory_in_bytes: float = -1.0,
                  random_seed: int = 123456, device=None,
                  feature_selector=None,
-                 num_threads: Optional[int] = None, **extra_hp):
+                 num_threads: Optional[int] = None,
+                 data_spec=None, extra_training_config=None, **extra_hp):
         self._extra_hp = {}
         for k, v in extra_hp.items():
             if k not in SHARED_TREE_PARAMS:
@@ -97,6 +98,14 @@ class GenericLearner:
             if v is not None:
                 self._extra_hp[k] = v
         self.feature_selector = feature_selector
+        # pre-built dataspec (reference data_spec constructor arg):
+        # skips dataspec inference; data is encoded with THIS spec
+        self.data_spec_override = data_spec
+        if extra_training_config is not None:
+            raise NotImplementedError(
+                "extra_training_config (raw TrainingConfig proto "
+                "extensions) is not supported; use the keyword "
+                "hyperparameters instead")
         self.allow_na_conditions = allow_na_conditions
         self.pure_serving_model = pure_serving_model
         if missing_value_policy not in ("GLOBAL_IMPUTATION",
@@ -422,13 +431,18 @@ class GenericLearner:
             max_bins = 255 if local_na else 256
             if nb:
                 max_bins = max(2, min(int(nb), max_bins))
-            ds = create_vertical_dataset(
-                cols, label=self.label, task=self._task,
-                features=features, max_vocab_count=self.max_vocab_count,
-                min_vocab_frequency=self.min_vocab_frequency,
-                allow_na_conditions=self.allow_na_conditions,
-                keep_na=local_na,
-                max_bins=max_bins)
+            if getattr(self, "data_spec_override", None) is not None:
+                ds = create_vertical_dataset(
+                    cols, dataspec=self.data_spec_override)
+            else:
+                ds = create_vertical_dataset(
+                    cols, label=self.label, task=self._task,
+                    features=features,
+                    max_vocab_count=self.max_vocab_count,
+                    min_vocab_frequency=self.min_vocab_frequency,
+                    allow_na_conditions=self.allow_na_conditions,
+                    keep_na=local_na,
+                    max_bins=max_bins)
             if self.weights_col is not None:
                 if self.weights_col not in cols:
                     raise ValueError(
