@@ -514,3 +514,52 @@ def test_tuner_optimize_metric(binary_data):
     best = m.tuner_logs.best_trial
     # accuracy objective: scores are accuracies in [0, 1]
     assert 0.5 < best.score <= 1.0
+
+
+def test_evaluate_keyword_surface(binary_data, regression_data):
+    """Reference evaluate() keyword surface: weighted=False, label/task
+    override, bootstrapping CIs, use_slow_engine, truncations."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=20, validation_ratio=0.0).train(
+        binary_data)
+    ev = m.evaluate(binary_data, weighted=False, use_slow_engine=True,
+                    num_threads=2)
+    assert ev.accuracy > 0.9
+    evb = m.evaluate(binary_data, bootstrapping=50)
+    lo, hi = evb.bootstrap_cis["accuracy"]
+    assert lo <= evb.accuracy <= hi
+
+    # label override: same values under a different column name
+    data2 = dict(binary_data)
+    data2["y2"] = binary_data["label"]
+    del data2["label"]
+    ev2 = m.evaluate(data2, label="y2")
+    assert abs(ev2.accuracy - ev.accuracy) < 1e-9
+
+    # task override: regression metrics on a regression model under
+    # an explicitly passed task
+    mr = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=20,
+        validation_ratio=0.0).train(regression_data)
+    evr = mr.evaluate(regression_data, task=ydf.Task.REGRESSION)
+    assert evr.rmse is not None
+
+
+def test_ranking_truncations():
+    rng = np.random.RandomState(5)
+    n = 2000
+    g = np.repeat(np.arange(n // 10), 10)
+    x = rng.randn(n).astype(np.float32)
+    rel = np.clip((x + rng.randn(n) * 0.4) * 2, 0, 4).astype(np.float32)
+    d = {"x": x, "rel": rel, "g": g}
+    m = ydf.GradientBoostedTreesLearner(
+        label="rel", task=ydf.Task.RANKING, ranking_group="g",
+        num_trees=20, validation_ratio=0.0).train(d)
+    e5 = m.evaluate(d)
+    e10 = m.evaluate(d, ndcg_truncation=10, mrr_truncation=10,
+                     map_truncation=10)
+    assert e5.ndcg != e10.ndcg  # truncation changes the metric
+    assert 0 < e10.mrr <= 1 and 0 < e10.map <= 1
+    # group override by explicit column name
+    e_g = m.evaluate(d, group="g")
+    assert abs(e_g.ndcg - e5.ndcg) < 1e-12

@@ -136,16 +136,17 @@ def mrr(labels: np.ndarray, scores: np.ndarray, groups: np.ndarray,
 
 
 def mean_average_precision(labels: np.ndarray, scores: np.ndarray,
-                           groups: np.ndarray) -> float:
-    """Ranking MAP: mean over groups of average precision with binary
-    relevance label > 0 (reference ranking metric "map")."""
+                           groups: np.ndarray,
+                           truncation: int = 5) -> float:
+    """Ranking MAP@truncation: mean over groups of average precision
+    with binary relevance label > 0 (reference ranking metric "map")."""
     total, n_groups = 0.0, 0
     for g in np.unique(groups):
         m = groups == g
         rel = labels[m] > 0
         if not rel.any():
             continue
-        order = np.argsort(-scores[m], kind="mergesort")
+        order = np.argsort(-scores[m], kind="mergesort")[:truncation]
         r = rel[order]
         hits = np.cumsum(r)
         prec_at_hit = hits[r] / (np.nonzero(r)[0] + 1)

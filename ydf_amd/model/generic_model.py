@@ -30,6 +30,9 @@ class ModelIOOptions:
     file_prefix: Optional[str] = None
 
 
+_WEIGHTS_DISABLED = object()  # evaluate(weighted=False) sentinel
+
+
 def default_device() -> torch.device:
     return torch.device("cuda") if torch.cuda.is_available() else \
         torch.device("cpu")
@@ -518,10 +521,17 @@ class GenericModel:
                     n_trees=T // C if C > 1 else T)
         return out
 
-    def predict(self, data, device=None) -> np.ndarray:
+    def predict(self, data, device=None, *, use_slow_engine=False,
+                num_threads=None) -> np.ndarray:
         """Predictions as numpy: binary classification -> P(class_2) [N];
         multi-class -> [N, C]; regression/anomaly -> [N]. (Mirrors
-        ydf GenericModel.predict semantics.)"""
+        ydf GenericModel.predict semantics; use_slow_engine forces the
+        generic CPU path instead of the packed GPU engines, and
+        num_threads bounds the CPU op parallelism.)"""
+        if num_threads is not None:
+            torch.set_num_threads(int(num_threads))
+        if use_slow_engine:
+            device = "cpu"
         dev = torch.device(device) if device is not None else default_device()
         from ydf_amd.utils import usage
 
@@ -639,29 +649,58 @@ class GenericModel:
         return m.T.contiguous()
 
     # ------------------------------------------------------------------
-    def evaluate(self, data, device=None, weights=None) -> Evaluation:
+    def evaluate(self, data, device=None, weights=None, *,
+                 weighted=None, task=None, label=None, group=None,
+                 bootstrapping=False, ndcg_truncation=5,
+                 mrr_truncation=5, map_truncation=5,
+                 use_slow_engine=False,
+                 num_threads=None) -> Evaluation:
         """`weights` names a column in `data` (or passes an array) for
         example-weighted metrics; defaults to the training weights
-        column if the model recorded one."""
+        column if the model recorded one. Reference keyword surface:
+        weighted=False disables the recorded weights; task/label/group
+        override the evaluation task, label column and ranking group;
+        bootstrapping=True|n attaches percentile-bootstrap CIs
+        (ev.bootstrap_cis); *_truncation set the ranking metric
+        cutoffs."""
+        if weighted is False:
+            weights = _WEIGHTS_DISABLED
+        task_override = task
         cols = _to_column_dict(data) if not isinstance(data, VerticalDataset) \
             else None
-        preds = self.predict(data, device=device)
+        preds = self.predict(data, device=device,
+                             use_slow_engine=use_slow_engine,
+                             num_threads=num_threads)
         if isinstance(data, VerticalDataset):
             labels = data.label_values
         else:
-            lname = self.dataspec.label
+            lname = label if label is not None else self.dataspec.label
             if lname is None or lname not in cols:
                 raise ValueError("dataset has no label column")
-            lspec = self.dataspec.label_column
-            if lspec.semantic == Semantic.CATEGORICAL:
+            use_spec = label is None or label == self.dataspec.label
+            if use_spec and self.dataspec.label_column.semantic \
+                    == Semantic.CATEGORICAL:
+                lspec = self.dataspec.label_column
                 lookup = {item: i for i, item in enumerate(lspec.vocab)}
                 labels = np.fromiter(
                     (lookup.get(s, 0) - 1 for s in cols[lname].astype(str)),
                     dtype=np.float32, count=len(cols[lname]))
             else:
-                labels = np.asarray(cols[lname], dtype=np.float32)
+                arr = np.asarray(cols[lname])
+                if arr.dtype.kind in "USO":
+                    # overridden label with string classes: map through
+                    # the model's class order
+                    lk = {c: i for i, c in enumerate(
+                        self.label_classes or [])}
+                    labels = np.fromiter(
+                        (lk.get(s, 0) for s in arr.astype(str)),
+                        dtype=np.float32, count=len(arr))
+                else:
+                    labels = arr.astype(np.float32)
         w = None
-        if weights is None:
+        if weights is _WEIGHTS_DISABLED:
+            weights = None
+        elif weights is None:
             weights = (self.metadata or {}).get("weights_column")
         if weights is not None and cols is not None:
             w = np.asarray(cols[weights], np.float64) \
@@ -669,7 +708,8 @@ class GenericModel:
                 else (np.asarray(weights, np.float64)
                       if not isinstance(weights, str) else None)
         n_classes = len(self.label_classes) if self.label_classes else 2
-        if self._task == Task.SURVIVAL_ANALYSIS:
+        etask = task_override if task_override is not None else self._task
+        if etask == Task.SURVIVAL_ANALYSIS:
             from ydf_amd.learner.survival import CoxData
             from ydf_amd.metric.survival import concordance_index
 
@@ -688,7 +728,7 @@ class GenericModel:
             ev.loss = cd.loss(torch.from_numpy(
                 np.asarray(preds, np.float32)))
             return ev
-        if self._task in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):
+        if etask in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):
             from ydf_amd.metric.uplift import auuc_qini
 
             tcol = (self.metadata or {}).get("uplift_treatment")
@@ -706,25 +746,35 @@ class GenericModel:
             ev = Evaluation(num_examples=len(labels))
             ev.auuc, ev.qini = auuc_qini(labels, treat, preds)
             return ev
-        ev = evaluate_predictions(preds, labels, self._task, n_classes,
+        ev = evaluate_predictions(preds, labels, etask, n_classes,
                                   weights=w)
-        if self._task == Task.CLASSIFICATION and self.label_classes:
+        if etask == Task.CLASSIFICATION and self.label_classes:
             ev.classes = tuple(self.label_classes)
-        if self._task == Task.RANKING:
-            gcol = (self.metadata or {}).get("ranking_group")
+        if etask == Task.RANKING:
+            gcol = group if group is not None else \
+                (self.metadata or {}).get("ranking_group")
             if gcol and cols is not None and gcol in cols:
                 from ydf_amd.metric.metric import mean_average_precision
                 from ydf_amd.metric.metric import mrr as mrr_fn
                 from ydf_amd.metric.metric import ndcg as ndcg_fn
 
                 g = np.asarray(cols[gcol])
-                ev.ndcg = ndcg_fn(
-                    labels, preds, g,
-                    truncation=(self.metadata or {}).get(
-                        "ndcg_truncation", 5))
-                ev.mrr = mrr_fn(labels, preds, g)
-                ev.map = mean_average_precision(labels, preds, g)
+                nt = ndcg_truncation if ndcg_truncation != 5 else \
+                    (self.metadata or {}).get("ndcg_truncation", 5)
+                ev.ndcg = ndcg_fn(labels, preds, g, truncation=nt)
+                ev.mrr = mrr_fn(labels, preds, g,
+                                truncation=mrr_truncation)
+                ev.map = mean_average_precision(
+                    labels, preds, g, truncation=map_truncation)
                 ev.loss = -ev.ndcg
+        if bootstrapping:
+            from ydf_amd.metric.metric import (
+                bootstrap_confidence_intervals)
+
+            n_boot = 2000 if bootstrapping is True \
+                else max(int(bootstrapping), 10)
+            ev.bootstrap_cis = bootstrap_confidence_intervals(
+                labels, preds, etask, n_samples=n_boot)
         return ev
 
     # ------------------------------------------------------------------
