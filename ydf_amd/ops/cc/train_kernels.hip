@@ -1478,9 +1478,19 @@ void gpu_hist_build(const uint8_t* bins, const float* gh,
     const char* e = getenv("YDFA_HIST_MAX_BLOCKS");
     max_blocks_env = e ? atoi(e) : 0;
   }
-  int chunks = row_chunks(
-      N, n_fgroups,
-      max_blocks_env > 0 ? max_blocks_env : 8192 * 256 / threads);
+  int chunks;
+  if (max_blocks_env > 0) {
+    // explicit override (sweeps): block-cap-driven chunking with the
+    // legacy 1k-row floor, bypassing the 64k-rows-per-chunk target
+    const int per_f = (max_blocks_env + n_fgroups - 1) / n_fgroups;
+    const int64_t mx = (N + 1023) >> 10;
+    chunks = (int)(mx < per_f ? mx : per_f);
+    const int min_c = (int)((N + (1 << 19) - 1) >> 19);
+    if (chunks < min_c) chunks = min_c;
+    if (chunks < 1) chunks = 1;
+  } else {
+    chunks = row_chunks(N, n_fgroups, 8192 * 256 / threads);
+  }
   {
     static int64_t max_rpb = -1;
     if (max_rpb < 0) {
