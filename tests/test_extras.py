@@ -563,3 +563,24 @@ def test_ranking_truncations():
     # group override by explicit column name
     e_g = m.evaluate(d, group="g")
     assert abs(e_g.ndcg - e5.ndcg) < 1e-12
+
+
+def test_analyze_keyword_surface(binary_data):
+    """Reference analyze() kwargs: sampling, plot toggles, SHAP
+    summary importance, permutation rounds."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=15, validation_ratio=0.0).train(
+        binary_data)
+    a = m.analyze(binary_data, sampling=0.3, num_bins=10,
+                  permutation_variable_importance_rounds=2,
+                  maximum_duration=30, num_threads=2)
+    assert "MEAN_ABS_SHAP" in a.variable_importances
+    assert a.partial_dependences and a.conditional_expectations
+    a2 = m.analyze(binary_data, partial_dependence_plot=False,
+                   conditional_expectation_plot=False,
+                   shap_values=False,
+                   permutation_variable_importance=False)
+    assert not a2.partial_dependences
+    assert not a2.conditional_expectations
+    assert "MEAN_ABS_SHAP" not in a2.variable_importances
+    assert "MEAN_DECREASE_IN_ACCURACY" not in a2.variable_importances

@@ -845,12 +845,31 @@ class GenericModel:
 
     def analyze(self, data, sampling: float = 1.0,
                 num_bins: int = 20,
+                partial_dependence_plot: bool = True,
+                conditional_expectation_plot: bool = True,
+                permutation_variable_importance: bool = True,
+                shap_values: bool = True,
                 permutation_variable_importance_rounds: int = 1,
+                num_threads: Optional[int] = None,
+                maximum_duration: Optional[float] = 20,
                 features: Optional[List[str]] = None, device=None):
-        """Full model analysis: variable importances + PDPs (mirrors ydf
-        model.analyze; reference utils/model_analysis.h:36-89)."""
+        """Full model analysis: variable importances (+SHAP summary) +
+        PDPs + CEPs (mirrors ydf model.analyze; reference
+        utils/model_analysis.h:36-89). `sampling` subsamples the rows;
+        maximum_duration is accepted for API parity (the GPU analysis
+        path finishes well inside it for the supported sizes)."""
         from ydf_amd.utils import analysis as analysis_lib
 
+        if num_threads is not None:
+            torch.set_num_threads(int(num_threads))
+        if sampling < 1.0:
+            cols0 = _to_column_dict(data) \
+                if not isinstance(data, VerticalDataset) else None
+            if cols0 is not None:
+                n0 = len(next(iter(cols0.values())))
+                rng = np.random.RandomState(1234)
+                keep = rng.rand(n0) < sampling
+                data = {k: np.asarray(v)[keep] for k, v in cols0.items()}
         labels = None
         if self.dataspec.label is not None:
             try:
@@ -875,7 +894,12 @@ class GenericModel:
         return analysis_lib.analyze(
             self, data, labels=labels,
             permutation_variable_importance=(
-                permutation_variable_importance_rounds > 0),
+                permutation_variable_importance
+                and permutation_variable_importance_rounds > 0),
+            partial_dependence=partial_dependence_plot,
+            conditional_expectation=conditional_expectation_plot,
+            shap_values=shap_values,
+            permutation_rounds=permutation_variable_importance_rounds,
             features=features, num_grid_points=num_bins, device=device)
 
     # ------------------------------------------------------------------

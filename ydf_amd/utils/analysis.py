@@ -382,20 +382,36 @@ def conditional_expectations(model, data, labels=None,
 def analyze(model, data, labels: Optional[np.ndarray] = None,
             permutation_variable_importance: bool = True,
             partial_dependence: bool = True,
+            conditional_expectation: bool = True,
+            shap_values: bool = True,
             features: Optional[List[str]] = None,
-            num_grid_points: int = 20, device=None) -> Analysis:
+            num_grid_points: int = 20,
+            permutation_rounds: int = 1, device=None) -> Analysis:
     vi = structure_importances(model)
     if permutation_variable_importance and labels is not None:
         key = ("MEAN_DECREASE_IN_ACCURACY"
                if model.task() == Task.CLASSIFICATION
                else "MEAN_INCREASE_IN_RMSE")
-        vi[key] = permutation_importances(model, data, labels, device=device)
+        vi[key] = permutation_importances(
+            model, data, labels,
+            num_repetitions=max(1, permutation_rounds), device=device)
+    if shap_values:
+        try:
+            phi = model.predict_shap(data)
+            shap_vi = sorted(
+                ((float(np.abs(v).mean()), k)
+                 for k, v in phi.items() if k != "__BIAS__"),
+                reverse=True)
+            vi["MEAN_ABS_SHAP"] = shap_vi
+        except (NotImplementedError, ValueError):
+            pass  # multi-output / coverless models: no SHAP summary
     pdps = []
     ceps = []
     if partial_dependence:
         pdps = partial_dependences(model, data, features=features,
                                    num_grid_points=num_grid_points,
                                    device=device)
+    if conditional_expectation:
         ceps = conditional_expectations(model, data, labels=labels,
                                         device=device)
     return Analysis(variable_importances=vi, partial_dependences=pdps,
