@@ -1298,3 +1298,17 @@ def test_data_spec_override_and_dgbt(binary_data):
     with pytest.raises(NotImplementedError, match="torch.distributed"):
         ydf.DistributedGradientBoostedTreesLearner(
             label="label", workers=["host:2001"])
+
+
+def test_predict_shap_reference_tuple_form(regression_data):
+    """predict_shap unpacks as (values, initial_value) like the
+    reference (generic_model.py:507) while the legacy mapping form
+    keeps working."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", task=ydf.Task.REGRESSION, num_trees=10,
+        validation_ratio=0).train(regression_data)
+    values, initial = m.predict_shap(regression_data, num_threads=2)
+    assert set(values) == set(m.input_feature_names())
+    total = np.stack(list(values.values()), 1).sum(1) + initial
+    np.testing.assert_allclose(total, m.predict(regression_data),
+                               atol=1e-4)

@@ -33,6 +33,34 @@ class ModelIOOptions:
 _WEIGHTS_DISABLED = object()  # evaluate(weighted=False) sentinel
 
 
+class ShapValues(tuple):
+    """predict_shap result: the reference tuple shape
+    (values: Dict[feature -> phi[N]], initial_value) that ALSO answers
+    the legacy mapping form (shap["feature"], shap["__BIAS__"],
+    .items()). Unpack `values, initial = model.predict_shap(data)` or
+    index by feature name."""
+
+    def __new__(cls, values, initial):
+        return super().__new__(cls, (values, initial))
+
+    def __getitem__(self, k):
+        if isinstance(k, str):
+            if k == "__BIAS__":
+                return tuple.__getitem__(self, 1)
+            return tuple.__getitem__(self, 0)[k]
+        return tuple.__getitem__(self, k)
+
+    def __contains__(self, k):
+        return k == "__BIAS__" or k in tuple.__getitem__(self, 0)
+
+    def items(self):
+        yield from tuple.__getitem__(self, 0).items()
+        yield "__BIAS__", tuple.__getitem__(self, 1)
+
+    def keys(self):
+        return list(tuple.__getitem__(self, 0).keys()) + ["__BIAS__"]
+
+
 def default_device() -> torch.device:
     return torch.device("cuda") if torch.cuda.is_available() else \
         torch.device("cpu")
@@ -778,7 +806,7 @@ class GenericModel:
         return ev
 
     # ------------------------------------------------------------------
-    def predict_shap(self, data) -> Dict[str, np.ndarray]:
+    def predict_shap(self, data, *, num_threads=None):
         """Path-dependent TreeSHAP values (reference utils/shap.h:83;
         mirrors ydf model.predict_shap). Returns {feature: phi [N]} plus
         the expected value under "__BIAS__"; margins (pre-activation)
@@ -821,10 +849,11 @@ class GenericModel:
             feat.ctypes.data, thr.ctypes.data, left.ctypes.data,
             cover.ctypes.data, roots.ctypes.data, 0, 1, f.n_trees, scale)
         phi[:, F] = float(self.init_predictions[0]) + ev
+        if num_threads is not None:
+            torch.set_num_threads(int(num_threads))
         out = {name: phi[:, i]
                for i, name in enumerate(self.input_feature_names())}
-        out["__BIAS__"] = phi[:, F]
-        return out
+        return ShapValues(out, phi[:, F])
 
     def analyze_prediction(self, single_example) -> Dict:
         """Per-example prediction analysis (mirrors PYDF
