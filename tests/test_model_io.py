@@ -296,3 +296,17 @@ def test_reference_api_compat_surface(trained, binary_data):
     with pytest.raises(ValueError):
         trained.set_node_format("TFE_RECORDIO")
     trained.set_data_spec(trained.data_spec())
+
+
+def test_describe_formats(trained):
+    """describe() output_format auto/text/html/notebook +
+    full_details (reference generic_model.py:277)."""
+    t = trained.describe()  # auto -> text
+    assert "GRADIENT_BOOSTED_TREES" in t
+    assert trained.describe("text") == t
+    h = trained.describe("html")
+    assert "<table>" in h
+    nb = trained.describe("notebook")
+    assert hasattr(nb, "_repr_html_") and "<table>" in nb._repr_html_()
+    fd = trained.describe("text", full_details=True)
+    assert "tree 0:" in fd and len(fd) > len(t)

@@ -971,9 +971,15 @@ class GenericModel:
         )
 
     # ------------------------------------------------------------------
-    def describe(self, output_format: str = "text") -> str:
+    def describe(self, output_format: str = "auto",
+                 full_details: bool = False):
         """Model card (reference model/describe.{h,cc}; ydf
-        model.describe()). output_format: "text" or "html"."""
+        model.describe()). output_format: "auto" (text on a terminal),
+        "text", "html", or "notebook" (object with _repr_html_);
+        full_details appends hyperparameter metadata and the first
+        tree."""
+        if output_format == "auto":
+            output_format = "text"
         lines = [
             f'type: "{self._model_type}"',
             f"task: {self._task.name}",
@@ -1014,7 +1020,15 @@ class GenericModel:
                 f"score {bt.score:.6g}, best hyperparameters "
                 + ", ".join(f"{k}={v}"
                             for k, v in bt.hyperparameters.items()))
-        if output_format == "html":
+        if full_details:
+            if self.metadata:
+                lines.append("metadata: " + ", ".join(
+                    f"{k}={v}" for k, v in sorted(self.metadata.items())
+                    if not isinstance(v, (dict, list))))
+            if self.num_trees() and hasattr(self, "print_tree"):
+                lines.append("tree 0:")
+                lines.append(self.print_tree(0))
+        if output_format in ("html", "notebook"):
             rows = "".join(f"<tr><td>{ln.split(':', 1)[0]}</td>"
                            f"<td>{ln.split(':', 1)[1] if ':' in ln else ''}"
                            "</td></tr>" for ln in lines)
@@ -1037,6 +1051,12 @@ class GenericModel:
                              f'<svg width="{w}" height="{h}">'
                              f'<polyline fill="none" stroke="steelblue" '
                              f'points="{poly}"/></svg>')
+            if output_format == "notebook":
+                class _Display(str):
+                    def _repr_html_(self):
+                        return str(self)
+
+                return _Display(html)
             return html
         return "\n".join(lines)
 
