@@ -310,3 +310,39 @@ def test_describe_formats(trained):
     assert hasattr(nb, "_repr_html_") and "<table>" in nb._repr_html_()
     fd = trained.describe("text", full_details=True)
     assert "tree 0:" in fd and len(fd) > len(t)
+
+
+def test_family_model_reference_methods(trained, binary_data):
+    """Family-specific reference methods: GBT num_trees_per_iteration /
+    output_logits / set_initial_predictions / validation_evaluation /
+    early_stopping_triggered; RF out_of_bag_evaluations +
+    winner_takes_all; IF num_examples_per_tree; activation duality."""
+    assert trained.num_trees_per_iteration() == 1
+    assert trained.activation == trained.activation()
+    assert trained.output_logits() is False
+    p0 = trained.predict(binary_data)
+    trained.set_output_logits(True)
+    logits = trained.predict(binary_data)
+    np.testing.assert_allclose(1 / (1 + np.exp(-logits)), p0, atol=1e-5)
+    trained.set_output_logits(False)
+    bias0 = list(trained.init_predictions)
+    trained.set_initial_predictions([float(bias0[0]) + 1.0])
+    p1 = trained.predict(binary_data)
+    assert (p1 >= p0 - 1e-6).all() and p1.mean() > p0.mean()
+    trained.set_initial_predictions(bias0)
+    assert trained.early_stopping_triggered() in (True, False, None)
+
+    mrf = ydf.RandomForestLearner(label="label", num_trees=8).train(
+        binary_data)
+    assert mrf.out_of_bag_evaluations() is not None
+    assert isinstance(mrf.winner_takes_all(), bool)
+    m_nw = ydf.RandomForestLearner(label="label", num_trees=8,
+                                   winner_take_all=False).train(
+        binary_data)
+    assert m_nw.winner_takes_all() is False
+
+    rng = np.random.RandomState(0)
+    mif = ydf.IsolationForestLearner(num_trees=10).train(
+        {"a": rng.randn(500).astype(np.float32)})
+    assert mif.num_examples_per_tree == 256
+    assert mif.num_examples_per_tree() == 256

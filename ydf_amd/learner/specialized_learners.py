@@ -768,6 +768,9 @@ class GradientBoostedTreesLearner(GenericLearner):
                 gains[k] = gains.get(k, 0.0) + v
         model = make_model(flat, init_preds, gains)
         model.training_logs = logs
+        model.metadata["early_stopping_triggered"] = bool(
+            len(trees) // max(C, 1) < hp["num_trees"]
+            and hp["early_stopping"] != "NONE")
         self._finalize_model(model)
         if not hp.get("keep_non_leaf_label_distribution", True):
             # drop non-leaf training distributions (reference

@@ -112,6 +112,16 @@ class _CallableDict(dict):
         return self
 
 
+class _CallableStr(str):
+    def __call__(self):
+        return self
+
+
+class _CallableInt(int):
+    def __call__(self):
+        return self
+
+
 class GenericModel:
     """Base decision-forest model."""
 
@@ -153,6 +163,14 @@ class GenericModel:
     @metadata.setter
     def metadata(self, v):
         self._metadata = _CallableDict(v) if isinstance(v, dict) else v
+
+    @property
+    def activation(self):
+        return self._activation
+
+    @activation.setter
+    def activation(self, v):
+        self._activation = _CallableStr(v) if isinstance(v, str) else v
 
     @property
     def training_logs(self):

@@ -168,6 +168,53 @@ class GradientBoostedTreesModel(DecisionForestModel):
     def initial_predictions(self) -> np.ndarray:
         return np.asarray(self.init_predictions, dtype=np.float32)
 
+    def num_trees_per_iteration(self) -> int:
+        """Trees per boosting iteration (PYDF
+        gradient_boosted_trees_model.py:256): 1, or the class count
+        for multi-class models."""
+        return int(self.num_trees_per_iter)
+
+    def early_stopping_triggered(self):
+        """Whether training stopped early on the validation loss
+        (None when unknown, e.g. imported models)."""
+        v = (self.metadata or {}).get("early_stopping_triggered")
+        return None if v is None else bool(v)
+
+    def set_initial_predictions(self, initial_predictions) -> None:
+        """Sets the model bias (PYDF set_initial_predictions)."""
+        self.init_predictions = [float(v) for v in initial_predictions]
+        self._dev_forest.clear()
+
+    def output_logits(self) -> bool:
+        """True when classification predictions are raw margins
+        instead of probabilities (PYDF output_logits)."""
+        return self.activation == "identity" \
+            and self._task == Task.CLASSIFICATION
+
+    def set_output_logits(self, output_logits: bool) -> None:
+        if self._task != Task.CLASSIFICATION:
+            raise ValueError(
+                "output_logits only applies to classification models")
+        if output_logits:
+            self.activation = "identity"
+        else:
+            self.activation = "softmax" if (
+                self.label_classes and len(self.label_classes) > 2) \
+                else "sigmoid"
+
+    def validation_evaluation(self):
+        """Evaluation on the training-time validation set, from the
+        last training-log entry (PYDF validation_evaluation)."""
+        if not self.training_logs:
+            return None
+        from ydf_amd.metric.metric import Evaluation
+
+        last = self.training_logs[-1]
+        ev = Evaluation(loss=last.get("valid_loss"))
+        ev.custom_metrics = {k: v for k, v in last.items()
+                             if k not in ("iteration", "valid_loss")}
+        return ev
+
     def validation_loss(self):
         if self.training_logs:
             return self.training_logs[-1].get("valid_loss")
@@ -176,6 +223,16 @@ class GradientBoostedTreesModel(DecisionForestModel):
 
 class RandomForestModel(DecisionForestModel):
     _model_type = "RANDOM_FOREST"
+
+    def out_of_bag_evaluations(self):
+        """OOB evaluation logs (PYDF random_forest_model.py:31; alias
+        of training_logs for Random Forests)."""
+        return self.training_logs or []
+
+    def winner_takes_all(self) -> bool:
+        """Whether classification aggregates one vote per tree instead
+        of summed leaf probabilities (PYDF winner_takes_all)."""
+        return bool((self.metadata or {}).get("winner_take_all", False))
 
     def _leaf_scale(self) -> float:
         C = self._n_outputs()
@@ -193,6 +250,18 @@ class IsolationForestModel(DecisionForestModel):
     depth + c(n_leaf))."""
 
     _model_type = "ISOLATION_FOREST"
+
+    @property
+    def num_examples_per_tree(self):
+        """Examples used to grow each tree (PYDF
+        isolation_forest_model.py:26; attribute AND method form)."""
+        return self._num_examples_per_tree
+
+    @num_examples_per_tree.setter
+    def num_examples_per_tree(self, v):
+        from ydf_amd.model.generic_model import _CallableInt
+
+        self._num_examples_per_tree = _CallableInt(int(v))
 
     def __init__(self, *args, num_examples_per_tree: int = 256, **kwargs):
         super().__init__(*args, **kwargs)
