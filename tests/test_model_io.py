@@ -346,3 +346,15 @@ def test_family_model_reference_methods(trained, binary_data):
         {"a": rng.randn(500).astype(np.float32)})
     assert mif.num_examples_per_tree == 256
     assert mif.num_examples_per_tree() == 256
+
+
+def test_training_log_entry_attr_access(binary_data):
+    """training_logs entries answer the reference attribute form
+    (entry.iteration, entry.evaluation) and the dict form."""
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=20, validation_ratio=0.2).train(
+        binary_data)
+    logs = m.training_logs()
+    assert logs and logs[-1].iteration == logs[-1]["iteration"]
+    ev = logs[-1].evaluation
+    assert ev.loss == logs[-1]["valid_loss"]

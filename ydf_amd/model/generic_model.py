@@ -122,6 +122,25 @@ class _CallableInt(int):
         return self
 
 
+class TrainingLogEntry(dict):
+    """One training-log record. Dict form ({"iteration", "valid_loss",
+    ...}) plus the reference attribute form (entry.iteration,
+    entry.evaluation — PYDF generic_model.TrainingLogEntry)."""
+
+    def __getattr__(self, k):
+        if k == "evaluation":
+            from ydf_amd.metric.metric import Evaluation
+
+            ev = Evaluation(loss=self.get("valid_loss"))
+            ev.custom_metrics = {kk: v for kk, v in self.items()
+                                 if kk not in ("iteration", "valid_loss")}
+            return ev
+        try:
+            return self[k]
+        except KeyError:
+            raise AttributeError(k) from None
+
+
 class GenericModel:
     """Base decision-forest model."""
 
@@ -178,8 +197,11 @@ class GenericModel:
 
     @training_logs.setter
     def training_logs(self, v):
-        self._training_logs = _CallableList(v) if isinstance(
-            v, (list, tuple)) else v
+        if isinstance(v, (list, tuple)):
+            v = _CallableList(
+                TrainingLogEntry(e) if isinstance(e, dict) else e
+                for e in v)
+        self._training_logs = v
 
     def __getstate__(self):
         # models pickle (reference PYDF __getstate__/__setstate__ via
