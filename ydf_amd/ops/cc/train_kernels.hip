@@ -1057,27 +1057,44 @@ __global__ void split_select_kernel(const float* __restrict__ hist,
   const int abs_node = w_abs;
   if (!is_cat) {
     const int bin = w_bin;
-    // block-parallel left-child sum over bins [0, bin] (the serial
-    // thread-0 loop was ~2/3 of this kernel's 14.5 us)
-    float part[3] = {0.f, 0.f, 0.f};
-    for (int bb = t; bb <= bin; bb += blockDim.x) {
-      part[0] += hp[bb * 3];
-      part[1] += hp[bb * 3 + 1];
-      part[2] += hp[bb * 3 + 2];
-    }
-    float red[3];
-    for (int c = 0; c < 3; ++c) {
-      rg[t] = part[c];
-      __syncthreads();
-      for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
-        if (t < off) rg[t] += rg[t + off];
+    // Small grids (shallow GBT levels: few slots = few blocks) are
+    // latency-bound on the serial thread-0 prefix — parallelize it
+    // across the block (~2/3 of the kernel's 14.5 us there). Large
+    // grids (deep RF levels: 1000+ blocks) already hide that latency
+    // with block parallelism and only pay the reduction's syncs, so
+    // they keep the serial form.
+    float GL, HL, CL;
+    if (gridDim.x <= 64) {
+      float part[3] = {0.f, 0.f, 0.f};
+      for (int bb = t; bb <= bin; bb += blockDim.x) {
+        part[0] += hp[bb * 3];
+        part[1] += hp[bb * 3 + 1];
+        part[2] += hp[bb * 3 + 2];
+      }
+      float red[3];
+      for (int c = 0; c < 3; ++c) {
+        rg[t] = part[c];
+        __syncthreads();
+        for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+          if (t < off) rg[t] += rg[t + off];
+          __syncthreads();
+        }
+        red[c] = rg[0];
         __syncthreads();
       }
-      red[c] = rg[0];
-      __syncthreads();
+      if (t != 0) return;
+      GL = red[0];
+      HL = red[1];
+      CL = red[2];
+    } else {
+      if (t != 0) return;
+      GL = HL = CL = 0.f;
+      for (int bb = 0; bb <= bin; ++bb) {
+        GL += hp[bb * 3];
+        HL += hp[bb * 3 + 1];
+        CL += hp[bb * 3 + 2];
+      }
     }
-    if (t != 0) return;
-    float GL = red[0], HL = red[1], CL = red[2];
     if (sp.na_mode && na_meanb_nf != nullptr) {
       // the scan folded the NA bin (255) into the node-local mean bin;
       // mirror that here and record the na direction for routing
