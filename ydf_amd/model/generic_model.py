@@ -469,11 +469,13 @@ class GenericModel:
                 and len(self.forest.masks) == 0 \
                 and len(self.forest.obl_ranges) == 0 \
                 and not self.forest.has_na_routing \
-                and X.shape[1] >= 65536 and self._thresholds_on_cuts():
+                and self._thresholds_on_cuts():
             # auto-select: the compact-node binned engine is ~2x the
             # flat engine on large batches (half the L2 node traffic,
             # profiles/serving_engines_r02.md) and bit-equivalent when
-            # thresholds sit on training cuts
+            # thresholds sit on training cuts; small batches route to
+            # its tree-parallel grid (binned4_tp) instead of serially
+            # walking the whole forest per thread
             eng = "binned8"
         if eng == "binned8" and X.is_cuda:
             return self._predict_margin_binned8(X)
@@ -971,6 +973,15 @@ class GenericModel:
             permutation_rounds=permutation_variable_importance_rounds,
             features=features, num_grid_points=num_bins, device=device)
 
+    def serving_session(self, batch_size: int, device=None):
+        """Graph-captured repeated inference at a fixed batch size —
+        the MI355X answer to the reference's latency-critical serving
+        engines: the eager bin+walk launch/sync overhead (~1 ms/batch
+        measured at small batches, profiles/serving_engines_r02.md) is
+        replaced by ONE hipGraph replay. Returns a ServingSession with
+        .predict(data) for repeated same-shape batches."""
+        return ServingSession(self, batch_size, device=device)
+
     # ------------------------------------------------------------------
     def benchmark(self, data, benchmark_duration: float = 3.0,
                   warmup_duration: float = 0.5, batch_size: int = 0,
@@ -1192,3 +1203,64 @@ class BenchmarkResult:
     def __str__(self) -> str:
         return (f"{self.examples_per_second:,.0f} examples/s "
                 f"({self.num_runs} runs over {self.duration_seconds:.2f}s)")
+
+
+class ServingSession:
+    """Fixed-batch-size inference with the whole pipeline (binning +
+    forest walk + activation) captured in one hipGraph. Feed
+    same-shaped batches through predict(); input upload and output
+    download stay outside the graph."""
+
+    def __init__(self, model, batch_size: int, device=None):
+        dev = torch.device(device) if device is not None \
+            else default_device()
+        if dev.type != "cuda":
+            raise ValueError(
+                "serving_session requires a GPU device (use predict() "
+                "on CPU)")
+        self._model = model
+        self._n = int(batch_size)
+        F = len(model.dataspec.feature_columns)
+        self._X = torch.zeros((F, self._n), dtype=torch.float32,
+                              device=dev)
+        self._pin = torch.empty((F, self._n), dtype=torch.float32,
+                                pin_memory=True)
+        # warm every lazy cache (device forest, packed engines,
+        # threshold checks) before capture
+        for _ in range(2):
+            model._apply_activation(model.predict_margin(self._X))
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._out = model._apply_activation(
+                model.predict_margin(self._X))
+        self._graph = g
+        # pre-bound pinned staging on both sides: predict() does zero
+        # tensor allocation
+        self._pin_np = self._pin.numpy()
+        self._out_pin = torch.empty(self._out.shape,
+                                    dtype=self._out.dtype,
+                                    pin_memory=True)
+        self._out_np = self._out_pin.numpy()
+
+    def predict(self, data) -> np.ndarray:
+        # fast path: a preformed feature-major [F, batch] f32 matrix
+        # skips per-call column encoding (the dominant Python cost at
+        # small batches)
+        if isinstance(data, np.ndarray) and data.ndim == 2 \
+                and data.shape == self._X.shape:
+            X_np = data
+        else:
+            X_np = self._model._encode_features(data)
+        if X_np.shape[1] != self._n:
+            raise ValueError(
+                f"serving_session captured for batch size {self._n}, "
+                f"got {X_np.shape[1]} examples")
+        np.copyto(self._pin_np, X_np)
+        self._X.copy_(self._pin, non_blocking=True)
+        self._graph.replay()
+        self._out_pin.copy_(self._out, non_blocking=True)
+        torch.cuda.synchronize(self._X.device)
+        return self._out_np.copy()
+
+    __call__ = predict

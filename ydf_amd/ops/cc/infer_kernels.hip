@@ -380,6 +380,80 @@ __global__ void predict_forest_binned4_kernel(
   out[base + tid] = init + (acc - init) * scale;
 }
 
+// Tree-parallel variant for SMALL batches: one thread per row walks a
+// serial chain of dependent node loads, so at N << 65k the row-only
+// grid leaves the chip idle while each thread walks ALL trees
+// (~250 x depth x L2-latency = the measured ~1 ms small-batch floor).
+// Here grid.y splits the forest into tree chunks; each block writes its
+// chunk's leaf-sum to a partial buffer and a tiny reduction combines
+// them in FIXED chunk order (deterministic — no float atomics).
+__global__ void predict_forest_binned4_tp_kernel(
+    const uint8_t* __restrict__ B, int64_t N, int F,
+    const uint32_t* __restrict__ nodes,
+    const float* __restrict__ leaf_vals,
+    const int32_t* __restrict__ roots, int tree_start, int tree_step,
+    int n_trees, int trees_per_chunk, float* __restrict__ partial) {
+  extern __shared__ uint8_t bs[];  // [F][kTile]
+  const int tid = threadIdx.x;
+  const int64_t base = (int64_t)blockIdx.x * kTile;
+  const int64_t n_here = min((int64_t)kTile, N - base);
+  if (n_here <= 0) return;
+  for (int idx = tid; idx < F * kTile; idx += blockDim.x) {
+    const int f = idx >> 8;
+    const int i = idx & 255;
+    bs[idx] = (i < n_here) ? B[(int64_t)f * N + base + i] : 0;
+  }
+  __syncthreads();
+  if (tid >= n_here) return;
+  const int t_lo = blockIdx.y * trees_per_chunk;
+  const int t_hi = min(t_lo + trees_per_chunk, n_trees);
+  float acc = 0.f;
+  int t = t_lo;
+  for (; t + 4 <= t_hi; t += 4) {
+    uint32_t nd[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      nd[u] = nodes[roots[tree_start + (int64_t)(t + u) * tree_step]];
+    bool done = false;
+    while (!done) {
+      done = true;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const int f = nd[u] & 63;
+        if (f != 63) {
+          const int right =
+              (int)bs[f * kTile + tid] > (int)((nd[u] >> 6) & 255);
+          nd[u] = nodes[(nd[u] >> 14) + right];
+          done &= (nd[u] & 63) == 63;
+        }
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc += leaf_vals[nd[u] >> 14];
+  }
+  for (; t < t_hi; ++t) {
+    uint32_t nd = nodes[roots[tree_start + (int64_t)t * tree_step]];
+    while ((nd & 63) != 63) {
+      const int right =
+          (int)bs[(nd & 63) * kTile + tid] > (int)((nd >> 6) & 255);
+      nd = nodes[(nd >> 14) + right];
+    }
+    acc += leaf_vals[nd >> 14];
+  }
+  partial[(int64_t)blockIdx.y * N + base + tid] = acc;
+}
+
+__global__ void reduce_partials_kernel(const float* __restrict__ partial,
+                                       int n_chunks, int64_t N,
+                                       float* __restrict__ out,
+                                       float init, float scale) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= N) return;
+  float s = 0.f;
+  for (int c = 0; c < n_chunks; ++c) s += partial[(int64_t)c * N + i];
+  out[i] = init + s * scale;
+}
+
 __global__ void sigmoid_kernel(const float* __restrict__ in,
                                float* __restrict__ out, int64_t N) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -483,6 +557,29 @@ void gpu_predict_forest_binned4(const uint8_t* B, int64_t N, int F,
                      dim3(kTile), lds, (hipStream_t)stream, B, N, F,
                      nodes4, leaf_vals, roots, tree_start, tree_step,
                      n_trees, out, init, scale);
+}
+
+void gpu_predict_forest_binned4_tp(const uint8_t* B, int64_t N, int F,
+                                   const uint32_t* nodes4,
+                                   const float* leaf_vals,
+                                   const int32_t* roots, int tree_start,
+                                   int tree_step, int n_trees,
+                                   int n_chunks, float* partial,
+                                   float* out, float init, float scale,
+                                   void* stream) {
+  const size_t lds = (size_t)F * kTile;
+  const int row_tiles = (int)((N + kTile - 1) / kTile);
+  const int tpc = (n_trees + n_chunks - 1) / n_chunks;
+  const int chunks = (n_trees + tpc - 1) / tpc;
+  hipLaunchKernelGGL(predict_forest_binned4_tp_kernel,
+                     dim3(row_tiles, chunks), dim3(kTile), lds,
+                     (hipStream_t)stream, B, N, F, nodes4, leaf_vals,
+                     roots, tree_start, tree_step, n_trees, tpc,
+                     partial);
+  const int rg = (int)((N + 255) / 256);
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rg), dim3(256), 0,
+                     (hipStream_t)stream, partial, chunks, N, out, init,
+                     scale);
 }
 
 void gpu_predict_forest_qs(const float* X, int64_t N, int F,
