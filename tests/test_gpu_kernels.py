@@ -727,3 +727,30 @@ def test_serving_session_graph_latency():
     print(f"# serving latency: graph {t_graph*1e6:.0f}us vs eager "
           f"{t_eager*1e6:.0f}us per {B}-row batch")
     assert t_graph < t_eager, (t_graph, t_eager)
+
+
+@pytest.mark.gpu
+def test_small_batch_engine_equality():
+    """The tree-parallel small-batch binned4 grid must match the flat
+    engine exactly across tiny batch sizes (deterministic fixed-order
+    partial reduction)."""
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(1)
+    n = 100000
+    d = {f"x{i}": rng.randn(n).astype(np.float32) for i in range(8)}
+    d["label"] = np.where(d["x0"] - d["x1"] * d["x2"] > 0, "a", "b")
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=500, max_depth=6,
+        validation_ratio=0.0, device="cuda:0").train(d)
+    for B in (1, 7, 100, 1000, 30000):
+        batch = {k: v[:B] for k, v in d.items() if k != "label"}
+        p_auto = m.predict(batch, device="cuda:0")  # binned4(+tp)
+        m.force_engine("flat")
+        p_flat = m.predict(batch, device="cuda:0")
+        m.force_engine(None)
+        np.testing.assert_allclose(p_auto, p_flat, rtol=1e-6,
+                                   atol=1e-7)
+        # determinism across repeated calls
+        np.testing.assert_array_equal(
+            p_auto, m.predict(batch, device="cuda:0"))
