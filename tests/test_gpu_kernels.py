@@ -749,8 +749,10 @@ def test_small_batch_engine_equality():
         m.force_engine("flat")
         p_flat = m.predict(batch, device="cuda:0")
         m.force_engine(None)
-        np.testing.assert_allclose(p_auto, p_flat, rtol=1e-6,
-                                   atol=1e-7)
-        # determinism across repeated calls
+        # chunked partial sums reduce in a different (fixed) order
+        # than the flat walk -> ulp-level float differences only
+        np.testing.assert_allclose(p_auto, p_flat, rtol=1e-5,
+                                   atol=1e-6)
+        # determinism across repeated calls is EXACT
         np.testing.assert_array_equal(
             p_auto, m.predict(batch, device="cuda:0"))
