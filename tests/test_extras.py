@@ -584,3 +584,12 @@ def test_analyze_keyword_surface(binary_data):
     assert not a2.conditional_expectations
     assert "MEAN_ABS_SHAP" not in a2.variable_importances
     assert "MEAN_DECREASE_IN_ACCURACY" not in a2.variable_importances
+
+
+def test_serving_session_requires_gpu(binary_data):
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=5, validation_ratio=0.0).train(
+        binary_data)
+    if not __import__("torch").cuda.is_available():
+        with pytest.raises(ValueError, match="GPU"):
+            m.serving_session(64, device="cpu")
