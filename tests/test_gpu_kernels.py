@@ -756,3 +756,36 @@ def test_small_batch_engine_equality():
         # determinism across repeated calls is EXACT
         np.testing.assert_array_equal(
             p_auto, m.predict(batch, device="cuda:0"))
+
+
+@pytest.mark.gpu
+def test_small_batch_binned8_wide_model():
+    """binned8 tree-parallel path for wide-feature models (>62
+    features exceed the 4-byte packing): small batches must match the
+    flat engine and stay deterministic."""
+    import ydf_amd as ydf
+
+    rng = np.random.RandomState(2)
+    n = 60000
+    F = 80
+    d = {f"x{i}": rng.randn(n).astype(np.float32) for i in range(F)}
+    d["label"] = np.where(
+        d["x0"] + d["x1"] - d["x70"] + 0.3 * rng.randn(n) > 0, "a", "b")
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=300, max_depth=5,
+        validation_ratio=0.0, device="cuda:0").train(d)
+    from ydf_amd.model.forest import pack_binned4_nodes, padded_boundaries
+    import pytest as _pt
+
+    with _pt.raises(ValueError):
+        pack_binned4_nodes(m.forest,
+                           padded_boundaries(m.dataspec.feature_columns))
+    for B in (100, 2000):
+        batch = {k: v[:B] for k, v in d.items() if k != "label"}
+        p_auto = m.predict(batch, device="cuda:0")
+        m.force_engine("flat")
+        p_flat = m.predict(batch, device="cuda:0")
+        m.force_engine(None)
+        np.testing.assert_allclose(p_auto, p_flat, rtol=1e-5, atol=1e-6)
+        np.testing.assert_array_equal(
+            p_auto, m.predict(batch, device="cuda:0"))

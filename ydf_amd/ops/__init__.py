@@ -470,6 +470,17 @@ def predict_forest_binned8(B: torch.Tensor, packed8: torch.Tensor,
     F, N = B.shape
     if n_trees < 0:
         n_trees = roots.numel()
+    row_tiles = (N + 255) // 256
+    if row_tiles * 4 < 512 and n_trees >= 64:
+        # small batch: tree-parallel grid (see predict_forest_binned4)
+        n_chunks = max(1, min(512 // max(row_tiles, 1), n_trees // 16))
+        partial = torch.empty((n_chunks, N), dtype=torch.float32,
+                              device=B.device)
+        _C.gpu_predict_forest_binned8_tp(
+            B.data_ptr(), N, F, packed8.data_ptr(), roots.data_ptr(),
+            tree_start, tree_step, n_trees, n_chunks,
+            partial.data_ptr(), out.data_ptr(), init, 1.0, _stream())
+        return out
     _C.gpu_predict_forest_binned8(B.data_ptr(), N, F,
                                   packed8.data_ptr(), roots.data_ptr(),
                                   tree_start, tree_step, n_trees,

@@ -87,6 +87,10 @@ void gpu_predict_forest_binned4_tp(const uint8_t*, int64_t, int,
                                    const uint32_t*, const float*,
                                    const int32_t*, int, int, int, int,
                                    float*, float*, float, float, void*);
+void gpu_predict_forest_binned8_tp(const uint8_t*, int64_t, int,
+                                   const uint32_t*, const int32_t*, int,
+                                   int, int, int, float*, float*, float,
+                                   float, void*);
 // cpu_ops.cpp
 void cpu_bin_data(const float*, const float*, uint8_t*, int64_t, int, int,
                   int);
@@ -395,6 +399,18 @@ PYBIND11_MODULE(_ydf_ops, m) {
                                      P<int32_t>(roots), tree_start,
                                      tree_step, n_trees, P<float>(out),
                                      init, scale, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_predict_forest_binned8_tp",
+        [](uintptr_t B, int64_t N, int F, uintptr_t nodes8,
+           uintptr_t roots, int tree_start, int tree_step, int n_trees,
+           int n_chunks, uintptr_t partial, uintptr_t out, float init,
+           float scale, uintptr_t stream) {
+          gpu_predict_forest_binned8_tp(
+              P<uint8_t>(B), N, F, P<uint32_t>(nodes8),
+              P<int32_t>(roots), tree_start, tree_step, n_trees,
+              n_chunks, P<float>(partial), P<float>(out), init, scale,
+              (void*)stream);
         },
         nogil);
   m.def("gpu_predict_forest_binned4_tp",

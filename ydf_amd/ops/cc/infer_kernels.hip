@@ -380,6 +380,62 @@ __global__ void predict_forest_binned4_kernel(
   out[base + tid] = init + (acc - init) * scale;
 }
 
+// Tree-parallel small-batch variant of the 8-byte-node engine (same
+// rationale as predict_forest_binned4_tp_kernel below; used for
+// wide-feature models that exceed the 4-byte packing limits).
+__global__ void predict_forest_binned8_tp_kernel(
+    const uint8_t* __restrict__ B, int64_t N, int F,
+    const Node8* __restrict__ nodes, const int32_t* __restrict__ roots,
+    int tree_start, int tree_step, int n_trees, int trees_per_chunk,
+    float* __restrict__ partial) {
+  extern __shared__ uint8_t bs[];  // [F][kTile]
+  const int tid = threadIdx.x;
+  const int64_t base = (int64_t)blockIdx.x * kTile;
+  const int64_t n_here = min((int64_t)kTile, N - base);
+  if (n_here <= 0) return;
+  for (int idx = tid; idx < F * kTile; idx += blockDim.x) {
+    const int f = idx >> 8;
+    const int i = idx & 255;
+    bs[idx] = (i < n_here) ? B[(int64_t)f * N + base + i] : 0;
+  }
+  __syncthreads();
+  if (tid >= n_here) return;
+  const int t_lo = blockIdx.y * trees_per_chunk;
+  const int t_hi = min(t_lo + trees_per_chunk, n_trees);
+  float acc = 0.f;
+  int t = t_lo;
+  for (; t + 4 <= t_hi; t += 4) {
+    Node8 nd[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      nd[u] = nodes[roots[tree_start + (int64_t)(t + u) * tree_step]];
+    bool done = false;
+    while (!done) {
+      done = true;
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        if (nd[u].feat != 0xFFFFu) {
+          const int right =
+              (int)bs[nd[u].feat * kTile + tid] > (int)nd[u].bin;
+          nd[u] = nodes[nd[u].left_or_val + right];
+          done &= nd[u].feat == 0xFFFFu;
+        }
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc += __uint_as_float(nd[u].left_or_val);
+  }
+  for (; t < t_hi; ++t) {
+    Node8 nd = nodes[roots[tree_start + (int64_t)t * tree_step]];
+    while (nd.feat != 0xFFFFu) {
+      const int right = (int)bs[nd.feat * kTile + tid] > (int)nd.bin;
+      nd = nodes[nd.left_or_val + right];
+    }
+    acc += __uint_as_float(nd.left_or_val);
+  }
+  partial[(int64_t)blockIdx.y * N + base + tid] = acc;
+}
+
 // Tree-parallel variant for SMALL batches: one thread per row walks a
 // serial chain of dependent node loads, so at N << 65k the row-only
 // grid leaves the chip idle while each thread walks ALL trees
@@ -557,6 +613,28 @@ void gpu_predict_forest_binned4(const uint8_t* B, int64_t N, int F,
                      dim3(kTile), lds, (hipStream_t)stream, B, N, F,
                      nodes4, leaf_vals, roots, tree_start, tree_step,
                      n_trees, out, init, scale);
+}
+
+void gpu_predict_forest_binned8_tp(const uint8_t* B, int64_t N, int F,
+                                   const uint32_t* nodes8_u,
+                                   const int32_t* roots, int tree_start,
+                                   int tree_step, int n_trees,
+                                   int n_chunks, float* partial,
+                                   float* out, float init, float scale,
+                                   void* stream) {
+  const size_t lds = (size_t)F * kTile;
+  const int row_tiles = (int)((N + kTile - 1) / kTile);
+  const int tpc = (n_trees + n_chunks - 1) / n_chunks;
+  const int chunks = (n_trees + tpc - 1) / tpc;
+  hipLaunchKernelGGL(predict_forest_binned8_tp_kernel,
+                     dim3(row_tiles, chunks), dim3(kTile), lds,
+                     (hipStream_t)stream, B, N, F,
+                     reinterpret_cast<const Node8*>(nodes8_u), roots,
+                     tree_start, tree_step, n_trees, tpc, partial);
+  const int rg = (int)((N + 255) / 256);
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rg), dim3(256), 0,
+                     (hipStream_t)stream, partial, chunks, N, out, init,
+                     scale);
 }
 
 void gpu_predict_forest_binned4_tp(const uint8_t* B, int64_t N, int F,
