@@ -375,6 +375,22 @@ def predict_forest(X: torch.Tensor, feat: torch.Tensor, thr: torch.Tensor,
             ci = cat_idx if cat_idx is not None else torch.full(
                 (feat.numel(),), -1, dtype=torch.int32, device=X.device)
             packed = pack_forest_nodes(feat, thr, left, ci)
+        row_tiles = (N + 255) // 256
+        lds_ok = F * 256 * 4 <= 96 * 1024
+        if lds_ok and row_tiles * 4 < 512 and n_trees >= 64:
+            # small batch: tree-parallel grid (deterministic per-chunk
+            # partials; see predict_forest_binned4)
+            n_chunks = max(1, min(512 // max(row_tiles, 1),
+                                  n_trees // 16))
+            partial = torch.empty((n_chunks, N), dtype=torch.float32,
+                                  device=X.device)
+            _C.gpu_predict_forest_tp(
+                X.data_ptr(), N, F, packed.data_ptr(),
+                roots.data_ptr(), mk, orr, oa, ow, nr,
+                1 if special else 0, tree_start, tree_step, n_trees,
+                n_chunks, partial.data_ptr(), out.data_ptr(), init,
+                scale, _stream())
+            return out
         _C.gpu_predict_forest(X.data_ptr(), N, F, packed.data_ptr(),
                               roots.data_ptr(), mk, orr, oa, ow, nr,
                               1 if special else 0, tree_start,

@@ -87,6 +87,11 @@ void gpu_predict_forest_binned4_tp(const uint8_t*, int64_t, int,
                                    const uint32_t*, const float*,
                                    const int32_t*, int, int, int, int,
                                    float*, float*, float, float, void*);
+void gpu_predict_forest_tp(const float*, int64_t, int, const int32_t*,
+                           const int32_t*, const unsigned long long*,
+                           const int32_t*, const int32_t*, const float*,
+                           const uint8_t*, int, int, int, int, int,
+                           float*, float*, float, float, void*);
 void gpu_predict_forest_binned8_tp(const uint8_t*, int64_t, int,
                                    const uint32_t*, const int32_t*, int,
                                    int, int, int, float*, float*, float,
@@ -399,6 +404,22 @@ PYBIND11_MODULE(_ydf_ops, m) {
                                      P<int32_t>(roots), tree_start,
                                      tree_step, n_trees, P<float>(out),
                                      init, scale, (void*)stream);
+        },
+        nogil);
+  m.def("gpu_predict_forest_tp",
+        [](uintptr_t X, int64_t N, int F, uintptr_t nodes,
+           uintptr_t roots, uintptr_t masks, uintptr_t obl_ranges,
+           uintptr_t obl_attr, uintptr_t obl_w, uintptr_t na_right,
+           int has_cats, int tree_start, int tree_step, int n_trees,
+           int n_chunks, uintptr_t partial, uintptr_t out, float init,
+           float scale, uintptr_t stream) {
+          gpu_predict_forest_tp(
+              P<float>(X), N, F, P<int32_t>(nodes), P<int32_t>(roots),
+              P<unsigned long long>(masks), P<int32_t>(obl_ranges),
+              P<int32_t>(obl_attr), P<float>(obl_w),
+              P<uint8_t>(na_right), has_cats, tree_start, tree_step,
+              n_trees, n_chunks, P<float>(partial), P<float>(out),
+              init, scale, (void*)stream);
         },
         nogil);
   m.def("gpu_predict_forest_binned8_tp",

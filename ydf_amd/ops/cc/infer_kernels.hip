@@ -45,7 +45,8 @@ __global__ void predict_forest_lds_kernel(
     const float* __restrict__ obl_w,
     const uint8_t* __restrict__ na_right, int has_cats,
     int tree_start, int tree_step, int n_trees, float* __restrict__ out,
-    float init, float scale) {
+    float init, float scale, int trees_per_chunk,
+    float* __restrict__ partial) {
   extern __shared__ float xs[];  // [F][kTile]
   const int64_t base = (int64_t)blockIdx.x * kTile;
   const int tid = threadIdx.x;
@@ -57,13 +58,18 @@ __global__ void predict_forest_lds_kernel(
   }
   __syncthreads();
   if (tid >= n_here) return;
+  // tree-chunk grid for small batches: blockIdx.y covers trees
+  // [t_lo, t_hi) and writes a leaf-sum partial; single-chunk launches
+  // (partial == nullptr) cover the whole forest and write `out`.
+  const int t_lo = blockIdx.y * trees_per_chunk;
+  const int t_hi = min(t_lo + trees_per_chunk, n_trees);
   float acc = init;
   if (!has_cats && na_right == nullptr) {
     // pure-numerical fast path: 4 trees walk in parallel per thread (a
     // single walk is a chain of DEPENDENT L2 gathers), one 16-B packed
     // node load per step.
-    int tt = 0;
-    for (; tt + 4 <= n_trees; tt += 4) {
+    int tt = t_lo;
+    for (; tt + 4 <= t_hi; tt += 4) {
       PackedNode nd[4];
 #pragma unroll
       for (int u = 0; u < 4; ++u)
@@ -84,14 +90,14 @@ __global__ void predict_forest_lds_kernel(
 #pragma unroll
       for (int u = 0; u < 4; ++u) acc += nd[u].thr;
     }
-    for (; tt < n_trees; ++tt) {
+    for (; tt < t_hi; ++tt) {
       PackedNode nd = nodes[roots[tree_start + (int64_t)tt * tree_step]];
       while (nd.feat >= 0)
         nd = nodes[nd.left + (xs[nd.feat * kTile + tid] > nd.thr ? 1 : 0)];
       acc += nd.thr;
     }
   } else {
-    for (int tt = 0; tt < n_trees; ++tt) {
+    for (int tt = t_lo; tt < t_hi; ++tt) {
       int ni = roots[tree_start + (int64_t)tt * tree_step];
       PackedNode nd = nodes[ni];
       while (nd.feat >= 0) {
@@ -124,7 +130,10 @@ __global__ void predict_forest_lds_kernel(
       acc += nd.thr;
     }
   }
-  out[base + tid] = init + (acc - init) * scale;
+  if (partial != nullptr)
+    partial[(int64_t)blockIdx.y * N + base + tid] = acc - init;
+  else
+    out[base + tid] = init + (acc - init) * scale;
 }
 
 // Fallback without the LDS tile (feature count too large to stage).
@@ -688,7 +697,8 @@ void gpu_predict_forest(const float* X, int64_t N, int F,
     hipLaunchKernelGGL(predict_forest_lds_kernel, dim3(grid), dim3(kTile), lds,
                        (hipStream_t)stream, X, N, F, nodes, roots, masks,
                        obl_ranges, obl_attr, obl_w, na_right, has_cats,
-                       tree_start, tree_step, n_trees, out, init, scale);
+                       tree_start, tree_step, n_trees, out, init, scale,
+                       n_trees, nullptr);
   } else {
     int grid = (int)((N + kTile - 1) / kTile);
     if (grid > 4096) grid = 4096;
@@ -698,6 +708,34 @@ void gpu_predict_forest(const float* X, int64_t N, int F,
                        obl_ranges, obl_attr, obl_w, na_right, has_cats,
                        tree_start, tree_step, n_trees, out, init, scale);
   }
+}
+
+void gpu_predict_forest_tp(const float* X, int64_t N, int F,
+                           const int32_t* packed_nodes,
+                           const int32_t* roots,
+                           const unsigned long long* masks,
+                           const int32_t* obl_ranges,
+                           const int32_t* obl_attr, const float* obl_w,
+                           const uint8_t* na_right, int has_cats,
+                           int tree_start, int tree_step, int n_trees,
+                           int n_chunks, float* partial, float* out,
+                           float init, float scale, void* stream) {
+  const PackedNode* nodes =
+      reinterpret_cast<const PackedNode*>(packed_nodes);
+  const size_t lds = (size_t)F * kTile * sizeof(float);
+  const int row_tiles = (int)((N + kTile - 1) / kTile);
+  const int tpc = (n_trees + n_chunks - 1) / n_chunks;
+  const int chunks = (n_trees + tpc - 1) / tpc;
+  hipLaunchKernelGGL(predict_forest_lds_kernel,
+                     dim3(row_tiles, chunks), dim3(kTile), lds,
+                     (hipStream_t)stream, X, N, F, nodes, roots, masks,
+                     obl_ranges, obl_attr, obl_w, na_right, has_cats,
+                     tree_start, tree_step, n_trees, out, 0.f, scale,
+                     tpc, partial);
+  const int rg = (int)((N + 255) / 256);
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(rg), dim3(256), 0,
+                     (hipStream_t)stream, partial, chunks, N, out, init,
+                     scale);
 }
 
 void gpu_sigmoid(const float* in, float* out, int64_t N, void* stream) {
