@@ -358,3 +358,21 @@ def test_training_log_entry_attr_access(binary_data):
     assert logs and logs[-1].iteration == logs[-1]["iteration"]
     ev = logs[-1].evaluation
     assert ev.loss == logs[-1]["valid_loss"]
+
+
+def test_leaf_indices_reconstruct_predictions(binary_data):
+    """Shapley-style consistency: summing leaf values at
+    leaf_indices(+init, scale, activation) must reproduce predict()
+    exactly — pins the walker against the serving kernels across
+    condition types (numerical + categorical masks)."""
+    data = dict(binary_data)
+    rng = np.random.RandomState(3)
+    data["cat"] = rng.choice(["u", "v", "w", "x"], len(data["x1"]))
+    m = ydf.GradientBoostedTreesLearner(
+        label="label", num_trees=25, validation_ratio=0.0).train(data)
+    li = m.leaf_indices(data)
+    margins = (m.init_predictions[0]
+               + m.forest.thr[li].astype(np.float64).sum(axis=1)
+               * m._leaf_scale())
+    prob = 1.0 / (1.0 + np.exp(-margins))
+    np.testing.assert_allclose(prob, m.predict(data), atol=2e-6)
