@@ -213,6 +213,15 @@ def main():
         print(f"# train logloss={float(s2[0] / s2[2]):.4f} "
               f"accuracy={float(s2[1] / s2[2]):.4f} after "
               f"{args.warmup + args.steps} trees", file=sys.stderr)
+    # orderly communicator teardown: without this, a rank can abort in
+    # a gloo/RCCL destructor at interpreter exit ("terminate called
+    # without an active exception") AFTER the results printed, failing
+    # the whole torchrun job
+    import torch.distributed as td
+
+    if td.is_initialized():
+        td.barrier()
+        td.destroy_process_group()
 
 
 if __name__ == "__main__":
