@@ -66,7 +66,8 @@ def f_msg(fn: int, payload: bytes) -> bytes:
 
 # --- data_spec.pb ----------------------------------------------------------
 _SEM_TO_TYPE = {Semantic.NUMERICAL: 1, Semantic.CATEGORICAL: 4,
-                Semantic.BOOLEAN: 7, Semantic.HASH: 10}
+                Semantic.BOOLEAN: 7, Semantic.HASH: 10,
+                Semantic.NUMERICAL_VECTOR_SEQUENCE: 11}
 
 
 def encode_data_spec(dataspec) -> bytes:
@@ -88,6 +89,10 @@ def encode_data_spec(dataspec) -> bytes:
             # (most-frequent value) with indicator counts.
             t = 1 if float(c.mean) >= 0.5 else 0
             body += f_msg(9, f_varint(1, t) + f_varint(2, 1 - t))
+        elif getattr(c, "vecseq_dim", 0):
+            # NumericalVectorSequenceSpec (data_spec.proto:237):
+            # vector_length=1; column field 13
+            body += f_msg(13, f_varint(1, int(c.vecseq_dim)))
         else:
             # NumericalSpec: mean=1 is a DOUBLE in the reference
             num = f_double(1, c.mean)
@@ -99,8 +104,28 @@ def encode_data_spec(dataspec) -> bytes:
 # --- nodes blob sequence ---------------------------------------------------
 def _encode_condition(feat: int, thr: float, mask, is_cat: bool,
                       is_bool: bool, oblique=None, cover: float = 0.0,
-                      na_right: bool = False, set_items=None) -> bytes:
-    if set_items is not None:
+                      na_right: bool = False, set_items=None,
+                      vecseq=None) -> bytes:
+    if vecseq is not None:
+        # NumericalVectorSequence (decision_tree.proto:133-161,
+        # Condition oneof field 8). Our virtual projection columns hold
+        #   dot : max_k <vec_k, anchor>     -> ProjectedMoreThan (>=)
+        #   dist: -min_k |vec_k - anchor|^2 -> CloserThan (d2 <=)
+        # strict-> semantics convert with one nextafter each way.
+        kind, anchor = vecseq
+        packed = struct.pack(f"<{len(anchor)}f",
+                             *[float(a) for a in anchor])
+        anchor_msg = f_msg(1, f_bytes(1, packed))  # Anchor.grounded
+        if kind == "dot":
+            # max_dot > thr  <=>  exists p >= nextafter(thr, +inf)
+            t = float(np.nextafter(np.float32(thr), np.float32("inf")))
+            inner = f_msg(8, f_msg(2, anchor_msg + f_float(2, t)))
+        else:
+            # -min_sq > thr  <=>  exists d2 <= nextafter(-thr, -inf)
+            t2 = float(np.nextafter(np.float32(-thr),
+                                    np.float32("-inf")))
+            inner = f_msg(8, f_msg(1, anchor_msg + f_float(2, t2)))
+    elif set_items is not None:
         # ContainsVector (decision_tree.proto Condition.contains=4):
         # elements=1 packed varints — categorical-SET conditions whose
         # vocab can exceed the 256-bit bitmap
@@ -166,9 +191,12 @@ def encode_forest_nodes(model, classifier_leaves: bool = False,
     recovered from the stored path-length contribution."""
     f = model.forest
     bool_feats = set()
+    vecseq_of_feat = {}
     for i, c in enumerate(model.dataspec.feature_columns):
         if c.semantic == Semantic.BOOLEAN:
             bool_feats.add(i)
+        if getattr(c, "vecseq_source", None):
+            vecseq_of_feat[i] = (c.vecseq_kind, c.vecseq_anchor)
     if feat_to_col is None:
         feat_to_col = list(range(len(model.dataspec.columns)))
 
@@ -212,7 +240,9 @@ def encode_forest_nodes(model, classifier_leaves: bool = False,
                 f.masks[ci] if ci >= 0 else None,
                 ci >= 0, fi in bool_feats and ci == -1 and si < 0,
                 oblique=obl, cover=float(f.cover[n]),
-                na_right=bool(f.na_right[n]), set_items=set_items)
+                na_right=bool(f.na_right[n]), set_items=set_items,
+                vecseq=vecseq_of_feat.get(fi)
+                if ci == -1 and si < 0 else None)
         records.append(body)
         if f.feat[n] >= 0:
             left = int(f.left[n])
@@ -292,23 +322,41 @@ def export_ydf_model(model, path: str) -> None:
             "models trained on expanded categorical-set token features "
             "use a virtual-column representation the reference data "
             "spec cannot express")
-    if any(getattr(c, "vecseq_source", None)
-           for c in model.dataspec.columns):
-        raise NotImplementedError(
-            "vector-sequence models: reference-format export of "
-            "NumericalVectorSequence conditions is not implemented yet")
     os.makedirs(path, exist_ok=True)
     # Column indexing: node conditions reference COLUMN indices in the
     # data spec, not dense feature indices (the label can sit anywhere —
     # ranking models put it first). input_features lists the feature
     # columns in the model's dense feature order; anomaly-detection
     # models have no label column (label_col_idx=-1).
+    # Vector-sequence models: the VIRTUAL projection columns are an
+    # internal representation — the exported spec carries ONE
+    # NUMERICAL_VECTOR_SEQUENCE column per source, and splits on
+    # virtual columns become NumericalVectorSequence conditions
+    # (ProjectedMoreThan / CloserThan) on that source column.
     cols = model.dataspec.columns
     label_name = model.dataspec.label
     has_label = label_name is not None
-    label_idx = next((i for i, c in enumerate(cols)
-                      if c.name == label_name), -1) if has_label else -1
-    feat_to_col = [i for i, c in enumerate(cols) if c.name != label_name]
+    exported = [c for c in cols
+                if not getattr(c, "vecseq_source", None)]
+    vec_sources = {}
+    for c in cols:
+        sname = getattr(c, "vecseq_source", None)
+        if sname and sname not in vec_sources:
+            vec_sources[sname] = int(len(c.vecseq_anchor))
+    if vec_sources:
+        from ydf_amd.dataset.dataspec import ColumnSpec as _CS
+        have = {c.name for c in exported}
+        for sname, dim in vec_sources.items():
+            if sname not in have:
+                exported.append(_CS(
+                    name=sname,
+                    semantic=Semantic.NUMERICAL_VECTOR_SEQUENCE,
+                    vecseq_dim=dim))
+    exp_idx = {c.name: i for i, c in enumerate(exported)}
+    label_idx = exp_idx.get(label_name, -1) if has_label else -1
+    feat_to_col = [
+        exp_idx[getattr(c, "vecseq_source", None) or c.name]
+        for c in model.dataspec.feature_columns]
     # AbstractModel: name=1, task=2, label_col_idx=3, input_features=5,
     # ranking_group_col_idx=6
     name = ("GRADIENT_BOOSTED_TREES" if is_gbt
@@ -316,7 +364,7 @@ def export_ydf_model(model, path: str) -> None:
     header = f_str(1, name)
     header += f_varint(2, _TASK[model.task()])
     header += f_varint(3, label_idx)
-    for i in feat_to_col:
+    for i in dict.fromkeys(feat_to_col):
         header += f_varint(5, i)
     rg = (model.metadata or {}).get("ranking_group")
     if rg:
@@ -326,8 +374,11 @@ def export_ydf_model(model, path: str) -> None:
                 break
     with open(os.path.join(path, "header.pb"), "wb") as fp:
         fp.write(header)
+    import types as _types
+
     with open(os.path.join(path, "data_spec.pb"), "wb") as fp:
-        fp.write(encode_data_spec(model.dataspec))
+        fp.write(encode_data_spec(
+            _types.SimpleNamespace(columns=exported)))
     if is_gbt:
         # GBT header (gradient_boosted_trees.proto:28-42):
         # num_node_shards=1, num_trees=2, loss=3, initial_predictions=4,

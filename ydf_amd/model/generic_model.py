@@ -372,6 +372,12 @@ class GenericModel:
             if spec.vecseq_source is not None:
                 has_vecseq = True
                 continue
+            if spec.semantic == Semantic.NUMERICAL_VECTOR_SEQUENCE:
+                # ragged source column (imported reference models keep
+                # it in the spec): its values feed the virtual
+                # projection columns, not a dense X row
+                X[i] = 0.0
+                continue
             src = spec.set_source or spec.name
             if src not in cols:
                 raise ValueError(f"missing input feature {src!r}")

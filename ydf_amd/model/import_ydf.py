@@ -125,7 +125,8 @@ def read_blob_sequence(path: str) -> Iterator[bytes]:
 # ---------------------------------------------------------------------------
 _COLTYPE = {1: Semantic.NUMERICAL, 4: Semantic.CATEGORICAL,
             7: Semantic.BOOLEAN, 9: Semantic.DISCRETIZED_NUMERICAL,
-            10: Semantic.HASH, 5: Semantic.CATEGORICAL_SET}
+            10: Semantic.HASH, 5: Semantic.CATEGORICAL_SET,
+            11: Semantic.NUMERICAL_VECTOR_SEQUENCE}
 
 
 def parse_data_spec(raw: bytes):
@@ -168,11 +169,15 @@ def parse_data_spec(raw: bytes):
             if 1 in dn:
                 disc_bounds[ci] = np.asarray(_packed_floats(dn[1][0]),
                                              dtype=np.float32)
+        vec_dim = 0
+        if sem == Semantic.NUMERICAL_VECTOR_SEQUENCE and 13 in col:
+            vs = _msg(col[13][0])
+            vec_dim = int(vs.get(1, [0])[0])
         columns.append(ColumnSpec(
             name=name,
             semantic=(Semantic.NUMERICAL
                       if sem == Semantic.DISCRETIZED_NUMERICAL else sem),
-            vocab=vocab, mean=mean))
+            vocab=vocab, mean=mean, vecseq_dim=vec_dim))
     return columns, disc_bounds
 
 
@@ -188,7 +193,7 @@ def parse_data_spec(raw: bytes):
 # ---------------------------------------------------------------------------
 class _NodeRec:
     __slots__ = ("is_leaf", "attr", "thr", "mask", "value", "cover",
-                 "na_value", "obl", "elements", "bitmap")
+                 "na_value", "obl", "elements", "bitmap", "vecseq")
 
 
 def parse_node(raw: bytes, disc_bounds, n_classes: int,
@@ -205,6 +210,7 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
     r.obl = None
     r.elements = None
     r.bitmap = None
+    r.vecseq = None
     if 5 in node:
         pass
     if 1 in node:  # classifier output
@@ -282,6 +288,32 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
             r.obl = (attrs, ws)
             # our kernels test strict >; dot >= t  <=>  dot > nextafter down
             r.thr = float(np.nextafter(np.float32(t), np.float32("-inf")))
+        elif 8 in inner:  # NumericalVectorSequence condition
+            vs = _msg(inner[8][0])
+            if 2 in vs:  # ProjectedMoreThan: exists <v|anchor> >= t
+                pm = _msg(vs[2][0])
+                anc = _msg(pm.get(1, [b""])[0])
+                anchor = np.asarray(
+                    _packed_floats(anc[1][0]) if anc.get(1) else [],
+                    dtype=np.float32)
+                t = _f32(pm.get(2, [0.0])[0]) if 2 in pm else 0.0
+                # internal virtual column holds max_dot with strict >
+                r.thr = float(np.nextafter(np.float32(t),
+                                           np.float32("-inf")))
+                r.vecseq = ("dot", anchor)
+            elif 1 in vs:  # CloserThan: exists |v-anchor|^2 <= t2
+                ct = _msg(vs[1][0])
+                anc = _msg(ct.get(1, [b""])[0])
+                anchor = np.asarray(
+                    _packed_floats(anc[1][0]) if anc.get(1) else [],
+                    dtype=np.float32)
+                t2 = _f32(ct.get(2, [0.0])[0]) if 2 in ct else 0.0
+                # internal column holds -min_sq with strict >
+                r.thr = float(np.nextafter(np.float32(-t2),
+                                           np.float32("-inf")))
+                r.vecseq = ("dist", anchor)
+            else:
+                raise ValueError("empty vector-sequence condition")
         elif 5 in inner:  # ContainsBitmap
             bm = _msg(inner[5][0]).get(1, [b""])[0]
             r.bitmap = bm
@@ -297,9 +329,12 @@ def parse_node(raw: bytes, disc_bounds, n_classes: int,
 # ---------------------------------------------------------------------------
 # Model assembly
 # ---------------------------------------------------------------------------
+_VECSEQ_BASE = 1 << 20  # sentinel feature ids for vecseq conditions
+
+
 def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
                 value_scale: float = 1.0, wta: bool = False,
-                set_feats=frozenset()):
+                set_feats=frozenset(), vecseq_reg=None):
     shards = sorted(p for p in os.listdir(model_dir)
                     if p.startswith(prefix + "nodes-"))
     records: List[bytes] = []
@@ -349,7 +384,15 @@ def _read_trees(model_dir: str, prefix: str, disc_bounds, n_classes,
             return
         feats[idx] = rec.attr
         na_right[idx] = 1 if rec.na_value else 0
-        if rec.mask is not None and (
+        if rec.vecseq is not None and vecseq_reg is not None:
+            kind, anchor = rec.vecseq
+            key = (int(rec.attr), kind, anchor.tobytes())
+            if key not in vecseq_reg:
+                vecseq_reg[key] = (len(vecseq_reg), int(rec.attr), kind,
+                                   anchor)
+            feats[idx] = _VECSEQ_BASE + vecseq_reg[key][0]
+            thrs[idx] = rec.thr
+        elif rec.mask is not None and (
                 rec.attr in set_feats
                 or (rec.elements is not None and len(rec.elements)
                     and max(rec.elements) >= 256)):
@@ -431,6 +474,27 @@ def load_ydf_model(path: str, file_prefix: str = ""):
         label=label_name)
     # remap attribute indices (original column idx -> dense feature idx)
     remap = {ci: i for i, ci in enumerate(input_features)}
+    # vecseq conditions materialize as virtual projection columns (one
+    # per distinct (source, kind, anchor)); _read_trees fills this
+    # registry with sentinel feature ids >= _VECSEQ_BASE
+    vecseq_reg = {}
+
+    def _finish(forest):
+        if vecseq_reg:
+            virtuals = []
+            base = len(feat_cols)
+            for key, (vid, attr, kind, anchor) in sorted(
+                    vecseq_reg.items(), key=lambda kv: kv[1][0]):
+                src_name = columns[attr].name
+                virtuals.append(ColumnSpec(
+                    name=f"{src_name}.{kind}.{vid}",
+                    semantic=Semantic.NUMERICAL,
+                    vecseq_source=src_name, vecseq_kind=kind,
+                    vecseq_anchor=np.asarray(anchor, np.float32).copy()))
+                remap[_VECSEQ_BASE + vid] = base + vid
+            dataspec.columns = feat_cols + virtuals + (
+                [columns[label_idx]] if has_label else [])
+        return _remap_forest(forest, remap)
 
     set_feats = frozenset(
         i for i, c in enumerate(columns)
@@ -465,7 +529,7 @@ def load_ydf_model(path: str, file_prefix: str = ""):
         loss = {1: 1, 2: 2, 3: 3, 4: 9, 5: 12, 6: 11, 7: 7, 8: 8,
                 9: 9, 10: 13}.get(int(loss_ref), int(loss_ref))
         forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
-                             set_feats=set_feats)
+                             set_feats=set_feats, vecseq_reg=vecseq_reg)
         ntpi = gh.get(5, [1])[0]
         activation = "identity"
         if task == Task.CLASSIFICATION:
@@ -476,7 +540,7 @@ def load_ydf_model(path: str, file_prefix: str = ""):
             loss = (3 if ntpi > 1 else 1) \
                 if task == Task.CLASSIFICATION else 2
         model = GradientBoostedTreesModel(
-            forest=_remap_forest(forest, remap), dataspec=dataspec,
+            forest=_finish(forest), dataspec=dataspec,
             task=task, label_classes=classes, init_predictions=init_preds,
             num_trees_per_iter=ntpi, activation=activation,
             metadata={"imported_from": "yggdrasil-decision-forests",
@@ -490,7 +554,7 @@ def load_ydf_model(path: str, file_prefix: str = ""):
         wta = bool(rh.get(3, [1])[0])
         forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
                              wta=wta and task == Task.CLASSIFICATION,
-                             set_feats=set_feats)
+                             set_feats=set_feats, vecseq_reg=vecseq_reg)
         meta = {"imported_from": "yggdrasil-decision-forests",
                 "winner_take_all": wta}
         if task in (Task.CATEGORICAL_UPLIFT, Task.NUMERICAL_UPLIFT):
@@ -501,7 +565,7 @@ def load_ydf_model(path: str, file_prefix: str = ""):
                 tv = columns[tcol].vocab
                 meta["treatment_vocab"] = list(tv[1:]) if tv else None
         model = RandomForestModel(
-            forest=_remap_forest(forest, remap), dataspec=dataspec,
+            forest=_finish(forest), dataspec=dataspec,
             task=task, label_classes=classes,
             init_predictions=[0.0],
             num_trees_per_iter=1, activation="identity",
@@ -515,9 +579,10 @@ def load_ydf_model(path: str, file_prefix: str = ""):
 
         with open(if_hdr_path, "rb") as f:
             ih = _msg(f.read())
-        forest = _read_trees(path, file_prefix, disc_bounds, n_classes)
+        forest = _read_trees(path, file_prefix, disc_bounds, n_classes,
+                             vecseq_reg=vecseq_reg)
         return IsolationForestModel(
-            forest=_remap_forest(forest, remap), dataspec=dataspec,
+            forest=_finish(forest), dataspec=dataspec,
             task=Task.ANOMALY_DETECTION, init_predictions=[0.0],
             num_trees_per_iter=1, activation="identity",
             num_examples_per_tree=int(ih.get(4, [256])[0]),
